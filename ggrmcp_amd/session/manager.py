@@ -1,0 +1,213 @@
+"""Session manager.
+
+Re-design of the reference's ``pkg/session/manager.go``: TTL-cached sessions
+(manager.go:53-66 — TTL 30 min, sweep 5 min, max 10k), per-session context with
+headers snapshot, timestamps, call counter, fixed-window rate limit and a
+blocked flag (manager.go:16-34), crypto-random 16-byte hex session ids with a
+timestamp fallback (manager.go:258-265), block/unblock (manager.go:146-175)
+and stats (manager.go:211-249).
+
+MI355X addition: every session carries a stable ``shard(n)`` — the GPU engine
+index its requests are routed to under data-parallel sharding.  The hash must
+be stable across processes (ranks compute it independently), so it uses
+blake2b of the session id, not Python's salted ``hash()``.
+"""
+
+from __future__ import annotations
+
+import hashlib
+import secrets
+import threading
+import time
+from collections import OrderedDict
+from dataclasses import dataclass, field
+from typing import Dict, List, Mapping, Optional
+
+DEFAULT_TTL_S = 30 * 60.0
+DEFAULT_CLEANUP_INTERVAL_S = 5 * 60.0
+DEFAULT_MAX_SESSIONS = 10_000
+DEFAULT_RATE_LIMIT_PER_MIN = 100
+DEFAULT_RATE_LIMIT_BURST = 20
+
+
+@dataclass
+class SessionContext:
+    """Per-session state (reference manager.go:16-34)."""
+
+    id: str
+    headers: Dict[str, str] = field(default_factory=dict)
+    created_at: float = field(default_factory=time.time)
+    last_accessed: float = field(default_factory=time.time)
+    call_count: int = 0
+    is_blocked: bool = False
+    # fixed-window rate limit state (manager.go:178-208)
+    window_start: float = 0.0
+    window_count: int = 0
+    _lock: threading.Lock = field(default_factory=threading.Lock, repr=False)
+
+    def increment_call_count(self) -> int:
+        with self._lock:
+            self.call_count += 1
+            return self.call_count
+
+    def update_last_accessed(self) -> None:
+        self.last_accessed = time.time()
+
+    def shard(self, n_shards: int) -> int:
+        """Stable session -> GPU-engine shard index (MI355X DP sharding)."""
+        if n_shards <= 1:
+            return 0
+        digest = hashlib.blake2b(self.id.encode(), digest_size=8).digest()
+        return int.from_bytes(digest, "little") % n_shards
+
+    def snapshot(self) -> Dict[str, object]:
+        """Per-session info for /metrics (reference manager.go:228-249)."""
+        return {
+            "id": self.id,
+            "createdAt": self.created_at,
+            "lastAccessed": self.last_accessed,
+            "callCount": self.call_count,
+            "isBlocked": self.is_blocked,
+            "headerCount": len(self.headers),
+        }
+
+
+def generate_session_id() -> str:
+    """crypto-rand 16-byte hex id with timestamp fallback (manager.go:258-265)."""
+    try:
+        return secrets.token_hex(16)
+    except Exception:  # pragma: no cover - entropy exhaustion is theoretical
+        return f"ts-{time.time_ns():x}"
+
+
+class SessionManager:
+    """TTL session cache (reference pkg/session/manager.go Manager)."""
+
+    def __init__(
+        self,
+        ttl_s: float = DEFAULT_TTL_S,
+        cleanup_interval_s: float = DEFAULT_CLEANUP_INTERVAL_S,
+        max_sessions: int = DEFAULT_MAX_SESSIONS,
+        rate_limit_per_min: int = DEFAULT_RATE_LIMIT_PER_MIN,
+        rate_limit_burst: int = DEFAULT_RATE_LIMIT_BURST,
+    ) -> None:
+        self.ttl_s = ttl_s
+        self.cleanup_interval_s = cleanup_interval_s
+        self.max_sessions = max_sessions
+        self.rate_limit_per_min = rate_limit_per_min
+        self.rate_limit_burst = rate_limit_burst
+        self._sessions: "OrderedDict[str, SessionContext]" = OrderedDict()
+        self._lock = threading.Lock()
+        self._last_sweep = time.monotonic()
+        self._expiry: Dict[str, float] = {}
+
+    # -- core (manager.go:69-143) -------------------------------------------
+
+    def get_or_create(
+        self, session_id: Optional[str], headers: Optional[Mapping[str, str]] = None
+    ) -> SessionContext:
+        """Reference GetOrCreateSession (manager.go:69-84)."""
+        now = time.monotonic()
+        with self._lock:
+            self._maybe_sweep(now)
+            if session_id:
+                ctx = self._get_locked(session_id, now)
+                if ctx is not None:
+                    ctx.update_last_accessed()
+                    if headers:
+                        ctx.headers = dict(headers)
+                    return ctx
+            ctx = SessionContext(id=session_id or generate_session_id())
+            if headers:
+                ctx.headers = dict(headers)
+            if len(self._sessions) >= self.max_sessions:
+                # evict oldest (go-cache rejects at max; we evict LRU instead,
+                # which keeps the gateway serving under session floods)
+                old_id, _ = self._sessions.popitem(last=False)
+                self._expiry.pop(old_id, None)
+            self._sessions[ctx.id] = ctx
+            self._expiry[ctx.id] = now + self.ttl_s
+            return ctx
+
+    def get(self, session_id: str) -> Optional[SessionContext]:
+        now = time.monotonic()
+        with self._lock:
+            return self._get_locked(session_id, now)
+
+    def _get_locked(self, session_id: str, now: float) -> Optional[SessionContext]:
+        exp = self._expiry.get(session_id)
+        if exp is None:
+            return None
+        if now >= exp:
+            self._sessions.pop(session_id, None)
+            self._expiry.pop(session_id, None)
+            return None
+        self._expiry[session_id] = now + self.ttl_s  # touch extends TTL
+        self._sessions.move_to_end(session_id)
+        return self._sessions.get(session_id)
+
+    def remove(self, session_id: str) -> bool:
+        with self._lock:
+            self._expiry.pop(session_id, None)
+            return self._sessions.pop(session_id, None) is not None
+
+    def _maybe_sweep(self, now: float) -> None:
+        if now - self._last_sweep < self.cleanup_interval_s:
+            return
+        self._last_sweep = now
+        dead = [sid for sid, exp in self._expiry.items() if now >= exp]
+        for sid in dead:
+            self._sessions.pop(sid, None)
+            self._expiry.pop(sid, None)
+
+    # -- rate limiting (manager.go:178-208, fixed window) --------------------
+
+    def check_rate_limit(self, ctx: SessionContext) -> bool:
+        """True if the call is allowed. Fixed 60s window of
+        rate_limit_per_min + burst (reference manager.go:178-208)."""
+        now = time.time()
+        with ctx._lock:
+            if now - ctx.window_start >= 60.0:
+                ctx.window_start = now
+                ctx.window_count = 0
+            limit = self.rate_limit_per_min + self.rate_limit_burst
+            if ctx.window_count >= limit:
+                return False
+            ctx.window_count += 1
+            return True
+
+    # -- block/unblock (manager.go:146-175) ----------------------------------
+
+    def block(self, session_id: str) -> bool:
+        ctx = self.get(session_id)
+        if ctx is None:
+            return False
+        ctx.is_blocked = True
+        return True
+
+    def unblock(self, session_id: str) -> bool:
+        ctx = self.get(session_id)
+        if ctx is None:
+            return False
+        ctx.is_blocked = False
+        return True
+
+    # -- stats (manager.go:211-249) ------------------------------------------
+
+    def stats(self) -> Dict[str, object]:
+        with self._lock:
+            sessions = list(self._sessions.values())
+        return {
+            "activeSessions": len(sessions),
+            "maxSessions": self.max_sessions,
+            "totalCalls": sum(s.call_count for s in sessions),
+            "blockedSessions": sum(1 for s in sessions if s.is_blocked),
+        }
+
+    def session_info(self, session_id: str) -> Optional[Dict[str, object]]:
+        ctx = self.get(session_id)
+        return None if ctx is None else ctx.snapshot()
+
+    def active_ids(self) -> List[str]:
+        with self._lock:
+            return list(self._sessions.keys())

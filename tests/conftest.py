@@ -1,0 +1,31 @@
+import sys
+from pathlib import Path
+
+import pytest
+
+# Make the in-tree package importable without installation.
+REPO_ROOT = Path(__file__).resolve().parent.parent
+if str(REPO_ROOT) not in sys.path:
+    sys.path.insert(0, str(REPO_ROOT))
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: test requires an MI355X GPU (run with -m gpu on a GPU box)"
+    )
+
+
+def pytest_collection_modifyitems(config, items):
+    """Skip gpu-marked tests automatically when no GPU is present."""
+    try:
+        import torch
+
+        has_gpu = torch.cuda.is_available()
+    except Exception:
+        has_gpu = False
+    if has_gpu:
+        return
+    skip = pytest.mark.skip(reason="no GPU available")
+    for item in items:
+        if "gpu" in item.keywords:
+            item.add_marker(skip)
