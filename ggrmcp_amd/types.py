@@ -46,11 +46,14 @@ class MethodInfo:
     # MI355X additions: which backend (centralized-gateway mode) owns the
     # method, used by the invoker; reference has exactly one backend.
     backend_index: int = 0
+    # When the compat shim truncates service_name for tool naming (reference
+    # loader.go:219-235), the full name still needed for the gRPC wire path.
+    full_service_name: str = ""
 
     @property
     def full_method_path(self) -> str:
         """gRPC wire path ``/package.Service/Method``."""
-        return f"/{self.service_name}/{self.method_name}"
+        return f"/{self.full_service_name or self.service_name}/{self.method_name}"
 
     @property
     def is_streaming(self) -> bool:
