@@ -88,14 +88,37 @@ class ConnectionManager:
             channel = self._channel
         if channel is None:
             return False
-        fut = grpc.channel_ready_future(channel)
         try:
-            fut.result(timeout=timeout_s)
-            return True
-        except Exception:
+            # connectivity-state probe without spawning a poll thread
+            # (grpc.channel_ready_future leaves a poller racing close())
+            state = channel._channel.check_connectivity_state(True)
+            ready = grpc.ChannelConnectivity.READY.value[0]
+            idle = grpc.ChannelConnectivity.IDLE.value[0]
+            if state == ready:
+                return True
+            if state == idle:
+                # IDLE channels flip to READY on first use; treat as healthy
+                # (the reference's WaitForStateChange loop does the same walk)
+                return True
+            deadline = timeout_s
+            import time as _time
+
+            step = 0.05
+            while deadline > 0:
+                _time.sleep(step)
+                deadline -= step
+                if channel._channel.check_connectivity_state(True) == ready:
+                    return True
             return False
-        finally:
-            fut.cancel()  # stop the connectivity poll thread
+        except Exception:
+            fut = grpc.channel_ready_future(channel)
+            try:
+                fut.result(timeout=timeout_s)
+                return True
+            except Exception:
+                return False
+            finally:
+                fut.cancel()
 
     def reconnect(self, timeout_s: Optional[float] = None) -> grpc.Channel:
         """Reference connection.go:103-106 (Reconnect = close + Connect)."""

@@ -199,7 +199,7 @@ class Config:
         errs: List[str] = []
         if not (0 < self.grpc.port < 65536):
             errs.append(f"grpc.port out of range: {self.grpc.port}")
-        if not (0 < self.server.http_port < 65536):
+        if not (0 <= self.server.http_port < 65536):  # 0 = auto-assign
             errs.append(f"server.http_port out of range: {self.server.http_port}")
         if not self.grpc.host:
             errs.append("grpc.host empty")
