@@ -1,0 +1,247 @@
+#!/usr/bin/env python3
+"""Flagship serving benchmark — MCP tool_call -> gRPC unary req/sec.
+
+Measures the BASELINE.json headline metric on the MI355X-native gateway: a
+batch of JSON-RPC tools/call requests per step goes through the full hot
+path — GPU envelope parse + validate + JSON->protobuf (k_json2pb), real gRPC
+unary invocations against a local backend process, GPU protobuf->JSON +
+response-envelope assembly (k_pb2json).  The reference (aalobaidi/ggRMCP)
+publishes no numbers, so vs_baseline is null.
+
+Contract (driver):
+  python bench.py --gpus N --steps K --warmup W
+launched for N>1 as one rank per GPU via torch.distributed.run; rank 0
+prints ONE JSON line; value = WHOLE-JOB req/s over all ranks (weak scaling:
+each rank runs its own sessions/backend shard).
+
+Configs (BASELINE.json):
+  --config hello1k   hello-service SayHello, 1 KB JSON payloads (default)
+  --config wide64    synthetic 64-field proto, 64 KB payloads, validation on
+  --config cpu       reference-equivalent CPU-only plumbing path (config 1)
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import random
+import statistics
+import subprocess
+import sys
+import tempfile
+import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent))
+
+from ggrmcp_amd.backend.discovery import ServiceDiscoverer  # noqa: E402
+from ggrmcp_amd.config import Config  # noqa: E402
+from ggrmcp_amd.utils.synthetic import (  # noqa: E402
+    hello_payload,
+    jsonrpc_body,
+    wide_payload,
+)
+
+
+def parse_args():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--config", default="hello1k", choices=["hello1k", "wide64", "cpu"])
+    ap.add_argument("--batch", type=int, default=1024, help="requests per step (= concurrent sessions)")
+    ap.add_argument("--payload-bytes", type=int, default=0)
+    ap.add_argument("--invoke-workers", type=int, default=64)
+    ap.add_argument("--backend-workers", type=int, default=32)
+    return ap.parse_args()
+
+
+def start_backend(rank: int, workers: int):
+    """Backend in its own process over a unix socket (no shared GIL)."""
+    sock = os.path.join(tempfile.gettempdir(), f"ggrmcp_bench_{os.getpid()}_{rank}.sock")
+    if os.path.exists(sock):
+        os.unlink(sock)
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "examples.bench_backend", "--uds", sock,
+         "--workers", str(workers)],
+        stdout=subprocess.PIPE, stderr=subprocess.DEVNULL,
+        cwd=str(Path(__file__).resolve().parent), text=True,
+    )
+    line = proc.stdout.readline()
+    if not line.startswith("READY"):
+        raise RuntimeError(f"backend failed to start: {line!r}")
+    return proc, sock
+
+
+def make_bodies(cfg_name: str, batch: int, payload_bytes: int, seed: int):
+    rng = random.Random(seed)
+    bodies = []
+    if cfg_name in ("hello1k", "cpu"):
+        tool = "hello_helloservice_sayhello"
+        size = payload_bytes or 1024
+        for i in range(batch):
+            bodies.append(jsonrpc_body(tool, hello_payload(rng, size), i + 1))
+    else:
+        tool = "bench_echoservice_echo"
+        size = payload_bytes or 64 * 1024
+        for i in range(batch):
+            bodies.append(jsonrpc_body(tool, wide_payload(rng, target_bytes=size), i + 1))
+    return bodies
+
+
+def main() -> None:
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    import torch
+
+    use_gpu = args.config != "cpu" and torch.cuda.is_available()
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        backend = "nccl" if use_gpu else "gloo"
+        if use_gpu:
+            torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend=backend)
+
+    backend_proc, sock = start_backend(rank, args.backend_workers)
+    try:
+        cfg = Config.default()
+        cfg.grpc.uds = sock
+        cfg.gpu.enabled = use_gpu
+        discoverer = ServiceDiscoverer(cfg)
+        discoverer.connect(timeout_s=30)
+        discoverer.discover()
+
+        bodies = make_bodies(args.config, args.batch, args.payload_bytes, seed=1234 + rank)
+
+        if use_gpu:
+            from ggrmcp_amd.engine.batch import GpuPipeline
+
+            pipeline = GpuPipeline(discoverer, cfg, device=local_rank,
+                                   invoke_workers=args.invoke_workers)
+
+            def step():
+                out = pipeline.process_batch(bodies, timeout_s=30.0)
+                return out
+        else:
+            # reference-equivalent CPU plumbing (BASELINE config 1)
+            from concurrent.futures import ThreadPoolExecutor
+
+            import ggrmcp_amd.mcp.types as mcp_types  # noqa: F401
+
+            pool = ThreadPoolExecutor(max_workers=args.invoke_workers)
+            parsed = []
+            for b in bodies:
+                d = json.loads(b)
+                parsed.append(
+                    (d["params"]["name"], json.dumps(d["params"].get("arguments", {})), d["id"])
+                )
+
+            def one(item):
+                tool, args_json, rid = item
+                out = discoverer.invoke_method_by_tool(tool, args_json, None, 30.0)
+                resp = {
+                    "jsonrpc": "2.0",
+                    "id": rid,
+                    "result": {
+                        "content": [{"type": "text", "text": out}],
+                        "isError": False,
+                    },
+                }
+                return json.dumps(resp).encode()
+
+            def step():
+                return list(pool.map(one, parsed, chunksize=8))
+
+        def sync():
+            if use_gpu:
+                torch.cuda.synchronize()
+            if dist is not None:
+                dist.barrier()
+
+        # warmup
+        for _ in range(args.warmup):
+            out = step()
+        # sanity: responses are well-formed
+        sample = json.loads(out[0])
+        assert sample.get("result", {}).get("isError") is False, sample
+
+        sync()
+        step_times = []
+        t0 = time.perf_counter()
+        for _ in range(args.steps):
+            s0 = time.perf_counter()
+            step()
+            step_times.append(time.perf_counter() - s0)
+        if use_gpu:
+            torch.cuda.synchronize()
+        t1 = time.perf_counter()
+        sync()
+
+        elapsed = t1 - t0
+        # max elapsed across ranks
+        if dist is not None:
+            t = torch.tensor([elapsed], dtype=torch.float64,
+                             device="cuda" if use_gpu and dist.get_backend() == "nccl" else "cpu")
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            elapsed = float(t.item())
+
+        total_requests = args.batch * args.steps * world
+        reqs_per_s = total_requests / elapsed
+        ms_per_step = elapsed / args.steps * 1e3
+        p50_ms = statistics.median(step_times) * 1e3
+
+        if rank == 0:
+            payload_size = args.payload_bytes or (1024 if args.config in ("hello1k", "cpu") else 65536)
+            result = {
+                "metric": "MCP tool_call→gRPC unary req/sec (whole node)",
+                "value": round(reqs_per_s, 1),
+                "unit": "req/s",
+                "n_gpus": world,
+                "steps": args.steps,
+                "warmup": args.warmup,
+                "ms_per_step": round(ms_per_step, 3),
+                "higher_is_better": True,
+                "scaling": "weak",
+                "vs_baseline": None,
+                "dtype": "uint8",
+                "data": "synthetic",
+                "config": {
+                    "model": (
+                        "hello-service SayHello"
+                        if args.config in ("hello1k", "cpu")
+                        else "bench.Wide64 64-field nested proto"
+                    ),
+                    "global_batch": args.batch * world,
+                    "seq_len": payload_size,
+                    "parallelism": f"dp{world}",
+                    "mode": "gpu" if use_gpu else "cpu-reference",
+                    "sessions": args.batch,
+                    "payload_bytes": payload_size,
+                    "p50_rtt_ms": round(p50_ms, 3),
+                    "backend": "local grpc (uds, separate process)",
+                },
+            }
+            if use_gpu:
+                result["config"]["engine_stats"] = pipeline.engine.stats.snapshot()
+            print(json.dumps(result), flush=True)
+    finally:
+        backend_proc.terminate()
+        try:
+            backend_proc.wait(timeout=5)
+        except Exception:
+            backend_proc.kill()
+        if os.path.exists(sock):
+            os.unlink(sock)
+        if dist is not None:
+            dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

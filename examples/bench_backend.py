@@ -1,0 +1,105 @@
+"""Benchmark backend: the demo services plus the synthetic bench.EchoService
+(BASELINE config 3's 64-field proto).  Runs as a separate process so gateway
+and backend don't share a GIL.
+
+    python -m examples.bench_backend --uds /tmp/bench.sock
+"""
+
+from __future__ import annotations
+
+import argparse
+import sys
+import time
+from pathlib import Path
+
+import grpc
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+from examples.hello_service import build_server  # noqa: E402
+from examples.protos import ALL_FDPS, SERVICE_NAMES  # noqa: E402
+from ggrmcp_amd.backend.reflection_server import ReflectionServicer  # noqa: E402
+from ggrmcp_amd.utils.protobuild import message_class, pool_for  # noqa: E402
+from ggrmcp_amd.utils.synthetic import synthetic_fdp  # noqa: E402
+
+BENCH_FDP = synthetic_fdp()
+ALL = ALL_FDPS + [BENCH_FDP]
+NAMES = SERVICE_NAMES + ["bench.EchoService"]
+
+
+def build_bench_server(max_workers: int = 32) -> grpc.Server:
+    from concurrent import futures
+
+    pool = pool_for(ALL)
+    Wide64 = message_class(pool, "bench.Wide64")
+
+    def _echo(request, context):
+        return request
+
+    server = grpc.server(futures.ThreadPoolExecutor(max_workers=max_workers))
+    # demo services (rebuild handlers against this server)
+    import examples.hello_service as hs
+
+    server.add_generic_rpc_handlers(
+        (
+            grpc.method_handlers_generic_handler(
+                "hello.HelloService",
+                {"SayHello": hs._unary(hs._say_hello, hs.HelloRequest)},
+            ),
+            grpc.method_handlers_generic_handler(
+                "complex.UserService",
+                {"GetUser": hs._unary(hs._get_user, hs.GetUserRequest)},
+            ),
+            grpc.method_handlers_generic_handler(
+                "complex.DocumentService",
+                {"PutDocument": hs._unary(hs._put_document, hs.Document)},
+            ),
+            grpc.method_handlers_generic_handler(
+                "complex.NodeService",
+                {
+                    "Echo": hs._unary(hs._echo_node, hs.NodeRequest),
+                    "StreamNodes": hs._server_stream(hs._stream_nodes, hs.NodeRequest),
+                },
+            ),
+            grpc.method_handlers_generic_handler(
+                "bench.EchoService", {"Echo": hs._unary(_echo, Wide64)}
+            ),
+        )
+    )
+    servicer = ReflectionServicer(NAMES, ALL)
+    server.add_generic_rpc_handlers(tuple(servicer.generic_handlers()))
+    return server
+
+
+def serve(target: str = "127.0.0.1:0", max_workers: int = 32):
+    server = build_bench_server(max_workers)
+    if target.startswith("unix:"):
+        server.add_insecure_port(target)
+        bound = target
+    else:
+        host, _, port = target.rpartition(":")
+        actual = server.add_insecure_port(target)
+        bound = f"{host}:{actual}"
+    server.start()
+    return server, bound
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--host", default="127.0.0.1")
+    ap.add_argument("--port", type=int, default=0)
+    ap.add_argument("--uds", default="")
+    ap.add_argument("--workers", type=int, default=32)
+    args = ap.parse_args()
+    target = f"unix:{args.uds}" if args.uds else f"{args.host}:{args.port}"
+    server, bound = serve(target, args.workers)
+    print(f"READY {bound}", flush=True)
+    try:
+        while True:
+            time.sleep(3600)
+    except KeyboardInterrupt:
+        server.stop(grace=1)
+
+
+if __name__ == "__main__":
+    main()

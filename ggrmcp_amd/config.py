@@ -37,9 +37,12 @@ class ConnectionConfig:
     max_recv_msg_bytes: int = 4 * 1024 * 1024
     request_timeout_s: float = 30.0  # config.go:235
     use_tls: bool = False
+    uds: str = ""  # unix-domain socket path (overrides host:port)
 
     @property
     def target(self) -> str:
+        if self.uds:
+            return f"unix:{self.uds}"
         return f"{self.host}:{self.port}"
 
 

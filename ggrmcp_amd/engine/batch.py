@@ -1,0 +1,526 @@
+"""Batch pipeline driver: GPU engine + gRPC I/O + host fallback.
+
+Two integration points replace the reference's per-request CPU hot path:
+
+* ``GpuPipeline.process_batch(bodies)`` — the full GPU path used by the raw
+  ingestion front end and bench.py: a batch of JSON-RPC request bodies goes
+  through k_json2pb (envelope parse + validate + tool resolve + JSON->pb),
+  the host fans the wire bytes out to the gRPC backend(s) concurrently
+  (conn.Invoke stays host-side, reflection.go:367-373), and k_pb2json turns
+  the response wire bytes into complete JSON-RPC response envelopes.
+
+* ``BatchEngineInvoker`` — the async invoker seam of the HTTP handler
+  (server/handler.py): concurrent tools/call coroutines are collected for up
+  to ``batch_window_us`` and executed as value-mode GPU transcode batches.
+
+Per-slot statuses from the kernels map to JSON-RPC errors; E_UNSUPPORTED /
+E_OVERFLOW slots are re-transcoded on the CPU oracle (cpu_ref.py) and
+counted — the GPU path never silently falls back wholesale.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import json
+import logging
+import threading
+import time
+from concurrent.futures import ThreadPoolExecutor
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional, Sequence, Tuple
+
+import numpy as np
+
+from ..config import Config
+from ..mcp import types as mcp
+from .cpu_ref import CpuTranscoder
+from .tables import CompiledTables, compile_tables
+
+log = logging.getLogger("ggrmcp.engine")
+
+# status codes (mirror ops/csrc/common.h)
+E_OK = 0
+E_PARSE = 1
+E_INVALID_REQUEST = 2
+E_METHOD_NOT_FOUND = 3
+E_INVALID_PARAMS = 4
+E_LIMIT = 5
+E_UNSUPPORTED = 6
+E_OVERFLOW = 7
+E_NOT_TOOLCALL = 8
+
+SR_ID_IS_MISSING = 1
+SR_SERVER_STREAMING = 2
+
+_STATUS_TO_RPC = {
+    E_PARSE: (mcp.PARSE_ERROR, "parse error"),
+    E_INVALID_REQUEST: (mcp.INVALID_REQUEST, "invalid request"),
+    E_METHOD_NOT_FOUND: (mcp.METHOD_NOT_FOUND, "tool not found"),
+    E_INVALID_PARAMS: (mcp.INVALID_PARAMS, "invalid params"),
+    E_LIMIT: (mcp.INVALID_PARAMS, "argument limits exceeded"),
+    E_NOT_TOOLCALL: (mcp.METHOD_NOT_FOUND, "method not supported on batch path"),
+}
+
+SLOT_DTYPE = np.dtype(
+    [
+        ("status", "<i4"),
+        ("tool_idx", "<i4"),
+        ("pb_off", "<u4"),
+        ("pb_len", "<u4"),
+        ("err_pos", "<u4"),
+        ("aux", "<i4"),
+        ("id_len", "<u4"),
+        ("flags", "<u4"),
+    ]
+)
+DECODE_DTYPE = np.dtype(
+    [("status", "<i4"), ("out_off", "<u4"), ("out_len", "<u4"), ("pad", "<u4")]
+)
+
+
+def _offsets(lengths: Sequence[int], pad: int = 0, align: int = 1) -> np.ndarray:
+    off = np.zeros(len(lengths) + 1, dtype=np.uint32)
+    acc = 0
+    for i, n in enumerate(lengths):
+        cap = n + pad
+        if align > 1:
+            cap = (cap + align - 1) // align * align
+        acc += cap
+        off[i + 1] = acc
+    return off
+
+
+@dataclass
+class EngineStats:
+    batches: int = 0
+    requests: int = 0
+    gpu_ok: int = 0
+    errors: int = 0
+    host_fallbacks: int = 0
+    encode_ns: int = 0
+    decode_ns: int = 0
+    invoke_ns: int = 0
+
+    def snapshot(self) -> Dict[str, Any]:
+        return {
+            "batches": self.batches,
+            "requests": self.requests,
+            "gpuOk": self.gpu_ok,
+            "errors": self.errors,
+            "hostFallbacks": self.host_fallbacks,
+            "encodeMs": self.encode_ns / 1e6,
+            "decodeMs": self.decode_ns / 1e6,
+            "invokeMs": self.invoke_ns / 1e6,
+        }
+
+
+class GpuEngine:
+    """One GPU device engine: compiled tables + the HIP extension Engine."""
+
+    def __init__(self, tools: Dict[str, Any], config: Optional[Config] = None,
+                 device: int = 0) -> None:
+        self.config = config or Config.default()
+        from .. import ops
+
+        self._mod = ops.load()
+        if self._mod.device_count() == 0:
+            raise ops.ExtensionUnavailable("no HIP device present")
+        self.device = device
+        self.tables: CompiledTables = compile_tables(tools)
+        self.tools = tools
+        gpu = self.config.gpu
+        self._eng = self._mod.Engine(
+            device,
+            self.tables.msg_table,
+            self.tables.field_table,
+            self.tables.enum_table,
+            self.tables.enum_values,
+            self.tables.tool_table,
+            self.tables.name_blob,
+            self.tables.n_msgs,
+            self.tables.n_tools,
+            max_batch=gpu.max_batch,
+            cap_in=gpu.pinned_pool_bytes // 4,
+            cap_pb=gpu.pinned_pool_bytes // 4,
+            cap_scratch=gpu.device_pool_bytes // 4,
+            cap_final=gpu.device_pool_bytes // 4,
+        )
+        self.stats = EngineStats()
+        self._lock = threading.Lock()  # one in-flight batch per engine
+
+    # -- low-level batch ops -------------------------------------------------
+
+    def encode_batch(
+        self,
+        payloads: Sequence[bytes],
+        mode: int,
+        msg_indices: Optional[Sequence[int]] = None,
+        enforce: bool = True,
+    ) -> Tuple[np.ndarray, List[Optional[bytes]]]:
+        """Run k_json2pb. Returns (slot results, per-request pb bytes)."""
+        data = b"".join(payloads)
+        in_off = _offsets([len(p) for p in payloads])
+        pb_off = _offsets([len(p) for p in payloads], pad=192, align=16)
+        msg_idx = (
+            np.asarray(msg_indices, dtype=np.int32) if msg_indices is not None else None
+        )
+        lim = self.config.session  # unused; limits from validation defaults
+        t0 = time.perf_counter_ns()
+        raw, pb_view = self._eng.encode(
+            data,
+            in_off,
+            pb_off,
+            msg_idx,
+            mode,
+            10,
+            1024,
+            1 << 20,
+            1 if enforce else 0,
+        )
+        self.stats.encode_ns += time.perf_counter_ns() - t0
+        results = np.frombuffer(raw.tobytes(), dtype=SLOT_DTYPE)
+        pb_mem = memoryview(pb_view)
+        out: List[Optional[bytes]] = []
+        for r in results:
+            if r["status"] == E_OK:
+                out.append(bytes(pb_mem[r["pb_off"] : r["pb_off"] + r["pb_len"]]))
+            else:
+                out.append(None)
+        return results, out
+
+    def decode_batch(
+        self,
+        payloads: Sequence[Optional[bytes]],
+        msg_indices: Sequence[int],
+        mode: int,
+        skip: Optional[Sequence[bool]] = None,
+    ) -> Tuple[np.ndarray, List[Optional[bytes]]]:
+        """Run k_pb2json. ``None`` payloads are auto-skipped slots."""
+        skips = [1 if (p is None or (skip is not None and skip[i])) else 0
+                 for i, p in enumerate(payloads)]
+        safe = [p if p is not None else b"" for p in payloads]
+        data = b"".join(safe)
+        lens = [len(p) for p in safe]
+        resp_off = _offsets(lens)
+        scratch_off = _offsets([n * 8 + 1024 for n in lens], align=16)
+        final_off = _offsets([n * 16 + 2048 for n in lens], align=16)
+        t0 = time.perf_counter_ns()
+        raw, out_view = self._eng.decode(
+            data,
+            resp_off,
+            scratch_off,
+            final_off,
+            np.asarray(msg_indices, dtype=np.int32),
+            np.asarray(skips, dtype=np.int32),
+            mode,
+        )
+        self.stats.decode_ns += time.perf_counter_ns() - t0
+        results = np.frombuffer(raw.tobytes(), dtype=DECODE_DTYPE)
+        mem = memoryview(out_view)
+        out: List[Optional[bytes]] = []
+        for i, r in enumerate(results):
+            if skips[i] or r["status"] != E_OK:
+                out.append(None)
+            else:
+                out.append(bytes(mem[r["out_off"] : r["out_off"] + r["out_len"]]))
+        return results, out
+
+
+class GpuPipeline:
+    """Full tools/call pipeline over one GpuEngine + a ServiceDiscoverer."""
+
+    def __init__(self, discoverer, config: Optional[Config] = None, device: int = 0,
+                 invoke_workers: int = 64) -> None:
+        self.discoverer = discoverer
+        self.config = config or Config.default()
+        self.engine = GpuEngine(discoverer.tools, self.config, device)
+        self.cpu = CpuTranscoder()
+        self._invoke_pool = ThreadPoolExecutor(
+            max_workers=invoke_workers, thread_name_prefix="ginvoke"
+        )
+        # tool idx -> MethodInfo
+        self._mi_by_idx = [
+            discoverer.tools[name] for name in self.engine.tables.tool_order
+        ]
+        self._out_msg_idx = np.asarray(
+            [
+                self.engine.tables.msg_index[mi.output_descriptor.full_name]
+                for mi in self._mi_by_idx
+            ],
+            dtype=np.int32,
+        )
+
+    def close(self) -> None:
+        self._invoke_pool.shutdown(wait=False)
+
+    # ---- the full batched hot path ----------------------------------------
+
+    def process_batch(
+        self,
+        bodies: Sequence[bytes],
+        headers: Optional[Sequence[Dict[str, str]]] = None,
+        timeout_s: Optional[float] = None,
+    ) -> List[bytes]:
+        """JSON-RPC request bodies -> JSON-RPC response bodies."""
+        st = self.engine.stats
+        st.batches += 1
+        st.requests += len(bodies)
+        enc, pbs = self.engine.encode_batch(bodies, mode=0)
+
+        # fan out gRPC invocations for OK slots (host-side I/O stage)
+        n = len(bodies)
+        resp_wire: List[Optional[bytes]] = [None] * n
+        rpc_error: List[Optional[Exception]] = [None] * n
+        out_idx = np.zeros(n, dtype=np.int32)
+
+        t0 = time.perf_counter_ns()
+        futures = {}
+        for i in range(n):
+            if enc[i]["status"] != E_OK:
+                continue
+            if enc[i]["flags"] & SR_SERVER_STREAMING:
+                continue  # streaming handled below via host assembly
+            mi = self._mi_by_idx[enc[i]["tool_idx"]]
+            out_idx[i] = self._out_msg_idx[enc[i]["tool_idx"]]
+            hdr = headers[i] if headers else None
+            futures[i] = self._invoke_pool.submit(
+                self.discoverer.invoke_wire, mi, pbs[i], hdr, timeout_s
+            )
+        for i, fut in futures.items():
+            try:
+                resp_wire[i] = fut.result()
+            except Exception as e:
+                rpc_error[i] = e
+        st.invoke_ns += time.perf_counter_ns() - t0
+
+        dec, finals = self.engine.decode_batch(resp_wire, out_idx, mode=0)
+
+        # assemble the batch: GPU envelopes where OK, host for the rest
+        out: List[bytes] = []
+        for i in range(n):
+            if finals[i] is not None:
+                st.gpu_ok += 1
+                out.append(finals[i])
+                continue
+            out.append(self._host_slot(bodies[i], enc[i], dec[i] if resp_wire[i] is not None else None,
+                                       resp_wire[i], rpc_error[i], headers[i] if headers else None,
+                                       timeout_s))
+        return out
+
+    # ---- host handling of non-GPU slots ------------------------------------
+
+    def _host_slot(self, body, enc_r, dec_r, wire, rpc_err, hdr, timeout_s) -> bytes:
+        st = self.engine.stats
+        rid, has_id = self._extract_id(body)
+        status = int(enc_r["status"])
+        flags = int(enc_r["flags"])
+
+        if status == E_OK and flags & SR_SERVER_STREAMING:
+            return self._host_streaming(body, enc_r, rid, hdr, timeout_s)
+
+        if status in (E_UNSUPPORTED, E_OVERFLOW) or (
+            status == E_OK and dec_r is not None and int(dec_r["status"]) in (E_UNSUPPORTED, E_OVERFLOW, E_PARSE, E_LIMIT)
+        ):
+            # full CPU transcode fallback (counted)
+            st.host_fallbacks += 1
+            return self._cpu_full(body, rid, hdr, timeout_s)
+
+        if status == E_OK and rpc_err is not None:
+            # gRPC failure -> isError tool result (handler.go:252-259)
+            import grpc
+
+            if isinstance(rpc_err, grpc.RpcError):
+                code = rpc_err.code().name if hasattr(rpc_err, "code") else "UNKNOWN"
+                detail = rpc_err.details() if hasattr(rpc_err, "details") else str(rpc_err)
+                text = f"gRPC error {code}: {detail}"
+            else:
+                text = str(rpc_err)
+            st.errors += 1
+            result = mcp.ToolCallResult(content=[mcp.TextContent(text)], is_error=True)
+            resp = mcp.JSONRPCResponse(id=rid, result=result.to_dict())
+            return json.dumps(resp.to_dict(), ensure_ascii=False).encode()
+
+        code, msg = _STATUS_TO_RPC.get(status, (mcp.INTERNAL_ERROR, "internal error"))
+        st.errors += 1
+        resp = mcp.JSONRPCResponse(id=rid, error=mcp.RPCError(code, msg))
+        return json.dumps(resp.to_dict(), ensure_ascii=False).encode()
+
+    def _cpu_full(self, body: bytes, rid, hdr, timeout_s) -> bytes:
+        try:
+            data = json.loads(body)
+            params = data.get("params") or {}
+            tool = params.get("name", "")
+            args_json = json.dumps(params.get("arguments", {}), ensure_ascii=False)
+            output = self.discoverer.invoke_method_by_tool(tool, args_json, hdr, timeout_s)
+            result = mcp.ToolCallResult(content=[mcp.TextContent(output)], is_error=False)
+            resp = mcp.JSONRPCResponse(id=rid, result=result.to_dict())
+        except Exception as e:
+            resp = mcp.JSONRPCResponse(
+                id=rid, error=mcp.RPCError(mcp.INTERNAL_ERROR, str(e)[:512])
+            )
+        return json.dumps(resp.to_dict(), ensure_ascii=False).encode()
+
+    def _host_streaming(self, body, enc_r, rid, hdr, timeout_s) -> bytes:
+        mi = self._mi_by_idx[enc_r["tool_idx"]]
+        try:
+            data = json.loads(body)
+            args_json = json.dumps(
+                (data.get("params") or {}).get("arguments", {}), ensure_ascii=False
+            )
+            chunks = list(
+                self.discoverer.invoke_streaming(mi.tool_name(), args_json, hdr, timeout_s)
+            )
+            result = mcp.ToolCallResult(
+                content=[mcp.TextContent(c) for c in chunks], is_error=False
+            )
+            resp = mcp.JSONRPCResponse(id=rid, result=result.to_dict())
+        except Exception as e:
+            resp = mcp.JSONRPCResponse(
+                id=rid, error=mcp.RPCError(mcp.INTERNAL_ERROR, str(e)[:512])
+            )
+        return json.dumps(resp.to_dict(), ensure_ascii=False).encode()
+
+    @staticmethod
+    def _extract_id(body: bytes):
+        try:
+            data = json.loads(body)
+            return data.get("id"), "id" in data
+        except Exception:
+            return None, False
+
+
+class BatchEngineInvoker:
+    """Async invoker seam for the HTTP handler: batches concurrent value-mode
+    transcodes onto the GPU (the handler already parsed the envelope)."""
+
+    def __init__(self, discoverer, config: Optional[Config] = None, device: int = 0):
+        self.config = config or Config.default()
+        self.pipeline = GpuPipeline(discoverer, self.config, device)
+        self.discoverer = discoverer
+        self._queue: List[Tuple[str, str, Dict[str, str], float, asyncio.Future]] = []
+        self._qlock = threading.Lock()
+        self._wakeup: Optional[asyncio.Event] = None
+        self._task: Optional[asyncio.Task] = None
+
+    def stats(self) -> Dict[str, Any]:
+        return self.pipeline.engine.stats.snapshot()
+
+    async def _ensure_worker(self) -> None:
+        if self._task is None or self._task.done():
+            self._wakeup = asyncio.Event()
+            self._task = asyncio.get_running_loop().create_task(self._worker())
+
+    async def invoke(self, tool_name, args_json, headers, timeout_s) -> str:
+        await self._ensure_worker()
+        fut = asyncio.get_running_loop().create_future()
+        with self._qlock:
+            self._queue.append((tool_name, args_json, headers, timeout_s, fut))
+        self._wakeup.set()
+        return await fut
+
+    async def invoke_streaming(self, tool_name, args_json, headers, timeout_s):
+        loop = asyncio.get_running_loop()
+        return await loop.run_in_executor(
+            self.pipeline._invoke_pool,
+            lambda: list(
+                self.discoverer.invoke_streaming(tool_name, args_json, headers, timeout_s)
+            ),
+        )
+
+    async def _worker(self) -> None:
+        window_s = self.config.gpu.batch_window_us / 1e6
+        while True:
+            await self._wakeup.wait()
+            await asyncio.sleep(window_s)  # batch collection window
+            with self._qlock:
+                batch = self._queue
+                self._queue = []
+                self._wakeup.clear()
+            if not batch:
+                continue
+            loop = asyncio.get_running_loop()
+            try:
+                results = await loop.run_in_executor(None, self._run_batch, batch)
+                for (_, _, _, _, fut), res in zip(batch, results):
+                    if not fut.done():
+                        if isinstance(res, Exception):
+                            fut.set_exception(res)
+                        else:
+                            fut.set_result(res)
+            except Exception as e:  # batch-level failure
+                for _, _, _, _, fut in batch:
+                    if not fut.done():
+                        fut.set_exception(e)
+
+    def _run_batch(self, batch) -> List[Any]:
+        """Value-mode GPU transcode for a set of already-validated calls."""
+        pipeline = self.pipeline
+        eng = pipeline.engine
+        eng.stats.batches += 1
+        eng.stats.requests += len(batch)
+        in_idx = np.zeros(len(batch), dtype=np.int32)
+        mis = []
+        payloads = []
+        for i, (tool_name, args_json, _h, _t, _f) in enumerate(batch):
+            mi = self.discoverer.tools.get(tool_name)
+            mis.append(mi)
+            if mi is None:
+                payloads.append(b"{}")
+                continue
+            in_idx[i] = eng.tables.msg_index[mi.input_descriptor.full_name]
+            payloads.append(args_json.encode())
+        enc, pbs = eng.encode_batch(payloads, mode=1, msg_indices=in_idx, enforce=True)
+        results: List[Any] = [None] * len(batch)
+        out_idx = np.zeros(len(batch), dtype=np.int32)
+        wires: List[Optional[bytes]] = [None] * len(batch)
+        futures = {}
+        t0 = time.perf_counter_ns()
+        for i, (tool_name, args_json, hdr, timeout_s, _f) in enumerate(batch):
+            mi = mis[i]
+            if mi is None:
+                from ..backend.discovery import MethodNotFoundError
+
+                results[i] = MethodNotFoundError(f"tool not found: {tool_name}")
+                continue
+            if enc[i]["status"] == E_OK:
+                out_idx[i] = eng.tables.msg_index[mi.output_descriptor.full_name]
+                futures[i] = pipeline._invoke_pool.submit(
+                    self.discoverer.invoke_wire, mi, pbs[i], hdr, timeout_s
+                )
+            elif enc[i]["status"] in (E_UNSUPPORTED, E_OVERFLOW):
+                eng.stats.host_fallbacks += 1
+                try:
+                    results[i] = self.discoverer.invoke_method_by_tool(
+                        tool_name, args_json, hdr, timeout_s
+                    )
+                except Exception as e:
+                    results[i] = e
+            else:
+                from ..mcp.validation import ValidationError
+
+                code, msg = _STATUS_TO_RPC.get(
+                    int(enc[i]["status"]), (mcp.INVALID_PARAMS, "invalid arguments")
+                )
+                eng.stats.errors += 1
+                results[i] = ValidationError("arguments", msg)
+        for i, fut in futures.items():
+            try:
+                wires[i] = fut.result()
+            except Exception as e:
+                results[i] = e
+        eng.stats.invoke_ns += time.perf_counter_ns() - t0
+        dec, jsons = eng.decode_batch(wires, out_idx, mode=1)
+        for i in list(futures.keys()):
+            if wires[i] is None:
+                continue
+            if jsons[i] is not None:
+                eng.stats.gpu_ok += 1
+                results[i] = jsons[i].decode()
+            else:
+                eng.stats.host_fallbacks += 1
+                mi = mis[i]
+                try:
+                    results[i] = pipeline.cpu.pb_to_json(mi.output_descriptor, wires[i])
+                except Exception as e:
+                    results[i] = e
+        return results

@@ -1,0 +1,262 @@
+// Shared table layouts, slot records and device helpers for the MI355X
+// batch-transcode kernels (gfx950 / CDNA4, wave64).
+//
+// The table formats mirror ggrmcp_amd/engine/tables.py exactly (packed,
+// little-endian); static_asserts below pin the sizes.  These kernels are the
+// GPU replacement for the reference gateway's per-request CPU hot path
+// (aalobaidi/ggRMCP pkg/server/handler.go:81-139 + pkg/grpc/
+// reflection.go:333-391): JSON-RPC envelope parse, tool-call validation,
+// JSON->protobuf encode, protobuf->JSON decode, response-envelope assembly.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+// ---------------------------------------------------------------------------
+// Table records (mirror engine/tables.py; packed little-endian)
+// ---------------------------------------------------------------------------
+
+struct __attribute__((packed)) FieldEntry {
+  uint64_t hash_json;   // FNV-1a64 of json_name
+  uint64_t hash_orig;   // FNV-1a64 of proto field name
+  uint32_t number;      // field number
+  uint32_t name_off;    // proto name offset in name_blob
+  uint32_t json_off;    // json name offset in name_blob
+  uint16_t name_len;
+  uint16_t json_len;
+  int32_t sub_index;    // message index / enum index / -1
+  uint8_t kind;         // google.protobuf FieldDescriptor TYPE_*
+  uint8_t flags;        // F_* below
+  uint8_t oneof_id;     // 255 = none
+  uint8_t pad;
+};
+static_assert(sizeof(FieldEntry) == 40, "FieldEntry layout");
+
+struct __attribute__((packed)) MsgEntry {
+  int32_t field_start;
+  int32_t field_count;
+  int32_t wkt_kind;  // WKT_* below
+  int32_t flags;
+};
+static_assert(sizeof(MsgEntry) == 16, "MsgEntry layout");
+
+struct __attribute__((packed)) EnumEntry {
+  int32_t val_start;
+  int32_t val_count;
+};
+static_assert(sizeof(EnumEntry) == 8, "EnumEntry layout");
+
+struct __attribute__((packed)) EnumValueEntry {
+  uint64_t hash;
+  int32_t number;
+  uint32_t name_off;
+  uint16_t name_len;
+  uint16_t pad;
+  int32_t pad2;
+};
+static_assert(sizeof(EnumValueEntry) == 24, "EnumValueEntry layout");
+
+struct __attribute__((packed)) ToolEntry {
+  uint64_t hash;
+  int32_t in_msg;
+  int32_t out_msg;
+  uint32_t name_off;
+  uint16_t name_len;
+  uint16_t flags;  // 1 = server streaming
+};
+static_assert(sizeof(ToolEntry) == 24, "ToolEntry layout");
+
+// field flags (tables.py F_*)
+enum : uint8_t {
+  F_REPEATED = 1,
+  F_PACKED = 2,
+  F_MAP = 4,
+  F_HAS_PRESENCE = 8,
+  F_ONEOF = 16,
+};
+
+// message well-known-type kinds (tables.py WKT_*)
+enum : int32_t {
+  WKT_NONE = 0,
+  WKT_TIMESTAMP = 1,
+  WKT_DURATION = 2,
+  WKT_STRUCT = 3,
+  WKT_VALUE = 4,
+  WKT_LISTVALUE = 5,
+  WKT_ANY = 6,
+  WKT_FIELDMASK = 7,
+  WKT_EMPTY = 8,
+  WKT_WRAPPER = 9,
+};
+
+// protobuf field kinds (FieldDescriptorProto.Type values)
+enum : uint8_t {
+  K_DOUBLE = 1, K_FLOAT = 2, K_INT64 = 3, K_UINT64 = 4, K_INT32 = 5,
+  K_FIXED64 = 6, K_FIXED32 = 7, K_BOOL = 8, K_STRING = 9, K_GROUP = 10,
+  K_MESSAGE = 11, K_BYTES = 12, K_UINT32 = 13, K_ENUM = 14,
+  K_SFIXED32 = 15, K_SFIXED64 = 16, K_SINT32 = 17, K_SINT64 = 18,
+};
+
+// protobuf wire types
+enum : uint32_t { W_VARINT = 0, W_I64 = 1, W_LEN = 2, W_I32 = 5 };
+
+// ---------------------------------------------------------------------------
+// Per-request records (kernel <-> host)
+// ---------------------------------------------------------------------------
+
+// status codes (mirror engine/batch.py STATUS_*)
+enum : int32_t {
+  E_OK = 0,
+  E_PARSE = 1,           // malformed JSON
+  E_INVALID_REQUEST = 2, // bad JSON-RPC envelope
+  E_METHOD_NOT_FOUND = 3,// unknown tool name
+  E_INVALID_PARAMS = 4,  // schema violation (unknown field, wrong type)
+  E_LIMIT = 5,           // depth/string/size limit exceeded
+  E_UNSUPPORTED = 6,     // valid but outside the GPU subset -> host fallback
+  E_OVERFLOW = 7,        // output buffer cap exceeded -> host fallback
+  E_NOT_TOOLCALL = 8,    // well-formed JSON-RPC but method != tools/call
+};
+
+// result of the encode (ingest) kernel, one per request slot
+struct __attribute__((packed)) SlotResult {
+  int32_t status;
+  int32_t tool_idx;     // resolved tool (valid when status==E_OK or stream)
+  uint32_t pb_off;      // request wire bytes in the pb arena
+  uint32_t pb_len;
+  uint32_t err_pos;     // input byte position of the error (diagnostics)
+  int32_t aux;          // error detail (field number / oneof id / limit kind)
+  uint32_t id_len;      // bytes of the JSON-RPC id copied into the id slot
+  uint32_t flags;       // SR_* below
+};
+static_assert(sizeof(SlotResult) == 32, "SlotResult layout");
+
+enum : uint32_t {
+  SR_ID_IS_MISSING = 1,   // request had no "id" member
+  SR_SERVER_STREAMING = 2,
+};
+
+// result of the decode (respond) kernel, one per request slot
+struct __attribute__((packed)) DecodeResult {
+  int32_t status;
+  uint32_t out_off;  // final response bytes in the output arena
+  uint32_t out_len;
+  uint32_t pad;
+};
+static_assert(sizeof(DecodeResult) == 16, "DecodeResult layout");
+
+// JSON-RPC id storage per slot (raw JSON token bytes, e.g. `17` or `"abc"`)
+constexpr int ID_SLOT_BYTES = 48;
+
+// limits passed to the encode kernel (defaults mirror mcp/validation.py)
+struct Limits {
+  uint32_t max_depth;       // nesting depth cap (reference: 10)
+  uint32_t max_string;      // per-string char cap (reference: 1024)
+  uint32_t max_args_bytes;  // whole-arguments byte cap (reference: 1 MB)
+  uint32_t enforce;         // 0 = transcode-only mode (no MCP limits)
+};
+
+// Table bundle passed to kernels
+struct Tables {
+  const MsgEntry* msgs;
+  const FieldEntry* fields;
+  const EnumEntry* enums;
+  const EnumValueEntry* enum_vals;
+  const ToolEntry* tools;
+  const uint8_t* names;
+  int32_t n_msgs;
+  int32_t n_tools;
+};
+
+// ---------------------------------------------------------------------------
+// Device helpers
+// ---------------------------------------------------------------------------
+
+#define WAVE 64
+#define DEV __device__ __forceinline__
+
+DEV uint64_t fnv1a64(const uint8_t* p, uint32_t n) {
+  uint64_t h = 0xCBF29CE484222325ull;
+  for (uint32_t i = 0; i < n; ++i) {
+    h ^= p[i];
+    h *= 0x100000001B3ull;
+  }
+  return h;
+}
+
+DEV int lane_id() { return threadIdx.x & (WAVE - 1); }
+
+// wave-uniform broadcast of a value held by lane 0
+DEV uint32_t bcast0_u32(uint32_t v) { return __shfl(v, 0, WAVE); }
+DEV uint64_t bcast0_u64(uint64_t v) { return __shfl(v, 0, WAVE); }
+
+DEV bool is_ws(uint8_t c) { return c == ' ' || c == '\t' || c == '\n' || c == '\r'; }
+
+// zigzag
+DEV uint64_t zigzag64(int64_t v) { return ((uint64_t)v << 1) ^ (uint64_t)(v >> 63); }
+DEV uint32_t zigzag32(int32_t v) { return ((uint32_t)v << 1) ^ (uint32_t)(v >> 31); }
+DEV int64_t unzigzag64(uint64_t v) { return (int64_t)(v >> 1) ^ -(int64_t)(v & 1); }
+
+// write a varint; returns byte count (<= 10)
+DEV uint32_t put_varint(uint8_t* out, uint64_t v) {
+  uint32_t n = 0;
+  while (v >= 0x80) {
+    out[n++] = (uint8_t)(v | 0x80);
+    v >>= 7;
+  }
+  out[n++] = (uint8_t)v;
+  return n;
+}
+
+// write v as a NON-MINIMAL varint occupying exactly `width` bytes (protobuf
+// parsers accept padded varints).  Used to backfill reserved length slots of
+// nested messages / packed arrays without a second sizing pass.
+DEV void put_varint_fixed(uint8_t* out, uint64_t v, uint32_t width) {
+  for (uint32_t i = 0; i + 1 < width; ++i) {
+    out[i] = (uint8_t)(v & 0x7F) | 0x80;
+    v >>= 7;
+  }
+  out[width - 1] = (uint8_t)(v & 0x7F);
+}
+
+// read a varint; advances *pos; returns false on truncation/overlong
+DEV bool get_varint(const uint8_t* p, uint32_t len, uint32_t* pos, uint64_t* out) {
+  uint64_t v = 0;
+  uint32_t shift = 0;
+  uint32_t i = *pos;
+  while (i < len && shift <= 63) {
+    uint8_t b = p[i++];
+    v |= (uint64_t)(b & 0x7F) << shift;
+    if (!(b & 0x80)) {
+      *pos = i;
+      *out = v;
+      return true;
+    }
+    shift += 7;
+  }
+  return false;
+}
+
+// fixed-width reserved length slots (supports nested payloads < 2^21)
+constexpr uint32_t LEN_SLOT = 3;
+constexpr uint32_t LEN_SLOT_MAX = (1u << 21) - 1;
+
+DEV uint32_t u64_to_dec(uint8_t* out, uint64_t v) {
+  uint8_t tmp[20];
+  uint32_t n = 0;
+  do {
+    tmp[n++] = (uint8_t)('0' + (v % 10));
+    v /= 10;
+  } while (v);
+  for (uint32_t i = 0; i < n; ++i) out[i] = tmp[n - 1 - i];
+  return n;
+}
+
+DEV uint32_t i64_to_dec(uint8_t* out, int64_t v) {
+  if (v < 0) {
+    out[0] = '-';
+    // careful with INT64_MIN
+    uint64_t mag = (uint64_t)(~v) + 1ull;
+    return 1 + u64_to_dec(out + 1, mag);
+  }
+  return u64_to_dec(out, (uint64_t)v);
+}

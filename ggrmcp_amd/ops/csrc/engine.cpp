@@ -1,0 +1,336 @@
+// _jsonproto — host runtime for the MI355X batch-transcode engine.
+//
+// Owns the device arenas (HBM), pinned host staging buffers and the HIP
+// stream for one engine instance; uploads the flat descriptor tables
+// compiled by ggrmcp_amd/engine/tables.py; launches the two kernels
+// (k_json2pb ingest, k_pb2json respond) over whole batches.  The reference
+// gateway has no equivalent — its hot path is per-request Go reflection
+// (aalobaidi/ggRMCP pkg/grpc/reflection.go:333-391).
+//
+// Single compilation unit: the kernels are #included below so no
+// relocatable-device-code link step is needed.  Build: ops/build.py
+// (hipcc --offload-arch=gfx950).
+//
+// Concurrency model: one Engine == one HIP stream == one in-flight batch.
+// encode()/decode() release the GIL while the GPU works, so Python-side
+// pipeline threads (gRPC I/O) overlap with kernels; multiple Engine
+// instances on the same device give copy/compute overlap across batches.
+
+#include <pybind11/numpy.h>
+#include <pybind11/pybind11.h>
+
+#include <cstring>
+#include <stdexcept>
+#include <string>
+
+#include "json2pb.hip"
+#include "pb2json.hip"
+
+namespace py = pybind11;
+
+#define HIP_CHECK(expr)                                                     \
+  do {                                                                      \
+    hipError_t _e = (expr);                                                 \
+    if (_e != hipSuccess)                                                   \
+      throw std::runtime_error(std::string("HIP error at " #expr ": ") +    \
+                               hipGetErrorString(_e));                      \
+  } while (0)
+
+namespace {
+
+struct DeviceBuf {
+  void* p = nullptr;
+  size_t n = 0;
+  void alloc(size_t bytes) {
+    HIP_CHECK(hipMalloc(&p, bytes));
+    n = bytes;
+  }
+  ~DeviceBuf() {
+    if (p) (void)hipFree(p);
+  }
+};
+
+struct PinnedBuf {
+  void* p = nullptr;
+  size_t n = 0;
+  void alloc(size_t bytes) {
+    HIP_CHECK(hipHostMalloc(&p, bytes, hipHostMallocDefault));
+    n = bytes;
+  }
+  ~PinnedBuf() {
+    if (p) (void)hipHostFree(p);
+  }
+};
+
+uint64_t cdiv(uint64_t a, uint64_t b) { return (a + b - 1) / b; }
+
+}  // namespace
+
+class Engine {
+ public:
+  Engine(int device, py::bytes msg_table, py::bytes field_table,
+         py::bytes enum_table, py::bytes enum_values, py::bytes tool_table,
+         py::bytes name_blob, int n_msgs, int n_tools, int max_batch,
+         size_t cap_in, size_t cap_pb, size_t cap_scratch, size_t cap_final)
+      : device_(device), max_batch_(max_batch) {
+    HIP_CHECK(hipSetDevice(device_));
+    HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
+
+    upload_blob(msg_table, d_msgs_);
+    upload_blob(field_table, d_fields_);
+    upload_blob(enum_table, d_enums_);
+    upload_blob(enum_values, d_enum_vals_);
+    upload_blob(tool_table, d_tools_);
+    upload_blob(name_blob, d_names_);
+    tables_.msgs = (const MsgEntry*)d_msgs_.p;
+    tables_.fields = (const FieldEntry*)d_fields_.p;
+    tables_.enums = (const EnumEntry*)d_enums_.p;
+    tables_.enum_vals = (const EnumValueEntry*)d_enum_vals_.p;
+    tables_.tools = (const ToolEntry*)d_tools_.p;
+    tables_.names = (const uint8_t*)d_names_.p;
+    tables_.n_msgs = n_msgs;
+    tables_.n_tools = n_tools;
+
+    d_in_.alloc(cap_in);
+    d_pb_.alloc(cap_pb);
+    d_resp_.alloc(cap_in);
+    d_scratch_.alloc(cap_scratch);
+    d_final_.alloc(cap_final);
+    size_t offs = (size_t)(max_batch + 1) * sizeof(uint32_t);
+    d_in_off_.alloc(offs);
+    d_pb_off_.alloc(offs);
+    d_resp_off_.alloc(offs);
+    d_scratch_off_.alloc(offs);
+    d_final_off_.alloc(offs);
+    d_msg_idx_.alloc((size_t)max_batch * sizeof(int32_t));
+    d_skip_.alloc((size_t)max_batch * sizeof(int32_t));
+    d_results_.alloc((size_t)max_batch * sizeof(SlotResult));
+    d_dec_results_.alloc((size_t)max_batch * sizeof(DecodeResult));
+    d_id_slots_.alloc((size_t)max_batch * ID_SLOT_BYTES);
+
+    h_in_.alloc(cap_in);
+    h_pb_.alloc(cap_pb);
+    h_resp_.alloc(cap_in);
+    h_final_.alloc(cap_final);
+    h_results_.alloc((size_t)max_batch * sizeof(SlotResult));
+    h_dec_results_.alloc((size_t)max_batch * sizeof(DecodeResult));
+    h_off_.alloc(offs * 2);
+    h_aux_.alloc((size_t)max_batch * sizeof(int32_t) * 2);
+  }
+
+  ~Engine() {
+    (void)hipStreamSynchronize(stream_);
+    (void)hipStreamDestroy(stream_);
+  }
+
+  // JSON(-RPC) -> protobuf.  mode 0: full envelope; mode 1: bare message.
+  // Returns (results u8[ n*32 ] structured, pb memoryview).
+  py::tuple encode(py::buffer data, py::array_t<uint32_t> in_off,
+                   py::array_t<uint32_t> pb_off,
+                   py::object msg_idx,  // int32 array or None
+                   int mode, uint32_t max_depth, uint32_t max_string,
+                   uint32_t max_args, int enforce) {
+    py::buffer_info din = data.request();
+    auto in_off_v = in_off.unchecked<1>();
+    auto pb_off_v = pb_off.unchecked<1>();
+    int n = (int)in_off_v.shape(0) - 1;
+    if (n < 0 || n > max_batch_) throw std::runtime_error("bad batch size");
+    size_t in_bytes = in_off_v(n);
+    size_t pb_bytes = pb_off_v(n);
+    if (in_bytes > h_in_.n || (size_t)din.size < in_bytes)
+      throw std::runtime_error("input exceeds engine cap_in");
+    if (pb_bytes > d_pb_.n) throw std::runtime_error("pb cap exceeded");
+
+    const int32_t* msg_idx_ptr = nullptr;
+    py::array_t<int32_t> msg_idx_arr;
+    if (!msg_idx.is_none()) {
+      msg_idx_arr = msg_idx.cast<py::array_t<int32_t>>();
+      msg_idx_ptr = msg_idx_arr.data();
+    }
+
+    std::memcpy(h_in_.p, din.ptr, in_bytes);
+    uint32_t* h_off = (uint32_t*)h_off_.p;
+    std::memcpy(h_off, in_off.data(), (n + 1) * sizeof(uint32_t));
+    std::memcpy(h_off + (n + 1), pb_off.data(), (n + 1) * sizeof(uint32_t));
+    int32_t* h_aux = (int32_t*)h_aux_.p;
+    if (msg_idx_ptr) std::memcpy(h_aux, msg_idx_ptr, n * sizeof(int32_t));
+
+    Limits lim{max_depth, max_string, max_args, (uint32_t)enforce};
+    {
+      py::gil_scoped_release rel;
+      HIP_CHECK(hipSetDevice(device_));
+      HIP_CHECK(hipMemcpyAsync(d_in_.p, h_in_.p, in_bytes,
+                               hipMemcpyHostToDevice, stream_));
+      HIP_CHECK(hipMemcpyAsync(d_in_off_.p, h_off, (n + 1) * sizeof(uint32_t),
+                               hipMemcpyHostToDevice, stream_));
+      HIP_CHECK(hipMemcpyAsync(d_pb_off_.p, h_off + (n + 1),
+                               (n + 1) * sizeof(uint32_t),
+                               hipMemcpyHostToDevice, stream_));
+      if (msg_idx_ptr)
+        HIP_CHECK(hipMemcpyAsync(d_msg_idx_.p, h_aux, n * sizeof(int32_t),
+                                 hipMemcpyHostToDevice, stream_));
+      int blocks = (int)cdiv(n, WPB);
+      if (blocks > 0) {
+        hipLaunchKernelGGL(k_json2pb, dim3(blocks), dim3(WPB * WAVE), 0,
+                           stream_, (const uint8_t*)d_in_.p,
+                           (const uint32_t*)d_in_off_.p, (uint8_t*)d_pb_.p,
+                           (const uint32_t*)d_pb_off_.p,
+                           (SlotResult*)d_results_.p, (uint8_t*)d_id_slots_.p,
+                           msg_idx_ptr ? (const int32_t*)d_msg_idx_.p : nullptr,
+                           tables_, lim, n, mode);
+        HIP_CHECK(hipGetLastError());
+      }
+      HIP_CHECK(hipMemcpyAsync(h_results_.p, d_results_.p,
+                               n * sizeof(SlotResult), hipMemcpyDeviceToHost,
+                               stream_));
+      HIP_CHECK(hipMemcpyAsync(h_pb_.p, d_pb_.p, pb_bytes,
+                               hipMemcpyDeviceToHost, stream_));
+      HIP_CHECK(hipStreamSynchronize(stream_));
+    }
+    last_batch_n_ = n;
+    py::array_t<uint8_t> results({(py::ssize_t)(n * sizeof(SlotResult))});
+    std::memcpy(results.mutable_data(), h_results_.p, n * sizeof(SlotResult));
+    py::memoryview pb_view = py::memoryview::from_memory(h_pb_.p, pb_bytes);
+    return py::make_tuple(results, pb_view);
+  }
+
+  // protobuf -> JSON(-RPC response).  mode 0: envelope using the id slots of
+  // the LAST encode() batch (slot-aligned); mode 1: bare JSON.
+  py::tuple decode(py::buffer data, py::array_t<uint32_t> resp_off,
+                   py::array_t<uint32_t> scratch_off,
+                   py::array_t<uint32_t> final_off,
+                   py::array_t<int32_t> msg_idx, py::object skip, int mode) {
+    py::buffer_info din = data.request();
+    auto resp_off_v = resp_off.unchecked<1>();
+    int n = (int)resp_off_v.shape(0) - 1;
+    if (n < 0 || n > max_batch_) throw std::runtime_error("bad batch size");
+    size_t resp_bytes = resp_off_v(n);
+    auto scratch_off_v = scratch_off.unchecked<1>();
+    auto final_off_v = final_off.unchecked<1>();
+    size_t scratch_bytes = scratch_off_v(n);
+    size_t final_bytes = final_off_v(n);
+    if (resp_bytes > h_resp_.n || (size_t)din.size < resp_bytes)
+      throw std::runtime_error("input exceeds engine cap_in");
+    if (scratch_bytes > d_scratch_.n) throw std::runtime_error("scratch cap");
+    if (final_bytes > d_final_.n) throw std::runtime_error("final cap");
+    if (mode == 0 && n != last_batch_n_)
+      throw std::runtime_error("envelope decode batch must match last encode");
+
+    const int32_t* skip_ptr = nullptr;
+    py::array_t<int32_t> skip_arr;
+    if (!skip.is_none()) {
+      skip_arr = skip.cast<py::array_t<int32_t>>();
+      skip_ptr = skip_arr.data();
+    }
+
+    std::memcpy(h_resp_.p, din.ptr, resp_bytes);
+    uint32_t* h_off = (uint32_t*)h_off_.p;
+    std::memcpy(h_off, resp_off.data(), (n + 1) * sizeof(uint32_t));
+    std::memcpy(h_off + (n + 1), final_off.data(), (n + 1) * sizeof(uint32_t));
+    int32_t* h_aux = (int32_t*)h_aux_.p;
+    std::memcpy(h_aux, msg_idx.data(), n * sizeof(int32_t));
+    if (skip_ptr) std::memcpy(h_aux + n, skip_ptr, n * sizeof(int32_t));
+    // scratch offsets travel via d_scratch_off_ (separate small copy)
+    {
+      py::gil_scoped_release rel;
+      HIP_CHECK(hipSetDevice(device_));
+      HIP_CHECK(hipMemcpyAsync(d_resp_.p, h_resp_.p, resp_bytes,
+                               hipMemcpyHostToDevice, stream_));
+      HIP_CHECK(hipMemcpyAsync(d_resp_off_.p, h_off, (n + 1) * sizeof(uint32_t),
+                               hipMemcpyHostToDevice, stream_));
+      HIP_CHECK(hipMemcpyAsync(d_final_off_.p, h_off + (n + 1),
+                               (n + 1) * sizeof(uint32_t),
+                               hipMemcpyHostToDevice, stream_));
+      HIP_CHECK(hipMemcpyAsync(d_scratch_off_.p, scratch_off.data(),
+                               (n + 1) * sizeof(uint32_t),
+                               hipMemcpyHostToDevice, stream_));
+      HIP_CHECK(hipMemcpyAsync(d_msg_idx_.p, h_aux, n * sizeof(int32_t),
+                               hipMemcpyHostToDevice, stream_));
+      if (skip_ptr)
+        HIP_CHECK(hipMemcpyAsync(d_skip_.p, h_aux + n, n * sizeof(int32_t),
+                                 hipMemcpyHostToDevice, stream_));
+      int blocks = (int)cdiv(n, WPB);
+      if (blocks > 0) {
+        hipLaunchKernelGGL(
+            k_pb2json, dim3(blocks), dim3(WPB * WAVE), 0, stream_,
+            (const uint8_t*)d_resp_.p, (const uint32_t*)d_resp_off_.p,
+            (const int32_t*)d_msg_idx_.p, (const uint8_t*)d_id_slots_.p,
+            mode == 0 ? (const SlotResult*)d_results_.p : nullptr,
+            (uint8_t*)d_scratch_.p, (const uint32_t*)d_scratch_off_.p,
+            (uint8_t*)d_final_.p, (const uint32_t*)d_final_off_.p,
+            (DecodeResult*)d_dec_results_.p,
+            skip_ptr ? (const int32_t*)d_skip_.p : nullptr, tables_, n, mode);
+        HIP_CHECK(hipGetLastError());
+      }
+      HIP_CHECK(hipMemcpyAsync(h_dec_results_.p, d_dec_results_.p,
+                               n * sizeof(DecodeResult), hipMemcpyDeviceToHost,
+                               stream_));
+      HIP_CHECK(hipMemcpyAsync(h_final_.p, d_final_.p, final_bytes,
+                               hipMemcpyDeviceToHost, stream_));
+      HIP_CHECK(hipStreamSynchronize(stream_));
+    }
+    py::array_t<uint8_t> results({(py::ssize_t)(n * sizeof(DecodeResult))});
+    std::memcpy(results.mutable_data(), h_dec_results_.p,
+                n * sizeof(DecodeResult));
+    py::memoryview out_view = py::memoryview::from_memory(h_final_.p, final_bytes);
+    return py::make_tuple(results, out_view);
+  }
+
+  int device() const { return device_; }
+  int max_batch() const { return max_batch_; }
+
+ private:
+  void upload_blob(py::bytes b, DeviceBuf& buf) {
+    std::string s = b;  // copy (init-time only)
+    size_t bytes = s.size() ? s.size() : 1;
+    buf.alloc(bytes);
+    if (s.size())
+      HIP_CHECK(hipMemcpy(buf.p, s.data(), s.size(), hipMemcpyHostToDevice));
+  }
+
+  int device_;
+  int max_batch_;
+  int last_batch_n_ = -1;
+  hipStream_t stream_;
+  Tables tables_{};
+  DeviceBuf d_msgs_, d_fields_, d_enums_, d_enum_vals_, d_tools_, d_names_;
+  DeviceBuf d_in_, d_pb_, d_resp_, d_scratch_, d_final_;
+  DeviceBuf d_in_off_, d_pb_off_, d_resp_off_, d_scratch_off_, d_final_off_;
+  DeviceBuf d_msg_idx_, d_skip_, d_results_, d_dec_results_, d_id_slots_;
+  PinnedBuf h_in_, h_pb_, h_resp_, h_final_, h_results_, h_dec_results_, h_off_,
+      h_aux_;
+};
+
+static int device_count() {
+  int n = 0;
+  hipError_t e = hipGetDeviceCount(&n);
+  if (e != hipSuccess) return 0;
+  return n;
+}
+
+PYBIND11_MODULE(_jsonproto, m) {
+  m.doc() = "MI355X batch JSON<->protobuf transcode engine (gfx950 HIP kernels)";
+  m.def("device_count", &device_count);
+  m.attr("ID_SLOT_BYTES") = ID_SLOT_BYTES;
+  m.attr("SLOT_RESULT_SIZE") = (int)sizeof(SlotResult);
+  m.attr("DECODE_RESULT_SIZE") = (int)sizeof(DecodeResult);
+  py::class_<Engine>(m, "Engine")
+      .def(py::init<int, py::bytes, py::bytes, py::bytes, py::bytes, py::bytes,
+                    py::bytes, int, int, int, size_t, size_t, size_t, size_t>(),
+           py::arg("device"), py::arg("msg_table"), py::arg("field_table"),
+           py::arg("enum_table"), py::arg("enum_values"), py::arg("tool_table"),
+           py::arg("name_blob"), py::arg("n_msgs"), py::arg("n_tools"),
+           py::arg("max_batch") = 4096, py::arg("cap_in") = 64u << 20,
+           py::arg("cap_pb") = 80u << 20, py::arg("cap_scratch") = 128u << 20,
+           py::arg("cap_final") = 160u << 20)
+      .def("encode", &Engine::encode, py::arg("data"), py::arg("in_off"),
+           py::arg("pb_off"), py::arg("msg_idx") = py::none(),
+           py::arg("mode") = 0, py::arg("max_depth") = 10,
+           py::arg("max_string") = 1024, py::arg("max_args") = 1u << 20,
+           py::arg("enforce") = 1)
+      .def("decode", &Engine::decode, py::arg("data"), py::arg("resp_off"),
+           py::arg("scratch_off"), py::arg("final_off"), py::arg("msg_idx"),
+           py::arg("skip") = py::none(), py::arg("mode") = 0)
+      .def_property_readonly("device", &Engine::device)
+      .def_property_readonly("max_batch", &Engine::max_batch);
+}

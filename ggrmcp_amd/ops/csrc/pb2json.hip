@@ -1,0 +1,1090 @@
+// k_pb2json — batched response kernel (gfx950, wave64).
+//
+// One 64-lane wave per request.  Replaces the reference gateway's CPU
+// response path: protojson.Marshal of the gRPC response message
+// (reflection.go:381) plus the JSON-RPC ToolCallResult envelope encode
+// (handler.go:252-270, 290): protobuf wire walk -> JSON text in a scratch
+// arena, then lane-parallel JSON-string escaping (exclusive wave scan of
+// per-byte expansion via __shfl) into the final envelope
+//   {"jsonrpc":"2.0","id":<id>,"result":{"content":[{"type":"text",
+//    "text":"<escaped>"}],"isError":false}}.
+//
+// protojson output semantics: camelCase field names, 64-bit ints quoted,
+// enums by name (unknown values by number), bytes as padded std base64,
+// default-valued non-presence fields omitted, Timestamp/Duration RFC3339 /
+// "Ns" forms with 3-digit-group nano trimming, wrappers unwrapped,
+// Struct/Value/ListValue as raw JSON, Empty as {}.  Fields are emitted in
+// wire order with adjacent-occurrence grouping for repeated/map fields;
+// out-of-order wire (no real serializer produces it) and google.protobuf.Any
+// flag E_UNSUPPORTED so the host transcodes that request (counted, never
+// silent).  Doubles print as 17-significant-digit shortest-trimmed decimals
+// (value-exact round trip; not always the minimal digit string).
+
+#include "common.h"
+
+#ifndef WPB
+#define WPB 4
+#endif
+
+struct DCtx {
+  const uint8_t* pb;
+  uint32_t len, pos;
+  uint8_t* out;
+  uint32_t opos, ocap;
+  Tables t;
+  int32_t status;
+  int lane;
+};
+
+DEV bool dfail(DCtx& c, int32_t code) {
+  if (c.status == E_OK) c.status = code;
+  return false;
+}
+
+DEV bool putc_(DCtx& c, uint8_t ch) {
+  if (c.opos >= c.ocap) return dfail(c, E_OVERFLOW);
+  if (!c.lane) c.out[c.opos] = ch;
+  c.opos++;
+  return true;
+}
+
+DEV bool puts_(DCtx& c, const char* s, uint32_t n) {
+  if (c.opos + n > c.ocap) return dfail(c, E_OVERFLOW);
+  for (uint32_t i = c.lane; i < n; i += WAVE) c.out[c.opos + i] = (uint8_t)s[i];
+  c.opos += n;
+  return true;
+}
+
+DEV bool put_u64_dec(DCtx& c, uint64_t v) {
+  if (c.opos + 20 > c.ocap) return dfail(c, E_OVERFLOW);
+  uint32_t n = 1;
+  {
+    uint64_t x = v;
+    while (x >= 10) {
+      x /= 10;
+      ++n;
+    }
+  }
+  if (!c.lane) u64_to_dec(c.out + c.opos, v);
+  c.opos += n;
+  return true;
+}
+
+DEV bool put_i64_dec(DCtx& c, int64_t v) {
+  if (v < 0) {
+    if (!putc_(c, '-')) return false;
+    return put_u64_dec(c, (uint64_t)(~v) + 1ull);
+  }
+  return put_u64_dec(c, (uint64_t)v);
+}
+
+// ---------------------------------------------------------------------------
+// JSON string escaping — lane-parallel with wave prefix scan
+// ---------------------------------------------------------------------------
+
+DEV uint32_t esc_len(uint8_t b) {
+  if (b == '"' || b == '\\') return 2;
+  if (b == '\b' || b == '\f' || b == '\n' || b == '\r' || b == '\t') return 2;
+  if (b < 0x20) return 6;  // \u00XX
+  return 1;
+}
+
+DEV void esc_write(uint8_t* dst, uint8_t b) {
+  static const char HEX[] = "0123456789abcdef";
+  switch (b) {
+    case '"': dst[0] = '\\'; dst[1] = '"'; return;
+    case '\\': dst[0] = '\\'; dst[1] = '\\'; return;
+    case '\b': dst[0] = '\\'; dst[1] = 'b'; return;
+    case '\f': dst[0] = '\\'; dst[1] = 'f'; return;
+    case '\n': dst[0] = '\\'; dst[1] = 'n'; return;
+    case '\r': dst[0] = '\\'; dst[1] = 'r'; return;
+    case '\t': dst[0] = '\\'; dst[1] = 't'; return;
+    default:
+      if (b < 0x20) {
+        dst[0] = '\\'; dst[1] = 'u'; dst[2] = '0'; dst[3] = '0';
+        dst[4] = HEX[b >> 4]; dst[5] = HEX[b & 15];
+      } else {
+        dst[0] = b;
+      }
+  }
+}
+
+// append src[0..n) JSON-escaped; lane-parallel
+DEV bool put_escaped(DCtx& c, const uint8_t* src, uint32_t n) {
+  for (uint32_t base = 0; base < n; base += WAVE) {
+    uint32_t i = base + c.lane;
+    uint8_t b = i < n ? src[i] : 'x';
+    uint32_t el = i < n ? esc_len(b) : 0;
+    // inclusive wave scan of el
+    uint32_t inc = el;
+    #pragma unroll
+    for (int d = 1; d < WAVE; d <<= 1) {
+      uint32_t up = __shfl_up(inc, d, WAVE);
+      if (c.lane >= d) inc += up;
+    }
+    uint32_t total = __shfl(inc, WAVE - 1, WAVE);
+    if (c.opos + total > c.ocap) return dfail(c, E_OVERFLOW);
+    if (i < n) {
+      uint8_t tmp[6];
+      esc_write(tmp, b);
+      uint32_t at = c.opos + inc - el;
+      for (uint32_t k = 0; k < el; ++k) c.out[at + k] = tmp[k];
+    }
+    c.opos += total;
+  }
+  return true;
+}
+
+// append a quoted escaped string
+DEV bool put_json_string(DCtx& c, const uint8_t* src, uint32_t n) {
+  if (!putc_(c, '"')) return false;
+  if (!put_escaped(c, src, n)) return false;
+  return putc_(c, '"');
+}
+
+// ---------------------------------------------------------------------------
+// base64 encode (bytes fields) — lane-parallel, 3 bytes/lane per window
+// ---------------------------------------------------------------------------
+
+DEV bool put_base64(DCtx& c, const uint8_t* src, uint32_t n) {
+  static const char B64[] =
+      "ABCDEFGHIJKLMNOPQRSTUVWXYZabcdefghijklmnopqrstuvwxyz0123456789+/";
+  uint32_t out_n = (n + 2) / 3 * 4;
+  if (c.opos + out_n > c.ocap) return dfail(c, E_OVERFLOW);
+  uint32_t groups = (n + 2) / 3;
+  for (uint32_t g0 = 0; g0 < groups; g0 += WAVE) {
+    uint32_t g = g0 + c.lane;
+    if (g < groups) {
+      uint32_t i = g * 3;
+      uint32_t b0 = src[i];
+      uint32_t b1 = i + 1 < n ? src[i + 1] : 0;
+      uint32_t b2 = i + 2 < n ? src[i + 2] : 0;
+      uint32_t v = (b0 << 16) | (b1 << 8) | b2;
+      uint8_t* dst = c.out + c.opos + g * 4;
+      dst[0] = B64[(v >> 18) & 63];
+      dst[1] = B64[(v >> 12) & 63];
+      dst[2] = i + 1 < n ? B64[(v >> 6) & 63] : '=';
+      dst[3] = i + 2 < n ? B64[v & 63] : '=';
+    }
+  }
+  c.opos += out_n;
+  return true;
+}
+
+// ---------------------------------------------------------------------------
+// double -> decimal text (17 significant digits, trailing zeros trimmed)
+// ---------------------------------------------------------------------------
+
+__constant__ double DPOW10[23] = {1e0,  1e1,  1e2,  1e3,  1e4,  1e5,  1e6,  1e7,
+                                  1e8,  1e9,  1e10, 1e11, 1e12, 1e13, 1e14, 1e15,
+                                  1e16, 1e17, 1e18, 1e19, 1e20, 1e21, 1e22};
+
+DEV uint32_t dtoa17(uint8_t* out, double d) {
+  // caller guarantees finite, non-negative handled via sign by caller
+  if (d == 0.0) {
+    out[0] = '0';
+    return 1;
+  }
+  // integral fast path (exact)
+  if (d == trunc(d) && d < 9.007199254740992e15) {
+    return u64_to_dec(out, (uint64_t)d);
+  }
+  // decimal exponent via log10, corrected
+  int e10 = (int)floor(log10(d));
+  // scale to [1e16, 1e17)
+  double scaled = d;
+  int shift = 16 - e10;
+  {
+    int s = shift;
+    while (s > 22) { scaled *= 1e22; s -= 22; }
+    while (s < -22) { scaled /= 1e22; s += 22; }
+    if (s >= 0) scaled *= DPOW10[s];
+    else scaled /= DPOW10[-s];
+  }
+  if (scaled >= 1e17) { scaled /= 10.0; e10 += 1; }
+  if (scaled < 1e16) { scaled *= 10.0; e10 -= 1; }
+  uint64_t digits = (uint64_t)(scaled + 0.5);
+  if (digits >= 100000000000000000ull) { digits /= 10; e10 += 1; }
+  uint8_t dig[17];
+  for (int i = 16; i >= 0; --i) {
+    dig[i] = (uint8_t)('0' + digits % 10);
+    digits /= 10;
+  }
+  int ndig = 17;
+  while (ndig > 1 && dig[ndig - 1] == '0') --ndig;
+  uint32_t o = 0;
+  if (e10 >= -6 && e10 <= 20) {
+    if (e10 >= 0) {
+      int ip = e10 + 1;  // digits before the point
+      for (int i = 0; i < ip; ++i) out[o++] = i < ndig ? dig[i] : '0';
+      if (ndig > ip) {
+        out[o++] = '.';
+        for (int i = ip; i < ndig; ++i) out[o++] = dig[i];
+      }
+    } else {
+      out[o++] = '0';
+      out[o++] = '.';
+      for (int i = 0; i < -e10 - 1; ++i) out[o++] = '0';
+      for (int i = 0; i < ndig; ++i) out[o++] = dig[i];
+    }
+  } else {
+    out[o++] = dig[0];
+    if (ndig > 1) {
+      out[o++] = '.';
+      for (int i = 1; i < ndig; ++i) out[o++] = dig[i];
+    }
+    out[o++] = 'e';
+    int e = e10;
+    if (e < 0) {
+      out[o++] = '-';
+      e = -e;
+    } else {
+      out[o++] = '+';
+    }
+    if (e >= 100) {
+      out[o++] = (uint8_t)('0' + e / 100);
+      e %= 100;
+      out[o++] = (uint8_t)('0' + e / 10);
+      out[o++] = (uint8_t)('0' + e % 10);
+    } else {
+      out[o++] = (uint8_t)('0' + e / 10);
+      out[o++] = (uint8_t)('0' + e % 10);
+    }
+  }
+  return o;
+}
+
+DEV bool put_double(DCtx& c, double d, bool as_float) {
+  if (as_float) d = (double)(float)d;
+  if (d != d) return puts_(c, "\"NaN\"", 5);
+  if (isinf(d))
+    return d > 0 ? puts_(c, "\"Infinity\"", 10) : puts_(c, "\"-Infinity\"", 11);
+  if (c.opos + 32 > c.ocap) return dfail(c, E_OVERFLOW);
+  uint32_t n = 0;
+  uint8_t buf[32];
+  if (d < 0 || (d == 0.0 && signbit(d))) {
+    buf[0] = '-';
+    n = 1 + dtoa17(buf + 1, -d);
+  } else {
+    n = dtoa17(buf, d);
+  }
+  if (!c.lane)
+    for (uint32_t i = 0; i < n; ++i) c.out[c.opos + i] = buf[i];
+  c.opos += n;
+  return true;
+}
+
+// ---------------------------------------------------------------------------
+// Timestamp / Duration formatting
+// ---------------------------------------------------------------------------
+
+DEV void civil_from_days(int64_t z, int64_t* y, int* m, int* d) {
+  z += 719468;
+  int64_t era = (z >= 0 ? z : z - 146096) / 146097;
+  uint64_t doe = (uint64_t)(z - era * 146097);
+  uint64_t yoe = (doe - doe / 1460 + doe / 36524 - doe / 146096) / 365;
+  int64_t yr = (int64_t)yoe + era * 400;
+  uint64_t doy = doe - (365 * yoe + yoe / 4 - yoe / 100);
+  uint64_t mp = (5 * doy + 2) / 153;
+  uint64_t dd = doy - (153 * mp + 2) / 5 + 1;
+  uint64_t mm = mp + (mp < 10 ? 3 : (uint64_t)-9);
+  *y = yr + (mm <= 2);
+  *m = (int)mm;
+  *d = (int)dd;
+}
+
+DEV uint32_t put2(uint8_t* o, int v) {
+  o[0] = (uint8_t)('0' + v / 10);
+  o[1] = (uint8_t)('0' + v % 10);
+  return 2;
+}
+
+// nanos -> ".fff[fff[fff]]" trimmed in 3-digit groups (protojson rule)
+DEV uint32_t put_nanos(uint8_t* o, int32_t nanos) {
+  if (nanos == 0) return 0;
+  uint32_t n = 0;
+  o[n++] = '.';
+  int digits = 9;
+  if (nanos % 1000000 == 0) {
+    digits = 3;
+    nanos /= 1000000;
+  } else if (nanos % 1000 == 0) {
+    digits = 6;
+    nanos /= 1000;
+  }
+  for (int i = digits - 1; i >= 0; --i) {
+    o[n + i] = (uint8_t)('0' + nanos % 10);
+    nanos /= 10;
+  }
+  return n + digits;
+}
+
+DEV bool put_timestamp(DCtx& c, int64_t secs, int32_t nanos) {
+  if (c.opos + 40 > c.ocap) return dfail(c, E_OVERFLOW);
+  uint8_t buf[40];
+  uint32_t n = 0;
+  int64_t days = secs >= 0 ? secs / 86400 : (secs - 86399) / 86400;
+  int64_t rem = secs - days * 86400;
+  int64_t y;
+  int mo, dd;
+  civil_from_days(days, &y, &mo, &dd);
+  buf[n++] = '"';
+  buf[n] = (uint8_t)('0' + (y / 1000) % 10);
+  buf[n + 1] = (uint8_t)('0' + (y / 100) % 10);
+  buf[n + 2] = (uint8_t)('0' + (y / 10) % 10);
+  buf[n + 3] = (uint8_t)('0' + y % 10);
+  n += 4;
+  buf[n++] = '-';
+  n += put2(buf + n, mo);
+  buf[n++] = '-';
+  n += put2(buf + n, dd);
+  buf[n++] = 'T';
+  n += put2(buf + n, (int)(rem / 3600));
+  buf[n++] = ':';
+  n += put2(buf + n, (int)((rem / 60) % 60));
+  buf[n++] = ':';
+  n += put2(buf + n, (int)(rem % 60));
+  n += put_nanos(buf + n, nanos);
+  buf[n++] = 'Z';
+  buf[n++] = '"';
+  if (!c.lane)
+    for (uint32_t i = 0; i < n; ++i) c.out[c.opos + i] = buf[i];
+  c.opos += n;
+  return true;
+}
+
+DEV bool put_duration(DCtx& c, int64_t secs, int32_t nanos) {
+  if (c.opos + 40 > c.ocap) return dfail(c, E_OVERFLOW);
+  uint8_t buf[40];
+  uint32_t n = 0;
+  buf[n++] = '"';
+  bool neg = secs < 0 || nanos < 0;
+  if (neg) buf[n++] = '-';
+  uint64_t s = secs < 0 ? (uint64_t)(-secs) : (uint64_t)secs;
+  int32_t ns = nanos < 0 ? -nanos : nanos;
+  n += u64_to_dec(buf + n, s);
+  n += put_nanos(buf + n, ns);
+  buf[n++] = 's';
+  buf[n++] = '"';
+  if (!c.lane)
+    for (uint32_t i = 0; i < n; ++i) c.out[c.opos + i] = buf[i];
+  c.opos += n;
+  return true;
+}
+
+// ---------------------------------------------------------------------------
+// wire helpers
+// ---------------------------------------------------------------------------
+
+DEV bool read_varint(DCtx& c, uint64_t* v) {
+  if (!get_varint(c.pb, c.len, &c.pos, v)) return dfail(c, E_PARSE);
+  return true;
+}
+
+DEV bool read_fixed32(DCtx& c, uint32_t* v) {
+  if (c.pos + 4 > c.len) return dfail(c, E_PARSE);
+  *v = (uint32_t)c.pb[c.pos] | ((uint32_t)c.pb[c.pos + 1] << 8) |
+       ((uint32_t)c.pb[c.pos + 2] << 16) | ((uint32_t)c.pb[c.pos + 3] << 24);
+  c.pos += 4;
+  return true;
+}
+
+DEV bool read_fixed64(DCtx& c, uint64_t* v) {
+  if (c.pos + 8 > c.len) return dfail(c, E_PARSE);
+  uint64_t r = 0;
+  for (int i = 0; i < 8; ++i) r |= (uint64_t)c.pb[c.pos + i] << (8 * i);
+  *v = r;
+  c.pos += 8;
+  return true;
+}
+
+DEV bool skip_wire(DCtx& c, uint32_t wt) {
+  uint64_t v;
+  switch (wt) {
+    case W_VARINT: return read_varint(c, &v);
+    case W_I64:
+      if (c.pos + 8 > c.len) return dfail(c, E_PARSE);
+      c.pos += 8;
+      return true;
+    case W_LEN: {
+      if (!read_varint(c, &v)) return false;
+      if (c.pos + v > c.len) return dfail(c, E_PARSE);
+      c.pos += (uint32_t)v;
+      return true;
+    }
+    case W_I32:
+      if (c.pos + 4 > c.len) return dfail(c, E_PARSE);
+      c.pos += 4;
+      return true;
+  }
+  return dfail(c, E_PARSE);
+}
+
+DEV const FieldEntry* find_field(DCtx& c, const MsgEntry& m, uint32_t number) {
+  for (int i = 0; i < m.field_count; ++i) {
+    const FieldEntry& f = c.t.fields[m.field_start + i];
+    if (f.number == number) return &f;
+  }
+  return nullptr;
+}
+
+DEV uint32_t expected_wire(const FieldEntry& f) {
+  switch (f.kind) {
+    case K_DOUBLE: case K_FIXED64: case K_SFIXED64: return W_I64;
+    case K_FLOAT: case K_FIXED32: case K_SFIXED32: return W_I32;
+    case K_STRING: case K_BYTES: case K_MESSAGE: return W_LEN;
+    default: return W_VARINT;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// decoder
+// ---------------------------------------------------------------------------
+
+DEV bool decode_message(DCtx& c, int msg_idx, uint32_t end, int depth);
+
+// is this scalar value the proto3 default? (for omit-default emission)
+DEV bool scalar_is_default_varint(const FieldEntry& f, uint64_t v) { return v == 0; }
+
+// emit one scalar/string/bytes value from the wire (no name)
+DEV bool emit_value(DCtx& c, const FieldEntry& f, int depth) {
+  uint64_t v;
+  uint32_t v32;
+  switch (f.kind) {
+    case K_DOUBLE:
+      if (!read_fixed64(c, &v)) return false;
+      return put_double(c, __builtin_bit_cast(double, v), false);
+    case K_FLOAT:
+      if (!read_fixed32(c, &v32)) return false;
+      return put_double(c, (double)__builtin_bit_cast(float, v32), false);
+    case K_INT64:
+    case K_SFIXED64: {
+      int64_t sv;
+      if (f.kind == K_INT64) {
+        if (!read_varint(c, &v)) return false;
+        sv = (int64_t)v;
+      } else {
+        if (!read_fixed64(c, &v)) return false;
+        sv = (int64_t)v;
+      }
+      if (!putc_(c, '"')) return false;
+      if (!put_i64_dec(c, sv)) return false;
+      return putc_(c, '"');
+    }
+    case K_SINT64: {
+      if (!read_varint(c, &v)) return false;
+      if (!putc_(c, '"')) return false;
+      if (!put_i64_dec(c, unzigzag64(v))) return false;
+      return putc_(c, '"');
+    }
+    case K_UINT64:
+    case K_FIXED64: {
+      if (f.kind == K_UINT64) {
+        if (!read_varint(c, &v)) return false;
+      } else {
+        if (!read_fixed64(c, &v)) return false;
+      }
+      if (!putc_(c, '"')) return false;
+      if (!put_u64_dec(c, v)) return false;
+      return putc_(c, '"');
+    }
+    case K_INT32: {
+      if (!read_varint(c, &v)) return false;
+      return put_i64_dec(c, (int64_t)(int32_t)(uint32_t)v);
+    }
+    case K_SINT32: {
+      if (!read_varint(c, &v)) return false;
+      return put_i64_dec(c, (int64_t)(int32_t)((uint32_t)(v >> 1) ^ (uint32_t)(0 - (v & 1))));
+    }
+    case K_SFIXED32: {
+      if (!read_fixed32(c, &v32)) return false;
+      return put_i64_dec(c, (int64_t)(int32_t)v32);
+    }
+    case K_UINT32: {
+      if (!read_varint(c, &v)) return false;
+      return put_u64_dec(c, (uint32_t)v);
+    }
+    case K_FIXED32: {
+      if (!read_fixed32(c, &v32)) return false;
+      return put_u64_dec(c, v32);
+    }
+    case K_BOOL: {
+      if (!read_varint(c, &v)) return false;
+      return v ? puts_(c, "true", 4) : puts_(c, "false", 5);
+    }
+    case K_ENUM: {
+      if (!read_varint(c, &v)) return false;
+      int32_t num = (int32_t)(int64_t)v;
+      const EnumEntry& ee = c.t.enums[f.sub_index];
+      for (int i = 0; i < ee.val_count; ++i) {
+        const EnumValueEntry& ev = c.t.enum_vals[ee.val_start + i];
+        if (ev.number == num)
+          return put_json_string(c, c.t.names + ev.name_off, ev.name_len);
+      }
+      return put_i64_dec(c, num);  // unknown enum value -> number (protojson)
+    }
+    case K_STRING: {
+      if (!read_varint(c, &v)) return false;
+      if (c.pos + v > c.len) return dfail(c, E_PARSE);
+      if (!put_json_string(c, c.pb + c.pos, (uint32_t)v)) return false;
+      c.pos += (uint32_t)v;
+      return true;
+    }
+    case K_BYTES: {
+      if (!read_varint(c, &v)) return false;
+      if (c.pos + v > c.len) return dfail(c, E_PARSE);
+      if (!putc_(c, '"')) return false;
+      if (!put_base64(c, c.pb + c.pos, (uint32_t)v)) return false;
+      c.pos += (uint32_t)v;
+      return putc_(c, '"');
+    }
+    case K_MESSAGE: {
+      if (!read_varint(c, &v)) return false;
+      if (c.pos + v > c.len) return dfail(c, E_PARSE);
+      uint32_t sub_end = c.pos + (uint32_t)v;
+      if (!decode_message(c, f.sub_index, sub_end, depth + 1)) return false;
+      c.pos = sub_end;
+      return true;
+    }
+  }
+  return dfail(c, E_UNSUPPORTED);
+}
+
+// packed array payload
+DEV bool emit_packed(DCtx& c, const FieldEntry& f, uint32_t end, bool* first) {
+  while (c.pos < end) {
+    if (!*first) {
+      if (!putc_(c, ',')) return false;
+    }
+    *first = false;
+    if (!emit_value(c, f, 0)) return false;
+  }
+  return c.pos == end || dfail(c, E_PARSE);
+}
+
+// map entry message {1: key, 2: value} -> `"key": value`
+DEV bool emit_map_entry(DCtx& c, const FieldEntry& f, uint32_t end, int depth) {
+  const MsgEntry& em = c.t.msgs[f.sub_index];
+  const FieldEntry& kf = c.t.fields[em.field_start];
+  const FieldEntry& vf = c.t.fields[em.field_start + 1];
+  // defaults if halves are missing on the wire
+  bool have_key = false, have_val = false;
+  uint32_t key_pos = 0, val_pos = 0;
+  // first pass: locate key/value spans
+  uint32_t save = c.pos;
+  while (c.pos < end) {
+    uint64_t tag;
+    if (!read_varint(c, &tag)) return false;
+    uint32_t num = (uint32_t)(tag >> 3), wt = (uint32_t)(tag & 7);
+    if (num == 1) {
+      have_key = true;
+      key_pos = c.pos;
+    } else if (num == 2) {
+      have_val = true;
+      val_pos = c.pos;
+    }
+    if (!skip_wire(c, wt)) return false;
+  }
+  // key -> JSON object key (always a string)
+  if (!putc_(c, '"')) return false;
+  if (have_key) {
+    uint32_t save2 = c.pos;
+    c.pos = key_pos;
+    uint64_t v;
+    switch (kf.kind) {
+      case K_STRING: {
+        if (!read_varint(c, &v)) return false;
+        if (!put_escaped(c, c.pb + c.pos, (uint32_t)v)) return false;
+        break;
+      }
+      case K_BOOL: {
+        if (!read_varint(c, &v)) return false;
+        if (!(v ? puts_(c, "true", 4) : puts_(c, "false", 5))) return false;
+        break;
+      }
+      case K_SINT64:
+      case K_SINT32: {
+        if (!read_varint(c, &v)) return false;
+        if (!put_i64_dec(c, unzigzag64(v))) return false;
+        break;
+      }
+      case K_UINT64:
+      case K_UINT32: {
+        if (!read_varint(c, &v)) return false;
+        if (!put_u64_dec(c, v)) return false;
+        break;
+      }
+      case K_FIXED64: {
+        if (!read_fixed64(c, &v)) return false;
+        if (!put_u64_dec(c, v)) return false;
+        break;
+      }
+      case K_SFIXED64: {
+        if (!read_fixed64(c, &v)) return false;
+        if (!put_i64_dec(c, (int64_t)v)) return false;
+        break;
+      }
+      case K_FIXED32: {
+        uint32_t v32;
+        if (!read_fixed32(c, &v32)) return false;
+        if (!put_u64_dec(c, v32)) return false;
+        break;
+      }
+      case K_SFIXED32: {
+        uint32_t v32;
+        if (!read_fixed32(c, &v32)) return false;
+        if (!put_i64_dec(c, (int32_t)v32)) return false;
+        break;
+      }
+      default: {  // int32/int64
+        if (!read_varint(c, &v)) return false;
+        if (!put_i64_dec(c, (int64_t)v)) return false;
+      }
+    }
+    c.pos = save2;
+  } else {
+    if (kf.kind != K_STRING && !putc_(c, '0')) return false;
+  }
+  if (!putc_(c, '"')) return false;
+  if (!putc_(c, ':')) return false;
+  // value
+  if (have_val) {
+    uint32_t save2 = c.pos;
+    c.pos = val_pos;
+    if (!emit_value(c, vf, depth)) return false;
+    c.pos = save2;
+  } else {
+    // default value for the value type
+    switch (vf.kind) {
+      case K_STRING: if (!puts_(c, "\"\"", 2)) return false; break;
+      case K_BYTES: if (!puts_(c, "\"\"", 2)) return false; break;
+      case K_BOOL: if (!puts_(c, "false", 5)) return false; break;
+      case K_MESSAGE: if (!puts_(c, "{}", 2)) return false; break;
+      case K_INT64: case K_UINT64: case K_SINT64: case K_FIXED64:
+      case K_SFIXED64:
+        if (!puts_(c, "\"0\"", 3)) return false;
+        break;
+      case K_ENUM: {
+        const EnumEntry& ee = c.t.enums[vf.sub_index];
+        bool done = false;
+        for (int i = 0; i < ee.val_count; ++i) {
+          const EnumValueEntry& ev = c.t.enum_vals[ee.val_start + i];
+          if (ev.number == 0) {
+            if (!put_json_string(c, c.t.names + ev.name_off, ev.name_len))
+              return false;
+            done = true;
+            break;
+          }
+        }
+        if (!done && !putc_(c, '0')) return false;
+        break;
+      }
+      default:
+        if (!putc_(c, '0')) return false;
+    }
+  }
+  c.pos = save;
+  // caller re-skips the entry; restore to start so it can
+  return true;
+}
+
+// decode message payload [c.pos, end) -> JSON (wkt-aware)
+DEV bool decode_message(DCtx& c, int msg_idx, uint32_t end, int depth) {
+  if (depth > 64) return dfail(c, E_LIMIT);
+  const MsgEntry& m = c.t.msgs[msg_idx];
+  // ---- WKTs ----
+  if (m.wkt_kind == WKT_TIMESTAMP || m.wkt_kind == WKT_DURATION) {
+    int64_t secs = 0;
+    int32_t nanos = 0;
+    while (c.pos < end) {
+      uint64_t tag;
+      if (!read_varint(c, &tag)) return false;
+      uint32_t num = (uint32_t)(tag >> 3);
+      if (num == 1 && (tag & 7) == W_VARINT) {
+        uint64_t v;
+        if (!read_varint(c, &v)) return false;
+        secs = (int64_t)v;
+      } else if (num == 2 && (tag & 7) == W_VARINT) {
+        uint64_t v;
+        if (!read_varint(c, &v)) return false;
+        nanos = (int32_t)(int64_t)v;
+      } else {
+        if (!skip_wire(c, (uint32_t)(tag & 7))) return false;
+      }
+    }
+    return m.wkt_kind == WKT_TIMESTAMP ? put_timestamp(c, secs, nanos)
+                                       : put_duration(c, secs, nanos);
+  }
+  if (m.wkt_kind == WKT_WRAPPER) {
+    const FieldEntry& inner = c.t.fields[m.field_start];
+    bool emitted = false;
+    while (c.pos < end) {
+      uint64_t tag;
+      if (!read_varint(c, &tag)) return false;
+      if ((uint32_t)(tag >> 3) == 1) {
+        if (!emit_value(c, inner, depth)) return false;
+        emitted = true;
+      } else {
+        if (!skip_wire(c, (uint32_t)(tag & 7))) return false;
+      }
+    }
+    if (!emitted) {
+      // default inner value
+      switch (inner.kind) {
+        case K_STRING: case K_BYTES: return puts_(c, "\"\"", 2);
+        case K_BOOL: return puts_(c, "false", 5);
+        case K_DOUBLE: case K_FLOAT: return putc_(c, '0');
+        case K_INT64: case K_UINT64: return puts_(c, "\"0\"", 3);
+        default: return putc_(c, '0');
+      }
+    }
+    return true;
+  }
+  if (m.wkt_kind == WKT_VALUE) {
+    // Value oneof; empty Value -> null
+    if (c.pos >= end) return puts_(c, "null", 4);
+    uint64_t tag;
+    if (!read_varint(c, &tag)) return false;
+    uint32_t num = (uint32_t)(tag >> 3);
+    uint64_t v;
+    switch (num) {
+      case 1:
+        if (!read_varint(c, &v)) return false;
+        return puts_(c, "null", 4);
+      case 2:
+        if (!read_fixed64(c, &v)) return false;
+        return put_double(c, __builtin_bit_cast(double, v), false);
+      case 3: {
+        if (!read_varint(c, &v)) return false;
+        if (c.pos + v > c.len) return dfail(c, E_PARSE);
+        if (!put_json_string(c, c.pb + c.pos, (uint32_t)v)) return false;
+        c.pos += (uint32_t)v;
+        return true;
+      }
+      case 4:
+        if (!read_varint(c, &v)) return false;
+        return v ? puts_(c, "true", 4) : puts_(c, "false", 5);
+      case 5: {  // struct_value
+        if (!read_varint(c, &v)) return false;
+        uint32_t sub_end = c.pos + (uint32_t)v;
+        // find the Struct message index via this Value's field table
+        const FieldEntry* sf = find_field(c, m, 5);
+        if (!sf) return dfail(c, E_UNSUPPORTED);
+        if (!decode_message(c, sf->sub_index, sub_end, depth + 1)) return false;
+        c.pos = sub_end;
+        return true;
+      }
+      case 6: {  // list_value
+        if (!read_varint(c, &v)) return false;
+        uint32_t sub_end = c.pos + (uint32_t)v;
+        const FieldEntry* lf = find_field(c, m, 6);
+        if (!lf) return dfail(c, E_UNSUPPORTED);
+        if (!decode_message(c, lf->sub_index, sub_end, depth + 1)) return false;
+        c.pos = sub_end;
+        return true;
+      }
+    }
+    return dfail(c, E_PARSE);
+  }
+  if (m.wkt_kind == WKT_STRUCT) {
+    // map<string, Value> on field 1 -> JSON object
+    if (!putc_(c, '{')) return false;
+    bool first = true;
+    while (c.pos < end) {
+      uint64_t tag;
+      if (!read_varint(c, &tag)) return false;
+      if ((uint32_t)(tag >> 3) != 1 || (tag & 7) != W_LEN) {
+        if (!skip_wire(c, (uint32_t)(tag & 7))) return false;
+        continue;
+      }
+      uint64_t elen;
+      if (!read_varint(c, &elen)) return false;
+      uint32_t eend = c.pos + (uint32_t)elen;
+      if (eend > c.len) return dfail(c, E_PARSE);
+      if (!first && !putc_(c, ',')) return false;
+      first = false;
+      const FieldEntry* f1 = &c.t.fields[m.field_start];  // entries field
+      if (!emit_map_entry(c, *f1, eend, depth)) return false;
+      c.pos = eend;
+    }
+    return putc_(c, '}');
+  }
+  if (m.wkt_kind == WKT_LISTVALUE) {
+    if (!putc_(c, '[')) return false;
+    bool first = true;
+    while (c.pos < end) {
+      uint64_t tag;
+      if (!read_varint(c, &tag)) return false;
+      if ((uint32_t)(tag >> 3) != 1 || (tag & 7) != W_LEN) {
+        if (!skip_wire(c, (uint32_t)(tag & 7))) return false;
+        continue;
+      }
+      uint64_t elen;
+      if (!read_varint(c, &elen)) return false;
+      uint32_t eend = c.pos + (uint32_t)elen;
+      if (eend > c.len) return dfail(c, E_PARSE);
+      if (!first && !putc_(c, ',')) return false;
+      first = false;
+      const FieldEntry* vf = &c.t.fields[m.field_start];
+      const MsgEntry& vm = c.t.msgs[vf->sub_index];
+      (void)vm;
+      if (!decode_message(c, vf->sub_index, eend, depth + 1)) return false;
+      c.pos = eend;
+    }
+    return putc_(c, ']');
+  }
+  if (m.wkt_kind == WKT_ANY) return dfail(c, E_UNSUPPORTED);
+  if (m.wkt_kind == WKT_FIELDMASK) {
+    // repeated string paths -> comma-joined camelCase
+    if (!putc_(c, '"')) return false;
+    bool first = true;
+    while (c.pos < end) {
+      uint64_t tag;
+      if (!read_varint(c, &tag)) return false;
+      if ((uint32_t)(tag >> 3) != 1 || (tag & 7) != W_LEN) {
+        if (!skip_wire(c, (uint32_t)(tag & 7))) return false;
+        continue;
+      }
+      uint64_t slen;
+      if (!read_varint(c, &slen)) return false;
+      if (c.pos + slen > c.len) return dfail(c, E_PARSE);
+      if (!first && !putc_(c, ',')) return false;
+      first = false;
+      // snake -> camel
+      bool up = false;
+      for (uint32_t i = 0; i < (uint32_t)slen; ++i) {
+        uint8_t ch = c.pb[c.pos + i];
+        if (ch == '_') {
+          up = true;
+          continue;
+        }
+        if (up && ch >= 'a' && ch <= 'z') ch -= 32;
+        up = false;
+        if (!putc_(c, ch)) return false;
+      }
+      c.pos += (uint32_t)slen;
+    }
+    return putc_(c, '"');
+  }
+
+  // ---- plain message ----
+  if (!putc_(c, '{')) return false;
+  bool first_member = true;
+  uint32_t prev_number = 0;
+  while (c.pos < end) {
+    uint64_t tag;
+    if (!read_varint(c, &tag)) return false;
+    uint32_t num = (uint32_t)(tag >> 3), wt = (uint32_t)(tag & 7);
+    const FieldEntry* f = find_field(c, m, num);
+    if (f == nullptr) {
+      if (!skip_wire(c, wt)) return false;  // unknown field: dropped
+      continue;
+    }
+    // out-of-order or non-adjacent duplicate wire fields (no standard
+    // serializer produces either; protobuf last-wins semantics would need a
+    // buffering pass) -> host transcodes this request
+    if (num <= prev_number) return dfail(c, E_UNSUPPORTED);
+    prev_number = num;
+    bool repeated = (f->flags & F_REPEATED) != 0;
+    bool is_map = (f->flags & F_MAP) != 0;
+    // default-omission for singular non-presence scalars (proto3 semantics:
+    // protojson prints nothing for unset/default non-presence fields)
+    if (!repeated && !(f->flags & F_HAS_PRESENCE) && wt == W_VARINT) {
+      uint32_t save = c.pos;
+      uint64_t v;
+      if (!read_varint(c, &v)) return false;
+      if (v == 0) continue;  // default -> omit
+      c.pos = save;
+    }
+    if (!repeated && !(f->flags & F_HAS_PRESENCE) && wt == W_LEN &&
+        (f->kind == K_STRING || f->kind == K_BYTES)) {
+      uint32_t save = c.pos;
+      uint64_t v;
+      if (!read_varint(c, &v)) return false;
+      if (v == 0) continue;  // empty string/bytes -> omit
+      c.pos = save;
+    }
+    if (!repeated && !(f->flags & F_HAS_PRESENCE) &&
+        (wt == W_I64 || wt == W_I32)) {
+      uint32_t save = c.pos;
+      uint64_t v = 1;
+      if (wt == W_I64) {
+        if (!read_fixed64(c, &v)) return false;
+      } else {
+        uint32_t v32;
+        if (!read_fixed32(c, &v32)) return false;
+        v = v32;
+      }
+      if (v == 0) continue;  // +0.0 / 0 fixed -> omit
+      c.pos = save;
+    }
+    if (!first_member && !putc_(c, ',')) return false;
+    first_member = false;
+    if (!putc_(c, '"')) return false;
+    if (!puts_(c, (const char*)(c.t.names + f->json_off), f->json_len))
+      return false;
+    if (!putc_(c, '"')) return false;
+    if (!putc_(c, ':')) return false;
+
+    if (is_map) {
+      if (!putc_(c, '{')) return false;
+      bool first = true;
+      // current + adjacent entries of the same field number
+      while (true) {
+        uint64_t elen;
+        if (!read_varint(c, &elen)) return false;
+        uint32_t eend = c.pos + (uint32_t)elen;
+        if (eend > c.len) return dfail(c, E_PARSE);
+        if (!first && !putc_(c, ',')) return false;
+        first = false;
+        if (!emit_map_entry(c, *f, eend, depth)) return false;
+        c.pos = eend;
+        if (c.pos >= end) break;
+        uint32_t save = c.pos;
+        uint64_t ntag;
+        if (!read_varint(c, &ntag)) return false;
+        if ((uint32_t)(ntag >> 3) != num) {
+          c.pos = save;
+          break;
+        }
+      }
+      if (!putc_(c, '}')) return false;
+    } else if (repeated) {
+      if (!putc_(c, '[')) return false;
+      bool first = true;
+      while (true) {
+        if (wt == W_LEN && expected_wire(*f) != W_LEN) {
+          // packed payload
+          uint64_t plen;
+          if (!read_varint(c, &plen)) return false;
+          uint32_t pend = c.pos + (uint32_t)plen;
+          if (pend > c.len) return dfail(c, E_PARSE);
+          if (!emit_packed(c, *f, pend, &first)) return false;
+        } else {
+          if (!first && !putc_(c, ',')) return false;
+          first = false;
+          if (!emit_value(c, *f, depth)) return false;
+        }
+        if (c.pos >= end) break;
+        uint32_t save = c.pos;
+        uint64_t ntag;
+        if (!read_varint(c, &ntag)) return false;
+        if ((uint32_t)(ntag >> 3) != num) {
+          c.pos = save;
+          break;
+        }
+        wt = (uint32_t)(ntag & 7);
+      }
+      if (!putc_(c, ']')) return false;
+    } else {
+      if (wt != expected_wire(*f)) return dfail(c, E_PARSE);
+      if (!emit_value(c, *f, depth)) return false;
+    }
+  }
+  return putc_(c, '}');
+}
+
+// ---------------------------------------------------------------------------
+// kernel
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ void __launch_bounds__(WPB * WAVE) k_pb2json(
+    const uint8_t* __restrict__ resp_bytes, const uint32_t* __restrict__ resp_off,
+    const int32_t* __restrict__ msg_idx_arr,
+    const uint8_t* __restrict__ id_slots, const SlotResult* __restrict__ enc_results,
+    uint8_t* __restrict__ scratch, const uint32_t* __restrict__ scratch_off,
+    uint8_t* __restrict__ final_out, const uint32_t* __restrict__ final_off,
+    DecodeResult* __restrict__ results, const int32_t* __restrict__ skip,
+    Tables t, int n_req, int mode) {
+  int wave_in_block = threadIdx.x / WAVE;
+  int lane = lane_id();
+  for (int req = blockIdx.x * WPB + wave_in_block; req < n_req;
+       req += gridDim.x * WPB) {
+    DecodeResult r;
+    r.status = E_OK;
+    r.out_off = final_off[req];
+    r.out_len = 0;
+    r.pad = 0;
+    if (skip && skip[req]) {
+      // host handles this slot (error / fallback); leave empty
+      if (!lane) results[req] = r;
+      continue;
+    }
+    // ---- phase 1: decode protobuf -> JSON in scratch ----
+    DCtx c;
+    c.pb = resp_bytes + resp_off[req];
+    c.len = resp_off[req + 1] - resp_off[req];
+    c.pos = 0;
+    c.out = scratch + scratch_off[req];
+    c.opos = 0;
+    c.ocap = scratch_off[req + 1] - scratch_off[req];
+    c.t = t;
+    c.status = E_OK;
+    c.lane = lane;
+    const MsgEntry& m = t.msgs[msg_idx_arr[req]];
+    bool top_is_scalar_wkt =
+        m.wkt_kind == WKT_TIMESTAMP || m.wkt_kind == WKT_DURATION ||
+        m.wkt_kind == WKT_WRAPPER || m.wkt_kind == WKT_VALUE ||
+        m.wkt_kind == WKT_FIELDMASK || m.wkt_kind == WKT_LISTVALUE;
+    (void)top_is_scalar_wkt;
+    decode_message(c, msg_idx_arr[req], c.len, 0);
+    if (c.status != E_OK) {
+      r.status = c.status;
+      if (!lane) results[req] = r;
+      continue;
+    }
+    uint32_t json_len = c.opos;
+
+    // ---- phase 2: output ----
+    DCtx o;
+    o.pb = nullptr;
+    o.len = 0;
+    o.pos = 0;
+    o.out = final_out + final_off[req];
+    o.opos = 0;
+    o.ocap = final_off[req + 1] - final_off[req];
+    o.t = t;
+    o.status = E_OK;
+    o.lane = lane;
+    if (mode == 1) {
+      // bare JSON (transcode tests / streaming chunks): copy scratch out
+      if (json_len > o.ocap) {
+        r.status = E_OVERFLOW;
+        if (!lane) results[req] = r;
+        continue;
+      }
+      for (uint32_t i = lane; i < json_len; i += WAVE)
+        o.out[i] = scratch[scratch_off[req] + i];
+      o.opos = json_len;
+    } else {
+      // JSON-RPC result envelope (handler.go:265-270 + TextContent wrap)
+      static const char P1[] = "{\"jsonrpc\":\"2.0\",\"id\":";
+      static const char P2[] =
+          ",\"result\":{\"content\":[{\"type\":\"text\",\"text\":\"";
+      static const char P3[] = "\"}],\"isError\":false}}";
+      bool ok = puts_(o, P1, sizeof(P1) - 1);
+      uint32_t idl = enc_results ? enc_results[req].id_len : 0;
+      if (ok && idl) {
+        if (o.opos + idl > o.ocap) ok = dfail(o, E_OVERFLOW);
+        if (ok) {
+          const uint8_t* idp = id_slots + (size_t)req * ID_SLOT_BYTES;
+          for (uint32_t i = lane; i < idl; i += WAVE) o.out[o.opos + i] = idp[i];
+          o.opos += idl;
+        }
+      } else if (ok) {
+        ok = puts_(o, "null", 4);
+      }
+      if (ok) ok = puts_(o, P2, sizeof(P2) - 1);
+      if (ok) ok = put_escaped(o, scratch + scratch_off[req], json_len);
+      if (ok) ok = puts_(o, P3, sizeof(P3) - 1);
+      if (!ok) {
+        r.status = o.status;
+        if (!lane) results[req] = r;
+        continue;
+      }
+    }
+    r.status = E_OK;
+    r.out_len = o.opos;
+    if (!lane) results[req] = r;
+  }
+}
