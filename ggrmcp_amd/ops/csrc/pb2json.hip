@@ -179,8 +179,45 @@ __constant__ double DPOW10[23] = {1e0,  1e1,  1e2,  1e3,  1e4,  1e5,  1e6,  1e7,
                                   1e8,  1e9,  1e10, 1e11, 1e12, 1e13, 1e14, 1e15,
                                   1e16, 1e17, 1e18, 1e19, 1e20, 1e21, 1e22};
 
-DEV uint32_t dtoa17(uint8_t* out, double d) {
-  // caller guarantees finite, non-negative handled via sign by caller
+// scale d by 10^e (composed; exact when |e|<=22 and mantissa fits)
+DEV double scale10(double d, int e) {
+  while (e > 22) { d *= 1e22; e -= 22; }
+  while (e < -22) { d /= 1e22; e += 22; }
+  if (e >= 0) return d * DPOW10[e];
+  return d / DPOW10[-e];
+}
+
+// digits * 10^(e10 - ndig + 1) as double — the parse-back for verification
+DEV double digits_to_double(uint64_t digits, int ndig, int e10) {
+  return scale10((double)digits, e10 - ndig + 1);
+}
+
+// Generate `prec` significant decimal digits of d (>0, finite); returns the
+// digit block and adjusts *e10_out.  Precision-search wrapper below picks the
+// shortest precision whose parse-back round-trips — matching protojson's
+// shortest-representation output for the common cases.
+DEV uint64_t gen_digits(double d, int prec, int* e10_out) {
+  int e10 = (int)floor(log10(d));
+  double scaled = scale10(d, (prec - 1) - e10);
+  uint64_t digits = (uint64_t)(scaled + 0.5);
+  uint64_t hi = 1;
+  for (int i = 0; i < prec; ++i) hi *= 10;
+  if (digits >= hi) {
+    digits /= 10;
+    e10 += 1;
+  } else if (digits < hi / 10) {
+    // log10 estimate was one high
+    e10 -= 1;
+    scaled = scale10(d, (prec - 1) - e10);
+    digits = (uint64_t)(scaled + 0.5);
+    if (digits >= hi) { digits /= 10; e10 += 1; }
+  }
+  *e10_out = e10;
+  return digits;
+}
+
+DEV uint32_t dtoa17(uint8_t* out, double d, bool as_float) {
+  // caller guarantees finite; d > 0 here (sign handled by caller)
   if (d == 0.0) {
     out[0] = '0';
     return 1;
@@ -189,28 +226,30 @@ DEV uint32_t dtoa17(uint8_t* out, double d) {
   if (d == trunc(d) && d < 9.007199254740992e15) {
     return u64_to_dec(out, (uint64_t)d);
   }
-  // decimal exponent via log10, corrected
-  int e10 = (int)floor(log10(d));
-  // scale to [1e16, 1e17)
-  double scaled = d;
-  int shift = 16 - e10;
-  {
-    int s = shift;
-    while (s > 22) { scaled *= 1e22; s -= 22; }
-    while (s < -22) { scaled /= 1e22; s += 22; }
-    if (s >= 0) scaled *= DPOW10[s];
-    else scaled /= DPOW10[-s];
+  // shortest precision whose parse-back round-trips
+  int p_lo = as_float ? 6 : 15;
+  int p_hi = as_float ? 9 : 17;
+  int e10 = 0;
+  uint64_t digits = 0;
+  int prec = p_hi;
+  for (int p = p_lo; p <= p_hi; ++p) {
+    int e;
+    uint64_t dg = gen_digits(d, p, &e);
+    double back = digits_to_double(dg, p, e);
+    bool ok = as_float ? ((float)back == (float)d) : (back == d);
+    if (ok || p == p_hi) {
+      digits = dg;
+      e10 = e;
+      prec = p;
+      break;
+    }
   }
-  if (scaled >= 1e17) { scaled /= 10.0; e10 += 1; }
-  if (scaled < 1e16) { scaled *= 10.0; e10 -= 1; }
-  uint64_t digits = (uint64_t)(scaled + 0.5);
-  if (digits >= 100000000000000000ull) { digits /= 10; e10 += 1; }
   uint8_t dig[17];
-  for (int i = 16; i >= 0; --i) {
+  for (int i = prec - 1; i >= 0; --i) {
     dig[i] = (uint8_t)('0' + digits % 10);
     digits /= 10;
   }
-  int ndig = 17;
+  int ndig = prec;
   while (ndig > 1 && dig[ndig - 1] == '0') --ndig;
   uint32_t o = 0;
   if (e10 >= -6 && e10 <= 20) {
@@ -264,9 +303,9 @@ DEV bool put_double(DCtx& c, double d, bool as_float) {
   uint8_t buf[32];
   if (d < 0 || (d == 0.0 && signbit(d))) {
     buf[0] = '-';
-    n = 1 + dtoa17(buf + 1, -d);
+    n = 1 + dtoa17(buf + 1, -d, as_float);
   } else {
-    n = dtoa17(buf, d);
+    n = dtoa17(buf, d, as_float);
   }
   if (!c.lane)
     for (uint32_t i = 0; i < n; ++i) c.out[c.opos + i] = buf[i];
@@ -456,7 +495,7 @@ DEV bool emit_value(DCtx& c, const FieldEntry& f, int depth) {
       return put_double(c, __builtin_bit_cast(double, v), false);
     case K_FLOAT:
       if (!read_fixed32(c, &v32)) return false;
-      return put_double(c, (double)__builtin_bit_cast(float, v32), false);
+      return put_double(c, (double)__builtin_bit_cast(float, v32), true);
     case K_INT64:
     case K_SFIXED64: {
       int64_t sv;
