@@ -54,17 +54,26 @@ def parse_args():
     ap.add_argument("--payload-bytes", type=int, default=0)
     ap.add_argument("--invoke-workers", type=int, default=64)
     ap.add_argument("--backend-workers", type=int, default=32)
+    ap.add_argument("--transport", default="native", choices=["native", "grpcio"],
+                    help="wire invoker: native C++ h2 client or grpcio threads")
+    ap.add_argument("--backend", default="native", choices=["native", "python"],
+                    help="local bench backend implementation")
+    ap.add_argument("--connections", type=int, default=8,
+                    help="native transport connections per backend")
     return ap.parse_args()
 
 
-def start_backend(rank: int, workers: int):
+def start_backend(rank: int, workers: int, native: bool):
     """Backend in its own process over a unix socket (no shared GIL)."""
     sock = os.path.join(tempfile.gettempdir(), f"ggrmcp_bench_{os.getpid()}_{rank}.sock")
     if os.path.exists(sock):
         os.unlink(sock)
+    cmd = [sys.executable, "-m", "examples.bench_backend", "--uds", sock,
+           "--workers", str(workers)]
+    if native:
+        cmd.append("--native")
     proc = subprocess.Popen(
-        [sys.executable, "-m", "examples.bench_backend", "--uds", sock,
-         "--workers", str(workers)],
+        cmd,
         stdout=subprocess.PIPE, stderr=subprocess.DEVNULL,
         cwd=str(Path(__file__).resolve().parent), text=True,
     )
@@ -109,14 +118,37 @@ def main() -> None:
             torch.cuda.set_device(local_rank)
         dist.init_process_group(backend=backend)
 
-    backend_proc, sock = start_backend(rank, args.backend_workers)
+    native_backend = args.backend == "native"
+    backend_proc, sock = start_backend(rank, args.backend_workers, native_backend)
     try:
         cfg = Config.default()
         cfg.grpc.uds = sock
         cfg.gpu.enabled = use_gpu
         discoverer = ServiceDiscoverer(cfg)
-        discoverer.connect(timeout_s=30)
-        discoverer.discover()
+        if native_backend:
+            # the native backend has no reflection service: discover from the
+            # in-repo descriptor blob (the descriptor-set path, loader.go route)
+            from examples.protos import ALL_FDPS
+            from ggrmcp_amd.utils.synthetic import synthetic_fdp
+            from google.protobuf import descriptor_pb2
+
+            fdset = descriptor_pb2.FileDescriptorSet()
+            fdset.file.extend(ALL_FDPS + [synthetic_fdp()])
+            discoverer.load_descriptor_blob(fdset.SerializeToString())
+            # a grpcio channel for the fallback/CPU paths (grpcio client
+            # interoperates with the nghttp2 server)
+            discoverer.connections[0].connect(timeout_s=15)
+        else:
+            discoverer.connect(timeout_s=30)
+            discoverer.discover()
+
+        wire_clients = None
+        if args.transport == "native":
+            from ggrmcp_amd.backend.native_invoker import NativeWireClient
+
+            wire_clients = [
+                NativeWireClient(f"unix:{sock}", connections=args.connections)
+            ]
 
         bodies = make_bodies(args.config, args.batch, args.payload_bytes, seed=1234 + rank)
 
@@ -124,7 +156,8 @@ def main() -> None:
             from ggrmcp_amd.engine.batch import GpuPipeline
 
             pipeline = GpuPipeline(discoverer, cfg, device=local_rank,
-                                   invoke_workers=args.invoke_workers)
+                                   invoke_workers=args.invoke_workers,
+                                   wire_clients=wire_clients)
 
             def step():
                 out = pipeline.process_batch(bodies, timeout_s=30.0)
@@ -225,7 +258,8 @@ def main() -> None:
                     "sessions": args.batch,
                     "payload_bytes": payload_size,
                     "p50_rtt_ms": round(p50_ms, 3),
-                    "backend": "local grpc (uds, separate process)",
+                    "backend": f"local grpc over uds, separate process ({args.backend})",
+                    "transport": args.transport,
                 },
             }
             if use_gpu:

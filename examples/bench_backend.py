@@ -84,14 +84,40 @@ def serve(target: str = "127.0.0.1:0", max_workers: int = 32):
     return server, bound
 
 
+def serve_native(target: str):
+    """C++ nghttp2 backend (ops/csrc/h2grpc.cpp H2Server): native handlers
+    for SayHello and echo routes; no Python in the serving path."""
+    from ggrmcp_amd.backend.native_invoker import load_module
+
+    mod = load_module()
+    srv = mod.Server(target)
+    srv.add_route("/hello.HelloService/SayHello", "hello")
+    srv.add_route("/bench.EchoService/Echo", "echo")
+    srv.add_route("/complex.DocumentService/PutDocument", "echo")
+    srv.add_route("/complex.NodeService/Echo", "echo")
+    bound = srv.start()
+    return srv, bound
+
+
 def main() -> None:
     ap = argparse.ArgumentParser()
     ap.add_argument("--host", default="127.0.0.1")
     ap.add_argument("--port", type=int, default=0)
     ap.add_argument("--uds", default="")
     ap.add_argument("--workers", type=int, default=32)
+    ap.add_argument("--native", action="store_true",
+                    help="serve with the C++ nghttp2 backend instead of grpcio")
     args = ap.parse_args()
     target = f"unix:{args.uds}" if args.uds else f"{args.host}:{args.port}"
+    if args.native:
+        server, bound = serve_native(target)
+        print(f"READY {bound}", flush=True)
+        try:
+            while True:
+                time.sleep(3600)
+        except KeyboardInterrupt:
+            server.stop()
+        return
     server, bound = serve(target, args.workers)
     print(f"READY {bound}", flush=True)
     try:

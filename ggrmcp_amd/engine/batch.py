@@ -230,11 +230,13 @@ class GpuPipeline:
     """Full tools/call pipeline over one GpuEngine + a ServiceDiscoverer."""
 
     def __init__(self, discoverer, config: Optional[Config] = None, device: int = 0,
-                 invoke_workers: int = 64) -> None:
+                 invoke_workers: int = 64, wire_clients=None) -> None:
         self.discoverer = discoverer
         self.config = config or Config.default()
         self.engine = GpuEngine(discoverer.tools, self.config, device)
         self.cpu = CpuTranscoder()
+        # native C++ h2 transport per backend index (None -> grpcio threads)
+        self.wire_clients = wire_clients
         self._invoke_pool = ThreadPoolExecutor(
             max_workers=invoke_workers, thread_name_prefix="ginvoke"
         )
@@ -274,23 +276,46 @@ class GpuPipeline:
         out_idx = np.zeros(n, dtype=np.int32)
 
         t0 = time.perf_counter_ns()
-        futures = {}
-        for i in range(n):
-            if enc[i]["status"] != E_OK:
-                continue
-            if enc[i]["flags"] & SR_SERVER_STREAMING:
-                continue  # streaming handled below via host assembly
-            mi = self._mi_by_idx[enc[i]["tool_idx"]]
-            out_idx[i] = self._out_msg_idx[enc[i]["tool_idx"]]
-            hdr = headers[i] if headers else None
-            futures[i] = self._invoke_pool.submit(
-                self.discoverer.invoke_wire, mi, pbs[i], hdr, timeout_s
-            )
-        for i, fut in futures.items():
-            try:
-                resp_wire[i] = fut.result()
-            except Exception as e:
-                rpc_error[i] = e
+        timeout = timeout_s if timeout_s is not None else self.config.grpc.request_timeout_s
+        if self.wire_clients:
+            # native transport: one blocking batch call (GIL released inside)
+            slots, paths, payloads, metas = [], [], [], []
+            for i in range(n):
+                if enc[i]["status"] != E_OK or enc[i]["flags"] & SR_SERVER_STREAMING:
+                    continue
+                tool = enc[i]["tool_idx"]
+                mi = self._mi_by_idx[tool]
+                out_idx[i] = self._out_msg_idx[tool]
+                slots.append(i)
+                paths.append(mi.full_method_path)
+                payloads.append(pbs[i])
+                hdr = headers[i] if headers else None
+                metas.append(list(hdr.items()) if hdr else [])
+            if slots:
+                res = self.wire_clients[0].invoke_batch(paths, payloads, timeout, metas)
+                for i, r in zip(slots, res):
+                    if isinstance(r, Exception):
+                        rpc_error[i] = r
+                    else:
+                        resp_wire[i] = r
+        else:
+            futures = {}
+            for i in range(n):
+                if enc[i]["status"] != E_OK:
+                    continue
+                if enc[i]["flags"] & SR_SERVER_STREAMING:
+                    continue  # streaming handled below via host assembly
+                mi = self._mi_by_idx[enc[i]["tool_idx"]]
+                out_idx[i] = self._out_msg_idx[enc[i]["tool_idx"]]
+                hdr = headers[i] if headers else None
+                futures[i] = self._invoke_pool.submit(
+                    self.discoverer.invoke_wire, mi, pbs[i], hdr, timeout_s
+                )
+            for i, fut in futures.items():
+                try:
+                    resp_wire[i] = fut.result()
+                except Exception as e:
+                    rpc_error[i] = e
         st.invoke_ns += time.perf_counter_ns() - t0
 
         dec, finals = self.engine.decode_batch(resp_wire, out_idx, mode=0)
@@ -326,11 +351,10 @@ class GpuPipeline:
             return self._cpu_full(body, rid, hdr, timeout_s)
 
         if status == E_OK and rpc_err is not None:
-            # gRPC failure -> isError tool result (handler.go:252-259)
-            import grpc
-
-            if isinstance(rpc_err, grpc.RpcError):
-                code = rpc_err.code().name if hasattr(rpc_err, "code") else "UNKNOWN"
+            # gRPC failure -> isError tool result (handler.go:252-259);
+            # NativeRpcError duck-types grpc.RpcError (code().name/details())
+            if hasattr(rpc_err, "code") and callable(rpc_err.code):
+                code = rpc_err.code().name
                 detail = rpc_err.details() if hasattr(rpc_err, "details") else str(rpc_err)
                 text = f"gRPC error {code}: {detail}"
             else:

@@ -18,51 +18,69 @@ from pathlib import Path
 OPS_DIR = Path(__file__).resolve().parent
 CSRC = OPS_DIR / "csrc"
 SO_PATH = OPS_DIR / "_jsonproto.so"
+H2_SO_PATH = OPS_DIR / "_h2grpc.so"
 
 GFX_ARCH = os.environ.get("GGRMCP_GFX_ARCH", "gfx950")
+NGHTTP2_INCLUDE = os.environ.get("GGRMCP_NGHTTP2_INCLUDE", "/opt/conda/include")
 
 
-def source_files():
-    return sorted(CSRC.glob("*.cpp")) + sorted(CSRC.glob("*.hip")) + sorted(
-        CSRC.glob("*.h")
-    )
-
-
-def needs_build() -> bool:
-    if not SO_PATH.exists():
+def _needs(so: Path, sources) -> bool:
+    if not so.exists():
         return True
-    so_mtime = SO_PATH.stat().st_mtime
-    return any(f.stat().st_mtime > so_mtime for f in source_files())
+    so_mtime = so.stat().st_mtime
+    return any(f.stat().st_mtime > so_mtime for f in sources)
 
 
-def build(verbose: bool = True, force: bool = False) -> Path:
-    if not force and not needs_build():
-        return SO_PATH
+def _common_flags():
     import pybind11
 
-    hipcc = os.environ.get("HIPCC", "hipcc")
     py_include = sysconfig.get_paths()["include"]
-    cmd = [
-        hipcc,
-        "-x", "hip",
-        str(CSRC / "engine.cpp"),
-        f"--offload-arch={GFX_ARCH}",
-        "-O3",
-        "-std=c++17",
-        "-fPIC",
-        "-shared",
-        "-fvisibility=hidden",
-        f"-I{pybind11.get_include()}",
-        f"-I{py_include}",
-        f"-I{CSRC}",
-        "-o", str(SO_PATH),
+    return [
+        "-O3", "-std=c++17", "-fPIC", "-shared", "-fvisibility=hidden",
+        f"-I{pybind11.get_include()}", f"-I{py_include}", f"-I{CSRC}",
     ]
+
+
+def build_jsonproto(verbose: bool = True, force: bool = False) -> Path:
+    sources = [CSRC / "engine.cpp", CSRC / "json2pb.hip", CSRC / "pb2json.hip",
+               CSRC / "common.h"]
+    if not force and not _needs(SO_PATH, sources):
+        return SO_PATH
+    hipcc = os.environ.get("HIPCC", "hipcc")
+    cmd = [hipcc, "-x", "hip", str(CSRC / "engine.cpp"),
+           f"--offload-arch={GFX_ARCH}"] + _common_flags() + ["-o", str(SO_PATH)]
     if verbose:
         print("[ggrmcp-amd build]", " ".join(cmd), file=sys.stderr, flush=True)
     subprocess.run(cmd, check=True)
     return SO_PATH
 
 
+def build_h2grpc(verbose: bool = True, force: bool = False) -> Path:
+    sources = [CSRC / "h2grpc.cpp"]
+    if not force and not _needs(H2_SO_PATH, sources):
+        return H2_SO_PATH
+    cxx = os.environ.get("CXX", "g++")
+    cmd = [cxx, str(CSRC / "h2grpc.cpp")] + _common_flags() + [
+        f"-I{NGHTTP2_INCLUDE}",
+        "-l:libnghttp2.so.14",
+        "-pthread",
+        "-o", str(H2_SO_PATH),
+    ]
+    if verbose:
+        print("[ggrmcp-amd build]", " ".join(cmd), file=sys.stderr, flush=True)
+    subprocess.run(cmd, check=True)
+    return H2_SO_PATH
+
+
+def build(verbose: bool = True, force: bool = False) -> Path:
+    build_h2grpc(verbose, force)
+    return build_jsonproto(verbose, force)
+
+
+def needs_build() -> bool:
+    return _needs(SO_PATH, list(CSRC.glob("*")))
+
+
 if __name__ == "__main__":
     build(force="--force" in sys.argv)
-    print(SO_PATH)
+    print(SO_PATH, H2_SO_PATH)
