@@ -124,6 +124,12 @@ def main() -> None:
         cfg = Config.default()
         cfg.grpc.uds = sock
         cfg.gpu.enabled = use_gpu
+        if args.config == "wide64":
+            # 64 KB payloads need bigger arenas; cap the default batch too
+            cfg.gpu.pinned_pool_bytes = 2 * 1024 * 1024 * 1024
+            cfg.gpu.device_pool_bytes = 4 * 1024 * 1024 * 1024
+            if args.batch == 1024:
+                args.batch = 256
         discoverer = ServiceDiscoverer(cfg)
         if native_backend:
             # the native backend has no reflection service: discover from the

@@ -1,0 +1,117 @@
+"""Full GPU pipeline tests (gfx950): process_batch end-to-end over the
+native transport + native backend, including error slots and fallbacks."""
+
+import json
+import random
+
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def pipeline_env():
+    from google.protobuf import descriptor_pb2
+
+    from examples.protos import ALL_FDPS
+    from ggrmcp_amd.backend.discovery import ServiceDiscoverer
+    from ggrmcp_amd.backend.native_invoker import NativeWireClient, load_module
+    from ggrmcp_amd.config import Config
+    from ggrmcp_amd.engine.batch import GpuPipeline
+    from ggrmcp_amd.utils.synthetic import synthetic_fdp
+
+    mod = load_module()
+    srv = mod.Server("127.0.0.1:0")
+    srv.add_route("/hello.HelloService/SayHello", "hello")
+    srv.add_route("/bench.EchoService/Echo", "echo")
+    bound = srv.start()
+
+    cfg = Config.default()
+    host, _, port = bound.rpartition(":")
+    cfg.grpc.host, cfg.grpc.port = host, int(port)
+    d = ServiceDiscoverer(cfg)
+    fdset = descriptor_pb2.FileDescriptorSet()
+    fdset.file.extend(ALL_FDPS + [synthetic_fdp()])
+    d.load_descriptor_blob(fdset.SerializeToString())
+    d.connections[0].connect(timeout_s=15)
+    wire = NativeWireClient(bound, connections=2)
+    pipeline = GpuPipeline(d, cfg, device=0, wire_clients=[wire])
+    yield pipeline
+    wire.close()
+    d.close()
+    srv.stop()
+
+
+def _body(tool, args, rid):
+    return json.dumps(
+        {"jsonrpc": "2.0", "id": rid, "method": "tools/call",
+         "params": {"name": tool, "arguments": args}}
+    ).encode()
+
+
+def test_process_batch_roundtrip(pipeline_env):
+    pipeline = pipeline_env
+    bodies = [
+        _body("hello_helloservice_sayhello", {"name": f"u{i}"}, i) for i in range(32)
+    ]
+    out = pipeline.process_batch(bodies, timeout_s=15.0)
+    assert len(out) == 32
+    for i, raw in enumerate(out):
+        resp = json.loads(raw)
+        assert resp["id"] == i
+        assert resp["result"]["isError"] is False
+        inner = json.loads(resp["result"]["content"][0]["text"])
+        assert inner == {"message": f"Hello, u{i}!"}
+    assert pipeline.engine.stats.gpu_ok >= 32
+
+
+def test_process_batch_mixed_errors(pipeline_env):
+    pipeline = pipeline_env
+    bodies = [
+        _body("hello_helloservice_sayhello", {"name": "ok"}, 1),
+        _body("hello_helloservice_sayhello", {"name": "error"}, 2),  # gRPC error
+        _body("missing_tool", {}, 3),                                # -32601
+        b"{broken json",                                             # -32700
+        json.dumps({"jsonrpc": "2.0", "id": 5, "method": "tools/list"}).encode(),
+        _body("hello_helloservice_sayhello", {"nope": 1}, 6),        # -32602
+    ]
+    out = pipeline.process_batch(bodies, timeout_s=15.0)
+    r = [json.loads(x) for x in out]
+    assert r[0]["result"]["isError"] is False
+    assert r[1]["result"]["isError"] is True
+    assert "INVALID_ARGUMENT" in r[1]["result"]["content"][0]["text"]
+    assert r[2]["error"]["code"] == -32601
+    assert r[3]["error"]["code"] == -32700
+    assert r[4]["error"]["code"] == -32601  # tools/list not on batch path
+    assert r[5]["error"]["code"] == -32602
+
+
+def test_process_batch_wide64(pipeline_env):
+    from ggrmcp_amd.utils.synthetic import wide_payload
+
+    pipeline = pipeline_env
+    rng = random.Random(3)
+    bodies = [
+        _body("bench_echoservice_echo", wide_payload(rng), i) for i in range(16)
+    ]
+    out = pipeline.process_batch(bodies, timeout_s=15.0)
+    for i, raw in enumerate(out):
+        resp = json.loads(raw)
+        assert resp["result"]["isError"] is False, resp
+        inner = json.loads(resp["result"]["content"][0]["text"])
+        assert inner["nested"]["value"] == "42"
+
+
+def test_process_batch_ids_preserved(pipeline_env):
+    pipeline = pipeline_env
+    bodies = [
+        _body("hello_helloservice_sayhello", {"name": "a"}, "string-id"),
+        _body("hello_helloservice_sayhello", {"name": "b"}, 3.5),
+        json.dumps({"jsonrpc": "2.0", "id": None, "method": "tools/call",
+                    "params": {"name": "hello_helloservice_sayhello",
+                               "arguments": {"name": "c"}}}).encode(),
+    ]
+    out = pipeline.process_batch(bodies, timeout_s=15.0)
+    assert json.loads(out[0])["id"] == "string-id"
+    assert json.loads(out[1])["id"] == 3.5
+    assert json.loads(out[2])["id"] is None
