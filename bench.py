@@ -131,22 +131,39 @@ def main() -> None:
             if args.batch == 1024:
                 args.batch = 256
         discoverer = ServiceDiscoverer(cfg)
-        if native_backend:
-            # the native backend has no reflection service: discover from the
-            # in-repo descriptor blob (the descriptor-set path, loader.go route)
-            from examples.protos import ALL_FDPS
-            from ggrmcp_amd.utils.synthetic import synthetic_fdp
-            from google.protobuf import descriptor_pb2
+        shard_group = None
+        if dist is not None:
+            from ggrmcp_amd.parallel.dist import ShardGroup
 
-            fdset = descriptor_pb2.FileDescriptorSet()
-            fdset.file.extend(ALL_FDPS + [synthetic_fdp()])
-            discoverer.load_descriptor_blob(fdset.SerializeToString())
+            shard_group = ShardGroup.attach(
+                dist, device=local_rank if use_gpu else None
+            )
+        if native_backend:
+            # the native backend has no reflection service: rank 0 builds the
+            # descriptor blob (the descriptor-set path, loader.go route) and
+            # broadcasts it to the other shards over RCCL (parallel/dist.py)
+            if rank == 0 or shard_group is None:
+                from examples.protos import ALL_FDPS
+                from ggrmcp_amd.utils.synthetic import synthetic_fdp
+                from google.protobuf import descriptor_pb2
+
+                fdset = descriptor_pb2.FileDescriptorSet()
+                fdset.file.extend(ALL_FDPS + [synthetic_fdp()])
+                discoverer.load_descriptor_blob(fdset.SerializeToString())
+            if shard_group is not None:
+                from ggrmcp_amd.parallel.dist import sync_discovery
+
+                sync_discovery(discoverer, shard_group, src=0)
             # a grpcio channel for the fallback/CPU paths (grpcio client
             # interoperates with the nghttp2 server)
             discoverer.connections[0].connect(timeout_s=15)
         else:
             discoverer.connect(timeout_s=30)
             discoverer.discover()
+            if shard_group is not None:
+                from ggrmcp_amd.parallel.dist import sync_discovery
+
+                sync_discovery(discoverer, shard_group, src=0)
 
         wire_clients = None
         if args.transport == "native":
