@@ -255,6 +255,27 @@ class ServiceDiscoverer:
             for chunk in callable_(wire, metadata=metadata or None, timeout=timeout):
                 yield self.decode_response(mi, chunk)
 
+    def invoke_streaming_wire(
+        self,
+        mi: MethodInfo,
+        request_wire: bytes,
+        headers: Optional[Mapping[str, str]] = None,
+        timeout_s: Optional[float] = None,
+    ) -> List[bytes]:
+        """Server-streaming invoke returning raw wire chunks — the host I/O
+        stage between the GPU encode kernel and a batched GPU decode of all
+        stream messages (BASELINE config 4: per-stream GPU decode batches)."""
+        channel = self.connections[mi.backend_index].channel()
+        callable_ = channel.unary_stream(
+            mi.full_method_path,
+            request_serializer=lambda b: b,
+            response_deserializer=lambda b: b,
+        )
+        metadata = tuple((k.lower(), v) for k, v in (headers or {}).items())
+        timeout = timeout_s if timeout_s is not None else self.config.grpc.request_timeout_s
+        with self._count():
+            return list(callable_(request_wire, metadata=metadata or None, timeout=timeout))
+
     def _count(self):
         disc = self
 

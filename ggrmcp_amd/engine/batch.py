@@ -275,24 +275,48 @@ class GpuPipeline:
         rpc_error: List[Optional[Exception]] = [None] * n
         out_idx = np.zeros(n, dtype=np.int32)
 
+        # server-streaming slots: fan out unary->stream invokes concurrently;
+        # their wire chunks decode in one extra GPU batch below (config 4)
+        stream_futs: Dict[int, Any] = {}
+        timeout0 = timeout_s if timeout_s is not None else self.config.grpc.request_timeout_s
+        for i in range(n):
+            if enc[i]["status"] == E_OK and enc[i]["flags"] & SR_SERVER_STREAMING:
+                mi = self._mi_by_idx[enc[i]["tool_idx"]]
+                hdr = headers[i] if headers else None
+                stream_futs[i] = self._invoke_pool.submit(
+                    self.discoverer.invoke_streaming_wire, mi, pbs[i], hdr, timeout0
+                )
+
         t0 = time.perf_counter_ns()
         timeout = timeout_s if timeout_s is not None else self.config.grpc.request_timeout_s
         if self.wire_clients:
-            # native transport: one blocking batch call (GIL released inside)
-            slots, paths, payloads, metas = [], [], [], []
+            # native transport: one blocking batch call per backend (GIL
+            # released inside); multiple backends dispatch concurrently
+            # (centralized-gateway mode, README.md:129-141)
+            per_be: Dict[int, List[Any]] = {}
             for i in range(n):
                 if enc[i]["status"] != E_OK or enc[i]["flags"] & SR_SERVER_STREAMING:
                     continue
                 tool = enc[i]["tool_idx"]
                 mi = self._mi_by_idx[tool]
                 out_idx[i] = self._out_msg_idx[tool]
-                slots.append(i)
-                paths.append(mi.full_method_path)
-                payloads.append(pbs[i])
                 hdr = headers[i] if headers else None
-                metas.append(list(hdr.items()) if hdr else [])
-            if slots:
-                res = self.wire_clients[0].invoke_batch(paths, payloads, timeout, metas)
+                be = mi.backend_index if mi.backend_index < len(self.wire_clients) else 0
+                g = per_be.setdefault(be, [[], [], [], []])
+                g[0].append(i)
+                g[1].append(mi.full_method_path)
+                g[2].append(pbs[i])
+                g[3].append(list(hdr.items()) if hdr else [])
+
+            def run_backend(be, g):
+                return g[0], self.wire_clients[be].invoke_batch(g[1], g[2], timeout, g[3])
+
+            batch_futs = [
+                self._invoke_pool.submit(run_backend, be, g)
+                for be, g in per_be.items()
+            ]
+            for fut in batch_futs:
+                slots, res = fut.result()
                 for i, r in zip(slots, res):
                     if isinstance(r, Exception):
                         rpc_error[i] = r
@@ -320,6 +344,12 @@ class GpuPipeline:
 
         dec, finals = self.engine.decode_batch(resp_wire, out_idx, mode=0)
 
+        # streaming: gather chunk lists, decode ALL chunks of ALL streams in
+        # one value-mode GPU batch, assemble envelopes host-side
+        stream_out: Dict[int, bytes] = {}
+        if stream_futs:
+            stream_out = self._decode_streams(stream_futs, enc, bodies)
+
         # assemble the batch: GPU envelopes where OK, host for the rest
         out: List[bytes] = []
         for i in range(n):
@@ -327,9 +357,79 @@ class GpuPipeline:
                 st.gpu_ok += 1
                 out.append(finals[i])
                 continue
+            if i in stream_out:
+                out.append(stream_out[i])
+                continue
             out.append(self._host_slot(bodies[i], enc[i], dec[i] if resp_wire[i] is not None else None,
                                        resp_wire[i], rpc_error[i], headers[i] if headers else None,
                                        timeout_s))
+        return out
+
+    def _decode_streams(self, stream_futs, enc, bodies) -> Dict[int, bytes]:
+        """Batch-decode every stream chunk on the GPU (mode 1) and wrap each
+        stream's chunks as the ToolCallResult content list (the capability
+        the reference rejects outright, discovery.go:354-356)."""
+        st = self.engine.stats
+        chunks_by_slot: Dict[int, List[bytes]] = {}
+        errors: Dict[int, Exception] = {}
+        for i, fut in stream_futs.items():
+            try:
+                chunks_by_slot[i] = fut.result()
+            except Exception as e:
+                errors[i] = e
+        flat: List[bytes] = []
+        flat_idx: List[int] = []
+        spans: Dict[int, Tuple[int, int]] = {}
+        for i, chunks in chunks_by_slot.items():
+            tool = int(enc[i]["tool_idx"])
+            spans[i] = (len(flat), len(chunks))
+            flat.extend(chunks)
+            flat_idx.extend([int(self._out_msg_idx[tool])] * len(chunks))
+        jsons: List[Optional[bytes]] = []
+        if flat:
+            _, jsons = self.engine.decode_batch(flat, flat_idx, mode=1)
+        out: Dict[int, bytes] = {}
+        for i in stream_futs:
+            rid, _ = self._extract_id(bodies[i])
+            if i in errors:
+                e = errors[i]
+                if hasattr(e, "code") and callable(e.code):
+                    text = f"gRPC error {e.code().name}: {e.details() if hasattr(e, 'details') else e}"
+                else:
+                    text = str(e)
+                st.errors += 1
+                result = mcp.ToolCallResult(content=[mcp.TextContent(text)], is_error=True)
+                resp = mcp.JSONRPCResponse(id=rid, result=result.to_dict())
+                out[i] = json.dumps(resp.to_dict(), ensure_ascii=False).encode()
+                continue
+            start, count = spans[i]
+            texts: List[str] = []
+            ok = True
+            mi = self._mi_by_idx[int(enc[i]["tool_idx"])]
+            for k in range(start, start + count):
+                if jsons[k] is not None:
+                    texts.append(jsons[k].decode())
+                else:
+                    # per-chunk CPU fallback (counted, never silent)
+                    st.host_fallbacks += 1
+                    try:
+                        texts.append(self.cpu.pb_to_json(mi.output_descriptor, flat[k]))
+                    except Exception as e:
+                        st.errors += 1
+                        resp = mcp.JSONRPCResponse(
+                            id=rid, error=mcp.RPCError(mcp.INTERNAL_ERROR, str(e)[:512])
+                        )
+                        out[i] = json.dumps(resp.to_dict(), ensure_ascii=False).encode()
+                        ok = False
+                        break
+            if not ok:
+                continue
+            st.gpu_ok += 1
+            result = mcp.ToolCallResult(
+                content=[mcp.TextContent(t) for t in texts], is_error=False
+            )
+            resp = mcp.JSONRPCResponse(id=rid, result=result.to_dict())
+            out[i] = json.dumps(resp.to_dict(), ensure_ascii=False).encode()
         return out
 
     # ---- host handling of non-GPU slots ------------------------------------

@@ -115,3 +115,54 @@ def test_process_batch_ids_preserved(pipeline_env):
     assert json.loads(out[0])["id"] == "string-id"
     assert json.loads(out[1])["id"] == 3.5
     assert json.loads(out[2])["id"] is None
+
+
+@pytest.fixture(scope="module")
+def grpcio_pipeline_env():
+    """Pipeline over the python grpcio backend (has server-streaming)."""
+    from google.protobuf import descriptor_pb2
+
+    from examples.bench_backend import serve
+    from examples.protos import ALL_FDPS
+    from ggrmcp_amd.backend.discovery import ServiceDiscoverer
+    from ggrmcp_amd.config import Config
+    from ggrmcp_amd.engine.batch import GpuPipeline
+    from ggrmcp_amd.utils.synthetic import synthetic_fdp
+
+    srv, bound = serve("127.0.0.1:0")
+    cfg = Config.default()
+    host, _, port = bound.rpartition(":")
+    cfg.grpc.host, cfg.grpc.port = host, int(port)
+    d = ServiceDiscoverer(cfg)
+    fdset = descriptor_pb2.FileDescriptorSet()
+    fdset.file.extend(ALL_FDPS + [synthetic_fdp()])
+    d.load_descriptor_blob(fdset.SerializeToString())
+    d.connections[0].connect(timeout_s=15)
+    pipeline = GpuPipeline(d, cfg, device=0)
+    yield pipeline
+    d.close()
+    srv.stop(grace=None)
+
+
+def test_process_batch_streaming_gpu_decode(grpcio_pipeline_env):
+    """Server-streaming slots decode their chunks in one GPU batch
+    (the reference rejects streaming outright, discovery.go:354-356)."""
+    pipeline = grpcio_pipeline_env
+    bodies = [
+        _body("complex_nodeservice_streamnodes", {"name": "s", "depth": 5}, 1),
+        _body("hello_helloservice_sayhello", {"name": "u"}, 2),
+        _body("complex_nodeservice_streamnodes", {"name": "t", "depth": 3}, 3),
+    ]
+    before_ok = pipeline.engine.stats.gpu_ok
+    out = pipeline.process_batch(bodies, timeout_s=15.0)
+    r0, r1, r2 = (json.loads(x) for x in out)
+    assert r0["result"]["isError"] is False
+    assert len(r0["result"]["content"]) == 5
+    chunks = [json.loads(c["text"]) for c in r0["result"]["content"]]
+    assert [c.get("depth", 0) for c in chunks] == [0, 1, 2, 3, 4]
+    assert all(c["name"] == "s" for c in chunks)
+    assert r1["result"]["isError"] is False
+    assert len(r2["result"]["content"]) == 3
+    # streams took the GPU decode path, not the host fallback
+    assert pipeline.engine.stats.gpu_ok >= before_ok + 3
+    assert pipeline.engine.stats.host_fallbacks == 0
