@@ -49,7 +49,12 @@ def parse_args():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--config", default="hello1k", choices=["hello1k", "wide64", "cpu"])
+    ap.add_argument("--config", default="hello1k",
+                    choices=["hello1k", "wide64", "stream", "multi", "cpu"])
+    ap.add_argument("--stream-depth", type=int, default=4096,
+                    help="messages per stream (config 4)")
+    ap.add_argument("--backends", type=int, default=4,
+                    help="backend count for the multi config (config 5)")
     ap.add_argument("--batch", type=int, default=1024, help="requests per step (= concurrent sessions)")
     ap.add_argument("--payload-bytes", type=int, default=0)
     ap.add_argument("--invoke-workers", type=int, default=64)
@@ -63,13 +68,16 @@ def parse_args():
     return ap.parse_args()
 
 
-def start_backend(rank: int, workers: int, native: bool):
+def start_backend(rank: int, workers: int, native: bool, package: str = "bench",
+                  index: int = 0):
     """Backend in its own process over a unix socket (no shared GIL)."""
-    sock = os.path.join(tempfile.gettempdir(), f"ggrmcp_bench_{os.getpid()}_{rank}.sock")
+    sock = os.path.join(
+        tempfile.gettempdir(), f"ggrmcp_bench_{os.getpid()}_{rank}_{index}.sock"
+    )
     if os.path.exists(sock):
         os.unlink(sock)
     cmd = [sys.executable, "-m", "examples.bench_backend", "--uds", sock,
-           "--workers", str(workers)]
+           "--workers", str(workers), "--package", package]
     if native:
         cmd.append("--native")
     proc = subprocess.Popen(
@@ -83,7 +91,8 @@ def start_backend(rank: int, workers: int, native: bool):
     return proc, sock
 
 
-def make_bodies(cfg_name: str, batch: int, payload_bytes: int, seed: int):
+def make_bodies(cfg_name: str, batch: int, payload_bytes: int, seed: int,
+                stream_depth: int = 4096, n_backends: int = 4):
     rng = random.Random(seed)
     bodies = []
     if cfg_name in ("hello1k", "cpu"):
@@ -91,6 +100,25 @@ def make_bodies(cfg_name: str, batch: int, payload_bytes: int, seed: int):
         size = payload_bytes or 1024
         for i in range(batch):
             bodies.append(jsonrpc_body(tool, hello_payload(rng, size), i + 1))
+    elif cfg_name == "stream":
+        # BASELINE config 4: server-streaming, N msgs/stream
+        size = payload_bytes or 1024
+        for i in range(batch):
+            args = wide_payload(rng, target_bytes=size)
+            args["f02Int32"] = stream_depth
+            bodies.append(jsonrpc_body("bench_echoservice_streamecho", args, i + 1))
+    elif cfg_name == "multi":
+        # BASELINE config 5: mixed unary+stream across N backends, headers on
+        size = payload_bytes or 4096
+        for i in range(batch):
+            be = i % n_backends
+            args = wide_payload(rng, target_bytes=size)
+            if i % 8 == 7:  # 1-in-8 requests is a stream
+                args["f02Int32"] = 16
+                tool = f"bench{be}_echoservice_streamecho"
+            else:
+                tool = f"bench{be}_echoservice_echo"
+            bodies.append(jsonrpc_body(tool, args, i + 1))
     else:
         tool = "bench_echoservice_echo"
         size = payload_bytes or 64 * 1024
@@ -118,8 +146,20 @@ def main() -> None:
             torch.cuda.set_device(local_rank)
         dist.init_process_group(backend=backend)
 
+    # streaming flows through the grpcio python backend (the native h2
+    # backend serves unary only); config 5 runs N distinct-package backends
+    if args.config in ("stream", "multi"):
+        args.backend = "python"
     native_backend = args.backend == "native"
-    backend_proc, sock = start_backend(rank, args.backend_workers, native_backend)
+    n_backends = args.backends if args.config == "multi" else 1
+    backend_procs, socks = [], []
+    for b in range(n_backends):
+        pkg = f"bench{b}" if args.config == "multi" else "bench"
+        proc, sock = start_backend(rank, args.backend_workers, native_backend,
+                                   package=pkg, index=b)
+        backend_procs.append(proc)
+        socks.append(sock)
+    backend_proc, sock = backend_procs[0], socks[0]
     try:
         cfg = Config.default()
         cfg.grpc.uds = sock
@@ -130,7 +170,18 @@ def main() -> None:
             cfg.gpu.device_pool_bytes = 4 * 1024 * 1024 * 1024
             if args.batch == 1024:
                 args.batch = 256
-        discoverer = ServiceDiscoverer(cfg)
+        if args.config == "stream" and args.batch == 1024:
+            args.batch = 64  # 64 streams x 4096 msgs per step
+        if args.config == "multi" and args.batch == 1024:
+            args.batch = 512
+        backend_cfgs = None
+        if n_backends > 1:
+            import dataclasses as _dc
+
+            backend_cfgs = [
+                _dc.replace(cfg.grpc, uds=s) for s in socks
+            ]
+        discoverer = ServiceDiscoverer(cfg, backends=backend_cfgs)
         shard_group = None
         if dist is not None:
             from ggrmcp_amd.parallel.dist import ShardGroup
@@ -138,25 +189,33 @@ def main() -> None:
             shard_group = ShardGroup.attach(
                 dist, device=local_rank if use_gpu else None
             )
-        if native_backend:
-            # the native backend has no reflection service: rank 0 builds the
-            # descriptor blob (the descriptor-set path, loader.go route) and
+        if native_backend or n_backends > 1:
+            # no-reflection path: rank 0 builds the descriptor blob per
+            # backend (the descriptor-set path, loader.go route) and
             # broadcasts it to the other shards over RCCL (parallel/dist.py)
             if rank == 0 or shard_group is None:
                 from examples.protos import ALL_FDPS
                 from ggrmcp_amd.utils.synthetic import synthetic_fdp
                 from google.protobuf import descriptor_pb2
 
-                fdset = descriptor_pb2.FileDescriptorSet()
-                fdset.file.extend(ALL_FDPS + [synthetic_fdp()])
-                discoverer.load_descriptor_blob(fdset.SerializeToString())
+                for b in range(n_backends):
+                    pkg = f"bench{b}" if n_backends > 1 else "bench"
+                    fdset = descriptor_pb2.FileDescriptorSet()
+                    fdps = [synthetic_fdp(package=pkg)]
+                    if b == 0:
+                        fdps = ALL_FDPS + fdps
+                    fdset.file.extend(fdps)
+                    discoverer.load_descriptor_blob(
+                        fdset.SerializeToString(), backend_index=b
+                    )
             if shard_group is not None:
                 from ggrmcp_amd.parallel.dist import sync_discovery
 
                 sync_discovery(discoverer, shard_group, src=0)
-            # a grpcio channel for the fallback/CPU paths (grpcio client
-            # interoperates with the nghttp2 server)
-            discoverer.connections[0].connect(timeout_s=15)
+            # grpcio channels for the fallback/CPU/streaming paths (grpcio
+            # client interoperates with the nghttp2 server)
+            for conn in discoverer.connections:
+                conn.connect(timeout_s=15)
         else:
             discoverer.connect(timeout_s=30)
             discoverer.discover()
@@ -170,7 +229,8 @@ def main() -> None:
             from ggrmcp_amd.backend.native_invoker import NativeWireClient
 
             wire_clients = [
-                NativeWireClient(f"unix:{sock}", connections=args.connections)
+                NativeWireClient(f"unix:{s}", connections=args.connections)
+                for s in socks
             ]
 
         bodies = make_bodies(args.config, args.batch, args.payload_bytes, seed=1234 + rank)
@@ -201,6 +261,20 @@ def main() -> None:
 
             def one(item):
                 tool, args_json, rid = item
+                mi = discoverer.get_method_by_tool(tool)
+                if mi.is_server_streaming:
+                    chunks = list(
+                        discoverer.invoke_streaming(tool, args_json, None, 30.0)
+                    )
+                    resp = {
+                        "jsonrpc": "2.0",
+                        "id": rid,
+                        "result": {
+                            "content": [{"type": "text", "text": c} for c in chunks],
+                            "isError": False,
+                        },
+                    }
+                    return json.dumps(resp).encode()
                 out = discoverer.invoke_method_by_tool(tool, args_json, None, 30.0)
                 resp = {
                     "jsonrpc": "2.0",
@@ -254,7 +328,10 @@ def main() -> None:
         p50_ms = statistics.median(step_times) * 1e3
 
         if rank == 0:
-            payload_size = args.payload_bytes or (1024 if args.config in ("hello1k", "cpu") else 65536)
+            payload_size = args.payload_bytes or {
+                "hello1k": 1024, "cpu": 1024, "stream": 1024,
+                "multi": 4096, "wide64": 65536,
+            }[args.config]
             result = {
                 "metric": "MCP tool_call→gRPC unary req/sec (whole node)",
                 "value": round(reqs_per_s, 1),
@@ -269,11 +346,13 @@ def main() -> None:
                 "dtype": "uint8",
                 "data": "synthetic",
                 "config": {
-                    "model": (
-                        "hello-service SayHello"
-                        if args.config in ("hello1k", "cpu")
-                        else "bench.Wide64 64-field nested proto"
-                    ),
+                    "model": {
+                        "hello1k": "hello-service SayHello",
+                        "cpu": "hello-service SayHello",
+                        "wide64": "bench.Wide64 64-field nested proto",
+                        "stream": "bench.EchoService/StreamEcho server-streaming",
+                        "multi": "4-backend centralized gateway, mixed unary+stream",
+                    }[args.config],
                     "global_batch": args.batch * world,
                     "seq_len": payload_size,
                     "parallelism": f"dp{world}",
@@ -281,8 +360,11 @@ def main() -> None:
                     "sessions": args.batch,
                     "payload_bytes": payload_size,
                     "p50_rtt_ms": round(p50_ms, 3),
-                    "backend": f"local grpc over uds, separate process ({args.backend})",
+                    "backend": f"local grpc over uds, separate process ({args.backend}) x{n_backends}",
                     "transport": args.transport,
+                    "stream_depth": args.stream_depth if args.config == "stream" else None,
+                    "messages_per_step": (args.batch * args.stream_depth
+                                          if args.config == "stream" else None),
                 },
             }
             if use_gpu:

@@ -27,14 +27,33 @@ ALL = ALL_FDPS + [BENCH_FDP]
 NAMES = SERVICE_NAMES + ["bench.EchoService"]
 
 
-def build_bench_server(max_workers: int = 32) -> grpc.Server:
+def build_bench_server(max_workers: int = 32, package: str = "bench") -> grpc.Server:
+    """Serve ``package``.EchoService (Echo unary + StreamEcho server-stream)
+    plus the demo services.  Distinct packages distinguish backends in
+    centralized-gateway mode (BASELINE config 5)."""
     from concurrent import futures
 
-    pool = pool_for(ALL)
-    Wide64 = message_class(pool, "bench.Wide64")
+    fdps = ALL if package == "bench" else ALL_FDPS + [synthetic_fdp(package=package)]
+    pool = pool_for(fdps)
+    Wide64 = message_class(pool, f"{package}.Wide64")
 
     def _echo(request, context):
         return request
+
+    def _stream_echo(request_bytes, context):
+        """Raw-bytes streaming echo: f02_int32 = requested message count
+        (BASELINE config 4: 4096 msgs/stream).  Bytes in, bytes out — the
+        per-chunk serialize cost stays off the bench's critical path."""
+        msg = Wide64.FromString(request_bytes)
+        n = min(max(msg.f02_int32, 1), 1 << 16)
+        for _ in range(n):
+            yield request_bytes
+
+    stream_echo_handler = grpc.unary_stream_rpc_method_handler(
+        _stream_echo,
+        request_deserializer=lambda b: b,
+        response_serializer=lambda b: b,
+    )
 
     server = grpc.server(futures.ThreadPoolExecutor(max_workers=max_workers))
     # demo services (rebuild handlers against this server)
@@ -62,17 +81,19 @@ def build_bench_server(max_workers: int = 32) -> grpc.Server:
                 },
             ),
             grpc.method_handlers_generic_handler(
-                "bench.EchoService", {"Echo": hs._unary(_echo, Wide64)}
+                f"{package}.EchoService",
+                {"Echo": hs._unary(_echo, Wide64), "StreamEcho": stream_echo_handler},
             ),
         )
     )
-    servicer = ReflectionServicer(NAMES, ALL)
+    names = SERVICE_NAMES + [f"{package}.EchoService"]
+    servicer = ReflectionServicer(names, fdps)
     server.add_generic_rpc_handlers(tuple(servicer.generic_handlers()))
     return server
 
 
-def serve(target: str = "127.0.0.1:0", max_workers: int = 32):
-    server = build_bench_server(max_workers)
+def serve(target: str = "127.0.0.1:0", max_workers: int = 32, package: str = "bench"):
+    server = build_bench_server(max_workers, package)
     if target.startswith("unix:"):
         server.add_insecure_port(target)
         bound = target
@@ -107,6 +128,8 @@ def main() -> None:
     ap.add_argument("--workers", type=int, default=32)
     ap.add_argument("--native", action="store_true",
                     help="serve with the C++ nghttp2 backend instead of grpcio")
+    ap.add_argument("--package", default="bench",
+                    help="synthetic EchoService package (centralized-gateway mode)")
     args = ap.parse_args()
     target = f"unix:{args.uds}" if args.uds else f"{args.host}:{args.port}"
     if args.native:
@@ -118,7 +141,7 @@ def main() -> None:
         except KeyboardInterrupt:
             server.stop()
         return
-    server, bound = serve(target, args.workers)
+    server, bound = serve(target, args.workers, args.package)
     print(f"READY {bound}", flush=True)
     try:
         while True:

@@ -145,24 +145,65 @@ class ServiceDiscoverer:
             raise MethodNotFoundError(f"tool not found: {tool_name}")
         return mi
 
-    def descriptor_blob(self) -> bytes:
-        """Deterministic serialized descriptor snapshot (for RCCL broadcast)."""
-        fdset = descriptor_pb2.FileDescriptorSet()
-        seen = set()
-        for backend_fdps in self._fdps:
-            for fdp in sorted(backend_fdps, key=lambda f: f.name):
-                if fdp.name not in seen:
-                    seen.add(fdp.name)
-                    fdset.file.append(fdp)
-        return fdset.SerializeToString()
+    _BLOB_MAGIC = b"GGDB"  # framed multi-backend snapshot
 
-    def load_descriptor_blob(self, blob: bytes, backend_index: int = 0) -> None:
-        """Rebuild the tool map from a broadcast snapshot (non-rank-0 shards)."""
-        fdset = descriptor_pb2.FileDescriptorSet.FromString(blob)
-        pool = build_pool(fdset.file)
-        infos = extract_method_infos(fdset.file, pool, backend_index, compat_names=False)
-        self.publish_tools({mi.tool_name(): mi for mi in infos})
-        self._fdps[backend_index] = list(fdset.file)
+    def descriptor_blob(self) -> bytes:
+        """Deterministic serialized descriptor snapshot (for RCCL broadcast).
+
+        Framed per backend so centralized-gateway mode survives the
+        broadcast: [magic][u32 n][u32 len_i][fdset_i]...  Each fdset keeps
+        its backend's files in sorted order."""
+        import struct
+
+        parts = []
+        for backend_fdps in self._fdps:
+            fdset = descriptor_pb2.FileDescriptorSet()
+            for fdp in sorted(backend_fdps, key=lambda f: f.name):
+                fdset.file.append(fdp)
+            parts.append(fdset.SerializeToString())
+        out = [self._BLOB_MAGIC, struct.pack("<I", len(parts))]
+        for p in parts:
+            out.append(struct.pack("<I", len(p)))
+            out.append(p)
+        return b"".join(out)
+
+    def load_descriptor_blob(
+        self, blob: bytes, backend_index: int = 0, merge: bool = True
+    ) -> None:
+        """Rebuild the tool map from a snapshot (non-rank-0 shards, or the
+        descriptor-set path for one backend).  Accepts a raw
+        FileDescriptorSet (single backend, placed at ``backend_index``) or
+        the framed multi-backend form produced by ``descriptor_blob``."""
+        import struct
+
+        per_backend: List[Tuple[int, bytes]] = []
+        if blob.startswith(self._BLOB_MAGIC):
+            off = len(self._BLOB_MAGIC)
+            (count,) = struct.unpack_from("<I", blob, off)
+            off += 4
+            for b in range(count):
+                (ln,) = struct.unpack_from("<I", blob, off)
+                off += 4
+                per_backend.append((b, blob[off : off + ln]))
+                off += ln
+        else:
+            per_backend.append((backend_index, blob))
+
+        tools: Dict[str, MethodInfo] = dict(self._tools) if merge else {}
+        for b, raw in per_backend:
+            fdset = descriptor_pb2.FileDescriptorSet.FromString(raw)
+            if not fdset.file:
+                continue
+            pool = build_pool(fdset.file)
+            infos = extract_method_infos(fdset.file, pool, b, compat_names=False)
+            for mi in infos:
+                tools[mi.tool_name()] = mi
+            while len(self._fdps) <= b:
+                self._fdps.append([])
+                self.connections.append(self.connections[0])
+                self.reflection_clients.append(None)
+            self._fdps[b] = list(fdset.file)
+        self.publish_tools(tools)
 
     # -- invocation: the hot string seam (discovery.go:346-369,
     #    reflection.go:333-391) — CPU reference path; the GPU engine replaces

@@ -386,8 +386,12 @@ class GpuPipeline:
             flat.extend(chunks)
             flat_idx.extend([int(self._out_msg_idx[tool])] * len(chunks))
         jsons: List[Optional[bytes]] = []
-        if flat:
-            _, jsons = self.engine.decode_batch(flat, flat_idx, mode=1)
+        cap = self.engine._eng.max_batch
+        for base in range(0, len(flat), cap):
+            _, part = self.engine.decode_batch(
+                flat[base : base + cap], flat_idx[base : base + cap], mode=1
+            )
+            jsons.extend(part)
         out: Dict[int, bytes] = {}
         for i in stream_futs:
             rid, _ = self._extract_id(bodies[i])

@@ -20,10 +20,11 @@ _SCALAR_CYCLE = [
 ]
 
 
-def synthetic_fdp(n_fields: int = 64):
+def synthetic_fdp(n_fields: int = 64, package: str = "bench"):
     """bench.proto: Wide64 (n_fields mixed scalars + nested + repeated +
-    enum + map) and an EchoService echoing it."""
-    fb = FileBuilder("bench/bench.proto", "bench")
+    enum + map) and an EchoService echoing it.  ``package`` distinguishes
+    per-backend copies in centralized-gateway mode (BASELINE config 5)."""
+    fb = FileBuilder(f"{package}/bench.proto", package)
     fb.enum("Level", [("LEVEL_UNSET", 0), ("LOW", 1), ("MID", 2), ("HIGH", 3)])
     inner = fb.message("Inner")
     inner.field("key", 1, "string")
@@ -44,7 +45,12 @@ def synthetic_fdp(n_fields: int = 64):
     num += 1
     msg.map_field("attrs", num, "string", "string")
     msg.done()
-    fb.service("EchoService").method("Echo", "Wide64", "Wide64").done()
+    (
+        fb.service("EchoService")
+        .method("Echo", "Wide64", "Wide64")
+        .method("StreamEcho", "Wide64", "Wide64", server_streaming=True)
+        .done()
+    )
     return fb.build()
 
 
