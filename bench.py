@@ -17,6 +17,8 @@ each rank runs its own sessions/backend shard).
 Configs (BASELINE.json):
   --config hello1k   hello-service SayHello, 1 KB JSON payloads (default)
   --config wide64    synthetic 64-field proto, 64 KB payloads, validation on
+  --config stream    server-streaming StreamEcho, N msgs/stream (config 4)
+  --config multi     4 backends, mixed unary+stream, headers on (config 5)
   --config cpu       reference-equivalent CPU-only plumbing path (config 1)
 """
 
@@ -233,7 +235,8 @@ def main() -> None:
                 for s in socks
             ]
 
-        bodies = make_bodies(args.config, args.batch, args.payload_bytes, seed=1234 + rank)
+        bodies = make_bodies(args.config, args.batch, args.payload_bytes, seed=1234 + rank,
+                             stream_depth=args.stream_depth, n_backends=n_backends)
 
         if use_gpu:
             from ggrmcp_amd.engine.batch import GpuPipeline
