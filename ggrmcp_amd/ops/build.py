@@ -72,8 +72,28 @@ def build_h2grpc(verbose: bool = True, force: bool = False) -> Path:
     return H2_SO_PATH
 
 
+HOSTSIM_SO_PATH = OPS_DIR / "_hostsim.so"
+
+
+def build_hostsim(verbose: bool = True, force: bool = False) -> Path:
+    """Single-lane CPU build of the kernels (csrc/host_shim.h)."""
+    sources = [CSRC / "host_sim.cpp", CSRC / "json2pb.hip", CSRC / "pb2json.hip",
+               CSRC / "common.h", CSRC / "host_shim.h"]
+    if not force and not _needs(HOSTSIM_SO_PATH, sources):
+        return HOSTSIM_SO_PATH
+    cxx = os.environ.get("CXX", "g++")
+    extra = os.environ.get("GGRMCP_HOSTSIM_FLAGS", "").split()
+    cmd = [cxx, str(CSRC / "host_sim.cpp"), "-DGGRMCP_HOST_SIM", "-DWAVE=1",
+           "-x", "c++"] + _common_flags() + extra + ["-o", str(HOSTSIM_SO_PATH)]
+    if verbose:
+        print("[ggrmcp-amd build]", " ".join(cmd), file=sys.stderr, flush=True)
+    subprocess.run(cmd, check=True)
+    return HOSTSIM_SO_PATH
+
+
 def build(verbose: bool = True, force: bool = False) -> Path:
     build_h2grpc(verbose, force)
+    build_hostsim(verbose, force)
     return build_jsonproto(verbose, force)
 
 

@@ -9,7 +9,11 @@
 // JSON->protobuf encode, protobuf->JSON decode, response-envelope assembly.
 #pragma once
 
+#ifdef GGRMCP_HOST_SIM
+#include "host_shim.h"
+#else
 #include <hip/hip_runtime.h>
+#endif
 #include <stdint.h>
 
 // ---------------------------------------------------------------------------
@@ -171,7 +175,9 @@ struct Tables {
 // Device helpers
 // ---------------------------------------------------------------------------
 
+#ifndef WAVE
 #define WAVE 64
+#endif
 #define DEV __device__ __forceinline__
 
 DEV uint64_t fnv1a64(const uint8_t* p, uint32_t n) {
@@ -235,6 +241,16 @@ DEV bool get_varint(const uint8_t* p, uint32_t len, uint32_t* pos, uint64_t* out
   }
   return false;
 }
+
+// recursion cap for the encode/decode message walkers.  The kernels are
+// recursive; the device stack is raised to KERNEL_STACK_BYTES per lane via
+// hipDeviceSetLimit (engine.cpp) and MAX_RECURSE bounds the frames so the
+// worst case fits (frame <= ~512 B; 48 * 512 = 24 KB < KERNEL_STACK_BYTES).
+// Anything deeper returns E_LIMIT / E_UNSUPPORTED and transcodes on the
+// host (counted).  The MCP validation limit is 10 (validation.go:163-184);
+// protojson itself errors near depth 100.
+constexpr int MAX_RECURSE = 48;
+constexpr size_t KERNEL_STACK_BYTES = 32 * 1024;
 
 // fixed-width reserved length slots (supports nested payloads < 2^21)
 constexpr uint32_t LEN_SLOT = 3;

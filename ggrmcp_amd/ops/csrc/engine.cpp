@@ -74,6 +74,12 @@ class Engine {
          size_t cap_in, size_t cap_pb, size_t cap_scratch, size_t cap_final)
       : device_(device), max_batch_(max_batch) {
     HIP_CHECK(hipSetDevice(device_));
+    // The kernels recurse through nested messages; the compiler's static
+    // private-segment estimate (~700 B/lane) cannot cover dynamic recursion,
+    // and overflow silently corrupts the NEIGHBORING wave's stack (observed:
+    // even slots fine, odd slots garbage at depth >= 3).  Raise the device
+    // stack so MAX_RECURSE frames always fit.
+    HIP_CHECK(hipDeviceSetLimit(hipLimitStackSize, KERNEL_STACK_BYTES));
     HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
 
     upload_blob(msg_table, d_msgs_);
@@ -114,7 +120,7 @@ class Engine {
     h_final_.alloc(cap_final);
     h_results_.alloc((size_t)max_batch * sizeof(SlotResult));
     h_dec_results_.alloc((size_t)max_batch * sizeof(DecodeResult));
-    h_off_.alloc(offs * 2);
+    h_off_.alloc(offs * 3);
     h_aux_.alloc((size_t)max_batch * sizeof(int32_t) * 2);
   }
 
@@ -227,10 +233,11 @@ class Engine {
     uint32_t* h_off = (uint32_t*)h_off_.p;
     std::memcpy(h_off, resp_off.data(), (n + 1) * sizeof(uint32_t));
     std::memcpy(h_off + (n + 1), final_off.data(), (n + 1) * sizeof(uint32_t));
+    std::memcpy(h_off + 2 * (n + 1), scratch_off.data(),
+                (n + 1) * sizeof(uint32_t));
     int32_t* h_aux = (int32_t*)h_aux_.p;
     std::memcpy(h_aux, msg_idx.data(), n * sizeof(int32_t));
     if (skip_ptr) std::memcpy(h_aux + n, skip_ptr, n * sizeof(int32_t));
-    // scratch offsets travel via d_scratch_off_ (separate small copy)
     {
       py::gil_scoped_release rel;
       HIP_CHECK(hipSetDevice(device_));
@@ -241,7 +248,7 @@ class Engine {
       HIP_CHECK(hipMemcpyAsync(d_final_off_.p, h_off + (n + 1),
                                (n + 1) * sizeof(uint32_t),
                                hipMemcpyHostToDevice, stream_));
-      HIP_CHECK(hipMemcpyAsync(d_scratch_off_.p, scratch_off.data(),
+      HIP_CHECK(hipMemcpyAsync(d_scratch_off_.p, h_off + 2 * (n + 1),
                                (n + 1) * sizeof(uint32_t),
                                hipMemcpyHostToDevice, stream_));
       HIP_CHECK(hipMemcpyAsync(d_msg_idx_.p, h_aux, n * sizeof(int32_t),

@@ -1132,7 +1132,7 @@ DEV bool encode_map_entries(Ctx& c, const FieldEntry& f, int depth) {
 // encode the JSON object at c.pos as message msg_idx (payload only; caller
 // owns any surrounding tag/len)
 DEV bool encode_message(Ctx& c, int msg_idx, int depth) {
-  if (depth > 64 || (c.lim.enforce && (uint32_t)depth > c.lim.max_depth))
+  if (depth > MAX_RECURSE || (c.lim.enforce && (uint32_t)depth > c.lim.max_depth))
     return fail(c, E_LIMIT, 1);
   const MsgEntry& m = c.t.msgs[msg_idx];
   if (m.wkt_kind == WKT_STRUCT) {
@@ -1359,7 +1359,7 @@ DEV bool encode_message(Ctx& c, int msg_idx, int depth) {
 
 // google.protobuf.Value encoder: any JSON value -> Value message payload
 DEV bool encode_json_value_as_value(Ctx& c, int depth) {
-  if (depth > 64 || (c.lim.enforce && (uint32_t)depth > c.lim.max_depth))
+  if (depth > MAX_RECURSE || (c.lim.enforce && (uint32_t)depth > c.lim.max_depth))
     return fail(c, E_LIMIT, 1);
   skip_ws(c);
   uint8_t ch = peek(c);

@@ -729,7 +729,7 @@ DEV bool emit_map_entry(DCtx& c, const FieldEntry& f, uint32_t end, int depth) {
 
 // decode message payload [c.pos, end) -> JSON (wkt-aware)
 DEV bool decode_message(DCtx& c, int msg_idx, uint32_t end, int depth) {
-  if (depth > 64) return dfail(c, E_LIMIT);
+  if (depth > MAX_RECURSE) return dfail(c, E_LIMIT);
   const MsgEntry& m = c.t.msgs[msg_idx];
   // ---- WKTs ----
   if (m.wkt_kind == WKT_TIMESTAMP || m.wkt_kind == WKT_DURATION) {
