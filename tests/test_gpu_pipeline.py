@@ -149,9 +149,9 @@ def test_process_batch_streaming_gpu_decode(grpcio_pipeline_env):
     (the reference rejects streaming outright, discovery.go:354-356)."""
     pipeline = grpcio_pipeline_env
     bodies = [
-        _body("complex_nodeservice_streamnodes", {"name": "s", "depth": 5}, 1),
+        _body("complex_nodeservice_streamnodes", {"root": {"value": "s"}, "depth": 5}, 1),
         _body("hello_helloservice_sayhello", {"name": "u"}, 2),
-        _body("complex_nodeservice_streamnodes", {"name": "t", "depth": 3}, 3),
+        _body("complex_nodeservice_streamnodes", {"root": {"value": "t"}, "depth": 3}, 3),
     ]
     before_ok = pipeline.engine.stats.gpu_ok
     out = pipeline.process_batch(bodies, timeout_s=15.0)
@@ -160,7 +160,7 @@ def test_process_batch_streaming_gpu_decode(grpcio_pipeline_env):
     assert len(r0["result"]["content"]) == 5
     chunks = [json.loads(c["text"]) for c in r0["result"]["content"]]
     assert [c.get("depth", 0) for c in chunks] == [0, 1, 2, 3, 4]
-    assert all(c["name"] == "s" for c in chunks)
+    assert all(c["root"]["value"] == "s" for c in chunks)
     assert r1["result"]["isError"] is False
     assert len(r2["result"]["content"]) == 3
     # streams took the GPU decode path, not the host fallback
