@@ -27,6 +27,8 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 #include <string.h>
+#include <stdio.h>
+#include <stdlib.h>
 #include <sys/eventfd.h>
 #include <sys/socket.h>
 #include <sys/un.h>
@@ -346,6 +348,40 @@ static void fail_call(ClientConn* conn, QueuedCall& qc, int status,
   qc.batch->complete_one();
 }
 
+
+// Pump nghttp2 output to the socket until the session has nothing more to
+// send or the socket blocks.  Returns false on fatal socket error; *blocked
+// is set when the socket is full (caller arms POLLOUT).  The subtle case
+// this exists for: a partial send() followed by a successful wbuf drain must
+// LOOP BACK into nghttp2_session_mem_send — stopping there leaves queued
+// frames unsent with nothing to wake the poll (observed as ~2 poll-timeout
+// stalls per 1024-batch over UDS: 205 ms/batch instead of 7).
+static bool flush_session(nghttp2_session* sess, int fd, std::string& wbuf,
+                          bool* blocked) {
+  *blocked = false;
+  while (true) {
+    if (!wbuf.empty()) {
+      ssize_t w = send(fd, wbuf.data(), wbuf.size(), MSG_NOSIGNAL);
+      if (w < 0) {
+        if (errno == EAGAIN || errno == EWOULDBLOCK) { *blocked = true; return true; }
+        return false;
+      }
+      wbuf.erase(0, (size_t)w);
+      if (!wbuf.empty()) continue;
+    }
+    const uint8_t* out = nullptr;
+    ssize_t n = nghttp2_session_mem_send(sess, &out);
+    if (n < 0) return false;
+    if (n == 0) return true;
+    ssize_t w = send(fd, out, (size_t)n, MSG_NOSIGNAL);
+    if (w < 0) {
+      if (errno == EAGAIN || errno == EWOULDBLOCK) w = 0;
+      else return false;
+    }
+    if (w < n) wbuf.assign((const char*)out + w, (size_t)(n - w));
+  }
+}
+
 static void conn_loop(ClientConn* conn) {
   while (!conn->stop.load()) {
     // submit pending
@@ -392,31 +428,12 @@ static void conn_loop(ClientConn* conn) {
     }
     // write
     bool write_blocked = false;
-    while (conn->wbuf.empty()) {
-      const uint8_t* out = nullptr;
-      ssize_t n = nghttp2_session_mem_send(conn->sess, &out);
-      if (n <= 0) break;
-      ssize_t w = send(conn->fd, out, n, MSG_NOSIGNAL);
-      if (w < 0) {
-        if (errno == EAGAIN || errno == EWOULDBLOCK) w = 0;
-        else { conn->broken = true; break; }
-      }
-      if (w < n) {
-        conn->wbuf.assign((const char*)out + w, n - w);
-        write_blocked = true;
-        break;
-      }
-    }
-    if (!conn->wbuf.empty() && !write_blocked) {
-      ssize_t w = send(conn->fd, conn->wbuf.data(), conn->wbuf.size(), MSG_NOSIGNAL);
-      if (w > 0) conn->wbuf.erase(0, w);
-      else if (w < 0 && errno != EAGAIN && errno != EWOULDBLOCK) conn->broken = true;
-      if (!conn->wbuf.empty()) write_blocked = true;
-    }
+    if (!flush_session(conn->sess, conn->fd, conn->wbuf, &write_blocked))
+      conn->broken = true;
     if (conn->broken) break;
 
     pollfd fds[2];
-    fds[0] = {conn->fd, (short)(POLLIN | (write_blocked || !conn->wbuf.empty() ? POLLOUT : 0)), 0};
+    fds[0] = {conn->fd, (short)(POLLIN | (write_blocked ? POLLOUT : 0)), 0};
     fds[1] = {conn->wake_fd, POLLIN, 0};
     int rc = poll(fds, 2, 100);
     if (rc < 0) break;
@@ -832,27 +849,11 @@ class H2Server {
     nghttp2_session_set_local_window_size(conn.sess, NGHTTP2_FLAG_NONE, 0, 1 << 30);
 
     while (!stop_.load() && !conn.broken) {
-      while (conn.wbuf.empty()) {
-        const uint8_t* out = nullptr;
-        ssize_t n = nghttp2_session_mem_send(conn.sess, &out);
-        if (n <= 0) break;
-        ssize_t w = send(conn.fd, out, n, MSG_NOSIGNAL);
-        if (w < 0) {
-          if (errno == EAGAIN || errno == EWOULDBLOCK) w = 0;
-          else { conn.broken = true; break; }
-        }
-        if (w < n) {
-          conn.wbuf.assign((const char*)out + w, n - w);
-          break;
-        }
-      }
-      if (!conn.wbuf.empty()) {
-        ssize_t w = send(conn.fd, conn.wbuf.data(), conn.wbuf.size(), MSG_NOSIGNAL);
-        if (w > 0) conn.wbuf.erase(0, w);
-        else if (w < 0 && errno != EAGAIN && errno != EWOULDBLOCK) conn.broken = true;
-      }
+      bool write_blocked = false;
+      if (!flush_session(conn.sess, conn.fd, conn.wbuf, &write_blocked))
+        conn.broken = true;
       if (conn.broken) break;
-      pollfd pfd{conn.fd, (short)(POLLIN | (conn.wbuf.empty() ? 0 : POLLOUT)), 0};
+      pollfd pfd{conn.fd, (short)(POLLIN | (write_blocked ? POLLOUT : 0)), 0};
       int rc = poll(&pfd, 1, 200);
       if (rc < 0) break;
       if (pfd.revents & (POLLIN | POLLERR | POLLHUP)) {
