@@ -118,7 +118,8 @@ class GpuEngine:
     """One GPU device engine: compiled tables + the HIP extension Engine."""
 
     def __init__(self, tools: Dict[str, Any], config: Optional[Config] = None,
-                 device: int = 0) -> None:
+                 device: int = 0, tables: Optional[CompiledTables] = None,
+                 stats: Optional["EngineStats"] = None) -> None:
         self.config = config or Config.default()
         from .. import ops
 
@@ -126,7 +127,7 @@ class GpuEngine:
         if self._mod.device_count() == 0:
             raise ops.ExtensionUnavailable("no HIP device present")
         self.device = device
-        self.tables: CompiledTables = compile_tables(tools)
+        self.tables: CompiledTables = tables if tables is not None else compile_tables(tools)
         self.tools = tools
         gpu = self.config.gpu
         self._eng = self._mod.Engine(
@@ -145,7 +146,7 @@ class GpuEngine:
             cap_scratch=gpu.device_pool_bytes // 4,
             cap_final=gpu.device_pool_bytes // 4,
         )
-        self.stats = EngineStats()
+        self.stats = stats if stats is not None else EngineStats()
         self._lock = threading.Lock()  # one in-flight batch per engine
 
     # -- low-level batch ops -------------------------------------------------
@@ -233,12 +234,26 @@ class GpuPipeline:
                  invoke_workers: int = 64, wire_clients=None) -> None:
         self.discoverer = discoverer
         self.config = config or Config.default()
-        self.engine = GpuEngine(discoverer.tools, self.config, device)
+        # config.gpu.streams engine instances (each its own HIP stream +
+        # arenas): concurrent chunks overlap copies/kernels/invokes, hiding
+        # the host I/O stage behind the GPU stages and vice versa
+        n_engines = max(1, self.config.gpu.streams)
+        shared_stats = EngineStats()
+        tables = compile_tables(discoverer.tools)
+        self.engines = [
+            GpuEngine(discoverer.tools, self.config, device, tables=tables,
+                      stats=shared_stats)
+            for _ in range(n_engines)
+        ]
+        self.engine = self.engines[0]
         self.cpu = CpuTranscoder()
         # native C++ h2 transport per backend index (None -> grpcio threads)
         self.wire_clients = wire_clients
         self._invoke_pool = ThreadPoolExecutor(
             max_workers=invoke_workers, thread_name_prefix="ginvoke"
+        )
+        self._chunk_pool = ThreadPoolExecutor(
+            max_workers=n_engines, thread_name_prefix="gchunk"
         )
         # tool idx -> MethodInfo
         self._mi_by_idx = [
@@ -254,6 +269,7 @@ class GpuPipeline:
 
     def close(self) -> None:
         self._invoke_pool.shutdown(wait=False)
+        self._chunk_pool.shutdown(wait=False)
 
     # ---- the full batched hot path ----------------------------------------
 
@@ -263,11 +279,44 @@ class GpuPipeline:
         headers: Optional[Sequence[Dict[str, str]]] = None,
         timeout_s: Optional[float] = None,
     ) -> List[bytes]:
-        """JSON-RPC request bodies -> JSON-RPC response bodies."""
-        st = self.engine.stats
+        """JSON-RPC request bodies -> JSON-RPC response bodies.
+
+        Large batches split across the engine instances; chunks run
+        concurrently so one chunk's gRPC invoke overlaps another's GPU
+        encode/decode (copy/compute/IO pipelining across HIP streams)."""
+        n = len(bodies)
+        n_eng = len(self.engines)
+        min_chunk = 64
+        if n_eng == 1 or n < 2 * min_chunk:
+            return self._process_span(self.engine, bodies, headers, timeout_s)
+        n_chunks = min(n_eng, (n + min_chunk - 1) // min_chunk)
+        bounds = [round(i * n / n_chunks) for i in range(n_chunks + 1)]
+        futs = []
+        for k in range(n_chunks):
+            lo, hi = bounds[k], bounds[k + 1]
+            hdr = headers[lo:hi] if headers else None
+            futs.append(
+                self._chunk_pool.submit(
+                    self._process_span, self.engines[k], bodies[lo:hi], hdr, timeout_s
+                )
+            )
+        out: List[bytes] = []
+        for f in futs:
+            out.extend(f.result())
+        return out
+
+    def _process_span(
+        self,
+        engine: GpuEngine,
+        bodies: Sequence[bytes],
+        headers: Optional[Sequence[Dict[str, str]]] = None,
+        timeout_s: Optional[float] = None,
+    ) -> List[bytes]:
+        st = engine.stats
         st.batches += 1
         st.requests += len(bodies)
-        enc, pbs = self.engine.encode_batch(bodies, mode=0)
+        with engine._lock:
+            enc, pbs = engine.encode_batch(bodies, mode=0)
 
         # fan out gRPC invocations for OK slots (host-side I/O stage)
         n = len(bodies)
@@ -342,13 +391,14 @@ class GpuPipeline:
                     rpc_error[i] = e
         st.invoke_ns += time.perf_counter_ns() - t0
 
-        dec, finals = self.engine.decode_batch(resp_wire, out_idx, mode=0)
+        with engine._lock:
+            dec, finals = engine.decode_batch(resp_wire, out_idx, mode=0)
 
         # streaming: gather chunk lists, decode ALL chunks of ALL streams in
         # one value-mode GPU batch, assemble envelopes host-side
         stream_out: Dict[int, bytes] = {}
         if stream_futs:
-            stream_out = self._decode_streams(stream_futs, enc, bodies)
+            stream_out = self._decode_streams(engine, stream_futs, enc, bodies)
 
         # assemble the batch: GPU envelopes where OK, host for the rest
         out: List[bytes] = []
@@ -360,16 +410,16 @@ class GpuPipeline:
             if i in stream_out:
                 out.append(stream_out[i])
                 continue
-            out.append(self._host_slot(bodies[i], enc[i], dec[i] if resp_wire[i] is not None else None,
+            out.append(self._host_slot(engine, bodies[i], enc[i], dec[i] if resp_wire[i] is not None else None,
                                        resp_wire[i], rpc_error[i], headers[i] if headers else None,
                                        timeout_s))
         return out
 
-    def _decode_streams(self, stream_futs, enc, bodies) -> Dict[int, bytes]:
+    def _decode_streams(self, engine, stream_futs, enc, bodies) -> Dict[int, bytes]:
         """Batch-decode every stream chunk on the GPU (mode 1) and wrap each
         stream's chunks as the ToolCallResult content list (the capability
         the reference rejects outright, discovery.go:354-356)."""
-        st = self.engine.stats
+        st = engine.stats
         chunks_by_slot: Dict[int, List[bytes]] = {}
         errors: Dict[int, Exception] = {}
         for i, fut in stream_futs.items():
@@ -386,11 +436,12 @@ class GpuPipeline:
             flat.extend(chunks)
             flat_idx.extend([int(self._out_msg_idx[tool])] * len(chunks))
         jsons: List[Optional[bytes]] = []
-        cap = self.engine._eng.max_batch
+        cap = engine._eng.max_batch
         for base in range(0, len(flat), cap):
-            _, part = self.engine.decode_batch(
-                flat[base : base + cap], flat_idx[base : base + cap], mode=1
-            )
+            with engine._lock:
+                _, part = engine.decode_batch(
+                    flat[base : base + cap], flat_idx[base : base + cap], mode=1
+                )
             jsons.extend(part)
         out: Dict[int, bytes] = {}
         for i in stream_futs:
@@ -438,8 +489,8 @@ class GpuPipeline:
 
     # ---- host handling of non-GPU slots ------------------------------------
 
-    def _host_slot(self, body, enc_r, dec_r, wire, rpc_err, hdr, timeout_s) -> bytes:
-        st = self.engine.stats
+    def _host_slot(self, engine, body, enc_r, dec_r, wire, rpc_err, hdr, timeout_s) -> bytes:
+        st = engine.stats
         rid, has_id = self._extract_id(body)
         status = int(enc_r["status"])
         flags = int(enc_r["flags"])
