@@ -109,6 +109,7 @@ class Engine {
     d_scratch_off_.alloc(offs);
     d_final_off_.alloc(offs);
     d_msg_idx_.alloc((size_t)max_batch * sizeof(int32_t));
+    d_tight_off_.alloc(offs);
     d_skip_.alloc((size_t)max_batch * sizeof(int32_t));
     d_results_.alloc((size_t)max_batch * sizeof(SlotResult));
     d_dec_results_.alloc((size_t)max_batch * sizeof(DecodeResult));
@@ -121,6 +122,7 @@ class Engine {
     h_results_.alloc((size_t)max_batch * sizeof(SlotResult));
     h_dec_results_.alloc((size_t)max_batch * sizeof(DecodeResult));
     h_off_.alloc(offs * 3);
+    h_tight_.alloc(offs);
     h_aux_.alloc((size_t)max_batch * sizeof(int32_t) * 2);
   }
 
@@ -272,14 +274,47 @@ class Engine {
       HIP_CHECK(hipMemcpyAsync(h_dec_results_.p, d_dec_results_.p,
                                n * sizeof(DecodeResult), hipMemcpyDeviceToHost,
                                stream_));
-      HIP_CHECK(hipMemcpyAsync(h_final_.p, d_final_.p, final_bytes,
-                               hipMemcpyDeviceToHost, stream_));
       HIP_CHECK(hipStreamSynchronize(stream_));
+      // Gather used bytes into a tight buffer (reusing d_pb_/h_pb_) so the
+      // D2H copy is sum(out_len), not the arena's worst-case capacity.
+      DecodeResult* rs = (DecodeResult*)h_dec_results_.p;
+      uint32_t* tight = (uint32_t*)h_tight_.p;
+      uint64_t acc = 0;
+      for (int i = 0; i < n; ++i) {
+        tight[i] = (uint32_t)acc;
+        acc += (rs[i].out_len + 3u) & ~3u;
+      }
+      if (n > 0 && acc <= d_pb_.n && acc <= h_pb_.n) {
+        HIP_CHECK(hipMemcpyAsync(d_tight_off_.p, tight, n * sizeof(uint32_t),
+                                 hipMemcpyHostToDevice, stream_));
+        hipLaunchKernelGGL(k_compact_out, dim3(n), dim3(256), 0, stream_,
+                           (const uint8_t*)d_final_.p,
+                           (const uint32_t*)d_final_off_.p,
+                           (const uint32_t*)d_tight_off_.p,
+                           (const DecodeResult*)d_dec_results_.p,
+                           (uint8_t*)d_pb_.p, n);
+        HIP_CHECK(hipGetLastError());
+        if (acc)
+          HIP_CHECK(hipMemcpyAsync(h_pb_.p, d_pb_.p, acc,
+                                   hipMemcpyDeviceToHost, stream_));
+        HIP_CHECK(hipStreamSynchronize(stream_));
+        for (int i = 0; i < n; ++i) rs[i].out_off = tight[i];
+        compact_bytes_ = acc;
+        compact_used_ = true;
+      } else {
+        HIP_CHECK(hipMemcpyAsync(h_final_.p, d_final_.p, final_bytes,
+                                 hipMemcpyDeviceToHost, stream_));
+        HIP_CHECK(hipStreamSynchronize(stream_));
+        compact_used_ = false;
+      }
     }
     py::array_t<uint8_t> results({(py::ssize_t)(n * sizeof(DecodeResult))});
     std::memcpy(results.mutable_data(), h_dec_results_.p,
                 n * sizeof(DecodeResult));
-    py::memoryview out_view = py::memoryview::from_memory(h_final_.p, final_bytes);
+    py::memoryview out_view =
+        compact_used_
+            ? py::memoryview::from_memory(h_pb_.p, compact_bytes_)
+            : py::memoryview::from_memory(h_final_.p, final_bytes);
     return py::make_tuple(results, out_view);
   }
 
@@ -298,14 +333,16 @@ class Engine {
   int device_;
   int max_batch_;
   int last_batch_n_ = -1;
+  bool compact_used_ = false;
+  size_t compact_bytes_ = 0;
   hipStream_t stream_;
   Tables tables_{};
   DeviceBuf d_msgs_, d_fields_, d_enums_, d_enum_vals_, d_tools_, d_names_;
   DeviceBuf d_in_, d_pb_, d_resp_, d_scratch_, d_final_;
   DeviceBuf d_in_off_, d_pb_off_, d_resp_off_, d_scratch_off_, d_final_off_;
-  DeviceBuf d_msg_idx_, d_skip_, d_results_, d_dec_results_, d_id_slots_;
+  DeviceBuf d_msg_idx_, d_skip_, d_results_, d_dec_results_, d_id_slots_, d_tight_off_;
   PinnedBuf h_in_, h_pb_, h_resp_, h_final_, h_results_, h_dec_results_, h_off_,
-      h_aux_;
+      h_aux_, h_tight_;
 };
 
 static int device_count() {
