@@ -121,7 +121,7 @@ def test_wide64_multislot_odd_offsets(env):
 
 
 def test_deep_recursion_returns_limit(env):
-    """Depth > MAX_RECURSE (48) -> E_LIMIT (5), never corruption."""
+    """Depth > MAX_RECURSE (16) -> E_LIMIT (5), never corruption."""
     e, c, p, _ = env
     node = {"value": "leaf"}
     for _ in range(60):
@@ -132,7 +132,7 @@ def test_deep_recursion_returns_limit(env):
     assert enc[0]["status"] == 5  # E_LIMIT
     # depth just under the cap still works
     node = {"value": "leaf"}
-    for _ in range(30):
+    for _ in range(12):
         node = {"value": "n", "children": [node]}
     text = json.dumps({"root": node}).encode()
     enc, pbs = e.encode_batch([text], mode=1, msg_indices=[idx], enforce=False)
