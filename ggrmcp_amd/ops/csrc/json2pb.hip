@@ -293,6 +293,7 @@ DEV bool backfill_len(Ctx& c, uint32_t slot) {
 struct NumVal {
   int cls;  // 0 = integer (neg,mag), 1 = double
   bool neg;
+  bool imprecise;  // range undecidable at double precision -> host transcodes
   uint64_t mag;
   double d;
 };
@@ -301,17 +302,21 @@ __constant__ double POW10[23] = {1e0,  1e1,  1e2,  1e3,  1e4,  1e5,  1e6,  1e7,
                                  1e8,  1e9,  1e10, 1e11, 1e12, 1e13, 1e14, 1e15,
                                  1e16, 1e17, 1e18, 1e19, 1e20, 1e21, 1e22};
 
-DEV double pow10d(int e) {
-  // exact for |e| <= 22; composed (<=1 ulp extra) beyond
-  double r = 1.0;
-  bool inv = e < 0;
-  if (inv) e = -e;
+// scale d by 10^e progressively: dividing step-by-step keeps subnormal
+// results reachable (computing 10^324 first overflows to inf and turned
+// every |exp|>308 input into 0.0 — found by the hostsim fuzzer).
+DEV double scale_by_pow10(double d, int e) {
   while (e > 22) {
-    r *= 1e22;
+    d *= 1e22;
     e -= 22;
+    if (d > 1.8e308) return d;  // inf soon; caller range-checks
   }
-  r *= POW10[e];
-  return inv ? 1.0 / r : r;  // note: inv path re-rounds; fast path handled by caller
+  while (e < -22) {
+    d /= 1e22;
+    e += 22;
+    if (d == 0.0) return d;
+  }
+  return e >= 0 ? d * POW10[e] : d / POW10[-e];
 }
 
 // parse decimal text [p, e) -> NumVal.  Returns false if malformed.
@@ -333,8 +338,11 @@ DEV bool parse_number_text(const uint8_t* s, uint32_t p, uint32_t e, NumVal* nv)
     uint8_t ch = s[p];
     if (ch >= '0' && ch <= '9') {
       any = true;
-      if (sig < 19) {
-        mag = mag * 10 + (ch - '0');
+      uint32_t dv = ch - '0';
+      // exact u64 accumulation (full 20 digits: int64/uint64 JSON strings
+      // use every bit; a 19-digit cap rounded 2^64-nearby values, fuzzer)
+      if (!overflow && mag <= (0xFFFFFFFFFFFFFFFFull - dv) / 10) {
+        mag = mag * 10 + dv;
         if (mag) sig++;
         if (frac) dec_exp--;
       } else {
@@ -366,6 +374,7 @@ DEV bool parse_number_text(const uint8_t* s, uint32_t p, uint32_t e, NumVal* nv)
     }
   }
   if (!any) return false;
+  nv->imprecise = false;
   if (!frac && !expseen && !overflow && dec_exp == 0) {
     nv->cls = 0;
     nv->neg = neg;
@@ -381,7 +390,18 @@ DEV bool parse_number_text(const uint8_t* s, uint32_t p, uint32_t e, NumVal* nv)
     } else if (dec_exp < 0 && dec_exp >= -22 && mag < (1ull << 53)) {
       d = d / POW10[-dec_exp];
     } else {
-      d = d * pow10d(dec_exp);
+      d = scale_by_pow10(d, dec_exp);
+      if (isinf(d) && mag != 0) {
+        // composed scaling carries <=~4 ulp of error; within a few ulp of
+        // DBL_MAX that is enough to round to inf while strtod stays finite.
+        // Decide by decimal magnitude: value ~= mag * 10^dec_exp, so its
+        // decimal exponent is dec_exp + floor(log10(mag)).
+        int mexp = 0;
+        for (uint64_t m = mag; m >= 10; m /= 10) ++mexp;
+        int vexp = dec_exp + mexp;
+        if (vexp <= 308) nv->imprecise = true;  // boundary -> host transcode
+        // vexp >= 309: genuinely out of range, keep inf (caller rejects)
+      }
     }
   }
   if (neg) d = -d;
@@ -736,12 +756,17 @@ DEV bool encode_single(Ctx& c, const FieldEntry& f, int depth) {
       else if (nonf == -1) d = -HUGE_VAL;
       else if (nonf == 2) d = nan("");
       else d = nv.cls == 0 ? (nv.neg ? -(double)nv.mag : (double)nv.mag) : nv.d;
+      if (nv.imprecise && !nonf) return fail(c, E_UNSUPPORTED, (int)f.number);
       if (f.kind == K_DOUBLE) {
+        if (isinf(d) && !nonf)  // finite text out of double range (protojson rejects)
+          return fail(c, E_INVALID_PARAMS, (int)f.number);
         if (!emit_tag(c, f.number, W_I64)) return false;
         uint64_t bits = __builtin_bit_cast(uint64_t, d);
         return emit_fixed64(c, bits);
       }
       float fv = (float)d;
+      if (isinf(fv) && !isinf(d) && !nonf)  // out of float range
+        return fail(c, E_INVALID_PARAMS, (int)f.number);
       if (!emit_tag(c, f.number, W_I32)) return false;
       return emit_fixed32(c, __builtin_bit_cast(uint32_t, fv));
     }

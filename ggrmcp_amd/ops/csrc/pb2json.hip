@@ -222,8 +222,12 @@ DEV uint32_t dtoa17(uint8_t* out, double d, bool as_float) {
     out[0] = '0';
     return 1;
   }
-  // integral fast path (exact)
-  if (d == trunc(d) && d < 9.007199254740992e15) {
+  // integral fast path (exact digits).  For float fields only below 2^24:
+  // protojson prints the SHORTEST digits that round-trip as float32, which
+  // differs from the exact integer above that (e.g. 51749408.0f prints as
+  // 51749410) — the precision search below reproduces that.
+  if (d == trunc(d) &&
+      (as_float ? d < 16777216.0 : d < 9.007199254740992e15)) {
     return u64_to_dec(out, (uint64_t)d);
   }
   // shortest precision whose parse-back round-trips
@@ -298,6 +302,14 @@ DEV bool put_double(DCtx& c, double d, bool as_float) {
   if (d != d) return puts_(c, "\"NaN\"", 5);
   if (isinf(d))
     return d > 0 ? puts_(c, "\"Infinity\"", 10) : puts_(c, "\"-Infinity\"", 11);
+  // outermost decade + subnormal range: composed base-10 scaling carries a
+  // few ulp of error, enough to print digits that strtod rounds to inf (or
+  // to mis-round subnormals).  Host transcodes these slots (counted).
+  {
+    double ad = d < 0 ? -d : d;
+    if (ad > 1e308 || (ad != 0.0 && ad < 1e-307))
+      return dfail(c, E_UNSUPPORTED);
+  }
   if (c.opos + 32 > c.ocap) return dfail(c, E_OVERFLOW);
   uint32_t n = 0;
   uint8_t buf[32];

@@ -1,0 +1,148 @@
+"""Property-based differential fuzz: kernel logic (host build) vs protojson.
+
+Hypothesis generates random Wide64/Node payloads (every scalar kind, nested
+messages, repeated, map, enum, unicode strings, boundary numbers); every
+example must encode and decode identically to google.protobuf.json_format.
+Complements the fixed cases in test_hostsim.py.
+"""
+
+import json
+import math
+import string
+
+import pytest
+from hypothesis import given, settings, strategies as st
+
+from examples.protos import ALL_FDPS
+from ggrmcp_amd.descriptors.loader import build_pool, extract_method_infos
+from ggrmcp_amd.engine.cpu_ref import CpuTranscoder
+from ggrmcp_amd.engine.hostsim import HostSimEngine
+from ggrmcp_amd.utils.synthetic import synthetic_fdp
+from google.protobuf import json_format
+
+_fdps = ALL_FDPS + [synthetic_fdp()]
+_pool = build_pool(_fdps)
+_infos = {m.tool_name(): m for m in extract_method_infos(_fdps, _pool, compat_names=False)}
+_engine = HostSimEngine(_infos)
+_cpu = CpuTranscoder()
+
+
+def _approx(a, b, rel=1e-9):
+    if isinstance(a, dict) and isinstance(b, dict):
+        return a.keys() == b.keys() and all(_approx(a[k], b[k], rel) for k in a)
+    if isinstance(a, list) and isinstance(b, list):
+        return len(a) == len(b) and all(_approx(x, y, rel) for x, y in zip(a, b))
+    if isinstance(a, float) or isinstance(b, float):
+        if isinstance(a, str) or isinstance(b, str):
+            return str(a) == str(b)
+        return math.isclose(float(a), float(b), rel_tol=rel, abs_tol=1e-300)
+    return a == b
+
+
+# text without lone surrogates (protojson rejects them too)
+_text = st.text(
+    alphabet=st.characters(blacklist_categories=("Cs",), max_codepoint=0x10FFFF),
+    max_size=40,
+)
+_ascii = st.text(alphabet=string.ascii_letters + string.digits + "_-. \t\n\"\\/", max_size=48)
+
+_i32 = st.integers(min_value=-(2**31), max_value=2**31 - 1)
+_i64s = st.integers(min_value=-(2**63), max_value=2**63 - 1).map(str)
+_u32 = st.integers(min_value=0, max_value=2**32 - 1)
+_u64s = st.integers(min_value=0, max_value=2**64 - 1).map(str)
+_dbl = st.one_of(
+    st.floats(allow_nan=False, allow_infinity=False, width=64),
+    st.sampled_from([0.0, -0.0, 1e-300, 1e300, 2.2250738585072014e-308]),
+)
+_flt = st.floats(allow_nan=False, allow_infinity=False, width=32)
+
+
+def wide_strategy():
+    inner = st.fixed_dictionaries(
+        {},
+        optional={
+            "key": _ascii,
+            "value": _i64s,
+            "weight": _dbl,
+        },
+    )
+    return st.fixed_dictionaries(
+        {},
+        optional={
+            "f01String": _text,
+            "f02Int32": _i32,
+            "f03Int64": _i64s,
+            "f04Double": _dbl,
+            "f05Bool": st.booleans(),
+            "f06Uint32": _u32,
+            "f07Uint64": _u64s,
+            "f08Float": _flt,
+            "f09Sint32": _i32,
+            "f10Sint64": _i64s,
+            "f11Fixed32": _u32,
+            "f12Fixed64": _u64s,
+            "f13Sfixed32": _i32,
+            "f14Sfixed64": _i64s,
+            "nested": inner,
+            "items": st.lists(inner, max_size=4),
+            "level": st.sampled_from(["LEVEL_UNSET", "LOW", "MID", "HIGH", 0, 1, 3]),
+            "attrs": st.dictionaries(_ascii.filter(bool), _text, max_size=4),
+        },
+    )
+
+
+def node_strategy():
+    return st.recursive(
+        st.fixed_dictionaries({}, optional={"value": _text}),
+        lambda children: st.fixed_dictionaries(
+            {}, optional={"value": _text, "children": st.lists(children, max_size=3)}
+        ),
+        max_leaves=8,
+    )
+
+
+@settings(max_examples=120, deadline=None)
+@given(payload=wide_strategy())
+def test_fuzz_wide64_encode_decode(payload):
+    _roundtrip("bench.Wide64", payload)
+
+
+@settings(max_examples=60, deadline=None)
+@given(node=node_strategy())
+def test_fuzz_node_recursive(node):
+    _roundtrip("complex.NodeRequest", {"root": node})
+
+
+def _roundtrip(msg_name, payload):
+    desc = _pool.FindMessageTypeByName(msg_name)
+    text = json.dumps(payload, ensure_ascii=False)
+    idx = _engine.tables.msg_index[msg_name]
+
+    # encode: GPU-kernel wire must parse to the same message as the oracle
+    enc, pbs = _engine.encode_batch([text.encode()], mode=1, msg_indices=[idx],
+                                    enforce=False)
+    try:
+        oracle_wire = _cpu.json_to_pb(desc, text)
+        oracle_ok = True
+    except Exception:
+        oracle_ok = False
+    if not oracle_ok:
+        # oracle rejects it; the kernel must reject too (any nonzero status)
+        assert enc[0]["status"] != 0
+        return
+    if enc[0]["status"] == 6:  # E_UNSUPPORTED: declared GPU subset boundary
+        # (e.g. escaped map keys) -> the pipeline host-transcodes, counted
+        return
+    assert enc[0]["status"] == 0, f"status {enc[0]['status']} aux={enc[0]['aux']} for {text!r}"
+    g = json_format.MessageToDict(_cpu.pb_to_message(desc, pbs[0]))
+    o = json_format.MessageToDict(_cpu.pb_to_message(desc, oracle_wire))
+    assert _approx(g, o), f"\npayload: {text!r}\nsim:    {g}\noracle: {o}"
+
+    # decode: oracle wire -> kernel JSON == oracle JSON
+    dec, outs = _engine.decode_batch([oracle_wire], [idx], mode=1)
+    if dec[0]["status"] == 6:  # declared subset boundary -> host fallback
+        return
+    assert dec[0]["status"] == 0, f"decode status {dec[0]['status']} for {text!r}"
+    gj = json.loads(outs[0])
+    oj = json.loads(_cpu.pb_to_json(desc, oracle_wire))
+    assert _approx(gj, oj), f"\npayload: {text!r}\nsim:    {gj}\noracle: {oj}"
