@@ -79,15 +79,12 @@ DECODE_DTYPE = np.dtype(
 
 
 def _offsets(lengths: Sequence[int], pad: int = 0, align: int = 1) -> np.ndarray:
-    off = np.zeros(len(lengths) + 1, dtype=np.uint32)
-    acc = 0
-    for i, n in enumerate(lengths):
-        cap = n + pad
-        if align > 1:
-            cap = (cap + align - 1) // align * align
-        acc += cap
-        off[i + 1] = acc
-    return off
+    caps = np.asarray(lengths, dtype=np.uint64) + pad
+    if align > 1:
+        caps = (caps + (align - 1)) // align * align
+    off = np.zeros(len(caps) + 1, dtype=np.uint64)
+    np.cumsum(caps, out=off[1:])
+    return off.astype(np.uint32)
 
 
 @dataclass
@@ -159,18 +156,12 @@ class GpuEngine:
         enforce: bool = True,
     ) -> Tuple[np.ndarray, List[Optional[bytes]]]:
         """Run k_json2pb. Returns (slot results, per-request pb bytes)."""
-        data = b"".join(payloads)
-        in_off = _offsets([len(p) for p in payloads])
-        pb_off = _offsets([len(p) for p in payloads], pad=192, align=16)
         msg_idx = (
             np.asarray(msg_indices, dtype=np.int32) if msg_indices is not None else None
         )
-        lim = self.config.session  # unused; limits from validation defaults
         t0 = time.perf_counter_ns()
-        raw, pb_view = self._eng.encode(
-            data,
-            in_off,
-            pb_off,
+        raw, pb_view = self._eng.encode_list(
+            list(payloads),
             msg_idx,
             mode,
             10,
@@ -199,18 +190,9 @@ class GpuEngine:
         """Run k_pb2json. ``None`` payloads are auto-skipped slots."""
         skips = [1 if (p is None or (skip is not None and skip[i])) else 0
                  for i, p in enumerate(payloads)]
-        safe = [p if p is not None else b"" for p in payloads]
-        data = b"".join(safe)
-        lens = [len(p) for p in safe]
-        resp_off = _offsets(lens)
-        scratch_off = _offsets([n * 8 + 1024 for n in lens], align=16)
-        final_off = _offsets([n * 16 + 2048 for n in lens], align=16)
         t0 = time.perf_counter_ns()
-        raw, out_view = self._eng.decode(
-            data,
-            resp_off,
-            scratch_off,
-            final_off,
+        raw, out_view = self._eng.decode_list(
+            list(payloads),
             np.asarray(msg_indices, dtype=np.int32),
             np.asarray(skips, dtype=np.int32),
             mode,

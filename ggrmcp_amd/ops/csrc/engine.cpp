@@ -12,9 +12,15 @@
 // (hipcc --offload-arch=gfx950).
 //
 // Concurrency model: one Engine == one HIP stream == one in-flight batch.
-// encode()/decode() release the GIL while the GPU works, so Python-side
+// encode*/decode* release the GIL while the GPU works, so Python-side
 // pipeline threads (gRPC I/O) overlap with kernels; multiple Engine
 // instances on the same device give copy/compute overlap across batches.
+//
+// Two call styles per direction:
+//   encode(data, in_off, pb_off, ...)  — caller-staged contiguous buffer
+//   encode_list([bytes, ...], ...)     — C++ gathers the Python list into
+//     pinned memory and builds the offsets (removes the join/slice loops
+//     from the Python hot path); same for decode/decode_list.
 
 #include <pybind11/numpy.h>
 #include <pybind11/pybind11.h>
@@ -63,6 +69,38 @@ struct PinnedBuf {
 };
 
 uint64_t cdiv(uint64_t a, uint64_t b) { return (a + b - 1) / b; }
+
+// borrow (ptr, len) from a bytes / bytearray / memoryview element
+inline bool view_of(py::handle el, const char** ptr, size_t* len) {
+  if (PyBytes_Check(el.ptr())) {
+    *ptr = PyBytes_AS_STRING(el.ptr());
+    *len = (size_t)PyBytes_GET_SIZE(el.ptr());
+    return true;
+  }
+  if (el.is_none()) {
+    *ptr = nullptr;
+    *len = 0;
+    return true;
+  }
+  Py_buffer view;
+  if (PyObject_GetBuffer(el.ptr(), &view, PyBUF_CONTIG_RO) != 0) {
+    PyErr_Clear();
+    return false;
+  }
+  *ptr = (const char*)view.buf;
+  *len = (size_t)view.len;
+  PyBuffer_Release(&view);  // borrowed pointers stay valid for bytes-likes
+  return true;
+}
+
+// per-request arena sizing rules (single source of truth, mirrored nowhere)
+inline size_t pb_cap(size_t in_len) { return (in_len + 192 + 15) & ~(size_t)15; }
+inline size_t scratch_cap(size_t wire_len) {
+  return (wire_len * 8 + 1024 + 15) & ~(size_t)15;
+}
+inline size_t final_cap(size_t wire_len) {
+  return (wire_len * 16 + 2048 + 15) & ~(size_t)15;
+}
 
 }  // namespace
 
@@ -131,13 +169,14 @@ class Engine {
     (void)hipStreamDestroy(stream_);
   }
 
-  // JSON(-RPC) -> protobuf.  mode 0: full envelope; mode 1: bare message.
-  // Returns (results u8[ n*32 ] structured, pb memoryview).
+  // ---- encode: JSON(-RPC) -> protobuf ------------------------------------
+  // mode 0: full envelope; mode 1: bare message.
+  // Returns (results u8[n*32] structured, pb memoryview into pinned mem).
+
   py::tuple encode(py::buffer data, py::array_t<uint32_t> in_off,
-                   py::array_t<uint32_t> pb_off,
-                   py::object msg_idx,  // int32 array or None
-                   int mode, uint32_t max_depth, uint32_t max_string,
-                   uint32_t max_args, int enforce) {
+                   py::array_t<uint32_t> pb_off, py::object msg_idx, int mode,
+                   uint32_t max_depth, uint32_t max_string, uint32_t max_args,
+                   int enforce) {
     py::buffer_info din = data.request();
     auto in_off_v = in_off.unchecked<1>();
     auto pb_off_v = pb_off.unchecked<1>();
@@ -148,66 +187,52 @@ class Engine {
     if (in_bytes > h_in_.n || (size_t)din.size < in_bytes)
       throw std::runtime_error("input exceeds engine cap_in");
     if (pb_bytes > d_pb_.n) throw std::runtime_error("pb cap exceeded");
-
-    const int32_t* msg_idx_ptr = nullptr;
-    py::array_t<int32_t> msg_idx_arr;
-    if (!msg_idx.is_none()) {
-      msg_idx_arr = msg_idx.cast<py::array_t<int32_t>>();
-      msg_idx_ptr = msg_idx_arr.data();
-    }
-
     std::memcpy(h_in_.p, din.ptr, in_bytes);
     uint32_t* h_off = (uint32_t*)h_off_.p;
     std::memcpy(h_off, in_off.data(), (n + 1) * sizeof(uint32_t));
     std::memcpy(h_off + (n + 1), pb_off.data(), (n + 1) * sizeof(uint32_t));
-    int32_t* h_aux = (int32_t*)h_aux_.p;
-    if (msg_idx_ptr) std::memcpy(h_aux, msg_idx_ptr, n * sizeof(int32_t));
-
+    bool has_idx = stage_msg_idx(msg_idx, n);
     Limits lim{max_depth, max_string, max_args, (uint32_t)enforce};
-    {
-      py::gil_scoped_release rel;
-      HIP_CHECK(hipSetDevice(device_));
-      HIP_CHECK(hipMemcpyAsync(d_in_.p, h_in_.p, in_bytes,
-                               hipMemcpyHostToDevice, stream_));
-      HIP_CHECK(hipMemcpyAsync(d_in_off_.p, h_off, (n + 1) * sizeof(uint32_t),
-                               hipMemcpyHostToDevice, stream_));
-      HIP_CHECK(hipMemcpyAsync(d_pb_off_.p, h_off + (n + 1),
-                               (n + 1) * sizeof(uint32_t),
-                               hipMemcpyHostToDevice, stream_));
-      if (msg_idx_ptr)
-        HIP_CHECK(hipMemcpyAsync(d_msg_idx_.p, h_aux, n * sizeof(int32_t),
-                                 hipMemcpyHostToDevice, stream_));
-      int blocks = (int)cdiv(n, WPB);
-      if (blocks > 0) {
-        hipLaunchKernelGGL(k_json2pb, dim3(blocks), dim3(WPB * WAVE), 0,
-                           stream_, (const uint8_t*)d_in_.p,
-                           (const uint32_t*)d_in_off_.p, (uint8_t*)d_pb_.p,
-                           (const uint32_t*)d_pb_off_.p,
-                           (SlotResult*)d_results_.p, (uint8_t*)d_id_slots_.p,
-                           msg_idx_ptr ? (const int32_t*)d_msg_idx_.p : nullptr,
-                           tables_, lim, n, mode);
-        HIP_CHECK(hipGetLastError());
-      }
-      HIP_CHECK(hipMemcpyAsync(h_results_.p, d_results_.p,
-                               n * sizeof(SlotResult), hipMemcpyDeviceToHost,
-                               stream_));
-      HIP_CHECK(hipMemcpyAsync(h_pb_.p, d_pb_.p, pb_bytes,
-                               hipMemcpyDeviceToHost, stream_));
-      HIP_CHECK(hipStreamSynchronize(stream_));
-    }
-    last_batch_n_ = n;
-    py::array_t<uint8_t> results({(py::ssize_t)(n * sizeof(SlotResult))});
-    std::memcpy(results.mutable_data(), h_results_.p, n * sizeof(SlotResult));
-    py::memoryview pb_view = py::memoryview::from_memory(h_pb_.p, pb_bytes);
-    return py::make_tuple(results, pb_view);
+    return run_encode(n, in_bytes, pb_bytes, has_idx, lim, mode);
   }
 
-  // protobuf -> JSON(-RPC response).  mode 0: envelope using the id slots of
-  // the LAST encode() batch (slot-aligned); mode 1: bare JSON.
+  py::tuple encode_list(py::sequence items, py::object msg_idx, int mode,
+                        uint32_t max_depth, uint32_t max_string,
+                        uint32_t max_args, int enforce) {
+    int n = (int)py::len(items);
+    if (n < 0 || n > max_batch_) throw std::runtime_error("bad batch size");
+    uint32_t* in_off = (uint32_t*)h_off_.p;
+    uint32_t* pb_off = in_off + (n + 1);
+    size_t acc = 0, pacc = 0;
+    uint8_t* dst = (uint8_t*)h_in_.p;
+    for (int i = 0; i < n; ++i) {
+      const char* ptr;
+      size_t len;
+      if (!view_of(items[i], &ptr, &len))
+        throw std::runtime_error("encode_list: unsupported element type");
+      if (acc + len > h_in_.n) throw std::runtime_error("input exceeds cap_in");
+      in_off[i] = (uint32_t)acc;
+      pb_off[i] = (uint32_t)pacc;
+      if (len) std::memcpy(dst + acc, ptr, len);
+      acc += len;
+      pacc += pb_cap(len);
+    }
+    in_off[n] = (uint32_t)acc;
+    pb_off[n] = (uint32_t)pacc;
+    if (pacc > d_pb_.n) throw std::runtime_error("pb cap exceeded");
+    bool has_idx = stage_msg_idx(msg_idx, n);
+    Limits lim{max_depth, max_string, max_args, (uint32_t)enforce};
+    return run_encode(n, acc, pacc, has_idx, lim, mode);
+  }
+
+  // ---- decode: protobuf -> JSON(-RPC response) ---------------------------
+  // mode 0: envelope using the id slots of the LAST encode batch
+  // (slot-aligned); mode 1: bare JSON.
+
   py::tuple decode(py::buffer data, py::array_t<uint32_t> resp_off,
                    py::array_t<uint32_t> scratch_off,
-                   py::array_t<uint32_t> final_off,
-                   py::array_t<int32_t> msg_idx, py::object skip, int mode) {
+                   py::array_t<uint32_t> final_off, py::array_t<int32_t> msg_idx,
+                   py::object skip, int mode) {
     py::buffer_info din = data.request();
     auto resp_off_v = resp_off.unchecked<1>();
     int n = (int)resp_off_v.shape(0) - 1;
@@ -223,14 +248,6 @@ class Engine {
     if (final_bytes > d_final_.n) throw std::runtime_error("final cap");
     if (mode == 0 && n != last_batch_n_)
       throw std::runtime_error("envelope decode batch must match last encode");
-
-    const int32_t* skip_ptr = nullptr;
-    py::array_t<int32_t> skip_arr;
-    if (!skip.is_none()) {
-      skip_arr = skip.cast<py::array_t<int32_t>>();
-      skip_ptr = skip_arr.data();
-    }
-
     std::memcpy(h_resp_.p, din.ptr, resp_bytes);
     uint32_t* h_off = (uint32_t*)h_off_.p;
     std::memcpy(h_off, resp_off.data(), (n + 1) * sizeof(uint32_t));
@@ -239,7 +256,121 @@ class Engine {
                 (n + 1) * sizeof(uint32_t));
     int32_t* h_aux = (int32_t*)h_aux_.p;
     std::memcpy(h_aux, msg_idx.data(), n * sizeof(int32_t));
-    if (skip_ptr) std::memcpy(h_aux + n, skip_ptr, n * sizeof(int32_t));
+    bool has_skip = false;
+    if (!skip.is_none()) {
+      auto skip_arr = skip.cast<py::array_t<int32_t>>();
+      std::memcpy(h_aux + n, skip_arr.data(), n * sizeof(int32_t));
+      has_skip = true;
+    }
+    return run_decode(n, resp_bytes, final_bytes, has_skip, mode);
+  }
+
+  py::tuple decode_list(py::sequence items, py::array_t<int32_t> msg_idx,
+                        py::object skip, int mode) {
+    int n = (int)py::len(items);
+    if (n < 0 || n > max_batch_) throw std::runtime_error("bad batch size");
+    if (mode == 0 && n != last_batch_n_)
+      throw std::runtime_error("envelope decode batch must match last encode");
+    uint32_t* resp_off = (uint32_t*)h_off_.p;
+    uint32_t* final_off = resp_off + (n + 1);
+    uint32_t* scratch_off = resp_off + 2 * (n + 1);
+    int32_t* h_aux = (int32_t*)h_aux_.p;
+    std::memcpy(h_aux, msg_idx.data(), n * sizeof(int32_t));
+    bool has_skip = false;
+    const int32_t* skip_src = nullptr;
+    py::array_t<int32_t> skip_arr;
+    if (!skip.is_none()) {
+      skip_arr = skip.cast<py::array_t<int32_t>>();
+      skip_src = skip_arr.data();
+      has_skip = true;
+    }
+    size_t acc = 0, sacc = 0, facc = 0;
+    uint8_t* dst = (uint8_t*)h_resp_.p;
+    for (int i = 0; i < n; ++i) {
+      const char* ptr;
+      size_t len;
+      if (!view_of(items[i], &ptr, &len))
+        throw std::runtime_error("decode_list: unsupported element type");
+      bool skipped = skip_src && skip_src[i];
+      if (skipped) len = 0;  // host handles this slot
+      if (acc + len > h_resp_.n) throw std::runtime_error("input exceeds cap_in");
+      resp_off[i] = (uint32_t)acc;
+      scratch_off[i] = (uint32_t)sacc;
+      final_off[i] = (uint32_t)facc;
+      if (len) std::memcpy(dst + acc, ptr, len);
+      acc += len;
+      sacc += scratch_cap(len);
+      facc += final_cap(len);
+      if (has_skip) h_aux[n + i] = skipped ? 1 : 0;
+    }
+    resp_off[n] = (uint32_t)acc;
+    scratch_off[n] = (uint32_t)sacc;
+    final_off[n] = (uint32_t)facc;
+    if (sacc > d_scratch_.n) throw std::runtime_error("scratch cap");
+    if (facc > d_final_.n) throw std::runtime_error("final cap");
+    return run_decode(n, acc, facc, has_skip, mode);
+  }
+
+  int device() const { return device_; }
+  int max_batch() const { return max_batch_; }
+
+ private:
+  bool stage_msg_idx(py::object msg_idx, int n) {
+    if (msg_idx.is_none()) return false;
+    auto arr = msg_idx.cast<py::array_t<int32_t>>();
+    std::memcpy(h_aux_.p, arr.data(), n * sizeof(int32_t));
+    return true;
+  }
+
+  // expects: h_in_ staged, h_off_[0..n]=in_off, h_off_[(n+1)..]=pb_off,
+  // h_aux_ = msg_idx when has_idx
+  py::tuple run_encode(int n, size_t in_bytes, size_t pb_bytes, bool has_idx,
+                       Limits lim, int mode) {
+    uint32_t* h_off = (uint32_t*)h_off_.p;
+    {
+      py::gil_scoped_release rel;
+      HIP_CHECK(hipSetDevice(device_));
+      HIP_CHECK(hipMemcpyAsync(d_in_.p, h_in_.p, in_bytes,
+                               hipMemcpyHostToDevice, stream_));
+      HIP_CHECK(hipMemcpyAsync(d_in_off_.p, h_off, (n + 1) * sizeof(uint32_t),
+                               hipMemcpyHostToDevice, stream_));
+      HIP_CHECK(hipMemcpyAsync(d_pb_off_.p, h_off + (n + 1),
+                               (n + 1) * sizeof(uint32_t),
+                               hipMemcpyHostToDevice, stream_));
+      if (has_idx)
+        HIP_CHECK(hipMemcpyAsync(d_msg_idx_.p, h_aux_.p, n * sizeof(int32_t),
+                                 hipMemcpyHostToDevice, stream_));
+      int blocks = (int)cdiv(n, WPB);
+      if (blocks > 0) {
+        hipLaunchKernelGGL(k_json2pb, dim3(blocks), dim3(WPB * WAVE), 0,
+                           stream_, (const uint8_t*)d_in_.p,
+                           (const uint32_t*)d_in_off_.p, (uint8_t*)d_pb_.p,
+                           (const uint32_t*)d_pb_off_.p,
+                           (SlotResult*)d_results_.p, (uint8_t*)d_id_slots_.p,
+                           has_idx ? (const int32_t*)d_msg_idx_.p : nullptr,
+                           tables_, lim, n, mode);
+        HIP_CHECK(hipGetLastError());
+      }
+      HIP_CHECK(hipMemcpyAsync(h_results_.p, d_results_.p,
+                               n * sizeof(SlotResult), hipMemcpyDeviceToHost,
+                               stream_));
+      HIP_CHECK(hipMemcpyAsync(h_pb_.p, d_pb_.p, pb_bytes,
+                               hipMemcpyDeviceToHost, stream_));
+      HIP_CHECK(hipStreamSynchronize(stream_));
+    }
+    last_batch_n_ = n;
+    py::array_t<uint8_t> results((py::ssize_t)(n * sizeof(SlotResult)));
+    std::memcpy(results.mutable_data(), h_results_.p, n * sizeof(SlotResult));
+    py::memoryview pb_view = py::memoryview::from_memory(h_pb_.p, pb_bytes);
+    return py::make_tuple(results, pb_view);
+  }
+
+  // expects: h_resp_ staged, h_off_ = resp/final/scratch offsets,
+  // h_aux_ = msg_idx (+ skip at offset n when has_skip)
+  py::tuple run_decode(int n, size_t resp_bytes, size_t final_bytes,
+                       bool has_skip, int mode) {
+    uint32_t* h_off = (uint32_t*)h_off_.p;
+    int32_t* h_aux = (int32_t*)h_aux_.p;
     {
       py::gil_scoped_release rel;
       HIP_CHECK(hipSetDevice(device_));
@@ -255,7 +386,7 @@ class Engine {
                                hipMemcpyHostToDevice, stream_));
       HIP_CHECK(hipMemcpyAsync(d_msg_idx_.p, h_aux, n * sizeof(int32_t),
                                hipMemcpyHostToDevice, stream_));
-      if (skip_ptr)
+      if (has_skip)
         HIP_CHECK(hipMemcpyAsync(d_skip_.p, h_aux + n, n * sizeof(int32_t),
                                  hipMemcpyHostToDevice, stream_));
       int blocks = (int)cdiv(n, WPB);
@@ -268,7 +399,7 @@ class Engine {
             (uint8_t*)d_scratch_.p, (const uint32_t*)d_scratch_off_.p,
             (uint8_t*)d_final_.p, (const uint32_t*)d_final_off_.p,
             (DecodeResult*)d_dec_results_.p,
-            skip_ptr ? (const int32_t*)d_skip_.p : nullptr, tables_, n, mode);
+            has_skip ? (const int32_t*)d_skip_.p : nullptr, tables_, n, mode);
         HIP_CHECK(hipGetLastError());
       }
       HIP_CHECK(hipMemcpyAsync(h_dec_results_.p, d_dec_results_.p,
@@ -308,7 +439,7 @@ class Engine {
         compact_used_ = false;
       }
     }
-    py::array_t<uint8_t> results({(py::ssize_t)(n * sizeof(DecodeResult))});
+    py::array_t<uint8_t> results((py::ssize_t)(n * sizeof(DecodeResult)));
     std::memcpy(results.mutable_data(), h_dec_results_.p,
                 n * sizeof(DecodeResult));
     py::memoryview out_view =
@@ -318,10 +449,6 @@ class Engine {
     return py::make_tuple(results, out_view);
   }
 
-  int device() const { return device_; }
-  int max_batch() const { return max_batch_; }
-
- private:
   void upload_blob(py::bytes b, DeviceBuf& buf) {
     std::string s = b;  // copy (init-time only)
     size_t bytes = s.size() ? s.size() : 1;
@@ -340,7 +467,8 @@ class Engine {
   DeviceBuf d_msgs_, d_fields_, d_enums_, d_enum_vals_, d_tools_, d_names_;
   DeviceBuf d_in_, d_pb_, d_resp_, d_scratch_, d_final_;
   DeviceBuf d_in_off_, d_pb_off_, d_resp_off_, d_scratch_off_, d_final_off_;
-  DeviceBuf d_msg_idx_, d_skip_, d_results_, d_dec_results_, d_id_slots_, d_tight_off_;
+  DeviceBuf d_msg_idx_, d_skip_, d_results_, d_dec_results_, d_id_slots_,
+      d_tight_off_;
   PinnedBuf h_in_, h_pb_, h_resp_, h_final_, h_results_, h_dec_results_, h_off_,
       h_aux_, h_tight_;
 };
@@ -372,9 +500,15 @@ PYBIND11_MODULE(_jsonproto, m) {
            py::arg("mode") = 0, py::arg("max_depth") = 10,
            py::arg("max_string") = 1024, py::arg("max_args") = 1u << 20,
            py::arg("enforce") = 1)
+      .def("encode_list", &Engine::encode_list, py::arg("items"),
+           py::arg("msg_idx") = py::none(), py::arg("mode") = 0,
+           py::arg("max_depth") = 10, py::arg("max_string") = 1024,
+           py::arg("max_args") = 1u << 20, py::arg("enforce") = 1)
       .def("decode", &Engine::decode, py::arg("data"), py::arg("resp_off"),
            py::arg("scratch_off"), py::arg("final_off"), py::arg("msg_idx"),
            py::arg("skip") = py::none(), py::arg("mode") = 0)
+      .def("decode_list", &Engine::decode_list, py::arg("items"),
+           py::arg("msg_idx"), py::arg("skip") = py::none(), py::arg("mode") = 0)
       .def_property_readonly("device", &Engine::device)
       .def_property_readonly("max_batch", &Engine::max_batch);
 }
