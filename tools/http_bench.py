@@ -1,0 +1,158 @@
+"""Serving-mode latency bench: per-request RTT through the REAL gateway.
+
+Unlike bench.py (which drives GpuPipeline.process_batch with pre-built
+batches), this spins the actual asyncio HTTP server + middleware +
+MCPHandler + BatchEngineInvoker (GPU batch window) against a local backend,
+then runs N concurrent MCP sessions each issuing sequential tools/call
+requests over keep-alive TCP connections — the shape a real MCP deployment
+sees.  Reports whole-gateway req/s and per-request p50/p90/p99 RTT.
+
+  python tools/http_bench.py --sessions 256 --requests 50 [--no-gpu]
+"""
+
+import argparse
+import asyncio
+import json
+import random
+import statistics
+import sys
+import time
+from pathlib import Path
+
+ROOT = Path(__file__).resolve().parent.parent
+sys.path.insert(0, str(ROOT))
+
+from examples.bench_backend import serve_native  # noqa: E402
+from ggrmcp_amd.cli import build_gateway  # noqa: E402
+from ggrmcp_amd.config import Config  # noqa: E402
+from ggrmcp_amd.server.http import HTTPServer  # noqa: E402
+from ggrmcp_amd.server.middleware import MetricsRecorder, default_middleware  # noqa: E402
+from ggrmcp_amd.utils.synthetic import hello_payload  # noqa: E402
+from examples.protos import ALL_FDPS  # noqa: E402
+from ggrmcp_amd.utils.synthetic import synthetic_fdp  # noqa: E402
+from google.protobuf import descriptor_pb2  # noqa: E402
+
+
+async def session_worker(port: int, sid: int, n_req: int, payload_bytes: int,
+                         lat: list) -> None:
+    rng = random.Random(sid)
+    reader, writer = await asyncio.open_connection("127.0.0.1", port)
+    try:
+        for i in range(n_req):
+            body = json.dumps({
+                "jsonrpc": "2.0", "id": i, "method": "tools/call",
+                "params": {"name": "hello_helloservice_sayhello",
+                           "arguments": hello_payload(rng, payload_bytes)},
+            }).encode()
+            req = (
+                b"POST / HTTP/1.1\r\nHost: l\r\nContent-Type: application/json\r\n"
+                + f"Mcp-Session-Id: bench-{sid}\r\nContent-Length: {len(body)}\r\n\r\n".encode()
+                + body
+            )
+            t0 = time.perf_counter()
+            writer.write(req)
+            await writer.drain()
+            # read headers
+            hdr = await reader.readuntil(b"\r\n\r\n")
+            clen = 0
+            for line in hdr.split(b"\r\n"):
+                if line.lower().startswith(b"content-length:"):
+                    clen = int(line.split(b":")[1])
+            data = await reader.readexactly(clen)
+            lat.append(time.perf_counter() - t0)
+            resp = json.loads(data)
+            assert resp.get("result", {}).get("isError") is False, resp
+    finally:
+        writer.close()
+
+
+async def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--sessions", type=int, default=256)
+    ap.add_argument("--requests", type=int, default=50)
+    ap.add_argument("--payload-bytes", type=int, default=1024)
+    ap.add_argument("--no-gpu", action="store_true")
+    ap.add_argument("--batch-window-us", type=int, default=200)
+    args = ap.parse_args()
+
+    srv, target = serve_native("127.0.0.1:0")
+    cfg = Config.default()
+    host, _, port = target.rpartition(":")
+    cfg.grpc.host, cfg.grpc.port = host, int(port)
+    cfg.gpu.enabled = not args.no_gpu
+    cfg.gpu.batch_window_us = args.batch_window_us
+    cfg.server.rate_limit_rps = 10_000_000  # measuring the engine, not the limiter
+    cfg.server.rate_limit_burst = 10_000_000
+
+    # native backend: no reflection; load the in-repo descriptor set
+    from ggrmcp_amd.backend.discovery import ServiceDiscoverer
+    from ggrmcp_amd.cli import build_gateway as _bg  # noqa: F401
+
+    # build_gateway does connect+discover; monkey-wire the descriptor path
+    fdset = descriptor_pb2.FileDescriptorSet()
+    fdset.file.extend(ALL_FDPS + [synthetic_fdp()])
+
+    import ggrmcp_amd.backend.discovery as disc_mod
+
+    orig_connect = disc_mod.ServiceDiscoverer.connect
+    orig_discover = disc_mod.ServiceDiscoverer.discover
+
+    def patched_connect(self, timeout_s=None):
+        self.connections[0].connect(timeout_s=15)
+
+    def patched_discover(self):
+        self.load_descriptor_blob(fdset.SerializeToString())
+        return self.tools
+
+    disc_mod.ServiceDiscoverer.connect = patched_connect
+    disc_mod.ServiceDiscoverer.discover = patched_discover
+    try:
+        handler, discoverer = build_gateway(cfg)
+    finally:
+        disc_mod.ServiceDiscoverer.connect = orig_connect
+        disc_mod.ServiceDiscoverer.discover = orig_discover
+
+    recorder = MetricsRecorder()
+    http = HTTPServer(handler.handle,
+                      middlewares=default_middleware(cfg.server, recorder),
+                      port=0)
+    await http.start()
+
+    lat: list = []
+    # warmup
+    await asyncio.gather(*[
+        session_worker(http.port, 10_000 + s, 3, args.payload_bytes, [])
+        for s in range(min(args.sessions, 64))
+    ])
+    lat.clear()
+    t0 = time.perf_counter()
+    await asyncio.gather(*[
+        session_worker(http.port, s, args.requests, args.payload_bytes, lat)
+        for s in range(args.sessions)
+    ])
+    dt = time.perf_counter() - t0
+
+    lat.sort()
+    n = len(lat)
+    stats = getattr(handler.invoker, "stats", None)
+    result = {
+        "mode": "http-serving",
+        "gpu": cfg.gpu.enabled,
+        "sessions": args.sessions,
+        "requests_per_session": args.requests,
+        "payload_bytes": args.payload_bytes,
+        "total_requests": n,
+        "req_per_s": round(n / dt, 1),
+        "p50_ms": round(lat[n // 2] * 1e3, 3),
+        "p90_ms": round(lat[int(n * 0.9)] * 1e3, 3),
+        "p99_ms": round(lat[int(n * 0.99)] * 1e3, 3),
+        "engine_stats": stats() if callable(stats) else None,
+    }
+    print(json.dumps(result), flush=True)
+    await http.stop(1.0)
+    discoverer.close()
+    srv.stop()
+
+
+if __name__ == "__main__":
+    asyncio.run(main())
