@@ -67,6 +67,8 @@ def parse_args():
                     help="local bench backend implementation")
     ap.add_argument("--connections", type=int, default=8,
                     help="native transport connections per backend")
+    ap.add_argument("--streams", type=int, default=0,
+                    help="override config.gpu.streams (engine instances)")
     return ap.parse_args()
 
 
@@ -166,6 +168,8 @@ def main() -> None:
         cfg = Config.default()
         cfg.grpc.uds = sock
         cfg.gpu.enabled = use_gpu
+        if args.streams > 0:
+            cfg.gpu.streams = args.streams
         if args.config == "wide64":
             # 64 KB payloads need bigger arenas; cap the default batch too
             cfg.gpu.pinned_pool_bytes = 2 * 1024 * 1024 * 1024
