@@ -550,13 +550,36 @@ class GpuPipeline:
             return None, False
 
 
+def build_wire_clients(discoverer, config: Config):
+    """Native h2 batch clients for every backend (serving hot path).  Falls
+    back to None (grpcio threads) when the extension is unavailable."""
+    try:
+        from ..backend.native_invoker import NativeWireClient
+    except Exception:  # pragma: no cover
+        return None
+    clients = []
+    try:
+        for conn in discoverer.connections:
+            clients.append(NativeWireClient(conn.target))
+    except Exception as e:  # pragma: no cover - fall back to grpcio
+        log.warning("native transport unavailable (%s); using grpcio", e)
+        for c in clients:
+            c.close()
+        return None
+    return clients
+
+
 class BatchEngineInvoker:
     """Async invoker seam for the HTTP handler: batches concurrent value-mode
     transcodes onto the GPU (the handler already parsed the envelope)."""
 
-    def __init__(self, discoverer, config: Optional[Config] = None, device: int = 0):
+    def __init__(self, discoverer, config: Optional[Config] = None, device: int = 0,
+                 wire_clients=None):
         self.config = config or Config.default()
-        self.pipeline = GpuPipeline(discoverer, self.config, device)
+        if wire_clients is None:
+            wire_clients = build_wire_clients(discoverer, self.config)
+        self.pipeline = GpuPipeline(discoverer, self.config, device,
+                                    wire_clients=wire_clients)
         self.discoverer = discoverer
         self._queue: List[Tuple[str, str, Dict[str, str], float, asyncio.Future]] = []
         self._qlock = threading.Lock()
@@ -635,6 +658,9 @@ class BatchEngineInvoker:
         out_idx = np.zeros(len(batch), dtype=np.int32)
         wires: List[Optional[bytes]] = [None] * len(batch)
         futures = {}
+        ok_slots = set()
+        per_be: Dict[int, List[Any]] = {}
+        wcs = pipeline.wire_clients
         t0 = time.perf_counter_ns()
         for i, (tool_name, args_json, hdr, timeout_s, _f) in enumerate(batch):
             mi = mis[i]
@@ -645,9 +671,19 @@ class BatchEngineInvoker:
                 continue
             if enc[i]["status"] == E_OK:
                 out_idx[i] = eng.tables.msg_index[mi.output_descriptor.full_name]
-                futures[i] = pipeline._invoke_pool.submit(
-                    self.discoverer.invoke_wire, mi, pbs[i], hdr, timeout_s
-                )
+                ok_slots.add(i)
+                if wcs:
+                    be = mi.backend_index if mi.backend_index < len(wcs) else 0
+                    g = per_be.setdefault(be, [[], [], [], [], []])
+                    g[0].append(i)
+                    g[1].append(mi.full_method_path)
+                    g[2].append(pbs[i])
+                    g[3].append(list(hdr.items()) if hdr else [])
+                    g[4].append(timeout_s)
+                else:
+                    futures[i] = pipeline._invoke_pool.submit(
+                        self.discoverer.invoke_wire, mi, pbs[i], hdr, timeout_s
+                    )
             elif enc[i]["status"] in (E_UNSUPPORTED, E_OVERFLOW):
                 eng.stats.host_fallbacks += 1
                 try:
@@ -664,6 +700,21 @@ class BatchEngineInvoker:
                 )
                 eng.stats.errors += 1
                 results[i] = ValidationError("arguments", msg)
+        if per_be:
+            def run_backend(be, g):
+                return g[0], wcs[be].invoke_batch(g[1], g[2], max(g[4]), g[3])
+
+            batch_futs = [
+                pipeline._invoke_pool.submit(run_backend, be, g)
+                for be, g in per_be.items()
+            ]
+            for fut in batch_futs:
+                slots, res = fut.result()
+                for i, r in zip(slots, res):
+                    if isinstance(r, Exception):
+                        results[i] = r
+                    else:
+                        wires[i] = r
         for i, fut in futures.items():
             try:
                 wires[i] = fut.result()
@@ -671,7 +722,7 @@ class BatchEngineInvoker:
                 results[i] = e
         eng.stats.invoke_ns += time.perf_counter_ns() - t0
         dec, jsons = eng.decode_batch(wires, out_idx, mode=1)
-        for i in list(futures.keys()):
+        for i in ok_slots:
             if wires[i] is None:
                 continue
             if jsons[i] is not None:
