@@ -49,6 +49,13 @@ def parse_args(argv: Optional[List[str]] = None) -> argparse.Namespace:
     )
     ap.add_argument("--gpus", type=int, default=None, help="GPU engines for DP sharding")
     ap.add_argument("--no-gpu", action="store_true", help="CPU-only hot path")
+    ap.add_argument(
+        "--frontend",
+        default="asyncio",
+        choices=["asyncio", "native"],
+        help="HTTP surface: asyncio (full middleware parity) or the C++ "
+             "batch reactor (production serving path)",
+    )
     return ap.parse_args(argv)
 
 
@@ -134,8 +141,34 @@ def build_gateway(cfg: Config):
     return handler, discoverer
 
 
-async def run(cfg: Config) -> None:
+async def run(cfg: Config, frontend: str = "asyncio") -> None:
     handler, discoverer = build_gateway(cfg)
+    if frontend == "native":
+        from .server.native_http import CpuBatchPipeline, NativeHTTPGateway
+
+        pipeline = (handler.invoker.pipeline
+                    if hasattr(handler.invoker, "pipeline")
+                    else CpuBatchPipeline(discoverer))
+        gw = NativeHTTPGateway(
+            pipeline, discoverer, cfg,
+            sessions=handler.sessions,
+            tool_builder=handler.tool_builder,
+            header_filter=handler.header_filter,
+            host="0.0.0.0", port=cfg.server.http_port,
+        )
+        port = gw.start()
+        stop = asyncio.Event()
+        loop = asyncio.get_running_loop()
+        for sig in (signal.SIGINT, signal.SIGTERM):
+            try:
+                loop.add_signal_handler(sig, stop.set)
+            except NotImplementedError:  # pragma: no cover
+                pass
+        log.info("native gateway ready on :%d (backend %s)", port, cfg.grpc.target)
+        await stop.wait()
+        gw.stop()
+        discoverer.close()
+        return
     recorder = MetricsRecorder()
     server = HTTPServer(
         handler.handle,
@@ -165,7 +198,7 @@ def main(argv: Optional[List[str]] = None) -> None:
     args = parse_args(argv)
     cfg = build_config(args)
     setup_logging(cfg)
-    asyncio.run(run(cfg))
+    asyncio.run(run(cfg, frontend=args.frontend))
 
 
 if __name__ == "__main__":
