@@ -201,12 +201,16 @@ class Frontend {
         }
       }
       drain_completions();
-      // dispatch batch when window expired or size reached
+      // Backpressure-clocked batching: while the worker is busy with the
+      // previous batch, arrivals accumulate; the moment it goes idle the
+      // whole backlog ships as one batch.  Batch size self-tunes to the
+      // service time (GPU + gRPC) with no artificial latency window —
+      // window_us_ only caps the wait when the worker is idle and a
+      // request just arrived (micro-coalescing across the same epoll wake).
+      (void)now;
       if (!pending_.empty() &&
           (pending_.size() >= (size_t)max_batch_ ||
-           std::chrono::duration_cast<std::chrono::microseconds>(
-               now - first_pending_)
-                   .count() >= window_us_)) {
+           !worker_busy_.load(std::memory_order_acquire))) {
         std::lock_guard<std::mutex> lk(batch_mu_);
         batches_.emplace_back(std::move(pending_));
         pending_.clear();
@@ -421,10 +425,20 @@ class Frontend {
       std::vector<PendingReq> batch;
       {
         std::unique_lock<std::mutex> lk(batch_mu_);
+        worker_busy_.store(false, std::memory_order_release);
         batch_cv_.wait(lk, [this] { return stop_.load() || !batches_.empty(); });
         if (stop_.load() && batches_.empty()) return;
         batch = std::move(batches_.front());
+        // coalesce any batches queued while we slept
         batches_.pop_front();
+        while (!batches_.empty() &&
+               batch.size() + batches_.front().size() <= (size_t)max_batch_) {
+          auto& nxt = batches_.front();
+          batch.insert(batch.end(), std::make_move_iterator(nxt.begin()),
+                       std::make_move_iterator(nxt.end()));
+          batches_.pop_front();
+        }
+        worker_busy_.store(true, std::memory_order_release);
       }
       std::vector<OutResp> out;
       out.reserve(batch.size());
@@ -515,6 +529,7 @@ class Frontend {
   std::deque<std::vector<PendingReq>> batches_;
   std::mutex done_mu_;
   std::deque<OutResp> done_;
+  std::atomic<bool> worker_busy_{false};
 };
 
 PYBIND11_MODULE(_frontend, m) {
