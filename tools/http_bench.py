@@ -8,6 +8,9 @@ requests over keep-alive TCP connections — the shape a real MCP deployment
 sees.  Reports whole-gateway req/s and per-request p50/p90/p99 RTT.
 
   python tools/http_bench.py --sessions 256 --requests 50 [--no-gpu]
+
+With --procs K the client side runs as K separate processes so the
+measurement isn't capped by one client event loop.
 """
 
 import argparse
@@ -75,6 +78,7 @@ async def main() -> None:
     ap.add_argument("--native-frontend", action="store_true",
                     help="serve through the C++ reactor + GpuPipeline instead of asyncio")
     ap.add_argument("--batch-window-us", type=int, default=200)
+    ap.add_argument("--procs", type=int, default=1, help="client processes")
     args = ap.parse_args()
 
     srv, target = serve_native("127.0.0.1:0")
@@ -147,10 +151,25 @@ async def main() -> None:
     ])
     lat.clear()
     t0 = time.perf_counter()
-    await asyncio.gather(*[
-        session_worker(http.port, s, args.requests, args.payload_bytes, lat)
-        for s in range(args.sessions)
-    ])
+    if args.procs > 1:
+        import subprocess as sp
+
+        per = args.sessions // args.procs
+        procs = [
+            sp.Popen([sys.executable, __file__, "--client-only",
+                      str(http.port), str(k * per), str(per),
+                      str(args.requests), str(args.payload_bytes)],
+                     stdout=sp.PIPE, text=True)
+            for k in range(args.procs)
+        ]
+        for p in procs:
+            out, _ = p.communicate(timeout=600)
+            lat.extend(json.loads(out.strip().splitlines()[-1]))
+    else:
+        await asyncio.gather(*[
+            session_worker(http.port, s, args.requests, args.payload_bytes, lat)
+            for s in range(args.sessions)
+        ])
     dt = time.perf_counter() - t0
 
     lat.sort()
@@ -175,5 +194,19 @@ async def main() -> None:
     srv.stop()
 
 
+async def client_only(argv) -> None:
+    port, s0, count, n_req, payload = (int(x) for x in argv)
+    lat: list = []
+    await asyncio.gather(*[
+        session_worker(port, s0 + s, n_req, payload, lat)
+        for s in range(count)
+    ])
+    print(json.dumps(lat))
+
+
 if __name__ == "__main__":
-    asyncio.run(main())
+    if "--client-only" in sys.argv:
+        i = sys.argv.index("--client-only")
+        asyncio.run(client_only(sys.argv[i + 1 : i + 6]))
+    else:
+        asyncio.run(main())
