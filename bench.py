@@ -171,11 +171,12 @@ def main() -> None:
         if args.streams > 0:
             cfg.gpu.streams = args.streams
         if args.config == "wide64":
-            # 64 KB payloads need bigger arenas; cap the default batch too
+            # 64 KB payloads need bigger arenas.  Keep the batch LARGE: the
+            # transcode kernels are per-wave latency-bound, so wall time per
+            # dispatch is nearly flat in wave count — more requests in
+            # flight is almost free GPU-side (profiles/wide64_kernels.txt)
             cfg.gpu.pinned_pool_bytes = 2 * 1024 * 1024 * 1024
-            cfg.gpu.device_pool_bytes = 4 * 1024 * 1024 * 1024
-            if args.batch == 1024:
-                args.batch = 256
+            cfg.gpu.device_pool_bytes = 6 * 1024 * 1024 * 1024
         if args.config == "stream" and args.batch == 1024:
             args.batch = 64  # 64 streams x 4096 msgs per step
         if args.config == "multi" and args.batch == 1024:

@@ -527,10 +527,64 @@ class H2GrpcClient {
     return false;
   }
 
+  // Batch server-streaming: like invoke_batch, but returns ALL gRPC-framed
+  // messages per call: list of (grpc_status, [payload, ...], message).
+  std::vector<std::tuple<int, std::vector<py::bytes>, std::string>>
+  invoke_stream_batch(
+      const std::vector<std::string>& paths, const std::vector<py::bytes>& payloads,
+      double timeout_s,
+      const std::vector<std::vector<std::pair<std::string, std::string>>>& metadata) {
+    auto raw = invoke_collect(paths, payloads, timeout_s, metadata);
+    std::vector<std::tuple<int, std::vector<py::bytes>, std::string>> out;
+    out.reserve(raw.size());
+    for (auto& r : raw) {
+      int status = std::get<0>(r);
+      const std::string& data = std::get<1>(r);
+      std::vector<py::bytes> msgs;
+      size_t pos = 0;
+      while (pos + 5 <= data.size()) {
+        uint32_t len;
+        memcpy(&len, data.data() + pos + 1, 4);
+        len = ntohl(len);
+        if (pos + 5 + len > data.size()) break;
+        msgs.emplace_back(data.substr(pos + 5, len));
+        pos += 5 + len;
+      }
+      if (status < 0) status = (msgs.empty() && data.empty()) ? 2 : 0;
+      out.emplace_back(status, std::move(msgs), std::get<2>(r));
+    }
+    return out;
+  }
+
   // Batch unary: returns list of (grpc_status, payload, message).
   // status -1 from the wire means "closed without decodable grpc-status":
   // treated as OK when a full unary message arrived, UNKNOWN otherwise.
   std::vector<std::tuple<int, py::bytes, std::string>> invoke_batch(
+      const std::vector<std::string>& paths, const std::vector<py::bytes>& payloads,
+      double timeout_s,
+      const std::vector<std::vector<std::pair<std::string, std::string>>>& metadata) {
+    auto raw = invoke_collect(paths, payloads, timeout_s, metadata);
+    std::vector<std::tuple<int, py::bytes, std::string>> out;
+    out.reserve(raw.size());
+    for (auto& r : raw) {
+      int status = std::get<0>(r);
+      const std::string& data = std::get<1>(r);
+      std::string payload;
+      if (data.size() >= 5) {
+        uint32_t len;
+        memcpy(&len, data.data() + 1, 4);
+        len = ntohl(len);
+        if (data.size() >= 5 + (size_t)len) payload = data.substr(5, len);
+      }
+      if (status < 0) status = payload.empty() && data.empty() ? 2 /*UNKNOWN*/ : 0;
+      out.emplace_back(status, py::bytes(payload), std::get<2>(r));
+    }
+    return out;
+  }
+
+ private:
+  // submit the batch, wait for completion, return raw DATA byte streams
+  std::vector<std::tuple<int, std::string, std::string>> invoke_collect(
       const std::vector<std::string>& paths, const std::vector<py::bytes>& payloads,
       double timeout_s,
       const std::vector<std::vector<std::pair<std::string, std::string>>>& metadata) {
@@ -568,29 +622,21 @@ class H2GrpcClient {
         if (batch->cv.wait_until(lk, hard) == std::cv_status::timeout) break;
       }
     }
-    std::vector<std::tuple<int, py::bytes, std::string>> out;
+    std::vector<std::tuple<int, std::string, std::string>> out;
     out.reserve(n);
     for (size_t i = 0; i < n; ++i) {
       Call& call = batch->calls[i];
       if (!call.done.load(std::memory_order_acquire)) {
-        out.emplace_back(4, py::bytes(), "deadline exceeded");
+        out.emplace_back(4, std::string(), "deadline exceeded");
         continue;
       }
-      int status = call.grpc_status;
-      std::string payload;
-      if (call.response.size() >= 5) {
-        uint32_t len;
-        memcpy(&len, call.response.data() + 1, 4);
-        len = ntohl(len);
-        if (call.response.size() >= 5 + (size_t)len)
-          payload = call.response.substr(5, len);
-      }
-      if (status < 0)
-        status = payload.empty() && call.response.empty() ? 2 /*UNKNOWN*/ : 0;
-      out.emplace_back(status, py::bytes(payload), call.grpc_message);
+      out.emplace_back(call.grpc_status, std::move(call.response),
+                       call.grpc_message);
     }
     return out;
   }
+
+ public:
 
  private:
   std::string target_;
@@ -622,7 +668,9 @@ struct ServerConn {
 
 class H2Server {
  public:
-  // kind: 0 = echo+validate, 1 = hello
+  // kind: 0 = echo+validate, 1 = hello, 2 = stream_echo (server-streaming:
+  // emits N copies of the request message, N = Wide64.f02_int32, the shape
+  // of BASELINE config 4)
   std::unordered_map<std::string, int> routes;
 
   H2Server(const std::string& target) : target_(target) {}
@@ -630,7 +678,7 @@ class H2Server {
   ~H2Server() { stop(); }
 
   void add_route(const std::string& path, const std::string& kind) {
-    routes[path] = (kind == "hello") ? 1 : 0;
+    routes[path] = (kind == "hello") ? 1 : (kind == "stream_echo") ? 2 : 0;
   }
 
   std::string start() {
@@ -697,6 +745,67 @@ class H2Server {
       return;
     }
     const uint8_t* msg = (const uint8_t*)st.body.data() + 5;
+    if (it->second == 2) {
+      // server-streaming echo: field 2 varint = message count
+      if (!pb_validate(msg, len)) {
+        st.grpc_status = 13;
+        st.grpc_message = "malformed protobuf";
+        return;
+      }
+      int64_t count = 1;
+      {
+        size_t pos = 0;
+        while (pos < len) {
+          uint64_t tag = 0;
+          int shift = 0;
+          while (pos < len && shift <= 63) {
+            uint8_t b = msg[pos++];
+            tag |= (uint64_t)(b & 0x7F) << shift;
+            if (!(b & 0x80)) break;
+            shift += 7;
+          }
+          uint32_t num = (uint32_t)(tag >> 3), wt = (uint32_t)(tag & 7);
+          if (num == 2 && wt == 0) {
+            uint64_t v = 0;
+            int s2 = 0;
+            while (pos < len && s2 <= 63) {
+              uint8_t b = msg[pos++];
+              v |= (uint64_t)(b & 0x7F) << s2;
+              if (!(b & 0x80)) break;
+              s2 += 7;
+            }
+            count = (int64_t)(int32_t)(uint32_t)v;
+            break;
+          }
+          // skip
+          if (wt == 0) {
+            while (pos < len && (msg[pos++] & 0x80)) {}
+          } else if (wt == 1) {
+            pos += 8;
+          } else if (wt == 5) {
+            pos += 4;
+          } else if (wt == 2) {
+            uint64_t l = 0;
+            int s2 = 0;
+            while (pos < len && s2 <= 35) {
+              uint8_t b = msg[pos++];
+              l |= (uint64_t)(b & 0x7F) << s2;
+              if (!(b & 0x80)) break;
+              s2 += 7;
+            }
+            pos += l;
+          } else {
+            break;
+          }
+        }
+      }
+      if (count < 1) count = 1;
+      if (count > 65536) count = 65536;
+      std::string one = grpc_frame(std::string((const char*)msg, len));
+      st.response.reserve(one.size() * count);
+      for (int64_t i = 0; i < count; ++i) st.response += one;
+      return;
+    }
     if (it->second == 1) {
       std::string name;
       if (!parse_hello_name(msg, len, &name)) {
@@ -897,6 +1006,9 @@ PYBIND11_MODULE(_h2grpc, m) {
       .def(py::init<const std::string&, int, const std::string&, int>(),
            py::arg("target"), py::arg("connections") = 4, py::arg("authority") = "",
            py::arg("max_inflight") = 512)
+      .def("invoke_stream_batch", &H2GrpcClient::invoke_stream_batch,
+           py::arg("paths"), py::arg("payloads"), py::arg("timeout_s") = 30.0,
+           py::arg("metadata") = std::vector<std::vector<std::pair<std::string, std::string>>>{})
       .def("invoke_batch", &H2GrpcClient::invoke_batch, py::arg("paths"),
            py::arg("payloads"), py::arg("timeout_s") = 30.0,
            py::arg("metadata") = std::vector<std::vector<std::pair<std::string, std::string>>>{})
