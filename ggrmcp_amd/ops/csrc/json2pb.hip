@@ -309,7 +309,7 @@ DEV double scale_by_pow10(double d, int e) {
   while (e > 22) {
     d *= 1e22;
     e -= 22;
-    if (d > 1.8e308) return d;  // inf soon; caller range-checks
+    if (d > 1.7e308) return d;  // about to overflow; caller range-checks
   }
   while (e < -22) {
     d /= 1e22;

@@ -22,6 +22,14 @@
 
 #include "common.h"
 
+// Mid-level emitters are deliberately NOT force-inlined: the iterative
+// walker calls them at many sites, and inlining every put_double/dtoa17
+// chain into the (already large) walker sent hipcc into a multi-minute
+// optimizer blowup.  Inside each noinline function the tiny primitives
+// (putc_/puts_/read_varint/put_escaped) still inline, so the per-byte hot
+// loops keep their state in registers; the call overhead is per-field.
+#define DEVN __device__ __noinline__
+
 #ifndef WPB
 #define WPB 4
 #endif
@@ -297,7 +305,7 @@ DEV uint32_t dtoa17(uint8_t* out, double d, bool as_float) {
   return o;
 }
 
-DEV bool put_double(DCtx& c, double d, bool as_float) {
+DEVN bool put_double(DCtx& c, double d, bool as_float) {
   if (as_float) d = (double)(float)d;
   if (d != d) return puts_(c, "\"NaN\"", 5);
   if (isinf(d))
@@ -370,7 +378,7 @@ DEV uint32_t put_nanos(uint8_t* o, int32_t nanos) {
   return n + digits;
 }
 
-DEV bool put_timestamp(DCtx& c, int64_t secs, int32_t nanos) {
+DEVN bool put_timestamp(DCtx& c, int64_t secs, int32_t nanos) {
   if (c.opos + 40 > c.ocap) return dfail(c, E_OVERFLOW);
   uint8_t buf[40];
   uint32_t n = 0;
@@ -404,7 +412,7 @@ DEV bool put_timestamp(DCtx& c, int64_t secs, int32_t nanos) {
   return true;
 }
 
-DEV bool put_duration(DCtx& c, int64_t secs, int32_t nanos) {
+DEVN bool put_duration(DCtx& c, int64_t secs, int32_t nanos) {
   if (c.opos + 40 > c.ocap) return dfail(c, E_OVERFLOW);
   uint8_t buf[40];
   uint32_t n = 0;
@@ -492,13 +500,21 @@ DEV uint32_t expected_wire(const FieldEntry& f) {
 // decoder
 // ---------------------------------------------------------------------------
 
-DEV bool decode_message(DCtx& c, int msg_idx, uint32_t end, int depth);
+// ---------------------------------------------------------------------------
+// decoder — ITERATIVE walker.
+//
+// The first implementation recursed through nested messages; the compiler
+// then had to keep the DCtx (and every frame) in scratch memory, so each
+// processed byte paid several private-memory round trips (~170 cycles/byte
+// measured).  This walker is a single flat function with an explicit frame
+// stack: the context never has its address escape, every helper inlines,
+// and the hot state lives in VGPRs.  It also removes the dynamic device
+// stack entirely for this kernel.
+// ---------------------------------------------------------------------------
 
-// is this scalar value the proto3 default? (for omit-default emission)
-DEV bool scalar_is_default_varint(const FieldEntry& f, uint64_t v) { return v == 0; }
-
-// emit one scalar/string/bytes value from the wire (no name)
-DEV bool emit_value(DCtx& c, const FieldEntry& f, int depth) {
+// emit ONE scalar/string/bytes/enum value from the wire (no name, not
+// K_MESSAGE — nested messages are the walker's job)
+DEVN bool emit_scalar_value(DCtx& c, const FieldEntry& f) {
   uint64_t v;
   uint32_t v32;
   switch (f.kind) {
@@ -589,44 +605,303 @@ DEV bool emit_value(DCtx& c, const FieldEntry& f, int depth) {
       c.pos += (uint32_t)v;
       return putc_(c, '"');
     }
-    case K_MESSAGE: {
-      if (!read_varint(c, &v)) return false;
-      if (c.pos + v > c.len) return dfail(c, E_PARSE);
-      uint32_t sub_end = c.pos + (uint32_t)v;
-      if (!decode_message(c, f.sub_index, sub_end, depth + 1)) return false;
-      c.pos = sub_end;
-      return true;
-    }
   }
   return dfail(c, E_UNSUPPORTED);
 }
 
-// packed array payload
-DEV bool emit_packed(DCtx& c, const FieldEntry& f, uint32_t end, bool* first) {
+// packed array payload (numeric/enum/bool kinds only per protobuf)
+DEVN bool emit_packed(DCtx& c, const FieldEntry& f, uint32_t end, bool* first) {
   while (c.pos < end) {
     if (!*first) {
       if (!putc_(c, ',')) return false;
     }
     *first = false;
-    if (!emit_value(c, f, 0)) return false;
+    if (!emit_scalar_value(c, f)) return false;
   }
   return c.pos == end || dfail(c, E_PARSE);
 }
 
-// map entry message {1: key, 2: value} -> `"key": value`
-DEV bool emit_map_entry(DCtx& c, const FieldEntry& f, uint32_t end, int depth) {
-  const MsgEntry& em = c.t.msgs[f.sub_index];
-  const FieldEntry& kf = c.t.fields[em.field_start];
-  const FieldEntry& vf = c.t.fields[em.field_start + 1];
-  // defaults if halves are missing on the wire
-  bool have_key = false, have_val = false;
-  uint32_t key_pos = 0, val_pos = 0;
-  // first pass: locate key/value spans
-  uint32_t save = c.pos;
+// map key bytes -> JSON object key (always quoted)
+DEVN bool emit_map_key(DCtx& c, const FieldEntry& kf, uint32_t key_pos,
+                      bool have_key) {
+  if (!putc_(c, '"')) return false;
+  if (have_key) {
+    uint32_t save = c.pos;
+    c.pos = key_pos;
+    uint64_t v;
+    bool ok = true;
+    switch (kf.kind) {
+      case K_STRING: {
+        ok = read_varint(c, &v);
+        if (ok && c.pos + v > c.len) ok = dfail(c, E_PARSE);
+        if (ok) ok = put_escaped(c, c.pb + c.pos, (uint32_t)v);
+        break;
+      }
+      case K_BOOL: {
+        ok = read_varint(c, &v);
+        if (ok) ok = v ? puts_(c, "true", 4) : puts_(c, "false", 5);
+        break;
+      }
+      case K_SINT64:
+      case K_SINT32: {
+        ok = read_varint(c, &v);
+        if (ok) ok = put_i64_dec(c, unzigzag64(v));
+        break;
+      }
+      case K_UINT64:
+      case K_UINT32: {
+        ok = read_varint(c, &v);
+        if (ok) ok = put_u64_dec(c, v);
+        break;
+      }
+      case K_FIXED64: {
+        ok = read_fixed64(c, &v);
+        if (ok) ok = put_u64_dec(c, v);
+        break;
+      }
+      case K_SFIXED64: {
+        ok = read_fixed64(c, &v);
+        if (ok) ok = put_i64_dec(c, (int64_t)v);
+        break;
+      }
+      case K_FIXED32: {
+        uint32_t v32;
+        ok = read_fixed32(c, &v32);
+        if (ok) ok = put_u64_dec(c, v32);
+        break;
+      }
+      case K_SFIXED32: {
+        uint32_t v32;
+        ok = read_fixed32(c, &v32);
+        if (ok) ok = put_i64_dec(c, (int32_t)v32);
+        break;
+      }
+      default: {  // int32/int64
+        ok = read_varint(c, &v);
+        if (ok) ok = put_i64_dec(c, (int64_t)v);
+      }
+    }
+    c.pos = save;
+    if (!ok) return false;
+  } else {
+    if (kf.kind != K_STRING && !putc_(c, '0')) return false;
+  }
+  if (!putc_(c, '"')) return false;
+  return putc_(c, ':');
+}
+
+// default JSON for a missing map VALUE half
+DEVN bool emit_map_default_value(DCtx& c, const FieldEntry& vf) {
+  switch (vf.kind) {
+    case K_STRING:
+    case K_BYTES: return puts_(c, "\"\"", 2);
+    case K_BOOL: return puts_(c, "false", 5);
+    case K_MESSAGE: return puts_(c, "{}", 2);
+    case K_INT64: case K_UINT64: case K_SINT64: case K_FIXED64:
+    case K_SFIXED64:
+      return puts_(c, "\"0\"", 3);
+    case K_ENUM: {
+      const EnumEntry& ee = c.t.enums[vf.sub_index];
+      for (int i = 0; i < ee.val_count; ++i) {
+        const EnumValueEntry& ev = c.t.enum_vals[ee.val_start + i];
+        if (ev.number == 0)
+          return put_json_string(c, c.t.names + ev.name_off, ev.name_len);
+      }
+      return putc_(c, '0');
+    }
+    default:
+      return putc_(c, '0');
+  }
+}
+
+// ---- scalar WKT leaves (no nesting) ----------------------------------------
+
+DEVN bool decode_ts_dur_body(DCtx& c, uint32_t end, bool is_ts) {
+  int64_t secs = 0;
+  int32_t nanos = 0;
   while (c.pos < end) {
     uint64_t tag;
     if (!read_varint(c, &tag)) return false;
-    uint32_t num = (uint32_t)(tag >> 3), wt = (uint32_t)(tag & 7);
+    uint32_t num = (uint32_t)(tag >> 3);
+    if (num == 1 && (tag & 7) == W_VARINT) {
+      uint64_t v;
+      if (!read_varint(c, &v)) return false;
+      secs = (int64_t)v;
+    } else if (num == 2 && (tag & 7) == W_VARINT) {
+      uint64_t v;
+      if (!read_varint(c, &v)) return false;
+      nanos = (int32_t)(int64_t)v;
+    } else {
+      if (!skip_wire(c, (uint32_t)(tag & 7))) return false;
+    }
+  }
+  return is_ts ? put_timestamp(c, secs, nanos) : put_duration(c, secs, nanos);
+}
+
+DEVN bool decode_wrapper_body(DCtx& c, const MsgEntry& m, uint32_t end) {
+  const FieldEntry& inner = c.t.fields[m.field_start];
+  bool emitted = false;
+  while (c.pos < end) {
+    uint64_t tag;
+    if (!read_varint(c, &tag)) return false;
+    if ((uint32_t)(tag >> 3) == 1) {
+      if (!emit_scalar_value(c, inner)) return false;
+      emitted = true;
+    } else {
+      if (!skip_wire(c, (uint32_t)(tag & 7))) return false;
+    }
+  }
+  if (!emitted) {
+    switch (inner.kind) {
+      case K_STRING: case K_BYTES: return puts_(c, "\"\"", 2);
+      case K_BOOL: return puts_(c, "false", 5);
+      case K_DOUBLE: case K_FLOAT: return putc_(c, '0');
+      case K_INT64: case K_UINT64: return puts_(c, "\"0\"", 3);
+      default: return putc_(c, '0');
+    }
+  }
+  return true;
+}
+
+DEVN bool decode_fieldmask_body(DCtx& c, uint32_t end) {
+  if (!putc_(c, '"')) return false;
+  bool first = true;
+  while (c.pos < end) {
+    uint64_t tag;
+    if (!read_varint(c, &tag)) return false;
+    if ((uint32_t)(tag >> 3) != 1 || (tag & 7) != W_LEN) {
+      if (!skip_wire(c, (uint32_t)(tag & 7))) return false;
+      continue;
+    }
+    uint64_t slen;
+    if (!read_varint(c, &slen)) return false;
+    if (c.pos + slen > c.len) return dfail(c, E_PARSE);
+    if (!first && !putc_(c, ',')) return false;
+    first = false;
+    bool up = false;  // snake -> camel
+    for (uint32_t i = 0; i < (uint32_t)slen; ++i) {
+      uint8_t ch = c.pb[c.pos + i];
+      if (ch == '_') {
+        up = true;
+        continue;
+      }
+      if (up && ch >= 'a' && ch <= 'z') ch -= 32;
+      up = false;
+      if (!putc_(c, ch)) return false;
+    }
+    c.pos += (uint32_t)slen;
+  }
+  return putc_(c, '"');
+}
+
+// ---- the walker ------------------------------------------------------------
+
+enum : uint8_t { FM_BODY = 0, FM_STRUCT = 1, FM_LIST = 2 };
+enum : uint8_t { CK_NONE = 0, CK_ARRAY = 1, CK_MAP = 2 };
+
+struct DFrame {
+  uint32_t end;           // wire end of this body
+  uint32_t prev_number;   // ascending-field check (FM_BODY)
+  uint32_t cont_end;      // current map-entry end (CK_MAP with pushed value)
+  int32_t msg_idx;
+  int32_t cont_field;     // field-table index of the active container
+  uint32_t cont_num;      // field number of the active container
+  uint8_t mode;
+  uint8_t first_member;
+  uint8_t cont_kind;
+  uint8_t cont_first;
+};
+
+DEVN bool map_entry_step(DCtx& c, DFrame& f, uint32_t eend, DFrame* stack,
+                        int& sp, int* pushed);
+
+
+// classify field f's message body starting at c.pos with length vlen.
+// Scalar WKTs are emitted inline (returns 1=done), container bodies are
+// described for a push (returns 2) with *push_mode set; error -> 0.
+DEVN int enter_body(DCtx& c, int msg_idx, uint32_t vend, uint8_t* push_mode,
+                   int32_t* push_idx, uint32_t* body_end) {
+  *push_idx = msg_idx;
+  *body_end = vend;
+  const MsgEntry& m = c.t.msgs[msg_idx];
+  switch (m.wkt_kind) {
+    case WKT_TIMESTAMP:
+      return decode_ts_dur_body(c, vend, true) ? 1 : 0;
+    case WKT_DURATION:
+      return decode_ts_dur_body(c, vend, false) ? 1 : 0;
+    case WKT_WRAPPER:
+      return decode_wrapper_body(c, m, vend) ? 1 : 0;
+    case WKT_FIELDMASK:
+      return decode_fieldmask_body(c, vend) ? 1 : 0;
+    case WKT_ANY:
+      return dfail(c, E_UNSUPPORTED) ? 1 : 0;
+    case WKT_VALUE: {
+      // one-shot oneof; nested struct/list push frames
+      if (c.pos >= vend) return puts_(c, "null", 4) ? 1 : 0;
+      uint64_t tag;
+      if (!read_varint(c, &tag)) return 0;
+      uint32_t num = (uint32_t)(tag >> 3);
+      uint64_t v;
+      switch (num) {
+        case 1:
+          if (!read_varint(c, &v)) return 0;
+          return puts_(c, "null", 4) ? 1 : 0;
+        case 2:
+          if (!read_fixed64(c, &v)) return 0;
+          return put_double(c, __builtin_bit_cast(double, v), false) ? 1 : 0;
+        case 3: {
+          if (!read_varint(c, &v)) return 0;
+          if (c.pos + v > c.len) return dfail(c, E_PARSE) ? 1 : 0;
+          if (!put_json_string(c, c.pb + c.pos, (uint32_t)v)) return 0;
+          c.pos += (uint32_t)v;
+          return 1;
+        }
+        case 4:
+          if (!read_varint(c, &v)) return 0;
+          return (v ? puts_(c, "true", 4) : puts_(c, "false", 5)) ? 1 : 0;
+        case 5:
+        case 6: {
+          if (!read_varint(c, &v)) return 0;
+          if (c.pos + v > c.len) return dfail(c, E_PARSE) ? 1 : 0;
+          const FieldEntry* sf = find_field(c, m, num);
+          if (!sf) return dfail(c, E_UNSUPPORTED) ? 1 : 0;
+          // unwrap the nested Struct/ListValue HERE: re-entering through a
+          // forced-inline recursive call sends the optimizer into runaway
+          // inlining (hipcc never finished compiling)
+          const MsgEntry& sm = c.t.msgs[sf->sub_index];
+          *push_mode = (sm.wkt_kind == WKT_LISTVALUE) ? FM_LIST : FM_STRUCT;
+          *push_idx = sf->sub_index;
+          *body_end = c.pos + (uint32_t)v;
+          return 2;
+        }
+      }
+      return dfail(c, E_PARSE) ? 1 : 0;
+    }
+    case WKT_STRUCT:
+      *push_mode = FM_STRUCT;
+      return 2;
+    case WKT_LISTVALUE:
+      *push_mode = FM_LIST;
+      return 2;
+    default:
+      *push_mode = FM_BODY;
+      return 2;
+  }
+}
+
+
+DEVN bool map_entry_step(DCtx& c, DFrame& f, uint32_t eend, DFrame* stack,
+                        int& sp, int* pushed) {
+  *pushed = 0;
+  const MsgEntry& em = c.t.msgs[c.t.fields[f.cont_field].sub_index];
+  const FieldEntry& kf = c.t.fields[em.field_start];
+  const FieldEntry& vf = c.t.fields[em.field_start + 1];
+  bool have_key = false, have_val = false;
+  uint32_t key_pos = 0, val_pos = 0;
+  while (c.pos < eend) {
+    uint64_t tag;
+    if (!read_varint(c, &tag)) return false;
+    uint32_t num = (uint32_t)(tag >> 3);
     if (num == 1) {
       have_key = true;
       key_pos = c.pos;
@@ -634,405 +909,411 @@ DEV bool emit_map_entry(DCtx& c, const FieldEntry& f, uint32_t end, int depth) {
       have_val = true;
       val_pos = c.pos;
     }
-    if (!skip_wire(c, wt)) return false;
+    if (!skip_wire(c, (uint32_t)(tag & 7))) return false;
   }
-  // key -> JSON object key (always a string)
-  if (!putc_(c, '"')) return false;
-  if (have_key) {
-    uint32_t save2 = c.pos;
-    c.pos = key_pos;
+  if (!emit_map_key(c, kf, key_pos, have_key)) return false;
+  if (!have_val) {
+    if (!emit_map_default_value(c, vf)) return false;
+    c.pos = eend;
+    return true;
+  }
+  c.pos = val_pos;
+  if (vf.kind == K_MESSAGE) {
     uint64_t v;
-    switch (kf.kind) {
-      case K_STRING: {
-        if (!read_varint(c, &v)) return false;
-        if (!put_escaped(c, c.pb + c.pos, (uint32_t)v)) return false;
-        break;
-      }
-      case K_BOOL: {
-        if (!read_varint(c, &v)) return false;
-        if (!(v ? puts_(c, "true", 4) : puts_(c, "false", 5))) return false;
-        break;
-      }
-      case K_SINT64:
-      case K_SINT32: {
-        if (!read_varint(c, &v)) return false;
-        if (!put_i64_dec(c, unzigzag64(v))) return false;
-        break;
-      }
-      case K_UINT64:
-      case K_UINT32: {
-        if (!read_varint(c, &v)) return false;
-        if (!put_u64_dec(c, v)) return false;
-        break;
-      }
-      case K_FIXED64: {
-        if (!read_fixed64(c, &v)) return false;
-        if (!put_u64_dec(c, v)) return false;
-        break;
-      }
-      case K_SFIXED64: {
-        if (!read_fixed64(c, &v)) return false;
-        if (!put_i64_dec(c, (int64_t)v)) return false;
-        break;
-      }
-      case K_FIXED32: {
-        uint32_t v32;
-        if (!read_fixed32(c, &v32)) return false;
-        if (!put_u64_dec(c, v32)) return false;
-        break;
-      }
-      case K_SFIXED32: {
-        uint32_t v32;
-        if (!read_fixed32(c, &v32)) return false;
-        if (!put_i64_dec(c, (int32_t)v32)) return false;
-        break;
-      }
-      default: {  // int32/int64
-        if (!read_varint(c, &v)) return false;
-        if (!put_i64_dec(c, (int64_t)v)) return false;
-      }
+    if (!read_varint(c, &v)) return false;
+    if (c.pos + v > c.len) return dfail(c, E_PARSE);
+    uint32_t vend = c.pos + (uint32_t)v;
+    uint8_t pm = FM_BODY;
+    int32_t pidx = vf.sub_index;
+    uint32_t bend = vend;
+    int r = enter_body(c, vf.sub_index, vend, &pm, &pidx, &bend);
+    if (r == 0) return false;
+    if (r == 1) {
+      c.pos = eend;
+      return true;
     }
-    c.pos = save2;
-  } else {
-    if (kf.kind != K_STRING && !putc_(c, '0')) return false;
+    if (sp >= MAX_RECURSE) return dfail(c, E_LIMIT);
+    f.cont_end = eend;  // resume point after the child pops
+    DFrame& nf = stack[sp++];
+    nf.end = bend;
+    nf.prev_number = 0;
+    nf.cont_end = 0;
+    nf.msg_idx = pidx;
+    nf.cont_field = -1;
+    nf.cont_num = 0;
+    nf.mode = pm;
+    nf.first_member = 1;
+    nf.cont_kind = CK_NONE;
+    nf.cont_first = 1;
+    if (!putc_(c, pm == FM_LIST ? '[' : '{')) return false;
+    *pushed = 1;
+    return true;
   }
-  if (!putc_(c, '"')) return false;
-  if (!putc_(c, ':')) return false;
-  // value
-  if (have_val) {
-    uint32_t save2 = c.pos;
-    c.pos = val_pos;
-    if (!emit_value(c, vf, depth)) return false;
-    c.pos = save2;
-  } else {
-    // default value for the value type
-    switch (vf.kind) {
-      case K_STRING: if (!puts_(c, "\"\"", 2)) return false; break;
-      case K_BYTES: if (!puts_(c, "\"\"", 2)) return false; break;
-      case K_BOOL: if (!puts_(c, "false", 5)) return false; break;
-      case K_MESSAGE: if (!puts_(c, "{}", 2)) return false; break;
-      case K_INT64: case K_UINT64: case K_SINT64: case K_FIXED64:
-      case K_SFIXED64:
-        if (!puts_(c, "\"0\"", 3)) return false;
-        break;
-      case K_ENUM: {
-        const EnumEntry& ee = c.t.enums[vf.sub_index];
-        bool done = false;
-        for (int i = 0; i < ee.val_count; ++i) {
-          const EnumValueEntry& ev = c.t.enum_vals[ee.val_start + i];
-          if (ev.number == 0) {
-            if (!put_json_string(c, c.t.names + ev.name_off, ev.name_len))
-              return false;
-            done = true;
-            break;
-          }
-        }
-        if (!done && !putc_(c, '0')) return false;
-        break;
-      }
-      default:
-        if (!putc_(c, '0')) return false;
-    }
-  }
-  c.pos = save;
-  // caller re-skips the entry; restore to start so it can
+  if (!emit_scalar_value(c, vf)) return false;
+  c.pos = eend;
   return true;
 }
 
-// decode message payload [c.pos, end) -> JSON (wkt-aware)
-DEV bool decode_message(DCtx& c, int msg_idx, uint32_t end, int depth) {
-  if (depth > MAX_RECURSE) return dfail(c, E_LIMIT);
-  const MsgEntry& m = c.t.msgs[msg_idx];
-  // ---- WKTs ----
-  if (m.wkt_kind == WKT_TIMESTAMP || m.wkt_kind == WKT_DURATION) {
-    int64_t secs = 0;
-    int32_t nanos = 0;
-    while (c.pos < end) {
-      uint64_t tag;
-      if (!read_varint(c, &tag)) return false;
-      uint32_t num = (uint32_t)(tag >> 3);
-      if (num == 1 && (tag & 7) == W_VARINT) {
-        uint64_t v;
-        if (!read_varint(c, &v)) return false;
-        secs = (int64_t)v;
-      } else if (num == 2 && (tag & 7) == W_VARINT) {
-        uint64_t v;
-        if (!read_varint(c, &v)) return false;
-        nanos = (int32_t)(int64_t)v;
-      } else {
-        if (!skip_wire(c, (uint32_t)(tag & 7))) return false;
-      }
-    }
-    return m.wkt_kind == WKT_TIMESTAMP ? put_timestamp(c, secs, nanos)
-                                       : put_duration(c, secs, nanos);
-  }
-  if (m.wkt_kind == WKT_WRAPPER) {
-    const FieldEntry& inner = c.t.fields[m.field_start];
-    bool emitted = false;
-    while (c.pos < end) {
-      uint64_t tag;
-      if (!read_varint(c, &tag)) return false;
-      if ((uint32_t)(tag >> 3) == 1) {
-        if (!emit_value(c, inner, depth)) return false;
-        emitted = true;
-      } else {
-        if (!skip_wire(c, (uint32_t)(tag & 7))) return false;
-      }
-    }
-    if (!emitted) {
-      // default inner value
-      switch (inner.kind) {
-        case K_STRING: case K_BYTES: return puts_(c, "\"\"", 2);
-        case K_BOOL: return puts_(c, "false", 5);
-        case K_DOUBLE: case K_FLOAT: return putc_(c, '0');
-        case K_INT64: case K_UINT64: return puts_(c, "\"0\"", 3);
-        default: return putc_(c, '0');
-      }
-    }
-    return true;
-  }
-  if (m.wkt_kind == WKT_VALUE) {
-    // Value oneof; empty Value -> null
-    if (c.pos >= end) return puts_(c, "null", 4);
-    uint64_t tag;
-    if (!read_varint(c, &tag)) return false;
-    uint32_t num = (uint32_t)(tag >> 3);
-    uint64_t v;
-    switch (num) {
-      case 1:
-        if (!read_varint(c, &v)) return false;
-        return puts_(c, "null", 4);
-      case 2:
-        if (!read_fixed64(c, &v)) return false;
-        return put_double(c, __builtin_bit_cast(double, v), false);
-      case 3: {
-        if (!read_varint(c, &v)) return false;
-        if (c.pos + v > c.len) return dfail(c, E_PARSE);
-        if (!put_json_string(c, c.pb + c.pos, (uint32_t)v)) return false;
-        c.pos += (uint32_t)v;
-        return true;
-      }
-      case 4:
-        if (!read_varint(c, &v)) return false;
-        return v ? puts_(c, "true", 4) : puts_(c, "false", 5);
-      case 5: {  // struct_value
-        if (!read_varint(c, &v)) return false;
-        uint32_t sub_end = c.pos + (uint32_t)v;
-        // find the Struct message index via this Value's field table
-        const FieldEntry* sf = find_field(c, m, 5);
-        if (!sf) return dfail(c, E_UNSUPPORTED);
-        if (!decode_message(c, sf->sub_index, sub_end, depth + 1)) return false;
-        c.pos = sub_end;
-        return true;
-      }
-      case 6: {  // list_value
-        if (!read_varint(c, &v)) return false;
-        uint32_t sub_end = c.pos + (uint32_t)v;
-        const FieldEntry* lf = find_field(c, m, 6);
-        if (!lf) return dfail(c, E_UNSUPPORTED);
-        if (!decode_message(c, lf->sub_index, sub_end, depth + 1)) return false;
-        c.pos = sub_end;
-        return true;
-      }
-    }
-    return dfail(c, E_PARSE);
-  }
-  if (m.wkt_kind == WKT_STRUCT) {
-    // map<string, Value> on field 1 -> JSON object
-    if (!putc_(c, '{')) return false;
-    bool first = true;
-    while (c.pos < end) {
-      uint64_t tag;
-      if (!read_varint(c, &tag)) return false;
-      if ((uint32_t)(tag >> 3) != 1 || (tag & 7) != W_LEN) {
-        if (!skip_wire(c, (uint32_t)(tag & 7))) return false;
-        continue;
-      }
-      uint64_t elen;
-      if (!read_varint(c, &elen)) return false;
-      uint32_t eend = c.pos + (uint32_t)elen;
-      if (eend > c.len) return dfail(c, E_PARSE);
-      if (!first && !putc_(c, ',')) return false;
-      first = false;
-      const FieldEntry* f1 = &c.t.fields[m.field_start];  // entries field
-      if (!emit_map_entry(c, *f1, eend, depth)) return false;
-      c.pos = eend;
-    }
-    return putc_(c, '}');
-  }
-  if (m.wkt_kind == WKT_LISTVALUE) {
-    if (!putc_(c, '[')) return false;
-    bool first = true;
-    while (c.pos < end) {
-      uint64_t tag;
-      if (!read_varint(c, &tag)) return false;
-      if ((uint32_t)(tag >> 3) != 1 || (tag & 7) != W_LEN) {
-        if (!skip_wire(c, (uint32_t)(tag & 7))) return false;
-        continue;
-      }
-      uint64_t elen;
-      if (!read_varint(c, &elen)) return false;
-      uint32_t eend = c.pos + (uint32_t)elen;
-      if (eend > c.len) return dfail(c, E_PARSE);
-      if (!first && !putc_(c, ',')) return false;
-      first = false;
-      const FieldEntry* vf = &c.t.fields[m.field_start];
-      const MsgEntry& vm = c.t.msgs[vf->sub_index];
-      (void)vm;
-      if (!decode_message(c, vf->sub_index, eend, depth + 1)) return false;
-      c.pos = eend;
-    }
-    return putc_(c, ']');
-  }
-  if (m.wkt_kind == WKT_ANY) return dfail(c, E_UNSUPPORTED);
-  if (m.wkt_kind == WKT_FIELDMASK) {
-    // repeated string paths -> comma-joined camelCase
-    if (!putc_(c, '"')) return false;
-    bool first = true;
-    while (c.pos < end) {
-      uint64_t tag;
-      if (!read_varint(c, &tag)) return false;
-      if ((uint32_t)(tag >> 3) != 1 || (tag & 7) != W_LEN) {
-        if (!skip_wire(c, (uint32_t)(tag & 7))) return false;
-        continue;
-      }
-      uint64_t slen;
-      if (!read_varint(c, &slen)) return false;
-      if (c.pos + slen > c.len) return dfail(c, E_PARSE);
-      if (!first && !putc_(c, ',')) return false;
-      first = false;
-      // snake -> camel
-      bool up = false;
-      for (uint32_t i = 0; i < (uint32_t)slen; ++i) {
-        uint8_t ch = c.pb[c.pos + i];
-        if (ch == '_') {
-          up = true;
-          continue;
-        }
-        if (up && ch >= 'a' && ch <= 'z') ch -= 32;
-        up = false;
-        if (!putc_(c, ch)) return false;
-      }
-      c.pos += (uint32_t)slen;
-    }
-    return putc_(c, '"');
+DEV bool decode_walk(DCtx& c, int top_msg_idx, uint32_t top_end) {
+  DFrame stack[MAX_RECURSE];
+  int sp = 0;
+
+  // enter the top-level body (responses are plain messages; transcode tests
+  // may decode WKT-typed messages directly)
+  {
+    uint8_t pm = FM_BODY;
+    int32_t pidx = top_msg_idx;
+    uint32_t bend = top_end;
+    int r = enter_body(c, top_msg_idx, top_end, &pm, &pidx, &bend);
+    if (r == 0) return false;
+    if (r == 1) return c.status == E_OK;
+    DFrame& f = stack[sp++];
+    f.end = bend;
+    f.prev_number = 0;
+    f.cont_end = 0;
+    f.msg_idx = pidx;
+    f.cont_field = -1;
+    f.cont_num = 0;
+    f.mode = pm;
+    f.first_member = 1;
+    f.cont_kind = CK_NONE;
+    f.cont_first = 1;
+    if (!putc_(c, pm == FM_LIST ? '[' : '{')) return false;
   }
 
-  // ---- plain message ----
-  if (!putc_(c, '{')) return false;
-  bool first_member = true;
-  uint32_t prev_number = 0;
-  while (c.pos < end) {
+  while (sp > 0) {
+    DFrame& f = stack[sp - 1];
+
+    // -- resume an active container after a child pop / entry finish ------
+    if (f.cont_kind == CK_ARRAY) {
+      bool more = false;
+      uint32_t wt = 0;
+      if (c.pos < f.end) {
+        uint32_t save = c.pos;
+        uint64_t ntag;
+        if (!read_varint(c, &ntag)) return false;
+        if ((uint32_t)(ntag >> 3) == f.cont_num) {
+          more = true;
+          wt = (uint32_t)(ntag & 7);
+        } else {
+          c.pos = save;
+        }
+      }
+      if (!more) {
+        if (!putc_(c, ']')) return false;
+        f.cont_kind = CK_NONE;
+        continue;
+      }
+      const FieldEntry& fe = c.t.fields[f.cont_field];
+      if (!putc_(c, ',')) return false;
+      if (wt == W_LEN && expected_wire(fe) != W_LEN) {
+        uint64_t plen;
+        if (!read_varint(c, &plen)) return false;
+        uint32_t pend = c.pos + (uint32_t)plen;
+        if (pend > c.len) return dfail(c, E_PARSE);
+        bool first = false;  // comma already emitted? emit_packed prepends
+        // emit_packed writes separators before each elem when !first; we
+        // already wrote one comma, so mark first=true for the first elem
+        first = true;
+        if (!emit_packed(c, fe, pend, &first)) return false;
+        continue;
+      }
+      if (fe.kind == K_MESSAGE) {
+        uint64_t v;
+        if (!read_varint(c, &v)) return false;
+        if (c.pos + v > c.len) return dfail(c, E_PARSE);
+        uint32_t vend = c.pos + (uint32_t)v;
+        uint8_t pm = FM_BODY;
+        int32_t pidx = fe.sub_index;
+        uint32_t bend = vend;
+        int r = enter_body(c, fe.sub_index, vend, &pm, &pidx, &bend);
+        if (r == 0) return false;
+        if (r == 1) {
+          c.pos = vend;
+          continue;
+        }
+        if (sp >= MAX_RECURSE) return dfail(c, E_LIMIT);
+        DFrame& nf = stack[sp++];
+        nf.end = bend;
+        nf.prev_number = 0;
+        nf.cont_end = 0;
+        nf.msg_idx = pidx;
+        nf.cont_field = -1;
+        nf.cont_num = 0;
+        nf.mode = pm;
+        nf.first_member = 1;
+        nf.cont_kind = CK_NONE;
+        nf.cont_first = 1;
+        if (!putc_(c, pm == FM_LIST ? '[' : '{')) return false;
+        continue;
+      }
+      if (wt != expected_wire(fe)) return dfail(c, E_PARSE);
+      if (!emit_scalar_value(c, fe)) return false;
+      continue;
+    }
+    if (f.cont_kind == CK_MAP) {
+      // a pushed map VALUE just finished: jump to the entry end, then check
+      // for an adjacent entry of the same field
+      if (f.cont_end) {
+        c.pos = f.cont_end;
+        f.cont_end = 0;
+      }
+      bool more = false;
+      if (c.pos < f.end) {
+        uint32_t save = c.pos;
+        uint64_t ntag;
+        if (!read_varint(c, &ntag)) return false;
+        if ((uint32_t)(ntag >> 3) == f.cont_num) {
+          more = true;
+        } else {
+          c.pos = save;
+        }
+      }
+      if (!more) {
+        if (!putc_(c, f.mode == FM_STRUCT ? '}' : '}')) return false;
+        f.cont_kind = CK_NONE;
+        if (f.mode == FM_STRUCT) {
+          // the struct body IS the map; pop the frame
+          // (fall through to the body loop which will see pos==end)
+        }
+        continue;
+      }
+      if (!putc_(c, ',')) return false;
+      // start the next entry
+      uint64_t elen;
+      if (!read_varint(c, &elen)) return false;
+      uint32_t eend = c.pos + (uint32_t)elen;
+      if (eend > c.len) return dfail(c, E_PARSE);
+      int pushed = 0;
+      if (!map_entry_step(c, f, eend, stack, sp, &pushed)) return false;
+      continue;
+    }
+
+    // -- body loops ------------------------------------------------------
+    if (f.mode == FM_LIST) {
+      if (c.pos >= f.end) {
+        if (!putc_(c, ']')) return false;
+        --sp;
+        continue;
+      }
+      uint64_t tag;
+      if (!read_varint(c, &tag)) return false;
+      if ((uint32_t)(tag >> 3) != 1 || (tag & 7) != W_LEN) {
+        if (!skip_wire(c, (uint32_t)(tag & 7))) return false;
+        continue;
+      }
+      uint64_t elen;
+      if (!read_varint(c, &elen)) return false;
+      uint32_t eend = c.pos + (uint32_t)elen;
+      if (eend > c.len) return dfail(c, E_PARSE);
+      if (!f.first_member && !putc_(c, ',')) return false;
+      f.first_member = 0;
+      const MsgEntry& m = c.t.msgs[f.msg_idx];
+      const FieldEntry& vf = c.t.fields[m.field_start];
+      uint8_t pm = FM_BODY;
+      int32_t pidx = vf.sub_index;
+      uint32_t bend = eend;
+      int r = enter_body(c, vf.sub_index, eend, &pm, &pidx, &bend);
+      if (r == 0) return false;
+      if (r == 1) {
+        c.pos = eend;
+        continue;
+      }
+      if (sp >= MAX_RECURSE) return dfail(c, E_LIMIT);
+      DFrame& nf = stack[sp++];
+      nf.end = bend;
+      nf.prev_number = 0;
+      nf.cont_end = 0;
+      nf.msg_idx = pidx;
+      nf.cont_field = -1;
+      nf.cont_num = 0;
+      nf.mode = pm;
+      nf.first_member = 1;
+      nf.cont_kind = CK_NONE;
+      nf.cont_first = 1;
+      if (!putc_(c, pm == FM_LIST ? '[' : '{')) return false;
+      continue;
+    }
+    if (f.mode == FM_STRUCT) {
+      // map<string, Value> on field 1 rendered as a bare object
+      if (c.pos >= f.end) {
+        if (!putc_(c, '}')) return false;
+        --sp;
+        continue;
+      }
+      uint64_t tag;
+      if (!read_varint(c, &tag)) return false;
+      if ((uint32_t)(tag >> 3) != 1 || (tag & 7) != W_LEN) {
+        if (!skip_wire(c, (uint32_t)(tag & 7))) return false;
+        continue;
+      }
+      uint64_t elen;
+      if (!read_varint(c, &elen)) return false;
+      uint32_t eend = c.pos + (uint32_t)elen;
+      if (eend > c.len) return dfail(c, E_PARSE);
+      if (!f.first_member && !putc_(c, ',')) return false;
+      f.first_member = 0;
+      const MsgEntry& m = c.t.msgs[f.msg_idx];
+      f.cont_field = m.field_start;  // entries field of the struct map
+      f.cont_num = 1;
+      f.cont_kind = CK_MAP;  // struct entries use map machinery, sans '{'
+      int pushed = 0;
+      if (!map_entry_step(c, f, eend, stack, sp, &pushed)) return false;
+      continue;
+    }
+
+    // FM_BODY
+    if (c.pos >= f.end) {
+      if (!putc_(c, '}')) return false;
+      --sp;
+      continue;
+    }
     uint64_t tag;
     if (!read_varint(c, &tag)) return false;
     uint32_t num = (uint32_t)(tag >> 3), wt = (uint32_t)(tag & 7);
-    const FieldEntry* f = find_field(c, m, num);
-    if (f == nullptr) {
+    const MsgEntry& m = c.t.msgs[f.msg_idx];
+    const FieldEntry* fe = find_field(c, m, num);
+    if (fe == nullptr) {
       if (!skip_wire(c, wt)) return false;  // unknown field: dropped
       continue;
     }
-    // out-of-order or non-adjacent duplicate wire fields (no standard
-    // serializer produces either; protobuf last-wins semantics would need a
-    // buffering pass) -> host transcodes this request
-    if (num <= prev_number) return dfail(c, E_UNSUPPORTED);
-    prev_number = num;
-    bool repeated = (f->flags & F_REPEATED) != 0;
-    bool is_map = (f->flags & F_MAP) != 0;
-    // default-omission for singular non-presence scalars (proto3 semantics:
-    // protojson prints nothing for unset/default non-presence fields)
-    if (!repeated && !(f->flags & F_HAS_PRESENCE) && wt == W_VARINT) {
-      uint32_t save = c.pos;
-      uint64_t v;
-      if (!read_varint(c, &v)) return false;
-      if (v == 0) continue;  // default -> omit
-      c.pos = save;
-    }
-    if (!repeated && !(f->flags & F_HAS_PRESENCE) && wt == W_LEN &&
-        (f->kind == K_STRING || f->kind == K_BYTES)) {
-      uint32_t save = c.pos;
-      uint64_t v;
-      if (!read_varint(c, &v)) return false;
-      if (v == 0) continue;  // empty string/bytes -> omit
-      c.pos = save;
-    }
-    if (!repeated && !(f->flags & F_HAS_PRESENCE) &&
-        (wt == W_I64 || wt == W_I32)) {
-      uint32_t save = c.pos;
-      uint64_t v = 1;
-      if (wt == W_I64) {
-        if (!read_fixed64(c, &v)) return false;
-      } else {
-        uint32_t v32;
-        if (!read_fixed32(c, &v32)) return false;
-        v = v32;
+    // out-of-order / non-adjacent duplicates -> host transcodes
+    if (num <= f.prev_number) return dfail(c, E_UNSUPPORTED);
+    f.prev_number = num;
+    bool repeated = (fe->flags & F_REPEATED) != 0;
+    bool is_map = (fe->flags & F_MAP) != 0;
+    // proto3 default-omission for singular non-presence fields
+    if (!repeated && !(fe->flags & F_HAS_PRESENCE)) {
+      if (wt == W_VARINT) {
+        uint32_t save = c.pos;
+        uint64_t v;
+        if (!read_varint(c, &v)) return false;
+        if (v == 0) continue;
+        c.pos = save;
+      } else if (wt == W_LEN && (fe->kind == K_STRING || fe->kind == K_BYTES)) {
+        uint32_t save = c.pos;
+        uint64_t v;
+        if (!read_varint(c, &v)) return false;
+        if (v == 0) continue;
+        c.pos = save;
+      } else if (wt == W_I64 || wt == W_I32) {
+        uint32_t save = c.pos;
+        uint64_t v = 1;
+        if (wt == W_I64) {
+          if (!read_fixed64(c, &v)) return false;
+        } else {
+          uint32_t v32;
+          if (!read_fixed32(c, &v32)) return false;
+          v = v32;
+        }
+        if (v == 0) continue;
+        c.pos = save;
       }
-      if (v == 0) continue;  // +0.0 / 0 fixed -> omit
-      c.pos = save;
     }
-    if (!first_member && !putc_(c, ',')) return false;
-    first_member = false;
+    if (!f.first_member && !putc_(c, ',')) return false;
+    f.first_member = 0;
     if (!putc_(c, '"')) return false;
-    if (!puts_(c, (const char*)(c.t.names + f->json_off), f->json_len))
+    if (!puts_(c, (const char*)(c.t.names + fe->json_off), fe->json_len))
       return false;
     if (!putc_(c, '"')) return false;
     if (!putc_(c, ':')) return false;
 
     if (is_map) {
       if (!putc_(c, '{')) return false;
-      bool first = true;
-      // current + adjacent entries of the same field number
-      while (true) {
-        uint64_t elen;
-        if (!read_varint(c, &elen)) return false;
-        uint32_t eend = c.pos + (uint32_t)elen;
-        if (eend > c.len) return dfail(c, E_PARSE);
-        if (!first && !putc_(c, ',')) return false;
-        first = false;
-        if (!emit_map_entry(c, *f, eend, depth)) return false;
-        c.pos = eend;
-        if (c.pos >= end) break;
-        uint32_t save = c.pos;
-        uint64_t ntag;
-        if (!read_varint(c, &ntag)) return false;
-        if ((uint32_t)(ntag >> 3) != num) {
-          c.pos = save;
-          break;
-        }
-      }
-      if (!putc_(c, '}')) return false;
-    } else if (repeated) {
-      if (!putc_(c, '[')) return false;
-      bool first = true;
-      while (true) {
-        if (wt == W_LEN && expected_wire(*f) != W_LEN) {
-          // packed payload
-          uint64_t plen;
-          if (!read_varint(c, &plen)) return false;
-          uint32_t pend = c.pos + (uint32_t)plen;
-          if (pend > c.len) return dfail(c, E_PARSE);
-          if (!emit_packed(c, *f, pend, &first)) return false;
-        } else {
-          if (!first && !putc_(c, ',')) return false;
-          first = false;
-          if (!emit_value(c, *f, depth)) return false;
-        }
-        if (c.pos >= end) break;
-        uint32_t save = c.pos;
-        uint64_t ntag;
-        if (!read_varint(c, &ntag)) return false;
-        if ((uint32_t)(ntag >> 3) != num) {
-          c.pos = save;
-          break;
-        }
-        wt = (uint32_t)(ntag & 7);
-      }
-      if (!putc_(c, ']')) return false;
-    } else {
-      if (wt != expected_wire(*f)) return dfail(c, E_PARSE);
-      if (!emit_value(c, *f, depth)) return false;
+      uint64_t elen;
+      if (!read_varint(c, &elen)) return false;
+      uint32_t eend = c.pos + (uint32_t)elen;
+      if (eend > c.len) return dfail(c, E_PARSE);
+      f.cont_field = (int32_t)(fe - c.t.fields);
+      f.cont_num = num;
+      f.cont_kind = CK_MAP;
+      int pushed = 0;
+      if (!map_entry_step(c, f, eend, stack, sp, &pushed)) return false;
+      continue;
     }
+    if (repeated) {
+      if (!putc_(c, '[')) return false;
+      f.cont_field = (int32_t)(fe - c.t.fields);
+      f.cont_num = num;
+      f.cont_kind = CK_ARRAY;
+      // first element (wt already read)
+      if (wt == W_LEN && expected_wire(*fe) != W_LEN) {
+        uint64_t plen;
+        if (!read_varint(c, &plen)) return false;
+        uint32_t pend = c.pos + (uint32_t)plen;
+        if (pend > c.len) return dfail(c, E_PARSE);
+        bool first = true;
+        if (!emit_packed(c, *fe, pend, &first)) return false;
+        continue;
+      }
+      if (fe->kind == K_MESSAGE) {
+        uint64_t v;
+        if (!read_varint(c, &v)) return false;
+        if (c.pos + v > c.len) return dfail(c, E_PARSE);
+        uint32_t vend = c.pos + (uint32_t)v;
+        uint8_t pm = FM_BODY;
+        int32_t pidx = fe->sub_index;
+        uint32_t bend = vend;
+        int r = enter_body(c, fe->sub_index, vend, &pm, &pidx, &bend);
+        if (r == 0) return false;
+        if (r == 1) {
+          c.pos = vend;
+          continue;
+        }
+        if (sp >= MAX_RECURSE) return dfail(c, E_LIMIT);
+        DFrame& nf = stack[sp++];
+        nf.end = bend;
+        nf.prev_number = 0;
+        nf.cont_end = 0;
+        nf.msg_idx = pidx;
+        nf.cont_field = -1;
+        nf.cont_num = 0;
+        nf.mode = pm;
+        nf.first_member = 1;
+        nf.cont_kind = CK_NONE;
+        nf.cont_first = 1;
+        if (!putc_(c, pm == FM_LIST ? '[' : '{')) return false;
+        continue;
+      }
+      if (wt != expected_wire(*fe)) return dfail(c, E_PARSE);
+      if (!emit_scalar_value(c, *fe)) return false;
+      continue;
+    }
+    // singular message or scalar
+    if (fe->kind == K_MESSAGE) {
+      if (wt != W_LEN) return dfail(c, E_PARSE);
+      uint64_t v;
+      if (!read_varint(c, &v)) return false;
+      if (c.pos + v > c.len) return dfail(c, E_PARSE);
+      uint32_t vend = c.pos + (uint32_t)v;
+      uint8_t pm = FM_BODY;
+      int32_t pidx = fe->sub_index;
+      uint32_t bend = vend;
+      int r = enter_body(c, fe->sub_index, vend, &pm, &pidx, &bend);
+      if (r == 0) return false;
+      if (r == 1) {
+        c.pos = vend;
+        continue;
+      }
+      if (sp >= MAX_RECURSE) return dfail(c, E_LIMIT);
+      DFrame& nf = stack[sp++];
+      nf.end = bend;
+      nf.prev_number = 0;
+      nf.cont_end = 0;
+      nf.msg_idx = pidx;
+      nf.cont_field = -1;
+      nf.cont_num = 0;
+      nf.mode = pm;
+      nf.first_member = 1;
+      nf.cont_kind = CK_NONE;
+      nf.cont_first = 1;
+      if (!putc_(c, pm == FM_LIST ? '[' : '{')) return false;
+      continue;
+    }
+    if (wt != expected_wire(*fe)) return dfail(c, E_PARSE);
+    if (!emit_scalar_value(c, *fe)) return false;
   }
-  return putc_(c, '}');
+  return c.status == E_OK;
 }
 
 // ---------------------------------------------------------------------------
@@ -1072,13 +1353,7 @@ extern "C" __global__ void __launch_bounds__(WPB * WAVE) k_pb2json(
     c.t = t;
     c.status = E_OK;
     c.lane = lane;
-    const MsgEntry& m = t.msgs[msg_idx_arr[req]];
-    bool top_is_scalar_wkt =
-        m.wkt_kind == WKT_TIMESTAMP || m.wkt_kind == WKT_DURATION ||
-        m.wkt_kind == WKT_WRAPPER || m.wkt_kind == WKT_VALUE ||
-        m.wkt_kind == WKT_FIELDMASK || m.wkt_kind == WKT_LISTVALUE;
-    (void)top_is_scalar_wkt;
-    decode_message(c, msg_idx_arr[req], c.len, 0);
+    decode_walk(c, msg_idx_arr[req], c.len);
     if (c.status != E_OK) {
       r.status = c.status;
       if (!lane) results[req] = r;
