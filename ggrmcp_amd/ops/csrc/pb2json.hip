@@ -28,7 +28,11 @@
 // optimizer blowup.  Inside each noinline function the tiny primitives
 // (putc_/puts_/read_varint/put_escaped) still inline, so the per-byte hot
 // loops keep their state in registers; the call overhead is per-field.
+#ifdef GGRMCP_HOST_SIM
+#define DEVN __attribute__((noinline))
+#else
 #define DEVN __device__ __noinline__
+#endif
 
 #ifndef WPB
 #define WPB 4
