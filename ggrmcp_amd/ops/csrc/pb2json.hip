@@ -1074,12 +1074,10 @@ DEV bool decode_walk(DCtx& c, int top_msg_idx, uint32_t top_end) {
         }
       }
       if (!more) {
-        if (!putc_(c, f.mode == FM_STRUCT ? '}' : '}')) return false;
+        // a Struct's entries ARE its body: the body step emits the closing
+        // brace when pos reaches end (emitting here double-closed it)
+        if (f.mode != FM_STRUCT && !putc_(c, '}')) return false;
         f.cont_kind = CK_NONE;
-        if (f.mode == FM_STRUCT) {
-          // the struct body IS the map; pop the frame
-          // (fall through to the body loop which will see pos==end)
-        }
         continue;
       }
       if (!putc_(c, ',')) return false;

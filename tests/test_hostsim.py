@@ -183,3 +183,38 @@ def test_envelope_with_real_reply(env):
     assert resp["result"]["isError"] is False
     inner = json.loads(resp["result"]["content"][0]["text"])
     assert inner == {"message": "Hello, w!"}
+
+
+def test_struct_value_roundtrip_hostsim():
+    """Same case as the GPU twin (test_gpu_transcode.test_struct_value_
+    roundtrip): nested Struct/Value/ListValue through both kernels."""
+    from ggrmcp_amd.utils.protobuild import FileBuilder
+    from ggrmcp_amd.descriptors.loader import build_pool as bp, extract_method_infos
+
+    fb = FileBuilder("t/struct.proto", "t")
+    fb.add_dependency("google/protobuf/struct.proto")
+    fb.message("Holder").field("data", 1, "message",
+                               message="google.protobuf.Struct").done()
+    fb.service("S").method("M", "Holder", "Holder").done()
+    fdp = fb.build()
+    pool2 = bp([fdp])
+    infos2 = {m.tool_name(): m
+              for m in extract_method_infos([fdp], pool2, compat_names=False)}
+    eng2 = HostSimEngine(infos2)
+    cpu2 = CpuTranscoder()
+    payload = {"data": {"s": "str", "n": 2.5, "b": True, "z": None,
+                        "arr": [1, "two", False, {"k": "v"}],
+                        "obj": {"nested": {"deep": [1, 2]}}}}
+    desc = pool2.FindMessageTypeByName("t.Holder")
+    text = json.dumps(payload)
+    idx = eng2.tables.msg_index["t.Holder"]
+    enc, pbs = eng2.encode_batch([text.encode()], mode=1, msg_indices=[idx])
+    assert enc[0]["status"] == 0, enc[0]
+    g = json_format.MessageToDict(cpu2.pb_to_message(desc, pbs[0]))
+    o = json_format.MessageToDict(cpu2.pb_to_message(desc, cpu2.json_to_pb(desc, text)))
+    assert _approx(g, o), f"\nsim: {g}\noracle: {o}"
+    wire = cpu2.json_to_pb(desc, text)
+    dec, outs = eng2.decode_batch([wire], [idx], mode=1)
+    assert dec[0]["status"] == 0, dec[0]
+    assert _approx(json.loads(outs[0]), json.loads(cpu2.pb_to_json(desc, wire))), \
+        f"\nsim: {outs[0]!r}\noracle: {cpu2.pb_to_json(desc, wire)!r}"

@@ -146,3 +146,57 @@ def _roundtrip(msg_name, payload):
     gj = json.loads(outs[0])
     oj = json.loads(_cpu.pb_to_json(desc, oracle_wire))
     assert _approx(gj, oj), f"\npayload: {text!r}\nsim:    {gj}\noracle: {oj}"
+
+
+# ---- arbitrary JSON through google.protobuf.Struct --------------------------
+
+from ggrmcp_amd.utils.protobuild import FileBuilder as _FB
+
+_fb = _FB("t/struct.proto", "t")
+_fb.add_dependency("google/protobuf/struct.proto")
+_fb.message("Holder").field("data", 1, "message",
+                            message="google.protobuf.Struct").done()
+_fb.service("S").method("M", "Holder", "Holder").done()
+_sfdp = _fb.build()
+_spool = build_pool([_sfdp])
+from ggrmcp_amd.descriptors.loader import extract_method_infos as _emi
+_sengine = HostSimEngine({m.tool_name(): m
+                          for m in _emi([_sfdp], _spool, compat_names=False)})
+_sdesc = _spool.FindMessageTypeByName("t.Holder")
+
+_json_value = st.recursive(
+    st.one_of(st.none(), st.booleans(),
+              st.floats(allow_nan=False, allow_infinity=False, width=64),
+              _text),
+    lambda v: st.one_of(st.lists(v, max_size=4),
+                        st.dictionaries(_ascii, v, max_size=4)),
+    max_leaves=12,
+)
+
+
+@settings(max_examples=100, deadline=None)
+@given(value=st.dictionaries(_ascii, _json_value, max_size=5))
+def test_fuzz_struct_value(value):
+    payload = {"data": value}
+    text = json.dumps(payload, ensure_ascii=False)
+    idx = _sengine.tables.msg_index["t.Holder"]
+    enc, pbs = _sengine.encode_batch([text.encode()], mode=1, msg_indices=[idx],
+                                     enforce=False)
+    try:
+        oracle_wire = _cpu.json_to_pb(_sdesc, text)
+    except Exception:
+        assert enc[0]["status"] != 0
+        return
+    if enc[0]["status"] == 6:
+        return
+    assert enc[0]["status"] == 0, f"status {enc[0]['status']} for {text!r}"
+    g = json_format.MessageToDict(_cpu.pb_to_message(_sdesc, pbs[0]))
+    o = json_format.MessageToDict(_cpu.pb_to_message(_sdesc, oracle_wire))
+    assert _approx(g, o), f"\npayload: {text!r}\nsim:    {g}\noracle: {o}"
+    dec, outs = _sengine.decode_batch([oracle_wire], [idx], mode=1)
+    if dec[0]["status"] == 6:
+        return
+    assert dec[0]["status"] == 0, f"decode status {dec[0]['status']} for {text!r}"
+    gj = json.loads(outs[0])
+    oj = json.loads(_cpu.pb_to_json(_sdesc, oracle_wire))
+    assert _approx(gj, oj), f"\npayload: {text!r}\nsim:    {gj}\noracle: {oj}"
