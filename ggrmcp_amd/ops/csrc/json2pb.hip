@@ -27,6 +27,14 @@
 
 #include "common.h"
 
+#ifndef DEVN
+#ifdef GGRMCP_HOST_SIM
+#define DEVN __attribute__((noinline))
+#else
+#define DEVN __device__ __noinline__
+#endif
+#endif
+
 #ifndef WPB
 #define WPB 4
 #endif  // waves (=requests) per 256-thread block
@@ -687,9 +695,6 @@ DEV bool parse_duration(const uint8_t* s, uint32_t n, int64_t* secs, int32_t* na
 // recursive encoder
 // ---------------------------------------------------------------------------
 
-DEV bool encode_message(Ctx& c, int msg_idx, int depth);
-DEV bool encode_json_value_as_value(Ctx& c, int depth);  // google.protobuf.Value
-
 // encode a string field's payload (tag already emitted): LEN slot + bytes
 DEV bool encode_string_payload(Ctx& c, uint32_t start, uint32_t rawlen, bool esc) {
   uint32_t slot;
@@ -707,8 +712,8 @@ DEV bool encode_string_payload(Ctx& c, uint32_t start, uint32_t rawlen, bool esc
   return backfill_len(c, slot);
 }
 
-// encode ONE singular value (not null) for field f; tag included.
-DEV bool encode_single(Ctx& c, const FieldEntry& f, int depth) {
+
+DEVN bool encode_scalar_field(Ctx& c, const FieldEntry& f) {
   switch (f.kind) {
     case K_STRING: {
       uint32_t st, rl;
@@ -808,59 +813,6 @@ DEV bool encode_single(Ctx& c, const FieldEntry& f, int depth) {
       if (!emit_tag(c, f.number, W_VARINT)) return false;
       return emit_varint(c, (uint64_t)(int64_t)number);
     }
-    case K_MESSAGE: {
-      const MsgEntry& sub = c.t.msgs[f.sub_index];
-      // WKTs that encode from non-object JSON
-      if (sub.wkt_kind == WKT_TIMESTAMP || sub.wkt_kind == WKT_DURATION) {
-        uint32_t st, rl;
-        bool esc;
-        if (!string_span(c, &st, &rl, &esc)) return false;
-        if (esc) return fail(c, E_UNSUPPORTED);
-        int64_t secs;
-        int32_t nanos;
-        bool ok = sub.wkt_kind == WKT_TIMESTAMP
-                      ? parse_timestamp(c, c.s + st, rl, &secs, &nanos)
-                      : parse_duration(c.s + st, rl, &secs, &nanos);
-        if (!ok) return fail(c, E_INVALID_PARAMS, (int)f.number);
-        if (!emit_tag(c, f.number, W_LEN)) return false;
-        uint32_t slot;
-        if (!reserve_len(c, &slot)) return false;
-        if (secs) {
-          if (!emit_varint(c, (1u << 3) | W_VARINT)) return false;
-          if (!emit_varint(c, (uint64_t)secs)) return false;
-        }
-        if (nanos) {
-          if (!emit_varint(c, (2u << 3) | W_VARINT)) return false;
-          if (!emit_varint(c, (uint64_t)(int64_t)nanos)) return false;
-        }
-        return backfill_len(c, slot);
-      }
-      if (sub.wkt_kind == WKT_WRAPPER) {
-        if (!emit_tag(c, f.number, W_LEN)) return false;
-        uint32_t slot;
-        if (!reserve_len(c, &slot)) return false;
-        const FieldEntry& inner = c.t.fields[sub.field_start];
-        if (!encode_single(c, inner, depth + 1)) return false;
-        return backfill_len(c, slot);
-      }
-      if (sub.wkt_kind == WKT_VALUE) {
-        if (!emit_tag(c, f.number, W_LEN)) return false;
-        uint32_t slot;
-        if (!reserve_len(c, &slot)) return false;
-        if (!encode_json_value_as_value(c, depth + 1)) return false;
-        return backfill_len(c, slot);
-      }
-      if (sub.wkt_kind == WKT_STRUCT || sub.wkt_kind == WKT_LISTVALUE ||
-          sub.wkt_kind == WKT_EMPTY || sub.wkt_kind == WKT_NONE ||
-          sub.wkt_kind == WKT_FIELDMASK) {
-        if (!emit_tag(c, f.number, W_LEN)) return false;
-        uint32_t slot;
-        if (!reserve_len(c, &slot)) return false;
-        if (!encode_message(c, f.sub_index, depth + 1)) return false;
-        return backfill_len(c, slot);
-      }
-      return fail(c, E_UNSUPPORTED);  // Any
-    }
     default: {  // integer kinds
       NumVal nv;
       int nonf;
@@ -941,6 +893,7 @@ DEV bool encode_single(Ctx& c, const FieldEntry& f, int depth) {
     }
   }
 }
+
 
 // packed element (no tag) for numeric repeated fields
 DEV bool encode_packed_element(Ctx& c, const FieldEntry& f) {
@@ -1061,364 +1014,662 @@ DEV bool is_packable(uint8_t kind) {
          kind != K_GROUP;
 }
 
-// map entry: key text (string span semantics per key kind) then value
-DEV bool encode_map_entries(Ctx& c, const FieldEntry& f, int depth) {
-  const MsgEntry& entry_msg = c.t.msgs[f.sub_index];
-  const FieldEntry& kf = c.t.fields[entry_msg.field_start];
-  const FieldEntry& vf = c.t.fields[entry_msg.field_start + 1];
-  if (!expect(c, '{')) return false;
-  skip_ws(c);
-  if (peek(c) == '}') {
-    c.pos++;
-    return true;
-  }
-  while (true) {
-    uint32_t kst, krl;
-    bool kesc;
-    if (!string_span(c, &kst, &krl, &kesc)) return false;
-    if (kesc) return fail(c, E_UNSUPPORTED);
-    if (!expect(c, ':')) return false;
-    if (!emit_tag(c, f.number, W_LEN)) return false;
-    uint32_t slot;
-    if (!reserve_len(c, &slot)) return false;
-    // key (field 1)
-    switch (kf.kind) {
-      case K_STRING: {
-        if (!emit_tag(c, 1, W_LEN)) return false;
-        if (!encode_string_payload(c, kst, krl, false)) return false;
-        break;
-      }
-      case K_BOOL: {
-        uint64_t v;
-        if (krl == 4 && c.s[kst] == 't') v = 1;
-        else if (krl == 5 && c.s[kst] == 'f') v = 0;
-        else return fail(c, E_INVALID_PARAMS, (int)f.number);
-        if (!emit_tag(c, 1, W_VARINT)) return false;
-        if (!emit_varint(c, v)) return false;
-        break;
-      }
-      default: {  // integer keys
-        NumVal nv;
-        if (!parse_number_text(c.s, kst, kst + krl, &nv))
-          return fail(c, E_INVALID_PARAMS, (int)f.number);
-        int64_t sv;
-        uint64_t uv;
-        switch (kf.kind) {
-          case K_INT64:
-          case K_SFIXED64:
-          case K_SINT64:
-          case K_INT32:
-          case K_SINT32:
-          case K_SFIXED32:
-            if (!num_to_i64(nv, &sv)) return fail(c, E_INVALID_PARAMS, (int)f.number);
-            break;
-          default:
-            if (!num_to_u64(nv, &uv)) return fail(c, E_INVALID_PARAMS, (int)f.number);
-            sv = (int64_t)uv;
-        }
-        if (kf.kind == K_SINT64) uv = zigzag64(sv);
-        else if (kf.kind == K_SINT32) uv = zigzag32((int32_t)sv);
-        else uv = (uint64_t)sv;
-        if (kf.kind == K_FIXED64 || kf.kind == K_SFIXED64) {
-          if (!emit_tag(c, 1, W_I64)) return false;
-          if (!emit_fixed64(c, uv)) return false;
-        } else if (kf.kind == K_FIXED32 || kf.kind == K_SFIXED32) {
-          if (!emit_tag(c, 1, W_I32)) return false;
-          if (!emit_fixed32(c, (uint32_t)uv)) return false;
-        } else {
-          if (!emit_tag(c, 1, W_VARINT)) return false;
-          if (!emit_varint(c, uv)) return false;
-        }
-      }
-    }
-    // value (field 2)
-    skip_ws(c);
-    if (literal_at(c, "null", 4) && vf.kind != K_MESSAGE) {
-      c.pos += 4;  // null value -> default (omit)
-    } else if (!encode_single(c, vf, depth + 1)) {
-      return false;
-    }
-    if (!backfill_len(c, slot)) return false;
-    skip_ws(c);
-    uint8_t ch = peek(c);
-    if (ch == ',') {
-      c.pos++;
-      skip_ws(c);
-      continue;
-    }
-    if (ch == '}') {
-      c.pos++;
-      return true;
-    }
-    return fail(c, E_PARSE);
-  }
-}
 
-// encode the JSON object at c.pos as message msg_idx (payload only; caller
-// owns any surrounding tag/len)
-DEV bool encode_message(Ctx& c, int msg_idx, int depth) {
-  if (depth > MAX_RECURSE || (c.lim.enforce && (uint32_t)depth > c.lim.max_depth))
-    return fail(c, E_LIMIT, 1);
-  const MsgEntry& m = c.t.msgs[msg_idx];
-  if (m.wkt_kind == WKT_STRUCT) {
-    // Struct = map<string, Value> on field 1
-    if (!expect(c, '{')) return false;
-    skip_ws(c);
-    if (peek(c) == '}') {
-      c.pos++;
-      return true;
-    }
-    while (true) {
-      uint32_t kst, krl;
-      bool kesc;
-      if (!string_span(c, &kst, &krl, &kesc)) return false;
-      if (!expect(c, ':')) return false;
-      if (!emit_tag(c, 1, W_LEN)) return false;
-      uint32_t slot;
-      if (!reserve_len(c, &slot)) return false;
-      if (!emit_tag(c, 1, W_LEN)) return false;  // entry key
-      if (!encode_string_payload(c, kst, krl, kesc)) return false;
-      if (!emit_tag(c, 2, W_LEN)) return false;  // entry value (Value)
-      uint32_t vslot;
-      if (!reserve_len(c, &vslot)) return false;
-      if (!encode_json_value_as_value(c, depth + 1)) return false;
-      if (!backfill_len(c, vslot)) return false;
-      if (!backfill_len(c, slot)) return false;
-      skip_ws(c);
-      uint8_t ch = peek(c);
-      if (ch == ',') {
-        c.pos++;
-        skip_ws(c);
-        continue;
-      }
-      if (ch == '}') {
-        c.pos++;
-        return true;
-      }
-      return fail(c, E_PARSE);
-    }
-  }
-  if (m.wkt_kind == WKT_LISTVALUE) {
-    if (!expect(c, '[')) return false;
-    skip_ws(c);
-    if (peek(c) == ']') {
-      c.pos++;
-      return true;
-    }
-    while (true) {
-      if (!emit_tag(c, 1, W_LEN)) return false;
-      uint32_t slot;
-      if (!reserve_len(c, &slot)) return false;
-      if (!encode_json_value_as_value(c, depth + 1)) return false;
-      if (!backfill_len(c, slot)) return false;
-      skip_ws(c);
-      uint8_t ch = peek(c);
-      if (ch == ',') {
-        c.pos++;
-        continue;
-      }
-      if (ch == ']') {
-        c.pos++;
-        return true;
-      }
-      return fail(c, E_PARSE);
-    }
-  }
-  if (m.wkt_kind == WKT_FIELDMASK) {
-    // JSON: comma-joined camelCase paths -> repeated snake_case strings
-    uint32_t st, rl;
-    bool esc;
-    if (!string_span(c, &st, &rl, &esc)) return false;
-    if (esc) return fail(c, E_UNSUPPORTED);
-    uint32_t i = 0;
-    while (i < rl) {
-      uint32_t j = i;
-      while (j < rl && c.s[st + j] != ',') ++j;
-      if (j > i) {
-        if (!emit_tag(c, 1, W_LEN)) return false;
-        uint32_t slot;
-        if (!reserve_len(c, &slot)) return false;
-        for (uint32_t k = i; k < j; ++k) {
-          uint8_t ch = c.s[st + k];
-          if (ch >= 'A' && ch <= 'Z') {
-            if (c.opos + 2 > c.ocap) return fail(c, E_OVERFLOW);
-            if (!c.lane) {
-              c.out[c.opos] = '_';
-              c.out[c.opos + 1] = ch + 32;
-            }
-            c.opos += 2;
-          } else {
-            if (c.opos + 1 > c.ocap) return fail(c, E_OVERFLOW);
-            if (!c.lane) c.out[c.opos] = ch;
-            c.opos += 1;
-          }
-        }
-        if (!backfill_len(c, slot)) return false;
-      }
-      i = j + 1;
-    }
-    return true;
-  }
-  if (m.wkt_kind == WKT_EMPTY) {
-    if (!expect(c, '{')) return false;
-    skip_ws(c);
-    if (!expect(c, '}')) return false;
-    return true;
-  }
-  // plain message (also reached for top-level Timestamp etc. via caller)
-  if (!expect(c, '{')) return false;
-  skip_ws(c);
-  uint64_t seen_fields = 0;  // duplicate-key rejection (first 64 fields)
-  uint32_t seen_oneofs = 0;  // oneof exclusivity (protojson errors)
-  if (peek(c) == '}') {
-    c.pos++;
-    return true;
-  }
-  while (true) {
-    // ---- key ----
-    uint32_t kst, krl;
-    bool kesc;
-    if (!string_span(c, &kst, &krl, &kesc)) return false;
-    const uint8_t* kptr = c.s + kst;
-    uint32_t klen = krl;
-    if (kesc) {
-      uint32_t n = unescape_serial(c, c.s + kst, krl, c.keybuf, 192);
-      if (n == 0xFFFFFFFF) return fail(c, E_PARSE);
-      __builtin_amdgcn_wave_barrier();
-      kptr = c.keybuf;
-      klen = n;
-    }
-    if (!expect(c, ':')) return false;
-    uint64_t h = fnv1a64(kptr, klen);
-    int fidx = -1;
-    for (int i = 0; i < m.field_count; ++i) {
-      const FieldEntry& f = c.t.fields[m.field_start + i];
-      if ((f.hash_json == h && f.json_len == klen &&
-           wave_equal(c, c.t.names + f.json_off, kptr, klen)) ||
-          (f.hash_orig == h && f.name_len == klen &&
-           wave_equal(c, c.t.names + f.name_off, kptr, klen))) {
-        fidx = i;
-        break;
-      }
-    }
-    if (fidx < 0) {
-      c.err_pos = kst;
-      return fail(c, E_INVALID_PARAMS, -1);  // unknown field: protojson rejects
-    }
-    const FieldEntry& f = c.t.fields[m.field_start + fidx];
-    if (fidx < 64) {
-      if (seen_fields & (1ull << fidx)) return fail(c, E_INVALID_PARAMS, (int)f.number);
-      seen_fields |= 1ull << fidx;
-    }
-    if (f.flags & F_ONEOF) {
-      if (seen_oneofs & (1u << f.oneof_id))
-        return fail(c, E_INVALID_PARAMS, (int)f.number);
-      seen_oneofs |= 1u << f.oneof_id;
-    }
-    // ---- value ----
-    skip_ws(c);
-    if (literal_at(c, "null", 4) &&
-        !(f.kind == K_MESSAGE && c.t.msgs[f.sub_index].wkt_kind == WKT_VALUE)) {
-      c.pos += 4;  // null -> unset (protojson), incl. null oneof member
-    } else if (f.flags & F_MAP) {
-      if (!encode_map_entries(c, f, depth)) return false;
-    } else if (f.flags & F_REPEATED) {
-      if (!expect(c, '[')) return false;
-      skip_ws(c);
-      if (peek(c) == ']') {
-        c.pos++;
-      } else if (is_packable(f.kind)) {
-        if (!emit_tag(c, f.number, W_LEN)) return false;
-        uint32_t slot;
-        if (!reserve_len(c, &slot)) return false;
-        while (true) {
-          skip_ws(c);
-          if (!encode_packed_element(c, f)) return false;
-          skip_ws(c);
-          uint8_t ch = peek(c);
-          if (ch == ',') {
-            c.pos++;
-            continue;
-          }
-          if (ch == ']') {
-            c.pos++;
-            break;
-          }
-          return fail(c, E_PARSE);
-        }
-        if (!backfill_len(c, slot)) return false;
-      } else {
-        while (true) {
-          skip_ws(c);
-          if (!encode_single(c, f, depth)) return false;
-          skip_ws(c);
-          uint8_t ch = peek(c);
-          if (ch == ',') {
-            c.pos++;
-            continue;
-          }
-          if (ch == ']') {
-            c.pos++;
-            break;
-          }
-          return fail(c, E_PARSE);
-        }
-      }
-    } else {
-      if (!encode_single(c, f, depth)) return false;
-    }
-    skip_ws(c);
-    uint8_t ch = peek(c);
-    if (ch == ',') {
-      c.pos++;
-      skip_ws(c);
-      continue;
-    }
-    if (ch == '}') {
-      c.pos++;
-      return true;
-    }
-    return fail(c, E_PARSE);
-  }
-}
+// ---------------------------------------------------------------------------
+// encoder — ITERATIVE walker (see pb2json.hip for the rationale: the
+// recursive form kept the whole context in scratch memory, costing ~170
+// cycles per processed byte; this flat walker keeps hot state in VGPRs and
+// removes the dynamic device stack).  Nested-message length prefixes are
+// 3-byte non-minimal varints backfilled on frame pop, so no sizing pass.
+// ---------------------------------------------------------------------------
 
-// google.protobuf.Value encoder: any JSON value -> Value message payload
-DEV bool encode_json_value_as_value(Ctx& c, int depth) {
-  if (depth > MAX_RECURSE || (c.lim.enforce && (uint32_t)depth > c.lim.max_depth))
-    return fail(c, E_LIMIT, 1);
+enum : uint8_t { EM_BODY = 0, EM_STRUCT = 1, EM_LIST = 2 };
+// resume states
+enum : uint8_t {
+  RB_KEY = 0,     // EM_BODY: at a member key
+  RB_SEP = 1,     // EM_BODY: after a member value (',' or '}')
+  RB_ARR = 2,     // EM_BODY: at a repeated element
+  RB_ARRSEP = 3,  // EM_BODY: after a repeated element
+  RB_MAPKEY = 4,  // EM_BODY: at a map entry key
+  RB_MAPSEP = 5,  // EM_BODY: after a map entry
+  RS_KEY = 6,     // EM_STRUCT: at an entry key
+  RS_SEP = 7,     // EM_STRUCT: after an entry
+  RL_ELEM = 8,    // EM_LIST: at an element
+  RL_SEP = 9,     // EM_LIST: after an element
+};
+
+struct EFrame {
+  uint64_t seen_fields;
+  uint32_t seen_oneofs;
+  uint32_t slots[3];   // len slots backfilled (reverse order) on pop
+  int32_t msg_idx;
+  int32_t cont_field;  // absolute field index of the active array/map
+  uint8_t n_slots;
+  uint8_t mode;
+  uint8_t resume;
+};
+
+// Value scalar one-shot: null/bool/string/number emit a complete Value
+// payload (returns 1); '{' -> emits struct_value tag+slot, consumes '{',
+// returns 2 (*slot set); '[' likewise returns 3.  0 = error.
+DEVN int encode_value_scalar_or_classify(Ctx& c, uint32_t* slot) {
   skip_ws(c);
   uint8_t ch = peek(c);
   if (literal_at(c, "null", 4)) {
     c.pos += 4;
-    if (!emit_tag(c, 1, W_VARINT)) return false;  // null_value = NULL_VALUE
-    return emit_varint(c, 0);
+    if (!emit_tag(c, 1, W_VARINT)) return 0;
+    return emit_varint(c, 0) ? 1 : 0;
   }
   if (literal_at(c, "true", 4)) {
     c.pos += 4;
-    if (!emit_tag(c, 4, W_VARINT)) return false;
-    return emit_varint(c, 1);
+    if (!emit_tag(c, 4, W_VARINT)) return 0;
+    return emit_varint(c, 1) ? 1 : 0;
   }
   if (literal_at(c, "false", 5)) {
     c.pos += 5;
-    if (!emit_tag(c, 4, W_VARINT)) return false;
-    return emit_varint(c, 0);
+    if (!emit_tag(c, 4, W_VARINT)) return 0;
+    return emit_varint(c, 0) ? 1 : 0;
   }
   if (ch == '"') {
     uint32_t st, rl;
     bool esc;
-    if (!string_span(c, &st, &rl, &esc)) return false;
-    if (!emit_tag(c, 3, W_LEN)) return false;
-    return encode_string_payload(c, st, rl, esc);
+    if (!string_span(c, &st, &rl, &esc)) return 0;
+    if (!emit_tag(c, 3, W_LEN)) return 0;
+    return encode_string_payload(c, st, rl, esc) ? 1 : 0;
   }
   if (ch == '{') {
-    if (!emit_tag(c, 5, W_LEN)) return false;
-    uint32_t slot;
-    if (!reserve_len(c, &slot)) return false;
-    // inline Struct body: iterate members as map<string,Value> field 1
-    if (!expect(c, '{')) return false;
+    if (!emit_tag(c, 5, W_LEN)) return 0;
+    if (!reserve_len(c, slot)) return 0;
+    c.pos++;
+    return 2;
+  }
+  if (ch == '[') {
+    if (!emit_tag(c, 6, W_LEN)) return 0;
+    if (!reserve_len(c, slot)) return 0;
+    c.pos++;
+    return 3;
+  }
+  NumVal nv;
+  int nonf;
+  if (!parse_numeric_value(c, &nv, false, &nonf)) return 0;
+  double d = nv.cls == 0 ? (nv.neg ? -(double)nv.mag : (double)nv.mag) : nv.d;
+  if (!emit_tag(c, 2, W_I64)) return 0;
+  return emit_fixed64(c, __builtin_bit_cast(uint64_t, d)) ? 1 : 0;
+}
+
+// FieldMask field: JSON "a.b,camelCase" -> repeated snake_case strings
+// (message payload only; caller owns tag/len)
+DEVN bool encode_fieldmask_payload(Ctx& c) {
+  uint32_t st, rl;
+  bool esc;
+  if (!string_span(c, &st, &rl, &esc)) return false;
+  if (esc) return fail(c, E_UNSUPPORTED);
+  uint32_t i = 0;
+  while (i < rl) {
+    uint32_t j = i;
+    while (j < rl && c.s[st + j] != ',') ++j;
+    if (j > i) {
+      if (!emit_tag(c, 1, W_LEN)) return false;
+      uint32_t slot;
+      if (!reserve_len(c, &slot)) return false;
+      for (uint32_t k = i; k < j; ++k) {
+        uint8_t ch = c.s[st + k];
+        if (ch >= 'A' && ch <= 'Z') {
+          if (c.opos + 2 > c.ocap) return fail(c, E_OVERFLOW);
+          if (!c.lane) {
+            c.out[c.opos] = '_';
+            c.out[c.opos + 1] = ch + 32;
+          }
+          c.opos += 2;
+        } else {
+          if (c.opos + 1 > c.ocap) return fail(c, E_OVERFLOW);
+          if (!c.lane) c.out[c.opos] = ch;
+          c.opos += 1;
+        }
+      }
+      if (!backfill_len(c, slot)) return false;
+    }
+    i = j + 1;
+  }
+  return true;
+}
+
+// Timestamp/Duration field payload from a JSON string (no tag/len)
+DEVN bool encode_tsdur_payload(Ctx& c, bool is_ts, uint32_t fnum) {
+  uint32_t st, rl;
+  bool esc;
+  if (!string_span(c, &st, &rl, &esc)) return false;
+  if (esc) return fail(c, E_UNSUPPORTED);
+  int64_t secs;
+  int32_t nanos;
+  bool ok = is_ts ? parse_timestamp(c, c.s + st, rl, &secs, &nanos)
+                  : parse_duration(c.s + st, rl, &secs, &nanos);
+  if (!ok) return fail(c, E_INVALID_PARAMS, (int)fnum);
+  if (secs) {
+    if (!emit_varint(c, (1u << 3) | W_VARINT)) return false;
+    if (!emit_varint(c, (uint64_t)secs)) return false;
+  }
+  if (nanos) {
+    if (!emit_varint(c, (2u << 3) | W_VARINT)) return false;
+    if (!emit_varint(c, (uint64_t)(int64_t)nanos)) return false;
+  }
+  return true;
+}
+
+// map entry key: parse the JSON key span, emit entry tag + len slot + key
+// field (1).  *eslot receives the entry slot.
+DEVN bool encode_map_entry_key(Ctx& c, const FieldEntry& f, const FieldEntry& kf,
+                               uint32_t* eslot) {
+  uint32_t kst, krl;
+  bool kesc;
+  if (!string_span(c, &kst, &krl, &kesc)) return false;
+  if (kesc) return fail(c, E_UNSUPPORTED);
+  if (!expect(c, ':')) return false;
+  if (!emit_tag(c, f.number, W_LEN)) return false;
+  if (!reserve_len(c, eslot)) return false;
+  switch (kf.kind) {
+    case K_STRING: {
+      if (!emit_tag(c, 1, W_LEN)) return false;
+      return encode_string_payload(c, kst, krl, false);
+    }
+    case K_BOOL: {
+      uint64_t v;
+      if (krl == 4 && c.s[kst] == 't') v = 1;
+      else if (krl == 5 && c.s[kst] == 'f') v = 0;
+      else return fail(c, E_INVALID_PARAMS, (int)f.number);
+      if (!emit_tag(c, 1, W_VARINT)) return false;
+      return emit_varint(c, v);
+    }
+    default: {  // integer keys
+      NumVal nv;
+      if (!parse_number_text(c.s, kst, kst + krl, &nv))
+        return fail(c, E_INVALID_PARAMS, (int)f.number);
+      int64_t sv;
+      uint64_t uv;
+      switch (kf.kind) {
+        case K_INT64:
+        case K_SFIXED64:
+        case K_SINT64:
+        case K_INT32:
+        case K_SINT32:
+        case K_SFIXED32:
+          if (!num_to_i64(nv, &sv)) return fail(c, E_INVALID_PARAMS, (int)f.number);
+          break;
+        default:
+          if (!num_to_u64(nv, &uv)) return fail(c, E_INVALID_PARAMS, (int)f.number);
+          sv = (int64_t)uv;
+      }
+      if (kf.kind == K_SINT64) uv = zigzag64(sv);
+      else if (kf.kind == K_SINT32) uv = zigzag32((int32_t)sv);
+      else uv = (uint64_t)sv;
+      if (kf.kind == K_FIXED64 || kf.kind == K_SFIXED64) {
+        if (!emit_tag(c, 1, W_I64)) return false;
+        return emit_fixed64(c, uv);
+      }
+      if (kf.kind == K_FIXED32 || kf.kind == K_SFIXED32) {
+        if (!emit_tag(c, 1, W_I32)) return false;
+        return emit_fixed32(c, (uint32_t)uv);
+      }
+      if (!emit_tag(c, 1, W_VARINT)) return false;
+      return emit_varint(c, uv);
+    }
+  }
+}
+
+// message-typed field value (tag NOT yet emitted).  Leaf WKTs complete
+// inline (returns 1).  Container bodies emit tag + slot(s), consume the
+// opening brace/bracket and return 2 with push parameters.  0 = error.
+DEVN int encode_msgfield_value(Ctx& c, const FieldEntry& f, uint8_t* push_mode,
+                               int32_t* push_idx, uint32_t* slots,
+                               uint8_t* n_slots) {
+  const MsgEntry& sub = c.t.msgs[f.sub_index];
+  *n_slots = 0;
+  switch (sub.wkt_kind) {
+    case WKT_TIMESTAMP:
+    case WKT_DURATION: {
+      if (!emit_tag(c, f.number, W_LEN)) return 0;
+      uint32_t slot;
+      if (!reserve_len(c, &slot)) return 0;
+      if (!encode_tsdur_payload(c, sub.wkt_kind == WKT_TIMESTAMP, f.number))
+        return 0;
+      return backfill_len(c, slot) ? 1 : 0;
+    }
+    case WKT_WRAPPER: {
+      if (!emit_tag(c, f.number, W_LEN)) return 0;
+      uint32_t slot;
+      if (!reserve_len(c, &slot)) return 0;
+      const FieldEntry& inner = c.t.fields[sub.field_start];
+      if (!encode_scalar_field(c, inner)) return 0;
+      return backfill_len(c, slot) ? 1 : 0;
+    }
+    case WKT_FIELDMASK: {
+      if (!emit_tag(c, f.number, W_LEN)) return 0;
+      uint32_t slot;
+      if (!reserve_len(c, &slot)) return 0;
+      if (!encode_fieldmask_payload(c)) return 0;
+      return backfill_len(c, slot) ? 1 : 0;
+    }
+    case WKT_EMPTY: {
+      if (!emit_tag(c, f.number, W_LEN)) return 0;
+      uint32_t slot;
+      if (!reserve_len(c, &slot)) return 0;
+      if (!expect(c, '{')) return 0;
+      skip_ws(c);
+      if (!expect(c, '}')) return 0;
+      return backfill_len(c, slot) ? 1 : 0;
+    }
+    case WKT_VALUE: {
+      if (!emit_tag(c, f.number, W_LEN)) return 0;
+      uint32_t vslot;
+      if (!reserve_len(c, &vslot)) return 0;
+      uint32_t inner_slot = 0;
+      int r = encode_value_scalar_or_classify(c, &inner_slot);
+      if (r == 0) return 0;
+      if (r == 1) return backfill_len(c, vslot) ? 1 : 0;
+      *push_mode = (r == 2) ? EM_STRUCT : EM_LIST;
+      *push_idx = -1;
+      slots[0] = inner_slot;
+      slots[1] = vslot;
+      *n_slots = 2;
+      return 2;
+    }
+    case WKT_STRUCT: {
+      if (!emit_tag(c, f.number, W_LEN)) return 0;
+      uint32_t slot;
+      if (!reserve_len(c, &slot)) return 0;
+      if (!expect(c, '{')) return 0;
+      *push_mode = EM_STRUCT;
+      *push_idx = -1;
+      slots[0] = slot;
+      *n_slots = 1;
+      return 2;
+    }
+    case WKT_LISTVALUE: {
+      if (!emit_tag(c, f.number, W_LEN)) return 0;
+      uint32_t slot;
+      if (!reserve_len(c, &slot)) return 0;
+      if (!expect(c, '[')) return 0;
+      *push_mode = EM_LIST;
+      *push_idx = -1;
+      slots[0] = slot;
+      *n_slots = 1;
+      return 2;
+    }
+    case WKT_ANY:
+      return fail(c, E_UNSUPPORTED) ? 1 : 0;
+    default: {  // plain nested message
+      if (!emit_tag(c, f.number, W_LEN)) return 0;
+      uint32_t slot;
+      if (!reserve_len(c, &slot)) return 0;
+      if (!expect(c, '{')) return 0;
+      *push_mode = EM_BODY;
+      *push_idx = f.sub_index;
+      slots[0] = slot;
+      *n_slots = 1;
+      return 2;
+    }
+  }
+}
+
+// member key lookup inside EM_BODY (hash + dup/oneof bookkeeping);
+// returns field index within the message or -1 (error already set)
+DEVN int encode_body_key(Ctx& c, EFrame& f) {
+  const MsgEntry& m = c.t.msgs[f.msg_idx];
+  uint32_t kst, krl;
+  bool kesc;
+  if (!string_span(c, &kst, &krl, &kesc)) return -1;
+  const uint8_t* kptr = c.s + kst;
+  uint32_t klen = krl;
+  if (kesc) {
+    uint32_t n = unescape_serial(c, c.s + kst, krl, c.keybuf, 192);
+    if (n == 0xFFFFFFFF) {
+      fail(c, E_PARSE);
+      return -1;
+    }
+    __builtin_amdgcn_wave_barrier();
+    kptr = c.keybuf;
+    klen = n;
+  }
+  if (!expect(c, ':')) return -1;
+  uint64_t h = fnv1a64(kptr, klen);
+  int fidx = -1;
+  for (int i = 0; i < m.field_count; ++i) {
+    const FieldEntry& fe = c.t.fields[m.field_start + i];
+    if ((fe.hash_json == h && fe.json_len == klen &&
+         wave_equal(c, c.t.names + fe.json_off, kptr, klen)) ||
+        (fe.hash_orig == h && fe.name_len == klen &&
+         wave_equal(c, c.t.names + fe.name_off, kptr, klen))) {
+      fidx = i;
+      break;
+    }
+  }
+  if (fidx < 0) {
+    c.err_pos = kst;
+    fail(c, E_INVALID_PARAMS, -1);  // unknown field: protojson rejects
+    return -1;
+  }
+  const FieldEntry& fe = c.t.fields[m.field_start + fidx];
+  if (fidx < 64) {
+    if (f.seen_fields & (1ull << fidx)) {
+      fail(c, E_INVALID_PARAMS, (int)fe.number);
+      return -1;
+    }
+    f.seen_fields |= 1ull << fidx;
+  }
+  if (fe.flags & F_ONEOF) {
+    if (f.seen_oneofs & (1u << fe.oneof_id)) {
+      fail(c, E_INVALID_PARAMS, (int)fe.number);
+      return -1;
+    }
+    f.seen_oneofs |= 1u << fe.oneof_id;
+  }
+  return m.field_start + fidx;
+}
+
+// whole packed repeated array "[e,e,...]" — '[' already consumed, at first
+// element; emits tag + len slot + packed payload + backfill
+DEVN bool encode_packed_array(Ctx& c, const FieldEntry& f) {
+  if (!emit_tag(c, f.number, W_LEN)) return false;
+  uint32_t slot;
+  if (!reserve_len(c, &slot)) return false;
+  while (true) {
     skip_ws(c);
-    if (peek(c) != '}') {
-      while (true) {
+    if (!encode_packed_element(c, f)) return false;
+    skip_ws(c);
+    uint8_t ch = peek(c);
+    if (ch == ',') {
+      c.pos++;
+      continue;
+    }
+    if (ch == ']') {
+      c.pos++;
+      break;
+    }
+    return fail(c, E_PARSE);
+  }
+  return backfill_len(c, slot);
+}
+
+DEV bool encode_walk(Ctx& c, int top_msg_idx) {
+  EFrame stack[MAX_RECURSE];
+  int sp = 0;
+  int depth_cap = (int)MAX_RECURSE;
+  if (c.lim.enforce && (int)c.lim.max_depth < depth_cap)
+    depth_cap = (int)c.lim.max_depth;
+
+#define EPUSH(MODE, MIDX, SLOTS_ARR, NS)                      \
+  do {                                                        \
+    if (sp >= depth_cap) return fail(c, E_LIMIT, 1);          \
+    EFrame& nf = stack[sp++];                                 \
+    nf.seen_fields = 0;                                       \
+    nf.seen_oneofs = 0;                                       \
+    nf.msg_idx = (MIDX);                                      \
+    nf.cont_field = -1;                                       \
+    nf.n_slots = (NS);                                        \
+    for (int _i = 0; _i < (NS); ++_i) nf.slots[_i] = (SLOTS_ARR)[_i]; \
+    nf.mode = (MODE);                                         \
+    nf.resume = (MODE) == EM_BODY ? RB_KEY                    \
+                : (MODE) == EM_STRUCT ? RS_KEY : RL_ELEM;     \
+  } while (0)
+
+  // ---- top-level entry ----
+  {
+    const MsgEntry& m = c.t.msgs[top_msg_idx];
+    uint32_t slots0[3];
+    if (m.wkt_kind == WKT_VALUE) {
+      uint32_t islot = 0;
+      int r = encode_value_scalar_or_classify(c, &islot);
+      if (r == 0) return false;
+      if (r == 1) return true;
+      slots0[0] = islot;
+      EPUSH(r == 2 ? EM_STRUCT : EM_LIST, -1, slots0, 1);
+      // fall into loop; empty-body handled by the frame steps
+      skip_ws(c);
+      if (r == 2 && peek(c) == '}') {
+        c.pos++;
+        if (!backfill_len(c, islot)) return false;
+        --sp;
+      } else if (r == 3 && peek(c) == ']') {
+        c.pos++;
+        if (!backfill_len(c, islot)) return false;
+        --sp;
+      }
+    } else if (m.wkt_kind == WKT_STRUCT) {
+      if (!expect(c, '{')) return false;
+      skip_ws(c);
+      if (peek(c) == '}') {
+        c.pos++;
+        return true;
+      }
+      EPUSH(EM_STRUCT, -1, slots0, 0);
+    } else if (m.wkt_kind == WKT_LISTVALUE) {
+      if (!expect(c, '[')) return false;
+      skip_ws(c);
+      if (peek(c) == ']') {
+        c.pos++;
+        return true;
+      }
+      EPUSH(EM_LIST, -1, slots0, 0);
+    } else if (m.wkt_kind == WKT_FIELDMASK) {
+      return encode_fieldmask_payload(c);
+    } else if (m.wkt_kind == WKT_EMPTY) {
+      if (!expect(c, '{')) return false;
+      skip_ws(c);
+      return expect(c, '}');
+    } else {
+      if (!expect(c, '{')) return false;
+      skip_ws(c);
+      if (peek(c) == '}') {
+        c.pos++;
+        return true;
+      }
+      EPUSH(EM_BODY, top_msg_idx, slots0, 0);
+    }
+  }
+
+  while (sp > 0) {
+    EFrame& f = stack[sp - 1];
+    switch (f.resume) {
+      case RB_KEY: {
+        int fidx = encode_body_key(c, f);
+        if (fidx < 0) return false;
+        const FieldEntry& fe = c.t.fields[fidx];
+        skip_ws(c);
+        bool is_value_wkt =
+            fe.kind == K_MESSAGE && c.t.msgs[fe.sub_index].wkt_kind == WKT_VALUE;
+        if (literal_at(c, "null", 4) && !is_value_wkt) {
+          c.pos += 4;  // null -> unset (protojson)
+          f.resume = RB_SEP;
+          continue;
+        }
+        if (fe.flags & F_MAP) {
+          if (!expect(c, '{')) return false;
+          skip_ws(c);
+          if (peek(c) == '}') {
+            c.pos++;
+            f.resume = RB_SEP;
+            continue;
+          }
+          f.cont_field = fidx;
+          f.resume = RB_MAPKEY;
+          continue;
+        }
+        if (fe.flags & F_REPEATED) {
+          if (!expect(c, '[')) return false;
+          skip_ws(c);
+          if (peek(c) == ']') {
+            c.pos++;
+            f.resume = RB_SEP;
+            continue;
+          }
+          if (is_packable(fe.kind)) {
+            if (!encode_packed_array(c, fe)) return false;
+            f.resume = RB_SEP;
+            continue;
+          }
+          f.cont_field = fidx;
+          f.resume = RB_ARR;
+          continue;
+        }
+        // singular
+        if (fe.kind == K_MESSAGE) {
+          uint8_t pm;
+          int32_t pidx;
+          uint32_t slots[3];
+          uint8_t ns;
+          int r = encode_msgfield_value(c, fe, &pm, &pidx, slots, &ns);
+          if (r == 0) return false;
+          if (r == 1) {
+            f.resume = RB_SEP;
+            continue;
+          }
+          f.resume = RB_SEP;
+          EPUSH(pm, pidx, slots, ns);
+          // empty body fast path
+          skip_ws(c);
+          if ((pm != EM_LIST && peek(c) == '}') ||
+              (pm == EM_LIST && peek(c) == ']')) {
+            c.pos++;
+            EFrame& nf = stack[sp - 1];
+            for (int i = 0; i < nf.n_slots; ++i)
+              if (!backfill_len(c, nf.slots[i])) return false;
+            --sp;
+          }
+          continue;
+        }
+        if (!encode_scalar_field(c, fe)) return false;
+        f.resume = RB_SEP;
+        continue;
+      }
+      case RB_SEP: {
+        skip_ws(c);
+        uint8_t ch = peek(c);
+        if (ch == ',') {
+          c.pos++;
+          skip_ws(c);
+          f.resume = RB_KEY;
+          continue;
+        }
+        if (ch == '}') {
+          c.pos++;
+          for (int i = 0; i < f.n_slots; ++i)
+            if (!backfill_len(c, f.slots[i])) return false;
+          --sp;
+          continue;
+        }
+        return fail(c, E_PARSE);
+      }
+      case RB_ARR: {
+        const FieldEntry& fe = c.t.fields[f.cont_field];
+        skip_ws(c);
+        if (fe.kind == K_MESSAGE) {
+          uint8_t pm;
+          int32_t pidx;
+          uint32_t slots[3];
+          uint8_t ns;
+          int r = encode_msgfield_value(c, fe, &pm, &pidx, slots, &ns);
+          if (r == 0) return false;
+          if (r == 1) {
+            f.resume = RB_ARRSEP;
+            continue;
+          }
+          f.resume = RB_ARRSEP;
+          EPUSH(pm, pidx, slots, ns);
+          skip_ws(c);
+          if ((pm != EM_LIST && peek(c) == '}') ||
+              (pm == EM_LIST && peek(c) == ']')) {
+            c.pos++;
+            EFrame& nf = stack[sp - 1];
+            for (int i = 0; i < nf.n_slots; ++i)
+              if (!backfill_len(c, nf.slots[i])) return false;
+            --sp;
+          }
+          continue;
+        }
+        if (!encode_scalar_field(c, fe)) return false;
+        f.resume = RB_ARRSEP;
+        continue;
+      }
+      case RB_ARRSEP: {
+        skip_ws(c);
+        uint8_t ch = peek(c);
+        if (ch == ',') {
+          c.pos++;
+          f.resume = RB_ARR;
+          continue;
+        }
+        if (ch == ']') {
+          c.pos++;
+          f.resume = RB_SEP;
+          continue;
+        }
+        return fail(c, E_PARSE);
+      }
+      case RB_MAPKEY: {
+        const FieldEntry& fe = c.t.fields[f.cont_field];
+        const MsgEntry& em = c.t.msgs[fe.sub_index];
+        const FieldEntry& kf = c.t.fields[em.field_start];
+        const FieldEntry& vf = c.t.fields[em.field_start + 1];
+        uint32_t eslot;
+        if (!encode_map_entry_key(c, fe, kf, &eslot)) return false;
+        skip_ws(c);
+        if (literal_at(c, "null", 4) && vf.kind != K_MESSAGE) {
+          c.pos += 4;  // null map value -> default (key-only entry)
+          if (!backfill_len(c, eslot)) return false;
+          f.resume = RB_MAPSEP;
+          continue;
+        }
+        if (vf.kind == K_MESSAGE) {
+          // value tag is field 2 inside the entry
+          FieldEntry vf2 = vf;  // value field with number 2 by construction
+          uint8_t pm;
+          int32_t pidx;
+          uint32_t slots[3];
+          uint8_t ns;
+          int r = encode_msgfield_value(c, vf2, &pm, &pidx, slots, &ns);
+          if (r == 0) return false;
+          if (r == 1) {
+            if (!backfill_len(c, eslot)) return false;
+            f.resume = RB_MAPSEP;
+            continue;
+          }
+          if (ns >= 3) return fail(c, E_LIMIT, 1);
+          slots[ns++] = eslot;  // entry slot backfilled last
+          f.resume = RB_MAPSEP;
+          EPUSH(pm, pidx, slots, ns);
+          skip_ws(c);
+          if ((pm != EM_LIST && peek(c) == '}') ||
+              (pm == EM_LIST && peek(c) == ']')) {
+            c.pos++;
+            EFrame& nf = stack[sp - 1];
+            for (int i = 0; i < nf.n_slots; ++i)
+              if (!backfill_len(c, nf.slots[i])) return false;
+            --sp;
+          }
+          continue;
+        }
+        if (!encode_scalar_field(c, vf)) return false;
+        if (!backfill_len(c, eslot)) return false;
+        f.resume = RB_MAPSEP;
+        continue;
+      }
+      case RB_MAPSEP: {
+        skip_ws(c);
+        uint8_t ch = peek(c);
+        if (ch == ',') {
+          c.pos++;
+          skip_ws(c);
+          f.resume = RB_MAPKEY;
+          continue;
+        }
+        if (ch == '}') {
+          c.pos++;
+          f.resume = RB_SEP;
+          continue;
+        }
+        return fail(c, E_PARSE);
+      }
+      case RS_KEY: {
         uint32_t kst, krl;
         bool kesc;
         if (!string_span(c, &kst, &krl, &kesc)) return false;
@@ -1431,56 +1682,92 @@ DEV bool encode_json_value_as_value(Ctx& c, int depth) {
         if (!emit_tag(c, 2, W_LEN)) return false;
         uint32_t vslot;
         if (!reserve_len(c, &vslot)) return false;
-        if (!encode_json_value_as_value(c, depth + 1)) return false;
-        if (!backfill_len(c, vslot)) return false;
-        if (!backfill_len(c, eslot)) return false;
-        skip_ws(c);
-        uint8_t ch2 = peek(c);
-        if (ch2 == ',') {
-          c.pos++;
-          skip_ws(c);
+        uint32_t islot = 0;
+        int r = encode_value_scalar_or_classify(c, &islot);
+        if (r == 0) return false;
+        if (r == 1) {
+          if (!backfill_len(c, vslot)) return false;
+          if (!backfill_len(c, eslot)) return false;
+          f.resume = RS_SEP;
           continue;
         }
-        if (ch2 == '}') break;
+        uint32_t slots[3] = {islot, vslot, eslot};
+        f.resume = RS_SEP;
+        EPUSH(r == 2 ? EM_STRUCT : EM_LIST, -1, slots, 3);
+        skip_ws(c);
+        if ((r == 2 && peek(c) == '}') || (r == 3 && peek(c) == ']')) {
+          c.pos++;
+          EFrame& nf = stack[sp - 1];
+          for (int i = 0; i < nf.n_slots; ++i)
+            if (!backfill_len(c, nf.slots[i])) return false;
+          --sp;
+        }
+        continue;
+      }
+      case RS_SEP: {
+        skip_ws(c);
+        uint8_t ch = peek(c);
+        if (ch == ',') {
+          c.pos++;
+          skip_ws(c);
+          f.resume = RS_KEY;
+          continue;
+        }
+        if (ch == '}') {
+          c.pos++;
+          for (int i = 0; i < f.n_slots; ++i)
+            if (!backfill_len(c, f.slots[i])) return false;
+          --sp;
+          continue;
+        }
         return fail(c, E_PARSE);
       }
-    }
-    c.pos++;  // '}'
-    return backfill_len(c, slot);
-  }
-  if (ch == '[') {
-    if (!emit_tag(c, 6, W_LEN)) return false;
-    uint32_t slot;
-    if (!reserve_len(c, &slot)) return false;
-    c.pos++;
-    skip_ws(c);
-    if (peek(c) != ']') {
-      while (true) {
+      case RL_ELEM: {
         if (!emit_tag(c, 1, W_LEN)) return false;
         uint32_t vslot;
         if (!reserve_len(c, &vslot)) return false;
-        if (!encode_json_value_as_value(c, depth + 1)) return false;
-        if (!backfill_len(c, vslot)) return false;
-        skip_ws(c);
-        uint8_t ch2 = peek(c);
-        if (ch2 == ',') {
-          c.pos++;
+        uint32_t islot = 0;
+        int r = encode_value_scalar_or_classify(c, &islot);
+        if (r == 0) return false;
+        if (r == 1) {
+          if (!backfill_len(c, vslot)) return false;
+          f.resume = RL_SEP;
           continue;
         }
-        if (ch2 == ']') break;
+        uint32_t slots[2] = {islot, vslot};
+        f.resume = RL_SEP;
+        EPUSH(r == 2 ? EM_STRUCT : EM_LIST, -1, slots, 2);
+        skip_ws(c);
+        if ((r == 2 && peek(c) == '}') || (r == 3 && peek(c) == ']')) {
+          c.pos++;
+          EFrame& nf = stack[sp - 1];
+          for (int i = 0; i < nf.n_slots; ++i)
+            if (!backfill_len(c, nf.slots[i])) return false;
+          --sp;
+        }
+        continue;
+      }
+      case RL_SEP: {
+        skip_ws(c);
+        uint8_t ch = peek(c);
+        if (ch == ',') {
+          c.pos++;
+          f.resume = RL_ELEM;
+          continue;
+        }
+        if (ch == ']') {
+          c.pos++;
+          for (int i = 0; i < f.n_slots; ++i)
+            if (!backfill_len(c, f.slots[i])) return false;
+          --sp;
+          continue;
+        }
         return fail(c, E_PARSE);
       }
     }
-    c.pos++;  // ']'
-    return backfill_len(c, slot);
   }
-  // number
-  NumVal nv;
-  int nonf;
-  if (!parse_numeric_value(c, &nv, false, &nonf)) return false;
-  double d = nv.cls == 0 ? (nv.neg ? -(double)nv.mag : (double)nv.mag) : nv.d;
-  if (!emit_tag(c, 2, W_I64)) return false;
-  return emit_fixed64(c, __builtin_bit_cast(uint64_t, d));
+#undef EPUSH
+  return c.status == E_OK;
 }
 
 // ---------------------------------------------------------------------------
@@ -1623,7 +1910,7 @@ DEV bool parse_envelope(Ctx& c, SlotResult& r, uint8_t* id_slot) {
   uint32_t save_len = c.len;
   c.pos = args_pos;
   c.len = args_end;
-  bool ok = encode_message(c, c.t.tools[tool].in_msg, 1);
+  bool ok = encode_walk(c, c.t.tools[tool].in_msg);
   c.len = save_len;
   return ok;
 }
@@ -1673,15 +1960,7 @@ extern "C" __global__ void __launch_bounds__(WPB * WAVE) k_json2pb(
     } else {
       int msg_idx = msg_idx_in ? msg_idx_in[req] : 0;
       const MsgEntry& m = t.msgs[msg_idx];
-      if (m.wkt_kind == WKT_TIMESTAMP || m.wkt_kind == WKT_DURATION ||
-          m.wkt_kind == WKT_WRAPPER || m.wkt_kind == WKT_VALUE) {
-        // top-level non-object WKT payloads are host-side concerns in
-        // transcode tests; treat Value specially, else expect message form
-        ok = m.wkt_kind == WKT_VALUE ? encode_json_value_as_value(c, 1)
-                                     : encode_message(c, msg_idx, 1);
-      } else {
-        ok = encode_message(c, msg_idx, 1);
-      }
+      ok = encode_walk(c, msg_idx);
       if (ok) {
         skip_ws(c);
         if (c.pos < c.len) ok = fail(c, E_PARSE);

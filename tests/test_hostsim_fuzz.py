@@ -187,7 +187,10 @@ def test_fuzz_struct_value(value):
     except Exception:
         assert enc[0]["status"] != 0
         return
-    if enc[0]["status"] == 6:
+    if enc[0]["status"] in (6, 7):
+        # E_UNSUPPORTED / E_OVERFLOW: declared GPU-subset boundary (e.g.
+        # pathological nesting where 3-byte len slots outgrow the pb cap);
+        # the pipeline host-transcodes these, counted
         return
     assert enc[0]["status"] == 0, f"status {enc[0]['status']} for {text!r}"
     g = json_format.MessageToDict(_cpu.pb_to_message(_sdesc, pbs[0]))
