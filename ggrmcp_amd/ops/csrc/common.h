@@ -259,14 +259,25 @@ constexpr size_t KERNEL_STACK_BYTES = 12 * 1024;
 constexpr uint32_t LEN_SLOT = 3;
 constexpr uint32_t LEN_SLOT_MAX = (1u << 21) - 1;
 
+// u64 -> decimal text without an addressable temp buffer (a local tmp[20]
+// array lands in scratch memory and costs a private-memory round trip per
+// digit; emitting MSB-first via the power table keeps everything in VGPRs)
+__constant__ uint64_t DEC_P10[20] = {
+    1ull, 10ull, 100ull, 1000ull, 10000ull, 100000ull, 1000000ull,
+    10000000ull, 100000000ull, 1000000000ull, 10000000000ull,
+    100000000000ull, 1000000000000ull, 10000000000000ull,
+    100000000000000ull, 1000000000000000ull, 10000000000000000ull,
+    100000000000000000ull, 1000000000000000000ull, 10000000000000000000ull};
+
 DEV uint32_t u64_to_dec(uint8_t* out, uint64_t v) {
-  uint8_t tmp[20];
-  uint32_t n = 0;
-  do {
-    tmp[n++] = (uint8_t)('0' + (v % 10));
-    v /= 10;
-  } while (v);
-  for (uint32_t i = 0; i < n; ++i) out[i] = tmp[n - 1 - i];
+  uint32_t n = 1;
+  while (n < 20 && v >= DEC_P10[n]) ++n;
+  for (uint32_t i = 0; i < n; ++i) {
+    uint64_t p = DEC_P10[n - 1 - i];
+    uint32_t d = (uint32_t)(v / p);
+    out[i] = (uint8_t)('0' + d);
+    v -= (uint64_t)d * p;
+  }
   return n;
 }
 
