@@ -48,7 +48,7 @@ class HostSimEngine:
     ) -> Tuple[np.ndarray, List[Optional[bytes]]]:
         data = b"".join(payloads)
         in_off = _offsets([len(p) for p in payloads])
-        pb_off = _offsets([len(p) for p in payloads], pad=192, align=16)
+        pb_off = _offsets([len(p) + len(p) // 4 for p in payloads], pad=192, align=16)
         msg_idx = (
             np.asarray(msg_indices, dtype=np.int32) if msg_indices is not None else None
         )

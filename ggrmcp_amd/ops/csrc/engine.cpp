@@ -94,7 +94,12 @@ inline bool view_of(py::handle el, const char** ptr, size_t* len) {
 }
 
 // per-request arena sizing rules (single source of truth, mirrored nowhere)
-inline size_t pb_cap(size_t in_len) { return (in_len + 192 + 15) & ~(size_t)15; }
+// +25% headroom: 3-byte non-minimal length slots exceed JSON syntax
+// overhead for small-entry maps/arrays (entry+key+value slots ~9B per
+// entry vs ~4 chars of JSON punctuation)
+inline size_t pb_cap(size_t in_len) {
+  return (in_len + in_len / 4 + 192 + 15) & ~(size_t)15;
+}
 inline size_t scratch_cap(size_t wire_len) {
   return (wire_len * 8 + 1024 + 15) & ~(size_t)15;
 }

@@ -238,19 +238,26 @@ DEV uint64_t gen_digits(double d, int prec, int* e10_out) {
   return digits;
 }
 
-DEV uint32_t dtoa17(uint8_t* out, double d, bool as_float) {
-  // caller guarantees finite; d > 0 here (sign handled by caller)
-  if (d == 0.0) {
-    out[0] = '0';
-    return 1;
-  }
+// emit d (>0, finite) as decimal text directly into the output (caller
+// guarantees >= 40 bytes of cap).  Digits live in a u64 the whole time —
+// the previous buffer-based form staged through scratch arrays (dig[17] +
+// buf[32]) and cost ~9 us per double.
+DEV void put_digit(DCtx& c, uint8_t ch) {
+  if (!c.lane) c.out[c.opos] = ch;
+  c.opos++;
+}
+
+DEVN bool emit_double_body(DCtx& c, double d, bool as_float) {
   // integral fast path (exact digits).  For float fields only below 2^24:
-  // protojson prints the SHORTEST digits that round-trip as float32, which
-  // differs from the exact integer above that (e.g. 51749408.0f prints as
-  // 51749410) — the precision search below reproduces that.
+  // protojson prints the SHORTEST digits that round-trip as float32.
   if (d == trunc(d) &&
       (as_float ? d < 16777216.0 : d < 9.007199254740992e15)) {
-    return u64_to_dec(out, (uint64_t)d);
+    uint32_t n = 1;
+    uint64_t v = (uint64_t)d;
+    while (n < 20 && v >= DEC_P10[n]) ++n;
+    for (uint32_t i = 0; i < n; ++i)
+      put_digit(c, (uint8_t)('0' + (uint32_t)((v / DEC_P10[n - 1 - i]) % 10)));
+    return true;
   }
   // shortest precision whose parse-back round-trips
   int p_lo = as_float ? 6 : 15;
@@ -270,53 +277,55 @@ DEV uint32_t dtoa17(uint8_t* out, double d, bool as_float) {
       break;
     }
   }
-  uint8_t dig[17];
-  for (int i = prec - 1; i >= 0; --i) {
-    dig[i] = (uint8_t)('0' + digits % 10);
-    digits /= 10;
-  }
+  // trim trailing zeros (integer form)
   int ndig = prec;
-  while (ndig > 1 && dig[ndig - 1] == '0') --ndig;
-  uint32_t o = 0;
+  while (ndig > 1 && digits % 10 == 0) {
+    digits /= 10;
+    --ndig;
+  }
+  // digit k (0-based, MSB first) of the ndig-digit block
+#define DIGIT_AT(k) ((uint8_t)('0' + (uint32_t)((digits / DEC_P10[ndig - 1 - (k)]) % 10)))
   if (e10 >= -6 && e10 <= 20) {
     if (e10 >= 0) {
       int ip = e10 + 1;  // digits before the point
-      for (int i = 0; i < ip; ++i) out[o++] = i < ndig ? dig[i] : '0';
+      for (int i = 0; i < ip; ++i)
+        put_digit(c, i < ndig ? DIGIT_AT(i) : (uint8_t)'0');
       if (ndig > ip) {
-        out[o++] = '.';
-        for (int i = ip; i < ndig; ++i) out[o++] = dig[i];
+        put_digit(c, '.');
+        for (int i = ip; i < ndig; ++i) put_digit(c, DIGIT_AT(i));
       }
     } else {
-      out[o++] = '0';
-      out[o++] = '.';
-      for (int i = 0; i < -e10 - 1; ++i) out[o++] = '0';
-      for (int i = 0; i < ndig; ++i) out[o++] = dig[i];
+      put_digit(c, '0');
+      put_digit(c, '.');
+      for (int i = 0; i < -e10 - 1; ++i) put_digit(c, '0');
+      for (int i = 0; i < ndig; ++i) put_digit(c, DIGIT_AT(i));
     }
   } else {
-    out[o++] = dig[0];
+    put_digit(c, DIGIT_AT(0));
     if (ndig > 1) {
-      out[o++] = '.';
-      for (int i = 1; i < ndig; ++i) out[o++] = dig[i];
+      put_digit(c, '.');
+      for (int i = 1; i < ndig; ++i) put_digit(c, DIGIT_AT(i));
     }
-    out[o++] = 'e';
+    put_digit(c, 'e');
     int e = e10;
     if (e < 0) {
-      out[o++] = '-';
+      put_digit(c, '-');
       e = -e;
     } else {
-      out[o++] = '+';
+      put_digit(c, '+');
     }
     if (e >= 100) {
-      out[o++] = (uint8_t)('0' + e / 100);
+      put_digit(c, (uint8_t)('0' + e / 100));
       e %= 100;
-      out[o++] = (uint8_t)('0' + e / 10);
-      out[o++] = (uint8_t)('0' + e % 10);
+      put_digit(c, (uint8_t)('0' + e / 10));
+      put_digit(c, (uint8_t)('0' + e % 10));
     } else {
-      out[o++] = (uint8_t)('0' + e / 10);
-      out[o++] = (uint8_t)('0' + e % 10);
+      put_digit(c, (uint8_t)('0' + e / 10));
+      put_digit(c, (uint8_t)('0' + e % 10));
     }
   }
-  return o;
+#undef DIGIT_AT
+  return true;
 }
 
 DEVN bool put_double(DCtx& c, double d, bool as_float) {
@@ -332,19 +341,17 @@ DEVN bool put_double(DCtx& c, double d, bool as_float) {
     if (ad > 1e308 || (ad != 0.0 && ad < 1e-307))
       return dfail(c, E_UNSUPPORTED);
   }
-  if (c.opos + 32 > c.ocap) return dfail(c, E_OVERFLOW);
-  uint32_t n = 0;
-  uint8_t buf[32];
-  if (d < 0 || (d == 0.0 && signbit(d))) {
-    buf[0] = '-';
-    n = 1 + dtoa17(buf + 1, -d, as_float);
-  } else {
-    n = dtoa17(buf, d, as_float);
+  if (c.opos + 40 > c.ocap) return dfail(c, E_OVERFLOW);
+  if (d == 0.0) {
+    if (signbit(d)) put_digit(c, '-');
+    put_digit(c, '0');
+    return true;
   }
-  if (!c.lane)
-    for (uint32_t i = 0; i < n; ++i) c.out[c.opos + i] = buf[i];
-  c.opos += n;
-  return true;
+  if (d < 0) {
+    put_digit(c, '-');
+    d = -d;
+  }
+  return emit_double_body(c, d, as_float);
 }
 
 // ---------------------------------------------------------------------------
