@@ -11,14 +11,4 @@ across up to 8 GPUs and shared discovery state broadcast via RCCL over xGMI.
 See SURVEY.md for the component-by-component map onto the reference.
 """
 
-import os as _os
-
-# Several engine instances (config.gpu.streams) dispatch on their own HIP
-# streams; ROCm provisions kernel-stack scratch PER HARDWARE QUEUE, and the
-# recursive transcode kernels' raised stack (common.h KERNEL_STACK_BYTES)
-# aborts with HSA_STATUS_ERROR_OUT_OF_RESOURCES when too many queues each
-# reserve it.  Two hardware queues keep copy/kernel overlap while bounding
-# the scratch reservation.  Set before any HIP/torch device initialization.
-_os.environ.setdefault("GPU_MAX_HW_QUEUES", "2")
-
 __version__ = "0.1.0"

@@ -242,18 +242,12 @@ DEV bool get_varint(const uint8_t* p, uint32_t len, uint32_t* pos, uint64_t* out
   return false;
 }
 
-// recursion cap for the encode/decode message walkers.  The kernels are
-// recursive; the device stack is raised to KERNEL_STACK_BYTES per lane via
-// hipDeviceSetLimit (engine.cpp) and MAX_RECURSE bounds the frames so the
-// worst case fits (frame <= ~600 B).  Scratch backing is provisioned PER
-// HARDWARE QUEUE, so the budget must stay small enough that several engine
-// streams can dispatch concurrently (32 KB x 4 queues aborted with
-// HSA_STATUS_ERROR_OUT_OF_RESOURCES); ggrmcp_amd caps GPU_MAX_HW_QUEUES as
-// well.  Anything deeper returns E_LIMIT and transcodes on the host
-// (counted).  The MCP validation limit is depth 10 (validation.go:163-184),
-// so 16 leaves headroom for transcode-mode payloads.
+// nesting cap for the iterative encode/decode walkers' explicit frame
+// stacks (statically-sized private arrays — no dynamic device stack).
+// Deeper nesting returns E_LIMIT and transcodes on the host (counted).
+// The MCP validation limit is depth 10 (validation.go:163-184), so 16
+// leaves headroom for transcode-mode payloads.
 constexpr int MAX_RECURSE = 16;
-constexpr size_t KERNEL_STACK_BYTES = 12 * 1024;
 
 // fixed-width reserved length slots (supports nested payloads < 2^21)
 constexpr uint32_t LEN_SLOT = 3;

@@ -117,12 +117,12 @@ class Engine {
          size_t cap_in, size_t cap_pb, size_t cap_scratch, size_t cap_final)
       : device_(device), max_batch_(max_batch) {
     HIP_CHECK(hipSetDevice(device_));
-    // The kernels recurse through nested messages; the compiler's static
-    // private-segment estimate (~700 B/lane) cannot cover dynamic recursion,
-    // and overflow silently corrupts the NEIGHBORING wave's stack (observed:
-    // even slots fine, odd slots garbage at depth >= 3).  Raise the device
-    // stack so MAX_RECURSE frames always fit.
-    HIP_CHECK(hipDeviceSetLimit(hipLimitStackSize, KERNEL_STACK_BYTES));
+    // Both kernels are ITERATIVE (explicit MAX_RECURSE-deep frame stacks in
+    // decode_walk/encode_walk): the private segment is statically sized by
+    // the compiler, so no hipLimitStackSize raise is needed — the earlier
+    // recursive versions overflowed the dynamic stack into the neighboring
+    // wave's scratch and, once raised, the per-queue scratch reservation
+    // capped how many HIP queues could dispatch concurrently.
     HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
 
     upload_blob(msg_table, d_msgs_);
