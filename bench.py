@@ -150,10 +150,9 @@ def main() -> None:
             torch.cuda.set_device(local_rank)
         dist.init_process_group(backend=backend)
 
-    # streaming flows through the grpcio python backend (the native h2
-    # backend serves unary only); config 5 runs N distinct-package backends
-    if args.config in ("stream", "multi"):
-        args.backend = "python"
+    # config 5 runs N distinct-package backends; both stream and multi can
+    # use the native backend (h2grpc stream_echo route) with the batched
+    # native streaming client
     native_backend = args.backend == "native"
     n_backends = args.backends if args.config == "multi" else 1
     backend_procs, socks = [], []

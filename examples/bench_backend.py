@@ -105,15 +105,17 @@ def serve(target: str = "127.0.0.1:0", max_workers: int = 32, package: str = "be
     return server, bound
 
 
-def serve_native(target: str):
+def serve_native(target: str, package: str = "bench"):
     """C++ nghttp2 backend (ops/csrc/h2grpc.cpp H2Server): native handlers
-    for SayHello and echo routes; no Python in the serving path."""
+    for SayHello, unary echo and server-streaming echo routes; no Python in
+    the serving path."""
     from ggrmcp_amd.backend.native_invoker import load_module
 
     mod = load_module()
     srv = mod.Server(target)
     srv.add_route("/hello.HelloService/SayHello", "hello")
-    srv.add_route("/bench.EchoService/Echo", "echo")
+    srv.add_route(f"/{package}.EchoService/Echo", "echo")
+    srv.add_route(f"/{package}.EchoService/StreamEcho", "stream_echo")
     srv.add_route("/complex.DocumentService/PutDocument", "echo")
     srv.add_route("/complex.NodeService/Echo", "echo")
     bound = srv.start()
@@ -133,7 +135,7 @@ def main() -> None:
     args = ap.parse_args()
     target = f"unix:{args.uds}" if args.uds else f"{args.host}:{args.port}"
     if args.native:
-        server, bound = serve_native(target)
+        server, bound = serve_native(target, args.package)
         print(f"READY {bound}", flush=True)
         try:
             while True:

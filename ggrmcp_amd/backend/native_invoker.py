@@ -86,6 +86,26 @@ class NativeWireClient:
                 out.append(NativeRpcError(status, message))
         return out
 
+    def invoke_stream_batch(
+        self,
+        paths: Sequence[str],
+        payloads: Sequence[bytes],
+        timeout_s: float,
+        metadata: Optional[Sequence[Sequence[Tuple[str, str]]]] = None,
+    ) -> List[object]:
+        """Server-streaming batch: each slot resolves to a list of wire
+        chunks (all messages of the stream) or NativeRpcError."""
+        res = self._cli.invoke_stream_batch(
+            list(paths), list(payloads), timeout_s, list(metadata or [])
+        )
+        out: List[object] = []
+        for status, msgs, message in res:
+            if status == 0:
+                out.append(list(msgs))
+            else:
+                out.append(NativeRpcError(status, message))
+        return out
+
     def healthy(self) -> bool:
         return self._cli.healthy()
 

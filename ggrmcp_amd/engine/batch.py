@@ -111,6 +111,21 @@ class EngineStats:
         }
 
 
+class _SlotOfBatch:
+    """Future-like view of one slot of a batched stream invoke."""
+
+    def __init__(self, fut, index: int) -> None:
+        self._fut = fut
+        self._index = index
+
+    def result(self):
+        slots, res = self._fut.result()
+        r = res[self._index]
+        if isinstance(r, Exception):
+            raise r
+        return r
+
+
 class GpuEngine:
     """One GPU device engine: compiled tables + the HIP extension Engine."""
 
@@ -311,16 +326,36 @@ class GpuPipeline:
         out_idx = np.zeros(n, dtype=np.int32)
 
         # server-streaming slots: fan out unary->stream invokes concurrently;
-        # their wire chunks decode in one extra GPU batch below (config 4)
+        # their wire chunks decode in one extra GPU batch below (config 4).
+        # With the native transport, all streams of a backend go out as ONE
+        # batched h2 call (multiplexed streams, GIL-free).
         stream_futs: Dict[int, Any] = {}
         timeout0 = timeout_s if timeout_s is not None else self.config.grpc.request_timeout_s
+        stream_per_be: Dict[int, List[Any]] = {}
         for i in range(n):
             if enc[i]["status"] == E_OK and enc[i]["flags"] & SR_SERVER_STREAMING:
                 mi = self._mi_by_idx[enc[i]["tool_idx"]]
                 hdr = headers[i] if headers else None
-                stream_futs[i] = self._invoke_pool.submit(
-                    self.discoverer.invoke_streaming_wire, mi, pbs[i], hdr, timeout0
+                if self.wire_clients:
+                    be = mi.backend_index if mi.backend_index < len(self.wire_clients) else 0
+                    g = stream_per_be.setdefault(be, [[], [], [], []])
+                    g[0].append(i)
+                    g[1].append(mi.full_method_path)
+                    g[2].append(pbs[i])
+                    g[3].append(list(hdr.items()) if hdr else [])
+                else:
+                    stream_futs[i] = self._invoke_pool.submit(
+                        self.discoverer.invoke_streaming_wire, mi, pbs[i], hdr, timeout0
+                    )
+        for be, g in stream_per_be.items():
+            def run_stream_backend(be=be, g=g):
+                res = g[0], self.wire_clients[be].invoke_stream_batch(
+                    g[1], g[2], timeout0, g[3]
                 )
+                return res
+            fut = self._invoke_pool.submit(run_stream_backend)
+            for k, i in enumerate(g[0]):
+                stream_futs[i] = _SlotOfBatch(fut, k)
 
         t0 = time.perf_counter_ns()
         timeout = timeout_s if timeout_s is not None else self.config.grpc.request_timeout_s
