@@ -94,14 +94,21 @@ class NativeWireClient:
         metadata: Optional[Sequence[Sequence[Tuple[str, str]]]] = None,
     ) -> List[object]:
         """Server-streaming batch: each slot resolves to a list of wire
-        chunks (all messages of the stream) or NativeRpcError."""
+        chunks — zero-copy memoryview slices of one blob per stream — or
+        NativeRpcError."""
         res = self._cli.invoke_stream_batch(
             list(paths), list(payloads), timeout_s, list(metadata or [])
         )
         out: List[object] = []
-        for status, msgs, message in res:
+        for status, blob, lens, message in res:
             if status == 0:
-                out.append(list(msgs))
+                mv = memoryview(blob)
+                chunks = []
+                off = 0
+                for ln in lens:
+                    chunks.append(mv[off : off + ln])
+                    off += ln
+                out.append(chunks)
             else:
                 out.append(NativeRpcError(status, message))
         return out
