@@ -461,7 +461,7 @@ class GpuPipeline:
         for base in range(0, len(flat), cap):
             with engine._lock:
                 _, part = engine.decode_batch(
-                    flat[base : base + cap], flat_idx[base : base + cap], mode=1
+                    flat[base : base + cap], flat_idx[base : base + cap], mode=2
                 )
             jsons.extend(part)
         out: Dict[int, bytes] = {}
@@ -479,17 +479,23 @@ class GpuPipeline:
                 out[i] = json.dumps(resp.to_dict(), ensure_ascii=False).encode()
                 continue
             start, count = spans[i]
-            texts: List[str] = []
+            items: List[bytes] = []
             ok = True
             mi = self._mi_by_idx[int(enc[i]["tool_idx"])]
             for k in range(start, start + count):
                 if jsons[k] is not None:
-                    texts.append(jsons[k].decode())
+                    # already a complete escaped {"type":"text","text":"..."}
+                    items.append(jsons[k])
                 else:
                     # per-chunk CPU fallback (counted, never silent)
                     st.host_fallbacks += 1
                     try:
-                        texts.append(self.cpu.pb_to_json(mi.output_descriptor, flat[k]))
+                        t = self.cpu.pb_to_json(mi.output_descriptor, flat[k])
+                        items.append(
+                            b'{"type":"text","text":'
+                            + json.dumps(t, ensure_ascii=False).encode()
+                            + b"}"
+                        )
                     except Exception as e:
                         st.errors += 1
                         resp = mcp.JSONRPCResponse(
@@ -501,11 +507,13 @@ class GpuPipeline:
             if not ok:
                 continue
             st.gpu_ok += 1
-            result = mcp.ToolCallResult(
-                content=[mcp.TextContent(t) for t in texts], is_error=False
+            out[i] = (
+                b'{"jsonrpc":"2.0","id":'
+                + json.dumps(rid).encode()
+                + b',"result":{"content":['
+                + b",".join(items)
+                + b'],"isError":false}}'
             )
-            resp = mcp.JSONRPCResponse(id=rid, result=result.to_dict())
-            out[i] = json.dumps(resp.to_dict(), ensure_ascii=False).encode()
         return out
 
     # ---- host handling of non-GPU slots ------------------------------------

@@ -1392,7 +1392,7 @@ extern "C" __global__ void __launch_bounds__(WPB * WAVE) k_pb2json(
     o.status = E_OK;
     o.lane = lane;
     if (mode == 1) {
-      // bare JSON (transcode tests / streaming chunks): copy scratch out
+      // bare JSON (transcode tests): copy scratch out
       if (json_len > o.ocap) {
         r.status = E_OVERFLOW;
         if (!lane) results[req] = r;
@@ -1401,6 +1401,21 @@ extern "C" __global__ void __launch_bounds__(WPB * WAVE) k_pb2json(
       for (uint32_t i = lane; i < json_len; i += WAVE)
         o.out[i] = scratch[scratch_off[req] + i];
       o.opos = json_len;
+    } else if (mode == 2) {
+      // streaming content item: the chunk pre-wrapped + escaped so the host
+      // assembles a multi-chunk ToolCallResult by joining byte slices
+      // (json-escaping 4096 chunk texts per stream in Python dominated the
+      // streaming step otherwise)
+      static const char CP1[] = "{\"type\":\"text\",\"text\":\"";
+      static const char CP2[] = "\"}";
+      bool ok = puts_(o, CP1, sizeof(CP1) - 1);
+      if (ok) ok = put_escaped(o, scratch + scratch_off[req], json_len);
+      if (ok) ok = puts_(o, CP2, sizeof(CP2) - 1);
+      if (!ok) {
+        r.status = o.status;
+        if (!lane) results[req] = r;
+        continue;
+      }
     } else {
       // JSON-RPC result envelope (handler.go:265-270 + TextContent wrap)
       static const char P1[] = "{\"jsonrpc\":\"2.0\",\"id\":";

@@ -218,3 +218,18 @@ def test_struct_value_roundtrip_hostsim():
     assert dec[0]["status"] == 0, dec[0]
     assert _approx(json.loads(outs[0]), json.loads(cpu2.pb_to_json(desc, wire))), \
         f"\nsim: {outs[0]!r}\noracle: {cpu2.pb_to_json(desc, wire)!r}"
+
+
+def test_mode2_content_item_wrapping(env):
+    """Decode mode 2 emits the chunk pre-wrapped as an escaped MCP content
+    item (the streaming path joins these byte-for-byte)."""
+    e, c, p, _ = env
+    desc = p.FindMessageTypeByName("hello.HelloResponse")
+    wire = c.json_to_pb(desc, json.dumps({"message": 'say "hi"\n'}))
+    idx = e.tables.msg_index["hello.HelloResponse"]
+    dec, outs = e.decode_batch([wire], [idx], mode=2)
+    assert dec[0]["status"] == 0
+    item = json.loads(outs[0])
+    assert item["type"] == "text"
+    inner = json.loads(item["text"])
+    assert inner == {"message": 'say "hi"\n'}
