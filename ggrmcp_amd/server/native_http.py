@@ -160,16 +160,50 @@ class NativeHTTPGateway:
         n = len(bodies)
         sids: List[str] = []
         fwd_headers: List[Optional[Dict[str, str]]] = []
+        rejected: Dict[int, bytes] = {}
+        rl_on = self.config.session.rate_limit_enabled
         for i in range(n):
             sess = self.sessions.get_or_create(session_ids[i], headers[i])
-            sess.increment_call_count()
             sids.append(sess.id)
+            # per-session guards (handler.go:219-226 parity)
+            if sess.is_blocked:
+                rejected[i] = self._session_error(bodies[i], "session is blocked")
+                continue
+            if rl_on and not self.sessions.check_rate_limit(sess):
+                rejected[i] = self._session_error(bodies[i], "session rate limit exceeded")
+                continue
+            sess.increment_call_count()
             fwd_headers.append(self.headers.filter_headers(headers[i]))
-        out = self.pipeline.process_batch(
-            bodies, headers=fwd_headers,
-            timeout_s=self.config.grpc.request_timeout_s,
-        )
+        if rejected:
+            live_idx = [i for i in range(n) if i not in rejected]
+            live = [bodies[i] for i in live_idx]
+            out_live = self.pipeline.process_batch(
+                live,
+                headers=fwd_headers,
+                timeout_s=self.config.grpc.request_timeout_s,
+            ) if live else []
+            out: List[bytes] = [b""] * n
+            for k, i in enumerate(live_idx):
+                out[i] = out_live[k]
+            for i, resp in rejected.items():
+                out[i] = resp
+        else:
+            out = self.pipeline.process_batch(
+                bodies, headers=fwd_headers,
+                timeout_s=self.config.grpc.request_timeout_s,
+            )
         return list(zip(out, sids))
+
+    @staticmethod
+    def _session_error(body: bytes, message: str) -> bytes:
+        try:
+            rid = json.loads(body).get("id")
+        except Exception:
+            rid = None
+        resp = mcp.JSONRPCResponse(
+            id=rid, error=mcp.RPCError(mcp.INVALID_REQUEST, message)
+        )
+        return json.dumps(resp.to_dict()).encode()
 
     # ---- non-tools/call JSON-RPC on the batch path --------------------------
 

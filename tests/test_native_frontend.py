@@ -192,3 +192,18 @@ def test_404(gateway):
     gw, port, pipe = gateway
     status, _, _ = _call(port, None, method="GET", path="/nope")
     assert status == 404
+
+
+def test_blocked_session_and_rate_limit(gateway):
+    gw, port, pipe = gateway
+    body = json.dumps({"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                       "params": {"name": "t", "arguments": {}}})
+    _, _, sid = _call(port, body)
+    gw.sessions.block(sid)
+    status, data, _ = _call(port, body, session=sid)
+    resp = json.loads(data)
+    assert resp["error"]["code"] == -32600
+    assert "blocked" in resp["error"]["message"]
+    gw.sessions.unblock(sid)
+    status, data, _ = _call(port, body, session=sid)
+    assert json.loads(data)["result"]["isError"] is False
