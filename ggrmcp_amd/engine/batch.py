@@ -159,7 +159,12 @@ class GpuEngine:
             cap_final=gpu.device_pool_bytes // 4,
         )
         self.stats = stats if stats is not None else EngineStats()
-        self._lock = threading.Lock()  # one in-flight batch per engine
+        # re-entrant: a span holds the engine for its whole encode ->
+        # invoke -> decode sequence (mode-0 decode consumes the id slots of
+        # THIS engine's last encode; interleaving another span's encode
+        # between them would silently mis-id responses), while the inner
+        # batch ops also take the lock for standalone callers
+        self._lock = threading.RLock()
 
     # -- low-level batch ops -------------------------------------------------
 
@@ -250,8 +255,9 @@ class GpuPipeline:
             max_workers=invoke_workers, thread_name_prefix="ginvoke"
         )
         self._chunk_pool = ThreadPoolExecutor(
-            max_workers=n_engines, thread_name_prefix="gchunk"
+            max_workers=max(2 * n_engines, 4), thread_name_prefix="gchunk"
         )
+        self._engine_rr = 0
         # optional: handles well-formed JSON-RPC envelopes whose method is
         # NOT tools/call (initialize, tools/list, ...) on the batch path;
         # (body: bytes, headers: dict|None) -> response bytes
@@ -292,13 +298,19 @@ class GpuPipeline:
             return self._process_span(self.engine, bodies, headers, timeout_s)
         n_chunks = min(n_eng, (n + min_chunk - 1) // min_chunk)
         bounds = [round(i * n / n_chunks) for i in range(n_chunks + 1)]
+        # rotate the starting engine so concurrent batches (two frontend
+        # workers) land on disjoint engines instead of all serializing on
+        # engines[0]
+        base = self._engine_rr
+        self._engine_rr = (base + n_chunks) % n_eng
         futs = []
         for k in range(n_chunks):
             lo, hi = bounds[k], bounds[k + 1]
             hdr = headers[lo:hi] if headers else None
             futs.append(
                 self._chunk_pool.submit(
-                    self._process_span, self.engines[k], bodies[lo:hi], hdr, timeout_s
+                    self._process_span, self.engines[(base + k) % n_eng],
+                    bodies[lo:hi], hdr, timeout_s
                 )
             )
         out: List[bytes] = []
@@ -313,11 +325,20 @@ class GpuPipeline:
         headers: Optional[Sequence[Dict[str, str]]] = None,
         timeout_s: Optional[float] = None,
     ) -> List[bytes]:
+        with engine._lock:
+            return self._process_span_locked(engine, bodies, headers, timeout_s)
+
+    def _process_span_locked(
+        self,
+        engine: GpuEngine,
+        bodies: Sequence[bytes],
+        headers: Optional[Sequence[Dict[str, str]]] = None,
+        timeout_s: Optional[float] = None,
+    ) -> List[bytes]:
         st = engine.stats
         st.batches += 1
         st.requests += len(bodies)
-        with engine._lock:
-            enc, pbs = engine.encode_batch(bodies, mode=0)
+        enc, pbs = engine.encode_batch(bodies, mode=0)
 
         # fan out gRPC invocations for OK slots (host-side I/O stage)
         n = len(bodies)
@@ -412,8 +433,7 @@ class GpuPipeline:
                     rpc_error[i] = e
         st.invoke_ns += time.perf_counter_ns() - t0
 
-        with engine._lock:
-            dec, finals = engine.decode_batch(resp_wire, out_idx, mode=0)
+        dec, finals = engine.decode_batch(resp_wire, out_idx, mode=0)
 
         # streaming: gather chunk lists, decode ALL chunks of ALL streams in
         # one value-mode GPU batch, assemble envelopes host-side

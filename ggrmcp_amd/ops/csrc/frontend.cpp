@@ -107,11 +107,13 @@ class Frontend {
  public:
   Frontend(const std::string& host, int port, py::function batch_cb,
            py::function slow_cb, int batch_window_us, int max_batch,
-           size_t max_body, double rate_rps, double rate_burst)
+           size_t max_body, double rate_rps, double rate_burst,
+           int workers)
       : host_(host), port_(port), batch_cb_(batch_cb), slow_cb_(slow_cb),
         window_us_(batch_window_us), max_batch_(max_batch),
         max_body_(max_body), rate_rps_(rate_rps),
-        tokens_(rate_burst), burst_(rate_burst) {}
+        tokens_(rate_burst), burst_(rate_burst),
+        n_workers_(workers < 1 ? 1 : workers) {}
 
   ~Frontend() { stop(); }
 
@@ -141,8 +143,10 @@ class Frontend {
     epoll_ctl(epfd_, EPOLL_CTL_ADD, wake_fd_, &ev);
 
     stop_.store(false);
+    idle_workers_.store(n_workers_);
     io_thread_ = std::thread([this] { io_loop(); });
-    worker_thread_ = std::thread([this] { worker_loop(); });
+    for (int i = 0; i < n_workers_; ++i)
+      worker_threads_.emplace_back([this] { worker_loop(); });
     return port_;
   }
 
@@ -155,16 +159,21 @@ class Frontend {
       batch_cv_.notify_all();
     }
     if (io_thread_.joinable()) io_thread_.join();
-    if (worker_thread_.joinable()) {
-      // The worker may be blocked acquiring the GIL; if this thread holds
-      // it, release while joining (stop() is reachable both from Python
-      // calls that hold the GIL and from C++ teardown that doesn't).
+    {
+      // Workers may be blocked acquiring the GIL; if this thread holds it,
+      // release while joining (stop() is reachable both from Python calls
+      // that hold the GIL and from C++ teardown that doesn't).
+      auto join_all = [this] {
+        for (auto& t : worker_threads_)
+          if (t.joinable()) t.join();
+      };
       if (PyGILState_Check()) {
         py::gil_scoped_release rel;
-        worker_thread_.join();
+        join_all();
       } else {
-        worker_thread_.join();
+        join_all();
       }
+      worker_threads_.clear();
     }
     for (auto& kv : conns_) close(kv.second->fd);
     conns_.clear();
@@ -210,7 +219,7 @@ class Frontend {
       (void)now;
       if (!pending_.empty() &&
           (pending_.size() >= (size_t)max_batch_ ||
-           !worker_busy_.load(std::memory_order_acquire))) {
+           idle_workers_.load(std::memory_order_acquire) > 0)) {
         std::lock_guard<std::mutex> lk(batch_mu_);
         batches_.emplace_back(std::move(pending_));
         pending_.clear();
@@ -425,7 +434,6 @@ class Frontend {
       std::vector<PendingReq> batch;
       {
         std::unique_lock<std::mutex> lk(batch_mu_);
-        worker_busy_.store(false, std::memory_order_release);
         batch_cv_.wait(lk, [this] { return stop_.load() || !batches_.empty(); });
         if (stop_.load() && batches_.empty()) return;
         batch = std::move(batches_.front());
@@ -438,7 +446,7 @@ class Frontend {
                        std::make_move_iterator(nxt.end()));
           batches_.pop_front();
         }
-        worker_busy_.store(true, std::memory_order_release);
+        idle_workers_.fetch_sub(1, std::memory_order_acq_rel);
       }
       std::vector<OutResp> out;
       out.reserve(batch.size());
@@ -502,6 +510,7 @@ class Frontend {
         std::lock_guard<std::mutex> lk(done_mu_);
         for (auto& r : out) done_.push_back(std::move(r));
       }
+      idle_workers_.fetch_add(1, std::memory_order_acq_rel);
       uint64_t one = 1;
       (void)!write(wake_fd_, &one, 8);
     }
@@ -518,7 +527,10 @@ class Frontend {
 
   int listen_fd_ = -1, epfd_ = -1, wake_fd_ = -1;
   std::atomic<bool> stop_{true};
-  std::thread io_thread_, worker_thread_;
+  std::thread io_thread_;
+  std::vector<std::thread> worker_threads_;
+  int n_workers_ = 1;
+  std::atomic<int> idle_workers_{0};
   uint64_t next_conn_id_ = 0;
   std::unordered_map<uint64_t, std::unique_ptr<Conn>> conns_;
 
@@ -529,18 +541,18 @@ class Frontend {
   std::deque<std::vector<PendingReq>> batches_;
   std::mutex done_mu_;
   std::deque<OutResp> done_;
-  std::atomic<bool> worker_busy_{false};
 };
 
 PYBIND11_MODULE(_frontend, m) {
   m.doc() = "native HTTP/1.1 batch ingestion front end for the MCP gateway";
   py::class_<Frontend>(m, "Frontend")
       .def(py::init<const std::string&, int, py::function, py::function, int,
-                    int, size_t, double, double>(),
+                    int, size_t, double, double, int>(),
            py::arg("host"), py::arg("port"), py::arg("batch_cb"),
            py::arg("slow_cb"), py::arg("batch_window_us") = 200,
            py::arg("max_batch") = 4096, py::arg("max_body") = 1 << 20,
-           py::arg("rate_rps") = 0.0, py::arg("rate_burst") = 0.0)
+           py::arg("rate_rps") = 0.0, py::arg("rate_burst") = 0.0,
+           py::arg("workers") = 2)
       .def("start", &Frontend::start)
       .def("stop", &Frontend::stop)
       .def_property_readonly("port", &Frontend::port);
