@@ -203,3 +203,59 @@ def test_fuzz_struct_value(value):
     gj = json.loads(outs[0])
     oj = json.loads(_cpu.pb_to_json(_sdesc, oracle_wire))
     assert _approx(gj, oj), f"\npayload: {text!r}\nsim:    {gj}\noracle: {oj}"
+
+
+# ---- envelope mode: JSON-RPC validation parity ------------------------------
+
+_env_engine = _engine  # same tool tables
+
+_tool_names = st.sampled_from([
+    "hello_helloservice_sayhello",        # valid unary
+    "complex_nodeservice_streamnodes",    # valid server-streaming
+    "no_such_tool",                       # -> E_METHOD_NOT_FOUND
+])
+_rpc_ids = st.one_of(st.integers(min_value=-(10**12), max_value=10**12),
+                     st.text(alphabet=string.ascii_letters + "-", max_size=24),
+                     st.none())
+_methods = st.sampled_from(["tools/call", "tools/list", "initialize", "bogus/x"])
+
+
+@settings(max_examples=150, deadline=None)
+@given(rid=_rpc_ids, method=_methods, tool=_tool_names,
+       args=st.dictionaries(st.sampled_from(["name"]), _text, max_size=1),
+       jsonrpc=st.sampled_from(["2.0", "1.0", None]),
+       drop_id=st.booleans())
+def test_fuzz_envelope_statuses(rid, method, tool, args, jsonrpc, drop_id):
+    body = {"method": method, "params": {"name": tool, "arguments": args}}
+    if jsonrpc is not None:
+        body["jsonrpc"] = jsonrpc
+    if not drop_id:
+        body["id"] = rid
+    text = json.dumps(body, ensure_ascii=False).encode()
+    enc, pbs = _env_engine.encode_batch([text], mode=0)
+    status = int(enc[0]["status"])
+    flags = int(enc[0]["flags"])
+    if jsonrpc != "2.0":
+        assert status == 2, (status, body)  # E_INVALID_REQUEST
+        return
+    if method != "tools/call":
+        assert status == 8, (status, body)  # E_NOT_TOOLCALL
+        return
+    if tool == "no_such_tool":
+        assert status == 3, (status, body)  # E_METHOD_NOT_FOUND
+        return
+    if tool == "complex_nodeservice_streamnodes":
+        if args:
+            # NodeRequest has no "name" field -> unknown field rejected
+            # (protojson parity, builder.go strictness)
+            assert status == 4, (status, body)
+            return
+        assert status == 0, (status, int(enc[0]["aux"]), body)
+        assert flags & 2, flags  # SR_SERVER_STREAMING
+    else:
+        assert status == 0, (status, int(enc[0]["aux"]), body)
+    if drop_id:
+        assert flags & 1, flags  # SR_ID_IS_MISSING
+    else:
+        # the raw JSON id token is captured for the response envelope
+        assert int(enc[0]["id_len"]) > 0, enc[0]
