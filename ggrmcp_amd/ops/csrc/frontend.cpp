@@ -552,7 +552,7 @@ PYBIND11_MODULE(_frontend, m) {
            py::arg("slow_cb"), py::arg("batch_window_us") = 200,
            py::arg("max_batch") = 4096, py::arg("max_body") = 1 << 20,
            py::arg("rate_rps") = 0.0, py::arg("rate_burst") = 0.0,
-           py::arg("workers") = 2)
+           py::arg("workers") = 1)
       .def("start", &Frontend::start)
       .def("stop", &Frontend::stop)
       .def_property_readonly("port", &Frontend::port);
