@@ -142,6 +142,14 @@ class GpuEngine:
         self.tables: CompiledTables = tables if tables is not None else compile_tables(tools)
         self.tools = tools
         gpu = self.config.gpu
+        # tool metadata for the fully-native span executor
+        mis = [tools[name] for name in self.tables.tool_order]
+        tool_paths = [mi.full_method_path for mi in mis]
+        tool_out_msg = np.asarray(
+            [self.tables.msg_index[mi.output_descriptor.full_name] for mi in mis],
+            dtype=np.int32,
+        )
+        tool_backend = np.asarray([mi.backend_index for mi in mis], dtype=np.int32)
         self._eng = self._mod.Engine(
             device,
             self.tables.msg_table,
@@ -157,6 +165,9 @@ class GpuEngine:
             cap_pb=gpu.pinned_pool_bytes // 4,
             cap_scratch=gpu.device_pool_bytes // 4,
             cap_final=gpu.device_pool_bytes // 4,
+            tool_paths=tool_paths,
+            tool_out_msg=tool_out_msg,
+            tool_backend=tool_backend,
         )
         self.stats = stats if stats is not None else EngineStats()
         # re-entrant: a span holds the engine for its whole encode ->
@@ -338,6 +349,8 @@ class GpuPipeline:
         st = engine.stats
         st.batches += 1
         st.requests += len(bodies)
+        if self.wire_clients:
+            return self._native_span(engine, bodies, headers, timeout_s)
         enc, pbs = engine.encode_batch(bodies, mode=0)
 
         # fan out gRPC invocations for OK slots (host-side I/O stage)
@@ -533,6 +546,87 @@ class GpuPipeline:
                 + b',"result":{"content":['
                 + b",".join(items)
                 + b'],"isError":false}}'
+            )
+        return out
+
+
+    def _native_span(
+        self,
+        engine: GpuEngine,
+        bodies: Sequence[bytes],
+        headers: Optional[Sequence[Dict[str, str]]],
+        timeout_s: Optional[float],
+    ) -> List[bytes]:
+        """encode -> invoke -> decode entirely in C++ (engine.process_span);
+        Python only assembles error/streaming/fallback slots."""
+        st = engine.stats
+        timeout = timeout_s if timeout_s is not None else self.config.grpc.request_timeout_s
+        t0 = time.perf_counter_ns()
+        enc_raw, dec_raw, out_view, pb_view, rpc_errs = engine._eng.process_span(
+            list(bodies),
+            list(headers) if headers else None,
+            [c._cli for c in self.wire_clients],
+            timeout,
+        )
+        st.invoke_ns += time.perf_counter_ns() - t0
+        enc = np.frombuffer(enc_raw.tobytes(), dtype=SLOT_DTYPE)
+        dec = np.frombuffer(dec_raw.tobytes(), dtype=DECODE_DTYPE)
+        n = len(bodies)
+        mem = memoryview(out_view)
+        pb_mem = memoryview(pb_view)
+
+        # streaming slots: batched native stream invoke + GPU chunk decode
+        stream_futs: Dict[int, Any] = {}
+        stream_per_be: Dict[int, List[Any]] = {}
+        for i in range(n):
+            if enc[i]["status"] == E_OK and enc[i]["flags"] & SR_SERVER_STREAMING:
+                mi = self._mi_by_idx[enc[i]["tool_idx"]]
+                hdr = headers[i] if headers else None
+                be = mi.backend_index if mi.backend_index < len(self.wire_clients) else 0
+                g = stream_per_be.setdefault(be, [[], [], [], []])
+                g[0].append(i)
+                g[1].append(mi.full_method_path)
+                g[2].append(bytes(pb_mem[enc[i]["pb_off"] : enc[i]["pb_off"] + enc[i]["pb_len"]]))
+                g[3].append(list(hdr.items()) if hdr else [])
+        for be, g in stream_per_be.items():
+            def run_stream_backend(be=be, g=g):
+                return g[0], self.wire_clients[be].invoke_stream_batch(
+                    g[1], g[2], timeout, g[3]
+                )
+            fut = self._invoke_pool.submit(run_stream_backend)
+            for k, i in enumerate(g[0]):
+                stream_futs[i] = _SlotOfBatch(fut, k)
+        stream_out: Dict[int, bytes] = {}
+        if stream_futs:
+            stream_out = self._decode_streams(engine, stream_futs, enc, bodies)
+
+        out: List[bytes] = []
+        for i in range(n):
+            r = dec[i]
+            if r["status"] == E_OK and r["out_len"] > 0 and rpc_errs[i] is None and \
+                    enc[i]["status"] == E_OK and not (enc[i]["flags"] & SR_SERVER_STREAMING):
+                st.gpu_ok += 1
+                out.append(bytes(mem[r["out_off"] : r["out_off"] + r["out_len"]]))
+                continue
+            if i in stream_out:
+                out.append(stream_out[i])
+                continue
+            err = rpc_errs[i]
+            rpc_exc = None
+            wire_present = None
+            if err is not None:
+                from ..backend.native_invoker import NativeRpcError
+
+                rpc_exc = NativeRpcError(int(err[0]), err[1])
+            elif enc[i]["status"] == E_OK and not (enc[i]["flags"] & SR_SERVER_STREAMING):
+                wire_present = b""  # decode failed on a delivered response
+            out.append(
+                self._host_slot(
+                    engine, bodies[i], enc[i],
+                    dec[i] if wire_present is not None else None,
+                    wire_present, rpc_exc,
+                    headers[i] if headers else None, timeout_s,
+                )
             )
         return out
 

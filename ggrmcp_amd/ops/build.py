@@ -43,12 +43,14 @@ def _common_flags():
 
 def build_jsonproto(verbose: bool = True, force: bool = False) -> Path:
     sources = [CSRC / "engine.cpp", CSRC / "json2pb.hip", CSRC / "pb2json.hip",
-               CSRC / "common.h"]
+               CSRC / "common.h", CSRC / "h2grpc_impl.h"]
     if not force and not _needs(SO_PATH, sources):
         return SO_PATH
     hipcc = os.environ.get("HIPCC", "hipcc")
     cmd = [hipcc, "-x", "hip", str(CSRC / "engine.cpp"),
-           f"--offload-arch={GFX_ARCH}"] + _common_flags() + ["-o", str(SO_PATH)]
+           f"--offload-arch={GFX_ARCH}"] + _common_flags() + [
+           f"-I{NGHTTP2_INCLUDE}", "-l:libnghttp2.so.14", "-pthread",
+           "-o", str(SO_PATH)]
     if verbose:
         print("[ggrmcp-amd build]", " ".join(cmd), file=sys.stderr, flush=True)
     subprocess.run(cmd, check=True)
@@ -56,7 +58,7 @@ def build_jsonproto(verbose: bool = True, force: bool = False) -> Path:
 
 
 def build_h2grpc(verbose: bool = True, force: bool = False) -> Path:
-    sources = [CSRC / "h2grpc.cpp"]
+    sources = [CSRC / "h2grpc.cpp", CSRC / "h2grpc_impl.h"]
     if not force and not _needs(H2_SO_PATH, sources):
         return H2_SO_PATH
     cxx = os.environ.get("CXX", "g++")
