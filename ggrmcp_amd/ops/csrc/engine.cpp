@@ -430,6 +430,7 @@ class Engine {
     int32_t* h_aux = (int32_t*)h_aux_.p;
     std::vector<const std::string*> resp_ptr(n, nullptr);
     py::list rpc_errors(n);
+    for (int i = 0; i < n; ++i) rpc_errors[i] = py::none();
     for (size_t b = 0; b < nb; ++b) {
       for (size_t k = 0; k < slots_be[b].size(); ++k) {
         int i = slots_be[b][k];
@@ -664,6 +665,13 @@ static int device_count() {
 PYBIND11_MODULE(_jsonproto, m) {
   m.doc() = "MI355X batch JSON<->protobuf transcode engine (gfx950 HIP kernels)";
   m.def("device_count", &device_count);
+  // diagnostic: verify the cross-module cast of an _h2grpc Client works in
+  // this process (pybind shares its type registry across extensions when
+  // the class typeinfo is default-visibility)
+  m.def("probe_client", [](py::object obj) {
+    H2GrpcClient* c = obj.cast<H2GrpcClient*>();
+    return c->healthy();
+  });
   m.attr("ID_SLOT_BYTES") = ID_SLOT_BYTES;
   m.attr("SLOT_RESULT_SIZE") = (int)sizeof(SlotResult);
   m.attr("DECODE_RESULT_SIZE") = (int)sizeof(DecodeResult);
