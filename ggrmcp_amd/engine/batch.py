@@ -574,6 +574,15 @@ class GpuPipeline:
         n = len(bodies)
         mem = memoryview(out_view)
         pb_mem = memoryview(pb_view)
+        # materialize unary responses NOW: the stream-chunk decode below
+        # reuses the engine's pinned output buffer and would overwrite them
+        finals: List[Optional[bytes]] = [None] * n
+        for i in range(n):
+            r = dec[i]
+            if (r["status"] == E_OK and r["out_len"] > 0 and rpc_errs[i] is None
+                    and enc[i]["status"] == E_OK
+                    and not (enc[i]["flags"] & SR_SERVER_STREAMING)):
+                finals[i] = bytes(mem[r["out_off"] : r["out_off"] + r["out_len"]])
 
         # streaming slots: batched native stream invoke + GPU chunk decode
         stream_futs: Dict[int, Any] = {}
@@ -602,11 +611,9 @@ class GpuPipeline:
 
         out: List[bytes] = []
         for i in range(n):
-            r = dec[i]
-            if r["status"] == E_OK and r["out_len"] > 0 and rpc_errs[i] is None and \
-                    enc[i]["status"] == E_OK and not (enc[i]["flags"] & SR_SERVER_STREAMING):
+            if finals[i] is not None:
                 st.gpu_ok += 1
-                out.append(bytes(mem[r["out_off"] : r["out_off"] + r["out_len"]]))
+                out.append(finals[i])
                 continue
             if i in stream_out:
                 out.append(stream_out[i])
