@@ -24,6 +24,7 @@ def pipeline_env():
     srv = mod.Server("127.0.0.1:0")
     srv.add_route("/hello.HelloService/SayHello", "hello")
     srv.add_route("/bench.EchoService/Echo", "echo")
+    srv.add_route("/bench.EchoService/StreamEcho", "stream_echo")
     bound = srv.start()
 
     cfg = Config.default()
@@ -166,3 +167,30 @@ def test_process_batch_streaming_gpu_decode(grpcio_pipeline_env):
     # streams took the GPU decode path, not the host fallback
     assert pipeline.engine.stats.gpu_ok >= before_ok + 3
     assert pipeline.engine.stats.host_fallbacks == 0
+
+
+def test_native_mixed_unary_and_stream(pipeline_env):
+    """Unary + server-streaming slots in ONE batch through the native span:
+    the stream-chunk decode must not clobber the unary responses
+    (regression: both shared the engine's pinned output buffer)."""
+    pipeline = pipeline_env
+    bodies = []
+    for i in range(12):
+        if i % 3 == 2:
+            bodies.append(_body("bench_echoservice_streamecho",
+                                {"f01String": f"s{i}", "f02Int32": 5}, i))
+        else:
+            bodies.append(_body("hello_helloservice_sayhello",
+                                {"name": f"u{i}"}, i))
+    out = pipeline.process_batch(bodies, timeout_s=15.0)
+    for i, raw in enumerate(out):
+        resp = json.loads(raw)
+        assert resp["id"] == i, resp
+        assert resp["result"]["isError"] is False, resp
+        if i % 3 == 2:
+            assert len(resp["result"]["content"]) == 5
+            inner = json.loads(resp["result"]["content"][0]["text"])
+            assert inner["f01String"] == f"s{i}"
+        else:
+            inner = json.loads(resp["result"]["content"][0]["text"])
+            assert inner == {"message": f"Hello, u{i}!"}
