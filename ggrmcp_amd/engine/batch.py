@@ -562,7 +562,7 @@ class GpuPipeline:
         st = engine.stats
         timeout = timeout_s if timeout_s is not None else self.config.grpc.request_timeout_s
         t0 = time.perf_counter_ns()
-        enc_raw, dec_raw, out_view, pb_view, rpc_errs = engine._eng.process_span(
+        enc_raw, dec_raw, out_view, stream_pbs, rpc_errs = engine._eng.process_span(
             list(bodies),
             list(headers) if headers else None,
             [c._cli for c in self.wire_clients],
@@ -573,7 +573,6 @@ class GpuPipeline:
         dec = np.frombuffer(dec_raw.tobytes(), dtype=DECODE_DTYPE)
         n = len(bodies)
         mem = memoryview(out_view)
-        pb_mem = memoryview(pb_view)
         # materialize unary responses NOW: the stream-chunk decode below
         # reuses the engine's pinned output buffer and would overwrite them
         finals: List[Optional[bytes]] = [None] * n
@@ -595,7 +594,7 @@ class GpuPipeline:
                 g = stream_per_be.setdefault(be, [[], [], [], []])
                 g[0].append(i)
                 g[1].append(mi.full_method_path)
-                g[2].append(bytes(pb_mem[enc[i]["pb_off"] : enc[i]["pb_off"] + enc[i]["pb_len"]]))
+                g[2].append(stream_pbs[i])
                 g[3].append(list(hdr.items()) if hdr else [])
         for be, g in stream_per_be.items():
             def run_stream_backend(be=be, g=g):

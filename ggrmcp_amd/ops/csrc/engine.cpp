@@ -410,6 +410,17 @@ class Engine {
       slots_be[be].push_back(i);
     }
 
+    // streaming slots: copy their request wire OUT now — run_decode's
+    // output compaction reuses h_pb_, which would clobber them before the
+    // Python streaming fan-out reads the bytes
+    py::list stream_pbs(n);
+    for (int i = 0; i < n; ++i) {
+      if (rs[i].status == E_OK && (rs[i].flags & SR_SERVER_STREAMING))
+        stream_pbs[i] = py::bytes((const char*)pb + rs[i].pb_off, rs[i].pb_len);
+      else
+        stream_pbs[i] = py::none();
+    }
+
     // invoke (GIL released; backends run concurrently inside the client's
     // connection threads — calls to different clients issue sequentially
     // but each returns only after ITS batch completes, so issue all, then
@@ -486,7 +497,7 @@ class Engine {
     if (facc > d_final_.n) throw std::runtime_error("final cap");
     py::tuple dec_out = run_decode(n, racc, facc, true, 0);
 
-    return py::make_tuple(enc_out[0], dec_out[0], dec_out[1], enc_out[1],
+    return py::make_tuple(enc_out[0], dec_out[0], dec_out[1], stream_pbs,
                           rpc_errors);
   }
 
