@@ -142,3 +142,66 @@ def test_descriptor_blob_roundtrip(discoverer, backend):
     other = ServiceDiscoverer(cfg)
     other.load_descriptor_blob(blob)
     assert set(other.tools) == set(discoverer.tools)
+
+
+def test_descriptor_set_file_discovery(tmp_path):
+    """--descriptor route: .binpb file -> loader -> tool map
+    (discovery.go:101-119 descriptor-set-first path)."""
+    from google.protobuf import descriptor_pb2
+
+    from examples.protos import ALL_FDPS
+    from ggrmcp_amd.config import Config
+    from ggrmcp_amd.backend.discovery import ServiceDiscoverer
+    from ggrmcp_amd.utils.synthetic import synthetic_fdp
+
+    fdset = descriptor_pb2.FileDescriptorSet()
+    fdset.file.extend(ALL_FDPS + [synthetic_fdp()])
+    path = tmp_path / "services.binpb"
+    path.write_bytes(fdset.SerializeToString())
+
+    cfg = Config.default()
+    cfg.descriptor_set.enabled = True
+    cfg.descriptor_set.path = str(path)
+    d = ServiceDiscoverer(cfg)
+    tools = d.discover()  # no connection needed for the descriptor path
+    assert "hello_helloservice_sayhello" in tools
+    assert "bench_echoservice_echo" in tools
+    assert tools["bench_echoservice_streamecho"].is_server_streaming
+
+
+def test_descriptor_set_missing_file_falls_back(tmp_path, backend):
+    """Bad descriptor path logs a warning and falls back to reflection
+    (discovery.go:107-111)."""
+    from ggrmcp_amd.config import Config
+    from ggrmcp_amd.backend.discovery import ServiceDiscoverer
+
+    target = backend
+    cfg = Config.default()
+    host, _, port = target.rpartition(":")
+    cfg.grpc.host, cfg.grpc.port = host, int(port)
+    cfg.descriptor_set.enabled = True
+    cfg.descriptor_set.path = str(tmp_path / "nope.binpb")
+    d = ServiceDiscoverer(cfg)
+    d.connect(timeout_s=10)
+    tools = d.discover()
+    assert "hello_helloservice_sayhello" in tools  # via reflection
+    d.close()
+
+
+def test_reconnect_with_retry(backend):
+    """discovery.go:186-235: bounded reconnect + full rediscovery."""
+    from ggrmcp_amd.config import Config
+    from ggrmcp_amd.backend.discovery import ServiceDiscoverer
+
+    target = backend
+    cfg = Config.default()
+    host, _, port = target.rpartition(":")
+    cfg.grpc.host, cfg.grpc.port = host, int(port)
+    d = ServiceDiscoverer(cfg)
+    d.connect(timeout_s=10)
+    d.discover()
+    v1 = d.tools_version
+    assert d.reconnect_with_retry(attempts=2, delay_s=0.1)
+    assert d.tools_version > v1  # rediscovered + atomic republish
+    assert "hello_helloservice_sayhello" in d.tools
+    d.close()
