@@ -119,3 +119,66 @@ def test_serving_tools_list_and_errors(native_gateway):
     assert resp["error"]["code"] == -32601
     status, resp, _ = _post(port, b"{broken")
     assert resp["error"]["code"] == -32700
+
+
+def test_serving_soak_mixed(native_gateway):
+    """Soak: concurrent sessions mixing unary, streaming, errors and
+    malformed bodies through the reactor + native span; every response must
+    match its request id and shape (order/captivity races would mispair)."""
+    import random
+
+    gw, port, pipeline = native_gateway
+    errs = []
+
+    def worker(t):
+        rng = random.Random(t)
+        try:
+            conn = http.client.HTTPConnection("127.0.0.1", port, timeout=20)
+            for i in range(25):
+                kind = rng.randrange(5)
+                rid = f"{t}-{i}"
+                if kind == 0:
+                    body = json.dumps({"jsonrpc": "2.0", "id": rid,
+                                       "method": "tools/call",
+                                       "params": {"name": "bench_echoservice_streamecho",
+                                                  "arguments": {"f01String": rid,
+                                                                "f02Int32": 3}}})
+                elif kind == 1:
+                    body = json.dumps({"jsonrpc": "2.0", "id": rid,
+                                       "method": "tools/call",
+                                       "params": {"name": "missing_tool",
+                                                  "arguments": {}}})
+                elif kind == 2:
+                    body = json.dumps({"jsonrpc": "2.0", "id": rid,
+                                       "method": "tools/list"})
+                else:
+                    body = json.dumps({"jsonrpc": "2.0", "id": rid,
+                                       "method": "tools/call",
+                                       "params": {"name": "hello_helloservice_sayhello",
+                                                  "arguments": {"name": rid}}})
+                conn.request("POST", "/", body=body,
+                             headers={"Content-Type": "application/json"})
+                r = conn.getresponse()
+                resp = json.loads(r.read())
+                assert resp["id"] == rid, (rid, resp)
+                if kind == 0:
+                    assert len(resp["result"]["content"]) == 3
+                    inner = json.loads(resp["result"]["content"][1]["text"])
+                    assert inner["f01String"] == rid
+                elif kind == 1:
+                    assert resp["error"]["code"] == -32601
+                elif kind == 2:
+                    assert "tools" in resp["result"]
+                else:
+                    inner = json.loads(resp["result"]["content"][0]["text"])
+                    assert inner == {"message": f"Hello, {rid}!"}
+            conn.close()
+        except Exception as e:  # pragma: no cover
+            errs.append((t, repr(e)))
+
+    ts = [threading.Thread(target=worker, args=(t,)) for t in range(32)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join()
+    assert not errs, errs[:3]
