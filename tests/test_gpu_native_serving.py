@@ -26,6 +26,7 @@ def native_gateway():
     srv = mod.Server("127.0.0.1:0")
     srv.add_route("/hello.HelloService/SayHello", "hello")
     srv.add_route("/bench.EchoService/Echo", "echo")
+    srv.add_route("/bench.EchoService/StreamEcho", "stream_echo")
     bound = srv.start()
 
     cfg = Config.default()
