@@ -87,6 +87,16 @@ def _worker(rank: int, world: int, port: int, results_dir: str):
     assert stats["requests"] == 30.0
     assert stats["errors"] == 2.0
 
+    # rediscovery: a CHANGED tool set must swap atomically on every shard
+    # (the cross-shard analogue of discovery.go:122-127's atomic publish)
+    d2 = StubDisc()
+    d2.descriptor_blob = lambda: b"NEWSET" * 32
+    sync_discovery(d2, g, src=0)
+    if rank != 0:
+        assert d2.loaded == b"NEWSET" * 32
+    # sync_discovery's checksum all-gather + barrier already proved every
+    # shard holds identical bytes before anyone proceeds
+
     with open(os.path.join(results_dir, f"rank{rank}.json"), "w") as f:
         json.dump({"ok": True}, f)
     dist.destroy_process_group()
