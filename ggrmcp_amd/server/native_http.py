@@ -222,7 +222,7 @@ class NativeHTTPGateway:
         elif method == "tools/list":
             # schema cache keyed by the tool-map version (the reference
             # declares a cache but never uses it, builder.go:18,29)
-            self.tools.set_cache_key(self.discoverer.tools_version)
+            self.tools.set_cache_key(getattr(self.discoverer, "tools_version", 0))
             result = {"tools": [t.to_dict() for t in
                                 self.tools.build_tools(self.discoverer.get_methods())]}
         elif method == "prompts/list":
