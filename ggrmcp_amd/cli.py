@@ -146,9 +146,18 @@ async def run(cfg: Config, frontend: str = "asyncio") -> None:
     if frontend == "native":
         from .server.native_http import CpuBatchPipeline, NativeHTTPGateway
 
-        pipeline = (handler.invoker.pipeline
-                    if hasattr(handler.invoker, "pipeline")
-                    else CpuBatchPipeline(discoverer))
+        if hasattr(handler.invoker, "pipeline"):
+            pipeline = handler.invoker.pipeline
+            if cfg.gpu.devices > 1:
+                # one pipeline per GPU; sessions shard across them
+                from .engine.batch import GpuPipeline
+
+                pipeline = [pipeline] + [
+                    GpuPipeline(discoverer, cfg, device=d)
+                    for d in range(1, cfg.gpu.devices)
+                ]
+        else:
+            pipeline = CpuBatchPipeline(discoverer)
         gw = NativeHTTPGateway(
             pipeline, discoverer, cfg,
             sessions=handler.sessions,

@@ -107,7 +107,13 @@ def load_module():
 
 
 class NativeHTTPGateway:
-    """Drop-in serving front end over a GpuPipeline."""
+    """Drop-in serving front end over one or more GpuPipelines.
+
+    With several pipelines (one per GPU device), sessions shard stably
+    across them by ``Mcp-Session-Id`` ("sessions hashed to 1
+    of 8 GPUs; per-GPU session tables"): each batch splits into per-shard
+    sub-batches that run concurrently, so every GPU owns its sessions
+    end-to-end and no request data crosses devices."""
 
     def __init__(
         self,
@@ -121,7 +127,8 @@ class NativeHTTPGateway:
         port: int = 0,
     ) -> None:
         self.config = config or Config.default()
-        self.pipeline = pipeline
+        self.pipelines = list(pipeline) if isinstance(pipeline, (list, tuple)) else [pipeline]
+        self.pipeline = self.pipelines[0]
         self.discoverer = discoverer
         self.sessions = sessions or SessionManager()
         self.tools = tool_builder or MCPToolBuilder()
@@ -130,7 +137,16 @@ class NativeHTTPGateway:
         )
         self.validator = Validator()
         self.start_time = time.time()
-        pipeline.non_toolcall_handler = self._handle_non_toolcall
+        for p in self.pipelines:
+            p.non_toolcall_handler = self._handle_non_toolcall
+        if len(self.pipelines) > 1:
+            from concurrent.futures import ThreadPoolExecutor
+
+            self._shard_pool = ThreadPoolExecutor(
+                max_workers=len(self.pipelines), thread_name_prefix="gshard"
+            )
+        else:
+            self._shard_pool = None
         mod = load_module()
         srv_cfg = self.config.server
         self._fe = mod.Frontend(
@@ -152,6 +168,8 @@ class NativeHTTPGateway:
 
     def stop(self) -> None:
         self._fe.stop()
+        if self._shard_pool is not None:
+            self._shard_pool.shutdown(wait=False)
 
     # ---- hot path: one call per collected batch -----------------------------
 
@@ -174,25 +192,51 @@ class NativeHTTPGateway:
                 continue
             sess.increment_call_count()
             fwd_headers.append(self.headers.filter_headers(headers[i]))
+        timeout = self.config.grpc.request_timeout_s
         if rejected:
             live_idx = [i for i in range(n) if i not in rejected]
             live = [bodies[i] for i in live_idx]
-            out_live = self.pipeline.process_batch(
-                live,
-                headers=fwd_headers,
-                timeout_s=self.config.grpc.request_timeout_s,
-            ) if live else []
+            out_live = self._run_sharded(live, fwd_headers, sids, live_idx, timeout) if live else []
             out: List[bytes] = [b""] * n
             for k, i in enumerate(live_idx):
                 out[i] = out_live[k]
             for i, resp in rejected.items():
                 out[i] = resp
         else:
-            out = self.pipeline.process_batch(
-                bodies, headers=fwd_headers,
-                timeout_s=self.config.grpc.request_timeout_s,
-            )
+            out = self._run_sharded(bodies, fwd_headers, sids, None, timeout)
         return list(zip(out, sids))
+
+    def _run_sharded(self, bodies, fwd_headers, sids, idx_map, timeout):
+        """Split the batch across session shards (one pipeline per GPU)."""
+        if self._shard_pool is None:
+            return self.pipelines[0].process_batch(
+                bodies, headers=fwd_headers, timeout_s=timeout
+            )
+        from ..parallel.dist import shard_for_session
+
+        n_shards = len(self.pipelines)
+        groups: List[List[int]] = [[] for _ in range(n_shards)]
+        for k in range(len(bodies)):
+            i = idx_map[k] if idx_map is not None else k
+            groups[shard_for_session(sids[i], n_shards)].append(k)
+        futs = []
+        for s, g in enumerate(groups):
+            if not g:
+                futs.append(None)
+                continue
+            sub_bodies = [bodies[k] for k in g]
+            sub_headers = [fwd_headers[k] for k in g] if fwd_headers else None
+            futs.append(self._shard_pool.submit(
+                self.pipelines[s].process_batch, sub_bodies,
+                headers=sub_headers, timeout_s=timeout))
+        out: List[bytes] = [b""] * len(bodies)
+        for s, g in enumerate(groups):
+            if futs[s] is None:
+                continue
+            sub_out = futs[s].result()
+            for j, k in enumerate(g):
+                out[k] = sub_out[j]
+        return out
 
     @staticmethod
     def _session_error(body: bytes, message: str) -> bytes:
@@ -264,6 +308,8 @@ class NativeHTTPGateway:
             stats["sessions"] = self.sessions.stats()
             stats["uptimeS"] = time.time() - self.start_time
             stats["engine"] = self.pipeline.engine.stats.snapshot()
+            if len(self.pipelines) > 1:
+                stats["shards"] = [p.engine.stats.snapshot() for p in self.pipelines]
             return 200, json.dumps(stats).encode(), ""
         if method == "GET" and path == "/":
             sess = self.sessions.get_or_create(
