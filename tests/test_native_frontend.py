@@ -241,3 +241,63 @@ def test_session_sharding_across_pipelines():
         assert sum(pipes[0].batches) + sum(pipes[1].batches) == 12
     finally:
         gw.stop()
+
+
+def test_raw_socket_abuse(gateway):
+    """Reactor robustness: split packets, HTTP pipelining, garbage request
+    lines, half-open connections — no hangs, ordered responses."""
+    import socket
+    import time as _t
+
+    gw, port, pipe = gateway
+    body = json.dumps({"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                       "params": {"name": "t", "arguments": {"a": 1}}}).encode()
+    req = (b"POST / HTTP/1.1\r\nHost: x\r\nContent-Type: application/json\r\n"
+           + b"Content-Length: %d\r\n\r\n" % len(body) + body)
+
+    # 1. byte-dribbled request
+    s = socket.create_connection(("127.0.0.1", port), timeout=10)
+    for i in range(0, len(req), 7):
+        s.sendall(req[i:i + 7])
+        _t.sleep(0.001)
+    resp = s.recv(65536)
+    assert b"200 OK" in resp and b'"isError": false' in resp or b"isError" in resp
+    s.close()
+
+    # 2. two pipelined requests in one write -> two responses, in order
+    body2 = json.dumps({"jsonrpc": "2.0", "id": 2, "method": "tools/call",
+                        "params": {"name": "t", "arguments": {"b": 2}}}).encode()
+    req2 = (b"POST / HTTP/1.1\r\nHost: x\r\nContent-Type: application/json\r\n"
+            + b"Content-Length: %d\r\n\r\n" % len(body2) + body2)
+    s = socket.create_connection(("127.0.0.1", port), timeout=10)
+    s.sendall(req + req2)
+    data = b""
+    deadline = _t.time() + 10
+    while data.count(b"HTTP/1.1 200") < 2 and _t.time() < deadline:
+        chunk = s.recv(65536)
+        if not chunk:
+            break
+        data += chunk
+    assert data.count(b"HTTP/1.1 200") == 2, data[:200]
+    first = data.index(b'"id": 1') if b'"id": 1' in data else data.index(b'"id":1')
+    second = data.index(b'"id": 2') if b'"id": 2' in data else data.index(b'"id":2')
+    assert first < second  # pipelined order preserved
+    s.close()
+
+    # 3. garbage request line -> connection dropped, server stays alive
+    s = socket.create_connection(("127.0.0.1", port), timeout=10)
+    s.sendall(b"NONSENSE\r\n\r\n")
+    _t.sleep(0.05)
+    s.close()
+
+    # 4. half-open (headers only, never the body) then abandon
+    s = socket.create_connection(("127.0.0.1", port), timeout=10)
+    s.sendall(b"POST / HTTP/1.1\r\nContent-Length: 50\r\n\r\n")
+    s.close()
+
+    # the gateway still serves
+    status, data, _ = _call(port, json.dumps(
+        {"jsonrpc": "2.0", "id": 9, "method": "tools/call",
+         "params": {"name": "t", "arguments": {}}}))
+    assert status == 200
+    assert json.loads(data)["id"] == 9
