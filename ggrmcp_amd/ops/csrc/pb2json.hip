@@ -917,6 +917,77 @@ DEVN bool map_entry_step(DCtx& c, DFrame& f, uint32_t eend, DFrame* stack,
   const MsgEntry& em = c.t.msgs[c.t.fields[f.cont_field].sub_index];
   const FieldEntry& kf = c.t.fields[em.field_start];
   const FieldEntry& vf = c.t.fields[em.field_start + 1];
+  // fast path: the canonical wire order every real serializer emits —
+  // key (field 1) then value (field 2).  ALL structural checks run before
+  // any byte is emitted, so falling back to the general two-pass walk
+  // never double-emits.
+  if (c.pos < eend) {
+    uint32_t save0 = c.pos;
+    bool committed = false;
+    uint64_t tag;
+    if (!read_varint(c, &tag)) return false;
+    if ((tag >> 3) == 1 && (uint32_t)(tag & 7) == expected_wire(kf)) {
+      uint32_t kp = c.pos;
+      if (!skip_wire(c, (uint32_t)(tag & 7))) return false;
+      if (c.pos < eend) {
+        uint64_t vtag;
+        if (!read_varint(c, &vtag)) return false;
+        if ((vtag >> 3) == 2) {
+          if (vf.kind == K_MESSAGE) {
+            uint64_t v;
+            if (!read_varint(c, &v)) return false;
+            if (c.pos + v > c.len) return dfail(c, E_PARSE);
+            uint32_t vend = c.pos + (uint32_t)v;
+            if (vend == eend) {
+              // commit: emit key, then enter/push the message value
+              uint32_t body_pos = c.pos;
+              if (!emit_map_key(c, kf, kp, true)) return false;
+              c.pos = body_pos;
+              uint8_t pm = FM_BODY;
+              int32_t pidx = vf.sub_index;
+              uint32_t bend = vend;
+              int r = enter_body(c, vf.sub_index, vend, &pm, &pidx, &bend);
+              if (r == 0) return false;
+              if (r == 1) {
+                c.pos = eend;
+                return true;
+              }
+              if (sp >= MAX_RECURSE) return dfail(c, E_LIMIT);
+              f.cont_end = eend;
+              DFrame& nf = stack[sp++];
+              nf.end = bend;
+              nf.prev_number = 0;
+              nf.cont_end = 0;
+              nf.msg_idx = pidx;
+              nf.cont_field = -1;
+              nf.cont_num = 0;
+              nf.mode = pm;
+              nf.first_member = 1;
+              nf.cont_kind = CK_NONE;
+              nf.cont_first = 1;
+              if (!putc_(c, pm == FM_LIST ? '[' : '{')) return false;
+              *pushed = 1;
+              return true;
+            }
+          } else if ((uint32_t)(vtag & 7) == expected_wire(vf)) {
+            // commit: key, value, then skip any trailing entry fields
+            uint32_t vpos = c.pos;
+            if (!emit_map_key(c, kf, kp, true)) return false;
+            c.pos = vpos;
+            if (!emit_scalar_value(c, vf)) return false;
+            while (c.pos < eend) {
+              uint64_t t2;
+              if (!read_varint(c, &t2)) return false;
+              if (!skip_wire(c, (uint32_t)(t2 & 7))) return false;
+            }
+            committed = true;
+          }
+        }
+      }
+    }
+    if (committed) return true;
+    c.pos = save0;
+  }
   bool have_key = false, have_val = false;
   uint32_t key_pos = 0, val_pos = 0;
   while (c.pos < eend) {

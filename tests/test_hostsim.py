@@ -233,3 +233,25 @@ def test_mode2_content_item_wrapping(env):
     assert item["type"] == "text"
     inner = json.loads(item["text"])
     assert inner == {"message": 'say "hi"\n'}
+
+
+def test_map_entry_value_before_key(env):
+    """Out-of-canonical-order map entry wire (value field 2 before key
+    field 1): no real serializer emits this, but protobuf allows it — the
+    decoder's general two-pass path must handle it (the fast path only
+    commits on canonical order)."""
+    e, c, p, _ = env
+    # complex.Document: metadata = map<string,string> field 4
+    entry = bytes([0x12, 0x01, ord("v"), 0x0A, 0x01, ord("k")])  # value, key
+    wire = bytes([0x22, len(entry)]) + entry  # field 4, LEN
+    idx = e.tables.msg_index["complex.Document"]
+    dec, outs = e.decode_batch([wire], [idx], mode=1)
+    assert dec[0]["status"] == 0, dec[0]
+    assert json.loads(outs[0]) == {"metadata": {"k": "v"}}
+
+    # key-only entry (missing value -> default)
+    entry2 = bytes([0x0A, 0x01, ord("x")])
+    wire2 = bytes([0x22, len(entry2)]) + entry2
+    dec, outs = e.decode_batch([wire2], [idx], mode=1)
+    assert dec[0]["status"] == 0
+    assert json.loads(outs[0]) == {"metadata": {"x": ""}}
