@@ -301,3 +301,55 @@ def test_raw_socket_abuse(gateway):
          "params": {"name": "t", "arguments": {}}}))
     assert status == 200
     assert json.loads(data)["id"] == 9
+
+
+def test_connection_churn_mid_batch(gateway):
+    """Clients that disconnect before their response arrives must not
+    crash the reactor or corrupt other connections' responses."""
+    import socket
+
+    gw, port, pipe = gateway
+    body = json.dumps({"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                       "params": {"name": "t", "arguments": {"x": 1}}}).encode()
+    req = (b"POST / HTTP/1.1\r\nHost: x\r\nContent-Type: application/json\r\n"
+           + b"Content-Length: %d\r\n\r\n" % len(body) + body)
+    # fire-and-abandon 40 requests
+    for _ in range(40):
+        s = socket.create_connection(("127.0.0.1", port), timeout=5)
+        s.sendall(req)
+        s.close()  # gone before the batch completes
+    # a healthy client interleaved with the churn still gets clean answers
+    errs = []
+
+    def worker(t):
+        try:
+            conn = http.client.HTTPConnection("127.0.0.1", port, timeout=10)
+            for i in range(10):
+                b = json.dumps({"jsonrpc": "2.0", "id": f"{t}-{i}",
+                                "method": "tools/call",
+                                "params": {"name": "t", "arguments": {"t": t, "i": i}}})
+                conn.request("POST", "/", body=b,
+                             headers={"Content-Type": "application/json"})
+                r = conn.getresponse()
+                resp = json.loads(r.read())
+                assert resp["id"] == f"{t}-{i}", resp
+                inner = json.loads(resp["result"]["content"][0]["text"])
+                assert inner["echo"] == {"t": t, "i": i}
+            conn.close()
+        except Exception as e:  # pragma: no cover
+            errs.append(repr(e))
+
+    ts = [threading.Thread(target=worker, args=(t,)) for t in range(8)]
+    for t in ts:
+        t.start()
+    # more churn while the workers run
+    for _ in range(40):
+        try:
+            s = socket.create_connection(("127.0.0.1", port), timeout=5)
+            s.sendall(req[: len(req) // 2])
+            s.close()
+        except OSError:
+            pass
+    for t in ts:
+        t.join()
+    assert not errs, errs[:3]
