@@ -259,3 +259,41 @@ def test_fuzz_envelope_statuses(rid, method, tool, args, jsonrpc, drop_id):
     else:
         # the raw JSON id token is captured for the response envelope
         assert int(enc[0]["id_len"]) > 0, enc[0]
+
+
+# ---- hostile input: arbitrary bytes must fail CLEANLY -----------------------
+
+@settings(max_examples=300, deadline=None)
+@given(data=st.binary(max_size=300))
+def test_fuzz_hostile_envelope_bytes(data):
+    """Random bytes through envelope mode: any E_* status is fine, crashes
+    and out-of-range statuses are not."""
+    enc, pbs = _engine.encode_batch([data], mode=0)
+    assert 0 <= int(enc[0]["status"]) <= 8
+
+
+@settings(max_examples=200, deadline=None)
+@given(prefix=st.binary(max_size=120), cut=st.integers(min_value=0, max_value=400))
+def test_fuzz_truncated_valid_envelope(prefix, cut):
+    """A valid envelope truncated at any byte, with optional garbage glued
+    on front, must fail cleanly (or parse if the cut lands at the end)."""
+    body = json.dumps({"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                       "params": {"name": "hello_helloservice_sayhello",
+                                  "arguments": {"name": "abc"}}}).encode()
+    data = body[: min(cut, len(body))]
+    enc, _ = _engine.encode_batch([data, prefix + data], mode=0)
+    assert 0 <= int(enc[0]["status"]) <= 8
+    assert 0 <= int(enc[1]["status"]) <= 8
+
+
+@settings(max_examples=200, deadline=None)
+@given(data=st.binary(max_size=200))
+def test_fuzz_hostile_wire_decode(data):
+    """Random bytes as protobuf wire through the decoder: clean status,
+    and successful decodes must emit valid JSON."""
+    idx = _engine.tables.msg_index["bench.Wide64"]
+    dec, outs = _engine.decode_batch([data], [idx], mode=1)
+    s = int(dec[0]["status"])
+    assert 0 <= s <= 8
+    if s == 0:
+        json.loads(outs[0])  # must be well-formed
