@@ -36,6 +36,7 @@
 #include <sys/socket.h>
 #include <unistd.h>
 
+#include <algorithm>
 #include <atomic>
 #include <chrono>
 #include <condition_variable>
@@ -608,6 +609,129 @@ class Frontend {
   std::deque<std::vector<PendingReq>> batches_;
 };
 
+// ---------------------------------------------------------------------------
+// load generator — C++ closed-loop HTTP client for measuring the gateway
+// without a Python client in the way.  N sessions across T threads, each
+// session a keep-alive connection issuing sequential POSTs; returns
+// (total_requests, elapsed_s, [p50_us, p90_us, p99_us], errors).
+// ---------------------------------------------------------------------------
+
+static py::tuple bench_client(const std::string& host, int port, int sessions,
+                              int requests, const std::string& body,
+                              int threads) {
+  if (threads < 1) threads = 1;
+  if (sessions < threads) threads = sessions;
+  std::string req_head =
+      "POST / HTTP/1.1\r\nHost: b\r\nContent-Type: application/json\r\n";
+  std::atomic<long> errors{0};
+  std::vector<std::vector<uint32_t>> lat(threads);
+  auto run = [&](int t) {
+    int per = sessions / threads + (t < sessions % threads ? 1 : 0);
+    std::vector<int> fds;
+    for (int s = 0; s < per; ++s) {
+      int fd = socket(AF_INET, SOCK_STREAM, 0);
+      sockaddr_in addr{};
+      addr.sin_family = AF_INET;
+      addr.sin_port = htons((uint16_t)port);
+      inet_pton(AF_INET, host.c_str(), &addr.sin_addr);
+      if (connect(fd, (sockaddr*)&addr, sizeof(addr)) != 0) {
+        close(fd);
+        errors++;
+        continue;
+      }
+      int one = 1;
+      setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+      fds.push_back(fd);
+    }
+    // closed-loop: each connection keeps exactly one request in flight;
+    // drive them round-robin with poll-free sequential turns per epoch
+    std::vector<std::string> reqs(fds.size());
+    for (size_t s = 0; s < fds.size(); ++s) {
+      std::string sid = "bench-" + std::to_string(t) + "-" + std::to_string(s);
+      reqs[s] = req_head + "Mcp-Session-Id: " + sid +
+                "\r\nContent-Length: " + std::to_string(body.size()) +
+                "\r\n\r\n" + body;
+    }
+    std::vector<Clock::time_point> t0(fds.size());
+    lat[t].reserve(fds.size() * requests);
+    // epoll-driven: send all, then read completions and resend
+    int ep = epoll_create1(0);
+    std::vector<int> remaining(fds.size(), requests);
+    std::vector<std::string> bufs(fds.size());
+    for (size_t s = 0; s < fds.size(); ++s) {
+      epoll_event ev{};
+      ev.events = EPOLLIN;
+      ev.data.u64 = s;
+      epoll_ctl(ep, EPOLL_CTL_ADD, fds[s], &ev);
+      t0[s] = Clock::now();
+      if (send(fds[s], reqs[s].data(), reqs[s].size(), MSG_NOSIGNAL) < 0) errors++;
+    }
+    size_t live = fds.size();
+    std::vector<epoll_event> events(64);
+    auto deadline = Clock::now() + std::chrono::seconds(120);
+    while (live > 0 && Clock::now() < deadline) {
+      int n = epoll_wait(ep, events.data(), (int)events.size(), 1000);
+      for (int i = 0; i < n; ++i) {
+        size_t s = events[i].data.u64;
+        char rb[1 << 16];
+        ssize_t r = recv(fds[s], rb, sizeof(rb), 0);
+        if (r <= 0) {
+          if (r < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) continue;
+          errors++;
+          epoll_ctl(ep, EPOLL_CTL_DEL, fds[s], nullptr);
+          close(fds[s]);
+          live--;
+          continue;
+        }
+        bufs[s].append(rb, r);
+        // complete iff we can see the whole response (headers+body)
+        size_t he = bufs[s].find("\r\n\r\n");
+        if (he == std::string::npos) continue;
+        size_t cl = 0;
+        size_t p = bufs[s].find("Content-Length:");
+        if (p != std::string::npos && p < he)
+          cl = (size_t)strtoull(bufs[s].c_str() + p + 15, nullptr, 10);
+        if (bufs[s].size() < he + 4 + cl) continue;
+        bufs[s].erase(0, he + 4 + cl);
+        lat[t].push_back((uint32_t)std::chrono::duration_cast<std::chrono::microseconds>(
+            Clock::now() - t0[s]).count());
+        if (--remaining[s] <= 0) {
+          epoll_ctl(ep, EPOLL_CTL_DEL, fds[s], nullptr);
+          close(fds[s]);
+          live--;
+          continue;
+        }
+        t0[s] = Clock::now();
+        if (send(fds[s], reqs[s].data(), reqs[s].size(), MSG_NOSIGNAL) < 0) {
+          errors++;
+          epoll_ctl(ep, EPOLL_CTL_DEL, fds[s], nullptr);
+          close(fds[s]);
+          live--;
+        }
+      }
+    }
+    close(ep);
+  };
+  auto t_start = Clock::now();
+  std::vector<std::thread> ts;
+  {
+    py::gil_scoped_release rel;
+    for (int t = 0; t < threads; ++t) ts.emplace_back(run, t);
+    for (auto& th : ts) th.join();
+  }
+  double elapsed = std::chrono::duration<double>(Clock::now() - t_start).count();
+  std::vector<uint32_t> all;
+  for (auto& v : lat) all.insert(all.end(), v.begin(), v.end());
+  std::sort(all.begin(), all.end());
+  py::list pct;
+  if (!all.empty()) {
+    pct.append(all[all.size() / 2]);
+    pct.append(all[(size_t)(all.size() * 0.9)]);
+    pct.append(all[(size_t)(all.size() * 0.99)]);
+  }
+  return py::make_tuple((long)all.size(), elapsed, pct, errors.load());
+}
+
 PYBIND11_MODULE(_frontend, m) {
   m.doc() = "native HTTP/1.1 batch ingestion front end for the MCP gateway";
   py::class_<Frontend>(m, "Frontend")
@@ -621,4 +745,7 @@ PYBIND11_MODULE(_frontend, m) {
       .def("start", &Frontend::start)
       .def("stop", &Frontend::stop)
       .def_property_readonly("port", &Frontend::port);
+  m.def("bench_client", &bench_client, py::arg("host"), py::arg("port"),
+        py::arg("sessions"), py::arg("requests"), py::arg("body"),
+        py::arg("threads") = 8);
 }

@@ -79,6 +79,8 @@ async def main() -> None:
                     help="serve through the C++ reactor + GpuPipeline instead of asyncio")
     ap.add_argument("--batch-window-us", type=int, default=200)
     ap.add_argument("--procs", type=int, default=1, help="client processes")
+    ap.add_argument("--cxx-client", action="store_true",
+                    help="drive load with the C++ epoll client (no Python client ceiling)")
     args = ap.parse_args()
 
     srv, target = serve_native("127.0.0.1:0")
@@ -142,6 +144,36 @@ async def main() -> None:
                           middlewares=default_middleware(cfg.server, recorder),
                           port=0)
         await http.start()
+
+    if args.cxx_client:
+        from ggrmcp_amd.server.native_http import load_module as _lm
+
+        femod = _lm()
+        body = json.dumps({
+            "jsonrpc": "2.0", "id": 1, "method": "tools/call",
+            "params": {"name": "hello_helloservice_sayhello",
+                       "arguments": hello_payload(random.Random(0), args.payload_bytes)},
+        })
+        femod.bench_client("127.0.0.1", http.port, min(args.sessions, 64), 3, body, 4)
+        total, elapsed, pct, errs = femod.bench_client(
+            "127.0.0.1", http.port, args.sessions, args.requests, body, 8)
+        result = {
+            "mode": "http-serving-native-cxxload",
+            "sessions": args.sessions,
+            "requests_per_session": args.requests,
+            "payload_bytes": args.payload_bytes,
+            "total_requests": total,
+            "req_per_s": round(total / elapsed, 1),
+            "p50_ms": round(pct[0] / 1e3, 3) if pct else None,
+            "p90_ms": round(pct[1] / 1e3, 3) if pct else None,
+            "p99_ms": round(pct[2] / 1e3, 3) if pct else None,
+            "client_errors": errs,
+        }
+        print(json.dumps(result), flush=True)
+        await http.stop(1.0)
+        discoverer.close()
+        srv.stop()
+        return
 
     lat: list = []
     # warmup
@@ -210,3 +242,7 @@ if __name__ == "__main__":
         asyncio.run(client_only(sys.argv[i + 1 : i + 6]))
     else:
         asyncio.run(main())
+
+
+# native C++ load generator entry (bypasses Python client limits):
+#   python tools/http_bench.py --cxx-client --sessions 2048 --requests 50
