@@ -501,9 +501,21 @@ DEV bool skip_wire(DCtx& c, uint32_t wt) {
 }
 
 DEV const FieldEntry* find_field(DCtx& c, const MsgEntry& m, uint32_t number) {
-  for (int i = 0; i < m.field_count; ++i) {
-    const FieldEntry& f = c.t.fields[m.field_start + i];
+  // fields are sorted by number (tables.py).  Densely-numbered messages
+  // (number == position+1, the overwhelmingly common schema shape) hit in
+  // O(1); otherwise binary search.  The linear scan this replaces cost
+  // ~300 cycles per field on 64-field messages (~150 ns/field measured).
+  if (number >= 1 && number <= (uint32_t)m.field_count) {
+    const FieldEntry& f = c.t.fields[m.field_start + number - 1];
     if (f.number == number) return &f;
+  }
+  int lo = 0, hi = m.field_count - 1;
+  while (lo <= hi) {
+    int mid = (lo + hi) >> 1;
+    const FieldEntry& f = c.t.fields[m.field_start + mid];
+    if (f.number == number) return &f;
+    if (f.number < number) lo = mid + 1;
+    else hi = mid - 1;
   }
   return nullptr;
 }

@@ -8,6 +8,7 @@ from ggrmcp_amd.utils.protobuild import FileBuilder
 from ggrmcp_amd.descriptors.loader import build_pool, extract_method_infos
 from ggrmcp_amd.engine.batch import GpuEngine
 from ggrmcp_amd.engine.cpu_ref import CpuTranscoder
+from ggrmcp_amd.utils.synthetic import synthetic_fdp
 
 fb = FileBuilder("p/probe.proto", "p")
 fb.message("Str").field("s", 1, "string").done()
@@ -17,10 +18,21 @@ fb.message("Ints").field("v", 1, "int64", repeated=True).done()
 (fb.service("S").method("M", "Str", "Str").method("M2", "Map", "Map")
    .method("M3", "Pack", "Pack").method("M4", "Ints", "Ints").done())
 fdp = fb.build()
-pool = build_pool([fdp])
-infos = {m.tool_name(): m for m in extract_method_infos([fdp], pool, compat_names=False)}
+fdps = [fdp, synthetic_fdp()]
+pool = build_pool(fdps)
+infos = {m.tool_name(): m for m in extract_method_infos(fdps, pool, compat_names=False)}
 eng = GpuEngine(infos, device=0)
 cpu = CpuTranscoder()
+
+# the real wide64 payload + components thereof
+import json as _json
+from ggrmcp_amd.utils.synthetic import synthetic_fdp, wide_payload
+import random as _random
+
+_rngW = _random.Random(7)
+_wp = wide_payload(_rngW, target_bytes=64 * 1024)
+_scalars = {k: v for k, v in _wp.items() if k.startswith("f")}
+_attrs_only = {"attrs": _wp["attrs"]}
 
 CASES = {
     "str64k": ("p.Str", {"s": "x" * 65536}),
@@ -29,6 +41,9 @@ CASES = {
     "map1024x64": ("p.Map", {"m": {f"key{i:04d}": "v" * 56 for i in range(1024)}}),
     "pack8k_doubles": ("p.Pack", {"d": [i * 1.5 for i in range(8192)]}),
     "ints4k": ("p.Ints", {"v": [str(i * 7) for i in range(4096)]}),
+    "wide_actual": ("bench.Wide64", _wp),
+    "wide_scalars": ("bench.Wide64", _scalars),
+    "wide_attrs": ("bench.Wide64", _attrs_only),
 }
 
 BATCH = 128
