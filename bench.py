@@ -140,15 +140,23 @@ def main() -> None:
     import torch
 
     use_gpu = args.config != "cpu" and torch.cuda.is_available()
+    # device index: clamp by the visible device count so a world-2 rehearsal
+    # on a 1-GPU box maps both ranks onto cuda:0 (no-op on a full node)
+    dev = (local_rank % max(1, torch.cuda.device_count())) if use_gpu else 0
     dist = None
+    dist_backend = None
     if world > 1:
         import torch.distributed as dist_mod
 
         dist = dist_mod
-        backend = "nccl" if use_gpu else "gloo"
+        # GGRMCP_DIST_BACKEND=gloo lets the multi-rank GPU path be smoked on
+        # a single GPU (RCCL refuses two ranks on one device)
+        dist_backend = os.environ.get("GGRMCP_DIST_BACKEND") or (
+            "nccl" if use_gpu else "gloo"
+        )
         if use_gpu:
-            torch.cuda.set_device(local_rank)
-        dist.init_process_group(backend=backend)
+            torch.cuda.set_device(dev)
+        dist.init_process_group(backend=dist_backend)
 
     # config 5 runs N distinct-package backends; both stream and multi can
     # use the native backend (h2grpc stream_echo route) with the batched
@@ -193,7 +201,7 @@ def main() -> None:
             from ggrmcp_amd.parallel.dist import ShardGroup
 
             shard_group = ShardGroup.attach(
-                dist, device=local_rank if use_gpu else None
+                dist, device=dev if (use_gpu and dist_backend == "nccl") else None
             )
         if native_backend or n_backends > 1:
             # no-reflection path: rank 0 builds the descriptor blob per
@@ -245,7 +253,7 @@ def main() -> None:
         if use_gpu:
             from ggrmcp_amd.engine.batch import GpuPipeline
 
-            pipeline = GpuPipeline(discoverer, cfg, device=local_rank,
+            pipeline = GpuPipeline(discoverer, cfg, device=dev,
                                    invoke_workers=args.invoke_workers,
                                    wire_clients=wire_clients)
 
