@@ -568,7 +568,13 @@ class GpuPipeline:
             [c._cli for c in self.wire_clients],
             timeout,
         )
-        st.invoke_ns += time.perf_counter_ns() - t0
+        total_ns = time.perf_counter_ns() - t0
+        # per-stage split from the C++ span (SURVEY §5 /metrics deliverable);
+        # Python call overhead stays booked under invoke so stages sum to total
+        enc_ms, _inv_ms, dec_ms = engine._eng.last_stage_ms()
+        st.encode_ns += int(enc_ms * 1e6)
+        st.decode_ns += int(dec_ms * 1e6)
+        st.invoke_ns += max(0, total_ns - int(enc_ms * 1e6) - int(dec_ms * 1e6))
         enc = np.frombuffer(enc_raw.tobytes(), dtype=SLOT_DTYPE)
         dec = np.frombuffer(dec_raw.tobytes(), dtype=DECODE_DTYPE)
         n = len(bodies)

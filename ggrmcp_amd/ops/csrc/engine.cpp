@@ -25,6 +25,7 @@
 #include <pybind11/numpy.h>
 #include <pybind11/pybind11.h>
 
+#include <chrono>
 #include <cstring>
 #include <stdexcept>
 #include <string>
@@ -388,7 +389,14 @@ class Engine {
       }
     }
     Limits lim{max_depth, max_string, max_args, (uint32_t)enforce};
+    // per-stage wall times for /metrics (parse/validate/encode
+    // live in run_encode; decode covers pb->JSON + envelope).  Read back
+    // via last_stage_ms() — safe because callers hold the engine lock.
+    auto t_enc0 = std::chrono::steady_clock::now();
     py::tuple enc_out = run_encode(n, acc, pacc, false, lim, 0);
+    last_enc_ms_ = std::chrono::duration<double, std::milli>(
+                       std::chrono::steady_clock::now() - t_enc0)
+                       .count();
 
     // route OK unary slots per backend
     SlotResult* rs = (SlotResult*)h_results_.p;
@@ -427,12 +435,16 @@ class Engine {
     // wait... simple path: sequential per backend; multi-backend batches
     // overlap because submission is async and the wait is per-batch)
     std::vector<std::vector<std::tuple<int, std::string, std::string>>> res_be(nb);
+    auto t_inv0 = std::chrono::steady_clock::now();
     {
       py::gil_scoped_release rel;
       for (size_t b = 0; b < nb; ++b)
         if (!per_be[b].empty())
           res_be[b] = clients[b]->invoke_raw(per_be[b], timeout_s);
     }
+    auto t_dec0 = std::chrono::steady_clock::now();
+    last_inv_ms_ =
+        std::chrono::duration<double, std::milli>(t_dec0 - t_inv0).count();
 
     // stage responses for decode
     uint32_t* resp_off = (uint32_t*)h_off_.p;
@@ -496,9 +508,18 @@ class Engine {
     if (sacc > d_scratch_.n) throw std::runtime_error("scratch cap");
     if (facc > d_final_.n) throw std::runtime_error("final cap");
     py::tuple dec_out = run_decode(n, racc, facc, true, 0);
+    last_dec_ms_ = std::chrono::duration<double, std::milli>(
+                       std::chrono::steady_clock::now() - t_dec0)
+                       .count();
 
     return py::make_tuple(enc_out[0], dec_out[0], dec_out[1], stream_pbs,
                           rpc_errors);
+  }
+
+  // (encode_ms, invoke_ms, decode_ms) of the LAST process_span call;
+  // callers serialize spans per engine, so no further synchronization
+  py::tuple last_stage_ms() const {
+    return py::make_tuple(last_enc_ms_, last_inv_ms_, last_dec_ms_);
   }
 
   int device() const { return device_; }
@@ -649,6 +670,7 @@ class Engine {
 
   int device_;
   int max_batch_;
+  double last_enc_ms_ = 0.0, last_inv_ms_ = 0.0, last_dec_ms_ = 0.0;
   int last_batch_n_ = -1;
   bool compact_used_ = false;
   size_t compact_bytes_ = 0;
@@ -718,6 +740,7 @@ PYBIND11_MODULE(_jsonproto, m) {
            py::arg("enforce") = 1)
       .def("decode_list", &Engine::decode_list, py::arg("items"),
            py::arg("msg_idx"), py::arg("skip") = py::none(), py::arg("mode") = 0)
+      .def("last_stage_ms", &Engine::last_stage_ms)
       .def_property_readonly("device", &Engine::device)
       .def_property_readonly("max_batch", &Engine::max_batch);
 }
