@@ -64,6 +64,12 @@ def test_process_batch_roundtrip(pipeline_env):
         inner = json.loads(resp["result"]["content"][0]["text"])
         assert inner == {"message": f"Hello, u{i}!"}
     assert pipeline.engine.stats.gpu_ok >= 32
+    # per-stage split (SURVEY §5 /metrics deliverable): the native span
+    # executor must attribute encode/decode time, not book it all as invoke
+    snap = pipeline.engine.stats.snapshot()
+    assert snap["encodeMs"] > 0
+    assert snap["decodeMs"] > 0
+    assert snap["invokeMs"] > 0
 
 
 def test_process_batch_mixed_errors(pipeline_env):
