@@ -97,6 +97,10 @@ class EngineStats:
     encode_ns: int = 0
     decode_ns: int = 0
     invoke_ns: int = 0
+    # device-only portions (copies+kernel+sync) of encode/decode — the
+    # difference vs encode_ns/decode_ns is GIL/thread contention
+    encode_gpu_ns: int = 0
+    decode_gpu_ns: int = 0
 
     def snapshot(self) -> Dict[str, Any]:
         return {
@@ -108,6 +112,8 @@ class EngineStats:
             "encodeMs": self.encode_ns / 1e6,
             "decodeMs": self.decode_ns / 1e6,
             "invokeMs": self.invoke_ns / 1e6,
+            "encodeGpuMs": self.encode_gpu_ns / 1e6,
+            "decodeGpuMs": self.decode_gpu_ns / 1e6,
         }
 
 
@@ -575,6 +581,9 @@ class GpuPipeline:
         st.encode_ns += int(enc_ms * 1e6)
         st.decode_ns += int(dec_ms * 1e6)
         st.invoke_ns += max(0, total_ns - int(enc_ms * 1e6) - int(dec_ms * 1e6))
+        enc_gpu_ms, dec_gpu_ms = engine._eng.last_gpu_ms()
+        st.encode_gpu_ns += int(enc_gpu_ms * 1e6)
+        st.decode_gpu_ns += int(dec_gpu_ms * 1e6)
         enc = np.frombuffer(enc_raw.tobytes(), dtype=SLOT_DTYPE)
         dec = np.frombuffer(dec_raw.tobytes(), dtype=DECODE_DTYPE)
         n = len(bodies)

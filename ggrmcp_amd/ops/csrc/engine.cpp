@@ -522,6 +522,13 @@ class Engine {
     return py::make_tuple(last_enc_ms_, last_inv_ms_, last_dec_ms_);
   }
 
+  // device-only spans (copies+kernel+sync, measured inside the released-GIL
+  // region) of the last run_encode/run_decode — the wall split above minus
+  // these is GIL/thread contention
+  py::tuple last_gpu_ms() const {
+    return py::make_tuple(last_enc_gpu_ms_, last_dec_gpu_ms_);
+  }
+
   int device() const { return device_; }
   int max_batch() const { return max_batch_; }
 
@@ -540,6 +547,7 @@ class Engine {
     uint32_t* h_off = (uint32_t*)h_off_.p;
     {
       py::gil_scoped_release rel;
+      auto g0 = std::chrono::steady_clock::now();
       HIP_CHECK(hipSetDevice(device_));
       HIP_CHECK(hipMemcpyAsync(d_in_.p, h_in_.p, in_bytes,
                                hipMemcpyHostToDevice, stream_));
@@ -568,6 +576,11 @@ class Engine {
       HIP_CHECK(hipMemcpyAsync(h_pb_.p, d_pb_.p, pb_bytes,
                                hipMemcpyDeviceToHost, stream_));
       HIP_CHECK(hipStreamSynchronize(stream_));
+      // pure device span (copies+kernel+sync, no GIL-reacquire wait):
+      // separates real GPU time from thread contention in the wall split
+      last_enc_gpu_ms_ = std::chrono::duration<double, std::milli>(
+                             std::chrono::steady_clock::now() - g0)
+                             .count();
     }
     last_batch_n_ = n;
     py::array_t<uint8_t> results((py::ssize_t)(n * sizeof(SlotResult)));
@@ -584,6 +597,7 @@ class Engine {
     int32_t* h_aux = (int32_t*)h_aux_.p;
     {
       py::gil_scoped_release rel;
+      auto g0 = std::chrono::steady_clock::now();
       HIP_CHECK(hipSetDevice(device_));
       HIP_CHECK(hipMemcpyAsync(d_resp_.p, h_resp_.p, resp_bytes,
                                hipMemcpyHostToDevice, stream_));
@@ -649,6 +663,9 @@ class Engine {
         HIP_CHECK(hipStreamSynchronize(stream_));
         compact_used_ = false;
       }
+      last_dec_gpu_ms_ = std::chrono::duration<double, std::milli>(
+                             std::chrono::steady_clock::now() - g0)
+                             .count();
     }
     py::array_t<uint8_t> results((py::ssize_t)(n * sizeof(DecodeResult)));
     std::memcpy(results.mutable_data(), h_dec_results_.p,
@@ -671,6 +688,7 @@ class Engine {
   int device_;
   int max_batch_;
   double last_enc_ms_ = 0.0, last_inv_ms_ = 0.0, last_dec_ms_ = 0.0;
+  double last_enc_gpu_ms_ = 0.0, last_dec_gpu_ms_ = 0.0;
   int last_batch_n_ = -1;
   bool compact_used_ = false;
   size_t compact_bytes_ = 0;
@@ -741,6 +759,7 @@ PYBIND11_MODULE(_jsonproto, m) {
       .def("decode_list", &Engine::decode_list, py::arg("items"),
            py::arg("msg_idx"), py::arg("skip") = py::none(), py::arg("mode") = 0)
       .def("last_stage_ms", &Engine::last_stage_ms)
+      .def("last_gpu_ms", &Engine::last_gpu_ms)
       .def_property_readonly("device", &Engine::device)
       .def_property_readonly("max_batch", &Engine::max_batch);
 }
