@@ -127,7 +127,11 @@ class GPUConfig:
     devices: int = 1  # engines (== visible GPUs) for DP session sharding
     batch_window_us: int = 200  # adaptive batch collection window
     max_batch: int = 4096  # requests per GPU batch
-    streams: int = 4  # HIP streams per engine (copy/compute overlap)
+    # Engine instances (each with its own HIP stream + arenas) per device.
+    # 2 measured best with the native span executor: overlap pays, but HIP
+    # runtime contention grows superlinearly with instance count
+    # (profiles/streams_sweep: 1->207k, 2->245k, 4->233k req/s same box)
+    streams: int = 2
     max_request_bytes: int = 1024 * 1024  # per-request staging cap
     pinned_pool_bytes: int = 256 * 1024 * 1024  # pinned host staging pool
     device_pool_bytes: int = 1024 * 1024 * 1024  # HBM arena per engine
