@@ -79,6 +79,8 @@ async def main() -> None:
                     help="serve through the C++ reactor + GpuPipeline instead of asyncio")
     ap.add_argument("--batch-window-us", type=int, default=200)
     ap.add_argument("--procs", type=int, default=1, help="client processes")
+    ap.add_argument("--streams", type=int, default=0,
+                    help="override cfg.gpu.streams (0 = config default)")
     ap.add_argument("--cxx-client", action="store_true",
                     help="drive load with the C++ epoll client (no Python client ceiling)")
     args = ap.parse_args()
@@ -89,6 +91,8 @@ async def main() -> None:
     cfg.grpc.host, cfg.grpc.port = host, int(port)
     cfg.gpu.enabled = not args.no_gpu
     cfg.gpu.batch_window_us = args.batch_window_us
+    if args.streams > 0:
+        cfg.gpu.streams = args.streams
     cfg.server.rate_limit_rps = 10_000_000  # measuring the engine, not the limiter
     cfg.server.rate_limit_burst = 10_000_000
 
