@@ -184,6 +184,11 @@ def main() -> None:
             # flight is almost free GPU-side (profiles/wide64_kernels.txt)
             cfg.gpu.pinned_pool_bytes = 2 * 1024 * 1024 * 1024
             cfg.gpu.device_pool_bytes = 6 * 1024 * 1024 * 1024
+            if args.streams <= 0:
+                # 128 MB/step of staging wants more copy overlap than the
+                # small-payload default (profiles/wide64_streams.log:
+                # 4 engines 30-34k vs 2 engines 27-29k req/s)
+                cfg.gpu.streams = 4
         if args.config == "stream" and args.batch == 1024:
             args.batch = 64  # 64 streams x 4096 msgs per step
         if args.config == "multi" and args.batch == 1024:
