@@ -140,6 +140,16 @@ class Engine {
     // wave's scratch and, once raised, the per-queue scratch reservation
     // capped how many HIP queues could dispatch concurrently.
     HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
+    // Blocking-sync event: hipStreamSynchronize busy-spins; with several
+    // engine instances per device that burns cores the HTTP reactors and
+    // nghttp2 clients need.  hipEventSynchronize on a hipEventBlockingSync
+    // event parks the thread instead, at equal measured throughput.
+    // (Multi-instance device-span inflation — 9.3/25/69 ms for the same
+    // work at 1/2/4 instances — is unchanged by this, so that cost is in
+    // the submission path / device-side interleaving, not the sync wait;
+    // see profiles/blocking_sync.log.)
+    HIP_CHECK(hipEventCreateWithFlags(
+        &sync_ev_, hipEventBlockingSync | hipEventDisableTiming));
 
     upload_blob(msg_table, d_msgs_);
     upload_blob(field_table, d_fields_);
@@ -187,7 +197,14 @@ class Engine {
 
   ~Engine() {
     (void)hipStreamSynchronize(stream_);
+    (void)hipEventDestroy(sync_ev_);
     (void)hipStreamDestroy(stream_);
+  }
+
+  // park-the-thread stream wait (see sync_ev_ above)
+  void sync_stream() {
+    HIP_CHECK(hipEventRecord(sync_ev_, stream_));
+    HIP_CHECK(hipEventSynchronize(sync_ev_));
   }
 
   // ---- encode: JSON(-RPC) -> protobuf ------------------------------------
@@ -575,7 +592,7 @@ class Engine {
                                stream_));
       HIP_CHECK(hipMemcpyAsync(h_pb_.p, d_pb_.p, pb_bytes,
                                hipMemcpyDeviceToHost, stream_));
-      HIP_CHECK(hipStreamSynchronize(stream_));
+      sync_stream();
       // pure device span (copies+kernel+sync, no GIL-reacquire wait):
       // separates real GPU time from thread contention in the wall split
       last_enc_gpu_ms_ = std::chrono::duration<double, std::milli>(
@@ -630,7 +647,7 @@ class Engine {
       HIP_CHECK(hipMemcpyAsync(h_dec_results_.p, d_dec_results_.p,
                                n * sizeof(DecodeResult), hipMemcpyDeviceToHost,
                                stream_));
-      HIP_CHECK(hipStreamSynchronize(stream_));
+      sync_stream();
       // Gather used bytes into a tight buffer (reusing d_pb_/h_pb_) so the
       // D2H copy is sum(out_len), not the arena's worst-case capacity.
       DecodeResult* rs = (DecodeResult*)h_dec_results_.p;
@@ -653,14 +670,14 @@ class Engine {
         if (acc)
           HIP_CHECK(hipMemcpyAsync(h_pb_.p, d_pb_.p, acc,
                                    hipMemcpyDeviceToHost, stream_));
-        HIP_CHECK(hipStreamSynchronize(stream_));
+        sync_stream();
         for (int i = 0; i < n; ++i) rs[i].out_off = tight[i];
         compact_bytes_ = acc;
         compact_used_ = true;
       } else {
         HIP_CHECK(hipMemcpyAsync(h_final_.p, d_final_.p, final_bytes,
                                  hipMemcpyDeviceToHost, stream_));
-        HIP_CHECK(hipStreamSynchronize(stream_));
+        sync_stream();
         compact_used_ = false;
       }
       last_dec_gpu_ms_ = std::chrono::duration<double, std::milli>(
@@ -693,6 +710,7 @@ class Engine {
   bool compact_used_ = false;
   size_t compact_bytes_ = 0;
   hipStream_t stream_;
+  hipEvent_t sync_ev_ = nullptr;
   Tables tables_{};
   DeviceBuf d_msgs_, d_fields_, d_enums_, d_enum_vals_, d_tools_, d_names_;
   DeviceBuf d_in_, d_pb_, d_resp_, d_scratch_, d_final_;
