@@ -172,14 +172,12 @@ class Engine {
     d_scratch_.alloc(cap_scratch);
     d_final_.alloc(cap_final);
     size_t offs = (size_t)(max_batch + 1) * sizeof(uint32_t);
-    d_in_off_.alloc(offs);
-    d_pb_off_.alloc(offs);
-    d_resp_off_.alloc(offs);
-    d_scratch_off_.alloc(offs);
-    d_final_off_.alloc(offs);
-    d_msg_idx_.alloc((size_t)max_batch * sizeof(int32_t));
+    // offsets + aux travel as ONE H2D copy each (they are contiguous in
+    // h_off_/h_aux_): fewer HIP API calls per chunk — the submission path
+    // is the measured multi-instance contention point
+    d_off3_.alloc(offs * 3);
+    d_aux2_.alloc((size_t)max_batch * sizeof(int32_t) * 2);
     d_tight_off_.alloc(offs);
-    d_skip_.alloc((size_t)max_batch * sizeof(int32_t));
     d_results_.alloc((size_t)max_batch * sizeof(SlotResult));
     d_dec_results_.alloc((size_t)max_batch * sizeof(DecodeResult));
     d_id_slots_.alloc((size_t)max_batch * ID_SLOT_BYTES);
@@ -568,22 +566,20 @@ class Engine {
       HIP_CHECK(hipSetDevice(device_));
       HIP_CHECK(hipMemcpyAsync(d_in_.p, h_in_.p, in_bytes,
                                hipMemcpyHostToDevice, stream_));
-      HIP_CHECK(hipMemcpyAsync(d_in_off_.p, h_off, (n + 1) * sizeof(uint32_t),
-                               hipMemcpyHostToDevice, stream_));
-      HIP_CHECK(hipMemcpyAsync(d_pb_off_.p, h_off + (n + 1),
-                               (n + 1) * sizeof(uint32_t),
+      uint32_t* d_off = (uint32_t*)d_off3_.p;  // in_off | pb_off contiguous
+      HIP_CHECK(hipMemcpyAsync(d_off, h_off, 2 * (n + 1) * sizeof(uint32_t),
                                hipMemcpyHostToDevice, stream_));
       if (has_idx)
-        HIP_CHECK(hipMemcpyAsync(d_msg_idx_.p, h_aux_.p, n * sizeof(int32_t),
+        HIP_CHECK(hipMemcpyAsync(d_aux2_.p, h_aux_.p, n * sizeof(int32_t),
                                  hipMemcpyHostToDevice, stream_));
       int blocks = (int)cdiv(n, WPB);
       if (blocks > 0) {
         hipLaunchKernelGGL(k_json2pb, dim3(blocks), dim3(WPB * WAVE), 0,
                            stream_, (const uint8_t*)d_in_.p,
-                           (const uint32_t*)d_in_off_.p, (uint8_t*)d_pb_.p,
-                           (const uint32_t*)d_pb_off_.p,
+                           (const uint32_t*)d_off, (uint8_t*)d_pb_.p,
+                           (const uint32_t*)d_off + (n + 1),
                            (SlotResult*)d_results_.p, (uint8_t*)d_id_slots_.p,
-                           has_idx ? (const int32_t*)d_msg_idx_.p : nullptr,
+                           has_idx ? (const int32_t*)d_aux2_.p : nullptr,
                            tables_, lim, n, mode);
         HIP_CHECK(hipGetLastError());
       }
@@ -618,30 +614,26 @@ class Engine {
       HIP_CHECK(hipSetDevice(device_));
       HIP_CHECK(hipMemcpyAsync(d_resp_.p, h_resp_.p, resp_bytes,
                                hipMemcpyHostToDevice, stream_));
-      HIP_CHECK(hipMemcpyAsync(d_resp_off_.p, h_off, (n + 1) * sizeof(uint32_t),
+      // resp_off | final_off | scratch_off contiguous, one copy; same for
+      // msg_idx | skip
+      uint32_t* d_off = (uint32_t*)d_off3_.p;
+      int32_t* d_aux = (int32_t*)d_aux2_.p;
+      HIP_CHECK(hipMemcpyAsync(d_off, h_off, 3 * (n + 1) * sizeof(uint32_t),
                                hipMemcpyHostToDevice, stream_));
-      HIP_CHECK(hipMemcpyAsync(d_final_off_.p, h_off + (n + 1),
-                               (n + 1) * sizeof(uint32_t),
+      HIP_CHECK(hipMemcpyAsync(d_aux, h_aux,
+                               (has_skip ? 2 : 1) * n * sizeof(int32_t),
                                hipMemcpyHostToDevice, stream_));
-      HIP_CHECK(hipMemcpyAsync(d_scratch_off_.p, h_off + 2 * (n + 1),
-                               (n + 1) * sizeof(uint32_t),
-                               hipMemcpyHostToDevice, stream_));
-      HIP_CHECK(hipMemcpyAsync(d_msg_idx_.p, h_aux, n * sizeof(int32_t),
-                               hipMemcpyHostToDevice, stream_));
-      if (has_skip)
-        HIP_CHECK(hipMemcpyAsync(d_skip_.p, h_aux + n, n * sizeof(int32_t),
-                                 hipMemcpyHostToDevice, stream_));
       int blocks = (int)cdiv(n, WPB);
       if (blocks > 0) {
         hipLaunchKernelGGL(
             k_pb2json, dim3(blocks), dim3(WPB * WAVE), 0, stream_,
-            (const uint8_t*)d_resp_.p, (const uint32_t*)d_resp_off_.p,
-            (const int32_t*)d_msg_idx_.p, (const uint8_t*)d_id_slots_.p,
+            (const uint8_t*)d_resp_.p, (const uint32_t*)d_off,
+            (const int32_t*)d_aux, (const uint8_t*)d_id_slots_.p,
             mode == 0 ? (const SlotResult*)d_results_.p : nullptr,
-            (uint8_t*)d_scratch_.p, (const uint32_t*)d_scratch_off_.p,
-            (uint8_t*)d_final_.p, (const uint32_t*)d_final_off_.p,
+            (uint8_t*)d_scratch_.p, (const uint32_t*)d_off + 2 * (n + 1),
+            (uint8_t*)d_final_.p, (const uint32_t*)d_off + (n + 1),
             (DecodeResult*)d_dec_results_.p,
-            has_skip ? (const int32_t*)d_skip_.p : nullptr, tables_, n, mode);
+            has_skip ? (const int32_t*)d_aux + n : nullptr, tables_, n, mode);
         HIP_CHECK(hipGetLastError());
       }
       HIP_CHECK(hipMemcpyAsync(h_dec_results_.p, d_dec_results_.p,
@@ -662,7 +654,7 @@ class Engine {
                                  hipMemcpyHostToDevice, stream_));
         hipLaunchKernelGGL(k_compact_out, dim3(n), dim3(256), 0, stream_,
                            (const uint8_t*)d_final_.p,
-                           (const uint32_t*)d_final_off_.p,
+                           (const uint32_t*)d_off + (n + 1),
                            (const uint32_t*)d_tight_off_.p,
                            (const DecodeResult*)d_dec_results_.p,
                            (uint8_t*)d_pb_.p, n);
@@ -714,8 +706,8 @@ class Engine {
   Tables tables_{};
   DeviceBuf d_msgs_, d_fields_, d_enums_, d_enum_vals_, d_tools_, d_names_;
   DeviceBuf d_in_, d_pb_, d_resp_, d_scratch_, d_final_;
-  DeviceBuf d_in_off_, d_pb_off_, d_resp_off_, d_scratch_off_, d_final_off_;
-  DeviceBuf d_msg_idx_, d_skip_, d_results_, d_dec_results_, d_id_slots_,
+  DeviceBuf d_off3_, d_aux2_;
+  DeviceBuf d_results_, d_dec_results_, d_id_slots_,
       d_tight_off_;
   PinnedBuf h_in_, h_pb_, h_resp_, h_final_, h_results_, h_dec_results_, h_off_,
       h_aux_, h_tight_;
