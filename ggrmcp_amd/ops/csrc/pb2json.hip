@@ -218,10 +218,21 @@ DEV double digits_to_double(uint64_t digits, int ndig, int e10) {
 // digit block and adjusts *e10_out.  Precision-search wrapper below picks the
 // shortest precision whose parse-back round-trips — matching protojson's
 // shortest-representation output for the common cases.
+// nearest integer with exact halves rounded to EVEN — shortest-round-trip
+// printers (Ryu, Go strconv; protojson uses them) break the both-candidates-
+// round-trip tie this way, e.g. float32 1048576.25 prints "1048576.2" not
+// "1048576.3" (fuzz-found divergence)
+DEV uint64_t round_half_even_u64(double scaled) {
+  double t = scaled + 0.5;
+  uint64_t dg = (uint64_t)t;
+  if ((double)dg == t && (dg & 1)) --dg;  // t integral <=> frac was exactly .5
+  return dg;
+}
+
 DEV uint64_t gen_digits(double d, int prec, int* e10_out) {
   int e10 = (int)floor(log10(d));
   double scaled = scale10(d, (prec - 1) - e10);
-  uint64_t digits = (uint64_t)(scaled + 0.5);
+  uint64_t digits = round_half_even_u64(scaled);
   uint64_t hi = 1;
   for (int i = 0; i < prec; ++i) hi *= 10;
   if (digits >= hi) {
@@ -231,7 +242,7 @@ DEV uint64_t gen_digits(double d, int prec, int* e10_out) {
     // log10 estimate was one high
     e10 -= 1;
     scaled = scale10(d, (prec - 1) - e10);
-    digits = (uint64_t)(scaled + 0.5);
+    digits = round_half_even_u64(scaled);
     if (digits >= hi) { digits /= 10; e10 += 1; }
   }
   *e10_out = e10;

@@ -91,6 +91,16 @@ def test_decode_cases(env, msg, payload):
     roundtrip_decode(e, c, p, msg, payload)
 
 
+def test_float_tie_rounds_to_even(env):
+    # fuzz-found: fp32 1048576.25 sits exactly between "1048576.2" and
+    # "1048576.3" (both round-trip); Ryu/protojson break the tie to EVEN.
+    # Exercises round_half_even_u64 in gen_digits (pb2json.hip).
+    e, c, p, _ = env
+    for v in (1048576.25, 2097152.5, 0.15625):
+        payload = {"f08Float": v, "f04Double": v}
+        roundtrip_decode(e, c, p, "bench.Wide64", payload)
+
+
 def test_wide64_roundtrips(env):
     e, c, p, _ = env
     rng = random.Random(3)
