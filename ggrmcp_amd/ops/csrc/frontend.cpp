@@ -47,6 +47,7 @@
 #include <string>
 #include <thread>
 #include <unordered_map>
+#include <unordered_set>
 #include <vector>
 
 namespace py = pybind11;
@@ -124,6 +125,28 @@ class Frontend {
         n_reactors_(reactors < 1 ? 1 : (reactors > 255 ? 255 : reactors)) {}
 
   ~Frontend() { stop(); }
+
+  // Header forwarding filter (reference pkg/headers/filter.go semantics:
+  // disabled -> none; blocked wins; forward_all still honors blocked; else
+  // allow-list).  Applied at parse time so the hot batch callback receives
+  // pre-filtered, LOWERCASED names (h2 requires lowercase on the wire).
+  // Call before start(); only used for case-insensitive configs — the
+  // Python fallback filter handles the case-sensitive variant.
+  void set_header_filter(bool enabled, bool forward_all,
+                         const std::vector<std::string>& allow,
+                         const std::vector<std::string>& block) {
+    auto lower = [](std::string s) {
+      for (auto& ch : s) ch = (char)tolower((unsigned char)ch);
+      return s;
+    };
+    hfilter_on_ = true;
+    hfwd_enabled_ = enabled;
+    hfwd_all_ = forward_all;
+    hallow_.clear();
+    hblock_.clear();
+    for (auto& s : allow) hallow_.insert(lower(s));
+    for (auto& s : block) hblock_.insert(lower(s));
+  }
 
   int start() {
     listen_fd_ = socket(AF_INET, SOCK_STREAM, 0);
@@ -374,8 +397,13 @@ class Frontend {
           is_json = value.find("application/json") != std::string::npos;
         else if (name.size() == 14 && iequal(name.data(), "mcp-session-id", 14))
           session = value;
-        else
+        else if (!hfilter_on_) {
           hdrs.emplace_back(std::move(name), std::move(value));
+        } else if (hfwd_enabled_) {
+          for (auto& ch : name) ch = (char)tolower((unsigned char)ch);
+          if (!hblock_.count(name) && (hfwd_all_ || hallow_.count(name)))
+            hdrs.emplace_back(std::move(name), std::move(value));
+        }
         pos = eol + 2;
       }
       if (clen > max_body_) {
@@ -588,6 +616,8 @@ class Frontend {
   std::string host_;
   int port_;
   py::function batch_cb_, slow_cb_;
+  bool hfilter_on_ = false, hfwd_enabled_ = true, hfwd_all_ = false;
+  std::unordered_set<std::string> hallow_, hblock_;
   int window_us_, max_batch_;
   size_t max_body_;
   double rate_rps_;
@@ -742,6 +772,10 @@ PYBIND11_MODULE(_frontend, m) {
            py::arg("max_batch") = 4096, py::arg("max_body") = 1 << 20,
            py::arg("rate_rps") = 0.0, py::arg("rate_burst") = 0.0,
            py::arg("workers") = 1, py::arg("reactors") = 4)
+      .def("set_header_filter", &Frontend::set_header_filter,
+           py::arg("enabled"), py::arg("forward_all"), py::arg("allow"),
+           py::arg("block"),
+           "install the forwarding filter in the parser (call before start)")
       .def("start", &Frontend::start)
       .def("stop", &Frontend::stop)
       .def_property_readonly("port", &Frontend::port);

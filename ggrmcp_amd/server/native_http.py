@@ -149,6 +149,14 @@ class NativeHTTPGateway:
             self._shard_pool = None
         mod = load_module()
         srv_cfg = self.config.server
+        # case-insensitive filters (the default) run inside the C++ parser:
+        # the batch callback then receives pre-filtered lowercase names and
+        # skips the per-request Python dict rebuild.  Case-sensitive configs
+        # keep the Python filter (C++ normalizes to lowercase).
+        self._cxx_header_filter = bool(
+            getattr(self.headers, "case_insensitive", False)
+            and hasattr(mod.Frontend, "set_header_filter")
+        )
         self._fe = mod.Frontend(
             host,
             port,
@@ -160,6 +168,13 @@ class NativeHTTPGateway:
             rate_rps=float(srv_cfg.rate_limit_rps),
             rate_burst=float(srv_cfg.rate_limit_burst),
         )
+        if self._cxx_header_filter:
+            self._fe.set_header_filter(
+                self.headers.enabled,
+                self.headers.forward_all,
+                sorted(self.headers.allowed),
+                sorted(self.headers.blocked),
+            )
         self.port = 0
 
     def start(self) -> int:
@@ -191,7 +206,8 @@ class NativeHTTPGateway:
                 rejected[i] = self._session_error(bodies[i], "session rate limit exceeded")
                 continue
             sess.increment_call_count()
-            fwd_headers.append(self.headers.filter_headers(headers[i]))
+            fwd_headers.append(headers[i] if self._cxx_header_filter
+                               else self.headers.filter_headers(headers[i]))
         timeout = self.config.grpc.request_timeout_s
         if rejected:
             live_idx = [i for i in range(n) if i not in rejected]

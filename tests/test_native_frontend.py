@@ -99,6 +99,30 @@ def test_tools_call_roundtrip(gateway):
     assert sid  # session issued
 
 
+def test_header_filter_in_cxx_parser(gateway):
+    # forwarding filter runs inside the C++ parser (filter.go semantics):
+    # allowed headers arrive LOWERCASED, blocked and unlisted ones never
+    # reach Python
+    gw, port, pipe = gateway
+    assert gw._cxx_header_filter  # default config is case-insensitive
+    body = json.dumps({"jsonrpc": "2.0", "id": 7, "method": "tools/call",
+                       "params": {"name": "t", "arguments": {}}})
+    conn = http.client.HTTPConnection("127.0.0.1", port, timeout=10)
+    conn.request("POST", "/", body=body, headers={
+        "Content-Type": "application/json",
+        "AUTHORIZATION": "Bearer tok",   # allowed, mixed case
+        "X-Trace-Id": "t-1",             # allowed
+        "Cookie": "secret=1",            # blocked
+        "X-Custom": "nope",              # not on the allow list
+    })
+    r = conn.getresponse()
+    data = r.read()
+    conn.close()
+    assert r.status == 200
+    inner = json.loads(json.loads(data)["result"]["content"][0]["text"])
+    assert inner["hdr"] == {"authorization": "Bearer tok", "x-trace-id": "t-1"}
+
+
 def test_session_reuse(gateway):
     gw, port, pipe = gateway
     body = json.dumps({"jsonrpc": "2.0", "id": 1, "method": "tools/call",
