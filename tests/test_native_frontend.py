@@ -123,6 +123,22 @@ def test_header_filter_in_cxx_parser(gateway):
     assert inner["hdr"] == {"authorization": "Bearer tok", "x-trace-id": "t-1"}
 
 
+def test_case_sensitive_filter_falls_back_to_python():
+    # case-sensitive forwarding configs keep the Python filter (the C++
+    # parser always normalizes names to lowercase)
+    from ggrmcp_amd.headers import HeaderFilter
+
+    cfg = Config.default()
+    hf = HeaderFilter(enabled=True, allowed=("X-Exact",), blocked=(),
+                      case_insensitive=False)
+    gw = NativeHTTPGateway(StubPipeline(), StubDiscoverer(), cfg,
+                           header_filter=hf)
+    try:
+        assert not gw._cxx_header_filter
+    finally:
+        gw.stop()
+
+
 def test_session_reuse(gateway):
     gw, port, pipe = gateway
     body = json.dumps({"jsonrpc": "2.0", "id": 1, "method": "tools/call",
