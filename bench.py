@@ -65,8 +65,10 @@ def parse_args():
                     help="wire invoker: native C++ h2 client or grpcio threads")
     ap.add_argument("--backend", default="native", choices=["native", "python"],
                     help="local bench backend implementation")
-    ap.add_argument("--connections", type=int, default=8,
-                    help="native transport connections per backend")
+    ap.add_argument("--connections", type=int, default=0,
+                    help="native transport connections per backend "
+                         "(0 = auto: 8, or 16 for wide64 whose 128 MB/step "
+                         "is UDS-syscall bound — profiles/wide64_conns.log)")
     ap.add_argument("--streams", type=int, default=0,
                     help="override config.gpu.streams (engine instances)")
     return ap.parse_args()
@@ -248,7 +250,8 @@ def main() -> None:
             from ggrmcp_amd.backend.native_invoker import NativeWireClient
 
             wire_clients = [
-                NativeWireClient(f"unix:{s}", connections=args.connections)
+                NativeWireClient(f"unix:{s}", connections=(
+                    args.connections or (16 if args.config == "wide64" else 8)))
                 for s in socks
             ]
 
