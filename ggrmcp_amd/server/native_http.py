@@ -196,16 +196,17 @@ class NativeHTTPGateway:
         rejected: Dict[int, bytes] = {}
         rl_on = self.config.session.rate_limit_enabled
         for i in range(n):
-            sess = self.sessions.get_or_create(session_ids[i], headers[i])
+            # per-session guards (handler.go:219-226 parity) in one
+            # manager-lock cycle (see SessionManager.guard)
+            sess, verdict = self.sessions.guard(session_ids[i], headers[i],
+                                                rate_limit=rl_on)
             sids.append(sess.id)
-            # per-session guards (handler.go:219-226 parity)
-            if sess.is_blocked:
+            if verdict == 1:
                 rejected[i] = self._session_error(bodies[i], "session is blocked")
                 continue
-            if rl_on and not self.sessions.check_rate_limit(sess):
+            if verdict == 2:
                 rejected[i] = self._session_error(bodies[i], "session rate limit exceeded")
                 continue
-            sess.increment_call_count()
             fwd_headers.append(headers[i] if self._cxx_header_filter
                                else self.headers.filter_headers(headers[i]))
         timeout = self.config.grpc.request_timeout_s

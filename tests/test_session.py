@@ -73,3 +73,28 @@ def test_shard_distribution():
     shards = [m.get_or_create(None).shard(8) for _ in range(800)]
     counts = [shards.count(i) for i in range(8)]
     assert all(c > 40 for c in counts)  # roughly uniform
+
+
+def test_guard_fast_path():
+    # one-lock serving fast lane == the 3-call sequence semantically
+    m = SessionManager(rate_limit_per_min=2, rate_limit_burst=0)
+    ctx, v = m.guard(None, {"user-agent": "x"})
+    assert v == 0 and ctx.id and ctx.call_count == 1
+    assert ctx.headers == {"user-agent": "x"}
+    # existing session: same ctx, counter advances
+    ctx2, v = m.guard(ctx.id)
+    assert v == 0 and ctx2 is ctx and ctx.call_count == 2
+    # window exhausted -> verdict 2, call_count NOT bumped
+    _, v = m.guard(ctx.id)
+    assert v == 2 and ctx.call_count == 2
+    # blocked wins over rate limit
+    m.block(ctx.id)
+    _, v = m.guard(ctx.id)
+    assert v == 1
+    # rate_limit=False skips the window but still counts the call
+    m.unblock(ctx.id)
+    _, v = m.guard(ctx.id, rate_limit=False)
+    assert v == 0 and ctx.call_count == 3
+    # expired/unknown id -> fresh session under the same id semantics
+    ctx3, v = m.guard("brand-new-id")
+    assert v == 0 and ctx3.id == "brand-new-id" and ctx3.call_count == 1
