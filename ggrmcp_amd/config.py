@@ -38,6 +38,10 @@ class ConnectionConfig:
     request_timeout_s: float = 30.0  # config.go:235
     use_tls: bool = False
     uds: str = ""  # unix-domain socket path (overrides host:port)
+    # h2 connections per backend for the native batch transport.  8 suits
+    # ~1 KB payloads; raise to 16 for multi-MB/step workloads, which are
+    # UDS-syscall bound (profiles/wide64_conns.log: +27% at 64 KB payloads)
+    native_connections: int = 8
 
     @property
     def target(self) -> str:
@@ -225,6 +229,8 @@ class Config:
                 errs.append("gpu.max_batch must be >= 1")
             if self.gpu.streams < 1:
                 errs.append("gpu.streams must be >= 1")
+        if self.grpc.native_connections < 1:
+            errs.append("grpc.native_connections must be >= 1")
         if self.logging.level not in ("debug", "info", "warn", "warning", "error"):
             errs.append(f"unknown logging.level: {self.logging.level}")
         if errs:

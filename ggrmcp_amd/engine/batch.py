@@ -750,9 +750,10 @@ def build_wire_clients(discoverer, config: Config):
     except Exception:  # pragma: no cover
         return None
     clients = []
+    n_conns = getattr(config.grpc, "native_connections", 8) or 8
     try:
         for conn in discoverer.connections:
-            clients.append(NativeWireClient(conn.target))
+            clients.append(NativeWireClient(conn.target, connections=n_conns))
     except Exception as e:  # pragma: no cover - fall back to grpcio
         log.warning("native transport unavailable (%s); using grpcio", e)
         for c in clients:

@@ -43,6 +43,10 @@ def test_validate_rejects_bad_values():
     cfg.gpu.devices = 9
     with pytest.raises(ValueError):
         cfg.validate()
+    cfg = Config.default()
+    cfg.grpc.native_connections = 0
+    with pytest.raises(ValueError):
+        cfg.validate()
 
 
 def test_from_dict_and_unknown_key():
