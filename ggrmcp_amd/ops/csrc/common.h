@@ -284,3 +284,56 @@ DEV uint32_t i64_to_dec(uint8_t* out, int64_t v) {
   }
   return u64_to_dec(out, (uint64_t)v);
 }
+
+// ---------------------------------------------------------------------------
+// Strict UTF-8 validation (RFC 3629), wave-parallel: each lane judges its
+// own byte.  Lead bytes verify their continuations + overlong/surrogate/
+// range rules; continuation bytes verify a lead 1-3 bytes back claims them
+// (a bad overlap always also fails the lead's own check).  protojson
+// rejects invalid UTF-8 in proto3 strings in BOTH directions (Go proto
+// wire unmarshal and protojson.Unmarshal), so the kernels must too:
+// decode -> E_UNSUPPORTED (host re-attempt surfaces the error), encode ->
+// E_PARSE.  Fuzz-found: wire 0x0A 0x0D 00*12 0x80 decoded to JSON that
+// json.loads could not UTF-8-decode.
+// ---------------------------------------------------------------------------
+
+DEV bool utf8_byte_ok(const uint8_t* s, uint32_t n, uint32_t i) {
+  uint8_t b = s[i];
+  if (b < 0x80) return true;
+  if (b < 0xC0) {  // continuation: must be claimed by a preceding lead
+    if (i >= 1) { uint8_t p = s[i - 1]; if (p >= 0xC2 && p <= 0xF4) return true; }
+    if (i >= 2) { uint8_t p = s[i - 2]; if (p >= 0xE0 && p <= 0xF4) return true; }
+    if (i >= 3) { uint8_t p = s[i - 3]; if (p >= 0xF0 && p <= 0xF4) return true; }
+    return false;
+  }
+  if (b < 0xC2) return false;  // C0/C1: overlong 2-byte form
+  uint8_t c1 = i + 1 < n ? s[i + 1] : 0;
+  bool cont1 = (c1 & 0xC0) == 0x80;
+  if (b < 0xE0) return cont1;  // 2-byte lead
+  uint8_t c2 = i + 2 < n ? s[i + 2] : 0;
+  bool cont2 = (c2 & 0xC0) == 0x80;
+  if (b < 0xF0) {  // 3-byte lead
+    if (!cont1 || !cont2) return false;
+    if (b == 0xE0 && c1 < 0xA0) return false;  // overlong
+    if (b == 0xED && c1 > 0x9F) return false;  // UTF-16 surrogate range
+    return true;
+  }
+  if (b > 0xF4) return false;  // beyond U+10FFFF
+  uint8_t c3 = i + 3 < n ? s[i + 3] : 0;
+  if (!cont1 || !cont2 || (c3 & 0xC0) != 0x80) return false;
+  if (b == 0xF0 && c1 < 0x90) return false;  // overlong
+  if (b == 0xF4 && c1 > 0x8F) return false;  // > U+10FFFF
+  return true;
+}
+
+// whole-span check; ASCII windows cost one ballot (the common case)
+DEV bool utf8_span_valid(const uint8_t* s, uint32_t n, uint32_t lane) {
+  for (uint32_t base = 0; base < n; base += WAVE) {
+    uint32_t i = base + lane;
+    uint8_t b = i < n ? s[i] : 0;
+    if (!__ballot(b >= 0x80)) continue;
+    bool ok = i < n ? utf8_byte_ok(s, n, i) : true;
+    if (__ballot(!ok)) return false;
+  }
+  return true;
+}

@@ -120,6 +120,10 @@ DEV bool string_span(Ctx& c, uint32_t* start, uint32_t* rawlen, bool* has_esc) {
   if (!string_end(c, &end, has_esc)) return false;
   *rawlen = end - *start;
   c.pos = end + 1;
+  // JSON text must be valid UTF-8 (protojson.Unmarshal rejects otherwise);
+  // escapes are pure ASCII so validating the raw span is exact
+  if (!utf8_span_valid(c.s + *start, *rawlen, c.lane))
+    return fail(c, E_PARSE);
   return true;
 }
 

@@ -122,8 +122,11 @@ DEV uint64_t esc_pack(uint8_t b, uint32_t el) {
   }
 }
 
-// append src[0..n) JSON-escaped; lane-parallel with a no-escape fast path
+// append src[0..n) JSON-escaped; lane-parallel with a no-escape fast path.
+// Invalid UTF-8 (proto3 string fields must be valid; protojson errors) ->
+// E_UNSUPPORTED so the host re-attempt surfaces the proper error.
 DEV bool put_escaped(DCtx& c, const uint8_t* src, uint32_t n) {
+  if (!utf8_span_valid(src, n, c.lane)) return dfail(c, E_UNSUPPORTED);
   for (uint32_t base = 0; base < n; base += WAVE) {
     uint32_t i = base + c.lane;
     uint32_t win = n - base < WAVE ? n - base : WAVE;

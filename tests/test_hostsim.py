@@ -91,6 +91,30 @@ def test_decode_cases(env, msg, payload):
     roundtrip_decode(e, c, p, msg, payload)
 
 
+def test_invalid_utf8_rejected_both_directions(env):
+    # proto3 strings must be valid UTF-8; protojson errors in both
+    # directions (fuzz-found: decoder emitted a JSON string json.loads
+    # could not UTF-8-decode)
+    e, c, p, _ = env
+    idx = e.tables.msg_index["bench.Wide64"]
+    # decode: field 1 (f01String) carrying a lone continuation byte
+    wire = b"\n\r" + b"\x00" * 12 + b"\x80"
+    dec, _ = e.decode_batch([wire], [idx], mode=1)
+    assert int(dec[0]["status"]) == 6  # E_UNSUPPORTED -> host error parity
+    # overlong + surrogate + truncated sequences
+    for bad in (b"\xc0\xaf", b"\xed\xa0\x80", b"\xf5\x80\x80\x80", b"\xe2\x82"):
+        dec, _ = e.decode_batch([b"\n" + bytes([len(bad)]) + bad], [idx], mode=1)
+        assert int(dec[0]["status"]) == 6, bad
+    # encode: raw invalid UTF-8 inside a JSON string -> E_PARSE
+    body = b'{"f01String": "\x80abc"}'
+    enc, _ = e.encode_batch([body], mode=1, msg_indices=[idx])
+    assert int(enc[0]["status"]) == 1  # E_PARSE
+    # valid multibyte still round-trips
+    payload = {"f01String": "héllo → \U0001f389"}
+    roundtrip_encode(e, c, p, "bench.Wide64", payload)
+    roundtrip_decode(e, c, p, "bench.Wide64", payload)
+
+
 def test_float_tie_rounds_to_even(env):
     # fuzz-found: fp32 1048576.25 sits exactly between "1048576.2" and
     # "1048576.3" (both round-trip); Ryu/protojson break the tie to EVEN.
