@@ -383,3 +383,17 @@ def test_big_batch(env):
             c.pb_to_message(desc, c.json_to_pb(desc, payloads[i].decode()))
         )
         assert _approx(g, o)
+
+
+def test_invalid_utf8_rejected(env):
+    # proto3 strings must be valid UTF-8 (protojson parity) — GPU build
+    e, c, p, _ = env
+    idx = e.tables.msg_index["bench.Wide64"]
+    wire = b"\n\r" + b"\x00" * 12 + b"\x80"
+    dec, _ = e.decode_batch([wire], [idx], mode=1)
+    assert int(dec[0]["status"]) == 6  # E_UNSUPPORTED -> host error parity
+    enc, _ = e.encode_batch([b'{"f01String": "\x80abc"}'], mode=1,
+                            msg_indices=[idx])
+    assert int(enc[0]["status"]) == 1  # E_PARSE
+    roundtrip_encode(e, c, p, "bench.Wide64", {"f01String": "hé \U0001f389"})
+    roundtrip_decode(e, c, p, "bench.Wide64", {"f01String": "hé \U0001f389"})
