@@ -184,6 +184,9 @@ DEV uint32_t unescape_serial(Ctx& c, const uint8_t* src, uint32_t n, uint8_t* ds
             i += 6;
           }
         }
+        // lone/unpaired surrogate: protojson rejects (would encode to
+        // invalid UTF-8)
+        if (cp >= 0xD800 && cp <= 0xDFFF) return 0xFFFFFFFF;
         // UTF-8 encode
         if (cp < 0x80) {
           if (o + 1 > cap) return 0xFFFFFFFF;

@@ -113,6 +113,16 @@ def test_invalid_utf8_rejected_both_directions(env):
     payload = {"f01String": "héllo → \U0001f389"}
     roundtrip_encode(e, c, p, "bench.Wide64", payload)
     roundtrip_decode(e, c, p, "bench.Wide64", payload)
+    # lone \u surrogate escape rejected (protojson parity); a proper
+    # pair still decodes to the astral char
+    enc, _ = e.encode_batch([b'{"f01String": "a\\ud800b"}'], mode=1,
+                            msg_indices=[idx])
+    assert int(enc[0]["status"]) != 0
+    enc, pbs = e.encode_batch([b'{"f01String": "\\ud83c\\udf89"}'], mode=1,
+                              msg_indices=[idx])
+    assert int(enc[0]["status"]) == 0
+    desc = p.FindMessageTypeByName("bench.Wide64")
+    assert c.pb_to_message(desc, pbs[0]).f01_string == "\U0001f389"
 
 
 def test_float_tie_rounds_to_even(env):
