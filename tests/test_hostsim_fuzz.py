@@ -299,11 +299,15 @@ def test_fuzz_truncated_valid_envelope(prefix, cut):
 
 
 @settings(max_examples=200, deadline=None)
-@given(data=st.binary(max_size=200))
-def test_fuzz_hostile_wire_decode(data):
-    """Random bytes as protobuf wire through the decoder: clean status,
-    and successful decodes must emit valid JSON."""
-    idx = _engine.tables.msg_index["bench.Wide64"]
+@given(data=st.binary(max_size=200),
+       msg=st.sampled_from(["bench.Wide64", "complex.NodeRequest",
+                            "complex.Document", "complex.UserProfile"]))
+def test_fuzz_hostile_wire_decode(data, msg):
+    """Random bytes as protobuf wire through the decoder (several message
+    shapes: wide scalars, recursion, maps+oneofs): clean status, and
+    successful decodes must emit valid JSON (incl. valid UTF-8 — this
+    fuzz caught the decoder forwarding invalid string-field bytes)."""
+    idx = _engine.tables.msg_index[msg]
     dec, outs = _engine.decode_batch([data], [idx], mode=1)
     s = int(dec[0]["status"])
     assert 0 <= s <= 8
