@@ -120,6 +120,10 @@ class ServerConfig:
     rate_limit_burst: int = 200
     rate_limit_enabled: bool = True
     cors_enabled: bool = True
+    # SO_REUSEPORT on the native frontend: run one gateway process per GPU
+    # rank all bound to the SAME port; the kernel balances connections
+    # (tools/serve_dp.py).  No reference equivalent (single process).
+    reuse_port: bool = False
     security_headers_enabled: bool = True
 
 

@@ -148,10 +148,17 @@ class Frontend {
     for (auto& s : block) hblock_.insert(lower(s));
   }
 
+  // allow N gateway processes (one per GPU rank) to share ONE port: the
+  // kernel load-balances accepted connections across listeners.  Call
+  // before start().
+  void set_reuse_port(bool on) { reuse_port_ = on; }
+
   int start() {
     listen_fd_ = socket(AF_INET, SOCK_STREAM, 0);
     int one = 1;
     setsockopt(listen_fd_, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+    if (reuse_port_)
+      setsockopt(listen_fd_, SOL_SOCKET, SO_REUSEPORT, &one, sizeof(one));
     sockaddr_in addr{};
     addr.sin_family = AF_INET;
     addr.sin_port = htons((uint16_t)port_);
@@ -616,6 +623,7 @@ class Frontend {
   std::string host_;
   int port_;
   py::function batch_cb_, slow_cb_;
+  bool reuse_port_ = false;
   bool hfilter_on_ = false, hfwd_enabled_ = true, hfwd_all_ = false;
   std::unordered_set<std::string> hallow_, hblock_;
   int window_us_, max_batch_;
@@ -772,6 +780,8 @@ PYBIND11_MODULE(_frontend, m) {
            py::arg("max_batch") = 4096, py::arg("max_body") = 1 << 20,
            py::arg("rate_rps") = 0.0, py::arg("rate_burst") = 0.0,
            py::arg("workers") = 1, py::arg("reactors") = 4)
+      .def("set_reuse_port", &Frontend::set_reuse_port, py::arg("on"),
+           "SO_REUSEPORT: N gateway ranks share one port (call before start)")
       .def("set_header_filter", &Frontend::set_header_filter,
            py::arg("enabled"), py::arg("forward_all"), py::arg("allow"),
            py::arg("block"),

@@ -168,6 +168,8 @@ class NativeHTTPGateway:
             rate_rps=float(srv_cfg.rate_limit_rps),
             rate_burst=float(srv_cfg.rate_limit_burst),
         )
+        if getattr(srv_cfg, "reuse_port", False) and hasattr(self._fe, "set_reuse_port"):
+            self._fe.set_reuse_port(True)
         if self._cxx_header_filter:
             self._fe.set_header_filter(
                 self.headers.enabled,

@@ -393,3 +393,36 @@ def test_connection_churn_mid_batch(gateway):
     for t in ts:
         t.join()
     assert not errs, errs[:3]
+
+
+def test_reuse_port_two_gateways_one_port():
+    """SO_REUSEPORT DP serving shape: N gateway processes share one port
+    (here: two gateways in-process); the kernel balances connections and
+    every request is answered correctly by whichever gateway got it."""
+    cfg = Config.default()
+    cfg.server.rate_limit_rps = 100000
+    cfg.server.rate_limit_burst = 100000
+    cfg.server.reuse_port = True
+    p1, p2 = StubPipeline(), StubPipeline()
+    gw1 = NativeHTTPGateway(p1, StubDiscoverer(), cfg)
+    port = gw1.start()
+    gw2 = NativeHTTPGateway(p2, StubDiscoverer(), cfg, port=port)
+    assert gw2.start() == port
+    try:
+        body = json.dumps({"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                           "params": {"name": "t", "arguments": {"k": 1}}})
+        ok = 0
+        for _ in range(48):  # one fresh connection each -> kernel balances
+            status, data, _ = _call(port, body)
+            assert status == 200
+            assert json.loads(data)["result"]["isError"] is False
+            ok += 1
+        assert ok == 48
+        served = [sum(p1.batches), sum(p2.batches)]
+        assert sum(served) == 48
+        # with 48 independent connections both listeners see traffic
+        # (kernel-balanced; P[all one side] ~ 2^-47)
+        assert min(served) >= 1, served
+    finally:
+        gw1.stop()
+        gw2.stop()

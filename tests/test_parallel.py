@@ -118,3 +118,82 @@ def test_world2_gloo(tmp_path):
     for rank in range(2):
         with open(tmp_path / f"rank{rank}.json") as f:
             assert json.load(f)["ok"]
+
+
+@pytest.mark.timeout(180)
+def test_serve_dp_world2_shared_port(tmp_path):
+    """tools/serve_dp.py end to end: two gateway ranks (gloo, CPU pipeline)
+    share ONE port via SO_REUSEPORT, rank 0 broadcasts the descriptor set,
+    and real HTTP tool calls round-trip through whichever rank the kernel
+    picks."""
+    import http.client
+    import socket
+    import subprocess
+    import sys
+    import time as _t
+
+    from examples.hello_service import serve as serve_hello
+    from examples.protos import ALL_FDPS
+    from google.protobuf import descriptor_pb2
+
+    sock_path = str(tmp_path / "hello.sock")
+    server, _ = serve_hello(target=f"unix:{sock_path}")
+    try:
+        fdset = descriptor_pb2.FileDescriptorSet()
+        fdset.file.extend(ALL_FDPS)
+        desc_path = tmp_path / "svc.binpb"
+        desc_path.write_bytes(fdset.SerializeToString())
+
+        def free_port():
+            s = socket.socket()
+            s.bind(("127.0.0.1", 0))
+            p = s.getsockname()[1]
+            s.close()
+            return p
+
+        http_port, master_port = free_port(), free_port()
+        proc = subprocess.Popen(
+            [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+             "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+             "--master-port", str(master_port), "tools/serve_dp.py",
+             "--port", str(http_port), "--uds", sock_path,
+             "--descriptor", str(desc_path), "--no-gpu",
+             "--run-seconds", "60"],
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+        )
+        try:
+            body = json.dumps({
+                "jsonrpc": "2.0", "id": 5, "method": "tools/call",
+                "params": {"name": "hello_helloservice_sayhello",
+                           "arguments": {"name": "dp"}},
+            })
+            deadline = _t.time() + 90
+            ok = 0
+            while _t.time() < deadline and ok < 6:
+                try:
+                    conn = http.client.HTTPConnection("127.0.0.1", http_port,
+                                                      timeout=5)
+                    conn.request("POST", "/", body=body,
+                                 headers={"Content-Type": "application/json"})
+                    r = conn.getresponse()
+                    data = json.loads(r.read())
+                    conn.close()
+                    if r.status == 200 and not data["result"]["isError"]:
+                        inner = json.loads(
+                            data["result"]["content"][0]["text"])
+                        assert inner == {"message": "Hello, dp!"}
+                        ok += 1
+                        continue
+                except (ConnectionError, OSError, TimeoutError):
+                    pass
+                _t.sleep(0.5)
+            assert ok >= 6, f"only {ok} round-trips; output:\n" + (
+                proc.stdout.read() if proc.poll() is not None else "")
+        finally:
+            proc.terminate()
+            try:
+                proc.wait(timeout=15)
+            except Exception:
+                proc.kill()
+    finally:
+        server.stop(0)
