@@ -17,6 +17,14 @@ Launch (8 GPUs):
 Each rank reads RANK/LOCAL_RANK/WORLD_SIZE from the environment.  With
 --no-gpu (or no CUDA) it serves the CPU reference pipeline — which is how
 the world-2 CPU test exercises this file end to end.
+
+Session-affinity caveat: SO_REUSEPORT balances CONNECTIONS, and session
+state (counters, blocks, rate windows) is per-rank, exactly like the
+reference's per-process in-memory sessions.  A client that keeps its
+HTTP connection (the normal MCP shape) stays on one rank; a client that
+reconnects mid-session may land on another rank and start a fresh
+session context there.  Front with an L4 source-hash balancer if strict
+cross-reconnect affinity matters.
 """
 
 from __future__ import annotations
