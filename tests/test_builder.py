@@ -113,3 +113,16 @@ def test_comment_index(env):
     idx = build_comment_index(ALL_FDPS)
     assert idx["hello.HelloService.SayHello"].startswith("SayHello returns")
     assert idx["hello.HelloRequest"].startswith("The request")
+
+
+def test_circular_proto_dependency_rejected():
+    # loader.go:67-134's recursive processFile would loop; ours raises
+    # (loader.py build_pool cycle detection)
+    from google.protobuf import descriptor_pb2
+
+    a = descriptor_pb2.FileDescriptorProto(name="a.proto", package="a",
+                                           dependency=["b.proto"])
+    b = descriptor_pb2.FileDescriptorProto(name="b.proto", package="b",
+                                           dependency=["a.proto"])
+    with pytest.raises(ValueError, match="circular"):
+        build_pool([a, b])
