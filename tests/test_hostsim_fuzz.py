@@ -53,6 +53,10 @@ _u64s = st.integers(min_value=0, max_value=2**64 - 1).map(str)
 _dbl = st.one_of(
     st.floats(allow_nan=False, allow_infinity=False, width=64),
     st.sampled_from([0.0, -0.0, 1e-300, 1e300, 2.2250738585072014e-308]),
+    # dyadic rationals: exact in fp64, tie-prone decimal digit strings
+    st.builds(lambda m, k: float(m) / (1 << k),
+              st.integers(min_value=1, max_value=2**40 - 1),
+              st.integers(min_value=0, max_value=40)),
 )
 # dyadic rationals m*2^-k land exactly between short decimal strings and
 # hammer the formatter's round-half-to-even tie-break (fuzz caught fp32
