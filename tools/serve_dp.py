@@ -93,15 +93,19 @@ def main() -> None:
         )
 
     discoverer = ServiceDiscoverer(cfg)
-    if args.descriptor and (rank == 0 or shard_group is None):
-        discoverer.load_descriptor_blob(Path(args.descriptor).read_bytes())
+    # rank 0 discovers (descriptor file, or live reflection) and broadcasts
+    # the framed snapshot so every rank serves IDENTICAL tools
+    # (discovery.go:122-127's atomic swap, node-wide)
+    if rank == 0 or shard_group is None:
+        if args.descriptor:
+            discoverer.load_descriptor_blob(Path(args.descriptor).read_bytes())
+        else:
+            discoverer.connect(timeout_s=30)
+            discoverer.discover()
     if shard_group is not None:
         from ggrmcp_amd.parallel.dist import sync_discovery
 
         sync_discovery(discoverer, shard_group, src=0)
-    if not discoverer.tools:
-        discoverer.connect(timeout_s=30)
-        discoverer.discover()
     discoverer.connections[0].connect(timeout_s=15)
 
     from ggrmcp_amd.server.native_http import (
