@@ -121,11 +121,13 @@ def test_world2_gloo(tmp_path):
 
 
 @pytest.mark.timeout(180)
-def test_serve_dp_world2_shared_port(tmp_path):
+@pytest.mark.parametrize("use_descriptor", [True, False],
+                         ids=["descriptor", "reflection"])
+def test_serve_dp_world2_shared_port(tmp_path, use_descriptor):
     """tools/serve_dp.py end to end: two gateway ranks (gloo, CPU pipeline)
-    share ONE port via SO_REUSEPORT, rank 0 broadcasts the descriptor set,
-    and real HTTP tool calls round-trip through whichever rank the kernel
-    picks."""
+    share ONE port via SO_REUSEPORT, rank 0 discovers (descriptor file OR
+    live reflection) and broadcasts the snapshot, and real HTTP tool calls
+    round-trip through whichever rank the kernel picks."""
     import http.client
     import socket
     import subprocess
@@ -152,14 +154,16 @@ def test_serve_dp_world2_shared_port(tmp_path):
             return p
 
         http_port, master_port = free_port(), free_port()
+        cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+               "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+               "--master-port", str(master_port), "tools/serve_dp.py",
+               "--port", str(http_port), "--uds", sock_path,
+               "--no-gpu", "--run-seconds", "60"]
+        if use_descriptor:
+            cmd += ["--descriptor", str(desc_path)]
+        # else: rank 0 discovers via live gRPC REFLECTION, then broadcasts
         proc = subprocess.Popen(
-            [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-             "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-             "--master-port", str(master_port), "tools/serve_dp.py",
-             "--port", str(http_port), "--uds", sock_path,
-             "--descriptor", str(desc_path), "--no-gpu",
-             "--run-seconds", "60"],
-            stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+            cmd, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
         )
         try:
             body = json.dumps({
