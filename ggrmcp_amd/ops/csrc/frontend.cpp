@@ -172,6 +172,8 @@ class Frontend {
     fcntl(listen_fd_, F_SETFL, fcntl(listen_fd_, F_GETFL, 0) | O_NONBLOCK);
 
     stop_.store(false);
+    draining_.store(false);
+    inflight_.store(0);
     reactors_.clear();
     for (int r = 0; r < n_reactors_; ++r) {
       auto re = std::make_unique<Reactor>();
@@ -196,6 +198,27 @@ class Frontend {
     for (int i = 0; i < n_workers_; ++i)
       worker_threads_.emplace_back([this] { worker_loop(); });
     return port_;
+  }
+
+  // Graceful drain (reference main.go:94-112's 30 s shutdown window, for
+  // the native path): stop accepting, let queued + in-flight requests
+  // finish, then a short grace for socket flushes.  Returns the number of
+  // requests still unanswered at timeout (0 = clean drain).  Call before
+  // stop().
+  long drain(double timeout_s) {
+    draining_.store(true, std::memory_order_relaxed);
+    auto deadline =
+        Clock::now() + std::chrono::duration_cast<Clock::duration>(
+                           std::chrono::duration<double>(timeout_s));
+    {
+      // workers need the GIL for batch callbacks — release while waiting
+      py::gil_scoped_release rel;
+      while (inflight_.load(std::memory_order_acquire) > 0 &&
+             Clock::now() < deadline)
+        std::this_thread::sleep_for(std::chrono::milliseconds(5));
+      std::this_thread::sleep_for(std::chrono::milliseconds(100));
+    }
+    return inflight_.load(std::memory_order_acquire);
   }
 
   void stop() {
@@ -295,6 +318,9 @@ class Frontend {
   }
 
   void accept_new(Reactor* re0) {
+    // draining: leave new connections in the backlog; they die with the
+    // listener at stop()
+    if (draining_.load(std::memory_order_relaxed)) return;
     while (true) {
       int cfd = accept(listen_fd_, nullptr, nullptr);
       if (cfd < 0) return;
@@ -447,6 +473,7 @@ class Frontend {
         continue;
       }
       if (re->pending.empty()) re->first_pending = Clock::now();
+      inflight_.fetch_add(1, std::memory_order_relaxed);
       re->pending.push_back(std::move(req));
     }
   }
@@ -517,6 +544,11 @@ class Frontend {
       done.swap(re->done);
     }
     for (auto& r : done) complete(re, r.conn_id, r.seq, std::move(r.payload));
+    // decrement AFTER complete(): when drain() sees zero the responses are
+    // already in connection write buffers (its grace period covers the
+    // final socket flush)
+    if (!done.empty())
+      inflight_.fetch_sub((long)done.size(), std::memory_order_release);
   }
 
   // ---- workers: one GIL crossing per batch --------------------------------
@@ -635,6 +667,8 @@ class Frontend {
 
   int listen_fd_ = -1;
   std::atomic<bool> stop_{true};
+  std::atomic<bool> draining_{false};
+  std::atomic<long> inflight_{0};
   std::atomic<uint64_t> rr_{0};
   std::vector<std::unique_ptr<Reactor>> reactors_;
   std::vector<std::thread> worker_threads_;
@@ -780,6 +814,9 @@ PYBIND11_MODULE(_frontend, m) {
            py::arg("max_batch") = 4096, py::arg("max_body") = 1 << 20,
            py::arg("rate_rps") = 0.0, py::arg("rate_burst") = 0.0,
            py::arg("workers") = 1, py::arg("reactors") = 4)
+      .def("drain", &Frontend::drain, py::arg("timeout_s"),
+           "graceful drain: stop accepting, finish in-flight requests; "
+           "returns requests still unanswered at timeout")
       .def("set_reuse_port", &Frontend::set_reuse_port, py::arg("on"),
            "SO_REUSEPORT: N gateway ranks share one port (call before start)")
       .def("set_header_filter", &Frontend::set_header_filter,

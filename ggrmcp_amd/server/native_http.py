@@ -183,7 +183,11 @@ class NativeHTTPGateway:
         self.port = self._fe.start()
         return self.port
 
-    def stop(self) -> None:
+    def stop(self, drain_s: float = 0.0) -> None:
+        """drain_s > 0: graceful shutdown — stop accepting, finish
+        in-flight requests (bounded), then stop (main.go:94-112 parity)."""
+        if drain_s > 0 and hasattr(self._fe, "drain"):
+            self._fe.drain(drain_s)
         self._fe.stop()
         if self._shard_pool is not None:
             self._shard_pool.shutdown(wait=False)

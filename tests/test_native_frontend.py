@@ -426,3 +426,45 @@ def test_reuse_port_two_gateways_one_port():
     finally:
         gw1.stop()
         gw2.stop()
+
+
+def test_graceful_drain():
+    """stop(drain_s=...) finishes in-flight requests before tearing down
+    (main.go:94-112's shutdown window, native path)."""
+    import time
+
+    cfg = Config.default()
+    cfg.server.rate_limit_rps = 100000
+    cfg.server.rate_limit_burst = 100000
+
+    class SlowPipeline(StubPipeline):
+        def process_batch(self, bodies, headers=None, timeout_s=None):
+            time.sleep(0.4)  # request is mid-flight when stop() arrives
+            return StubPipeline.process_batch(self, bodies, headers, timeout_s)
+
+    pipe = SlowPipeline()
+    gw = NativeHTTPGateway(pipe, StubDiscoverer(), cfg)
+    port = gw.start()
+    body = json.dumps({"jsonrpc": "2.0", "id": 3, "method": "tools/call",
+                       "params": {"name": "t", "arguments": {"x": 9}}})
+    result = {}
+
+    def client():
+        try:
+            result["resp"] = _call(port, body)
+        except Exception as e:  # pragma: no cover
+            result["err"] = e
+
+    t = threading.Thread(target=client)
+    t.start()
+    time.sleep(0.15)  # batch dispatched, worker sleeping inside batch_cb
+    t0 = time.time()
+    gw.stop(drain_s=10.0)
+    drained_in = time.time() - t0
+    t.join(timeout=5)
+    assert "err" not in result, result
+    status, data, _ = result["resp"]
+    assert status == 200
+    inner = json.loads(json.loads(data)["result"]["content"][0]["text"])
+    assert inner["echo"] == {"x": 9}
+    assert drained_in < 5.0  # waited for the request, not the full timeout

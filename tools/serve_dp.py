@@ -136,7 +136,7 @@ def main() -> None:
                 break
             time.sleep(0.2)
     finally:
-        gw.stop()
+        gw.stop(drain_s=cfg.server.shutdown_drain_s)
         if dist is not None:
             dist.destroy_process_group()
 
