@@ -93,8 +93,13 @@ std::string http_response(int status, const std::string& body,
   out.reserve(body.size() + 256);
   out += "HTTP/1.1 " + std::to_string(status) + " " + reason + "\r\n";
   out += "Content-Type: application/json\r\n";
-  // security headers (middleware.go:65-86)
+  // security + CORS headers (middleware.go:65-86, 46-62); Mcp-Session-Id
+  // must be exposed for browser MCP clients
   out += "X-Content-Type-Options: nosniff\r\nX-Frame-Options: DENY\r\n";
+  out += "Access-Control-Allow-Origin: *\r\n";
+  out += "Access-Control-Allow-Methods: GET, POST, OPTIONS\r\n";
+  out += "Access-Control-Allow-Headers: Content-Type, Mcp-Session-Id, Authorization\r\n";
+  out += "Access-Control-Expose-Headers: Mcp-Session-Id\r\n";
   if (!session_id.empty()) out += "Mcp-Session-Id: " + session_id + "\r\n";
   out += "Content-Length: " + std::to_string(body.size()) + "\r\n\r\n";
   out += body;

@@ -139,6 +139,27 @@ def test_case_sensitive_filter_falls_back_to_python():
         gw.stop()
 
 
+def test_cors_and_security_headers(gateway):
+    # middleware.go:46-62 + 65-86 parity on the native path: browser MCP
+    # clients need Access-Control-* and the exposed session header
+    gw, port, pipe = gateway
+    body = json.dumps({"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                       "params": {"name": "t", "arguments": {}}})
+    conn = http.client.HTTPConnection("127.0.0.1", port, timeout=10)
+    conn.request("POST", "/", body=body,
+                 headers={"Content-Type": "application/json"})
+    r = conn.getresponse()
+    r.read()
+    assert r.getheader("Access-Control-Allow-Origin") == "*"
+    assert r.getheader("Access-Control-Expose-Headers") == "Mcp-Session-Id"
+    assert r.getheader("X-Frame-Options") == "DENY"
+    assert r.getheader("X-Content-Type-Options") == "nosniff"
+    conn.close()
+    # OPTIONS preflight
+    status, _, _ = _call(port, None, method="OPTIONS")
+    assert status == 200
+
+
 def test_session_reuse(gateway):
     gw, port, pipe = gateway
     body = json.dumps({"jsonrpc": "2.0", "id": 1, "method": "tools/call",
