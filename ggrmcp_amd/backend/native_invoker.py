@@ -60,12 +60,13 @@ class NativeWireClient:
     """Batch unary invoker over one backend target."""
 
     def __init__(self, target: str, connections: int = 8, authority: str = "",
-                 max_inflight: int = 512) -> None:
+                 max_inflight: int = 512, max_resp_bytes: int = 0) -> None:
         mod = load_module()
         self.target = target
         self._cli = mod.Client(target, connections=connections,
                                authority=authority or "localhost",
-                               max_inflight=max_inflight)
+                               max_inflight=max_inflight,
+                               max_resp_bytes=max_resp_bytes)
 
     def invoke_batch(
         self,

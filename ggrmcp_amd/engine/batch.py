@@ -751,9 +751,14 @@ def build_wire_clients(discoverer, config: Config):
         return None
     clients = []
     n_conns = getattr(config.grpc, "native_connections", 8) or 8
+    # reference connection.go:55-57's 4 MB recv cap, enforced per CALL so
+    # one oversized backend response errors alone instead of blowing the
+    # whole batch's response arena
+    max_resp = getattr(config.grpc, "max_recv_msg_bytes", 0) or 0
     try:
         for conn in discoverer.connections:
-            clients.append(NativeWireClient(conn.target, connections=n_conns))
+            clients.append(NativeWireClient(conn.target, connections=n_conns,
+                                            max_resp_bytes=max_resp))
     except Exception as e:  # pragma: no cover - fall back to grpcio
         log.warning("native transport unavailable (%s); using grpcio", e)
         for c in clients:

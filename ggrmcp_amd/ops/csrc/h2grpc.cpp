@@ -5,9 +5,9 @@
 PYBIND11_MODULE(_h2grpc, m) {
   m.doc() = "native gRPC-over-HTTP/2 transport (nghttp2): batch client + bench server";
   py::class_<H2GrpcClient>(m, "Client")
-      .def(py::init<const std::string&, int, const std::string&, int>(),
+      .def(py::init<const std::string&, int, const std::string&, int, size_t>(),
            py::arg("target"), py::arg("connections") = 4, py::arg("authority") = "",
-           py::arg("max_inflight") = 512)
+           py::arg("max_inflight") = 512, py::arg("max_resp_bytes") = 0)
       .def("invoke_stream_batch", &H2GrpcClient::invoke_stream_batch,
            py::arg("paths"), py::arg("payloads"), py::arg("timeout_s") = 30.0,
            py::arg("metadata") = std::vector<std::vector<std::pair<std::string, std::string>>>{})

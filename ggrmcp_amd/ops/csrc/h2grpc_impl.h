@@ -257,6 +257,9 @@ struct Call {
   std::atomic<bool> done{false};
   Clock::time_point deadline;
   int slot = -1;                // caller's index
+  bool no_cap = false;          // server-streaming: whole-stream bytes are
+                                // unbounded by design (cap is per message)
+  bool overflow = false;        // unary response exceeded max_resp
   std::vector<std::pair<std::string, std::string>> metadata;
 };
 
@@ -291,6 +294,8 @@ struct ClientConn {
   std::atomic<bool> stop{false};
   std::atomic<int> inflight{0};
   int max_inflight = 512;  // streams submitted concurrently per connection
+  size_t max_resp = 0;     // unary response byte cap (0 = unlimited);
+                           // reference connection.go:55-57's 4 MB recv cap
   std::string authority;
   bool broken = false;
 };
@@ -326,7 +331,17 @@ static int client_on_data(nghttp2_session*, uint8_t, int32_t stream_id,
                           const uint8_t* data, size_t len, void* user) {
   ClientConn* conn = (ClientConn*)user;
   auto it = conn->live.find(stream_id);
-  if (it != conn->live.end()) it->second.call->response.append((const char*)data, len);
+  if (it != conn->live.end()) {
+    Call* c = it->second.call;
+    if (!c->no_cap && conn->max_resp &&
+        c->response.size() + len > conn->max_resp) {
+      // cap ONE oversized unary response without failing the whole batch
+      c->overflow = true;
+      c->response.clear();
+    } else if (!c->overflow) {
+      c->response.append((const char*)data, len);
+    }
+  }
   return 0;
 }
 
@@ -476,12 +491,14 @@ static void conn_loop(ClientConn* conn) {
 class __attribute__((visibility("default"))) H2GrpcClient {
  public:
   H2GrpcClient(const std::string& target, int n_connections,
-               const std::string& authority, int max_inflight = 512)
+               const std::string& authority, int max_inflight = 512,
+               size_t max_resp_bytes = 0)
       : target_(target) {
     if (n_connections < 1) n_connections = 1;
     for (int i = 0; i < n_connections; ++i) {
       auto conn = std::make_unique<ClientConn>();
       conn->max_inflight = max_inflight < 1 ? 1 : max_inflight;
+      conn->max_resp = max_resp_bytes;
       conn->fd = connect_target(target);
       conn->wake_fd = eventfd(0, EFD_NONBLOCK);
       conn->authority = authority.empty() ? "localhost" : authority;
@@ -541,7 +558,7 @@ class __attribute__((visibility("default"))) H2GrpcClient {
       const std::vector<std::string>& paths, const std::vector<py::bytes>& payloads,
       double timeout_s,
       const std::vector<std::vector<std::pair<std::string, std::string>>>& metadata) {
-    auto raw = invoke_collect(paths, payloads, timeout_s, metadata);
+    auto raw = invoke_collect(paths, payloads, timeout_s, metadata, /*no_cap=*/true);
     std::vector<std::tuple<int, py::bytes, std::vector<uint32_t>, std::string>> out;
     out.reserve(raw.size());
     std::string blob;
@@ -645,6 +662,11 @@ class __attribute__((visibility("default"))) H2GrpcClient {
         out.emplace_back(4, std::string(), "deadline exceeded");
         continue;
       }
+      if (call.overflow) {  // RESOURCE_EXHAUSTED, this slot only
+        out.emplace_back(8, std::string(),
+                         "response exceeds max_recv_msg_bytes");
+        continue;
+      }
       out.emplace_back(call.grpc_status, std::move(call.response),
                        call.grpc_message);
     }
@@ -656,7 +678,8 @@ class __attribute__((visibility("default"))) H2GrpcClient {
   std::vector<std::tuple<int, std::string, std::string>> invoke_collect(
       const std::vector<std::string>& paths, const std::vector<py::bytes>& payloads,
       double timeout_s,
-      const std::vector<std::vector<std::pair<std::string, std::string>>>& metadata) {
+      const std::vector<std::vector<std::pair<std::string, std::string>>>& metadata,
+      bool no_cap = false) {
     size_t n = paths.size();
     if (payloads.size() != n) throw std::runtime_error("paths/payloads mismatch");
     auto batch = std::make_shared<Batch>();
@@ -668,6 +691,7 @@ class __attribute__((visibility("default"))) H2GrpcClient {
       call.path = paths[i];
       call.deadline = deadline;
       call.slot = (int)i;
+      call.no_cap = no_cap;
       if (i < metadata.size()) call.metadata = metadata[i];
       std::string raw = payloads[i];  // needs GIL; held here
       call.payload = grpc_frame(raw);
@@ -697,6 +721,11 @@ class __attribute__((visibility("default"))) H2GrpcClient {
       Call& call = batch->calls[i];
       if (!call.done.load(std::memory_order_acquire)) {
         out.emplace_back(4, std::string(), "deadline exceeded");
+        continue;
+      }
+      if (call.overflow) {  // RESOURCE_EXHAUSTED, this slot only
+        out.emplace_back(8, std::string(),
+                         "response exceeds max_recv_msg_bytes");
         continue;
       }
       out.emplace_back(call.grpc_status, std::move(call.response),

@@ -224,3 +224,46 @@ def test_native_stream_batch_grpcio_server():
         cli.close()
     finally:
         server.stop(grace=None)
+
+
+def test_response_cap_fails_one_slot_only(native_server):
+    """connection.go:55-57's 4 MB recv cap, per CALL: one oversized unary
+    response becomes RESOURCE_EXHAUSTED for that slot; the rest of the
+    batch is unaffected (the engine's batch arena never sees the blob)."""
+    cli = NativeWireClient(native_server, connections=1,
+                           max_resp_bytes=64 * 1024)
+    try:
+        # echo route reflects the request: 200 KB echo > 64 KB cap
+        big = b"\x0a" + b"\x80\x9a\x0c" + b"x" * 200_000  # field1, len 200k
+        out = cli.invoke_batch(
+            ["/bench.EchoService/Echo", HELLO],
+            [big, REQ], 15.0,
+        )
+        assert isinstance(out[0], NativeRpcError)
+        assert out[0].code().name == "RESOURCE_EXHAUSTED"
+        assert out[1] == b"\x0a\x0dHello, world!"  # neighbor unaffected
+    finally:
+        cli.close()
+
+
+def test_streaming_exempt_from_response_cap():
+    """server-streaming responses are unbounded by design (the gRPC cap is
+    per MESSAGE, not per stream): a stream larger than max_resp_bytes
+    still arrives complete."""
+    mod = load_module()
+    srv = mod.Server("127.0.0.1:0")
+    srv.add_route("/bench.EchoService/StreamEcho", "stream_echo")
+    bound = srv.start()
+    cli = NativeWireClient(bound, connections=1, max_resp_bytes=4096)
+    try:
+        # 64 messages of ~200 B >> the 4 KB cap
+        req = b"\x0a\xc8\x01" + b"y" * 200 + b"\x10\x40"  # payload + count=64
+        res = cli._cli.invoke_stream_batch(
+            ["/bench.EchoService/StreamEcho"], [req], 15.0, [[]])
+        status, blob, lens, msg = res[0]
+        assert status == 0, (status, msg)
+        assert len(lens) == 64
+        assert sum(lens) > 4096  # exceeded the unary cap, by design
+    finally:
+        cli.close()
+        srv.stop()
