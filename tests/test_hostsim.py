@@ -299,3 +299,14 @@ def test_map_entry_value_before_key(env):
     dec, outs = e.decode_batch([wire2], [idx], mode=1)
     assert dec[0]["status"] == 0
     assert json.loads(outs[0]) == {"metadata": {"x": ""}}
+
+
+def test_jsonrpc_batch_arrays_rejected(env):
+    # JSON-RPC 2.0 batch arrays: the reference decodes into a single
+    # request struct and errors (handler.go:84); the kernel matches with
+    # E_PARSE -> -32700 envelope
+    e, _, _, _ = env
+    for body in (b"[]", b'[{"jsonrpc":"2.0","id":1,"method":"tools/call",'
+                        b'"params":{"name":"x","arguments":{}}}]'):
+        enc, _ = e.encode_batch([body], mode=0)
+        assert int(enc[0]["status"]) == 1  # E_PARSE
