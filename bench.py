@@ -394,13 +394,16 @@ def main() -> None:
                 result["config"]["engine_stats"] = pipeline.engine.stats.snapshot()
             print(json.dumps(result), flush=True)
     finally:
-        backend_proc.terminate()
-        try:
-            backend_proc.wait(timeout=5)
-        except Exception:
-            backend_proc.kill()
-        if os.path.exists(sock):
-            os.unlink(sock)
+        for bp in backend_procs:
+            bp.terminate()
+        for bp in backend_procs:
+            try:
+                bp.wait(timeout=5)
+            except Exception:
+                bp.kill()
+        for s in socks:
+            if os.path.exists(s):
+                os.unlink(s)
         if dist is not None:
             dist.destroy_process_group()
 

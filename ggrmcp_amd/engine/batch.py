@@ -367,89 +367,37 @@ class GpuPipeline:
 
         # server-streaming slots: fan out unary->stream invokes concurrently;
         # their wire chunks decode in one extra GPU batch below (config 4).
-        # With the native transport, all streams of a backend go out as ONE
-        # batched h2 call (multiplexed streams, GIL-free).
+        # (The native-transport batching variant lives in _native_span; this
+        # path always runs on grpcio threads — _process_span_locked returns
+        # early when wire_clients is set.)
         stream_futs: Dict[int, Any] = {}
         timeout0 = timeout_s if timeout_s is not None else self.config.grpc.request_timeout_s
-        stream_per_be: Dict[int, List[Any]] = {}
         for i in range(n):
             if enc[i]["status"] == E_OK and enc[i]["flags"] & SR_SERVER_STREAMING:
                 mi = self._mi_by_idx[enc[i]["tool_idx"]]
                 hdr = headers[i] if headers else None
-                if self.wire_clients:
-                    be = mi.backend_index if mi.backend_index < len(self.wire_clients) else 0
-                    g = stream_per_be.setdefault(be, [[], [], [], []])
-                    g[0].append(i)
-                    g[1].append(mi.full_method_path)
-                    g[2].append(pbs[i])
-                    g[3].append(list(hdr.items()) if hdr else [])
-                else:
-                    stream_futs[i] = self._invoke_pool.submit(
-                        self.discoverer.invoke_streaming_wire, mi, pbs[i], hdr, timeout0
-                    )
-        for be, g in stream_per_be.items():
-            def run_stream_backend(be=be, g=g):
-                res = g[0], self.wire_clients[be].invoke_stream_batch(
-                    g[1], g[2], timeout0, g[3]
+                stream_futs[i] = self._invoke_pool.submit(
+                    self.discoverer.invoke_streaming_wire, mi, pbs[i], hdr, timeout0
                 )
-                return res
-            fut = self._invoke_pool.submit(run_stream_backend)
-            for k, i in enumerate(g[0]):
-                stream_futs[i] = _SlotOfBatch(fut, k)
 
         t0 = time.perf_counter_ns()
-        timeout = timeout_s if timeout_s is not None else self.config.grpc.request_timeout_s
-        if self.wire_clients:
-            # native transport: one blocking batch call per backend (GIL
-            # released inside); multiple backends dispatch concurrently
-            # (centralized-gateway mode, README.md:129-141)
-            per_be: Dict[int, List[Any]] = {}
-            for i in range(n):
-                if enc[i]["status"] != E_OK or enc[i]["flags"] & SR_SERVER_STREAMING:
-                    continue
-                tool = enc[i]["tool_idx"]
-                mi = self._mi_by_idx[tool]
-                out_idx[i] = self._out_msg_idx[tool]
-                hdr = headers[i] if headers else None
-                be = mi.backend_index if mi.backend_index < len(self.wire_clients) else 0
-                g = per_be.setdefault(be, [[], [], [], []])
-                g[0].append(i)
-                g[1].append(mi.full_method_path)
-                g[2].append(pbs[i])
-                g[3].append(list(hdr.items()) if hdr else [])
-
-            def run_backend(be, g):
-                return g[0], self.wire_clients[be].invoke_batch(g[1], g[2], timeout, g[3])
-
-            batch_futs = [
-                self._invoke_pool.submit(run_backend, be, g)
-                for be, g in per_be.items()
-            ]
-            for fut in batch_futs:
-                slots, res = fut.result()
-                for i, r in zip(slots, res):
-                    if isinstance(r, Exception):
-                        rpc_error[i] = r
-                    else:
-                        resp_wire[i] = r
-        else:
-            futures = {}
-            for i in range(n):
-                if enc[i]["status"] != E_OK:
-                    continue
-                if enc[i]["flags"] & SR_SERVER_STREAMING:
-                    continue  # streaming handled below via host assembly
-                mi = self._mi_by_idx[enc[i]["tool_idx"]]
-                out_idx[i] = self._out_msg_idx[enc[i]["tool_idx"]]
-                hdr = headers[i] if headers else None
-                futures[i] = self._invoke_pool.submit(
-                    self.discoverer.invoke_wire, mi, pbs[i], hdr, timeout_s
-                )
-            for i, fut in futures.items():
-                try:
-                    resp_wire[i] = fut.result()
-                except Exception as e:
-                    rpc_error[i] = e
+        futures = {}
+        for i in range(n):
+            if enc[i]["status"] != E_OK:
+                continue
+            if enc[i]["flags"] & SR_SERVER_STREAMING:
+                continue  # streaming handled below via host assembly
+            mi = self._mi_by_idx[enc[i]["tool_idx"]]
+            out_idx[i] = self._out_msg_idx[enc[i]["tool_idx"]]
+            hdr = headers[i] if headers else None
+            futures[i] = self._invoke_pool.submit(
+                self.discoverer.invoke_wire, mi, pbs[i], hdr, timeout_s
+            )
+        for i, fut in futures.items():
+            try:
+                resp_wire[i] = fut.result()
+            except Exception as e:
+                rpc_error[i] = e
         st.invoke_ns += time.perf_counter_ns() - t0
 
         dec, finals = engine.decode_batch(resp_wire, out_idx, mode=0)
@@ -568,7 +516,8 @@ class GpuPipeline:
         st = engine.stats
         timeout = timeout_s if timeout_s is not None else self.config.grpc.request_timeout_s
         t0 = time.perf_counter_ns()
-        enc_raw, dec_raw, out_view, stream_pbs, rpc_errs = engine._eng.process_span(
+        (enc_raw, dec_raw, out_view, stream_pbs, rpc_errs,
+         resp_wires) = engine._eng.process_span(
             list(bodies),
             list(headers) if headers else None,
             [c._cli for c in self.wire_clients],
@@ -634,18 +583,17 @@ class GpuPipeline:
                 continue
             err = rpc_errs[i]
             rpc_exc = None
-            wire_present = None
+            wire = resp_wires[i]  # delivered bytes when GPU decode failed
             if err is not None:
                 from ..backend.native_invoker import NativeRpcError
 
                 rpc_exc = NativeRpcError(int(err[0]), err[1])
-            elif enc[i]["status"] == E_OK and not (enc[i]["flags"] & SR_SERVER_STREAMING):
-                wire_present = b""  # decode failed on a delivered response
+                wire = None
             out.append(
                 self._host_slot(
                     engine, bodies[i], enc[i],
-                    dec[i] if wire_present is not None else None,
-                    wire_present, rpc_exc,
+                    dec[i] if wire is not None else None,
+                    wire, rpc_exc,
                     headers[i] if headers else None, timeout_s,
                 )
             )
@@ -672,12 +620,18 @@ class GpuPipeline:
                 )
                 return json.dumps(resp.to_dict(), ensure_ascii=False).encode()
 
-        if status in (E_UNSUPPORTED, E_OVERFLOW) or (
-            status == E_OK and dec_r is not None and int(dec_r["status"]) in (E_UNSUPPORTED, E_OVERFLOW, E_PARSE, E_LIMIT)
-        ):
-            # full CPU transcode fallback (counted)
+        if status in (E_UNSUPPORTED, E_OVERFLOW):
+            # encode-side fallback: nothing was invoked yet, so the CPU path
+            # performs the one and only invoke (counted)
             st.host_fallbacks += 1
             return self._cpu_full(body, rid, hdr, timeout_s)
+
+        if status == E_OK and rpc_err is None and wire is not None:
+            # decode-side fallback: the response WAS delivered; transcode the
+            # received bytes on the CPU oracle — never re-invoke, which would
+            # duplicate side effects on non-idempotent methods
+            st.host_fallbacks += 1
+            return self._cpu_decode(enc_r, bytes(wire), rid)
 
         if status == E_OK and rpc_err is not None:
             # gRPC failure -> isError tool result (handler.go:252-259);
@@ -696,6 +650,24 @@ class GpuPipeline:
         code, msg = _STATUS_TO_RPC.get(status, (mcp.INTERNAL_ERROR, "internal error"))
         st.errors += 1
         resp = mcp.JSONRPCResponse(id=rid, error=mcp.RPCError(code, msg))
+        return json.dumps(resp.to_dict(), ensure_ascii=False).encode()
+
+    def _cpu_decode(self, enc_r, wire: bytes, rid) -> bytes:
+        """Transcode an already-received response wire on the CPU (protojson
+        semantics via cpu_ref) and wrap it as the tool-call result envelope.
+        Used when the GPU decode stage rejects a delivered response (invalid
+        UTF-8, over-cap, E_PARSE/E_LIMIT) — the RPC is NOT repeated."""
+        mi = self._mi_by_idx[int(enc_r["tool_idx"])]
+        try:
+            text = self.cpu.pb_to_json(mi.output_descriptor, wire)
+            result = mcp.ToolCallResult(
+                content=[mcp.TextContent(text)], is_error=False
+            )
+            resp = mcp.JSONRPCResponse(id=rid, result=result.to_dict())
+        except Exception as e:
+            resp = mcp.JSONRPCResponse(
+                id=rid, error=mcp.RPCError(mcp.INTERNAL_ERROR, str(e)[:512])
+            )
         return json.dumps(resp.to_dict(), ensure_ascii=False).encode()
 
     def _cpu_full(self, body: bytes, rid, hdr, timeout_s) -> bytes:

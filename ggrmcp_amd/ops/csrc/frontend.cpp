@@ -82,11 +82,12 @@ struct Conn {
 };
 
 std::string http_response(int status, const std::string& body,
-                          const std::string& session_id) {
+                          const std::string& session_id, bool close = false) {
   const char* reason = status == 200 ? "OK"
                        : status == 429 ? "Too Many Requests"
                        : status == 404 ? "Not Found"
                        : status == 413 ? "Payload Too Large"
+                       : status == 501 ? "Not Implemented"
                        : status == 503 ? "Service Unavailable"
                                        : "Error";
   std::string out;
@@ -101,6 +102,7 @@ std::string http_response(int status, const std::string& body,
   out += "Access-Control-Allow-Headers: Content-Type, Mcp-Session-Id, Authorization\r\n";
   out += "Access-Control-Expose-Headers: Mcp-Session-Id\r\n";
   if (!session_id.empty()) out += "Mcp-Session-Id: " + session_id + "\r\n";
+  if (close) out += "Connection: close\r\n";
   out += "Content-Length: " + std::to_string(body.size()) + "\r\n\r\n";
   out += body;
   return out;
@@ -309,11 +311,16 @@ class Frontend {
       drain_completions(re);
       // Backpressure-clocked batching: while all workers are busy,
       // arrivals accumulate; dispatch as soon as a worker is idle (or the
-      // batch is full).  Batches from every reactor coalesce in the
-      // worker, so sharding the reactors doesn't shrink GPU batches.
+      // batch is full).  batch_window_us additionally caps how long a
+      // request may sit in this reactor's pending list before it is queued
+      // to the workers even with every worker busy — queued batches from
+      // every reactor coalesce at worker pop, so sharding the reactors
+      // doesn't shrink GPU batches.
       if (!re->pending.empty() &&
           (re->pending.size() >= (size_t)max_batch_ ||
-           idle_workers_.load(std::memory_order_acquire) > 0)) {
+           idle_workers_.load(std::memory_order_acquire) > 0 ||
+           Clock::now() - re->first_pending >=
+               std::chrono::microseconds(window_us_))) {
         std::lock_guard<std::mutex> lk(batch_mu_);
         batches_.emplace_back(std::move(re->pending));
         re->pending.clear();
@@ -419,6 +426,7 @@ class Frontend {
       size_t clen = 0;
       std::string session;
       bool is_json = false;
+      bool has_te = false;
       std::vector<std::pair<std::string, std::string>> hdrs;
       size_t pos = line_end + 2;
       while (pos < hdr_end) {
@@ -431,6 +439,9 @@ class Frontend {
         std::string value = c->rbuf.substr(vstart, eol - vstart);
         if (name.size() == 14 && iequal(name.data(), "content-length", 14))
           clen = (size_t)strtoull(value.c_str(), nullptr, 10);
+        else if (name.size() == 17 &&
+                 iequal(name.data(), "transfer-encoding", 17))
+          has_te = true;
         else if (name.size() == 12 && iequal(name.data(), "content-type", 12))
           is_json = value.find("application/json") != std::string::npos;
         else if (name.size() == 14 && iequal(name.data(), "mcp-session-id", 14))
@@ -443,6 +454,17 @@ class Frontend {
             hdrs.emplace_back(std::move(name), std::move(value));
         }
         pos = eol + 2;
+      }
+      // chunked (or any Transfer-Encoding) bodies are out of scope; a
+      // body parsed as zero-length would desync the pipelined stream
+      // (request-smuggling-style misparse — ADVICE r1), so refuse and
+      // close instead of misreading the connection.
+      if (has_te) {
+        c->wbuf += http_response(
+            501, "{\"error\":\"transfer-encoding not supported\"}", "", true);
+        c->closing = true;
+        flush_conn(re, id, c);
+        return;
       }
       if (clen > max_body_) {
         c->wbuf += http_response(413, "{\"error\":\"body too large\"}", "");

@@ -19,6 +19,7 @@ PYBIND11_MODULE(_h2grpc, m) {
   py::class_<H2Server>(m, "Server")
       .def(py::init<const std::string&>(), py::arg("target"))
       .def("add_route", &H2Server::add_route, py::arg("path"), py::arg("kind"))
+      .def("request_count", &H2Server::request_count)
       .def("start", &H2Server::start)
       .def("stop", &H2Server::stop);
 }

@@ -569,7 +569,12 @@ class __attribute__((visibility("default"))) H2GrpcClient {
       blob.clear();
       blob.reserve(data.size());
       size_t pos = 0;
+      bool compressed = false;
       while (pos + 5 <= data.size()) {
+        if (data[pos] != 0) {  // compressed-flag frame: not negotiated
+          compressed = true;
+          break;
+        }
         uint32_t len;
         memcpy(&len, data.data() + pos + 1, 4);
         len = ntohl(len);
@@ -578,7 +583,18 @@ class __attribute__((visibility("default"))) H2GrpcClient {
         lens.push_back(len);
         pos += 5 + len;
       }
-      if (status < 0) status = (lens.empty() && data.empty()) ? 2 : 0;
+      if (status == 0 && compressed) {
+        status = 13;  // INTERNAL
+        std::get<2>(r) = "compressed gRPC response frame not supported";
+      }
+      // No decodable grpc-status trailer = malformed close: surface it even
+      // when DATA frames arrived (never silently treat a half-delivered
+      // stream as complete).
+      if (status < 0) {
+        status = 2;  // UNKNOWN
+        if (std::get<2>(r).empty())
+          std::get<2>(r) = "stream closed without grpc-status";
+      }
       out.emplace_back(status, py::bytes(blob), std::move(lens), std::get<2>(r));
     }
     return out;
@@ -586,7 +602,7 @@ class __attribute__((visibility("default"))) H2GrpcClient {
 
   // Batch unary: returns list of (grpc_status, payload, message).
   // status -1 from the wire means "closed without decodable grpc-status":
-  // treated as OK when a full unary message arrived, UNKNOWN otherwise.
+  // always surfaced as UNKNOWN (gRPC requires trailers).
   std::vector<std::tuple<int, py::bytes, std::string>> invoke_batch(
       const std::vector<std::string>& paths, const std::vector<py::bytes>& payloads,
       double timeout_s,
@@ -598,13 +614,25 @@ class __attribute__((visibility("default"))) H2GrpcClient {
       int status = std::get<0>(r);
       const std::string& data = std::get<1>(r);
       std::string payload;
+      bool compressed = false;
       if (data.size() >= 5) {
+        compressed = data[0] != 0;
         uint32_t len;
         memcpy(&len, data.data() + 1, 4);
         len = ntohl(len);
         if (data.size() >= 5 + (size_t)len) payload = data.substr(5, len);
       }
-      if (status < 0) status = payload.empty() && data.empty() ? 2 /*UNKNOWN*/ : 0;
+      if (status < 0) {
+        // malformed close (no grpc-status trailer): an error, even if a
+        // full message arrived — gRPC requires trailers
+        status = 2;  // UNKNOWN
+        if (std::get<2>(r).empty())
+          std::get<2>(r) = "stream closed without grpc-status";
+      } else if (status == 0 && compressed) {
+        status = 13;  // INTERNAL
+        std::get<2>(r) = "compressed gRPC response frame not supported";
+        payload.clear();
+      }
       out.emplace_back(status, py::bytes(payload), std::get<2>(r));
     }
     return out;
@@ -785,6 +813,8 @@ class H2Server {
       sockaddr_un addr{};
       addr.sun_family = AF_UNIX;
       std::string path = target_.substr(5);
+      if (path.size() >= sizeof(addr.sun_path))
+        throw std::runtime_error("uds path too long: " + path);
       unlink(path.c_str());
       memcpy(addr.sun_path, path.c_str(), path.size() + 1);
       if (bind(listen_fd_, (sockaddr*)&addr, sizeof(addr)) != 0)
@@ -822,7 +852,12 @@ class H2Server {
     listen_fd_ = -1;
   }
 
+  // total requests handled (all routes) — lets tests assert exactly-once
+  // invocation semantics of the gateway's fallback paths
+  long request_count() const { return requests_.load(); }
+
   void handle(ServerStream& st) {
+    requests_.fetch_add(1, std::memory_order_relaxed);
     auto it = routes.find(st.path);
     if (it == routes.end()) {
       st.grpc_status = 12;  // UNIMPLEMENTED
@@ -914,6 +949,19 @@ class H2Server {
       if (name == "error") {
         st.grpc_status = 3;  // INVALID_ARGUMENT (mirrors the demo backend)
         st.grpc_message = "name must not be 'error'";
+        return;
+      }
+      if (name == "badutf8") {
+        // hostile-backend fixture: a HelloResponse whose message field
+        // contains invalid UTF-8 (0xFF 0xFE).  The GPU decode stage must
+        // reject it and the gateway must transcode the received bytes on
+        // the host WITHOUT re-invoking (request_count() stays at 1).
+        std::string msg = "bad\xff\xfe";
+        std::string out;
+        out.push_back('\x0a');
+        out.push_back((char)msg.size());
+        out += msg;
+        st.response = grpc_frame(out);
         return;
       }
       st.response = grpc_frame(make_hello_response(name));
@@ -1096,5 +1144,6 @@ class H2Server {
   int listen_fd_ = -1;
   std::thread accept_thread_;
   std::atomic<bool> stop_{true};
+  std::atomic<long> requests_{0};
 };
 

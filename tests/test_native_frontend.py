@@ -489,3 +489,34 @@ def test_graceful_drain():
     inner = json.loads(json.loads(data)["result"]["content"][0]["text"])
     assert inner["echo"] == {"x": 9}
     assert drained_in < 5.0  # waited for the request, not the full timeout
+
+
+def test_chunked_transfer_encoding_rejected(gateway):
+    """Transfer-Encoding bodies are refused with 501 + Connection: close
+    instead of being parsed as zero-length (which would desync pipelined
+    parsing, a request-smuggling-style misparse — ADVICE r1)."""
+    import socket
+
+    gw, port, pipe = gateway
+    s = socket.create_connection(("127.0.0.1", port), timeout=10)
+    # chunked body whose bytes spell a second, smuggled request
+    smuggled = (b"POST / HTTP/1.1\r\nContent-Type: application/json\r\n"
+                b"Content-Length: 2\r\n\r\n{}")
+    req = (b"POST / HTTP/1.1\r\nHost: x\r\nContent-Type: application/json\r\n"
+           b"Transfer-Encoding: chunked\r\n\r\n"
+           + b"%x\r\n" % len(smuggled) + smuggled + b"\r\n0\r\n\r\n")
+    s.sendall(req)
+    data = b""
+    while True:
+        chunk = s.recv(65536)
+        if not chunk:
+            break
+        data += chunk
+    s.close()
+    # exactly ONE response: the 501 refusal; the smuggled bytes were never
+    # parsed as a request (connection closed instead)
+    assert data.count(b"HTTP/1.1 ") == 1, data
+    assert data.startswith(b"HTTP/1.1 501"), data[:80]
+    assert b"Connection: close" in data
+    # and no batch was dispatched for the smuggled payload
+    assert sum(pipe.batches) == 0
