@@ -102,6 +102,11 @@ class SessionConfig:
     rate_limit_per_min: int = 100
     rate_limit_burst: int = 20
     rate_limit_enabled: bool = False  # ref default stack never calls it
+    # /dev/shm path for the C++ shared-memory session table; when set, every
+    # serve_dp rank on the port maps the SAME table, so a session keeps its
+    # rate-limit/block/call-count state no matter which rank the kernel's
+    # SO_REUSEPORT balancing lands its reconnect on.  Empty = per-process.
+    shared_table_path: str = ""
 
 
 @dataclass

@@ -97,7 +97,7 @@ FRONTEND_SO_PATH = OPS_DIR / "_frontend.so"
 
 
 def build_frontend(verbose: bool = True, force: bool = False) -> Path:
-    sources = [CSRC / "frontend.cpp"]
+    sources = [CSRC / "frontend.cpp", CSRC / "session_table.h"]
     if not force and not _needs(FRONTEND_SO_PATH, sources):
         return FRONTEND_SO_PATH
     cxx = os.environ.get("CXX", "g++")

@@ -36,6 +36,8 @@
 #include <sys/socket.h>
 #include <unistd.h>
 
+#include "session_table.h"
+
 #include <algorithm>
 #include <atomic>
 #include <chrono>
@@ -59,9 +61,10 @@ struct PendingReq {
   uint64_t conn_id;
   uint64_t seq;        // per-connection arrival order
   std::string body;
-  std::string session; // Mcp-Session-Id or empty
+  std::string session; // Mcp-Session-Id or empty; effective id after guard
   std::vector<std::pair<std::string, std::string>> headers;
   bool batchable;      // POST "/" with json content
+  int verdict = 0;     // sesstab::V_* when the C++ session guard ran
   std::string method, path;
 };
 
@@ -159,6 +162,17 @@ class Frontend {
   // kernel load-balances accepted connections across listeners.  Call
   // before start().
   void set_reuse_port(bool on) { reuse_port_ = on; }
+
+  // C++ session guard (pkg/session/manager.go semantics, see
+  // session_table.h): the reactor resolves/creates the session and applies
+  // blocked/rate-limit verdicts at parse time, so the batch callback never
+  // runs per-request Python session code.  With a /dev/shm-backed table the
+  // SAME state is shared by every serve_dp rank on the port.  Call before
+  // start(); the table must outlive the frontend (pybind keep_alive).
+  void set_session_table(sesstab::SessionTable* st, bool rate_limit_enabled) {
+    sess_table_ = st;
+    sess_rate_limit_ = rate_limit_enabled;
+  }
 
   int start() {
     listen_fd_ = socket(AF_INET, SOCK_STREAM, 0);
@@ -492,6 +506,13 @@ class Frontend {
       req.method = method;
       req.path = path;
       req.batchable = (method == "POST" && path == "/" && is_json);
+      if (req.batchable && sess_table_) {
+        bool created = false;
+        std::string sid;
+        req.verdict = sess_table_->guard(req.session.data(), req.session.size(),
+                                         sess_rate_limit_, &sid, &created);
+        req.session = std::move(sid);
+      }
       if (!req.batchable && method == "POST" && path == "/") {
         complete(re, id, seq,
                  http_response(
@@ -604,8 +625,9 @@ class Frontend {
       out.reserve(batch.size());
       {
         py::gil_scoped_acquire gil;
-        py::list bodies, sessions, headers;
+        py::list bodies, sessions, headers, verdicts;
         std::vector<size_t> batch_idx;
+        bool guarded = sess_table_ != nullptr;
         for (size_t i = 0; i < batch.size(); ++i) {
           if (!batch[i].batchable) continue;
           batch_idx.push_back(i);
@@ -613,6 +635,7 @@ class Frontend {
           sessions.append(batch[i].session.empty()
                               ? py::object(py::none())
                               : py::object(py::str(batch[i].session)));
+          if (guarded) verdicts.append(py::int_(batch[i].verdict));
           py::dict h;
           for (auto& kv : batch[i].headers)
             h[py::str(kv.first)] = py::str(kv.second);
@@ -620,7 +643,9 @@ class Frontend {
         }
         if (py::len(bodies) > 0) {
           try {
-            py::list res = batch_cb_(bodies, sessions, headers);
+            py::list res = batch_cb_(bodies, sessions, headers,
+                                     guarded ? py::object(verdicts)
+                                             : py::object(py::none()));
             for (size_t k = 0; k < batch_idx.size(); ++k) {
               py::tuple t = res[k].cast<py::tuple>();
               std::string body = t[0].cast<std::string>();
@@ -683,6 +708,8 @@ class Frontend {
   int port_;
   py::function batch_cb_, slow_cb_;
   bool reuse_port_ = false;
+  sesstab::SessionTable* sess_table_ = nullptr;
+  bool sess_rate_limit_ = true;
   bool hfilter_on_ = false, hfwd_enabled_ = true, hfwd_all_ = false;
   std::unordered_set<std::string> hallow_, hblock_;
   int window_us_, max_batch_;
@@ -833,6 +860,69 @@ static py::tuple bench_client(const std::string& host, int port, int sessions,
 
 PYBIND11_MODULE(_frontend, m) {
   m.doc() = "native HTTP/1.1 batch ingestion front end for the MCP gateway";
+  py::class_<sesstab::SessionTable>(m, "SessionTable")
+      .def(py::init<uint64_t, double, const std::string&, uint32_t, uint32_t>(),
+           py::arg("capacity") = 16384, py::arg("ttl_s") = 1800.0,
+           py::arg("path") = std::string(), py::arg("rate_per_min") = 100,
+           py::arg("rate_burst") = 20,
+           "Shared-memory session table (manager.go semantics).  path='' -> "
+           "anonymous (single process); a /dev/shm path is shared by every "
+           "rank that opens it (serve_dp session affinity).")
+      .def("guard",
+           [](sesstab::SessionTable& t, const std::string& id, bool rate_limit) {
+             std::string out;
+             bool created;
+             int v = t.guard(id.data(), id.size(), rate_limit, &out, &created);
+             return py::make_tuple(py::str(out), v, created);
+           },
+           py::arg("session_id") = std::string(), py::arg("rate_limit") = true,
+           "-> (effective_id, verdict 0=ok/1=blocked/2=rate-limited, created)")
+      .def("get_or_create",
+           [](sesstab::SessionTable& t, const std::string& id) {
+             return t.get_or_create(id.data(), id.size());
+           },
+           py::arg("session_id") = std::string())
+      .def("block",
+           [](sesstab::SessionTable& t, const std::string& id) {
+             return t.set_blocked(id.data(), id.size(), true);
+           })
+      .def("unblock",
+           [](sesstab::SessionTable& t, const std::string& id) {
+             return t.set_blocked(id.data(), id.size(), false);
+           })
+      .def("remove",
+           [](sesstab::SessionTable& t, const std::string& id) {
+             return t.remove(id.data(), id.size());
+           })
+      .def("info",
+           [](sesstab::SessionTable& t, const std::string& id) -> py::object {
+             int64_t created_us, last_us;
+             uint64_t calls;
+             bool blocked;
+             if (!t.info(id.data(), id.size(), &created_us, &last_us, &calls,
+                         &blocked))
+               return py::none();
+             py::dict d;
+             d["id"] = id;
+             d["createdAt"] = created_us / 1e6;
+             d["lastAccessed"] = last_us / 1e6;
+             d["callCount"] = (long long)calls;
+             d["isBlocked"] = blocked;
+             return d;
+           })
+      .def("stats",
+           [](sesstab::SessionTable& t) {
+             uint64_t active, calls, blocked, created;
+             t.stats(&active, &calls, &blocked, &created);
+             py::dict d;
+             d["activeSessions"] = (long long)active;
+             d["totalCalls"] = (long long)calls;
+             d["blockedSessions"] = (long long)blocked;
+             d["createdTotal"] = (long long)created;
+             d["maxSessions"] = (long long)t.capacity();
+             return d;
+           })
+      .def_property_readonly("capacity", &sesstab::SessionTable::capacity);
   py::class_<Frontend>(m, "Frontend")
       .def(py::init<const std::string&, int, py::function, py::function, int,
                     int, size_t, double, double, int, int>(),
@@ -846,6 +936,11 @@ PYBIND11_MODULE(_frontend, m) {
            "returns requests still unanswered at timeout")
       .def("set_reuse_port", &Frontend::set_reuse_port, py::arg("on"),
            "SO_REUSEPORT: N gateway ranks share one port (call before start)")
+      .def("set_session_table", &Frontend::set_session_table,
+           py::arg("table"), py::arg("rate_limit_enabled") = true,
+           py::keep_alive<1, 2>(),
+           "run the per-request session guard in the C++ reactor "
+           "(call before start)")
       .def("set_header_filter", &Frontend::set_header_filter,
            py::arg("enabled"), py::arg("forward_all"), py::arg("allow"),
            py::arg("block"),

@@ -106,6 +106,55 @@ def load_module():
     return importlib.import_module("_frontend")
 
 
+class _SidContext:
+    """Minimal SessionContext stand-in for slow paths that only need .id."""
+
+    __slots__ = ("id",)
+
+    def __init__(self, sid: str) -> None:
+        self.id = sid
+
+
+class NativeSessionStore:
+    """SessionManager facade over the C++ shared-memory SessionTable
+    (ops/csrc/session_table.h; reference semantics pkg/session/manager.go).
+
+    The serving hot path never calls into this — the guard runs inside the
+    C++ reactor at parse time.  This facade gives the slow paths (GET /
+    session issuance, /metrics, admin block/unblock) the same view of the
+    same state.  Divergence from the Python SessionManager: the table keeps
+    counters/flags only, not header snapshots (headers travel per-request
+    on this path), so session_info reports headerCount 0."""
+
+    def __init__(self, table) -> None:
+        self.table = table
+
+    def guard(self, session_id, headers=None, rate_limit: bool = True):
+        sid, verdict, _created = self.table.guard(session_id or "", rate_limit)
+        return _SidContext(sid), verdict
+
+    def get_or_create(self, session_id, headers=None) -> _SidContext:
+        return _SidContext(self.table.get_or_create(session_id or ""))
+
+    def block(self, session_id: str) -> bool:
+        return self.table.block(session_id)
+
+    def unblock(self, session_id: str) -> bool:
+        return self.table.unblock(session_id)
+
+    def remove(self, session_id: str) -> bool:
+        return self.table.remove(session_id)
+
+    def session_info(self, session_id: str):
+        info = self.table.info(session_id)
+        if info is not None:
+            info["headerCount"] = 0
+        return info
+
+    def stats(self):
+        return self.table.stats()
+
+
 class NativeHTTPGateway:
     """Drop-in serving front end over one or more GpuPipelines.
 
@@ -130,7 +179,6 @@ class NativeHTTPGateway:
         self.pipelines = list(pipeline) if isinstance(pipeline, (list, tuple)) else [pipeline]
         self.pipeline = self.pipelines[0]
         self.discoverer = discoverer
-        self.sessions = sessions or SessionManager()
         self.tools = tool_builder or MCPToolBuilder()
         self.headers = header_filter or HeaderFilter.from_config(
             self.config.header_forwarding
@@ -149,6 +197,28 @@ class NativeHTTPGateway:
             self._shard_pool = None
         mod = load_module()
         srv_cfg = self.config.server
+        sess_cfg = self.config.session
+        # session state: C++ shared-memory table (reactor-side guard, shared
+        # across serve_dp ranks when shared_table_path is set) unless the
+        # caller injected a Python SessionManager explicitly
+        self._cxx_sessions = sessions is None and hasattr(mod, "SessionTable")
+        if self._cxx_sessions:
+            self._sess_table = mod.SessionTable(
+                capacity=max(1024, 2 * sess_cfg.max_sessions),
+                ttl_s=sess_cfg.ttl_s,
+                path=sess_cfg.shared_table_path,
+                rate_per_min=sess_cfg.rate_limit_per_min,
+                rate_burst=sess_cfg.rate_limit_burst,
+            )
+            self.sessions = NativeSessionStore(self._sess_table)
+        else:
+            self._sess_table = None
+            self.sessions = sessions or SessionManager(
+                ttl_s=sess_cfg.ttl_s,
+                max_sessions=sess_cfg.max_sessions,
+                rate_limit_per_min=sess_cfg.rate_limit_per_min,
+                rate_limit_burst=sess_cfg.rate_limit_burst,
+            )
         # case-insensitive filters (the default) run inside the C++ parser:
         # the batch callback then receives pre-filtered lowercase names and
         # skips the per-request Python dict rebuild.  Case-sensitive configs
@@ -170,6 +240,10 @@ class NativeHTTPGateway:
         )
         if getattr(srv_cfg, "reuse_port", False) and hasattr(self._fe, "set_reuse_port"):
             self._fe.set_reuse_port(True)
+        if self._sess_table is not None:
+            self._fe.set_session_table(
+                self._sess_table, sess_cfg.rate_limit_enabled
+            )
         if self._cxx_header_filter:
             self._fe.set_header_filter(
                 self.headers.enabled,
@@ -195,26 +269,49 @@ class NativeHTTPGateway:
     # ---- hot path: one call per collected batch -----------------------------
 
     def _batch_cb(self, bodies: List[bytes], session_ids: List[Optional[str]],
-                  headers: List[Dict[str, str]]) -> List[Tuple[bytes, str]]:
+                  headers: List[Dict[str, str]],
+                  verdicts: Optional[List[int]] = None) -> List[Tuple[bytes, str]]:
         n = len(bodies)
-        sids: List[str] = []
-        fwd_headers: List[Optional[Dict[str, str]]] = []
         rejected: Dict[int, bytes] = {}
-        rl_on = self.config.session.rate_limit_enabled
-        for i in range(n):
-            # per-session guards (handler.go:219-226 parity) in one
-            # manager-lock cycle (see SessionManager.guard)
-            sess, verdict = self.sessions.guard(session_ids[i], headers[i],
-                                                rate_limit=rl_on)
-            sids.append(sess.id)
-            if verdict == 1:
-                rejected[i] = self._session_error(bodies[i], "session is blocked")
-                continue
-            if verdict == 2:
-                rejected[i] = self._session_error(bodies[i], "session rate limit exceeded")
-                continue
-            fwd_headers.append(headers[i] if self._cxx_header_filter
-                               else self.headers.filter_headers(headers[i]))
+        if verdicts is not None:
+            # C++ reactor already ran the session guard (session_table.h):
+            # session_ids are final and verdicts carry blocked/rate-limit
+            # decisions — the common all-allowed batch does ZERO per-request
+            # Python session work here
+            sids: List[str] = session_ids  # type: ignore[assignment]
+            if self._cxx_header_filter:
+                fwd_headers: List[Optional[Dict[str, str]]] = headers
+            else:
+                fwd_headers = [self.headers.filter_headers(h) for h in headers]
+            for i, v in enumerate(verdicts):
+                if v == 1:
+                    rejected[i] = self._session_error(bodies[i], "session is blocked")
+                elif v == 2:
+                    rejected[i] = self._session_error(
+                        bodies[i], "session rate limit exceeded")
+            if rejected and not self._cxx_header_filter:
+                fwd_headers = [h for i, h in enumerate(fwd_headers)
+                               if i not in rejected]
+            elif rejected:
+                fwd_headers = [headers[i] for i in range(n) if i not in rejected]
+        else:
+            sids = []
+            fwd_headers = []
+            rl_on = self.config.session.rate_limit_enabled
+            for i in range(n):
+                # per-session guards (handler.go:219-226 parity) in one
+                # manager-lock cycle (see SessionManager.guard)
+                sess, verdict = self.sessions.guard(session_ids[i], headers[i],
+                                                    rate_limit=rl_on)
+                sids.append(sess.id)
+                if verdict == 1:
+                    rejected[i] = self._session_error(bodies[i], "session is blocked")
+                    continue
+                if verdict == 2:
+                    rejected[i] = self._session_error(bodies[i], "session rate limit exceeded")
+                    continue
+                fwd_headers.append(headers[i] if self._cxx_header_filter
+                                   else self.headers.filter_headers(headers[i]))
         timeout = self.config.grpc.request_timeout_s
         if rejected:
             live_idx = [i for i in range(n) if i not in rejected]

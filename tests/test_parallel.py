@@ -193,6 +193,44 @@ def test_serve_dp_world2_shared_port(tmp_path, use_descriptor):
                 _t.sleep(0.5)
             assert ok >= 6, f"only {ok} round-trips; output:\n" + (
                 proc.stdout.read() if proc.poll() is not None else "")
+
+            if use_descriptor:
+                # cross-rank session affinity: the same
+                # Mcp-Session-Id over FRESH connections (kernel may land each
+                # on either rank) must hit ONE shared session — the
+                # /dev/shm-backed C++ table both ranks map.
+                def post(sid=None):
+                    hdrs = {"Content-Type": "application/json"}
+                    if sid:
+                        hdrs["Mcp-Session-Id"] = sid
+                    conn = http.client.HTTPConnection("127.0.0.1", http_port,
+                                                      timeout=5)
+                    conn.request("POST", "/", body=body, headers=hdrs)
+                    r = conn.getresponse()
+                    r.read()
+                    out_sid = r.getheader("Mcp-Session-Id")
+                    conn.close()
+                    return out_sid
+
+                def metrics():
+                    conn = http.client.HTTPConnection("127.0.0.1", http_port,
+                                                      timeout=5)
+                    conn.request("GET", "/metrics")
+                    r = conn.getresponse()
+                    data = json.loads(r.read())
+                    conn.close()
+                    return data["sessions"]
+
+                m0 = metrics()
+                sid = post()
+                assert sid
+                for _ in range(5):
+                    assert post(sid) == sid
+                m1 = metrics()
+                # 6 calls booked on ONE new session, regardless of which
+                # rank each fresh connection landed on
+                assert m1["totalCalls"] - m0["totalCalls"] == 6
+                assert m1["activeSessions"] - m0["activeSessions"] == 1
         finally:
             proc.terminate()
             try:

@@ -18,13 +18,13 @@ Each rank reads RANK/LOCAL_RANK/WORLD_SIZE from the environment.  With
 --no-gpu (or no CUDA) it serves the CPU reference pipeline — which is how
 the world-2 CPU test exercises this file end to end.
 
-Session-affinity caveat: SO_REUSEPORT balances CONNECTIONS, and session
-state (counters, blocks, rate windows) is per-rank, exactly like the
-reference's per-process in-memory sessions.  A client that keeps its
-HTTP connection (the normal MCP shape) stays on one rank; a client that
-reconnects mid-session may land on another rank and start a fresh
-session context there.  Front with an L4 source-hash balancer if strict
-cross-reconnect affinity matters.
+Session affinity: SO_REUSEPORT balances CONNECTIONS, but session state
+(counters, blocks, rate windows) lives in ONE /dev/shm-backed C++ table
+mapped by every rank (ops/csrc/session_table.h, wired below via
+config.session.shared_table_path), so a client that reconnects and lands
+on a different rank keeps its session context — the round-1 affinity
+caveat is closed.  Pass --no-shared-sessions to
+revert to per-rank state.
 """
 
 from __future__ import annotations
@@ -51,6 +51,11 @@ def parse_args():
     ap.add_argument("--uds", default="", help="backend unix socket")
     ap.add_argument("--descriptor", default="", help=".binpb FileDescriptorSet")
     ap.add_argument("--no-gpu", action="store_true")
+    ap.add_argument("--no-shared-sessions", action="store_true",
+                    help="per-rank session state (round-1 behavior)")
+    ap.add_argument("--session-table", default="",
+                    help="explicit shared session table path "
+                         "(default: /dev/shm/ggrmcp_sessions_<port>)")
     ap.add_argument("--run-seconds", type=float, default=0.0,
                     help="exit after N seconds (0 = run until SIGTERM)")
     return ap.parse_args()
@@ -73,6 +78,12 @@ def main() -> None:
     cfg.gpu.enabled = use_gpu
     cfg.server.http_port = args.port
     cfg.server.reuse_port = world > 1  # all ranks share the port
+    if world > 1 and not args.no_shared_sessions:
+        # one session table for the whole node: reconnects keep their
+        # session no matter which rank SO_REUSEPORT lands them on
+        cfg.session.shared_table_path = args.session_table or (
+            f"/dev/shm/ggrmcp_sessions_{args.port}"
+        )
 
     dist = None
     shard_group = None
@@ -91,6 +102,12 @@ def main() -> None:
         shard_group = ShardGroup.attach(
             dist, device=dev if (use_gpu and backend == "nccl") else None
         )
+        if cfg.session.shared_table_path:
+            # rank 0 clears any stale table from a previous run BEFORE any
+            # rank maps it (barrier orders the unlink against the opens)
+            if rank == 0 and os.path.exists(cfg.session.shared_table_path):
+                os.unlink(cfg.session.shared_table_path)
+            dist.barrier()
 
     discoverer = ServiceDiscoverer(cfg)
     # rank 0 discovers (descriptor file, or live reflection) and broadcasts
