@@ -43,7 +43,7 @@ def _common_flags():
 
 def build_jsonproto(verbose: bool = True, force: bool = False) -> Path:
     sources = [CSRC / "engine.cpp", CSRC / "json2pb.hip", CSRC / "pb2json.hip",
-               CSRC / "common.h", CSRC / "h2grpc_impl.h"]
+               CSRC / "common.h", CSRC / "h2grpc_impl.h", CSRC / "span_api.h"]
     if not force and not _needs(SO_PATH, sources):
         return SO_PATH
     hipcc = os.environ.get("HIPCC", "hipcc")
@@ -97,7 +97,8 @@ FRONTEND_SO_PATH = OPS_DIR / "_frontend.so"
 
 
 def build_frontend(verbose: bool = True, force: bool = False) -> Path:
-    sources = [CSRC / "frontend.cpp", CSRC / "session_table.h"]
+    sources = [CSRC / "frontend.cpp", CSRC / "session_table.h",
+               CSRC / "span_api.h"]
     if not force and not _needs(FRONTEND_SO_PATH, sources):
         return FRONTEND_SO_PATH
     cxx = os.environ.get("CXX", "g++")

@@ -33,6 +33,7 @@
 #include "h2grpc_impl.h"
 #include "json2pb.hip"
 #include "pb2json.hip"
+#include "span_api.h"
 
 namespace py = pybind11;
 
@@ -109,9 +110,58 @@ inline size_t final_cap(size_t wire_len) {
   return (wire_len * 16 + 2048 + 15) & ~(size_t)15;
 }
 
+// JSON string escaping for C++-assembled error envelopes (error details may
+// contain quotes/control bytes)
+void json_escape_append(std::string& out, const char* s, size_t n) {
+  static const char* hex = "0123456789abcdef";
+  for (size_t i = 0; i < n; ++i) {
+    unsigned char c = (unsigned char)s[i];
+    if (c == '"' || c == '\\') {
+      out.push_back('\\');
+      out.push_back((char)c);
+    } else if (c == '\n') {
+      out += "\\n";
+    } else if (c == '\r') {
+      out += "\\r";
+    } else if (c == '\t') {
+      out += "\\t";
+    } else if (c < 0x20) {
+      out += "\\u00";
+      out.push_back(hex[c >> 4]);
+      out.push_back(hex[c & 15]);
+    } else {
+      out.push_back((char)c);
+    }
+  }
+}
+
+// grpc status code -> canonical name (mirrors backend/native_invoker.py
+// _CODE_NAMES so native and Python error texts match)
+const char* grpc_status_name(int s) {
+  static const char* names[] = {
+      "OK", "CANCELLED", "UNKNOWN", "INVALID_ARGUMENT", "DEADLINE_EXCEEDED",
+      "NOT_FOUND", "ALREADY_EXISTS", "PERMISSION_DENIED", "RESOURCE_EXHAUSTED",
+      "FAILED_PRECONDITION", "ABORTED", "OUT_OF_RANGE", "UNIMPLEMENTED",
+      "INTERNAL", "UNAVAILABLE", "DATA_LOSS", "UNAUTHENTICATED"};
+  return (s >= 0 && s < 17) ? names[s] : "CODE_?";
+}
+
+// kernel status -> JSON-RPC (code, message); mirrors engine/batch.py
+// _STATUS_TO_RPC (reference error mapping handler.go:117-127)
+void status_to_rpc(int status, int* code, const char** msg) {
+  switch (status) {
+    case E_PARSE: *code = -32700; *msg = "parse error"; break;
+    case E_INVALID_REQUEST: *code = -32600; *msg = "invalid request"; break;
+    case E_METHOD_NOT_FOUND: *code = -32601; *msg = "tool not found"; break;
+    case E_INVALID_PARAMS: *code = -32602; *msg = "invalid params"; break;
+    case E_LIMIT: *code = -32602; *msg = "argument limits exceeded"; break;
+    default: *code = -32603; *msg = "internal error"; break;
+  }
+}
+
 }  // namespace
 
-class Engine {
+class Engine : public spanapi::ISpanExecutor {
  public:
   Engine(int device, py::bytes msg_table, py::bytes field_table,
          py::bytes enum_table, py::bytes enum_values, py::bytes tool_table,
@@ -191,6 +241,9 @@ class Engine {
     h_off_.alloc(offs * 3);
     h_tight_.alloc(offs);
     h_aux_.alloc((size_t)max_batch * sizeof(int32_t) * 2);
+    // JSON-RPC id tokens on the host: the native span executor assembles
+    // error envelopes in C++ and needs the ids the encode kernel captured
+    h_id_.alloc((size_t)max_batch * ID_SLOT_BYTES);
   }
 
   ~Engine() {
@@ -566,6 +619,243 @@ class Engine {
                           rpc_errors, resp_wires);
   }
 
+  // ---- fully-native serving span (span_api.h) -----------------------------
+  // The GIL-free twin of process_span for the C++ HTTP frontend: the whole
+  // tools/call hot path — encode kernel, gRPC invoke, decode kernel, and
+  // the response envelopes — with Python needed only for the slot kinds
+  // the span reports back (streaming / non-tools-call / host fallbacks).
+  // Replaces the reference's per-request broker end-to-end
+  // (handler.go:81-139 + reflection.go:333-391).
+  bool run_span(const spanapi::SpanIn& in, spanapi::SpanOut* out,
+                std::string* err) override {
+    using namespace spanapi;
+    try {
+      int n = (int)in.n;
+      if (n <= 0 || n > max_batch_) {
+        *err = "bad batch size";
+        return false;
+      }
+      out->slots.assign(n, SlotOut());
+      // stage bodies into pinned memory
+      uint32_t* in_off = (uint32_t*)h_off_.p;
+      uint32_t* pb_off = in_off + (n + 1);
+      size_t acc = 0, pacc = 0;
+      uint8_t* dst = (uint8_t*)h_in_.p;
+      for (int i = 0; i < n; ++i) {
+        size_t len = in.body_lens[i];
+        if (acc + len > h_in_.n) {
+          *err = "input exceeds cap_in";
+          return false;
+        }
+        in_off[i] = (uint32_t)acc;
+        pb_off[i] = (uint32_t)pacc;
+        if (len) std::memcpy(dst + acc, in.bodies[i], len);
+        acc += len;
+        pacc += pb_cap(len);
+      }
+      in_off[n] = (uint32_t)acc;
+      pb_off[n] = (uint32_t)pacc;
+      if (pacc > d_pb_.n) {
+        *err = "pb cap exceeded";
+        return false;
+      }
+      Limits lim{in.max_depth, in.max_string, in.max_args,
+                 (uint32_t)in.enforce};
+      auto t0 = std::chrono::steady_clock::now();
+      run_encode_device(n, acc, pacc, false, lim, 0);
+      auto t1 = std::chrono::steady_clock::now();
+      out->enc_ms =
+          std::chrono::duration<double, std::milli>(t1 - t0).count();
+      out->enc_gpu_ms = last_enc_gpu_ms_;
+
+      SlotResult* rs = (SlotResult*)h_results_.p;
+      uint8_t* pb = (uint8_t*)h_pb_.p;
+
+      // classify slots + route OK unary calls per backend
+      size_t nb = in.n_clients;
+      std::vector<std::vector<H2GrpcClient::RawCall>> per_be(nb);
+      std::vector<std::vector<int>> slots_be(nb);
+      for (int i = 0; i < n; ++i) {
+        SlotOut& so = out->slots[i];
+        so.tool_idx = rs[i].tool_idx;
+        int st = rs[i].status;
+        if (st == E_OK && (rs[i].flags & SR_SERVER_STREAMING)) {
+          // request wire OUT now: decode compaction reuses h_pb_
+          so.kind = K_PY_STREAM;
+          so.aux.assign((const char*)pb + rs[i].pb_off, rs[i].pb_len);
+          continue;
+        }
+        if (st == E_OK) {
+          int tool = rs[i].tool_idx;
+          if (tool < 0 || tool >= (int)tool_paths_.size() || nb == 0) {
+            so.kind = K_PY_ENC_FALLBACK;
+            continue;
+          }
+          size_t be = tool < (int)tool_backend_.size()
+                          ? (size_t)tool_backend_[tool] % nb
+                          : 0;
+          const std::string& path = tool_paths_[tool];
+          per_be[be].push_back(H2GrpcClient::RawCall{
+              path.data(), path.size(), pb + rs[i].pb_off, rs[i].pb_len,
+              (in.metas && in.metas[i] && !in.metas[i]->empty())
+                  ? in.metas[i]
+                  : nullptr});
+          slots_be[be].push_back(i);
+          so.kind = K_FINAL;  // provisional until decode verdict
+          continue;
+        }
+        if (st == E_NOT_TOOLCALL) {
+          so.kind = K_PY_NOT_TOOLCALL;
+          continue;
+        }
+        if (st == E_UNSUPPORTED || st == E_OVERFLOW) {
+          so.kind = K_PY_ENC_FALLBACK;
+          continue;
+        }
+        so.kind = K_ERR_FINAL;
+        int code;
+        const char* msg;
+        status_to_rpc(st, &code, &msg);
+        build_rpc_error(so.aux, i, code, msg);
+      }
+
+      // invoke (pure C++; the h2 clients block until their batch resolves)
+      std::vector<std::vector<std::tuple<int, std::string, std::string>>>
+          res_be(nb);
+      for (size_t b = 0; b < nb; ++b)
+        if (!per_be[b].empty())
+          res_be[b] = ((H2GrpcClient*)in.clients[b])
+                          ->invoke_raw(per_be[b], in.timeout_s);
+      auto t2 = std::chrono::steady_clock::now();
+      out->inv_ms =
+          std::chrono::duration<double, std::milli>(t2 - t1).count();
+
+      // parse + stage responses for decode (frame rules match process_span)
+      uint32_t* resp_off = (uint32_t*)h_off_.p;
+      uint32_t* final_off = resp_off + (n + 1);
+      uint32_t* scratch_off = resp_off + 2 * (n + 1);
+      int32_t* h_aux = (int32_t*)h_aux_.p;
+      std::vector<const std::string*> resp_ptr(n, nullptr);
+      for (size_t b = 0; b < nb; ++b) {
+        for (size_t k = 0; k < slots_be[b].size(); ++k) {
+          int i = slots_be[b][k];
+          auto& r = res_be[b][k];
+          int status = std::get<0>(r);
+          const std::string& data = std::get<1>(r);
+          size_t plen = 0, poff = 0;
+          bool compressed = false, truncated = false;
+          if (!data.empty()) {
+            if (data.size() >= 5) {
+              compressed = data[0] != 0;
+              uint32_t len;
+              memcpy(&len, data.data() + 1, 4);
+              len = ntohl(len);
+              if (data.size() >= 5 + (size_t)len) {
+                poff = 5;
+                plen = len;
+              } else {
+                truncated = true;
+              }
+            } else {
+              truncated = true;
+            }
+          }
+          const char* local_err = nullptr;
+          if (status < 0) {
+            status = 2;
+            local_err = std::get<2>(r).empty()
+                            ? "stream closed without grpc-status"
+                            : std::get<2>(r).c_str();
+          } else if (status == 0 && compressed) {
+            status = 13;
+            local_err = "compressed gRPC response frame not supported";
+          } else if (status == 0 && truncated) {
+            status = 13;
+            local_err = "truncated gRPC frame";
+          }
+          if (status == 0) {
+            resp_ptr[i] = &std::get<1>(res_be[b][k]);
+            rs[i].err_pos = (uint32_t)poff;  // reuse: payload offset
+            rs[i].aux = (int32_t)plen;       // reuse: payload length
+          } else {
+            // gRPC failure -> isError tool result, HTTP 200
+            // (handler.go:252-259 semantics)
+            SlotOut& so = out->slots[i];
+            so.kind = K_ERR_FINAL;
+            build_grpc_error(so.aux, i, status,
+                             local_err ? local_err
+                                       : std::get<2>(r).c_str());
+          }
+        }
+      }
+      size_t racc = 0, sacc = 0, facc = 0;
+      uint8_t* rdst = (uint8_t*)h_resp_.p;
+      for (int i = 0; i < n; ++i) {
+        size_t len = resp_ptr[i] ? (size_t)rs[i].aux : 0;
+        // a response that would overflow an arena falls back PER SLOT with
+        // its delivered wire: failing the whole span here would force the
+        // caller to either drop or re-invoke already-invoked slots
+        if (resp_ptr[i] &&
+            (racc + len > h_resp_.n || sacc + scratch_cap(len) > d_scratch_.n ||
+             facc + final_cap(len) > d_final_.n)) {
+          SlotOut& so = out->slots[i];
+          so.kind = K_PY_DEC_FALLBACK;
+          so.aux.assign(resp_ptr[i]->data() + rs[i].err_pos, len);
+          resp_ptr[i] = nullptr;
+          len = 0;
+        }
+        resp_off[i] = (uint32_t)racc;
+        scratch_off[i] = (uint32_t)sacc;
+        final_off[i] = (uint32_t)facc;
+        h_aux[i] = (resp_ptr[i] &&
+                    tool_out_msg_.size() > (size_t)rs[i].tool_idx)
+                       ? tool_out_msg_[rs[i].tool_idx]
+                       : 0;
+        h_aux[n + i] = resp_ptr[i] ? 0 : 1;  // skip slots with no response
+        if (len)
+          std::memcpy(rdst + racc, resp_ptr[i]->data() + rs[i].err_pos, len);
+        racc += len;
+        sacc += scratch_cap(len);
+        facc += final_cap(len);
+      }
+      resp_off[n] = (uint32_t)racc;
+      scratch_off[n] = (uint32_t)sacc;
+      final_off[n] = (uint32_t)facc;
+      run_decode_device(n, racc, facc, true, 0);
+      auto t3 = std::chrono::steady_clock::now();
+      out->dec_ms =
+          std::chrono::duration<double, std::milli>(t3 - t2).count();
+      out->dec_gpu_ms = last_dec_gpu_ms_;
+
+      DecodeResult* dr = (DecodeResult*)h_dec_results_.p;
+      out->blob = compact_used_ ? (const uint8_t*)h_pb_.p
+                                : (const uint8_t*)h_final_.p;
+      out->blob_len = compact_used_ ? compact_bytes_ : last_final_bytes_;
+      for (int i = 0; i < n; ++i) {
+        SlotOut& so = out->slots[i];
+        if (so.kind != K_FINAL) continue;
+        if (resp_ptr[i] && dr[i].status == E_OK && dr[i].out_len > 0) {
+          so.off = dr[i].out_off;
+          so.len = dr[i].out_len;
+        } else if (resp_ptr[i]) {
+          // decode rejected a DELIVERED response: hand the received wire
+          // to the host transcoder — never re-invoke
+          so.kind = K_PY_DEC_FALLBACK;
+          so.aux.assign(resp_ptr[i]->data() + rs[i].err_pos,
+                        (size_t)rs[i].aux);
+        } else {
+          // routed but no response tuple (client size mismatch): internal
+          so.kind = K_ERR_FINAL;
+          build_rpc_error(so.aux, i, -32603, "internal error");
+        }
+      }
+      return true;
+    } catch (const std::exception& e) {
+      *err = e.what();
+      return false;
+    }
+  }
+
   // (encode_ms, invoke_ms, decode_ms) of the LAST process_span call;
   // callers serialize spans per engine, so no further synchronization
   py::tuple last_stage_ms() const {
@@ -583,6 +873,45 @@ class Engine {
   int max_batch() const { return max_batch_; }
 
  private:
+  // JSON-RPC id token of a slot (raw JSON captured by the encode kernel;
+  // "null" when absent/unparsed — matches the Python path's best effort)
+  void append_id(std::string& out, int slot) {
+    const SlotResult* rs = (const SlotResult*)h_results_.p;
+    uint32_t len = rs[slot].id_len;
+    if ((rs[slot].flags & SR_ID_IS_MISSING) || len == 0 ||
+        len > (uint32_t)ID_SLOT_BYTES) {
+      out += "null";
+      return;
+    }
+    out.append((const char*)h_id_.p + (size_t)slot * ID_SLOT_BYTES, len);
+  }
+
+  // {"jsonrpc":"2.0","id":ID,"error":{"code":C,"message":"M"}}
+  void build_rpc_error(std::string& out, int slot, int code, const char* msg) {
+    out.clear();
+    out += "{\"jsonrpc\":\"2.0\",\"id\":";
+    append_id(out, slot);
+    out += ",\"error\":{\"code\":";
+    out += std::to_string(code);
+    out += ",\"message\":\"";
+    json_escape_append(out, msg, strlen(msg));
+    out += "\"}}";
+  }
+
+  // gRPC failure -> isError tool result with HTTP 200 (handler.go:252-259;
+  // text format matches engine/batch.py _host_slot)
+  void build_grpc_error(std::string& out, int slot, int status,
+                        const char* detail) {
+    out.clear();
+    out += "{\"jsonrpc\":\"2.0\",\"id\":";
+    append_id(out, slot);
+    out += ",\"result\":{\"content\":[{\"type\":\"text\",\"text\":\"gRPC error ";
+    out += grpc_status_name(status);
+    out += ": ";
+    json_escape_append(out, detail, strlen(detail));
+    out += "\"}],\"isError\":true}}";
+  }
+
   bool stage_msg_idx(py::object msg_idx, int n) {
     if (msg_idx.is_none()) return false;
     auto arr = msg_idx.cast<py::array_t<int32_t>>();
@@ -590,126 +919,144 @@ class Engine {
     return true;
   }
 
+  // Device-only encode (no GIL use, callable from native threads).
   // expects: h_in_ staged, h_off_[0..n]=in_off, h_off_[(n+1)..]=pb_off,
   // h_aux_ = msg_idx when has_idx
+  void run_encode_device(int n, size_t in_bytes, size_t pb_bytes, bool has_idx,
+                         Limits lim, int mode) {
+    uint32_t* h_off = (uint32_t*)h_off_.p;
+    auto g0 = std::chrono::steady_clock::now();
+    HIP_CHECK(hipSetDevice(device_));
+    HIP_CHECK(hipMemcpyAsync(d_in_.p, h_in_.p, in_bytes,
+                             hipMemcpyHostToDevice, stream_));
+    uint32_t* d_off = (uint32_t*)d_off3_.p;  // in_off | pb_off contiguous
+    HIP_CHECK(hipMemcpyAsync(d_off, h_off, 2 * (n + 1) * sizeof(uint32_t),
+                             hipMemcpyHostToDevice, stream_));
+    if (has_idx)
+      HIP_CHECK(hipMemcpyAsync(d_aux2_.p, h_aux_.p, n * sizeof(int32_t),
+                               hipMemcpyHostToDevice, stream_));
+    int blocks = (int)cdiv(n, WPB);
+    if (blocks > 0) {
+      hipLaunchKernelGGL(k_json2pb, dim3(blocks), dim3(WPB * WAVE), 0,
+                         stream_, (const uint8_t*)d_in_.p,
+                         (const uint32_t*)d_off, (uint8_t*)d_pb_.p,
+                         (const uint32_t*)d_off + (n + 1),
+                         (SlotResult*)d_results_.p, (uint8_t*)d_id_slots_.p,
+                         has_idx ? (const int32_t*)d_aux2_.p : nullptr,
+                         tables_, lim, n, mode);
+      HIP_CHECK(hipGetLastError());
+    }
+    HIP_CHECK(hipMemcpyAsync(h_results_.p, d_results_.p,
+                             n * sizeof(SlotResult), hipMemcpyDeviceToHost,
+                             stream_));
+    // id tokens to the host: the native span's C++ error envelopes need
+    // them (mode-0 decode keeps reading the device copy)
+    HIP_CHECK(hipMemcpyAsync(h_id_.p, d_id_slots_.p,
+                             (size_t)n * ID_SLOT_BYTES, hipMemcpyDeviceToHost,
+                             stream_));
+    HIP_CHECK(hipMemcpyAsync(h_pb_.p, d_pb_.p, pb_bytes,
+                             hipMemcpyDeviceToHost, stream_));
+    sync_stream();
+    // pure device span (copies+kernel+sync, no GIL-reacquire wait):
+    // separates real GPU time from thread contention in the wall split
+    last_enc_gpu_ms_ = std::chrono::duration<double, std::milli>(
+                           std::chrono::steady_clock::now() - g0)
+                           .count();
+    last_batch_n_ = n;
+  }
+
   py::tuple run_encode(int n, size_t in_bytes, size_t pb_bytes, bool has_idx,
                        Limits lim, int mode) {
-    uint32_t* h_off = (uint32_t*)h_off_.p;
     {
       py::gil_scoped_release rel;
-      auto g0 = std::chrono::steady_clock::now();
-      HIP_CHECK(hipSetDevice(device_));
-      HIP_CHECK(hipMemcpyAsync(d_in_.p, h_in_.p, in_bytes,
-                               hipMemcpyHostToDevice, stream_));
-      uint32_t* d_off = (uint32_t*)d_off3_.p;  // in_off | pb_off contiguous
-      HIP_CHECK(hipMemcpyAsync(d_off, h_off, 2 * (n + 1) * sizeof(uint32_t),
-                               hipMemcpyHostToDevice, stream_));
-      if (has_idx)
-        HIP_CHECK(hipMemcpyAsync(d_aux2_.p, h_aux_.p, n * sizeof(int32_t),
-                                 hipMemcpyHostToDevice, stream_));
-      int blocks = (int)cdiv(n, WPB);
-      if (blocks > 0) {
-        hipLaunchKernelGGL(k_json2pb, dim3(blocks), dim3(WPB * WAVE), 0,
-                           stream_, (const uint8_t*)d_in_.p,
-                           (const uint32_t*)d_off, (uint8_t*)d_pb_.p,
-                           (const uint32_t*)d_off + (n + 1),
-                           (SlotResult*)d_results_.p, (uint8_t*)d_id_slots_.p,
-                           has_idx ? (const int32_t*)d_aux2_.p : nullptr,
-                           tables_, lim, n, mode);
-        HIP_CHECK(hipGetLastError());
-      }
-      HIP_CHECK(hipMemcpyAsync(h_results_.p, d_results_.p,
-                               n * sizeof(SlotResult), hipMemcpyDeviceToHost,
-                               stream_));
-      HIP_CHECK(hipMemcpyAsync(h_pb_.p, d_pb_.p, pb_bytes,
-                               hipMemcpyDeviceToHost, stream_));
-      sync_stream();
-      // pure device span (copies+kernel+sync, no GIL-reacquire wait):
-      // separates real GPU time from thread contention in the wall split
-      last_enc_gpu_ms_ = std::chrono::duration<double, std::milli>(
-                             std::chrono::steady_clock::now() - g0)
-                             .count();
+      run_encode_device(n, in_bytes, pb_bytes, has_idx, lim, mode);
     }
-    last_batch_n_ = n;
     py::array_t<uint8_t> results((py::ssize_t)(n * sizeof(SlotResult)));
     std::memcpy(results.mutable_data(), h_results_.p, n * sizeof(SlotResult));
     py::memoryview pb_view = py::memoryview::from_memory(h_pb_.p, pb_bytes);
     return py::make_tuple(results, pb_view);
   }
 
+  // Device-only decode (no GIL use, callable from native threads).
   // expects: h_resp_ staged, h_off_ = resp/final/scratch offsets,
   // h_aux_ = msg_idx (+ skip at offset n when has_skip)
-  py::tuple run_decode(int n, size_t resp_bytes, size_t final_bytes,
-                       bool has_skip, int mode) {
+  void run_decode_device(int n, size_t resp_bytes, size_t final_bytes,
+                         bool has_skip, int mode) {
     uint32_t* h_off = (uint32_t*)h_off_.p;
     int32_t* h_aux = (int32_t*)h_aux_.p;
+    auto g0 = std::chrono::steady_clock::now();
+    HIP_CHECK(hipSetDevice(device_));
+    HIP_CHECK(hipMemcpyAsync(d_resp_.p, h_resp_.p, resp_bytes,
+                             hipMemcpyHostToDevice, stream_));
+    // resp_off | final_off | scratch_off contiguous, one copy; same for
+    // msg_idx | skip
+    uint32_t* d_off = (uint32_t*)d_off3_.p;
+    int32_t* d_aux = (int32_t*)d_aux2_.p;
+    HIP_CHECK(hipMemcpyAsync(d_off, h_off, 3 * (n + 1) * sizeof(uint32_t),
+                             hipMemcpyHostToDevice, stream_));
+    HIP_CHECK(hipMemcpyAsync(d_aux, h_aux,
+                             (has_skip ? 2 : 1) * n * sizeof(int32_t),
+                             hipMemcpyHostToDevice, stream_));
+    int blocks = (int)cdiv(n, WPB);
+    if (blocks > 0) {
+      hipLaunchKernelGGL(
+          k_pb2json, dim3(blocks), dim3(WPB * WAVE), 0, stream_,
+          (const uint8_t*)d_resp_.p, (const uint32_t*)d_off,
+          (const int32_t*)d_aux, (const uint8_t*)d_id_slots_.p,
+          mode == 0 ? (const SlotResult*)d_results_.p : nullptr,
+          (uint8_t*)d_scratch_.p, (const uint32_t*)d_off + 2 * (n + 1),
+          (uint8_t*)d_final_.p, (const uint32_t*)d_off + (n + 1),
+          (DecodeResult*)d_dec_results_.p,
+          has_skip ? (const int32_t*)d_aux + n : nullptr, tables_, n, mode);
+      HIP_CHECK(hipGetLastError());
+    }
+    HIP_CHECK(hipMemcpyAsync(h_dec_results_.p, d_dec_results_.p,
+                             n * sizeof(DecodeResult), hipMemcpyDeviceToHost,
+                             stream_));
+    sync_stream();
+    // Gather used bytes into a tight buffer (reusing d_pb_/h_pb_) so the
+    // D2H copy is sum(out_len), not the arena's worst-case capacity.
+    DecodeResult* rs = (DecodeResult*)h_dec_results_.p;
+    uint32_t* tight = (uint32_t*)h_tight_.p;
+    uint64_t acc = 0;
+    for (int i = 0; i < n; ++i) {
+      tight[i] = (uint32_t)acc;
+      acc += (rs[i].out_len + 3u) & ~3u;
+    }
+    if (n > 0 && acc <= d_pb_.n && acc <= h_pb_.n) {
+      HIP_CHECK(hipMemcpyAsync(d_tight_off_.p, tight, n * sizeof(uint32_t),
+                               hipMemcpyHostToDevice, stream_));
+      hipLaunchKernelGGL(k_compact_out, dim3(n), dim3(256), 0, stream_,
+                         (const uint8_t*)d_final_.p,
+                         (const uint32_t*)d_off + (n + 1),
+                         (const uint32_t*)d_tight_off_.p,
+                         (const DecodeResult*)d_dec_results_.p,
+                         (uint8_t*)d_pb_.p, n);
+      HIP_CHECK(hipGetLastError());
+      if (acc)
+        HIP_CHECK(hipMemcpyAsync(h_pb_.p, d_pb_.p, acc,
+                                 hipMemcpyDeviceToHost, stream_));
+      sync_stream();
+      for (int i = 0; i < n; ++i) rs[i].out_off = tight[i];
+      compact_bytes_ = acc;
+      compact_used_ = true;
+    } else {
+      HIP_CHECK(hipMemcpyAsync(h_final_.p, d_final_.p, final_bytes,
+                               hipMemcpyDeviceToHost, stream_));
+      sync_stream();
+      compact_used_ = false;
+    }
+    last_dec_gpu_ms_ = std::chrono::duration<double, std::milli>(
+                           std::chrono::steady_clock::now() - g0)
+                           .count();
+    last_final_bytes_ = final_bytes;
+  }
+
+  py::tuple run_decode(int n, size_t resp_bytes, size_t final_bytes,
+                       bool has_skip, int mode) {
     {
       py::gil_scoped_release rel;
-      auto g0 = std::chrono::steady_clock::now();
-      HIP_CHECK(hipSetDevice(device_));
-      HIP_CHECK(hipMemcpyAsync(d_resp_.p, h_resp_.p, resp_bytes,
-                               hipMemcpyHostToDevice, stream_));
-      // resp_off | final_off | scratch_off contiguous, one copy; same for
-      // msg_idx | skip
-      uint32_t* d_off = (uint32_t*)d_off3_.p;
-      int32_t* d_aux = (int32_t*)d_aux2_.p;
-      HIP_CHECK(hipMemcpyAsync(d_off, h_off, 3 * (n + 1) * sizeof(uint32_t),
-                               hipMemcpyHostToDevice, stream_));
-      HIP_CHECK(hipMemcpyAsync(d_aux, h_aux,
-                               (has_skip ? 2 : 1) * n * sizeof(int32_t),
-                               hipMemcpyHostToDevice, stream_));
-      int blocks = (int)cdiv(n, WPB);
-      if (blocks > 0) {
-        hipLaunchKernelGGL(
-            k_pb2json, dim3(blocks), dim3(WPB * WAVE), 0, stream_,
-            (const uint8_t*)d_resp_.p, (const uint32_t*)d_off,
-            (const int32_t*)d_aux, (const uint8_t*)d_id_slots_.p,
-            mode == 0 ? (const SlotResult*)d_results_.p : nullptr,
-            (uint8_t*)d_scratch_.p, (const uint32_t*)d_off + 2 * (n + 1),
-            (uint8_t*)d_final_.p, (const uint32_t*)d_off + (n + 1),
-            (DecodeResult*)d_dec_results_.p,
-            has_skip ? (const int32_t*)d_aux + n : nullptr, tables_, n, mode);
-        HIP_CHECK(hipGetLastError());
-      }
-      HIP_CHECK(hipMemcpyAsync(h_dec_results_.p, d_dec_results_.p,
-                               n * sizeof(DecodeResult), hipMemcpyDeviceToHost,
-                               stream_));
-      sync_stream();
-      // Gather used bytes into a tight buffer (reusing d_pb_/h_pb_) so the
-      // D2H copy is sum(out_len), not the arena's worst-case capacity.
-      DecodeResult* rs = (DecodeResult*)h_dec_results_.p;
-      uint32_t* tight = (uint32_t*)h_tight_.p;
-      uint64_t acc = 0;
-      for (int i = 0; i < n; ++i) {
-        tight[i] = (uint32_t)acc;
-        acc += (rs[i].out_len + 3u) & ~3u;
-      }
-      if (n > 0 && acc <= d_pb_.n && acc <= h_pb_.n) {
-        HIP_CHECK(hipMemcpyAsync(d_tight_off_.p, tight, n * sizeof(uint32_t),
-                                 hipMemcpyHostToDevice, stream_));
-        hipLaunchKernelGGL(k_compact_out, dim3(n), dim3(256), 0, stream_,
-                           (const uint8_t*)d_final_.p,
-                           (const uint32_t*)d_off + (n + 1),
-                           (const uint32_t*)d_tight_off_.p,
-                           (const DecodeResult*)d_dec_results_.p,
-                           (uint8_t*)d_pb_.p, n);
-        HIP_CHECK(hipGetLastError());
-        if (acc)
-          HIP_CHECK(hipMemcpyAsync(h_pb_.p, d_pb_.p, acc,
-                                   hipMemcpyDeviceToHost, stream_));
-        sync_stream();
-        for (int i = 0; i < n; ++i) rs[i].out_off = tight[i];
-        compact_bytes_ = acc;
-        compact_used_ = true;
-      } else {
-        HIP_CHECK(hipMemcpyAsync(h_final_.p, d_final_.p, final_bytes,
-                                 hipMemcpyDeviceToHost, stream_));
-        sync_stream();
-        compact_used_ = false;
-      }
-      last_dec_gpu_ms_ = std::chrono::duration<double, std::milli>(
-                             std::chrono::steady_clock::now() - g0)
-                             .count();
+      run_decode_device(n, resp_bytes, final_bytes, has_skip, mode);
     }
     py::array_t<uint8_t> results((py::ssize_t)(n * sizeof(DecodeResult)));
     std::memcpy(results.mutable_data(), h_dec_results_.p,
@@ -736,6 +1083,7 @@ class Engine {
   int last_batch_n_ = -1;
   bool compact_used_ = false;
   size_t compact_bytes_ = 0;
+  size_t last_final_bytes_ = 0;
   hipStream_t stream_;
   hipEvent_t sync_ev_ = nullptr;
   Tables tables_{};
@@ -745,7 +1093,7 @@ class Engine {
   DeviceBuf d_results_, d_dec_results_, d_id_slots_,
       d_tight_off_;
   PinnedBuf h_in_, h_pb_, h_resp_, h_final_, h_results_, h_dec_results_, h_off_,
-      h_aux_, h_tight_;
+      h_aux_, h_tight_, h_id_;
   std::vector<std::string> tool_paths_;
   std::vector<int32_t> tool_out_msg_;
   std::vector<int32_t> tool_backend_;
@@ -805,6 +1153,10 @@ PYBIND11_MODULE(_jsonproto, m) {
            py::arg("msg_idx"), py::arg("skip") = py::none(), py::arg("mode") = 0)
       .def("last_stage_ms", &Engine::last_stage_ms)
       .def("last_gpu_ms", &Engine::last_gpu_ms)
+      .def("span_handle",
+           [](Engine& e) { return (uintptr_t)(spanapi::ISpanExecutor*)&e; },
+           "opaque ISpanExecutor* for the native HTTP frontend "
+           "(span_api.h); the engine must outlive the frontend")
       .def_property_readonly("device", &Engine::device)
       .def_property_readonly("max_batch", &Engine::max_batch);
 }

@@ -37,6 +37,7 @@
 #include <unistd.h>
 
 #include "session_table.h"
+#include "span_api.h"
 
 #include <algorithm>
 #include <atomic>
@@ -172,6 +173,56 @@ class Frontend {
   void set_session_table(sesstab::SessionTable* st, bool rate_limit_enabled) {
     sess_table_ = st;
     sess_rate_limit_ = rate_limit_enabled;
+  }
+
+  // Fully-native serving span (span_api.h): workers run the tools/call hot
+  // path — GPU encode, gRPC invoke, GPU decode, response envelopes —
+  // entirely in C++; fallback_cb is entered ONLY for the slots the span
+  // reports (streaming / non-tools-call / host fallbacks / rejected
+  // sessions).  fallback_cb(items) -> list[bytes], item = (kind, body,
+  // session_id, headers, aux, tool_idx); kind -1 = blocked session,
+  // -2 = rate-limited, else spanapi K_PY_*.  Executors and clients are
+  // opaque handles (Engine.span_handle() / Client.raw_handle()); their
+  // owners must outlive the frontend.  Call before start().
+  void set_native_span(const std::vector<uintptr_t>& execs,
+                       const std::vector<uintptr_t>& clients,
+                       double timeout_s, int max_depth, int max_string,
+                       long max_args, int enforce, int max_span_batch,
+                       long max_span_bytes, py::function fallback_cb) {
+    span_execs_.clear();
+    span_mu_.clear();
+    for (auto h : execs) {
+      span_execs_.push_back((spanapi::ISpanExecutor*)h);
+      span_mu_.push_back(std::make_unique<std::mutex>());
+    }
+    span_clients_.clear();
+    for (auto h : clients) span_clients_.push_back((void*)h);
+    span_timeout_s_ = timeout_s;
+    span_max_depth_ = (uint32_t)max_depth;
+    span_max_string_ = (uint32_t)max_string;
+    span_max_args_ = (uint32_t)max_args;
+    span_enforce_ = enforce;
+    span_max_batch_ = max_span_batch > 0 ? max_span_batch : 4096;
+    span_max_bytes_ = max_span_bytes > 0 ? (size_t)max_span_bytes : (8u << 20);
+    fallback_cb_ = fallback_cb;
+  }
+
+  // per-stage counters of the native span path (for /metrics; mirrors the
+  // Python EngineStats snapshot keys)
+  py::dict native_stats() const {
+    py::dict d;
+    d["batches"] = ns_batches_.load();
+    d["requests"] = ns_requests_.load();
+    d["gpuOk"] = ns_final_.load();
+    d["errors"] = ns_err_final_.load();
+    d["hostFallbacks"] = ns_py_slots_.load();
+    d["spanFailures"] = ns_span_fail_.load();
+    d["encodeMs"] = ns_enc_us_.load() / 1e3;
+    d["invokeMs"] = ns_inv_us_.load() / 1e3;
+    d["decodeMs"] = ns_dec_us_.load() / 1e3;
+    d["encodeGpuMs"] = ns_enc_gpu_us_.load() / 1e3;
+    d["decodeGpuMs"] = ns_dec_gpu_us_.load() / 1e3;
+    return d;
   }
 
   int start() {
@@ -623,45 +674,210 @@ class Frontend {
       }
       std::vector<OutResp> out;
       out.reserve(batch.size());
-      {
-        py::gil_scoped_acquire gil;
-        py::list bodies, sessions, headers, verdicts;
-        std::vector<size_t> batch_idx;
-        bool guarded = sess_table_ != nullptr;
+      bool native = !span_execs_.empty();
+      bool any_nonbatch = false;
+      // slots the native span hands back to Python:
+      // (batch index, kind, aux, tool_idx)
+      std::vector<size_t> py_items;
+      std::vector<int> py_kinds;
+      std::vector<std::string> py_aux;
+      std::vector<int32_t> py_tool;
+      if (native) {
+        // ---- GIL-FREE hot path: session verdicts already assigned by the
+        // reactor; allowed tools/call slots run through the span executor
+        // (GPU encode -> gRPC -> GPU decode -> envelope) in C++ ----------
+        std::vector<size_t> ok_idx;
         for (size_t i = 0; i < batch.size(); ++i) {
-          if (!batch[i].batchable) continue;
-          batch_idx.push_back(i);
-          bodies.append(py::bytes(batch[i].body));
-          sessions.append(batch[i].session.empty()
-                              ? py::object(py::none())
-                              : py::object(py::str(batch[i].session)));
-          if (guarded) verdicts.append(py::int_(batch[i].verdict));
-          py::dict h;
-          for (auto& kv : batch[i].headers)
-            h[py::str(kv.first)] = py::str(kv.second);
-          headers.append(h);
+          if (!batch[i].batchable) {
+            any_nonbatch = true;
+            continue;
+          }
+          if (batch[i].verdict != 0) {
+            py_items.push_back(i);
+            py_kinds.push_back(-batch[i].verdict);
+            py_aux.emplace_back();
+            py_tool.push_back(-1);
+            continue;
+          }
+          ok_idx.push_back(i);
         }
-        if (py::len(bodies) > 0) {
+        ns_batches_.fetch_add(ok_idx.empty() ? 0 : 1,
+                              std::memory_order_relaxed);
+        ns_requests_.fetch_add((long)ok_idx.size(), std::memory_order_relaxed);
+        // chunk by count and staged bytes (engine arena caps)
+        size_t base = 0;
+        while (base < ok_idx.size()) {
+          size_t end = base, bytes = 0;
+          while (end < ok_idx.size() && (end - base) < (size_t)span_max_batch_ &&
+                 (end == base ||
+                  bytes + batch[ok_idx[end]].body.size() <= span_max_bytes_)) {
+            bytes += batch[ok_idx[end]].body.size();
+            ++end;
+          }
+          size_t cn = end - base;
+          std::vector<const char*> bptr(cn);
+          std::vector<size_t> blen(cn);
+          std::vector<
+              const std::vector<std::pair<std::string, std::string>>*>
+              metas(cn);
+          for (size_t k = 0; k < cn; ++k) {
+            PendingReq& r = batch[ok_idx[base + k]];
+            bptr[k] = r.body.data();
+            blen[k] = r.body.size();
+            metas[k] = r.headers.empty() ? nullptr : &r.headers;
+          }
+          spanapi::SpanIn sin{bptr.data(),
+                              blen.data(),
+                              cn,
+                              metas.data(),
+                              span_clients_.data(),
+                              span_clients_.size(),
+                              span_timeout_s_,
+                              span_max_depth_,
+                              span_max_string_,
+                              span_max_args_,
+                              span_enforce_};
+          spanapi::SpanOut sout;
+          std::string serr;
+          size_t ei = span_rr_.fetch_add(1) % span_execs_.size();
+          bool okrun;
+          {
+            std::lock_guard<std::mutex> lk(*span_mu_[ei]);
+            okrun = span_execs_[ei]->run_span(sin, &sout, &serr);
+            if (okrun) {
+              // materialize responses while the executor is held: the blob
+              // belongs to the engine and is reused by its next span
+              for (size_t k = 0; k < cn; ++k) {
+                spanapi::SlotOut& so = sout.slots[k];
+                size_t i = ok_idx[base + k];
+                if (so.kind == spanapi::K_FINAL) {
+                  ns_final_.fetch_add(1, std::memory_order_relaxed);
+                  out.push_back(
+                      {batch[i].conn_id, batch[i].seq,
+                       http_response(
+                           200,
+                           std::string((const char*)sout.blob + so.off,
+                                       so.len),
+                           batch[i].session)});
+                } else if (so.kind == spanapi::K_ERR_FINAL) {
+                  ns_err_final_.fetch_add(1, std::memory_order_relaxed);
+                  out.push_back({batch[i].conn_id, batch[i].seq,
+                                 http_response(200, so.aux,
+                                               batch[i].session)});
+                } else {
+                  py_items.push_back(i);
+                  py_kinds.push_back(so.kind);
+                  py_aux.push_back(std::move(so.aux));
+                  py_tool.push_back(so.tool_idx);
+                }
+              }
+              ns_enc_us_.fetch_add((long)(sout.enc_ms * 1e3),
+                                   std::memory_order_relaxed);
+              ns_inv_us_.fetch_add((long)(sout.inv_ms * 1e3),
+                                   std::memory_order_relaxed);
+              ns_dec_us_.fetch_add((long)(sout.dec_ms * 1e3),
+                                   std::memory_order_relaxed);
+              ns_enc_gpu_us_.fetch_add((long)(sout.enc_gpu_ms * 1e3),
+                                       std::memory_order_relaxed);
+              ns_dec_gpu_us_.fetch_add((long)(sout.dec_gpu_ms * 1e3),
+                                       std::memory_order_relaxed);
+            }
+          }
+          if (!okrun) {
+            // engine failure: answer -32603 rather than guessing whether
+            // the invoke stage ran (a blind retry could double side
+            // effects on non-idempotent methods)
+            ns_span_fail_.fetch_add(1, std::memory_order_relaxed);
+            std::string env =
+                "{\"jsonrpc\":\"2.0\",\"id\":null,\"error\":{\"code\":-32603,"
+                "\"message\":\"engine failure\"}}";
+            for (size_t k = 0; k < cn; ++k) {
+              size_t i = ok_idx[base + k];
+              out.push_back({batch[i].conn_id, batch[i].seq,
+                             http_response(200, env, batch[i].session)});
+            }
+          }
+          base = end;
+        }
+        ns_py_slots_.fetch_add((long)py_items.size(),
+                               std::memory_order_relaxed);
+      } else {
+        for (size_t i = 0; i < batch.size(); ++i)
+          if (!batch[i].batchable) any_nonbatch = true;
+      }
+      bool need_python =
+          any_nonbatch || !py_items.empty() || (!native && !batch.empty());
+      if (need_python) {
+        py::gil_scoped_acquire gil;
+        if (!native) {
+          py::list bodies, sessions, headers, verdicts;
+          std::vector<size_t> batch_idx;
+          bool guarded = sess_table_ != nullptr;
+          for (size_t i = 0; i < batch.size(); ++i) {
+            if (!batch[i].batchable) continue;
+            batch_idx.push_back(i);
+            bodies.append(py::bytes(batch[i].body));
+            sessions.append(batch[i].session.empty()
+                                ? py::object(py::none())
+                                : py::object(py::str(batch[i].session)));
+            if (guarded) verdicts.append(py::int_(batch[i].verdict));
+            py::dict h;
+            for (auto& kv : batch[i].headers)
+              h[py::str(kv.first)] = py::str(kv.second);
+            headers.append(h);
+          }
+          if (py::len(bodies) > 0) {
+            try {
+              py::list res = batch_cb_(bodies, sessions, headers,
+                                       guarded ? py::object(verdicts)
+                                               : py::object(py::none()));
+              for (size_t k = 0; k < batch_idx.size(); ++k) {
+                py::tuple t = res[k].cast<py::tuple>();
+                std::string body = t[0].cast<std::string>();
+                std::string sid = t[1].cast<std::string>();
+                out.push_back({batch[batch_idx[k]].conn_id,
+                               batch[batch_idx[k]].seq,
+                               http_response(200, body, sid)});
+              }
+            } catch (const std::exception& e) {
+              std::string err = std::string(
+                  "{\"jsonrpc\":\"2.0\",\"id\":null,\"error\":"
+                  "{\"code\":-32603,\"message\":\"internal\"}}");
+              for (size_t k = 0; k < batch_idx.size(); ++k)
+                out.push_back({batch[batch_idx[k]].conn_id,
+                               batch[batch_idx[k]].seq,
+                               http_response(200, err, "")});
+            }
+          }
+        } else if (!py_items.empty()) {
+          // rare slots the span could not finish natively
           try {
-            py::list res = batch_cb_(bodies, sessions, headers,
-                                     guarded ? py::object(verdicts)
-                                             : py::object(py::none()));
-            for (size_t k = 0; k < batch_idx.size(); ++k) {
-              py::tuple t = res[k].cast<py::tuple>();
-              std::string body = t[0].cast<std::string>();
-              std::string sid = t[1].cast<std::string>();
-              out.push_back({batch[batch_idx[k]].conn_id,
-                             batch[batch_idx[k]].seq,
-                             http_response(200, body, sid)});
+            py::list items;
+            for (size_t k = 0; k < py_items.size(); ++k) {
+              PendingReq& r = batch[py_items[k]];
+              py::dict h;
+              for (auto& kv : r.headers)
+                h[py::str(kv.first)] = py::str(kv.second);
+              items.append(py::make_tuple(
+                  py_kinds[k], py::bytes(r.body), py::str(r.session), h,
+                  py::bytes(py_aux[k]), py_tool[k]));
+            }
+            py::list res = fallback_cb_(items);
+            for (size_t k = 0; k < py_items.size(); ++k) {
+              PendingReq& r = batch[py_items[k]];
+              out.push_back({r.conn_id, r.seq,
+                             http_response(200, res[k].cast<std::string>(),
+                                           r.session)});
             }
           } catch (const std::exception& e) {
-            std::string err = std::string(
-                "{\"jsonrpc\":\"2.0\",\"id\":null,\"error\":"
-                "{\"code\":-32603,\"message\":\"internal\"}}");
-            for (size_t k = 0; k < batch_idx.size(); ++k)
-              out.push_back({batch[batch_idx[k]].conn_id,
-                             batch[batch_idx[k]].seq,
-                             http_response(200, err, "")});
+            std::string env =
+                "{\"jsonrpc\":\"2.0\",\"id\":null,\"error\":{\"code\":-32603,"
+                "\"message\":\"internal\"}}";
+            for (size_t k = 0; k < py_items.size(); ++k) {
+              PendingReq& r = batch[py_items[k]];
+              out.push_back(
+                  {r.conn_id, r.seq, http_response(200, env, r.session)});
+            }
           }
         }
         for (size_t i = 0; i < batch.size(); ++i) {
@@ -707,6 +923,22 @@ class Frontend {
   std::string host_;
   int port_;
   py::function batch_cb_, slow_cb_;
+  // native span state (set_native_span)
+  std::vector<spanapi::ISpanExecutor*> span_execs_;
+  std::vector<std::unique_ptr<std::mutex>> span_mu_;
+  std::vector<void*> span_clients_;
+  double span_timeout_s_ = 30.0;
+  uint32_t span_max_depth_ = 10, span_max_string_ = 1024,
+           span_max_args_ = 1u << 20;
+  int span_enforce_ = 1;
+  int span_max_batch_ = 4096;
+  size_t span_max_bytes_ = 8u << 20;
+  py::function fallback_cb_;
+  std::atomic<size_t> span_rr_{0};
+  std::atomic<long> ns_batches_{0}, ns_requests_{0}, ns_final_{0},
+      ns_err_final_{0}, ns_py_slots_{0}, ns_span_fail_{0};
+  std::atomic<long> ns_enc_us_{0}, ns_inv_us_{0}, ns_dec_us_{0},
+      ns_enc_gpu_us_{0}, ns_dec_gpu_us_{0};
   bool reuse_port_ = false;
   sesstab::SessionTable* sess_table_ = nullptr;
   bool sess_rate_limit_ = true;
@@ -733,6 +965,72 @@ class Frontend {
   std::mutex batch_mu_;
   std::condition_variable batch_cv_;
   std::deque<std::vector<PendingReq>> batches_;
+};
+
+// ---------------------------------------------------------------------------
+// MockSpanExecutor — CPU stand-in for the GPU engine so the frontend's
+// native-span plumbing (chunking, per-kind fallback routing, blob copies,
+// error paths) is testable without a GPU.  Body markers select the slot
+// kind; anything else echoes a canned K_FINAL response.
+// ---------------------------------------------------------------------------
+
+class MockSpanExecutor : public spanapi::ISpanExecutor {
+ public:
+  bool run_span(const spanapi::SpanIn& in, spanapi::SpanOut* out,
+                std::string* err) override {
+    using namespace spanapi;
+    calls_++;
+    if (fail_next_) {
+      fail_next_ = false;
+      *err = "mock failure";
+      return false;
+    }
+    blob_.clear();
+    out->slots.assign(in.n, SlotOut());
+    for (size_t i = 0; i < in.n; ++i) {
+      std::string body(in.bodies[i], in.body_lens[i]);
+      SlotOut& so = out->slots[i];
+      so.tool_idx = (int32_t)i;
+      if (body.find("__stream__") != std::string::npos) {
+        so.kind = K_PY_STREAM;
+        so.aux = "PB";
+      } else if (body.find("__notool__") != std::string::npos) {
+        so.kind = K_PY_NOT_TOOLCALL;
+      } else if (body.find("__encfb__") != std::string::npos) {
+        so.kind = K_PY_ENC_FALLBACK;
+      } else if (body.find("__decfb__") != std::string::npos) {
+        so.kind = K_PY_DEC_FALLBACK;
+        so.aux = "WIRE";
+      } else if (body.find("__err__") != std::string::npos) {
+        so.kind = K_ERR_FINAL;
+        so.aux =
+            "{\"jsonrpc\":\"2.0\",\"id\":null,\"error\":{\"code\":-32600,"
+            "\"message\":\"mock\"}}";
+      } else {
+        so.kind = K_FINAL;
+        so.off = (uint32_t)blob_.size();
+        std::string resp = "{\"jsonrpc\":\"2.0\",\"id\":1,\"result\":{"
+                           "\"content\":[{\"type\":\"text\",\"text\":\"n=" +
+                           std::to_string(in.body_lens[i]) +
+                           "\"}],\"isError\":false}}";
+        blob_ += resp;
+        so.len = (uint32_t)resp.size();
+      }
+    }
+    out->blob = (const uint8_t*)blob_.data();
+    out->blob_len = blob_.size();
+    out->enc_ms = 0.01;
+    out->inv_ms = 0.01;
+    out->dec_ms = 0.01;
+    return true;
+  }
+  void fail_next() { fail_next_ = true; }
+  long calls() const { return calls_; }
+
+ private:
+  std::string blob_;
+  bool fail_next_ = false;
+  long calls_ = 0;
 };
 
 // ---------------------------------------------------------------------------
@@ -945,10 +1243,28 @@ PYBIND11_MODULE(_frontend, m) {
            py::arg("enabled"), py::arg("forward_all"), py::arg("allow"),
            py::arg("block"),
            "install the forwarding filter in the parser (call before start)")
+      .def("set_native_span", &Frontend::set_native_span, py::arg("execs"),
+           py::arg("clients"), py::arg("timeout_s") = 30.0,
+           py::arg("max_depth") = 10, py::arg("max_string") = 1024,
+           py::arg("max_args") = (long)(1u << 20), py::arg("enforce") = 1,
+           py::arg("max_span_batch") = 4096,
+           py::arg("max_span_bytes") = (long)(8u << 20),
+           py::arg("fallback_cb"),
+           "run tools/call batches through ISpanExecutor handles in C++ "
+           "(no GIL on the hot path); call before start")
+      .def("native_stats", &Frontend::native_stats)
       .def("start", &Frontend::start)
       .def("stop", &Frontend::stop)
       .def_property_readonly("port", &Frontend::port);
   m.def("bench_client", &bench_client, py::arg("host"), py::arg("port"),
         py::arg("sessions"), py::arg("requests"), py::arg("body"),
         py::arg("threads") = 8);
+  py::class_<MockSpanExecutor>(m, "MockSpanExecutor")
+      .def(py::init<>())
+      .def("span_handle",
+           [](MockSpanExecutor& e) {
+             return (uintptr_t)(spanapi::ISpanExecutor*)&e;
+           })
+      .def("fail_next", &MockSpanExecutor::fail_next)
+      .def("calls", &MockSpanExecutor::calls);
 }

@@ -15,6 +15,10 @@ PYBIND11_MODULE(_h2grpc, m) {
            py::arg("payloads"), py::arg("timeout_s") = 30.0,
            py::arg("metadata") = std::vector<std::vector<std::pair<std::string, std::string>>>{})
       .def("healthy", &H2GrpcClient::healthy)
+      .def("raw_handle",
+           [](H2GrpcClient& c) { return (uintptr_t)&c; },
+           "opaque H2GrpcClient* for the native span executor; the client "
+           "must outlive every frontend/engine holding the handle")
       .def("close", &H2GrpcClient::close_all);
   py::class_<H2Server>(m, "Server")
       .def(py::init<const std::string&>(), py::arg("target"))

@@ -227,6 +227,19 @@ class NativeHTTPGateway:
             getattr(self.headers, "case_insensitive", False)
             and hasattr(mod.Frontend, "set_header_filter")
         )
+        # fully-native span: the frontend workers call the GPU engine's
+        # ISpanExecutor directly (no GIL on the tools/call hot path).
+        # Requires the C++ session guard + header filter (so no per-request
+        # Python is needed) and a single GPU pipeline with native transport.
+        native_span_ready = (
+            self._cxx_sessions
+            and self._cxx_header_filter
+            and len(self.pipelines) == 1
+            and getattr(self.pipeline, "wire_clients", None)
+            and hasattr(self.pipeline, "engines")
+            and hasattr(mod.Frontend, "set_native_span")
+        )
+        n_workers = max(1, self.config.gpu.streams) if native_span_ready else 1
         self._fe = mod.Frontend(
             host,
             port,
@@ -237,6 +250,7 @@ class NativeHTTPGateway:
             max_body=srv_cfg.max_body_bytes,
             rate_rps=float(srv_cfg.rate_limit_rps),
             rate_burst=float(srv_cfg.rate_limit_burst),
+            workers=n_workers,
         )
         if getattr(srv_cfg, "reuse_port", False) and hasattr(self._fe, "set_reuse_port"):
             self._fe.set_reuse_port(True)
@@ -244,6 +258,12 @@ class NativeHTTPGateway:
             self._fe.set_session_table(
                 self._sess_table, sess_cfg.rate_limit_enabled
             )
+        self._span_engines: List = []
+        if native_span_ready:
+            try:
+                self._setup_native_span()
+            except Exception as e:  # pragma: no cover - env-specific
+                log.warning("native span unavailable (%s); using batch_cb", e)
         if self._cxx_header_filter:
             self._fe.set_header_filter(
                 self.headers.enabled,
@@ -252,6 +272,81 @@ class NativeHTTPGateway:
                 sorted(self.headers.blocked),
             )
         self.port = 0
+
+    def _setup_native_span(self) -> None:
+        """Dedicated span engines for the C++ workers (separate from the
+        Python pipeline's engines so worker-held arenas and Python calls
+        never interleave on one engine; HBM3E has room to spare)."""
+        from ..engine.batch import GpuEngine
+
+        cfg = self.config
+        n = max(1, cfg.gpu.streams)
+        tables = self.pipeline.engine.tables
+        stats = self.pipeline.engine.stats
+        self._span_engines = [
+            GpuEngine(self.discoverer.tools, cfg, self.pipeline.engine.device,
+                      tables=tables, stats=stats)
+            for _ in range(n)
+        ]
+        self._fe.set_native_span(
+            [e._eng.span_handle() for e in self._span_engines],
+            [c._cli.raw_handle() for c in self.pipeline.wire_clients],
+            timeout_s=cfg.grpc.request_timeout_s,
+            max_span_batch=cfg.gpu.max_batch,
+            max_span_bytes=cfg.gpu.pinned_pool_bytes // 8,
+            fallback_cb=self._fallback_cb,
+        )
+        log.info("native span enabled: %d engines, %d backends", n,
+                 len(self.pipeline.wire_clients))
+
+    def _fallback_cb(self, items) -> List[bytes]:
+        """Rare slots the native span hands back: (kind, body, sid, headers,
+        aux, tool_idx) per item — see span_api.h kinds.  Streaming and
+        encode-side fallbacks re-enter the regular pipeline (nothing was
+        invoked for them); decode-side fallbacks transcode the DELIVERED
+        wire bytes in aux and never re-invoke."""
+        from ..engine.batch import GpuPipeline
+
+        out: List[bytes] = []
+        timeout = self.config.grpc.request_timeout_s
+        for kind, body, sid, hdr, aux, tool_idx in items:
+            try:
+                if kind == -1:
+                    out.append(self._session_error(body, "session is blocked"))
+                elif kind == -2:
+                    out.append(self._session_error(
+                        body, "session rate limit exceeded"))
+                elif kind == 2:  # K_PY_NOT_TOOLCALL
+                    out.append(self._handle_non_toolcall(body, hdr))
+                elif kind == 5:  # K_PY_DEC_FALLBACK: delivered wire in aux
+                    rid, _ = GpuPipeline._extract_id(body)
+                    cpu_decode = getattr(self.pipeline, "_cpu_decode", None)
+                    if cpu_decode is None:
+                        raise RuntimeError("no decode fallback available")
+                    st = self.pipeline.engine.stats
+                    if hasattr(st, "host_fallbacks"):
+                        st.host_fallbacks += 1
+                    out.append(cpu_decode({"tool_idx": tool_idx}, bytes(aux),
+                                          rid))
+                elif kind in (3, 4):  # stream / encode fallback: not yet
+                    # invoked — run the slot through the regular pipeline
+                    out.append(self.pipeline.process_batch(
+                        [body], headers=[hdr], timeout_s=timeout)[0])
+                else:
+                    out.append(self._session_error(body, "internal error"))
+            except Exception as e:  # noqa: BLE001 - per-slot isolation
+                from ..mcp import types as _mcp
+
+                try:
+                    rid, _ = GpuPipeline._extract_id(body)
+                except Exception:
+                    rid = None
+                resp = _mcp.JSONRPCResponse(
+                    id=rid,
+                    error=_mcp.RPCError(_mcp.INTERNAL_ERROR, str(e)[:256]),
+                )
+                out.append(json.dumps(resp.to_dict()).encode())
+        return out
 
     def start(self) -> int:
         self.port = self._fe.start()
@@ -428,6 +523,9 @@ class NativeHTTPGateway:
             stats["sessions"] = self.sessions.stats()
             stats["uptimeS"] = time.time() - self.start_time
             stats["engine"] = self.pipeline.engine.stats.snapshot()
+            if self._span_engines:
+                # per-stage timers of the GIL-free serving path
+                stats["nativeSpan"] = self._fe.native_stats()
             if len(self.pipelines) > 1:
                 stats["shards"] = [p.engine.stats.snapshot() for p in self.pipelines]
             return 200, json.dumps(stats).encode(), ""

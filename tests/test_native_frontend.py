@@ -520,3 +520,103 @@ def test_chunked_transfer_encoding_rejected(gateway):
     assert b"Connection: close" in data
     # and no batch was dispatched for the smuggled payload
     assert sum(pipe.batches) == 0
+
+
+class SpanStubPipeline(StubPipeline):
+    """StubPipeline + the decode-fallback hook the span fallback_cb uses."""
+
+    def __init__(self):
+        super().__init__()
+        self.decode_calls = []
+
+    def _cpu_decode(self, enc_r, wire, rid):
+        self.decode_calls.append((int(enc_r["tool_idx"]), bytes(wire)))
+        return json.dumps({"jsonrpc": "2.0", "id": rid,
+                           "result": {"content": [{"type": "text",
+                                                   "text": "decoded-wire"}],
+                                      "isError": False}}).encode()
+
+
+def _span_gateway():
+    from ggrmcp_amd.server.native_http import load_module
+
+    cfg = Config.default()
+    cfg.server.rate_limit_rps = 100000
+    cfg.server.rate_limit_burst = 100000
+    pipe = SpanStubPipeline()
+    gw = NativeHTTPGateway(pipe, StubDiscoverer(), cfg)
+    mod = load_module()
+    mock = mod.MockSpanExecutor()
+    gw._mock_span = mock  # keep alive: frontend holds a raw pointer
+    gw._fe.set_native_span([mock.span_handle()], [],
+                           fallback_cb=gw._fallback_cb)
+    port = gw.start()
+    return gw, port, pipe, mock
+
+
+def _tc_body(rid, args):
+    return json.dumps({"jsonrpc": "2.0", "id": rid, "method": "tools/call",
+                       "params": {"name": "t", "arguments": args}})
+
+
+def test_native_span_final_and_error_paths():
+    """Frontend native-span plumbing (CPU, mock executor): K_FINAL blob
+    spans, K_ERR_FINAL envelopes, and every K_PY_* fallback kind route
+    through the right handler — with zero Python on the K_FINAL path."""
+    gw, port, pipe, mock = _span_gateway()
+    try:
+        # K_FINAL: canned response copied out of the executor blob
+        status, data, sid = _call(port, _tc_body(1, {"a": 1}))
+        assert status == 200
+        resp = json.loads(data)
+        assert resp["result"]["isError"] is False
+        assert resp["result"]["content"][0]["text"].startswith("n=")
+        assert sid  # session id still issued by the C++ guard
+
+        # K_ERR_FINAL: engine-assembled error envelope passes through
+        status, data, _ = _call(port, _tc_body(2, {"m": "__err__"}))
+        assert json.loads(data)["error"]["code"] == -32600
+
+        # K_PY_NOT_TOOLCALL -> MCP handler (initialize)
+        body = json.dumps({"jsonrpc": "2.0", "id": 3, "method": "initialize",
+                           "params": {"note": "__notool__"}})
+        status, data, _ = _call(port, body)
+        assert json.loads(data)["result"]["protocolVersion"]
+
+        # K_PY_DEC_FALLBACK -> _cpu_decode with the DELIVERED wire (no
+        # re-invoke; the stub records the call)
+        status, data, _ = _call(port, _tc_body(4, {"m": "__decfb__"}))
+        assert json.loads(data)["result"]["content"][0]["text"] == "decoded-wire"
+        assert pipe.decode_calls and pipe.decode_calls[-1][1] == b"WIRE"
+
+        # K_PY_ENC_FALLBACK -> one-slot pipeline batch
+        status, data, _ = _call(port, _tc_body(5, {"m": "__encfb__"}))
+        assert json.loads(data)["result"]["isError"] is False
+
+        # engine failure -> honest -32603, never a blind retry
+        mock.fail_next()
+        status, data, _ = _call(port, _tc_body(6, {"a": 2}))
+        assert json.loads(data)["error"]["code"] == -32603
+
+        # stats flowed
+        st = gw._fe.native_stats()
+        assert st["gpuOk"] >= 1 and st["requests"] >= 5
+    finally:
+        gw.stop()
+
+
+def test_native_span_blocked_session():
+    gw, port, pipe, mock = _span_gateway()
+    try:
+        _, _, sid = _call(port, _tc_body(1, {}))
+        assert gw.sessions.block(sid)
+        status, data, sid2 = _call(port, _tc_body(2, {}), session=sid)
+        resp = json.loads(data)
+        assert resp["error"]["code"] == -32600
+        assert "blocked" in resp["error"]["message"]
+        assert sid2 == sid
+        gw.sessions.unblock(sid)
+        status, data, _ = _call(port, _tc_body(3, {}), session=sid)
+        assert json.loads(data)["result"]["isError"] is False
+    finally:
+        gw.stop()
