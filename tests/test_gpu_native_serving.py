@@ -43,7 +43,7 @@ def native_gateway():
     pipeline = GpuPipeline(d, cfg, device=0, wire_clients=[wire])
     gw = NativeHTTPGateway(pipeline, d, cfg)
     port_http = gw.start()
-    yield gw, port_http, pipeline
+    yield gw, port_http, pipeline, srv
     gw.stop()
     wire.close()
     d.close()
@@ -63,9 +63,18 @@ def _post(port, body, session=None):
     return r.status, json.loads(data), sid
 
 
+def _gpu_ok(gw, pipeline):
+    """GPU-completed requests across both serving paths: the Python
+    batch_cb pipeline and the GIL-free native span (C++ counters)."""
+    n = pipeline.engine.stats.gpu_ok
+    if getattr(gw, "_span_engines", None):
+        n += gw._fe.native_stats()["gpuOk"]
+    return n
+
+
 def test_serving_roundtrip_on_gpu(native_gateway):
-    gw, port, pipeline = native_gateway
-    before = pipeline.engine.stats.gpu_ok
+    gw, port, pipeline = native_gateway[:3]
+    before = _gpu_ok(gw, pipeline)
     body = json.dumps({"jsonrpc": "2.0", "id": 1, "method": "tools/call",
                        "params": {"name": "hello_helloservice_sayhello",
                                   "arguments": {"name": "serve"}}})
@@ -73,11 +82,63 @@ def test_serving_roundtrip_on_gpu(native_gateway):
     assert status == 200 and sid
     inner = json.loads(resp["result"]["content"][0]["text"])
     assert inner == {"message": "Hello, serve!"}
-    assert pipeline.engine.stats.gpu_ok > before  # GPU path, not fallback
+    assert _gpu_ok(gw, pipeline) > before  # GPU path, not fallback
+
+
+def test_native_span_active_on_gpu(native_gateway):
+    """The serving hot path must be the GIL-free C++ span (span_api.h) —
+    not the Python batch_cb fallback (make the serving number
+    the native number)."""
+    gw, port, pipeline = native_gateway[:3]
+    assert gw._span_engines, "native span must be enabled on the GPU path"
+    st0 = gw._fe.native_stats()
+    body = json.dumps({"jsonrpc": "2.0", "id": 11, "method": "tools/call",
+                       "params": {"name": "hello_helloservice_sayhello",
+                                  "arguments": {"name": "span"}}})
+    status, resp, _ = _post(port, body)
+    assert status == 200
+    st1 = gw._fe.native_stats()
+    assert st1["gpuOk"] > st0["gpuOk"]
+    assert st1["requests"] > st0["requests"]
+    assert st1["encodeMs"] > 0 and st1["decodeMs"] > 0
+
+
+def test_native_span_grpc_error_envelope(native_gateway):
+    """gRPC failure -> C++-assembled isError result (handler.go:252-259
+    semantics) with the request id preserved from the kernel's id slot."""
+    gw, port, pipeline = native_gateway[:3]
+    body = json.dumps({"jsonrpc": "2.0", "id": "err-1", "method": "tools/call",
+                       "params": {"name": "hello_helloservice_sayhello",
+                                  "arguments": {"name": "error"}}})
+    status, resp, _ = _post(port, body)
+    assert status == 200
+    assert resp["id"] == "err-1"
+    assert resp["result"]["isError"] is True
+    text = resp["result"]["content"][0]["text"]
+    assert text.startswith("gRPC error INVALID_ARGUMENT")
+
+
+def test_native_span_decode_fallback_single_invoke(native_gateway):
+    """badutf8 response through the SERVING path: decode rejects, the
+    fallback transcodes the delivered wire — zero duplicate invokes."""
+    gw, port, pipeline = native_gateway[:3]
+    body = json.dumps({"jsonrpc": "2.0", "id": 21, "method": "tools/call",
+                       "params": {"name": "hello_helloservice_sayhello",
+                                  "arguments": {"name": "badutf8"}}})
+    srv = native_gateway[3]
+    before = srv.request_count()
+    status, resp, _ = _post(port, body)
+    after = srv.request_count()
+    assert status == 200
+    assert resp["id"] == 21
+    # both GPU and the protojson oracle reject invalid UTF-8 -> error
+    # envelope, produced WITHOUT a second invoke
+    assert "error" in resp
+    assert after - before == 1, "fallback must not re-invoke" 
 
 
 def test_serving_concurrent_sessions(native_gateway):
-    gw, port, pipeline = native_gateway
+    gw, port, pipeline = native_gateway[:3]
     errs = []
 
     def worker(t):
@@ -108,7 +169,7 @@ def test_serving_concurrent_sessions(native_gateway):
 
 
 def test_serving_tools_list_and_errors(native_gateway):
-    gw, port, pipeline = native_gateway
+    gw, port, pipeline = native_gateway[:3]
     status, resp, _ = _post(port, json.dumps(
         {"jsonrpc": "2.0", "id": 5, "method": "tools/list"}))
     assert status == 200
@@ -128,7 +189,7 @@ def test_serving_soak_mixed(native_gateway):
     match its request id and shape (order/captivity races would mispair)."""
     import random
 
-    gw, port, pipeline = native_gateway
+    gw, port, pipeline = native_gateway[:3]
     errs = []
 
     def worker(t):
