@@ -87,6 +87,8 @@ def parse_args():
                          "is UDS-syscall bound — profiles/wide64_conns.log)")
     ap.add_argument("--streams", type=int, default=0,
                     help="override config.gpu.streams (engine instances)")
+    ap.add_argument("--reactors", type=int, default=0,
+                    help="override config.server.reactors (serving config)")
     return ap.parse_args()
 
 
@@ -208,8 +210,16 @@ def main() -> None:
         cfg = Config.default()
         cfg.grpc.uds = sock
         cfg.gpu.enabled = use_gpu
+        if args.config == "hello1k" and use_gpu:
+            # tuned serving defaults (profiles/serving_sweep_r02.log):
+            # reactors 4->8 and span engines 2->6 lifted 1-GPU serving
+            # 242k -> 363k req/s; explicit flags still override
+            cfg.gpu.streams = 6
+            cfg.server.reactors = 8
         if args.streams > 0:
             cfg.gpu.streams = args.streams
+        if args.reactors > 0:
+            cfg.server.reactors = args.reactors
         if args.config == "wide64":
             # 64 KB payloads need bigger arenas.  Keep the batch LARGE: the
             # transcode kernels are per-wave latency-bound, so wall time per

@@ -129,6 +129,9 @@ class ServerConfig:
     # rank all bound to the SAME port; the kernel balances connections
     # (tools/serve_dp.py).  No reference equivalent (single process).
     reuse_port: bool = False
+    # native frontend epoll reactor shards (HTTP parse + socket I/O);
+    # a single reactor measured ~25 us/request of parse+epoll+write
+    reactors: int = 4
     security_headers_enabled: bool = True
 
 

@@ -251,6 +251,7 @@ class NativeHTTPGateway:
             rate_rps=float(srv_cfg.rate_limit_rps),
             rate_burst=float(srv_cfg.rate_limit_burst),
             workers=n_workers,
+            reactors=max(1, getattr(srv_cfg, "reactors", 4)),
         )
         if getattr(srv_cfg, "reuse_port", False) and hasattr(self._fe, "set_reuse_port"):
             self._fe.set_reuse_port(True)
