@@ -61,7 +61,7 @@ def parse_args():
     # requests), so the driver's GPU-busy sampler sees the run (VERDICT r1
     # item 8: 20 x 4 ms was a 0.08 s window)
     ap.add_argument("--steps", type=int, default=-1,
-                    help="-1 = config default (serving 400, pipeline 200)")
+                    help="-1 = config default (serving 800, pipeline 200)")
     ap.add_argument("--warmup", type=int, default=-1,
                     help="-1 = config default (serving 40, pipeline 20)")
     ap.add_argument("--config", default="hello1k",
@@ -163,7 +163,7 @@ def main() -> None:
     # per-config step defaults sized so the timed region spans seconds on
     # the GPU box yet finishes within minutes anywhere
     step_defaults = {
-        "hello1k": (400, 40) if use_gpu else (5, 1),
+        "hello1k": (800, 40) if use_gpu else (5, 1),
         "pipeline": (200, 20),
         "wide64": (100, 10),
         "stream": (8, 2),
