@@ -253,6 +253,23 @@ constexpr int MAX_RECURSE = 16;
 constexpr uint32_t LEN_SLOT = 3;
 constexpr uint32_t LEN_SLOT_MAX = (1u << 21) - 1;
 
+// ---------------------------------------------------------------------------
+// Workgroup-cooperative decode (k_pb2json_wg): responses at least
+// WG_DEC_MIN_BYTES long are decoded one WORKGROUP per request — wave 0
+// scans the top-level field runs, then the block's waves decode runs
+// concurrently into per-item scratch regions and escape-compact them into
+// the final envelope.  Cuts the per-wave ~26 us/KB serialization floor by
+// ~the wave count for large payloads (BASELINE config 3).  The engine
+// provisions WG_DEC_EXTRA extra scratch per eligible slot (per-item
+// regions are padded by WG_DEC_ITEM_PAD for field-name/syntax overhead).
+// ---------------------------------------------------------------------------
+constexpr uint32_t WG_DEC_MIN_BYTES = 16384;
+constexpr int WG_DEC_MAX_ITEMS = 768;
+constexpr uint32_t WG_DEC_ITEM_PAD = 256;
+constexpr uint32_t WG_DEC_EXTRA =
+    WG_DEC_ITEM_PAD * (uint32_t)WG_DEC_MAX_ITEMS + 1024;
+constexpr int WG_DEC_WAVES = 8;  // waves per request workgroup
+
 // u64 -> decimal text without an addressable temp buffer (a local tmp[20]
 // array lands in scratch memory and costs a private-memory round trip per
 // digit; emitting MSB-first via the power table keeps everything in VGPRs)

@@ -73,6 +73,13 @@ struct PinnedBuf {
 
 uint64_t cdiv(uint64_t a, uint64_t b) { return (a + b - 1) / b; }
 
+// workgroup-decode routing threshold; GGRMCP_WG_DEC_MIN overrides for
+// classic-vs-cooperative differential tests (read per batch: ~100 ns)
+inline uint32_t wg_dec_min() {
+  const char* e = getenv("GGRMCP_WG_DEC_MIN");
+  return e ? (uint32_t)strtoul(e, nullptr, 10) : WG_DEC_MIN_BYTES;
+}
+
 // borrow (ptr, len) from a bytes / bytearray / memoryview element
 inline bool view_of(py::handle el, const char** ptr, size_t* len) {
   if (PyBytes_Check(el.ptr())) {
@@ -104,7 +111,12 @@ inline size_t pb_cap(size_t in_len) {
   return (in_len + in_len / 4 + 192 + 15) & ~(size_t)15;
 }
 inline size_t scratch_cap(size_t wire_len) {
-  return (wire_len * 8 + 1024 + 15) & ~(size_t)15;
+  // +WG_DEC_EXTRA for large responses: the workgroup-cooperative decode
+  // pads each top-level item's scratch region (common.h); keyed on the
+  // same (env-overridable) threshold the routing uses
+  return (wire_len * 8 + 1024 +
+          (wire_len >= wg_dec_min() ? (size_t)WG_DEC_EXTRA : 0) + 15) &
+         ~(size_t)15;
 }
 inline size_t final_cap(size_t wire_len) {
   return (wire_len * 16 + 2048 + 15) & ~(size_t)15;
@@ -984,6 +996,19 @@ class Engine : public spanapi::ISpanExecutor {
                          bool has_skip, int mode) {
     uint32_t* h_off = (uint32_t*)h_off_.p;
     int32_t* h_aux = (int32_t*)h_aux_.p;
+    // route big mode-0 responses to the workgroup-cooperative kernel
+    // (skip tag 2): the classic kernel skips them, k_pb2json_wg decodes
+    // one request per workgroup (common.h WG_DEC_*)
+    int n_wg = 0;
+    if (mode == 0 && has_skip) {
+      uint32_t thr = wg_dec_min();
+      for (int i = 0; i < n; ++i) {
+        if (h_aux[n + i] == 0 && h_off[i + 1] - h_off[i] >= thr) {
+          h_aux[n + i] = 2;
+          ++n_wg;
+        }
+      }
+    }
     auto g0 = std::chrono::steady_clock::now();
     HIP_CHECK(hipSetDevice(device_));
     HIP_CHECK(hipMemcpyAsync(d_resp_.p, h_resp_.p, resp_bytes,
@@ -1008,6 +1033,17 @@ class Engine : public spanapi::ISpanExecutor {
           (uint8_t*)d_final_.p, (const uint32_t*)d_off + (n + 1),
           (DecodeResult*)d_dec_results_.p,
           has_skip ? (const int32_t*)d_aux + n : nullptr, tables_, n, mode);
+      HIP_CHECK(hipGetLastError());
+    }
+    if (n_wg > 0) {
+      hipLaunchKernelGGL(
+          k_pb2json_wg, dim3(n), dim3(WG_DEC_WAVES * WAVE), 0, stream_,
+          (const uint8_t*)d_resp_.p, (const uint32_t*)d_off,
+          (const int32_t*)d_aux, (const uint8_t*)d_id_slots_.p,
+          (const SlotResult*)d_results_.p, (uint8_t*)d_scratch_.p,
+          (const uint32_t*)d_off + 2 * (n + 1), (uint8_t*)d_final_.p,
+          (const uint32_t*)d_off + (n + 1), (DecodeResult*)d_dec_results_.p,
+          (const int32_t*)d_aux + n, tables_, n);
       HIP_CHECK(hipGetLastError());
     }
     HIP_CHECK(hipMemcpyAsync(h_dec_results_.p, d_dec_results_.p,

@@ -1072,13 +1072,31 @@ DEVN bool map_entry_step(DCtx& c, DFrame& f, uint32_t eend, DFrame* stack,
   return true;
 }
 
-DEV bool decode_walk(DCtx& c, int top_msg_idx, uint32_t top_end) {
+// bare_top: the top frame emits NO surrounding braces — used by the
+// workgroup-cooperative kernel, whose per-item walks produce the comma-
+// joined members of an enclosing object the caller writes itself.
+DEV bool decode_walk(DCtx& c, int top_msg_idx, uint32_t top_end,
+                     bool bare_top = false) {
   DFrame stack[MAX_RECURSE];
   int sp = 0;
 
   // enter the top-level body (responses are plain messages; transcode tests
   // may decode WKT-typed messages directly)
-  {
+  if (bare_top) {
+    // known-plain body (the wg scanner rejects WKT tops): push the frame
+    // directly, no opening brace
+    DFrame& f = stack[sp++];
+    f.end = top_end;
+    f.prev_number = 0;
+    f.cont_end = 0;
+    f.msg_idx = top_msg_idx;
+    f.cont_field = -1;
+    f.cont_num = 0;
+    f.mode = FM_BODY;
+    f.first_member = 1;
+    f.cont_kind = CK_NONE;
+    f.cont_first = 1;
+  } else {
     uint8_t pm = FM_BODY;
     int32_t pidx = top_msg_idx;
     uint32_t bend = top_end;
@@ -1280,7 +1298,7 @@ DEV bool decode_walk(DCtx& c, int top_msg_idx, uint32_t top_end) {
 
     // FM_BODY
     if (c.pos >= f.end) {
-      if (!putc_(c, '}')) return false;
+      if (!(bare_top && sp == 1) && !putc_(c, '}')) return false;
       --sp;
       continue;
     }
@@ -1568,3 +1586,313 @@ extern "C" __global__ void k_compact_out(
   for (uint32_t i = t; i < words; i += stride) d4[i] = s4[i];
   for (uint32_t i = (words << 2) + t; i < len; i += stride) d[i] = s[i];
 }
+
+#ifndef GGRMCP_HOST_SIM
+// ---------------------------------------------------------------------------
+// k_pb2json_wg — workgroup-per-request decode for LARGE responses
+// (BASELINE config 3: 64 KB payloads).
+//
+// The per-wave walker serializes one request through 64 lanes at a
+// ~26 us/KB floor (a 64 KB response costs ~1.7 ms of wave latency and the
+// whole batch inherits it).  Here one WORKGROUP (WG_DEC_WAVES wave64s)
+// owns one request:
+//   A. one thread scans the top-level wire for contiguous field runs
+//      ("items") — O(#fields), just varint/tag walking;
+//   B. waves grab items dynamically and decode each item with the SAME
+//      decode_walk machinery (bare_top: no surrounding braces) into a
+//      private scratch region, then compute its escaped length;
+//   C. one thread prefix-sums the escaped item lengths into final
+//      offsets (comma-joined, skipping default-omitted empty items);
+//   D. wave 0 writes the JSON-RPC envelope framing while every wave
+//      escape-copies its items into place.
+// Requests the scanner cannot split safely (WKT-typed top message,
+// out-of-order top fields, > WG_DEC_MAX_ITEMS runs) fall back to the
+// classic single-wave mode-0 path inside the same block, preserving
+// exact semantics.  The engine routes only mode-0 slots with
+// wire_len >= WG_DEC_MIN_BYTES here (skip tag == 2).
+// ---------------------------------------------------------------------------
+
+static __device__ const char WG_P1[] = "{\"jsonrpc\":\"2.0\",\"id\":";
+static __device__ const char WG_P2[] =
+    ",\"result\":{\"content\":[{\"type\":\"text\",\"text\":\"";
+static __device__ const char WG_P3[] = "\"}],\"isError\":false}}";
+
+extern "C" __global__ void __launch_bounds__(WG_DEC_WAVES * WAVE) k_pb2json_wg(
+    const uint8_t* __restrict__ resp_bytes,
+    const uint32_t* __restrict__ resp_off,
+    const int32_t* __restrict__ msg_idx_arr,
+    const uint8_t* __restrict__ id_slots,
+    const SlotResult* __restrict__ enc_results, uint8_t* __restrict__ scratch,
+    const uint32_t* __restrict__ scratch_off, uint8_t* __restrict__ final_out,
+    const uint32_t* __restrict__ final_off, DecodeResult* __restrict__ results,
+    const int32_t* __restrict__ skip, Tables t, int n_req) {
+  int req = blockIdx.x;
+  if (req >= n_req) return;
+  if (!skip || skip[req] != 2) return;  // 2 = routed to this kernel
+
+  __shared__ uint32_t s_start[WG_DEC_MAX_ITEMS + 1];
+  __shared__ uint32_t s_outlen[WG_DEC_MAX_ITEMS];
+  __shared__ uint32_t s_esclen[WG_DEC_MAX_ITEMS];
+  __shared__ uint32_t s_finoff[WG_DEC_MAX_ITEMS];
+  __shared__ uint8_t s_comma[WG_DEC_MAX_ITEMS];
+  __shared__ int s_nitems, s_next, s_err, s_mode;
+  __shared__ uint32_t s_total;
+
+  const uint8_t* pb = resp_bytes + resp_off[req];
+  const uint32_t wire_len = resp_off[req + 1] - resp_off[req];
+  uint8_t* scr = scratch + scratch_off[req];
+  const uint32_t scr_cap = scratch_off[req + 1] - scratch_off[req];
+  uint8_t* fout = final_out + final_off[req];
+  const uint32_t fcap = final_off[req + 1] - final_off[req];
+  const int lane = lane_id();
+  const int wave = threadIdx.x / WAVE;
+  const int msg_idx = msg_idx_arr[req];
+
+  // ---- phase A: top-level field-run scan (one thread) ---------------------
+  if (threadIdx.x == 0) {
+    s_next = 0;
+    s_err = E_OK;
+    s_mode = 1;
+    int n = 0;
+    uint32_t pos = 0, prev_num = 0, cur = 0xFFFFFFFFu;
+    bool ok = t.msgs[msg_idx].wkt_kind == WKT_NONE;
+    while (ok && pos < wire_len) {
+      uint32_t tag_start = pos;
+      uint64_t tag;
+      if (!get_varint(pb, wire_len, &pos, &tag)) { ok = false; break; }
+      uint32_t num = (uint32_t)(tag >> 3), wt = (uint32_t)(tag & 7);
+      if (num == 0) { ok = false; break; }
+      if (num != cur) {
+        // new run must be strictly ascending (non-adjacent duplicates are
+        // the classic path's E_UNSUPPORTED -> host fallback)
+        if (num < prev_num || n >= WG_DEC_MAX_ITEMS) { ok = false; break; }
+        s_start[n++] = tag_start;
+        prev_num = num;
+        cur = num;
+      }
+      if (wt == W_VARINT) {
+        uint64_t v;
+        if (!get_varint(pb, wire_len, &pos, &v)) { ok = false; break; }
+      } else if (wt == W_I64) {
+        if (pos + 8 > wire_len) { ok = false; break; }
+        pos += 8;
+      } else if (wt == W_I32) {
+        if (pos + 4 > wire_len) { ok = false; break; }
+        pos += 4;
+      } else if (wt == W_LEN) {
+        uint64_t v;
+        if (!get_varint(pb, wire_len, &pos, &v)) { ok = false; break; }
+        if (pos + v > wire_len) { ok = false; break; }
+        pos += (uint32_t)v;
+      } else {
+        ok = false;
+        break;
+      }
+    }
+    if (ok && pos == wire_len && n > 0) {
+      s_start[n] = wire_len;
+      s_nitems = n;
+    } else {
+      s_mode = 0;  // classic single-wave fallback below
+      s_nitems = 0;
+    }
+  }
+  __syncthreads();
+  const int n_items = s_nitems;
+
+  // ---- phase B: decode items (dynamic wave grabs) -------------------------
+  if (s_mode) {
+    while (true) {
+      int idx = 0;
+      if (!lane) idx = atomicAdd(&s_next, 1);
+      idx = __shfl(idx, 0, WAVE);
+      if (idx >= n_items) break;
+      uint32_t ist = s_start[idx], ien = s_start[idx + 1];
+      uint32_t soff = 8u * ist + WG_DEC_ITEM_PAD * (uint32_t)idx;
+      uint32_t scap = 8u * (ien - ist) + WG_DEC_ITEM_PAD;
+      if (soff + scap > scr_cap) {
+        if (!lane) {
+          atomicCAS(&s_err, E_OK, E_OVERFLOW);
+          s_outlen[idx] = 0;
+          s_esclen[idx] = 0;
+        }
+        continue;
+      }
+      DCtx c;
+      c.pb = pb;
+      c.len = ien;
+      c.pos = ist;
+      c.out = scr + soff;
+      c.opos = 0;
+      c.ocap = scap;
+      c.t = t;
+      c.status = E_OK;
+      c.lane = lane;
+      bool ok = decode_walk(c, msg_idx, ien, /*bare_top=*/true);
+      if (!ok || c.status != E_OK) {
+        if (!lane) {
+          atomicCAS(&s_err, E_OK, c.status == E_OK ? E_PARSE : c.status);
+          s_outlen[idx] = 0;
+          s_esclen[idx] = 0;
+        }
+        continue;
+      }
+      // escaped length of the produced text (wave-parallel reduce)
+      uint32_t acc = 0;
+      for (uint32_t i = lane; i < c.opos; i += WAVE) acc += esc_len(c.out[i]);
+      for (int d = WAVE / 2; d > 0; d >>= 1) acc += __shfl_down(acc, d, WAVE);
+      acc = __shfl(acc, 0, WAVE);
+      if (!lane) {
+        s_outlen[idx] = c.opos;
+        s_esclen[idx] = acc;
+      }
+    }
+  }
+  __syncthreads();
+
+  // ---- phase C: escaped prefix sums + envelope sizing (one thread) --------
+  if (s_mode && threadIdx.x == 0) {
+    s_next = 0;  // re-used by phase D's grab loop
+    if (s_err == E_OK) {
+      uint32_t idl = enc_results ? enc_results[req].id_len : 0;
+      uint32_t pre = (uint32_t)(sizeof(WG_P1) - 1) + (idl ? idl : 4) +
+                     (uint32_t)(sizeof(WG_P2) - 1) + 1 /* '{' */;
+      uint32_t off = pre;
+      int first = 1;
+      for (int i = 0; i < n_items; ++i) {
+        uint8_t comma = 0;
+        if (s_outlen[i]) {
+          if (!first) comma = 1;
+          first = 0;
+        }
+        s_comma[i] = comma;
+        s_finoff[i] = off + comma;
+        off += s_outlen[i] ? s_esclen[i] + comma : 0;
+      }
+      uint32_t total = off + 1 /* '}' */ + (uint32_t)(sizeof(WG_P3) - 1);
+      if (total > fcap) s_err = E_OVERFLOW;
+      s_total = total;
+    }
+  }
+  __syncthreads();
+
+  // ---- phase D: envelope framing (wave 0) + item escapes (all waves) ------
+  if (s_mode && s_err == E_OK) {
+    if (wave == 0) {
+      DCtx o;
+      o.pb = nullptr;
+      o.len = 0;
+      o.pos = 0;
+      o.out = fout;
+      o.opos = 0;
+      o.ocap = fcap;
+      o.t = t;
+      o.status = E_OK;
+      o.lane = lane;
+      bool ok = puts_(o, WG_P1, sizeof(WG_P1) - 1);
+      uint32_t idl = enc_results ? enc_results[req].id_len : 0;
+      if (ok && idl) {
+        const uint8_t* idp = id_slots + (size_t)req * ID_SLOT_BYTES;
+        for (uint32_t i = lane; i < idl; i += WAVE) o.out[o.opos + i] = idp[i];
+        o.opos += idl;
+      } else if (ok) {
+        ok = puts_(o, "null", 4);
+      }
+      if (ok) ok = puts_(o, WG_P2, sizeof(WG_P2) - 1);
+      if (ok) ok = putc_(o, '{');
+      // closing '}' + P3 at the precomputed end
+      uint32_t cpos = s_total - (uint32_t)(sizeof(WG_P3) - 1) - 1;
+      if (!lane) fout[cpos] = '}';
+      for (uint32_t i = lane; i < sizeof(WG_P3) - 1; i += WAVE)
+        fout[cpos + 1 + i] = (uint8_t)WG_P3[i];
+      if (!ok && !lane) atomicCAS(&s_err, E_OK, o.status);
+    }
+    while (true) {
+      int idx = 0;
+      if (!lane) idx = atomicAdd(&s_next, 1);
+      idx = __shfl(idx, 0, WAVE);
+      if (idx >= n_items) break;
+      if (!s_outlen[idx]) continue;
+      uint32_t soff = 8u * s_start[idx] + WG_DEC_ITEM_PAD * (uint32_t)idx;
+      if (s_comma[idx] && !lane) fout[s_finoff[idx] - 1] = ',';
+      DCtx e;
+      e.pb = nullptr;
+      e.len = 0;
+      e.pos = 0;
+      e.out = fout + s_finoff[idx];
+      e.opos = 0;
+      e.ocap = s_esclen[idx];
+      e.t = t;
+      e.status = E_OK;
+      e.lane = lane;
+      if (!put_escaped(e, scr + soff, s_outlen[idx])) {
+        if (!lane) atomicCAS(&s_err, E_OK, e.status);
+      }
+    }
+  } else if (!s_mode && wave == 0) {
+    // ---- classic single-wave fallback (exact k_pb2json mode-0 body) ------
+    DecodeResult r;
+    r.status = E_OK;
+    r.out_off = final_off[req];
+    r.out_len = 0;
+    r.pad = 0;
+    DCtx c;
+    c.pb = pb;
+    c.len = wire_len;
+    c.pos = 0;
+    c.out = scr;
+    c.opos = 0;
+    c.ocap = scr_cap;
+    c.t = t;
+    c.status = E_OK;
+    c.lane = lane;
+    decode_walk(c, msg_idx, c.len);
+    if (c.status != E_OK) {
+      r.status = c.status;
+      if (!lane) results[req] = r;
+    } else {
+      uint32_t json_len = c.opos;
+      DCtx o;
+      o.pb = nullptr;
+      o.len = 0;
+      o.pos = 0;
+      o.out = fout;
+      o.opos = 0;
+      o.ocap = fcap;
+      o.t = t;
+      o.status = E_OK;
+      o.lane = lane;
+      bool ok = puts_(o, WG_P1, sizeof(WG_P1) - 1);
+      uint32_t idl = enc_results ? enc_results[req].id_len : 0;
+      if (ok && idl) {
+        const uint8_t* idp = id_slots + (size_t)req * ID_SLOT_BYTES;
+        for (uint32_t i = lane; i < idl; i += WAVE) o.out[o.opos + i] = idp[i];
+        o.opos += idl;
+      } else if (ok) {
+        ok = puts_(o, "null", 4);
+      }
+      if (ok) ok = puts_(o, WG_P2, sizeof(WG_P2) - 1);
+      if (ok) ok = put_escaped(o, scr, json_len);
+      if (ok) ok = puts_(o, WG_P3, sizeof(WG_P3) - 1);
+      if (!ok) {
+        r.status = o.status;
+      } else {
+        r.status = E_OK;
+        r.out_len = o.opos;
+      }
+      if (!lane) results[req] = r;
+    }
+  }
+  __syncthreads();
+
+  // ---- finalize (item mode) ------------------------------------------------
+  if (s_mode && threadIdx.x == 0) {
+    DecodeResult r;
+    r.out_off = final_off[req];
+    r.pad = 0;
+    r.status = s_err;
+    r.out_len = s_err == E_OK ? s_total : 0;
+    results[req] = r;
+  }
+}
+#endif  // GGRMCP_HOST_SIM
