@@ -125,8 +125,13 @@ DEV uint64_t esc_pack(uint8_t b, uint32_t el) {
 // append src[0..n) JSON-escaped; lane-parallel with a no-escape fast path.
 // Invalid UTF-8 (proto3 string fields must be valid; protojson errors) ->
 // E_UNSUPPORTED so the host re-attempt surfaces the proper error.
-DEV bool put_escaped(DCtx& c, const uint8_t* src, uint32_t n) {
-  if (!utf8_span_valid(src, n, c.lane)) return dfail(c, E_UNSUPPORTED);
+// validate=false skips the UTF-8 re-check: used on SLICES of text the
+// walker already validated (a slice boundary can split a multi-byte
+// sequence, which would false-reject; escaping itself is per-byte safe)
+DEV bool put_escaped(DCtx& c, const uint8_t* src, uint32_t n,
+                     bool validate = true) {
+  if (validate && !utf8_span_valid(src, n, c.lane))
+    return dfail(c, E_UNSUPPORTED);
   for (uint32_t base = 0; base < n; base += WAVE) {
     uint32_t i = base + c.lane;
     uint32_t win = n - base < WAVE ? n - base : WAVE;
@@ -1630,13 +1635,19 @@ extern "C" __global__ void __launch_bounds__(WG_DEC_WAVES * WAVE) k_pb2json_wg(
   if (req >= n_req) return;
   if (!skip || skip[req] != 2) return;  // 2 = routed to this kernel
 
+  // items = top-level field runs; slices = fixed-size spans of the items'
+  // produced text, so the escape passes load-balance across waves even
+  // when one map/array run carries most of the payload
+  constexpr int MAX_SLICES = 1024;
   __shared__ uint32_t s_start[WG_DEC_MAX_ITEMS + 1];
   __shared__ uint32_t s_outlen[WG_DEC_MAX_ITEMS];
-  __shared__ uint32_t s_esclen[WG_DEC_MAX_ITEMS];
-  __shared__ uint32_t s_finoff[WG_DEC_MAX_ITEMS];
-  __shared__ uint8_t s_comma[WG_DEC_MAX_ITEMS];
-  __shared__ int s_nitems, s_next, s_err, s_mode;
-  __shared__ uint32_t s_total;
+  __shared__ uint16_t s_sl_item[MAX_SLICES];
+  __shared__ uint32_t s_sl_off[MAX_SLICES];   // offset within the item text
+  __shared__ uint32_t s_sl_esc[MAX_SLICES];   // escaped length of the slice
+  __shared__ uint32_t s_sl_fin[MAX_SLICES];   // final-buffer offset
+  __shared__ uint8_t s_sl_comma[MAX_SLICES];  // prepend ',' (item starts)
+  __shared__ int s_nitems, s_nslices, s_next, s_err, s_mode;
+  __shared__ uint32_t s_total, s_slice_bytes;
 
   const uint8_t* pb = resp_bytes + resp_off[req];
   const uint32_t wire_len = resp_off[req + 1] - resp_off[req];
@@ -1714,7 +1725,6 @@ extern "C" __global__ void __launch_bounds__(WG_DEC_WAVES * WAVE) k_pb2json_wg(
         if (!lane) {
           atomicCAS(&s_err, E_OK, E_OVERFLOW);
           s_outlen[idx] = 0;
-          s_esclen[idx] = 0;
         }
         continue;
       }
@@ -1733,24 +1743,66 @@ extern "C" __global__ void __launch_bounds__(WG_DEC_WAVES * WAVE) k_pb2json_wg(
         if (!lane) {
           atomicCAS(&s_err, E_OK, c.status == E_OK ? E_PARSE : c.status);
           s_outlen[idx] = 0;
-          s_esclen[idx] = 0;
         }
         continue;
       }
-      // escaped length of the produced text (wave-parallel reduce)
-      uint32_t acc = 0;
-      for (uint32_t i = lane; i < c.opos; i += WAVE) acc += esc_len(c.out[i]);
-      for (int d = WAVE / 2; d > 0; d >>= 1) acc += __shfl_down(acc, d, WAVE);
-      acc = __shfl(acc, 0, WAVE);
-      if (!lane) {
-        s_outlen[idx] = c.opos;
-        s_esclen[idx] = acc;
-      }
+      if (!lane) s_outlen[idx] = c.opos;
     }
   }
   __syncthreads();
 
-  // ---- phase C: escaped prefix sums + envelope sizing (one thread) --------
+  // ---- phase C1: slice the item texts (one thread) ------------------------
+  if (s_mode && threadIdx.x == 0) {
+    s_next = 0;  // re-used by phase C2's grab loop
+    int ns = 0;
+    if (s_err == E_OK) {
+      uint32_t sum = 0;
+      for (int i = 0; i < n_items; ++i) sum += s_outlen[i];
+      uint32_t sb = 4096;
+      uint32_t budget = (uint32_t)(MAX_SLICES - n_items);
+      if (budget > 0 && sum / budget + 1 > sb) sb = sum / budget + 1;
+      s_slice_bytes = sb;
+      bool clipped = false;
+      for (int i = 0; i < n_items && !clipped; ++i) {
+        for (uint32_t off = 0; off < s_outlen[i]; off += sb) {
+          if (ns >= MAX_SLICES) {  // unreachable by the budget math above,
+            clipped = true;        // but NEVER truncate output silently
+            break;
+          }
+          s_sl_item[ns] = (uint16_t)i;
+          s_sl_off[ns] = off;
+          ++ns;
+        }
+      }
+      if (clipped) s_err = E_OVERFLOW;
+    }
+    s_nslices = ns;
+  }
+  __syncthreads();
+  const int n_slices = s_nslices;
+
+  // ---- phase C2: escaped length per slice (wave-parallel) -----------------
+  if (s_mode && s_err == E_OK) {
+    while (true) {
+      int s = 0;
+      if (!lane) s = atomicAdd(&s_next, 1);
+      s = __shfl(s, 0, WAVE);
+      if (s >= n_slices) break;
+      int item = s_sl_item[s];
+      uint32_t ioff = s_sl_off[s];
+      uint32_t ilen = s_outlen[item] - ioff;
+      if (ilen > s_slice_bytes) ilen = s_slice_bytes;
+      const uint8_t* src =
+          scr + 8u * s_start[item] + WG_DEC_ITEM_PAD * (uint32_t)item + ioff;
+      uint32_t acc = 0;
+      for (uint32_t i = lane; i < ilen; i += WAVE) acc += esc_len(src[i]);
+      for (int d = WAVE / 2; d > 0; d >>= 1) acc += __shfl_down(acc, d, WAVE);
+      if (!lane) s_sl_esc[s] = acc;
+    }
+  }
+  __syncthreads();
+
+  // ---- phase C3: prefix sums + envelope sizing (one thread) ---------------
   if (s_mode && threadIdx.x == 0) {
     s_next = 0;  // re-used by phase D's grab loop
     if (s_err == E_OK) {
@@ -1759,15 +1811,16 @@ extern "C" __global__ void __launch_bounds__(WG_DEC_WAVES * WAVE) k_pb2json_wg(
                      (uint32_t)(sizeof(WG_P2) - 1) + 1 /* '{' */;
       uint32_t off = pre;
       int first = 1;
-      for (int i = 0; i < n_items; ++i) {
+      for (int s = 0; s < n_slices; ++s) {
         uint8_t comma = 0;
-        if (s_outlen[i]) {
+        if (s_sl_off[s] == 0) {  // first slice of its item
           if (!first) comma = 1;
           first = 0;
         }
-        s_comma[i] = comma;
-        s_finoff[i] = off + comma;
-        off += s_outlen[i] ? s_esclen[i] + comma : 0;
+        s_sl_comma[s] = comma;
+        off += comma;
+        s_sl_fin[s] = off;
+        off += s_sl_esc[s];
       }
       uint32_t total = off + 1 /* '}' */ + (uint32_t)(sizeof(WG_P3) - 1);
       if (total > fcap) s_err = E_OVERFLOW;
@@ -1808,24 +1861,30 @@ extern "C" __global__ void __launch_bounds__(WG_DEC_WAVES * WAVE) k_pb2json_wg(
       if (!ok && !lane) atomicCAS(&s_err, E_OK, o.status);
     }
     while (true) {
-      int idx = 0;
-      if (!lane) idx = atomicAdd(&s_next, 1);
-      idx = __shfl(idx, 0, WAVE);
-      if (idx >= n_items) break;
-      if (!s_outlen[idx]) continue;
-      uint32_t soff = 8u * s_start[idx] + WG_DEC_ITEM_PAD * (uint32_t)idx;
-      if (s_comma[idx] && !lane) fout[s_finoff[idx] - 1] = ',';
+      int s = 0;
+      if (!lane) s = atomicAdd(&s_next, 1);
+      s = __shfl(s, 0, WAVE);
+      if (s >= n_slices) break;
+      int item = s_sl_item[s];
+      uint32_t ioff = s_sl_off[s];
+      uint32_t ilen = s_outlen[item] - ioff;
+      if (ilen > s_slice_bytes) ilen = s_slice_bytes;
+      const uint8_t* src =
+          scr + 8u * s_start[item] + WG_DEC_ITEM_PAD * (uint32_t)item + ioff;
+      if (s_sl_comma[s] && !lane) fout[s_sl_fin[s] - 1] = ',';
       DCtx e;
       e.pb = nullptr;
       e.len = 0;
       e.pos = 0;
-      e.out = fout + s_finoff[idx];
+      e.out = fout + s_sl_fin[s];
       e.opos = 0;
-      e.ocap = s_esclen[idx];
+      e.ocap = s_sl_esc[s];
       e.t = t;
       e.status = E_OK;
       e.lane = lane;
-      if (!put_escaped(e, scr + soff, s_outlen[idx])) {
+      // slice of already-validated walker output: skip the UTF-8 re-check
+      // (a boundary can split a multi-byte char; escaping is per-byte)
+      if (!put_escaped(e, src, ilen, /*validate=*/false)) {
         if (!lane) atomicCAS(&s_err, E_OK, e.status);
       }
     }

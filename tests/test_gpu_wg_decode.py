@@ -63,33 +63,35 @@ def _shapes():
 
     rng = random.Random(7)
     shapes = []
-    # config-3 standard: 64-field nested proto, ~64 KB
+    # config-3 standard: 64-field nested proto, ~64 KB (strings stay under
+    # the MCP 1024-char validation cap — that cap applies on ENCODE)
     shapes.append(wide_payload(rng, target_bytes=64 * 1024))
-    # single huge string (ONE top-level item)
-    shapes.append({"f01String": "x" * 40000})
+    # ONE top-level item carrying the whole payload (a single map run)
+    shapes.append({"attrs": {f"p{j}": "x" * 1000 for j in range(40)}})
     # escapes + unicode spread across items
     shapes.append({
-        "f01String": ('he said "hi"\n\t\\' + "é中\U0001f600") * 900,
+        "f01String": ('he said "hi"\n\t\\' + "é中\U0001f600") * 40,
         "f02Int32": -7,
         "f05Bool": True,
         "items": [{"key": f"i{j}", "value": str(j), "weight": j / 3}
-                  for j in range(200)],
+                  for j in range(400)],
+        "attrs": {f"q{j}": "y" * 500 for j in range(20)},
     })
     # many items: scalars + repeated + map + nested around the threshold
-    w = wide_payload(rng, target_bytes=20 * 1024)
-    shapes.append(w)
+    shapes.append(wide_payload(rng, target_bytes=20 * 1024))
     # default-valued singulars between real fields (empty items -> no
     # stray commas)
     shapes.append({
-        "f01String": "a" * 17000,
+        "f01String": "a" * 1000,
         "f02Int32": 0,          # proto3 default: omitted from output
         "f03Int64": "0",
         "f05Bool": False,
         "f04Double": 1.5,
+        "attrs": {f"z{j}": "w" * 900 for j in range(16)},
     })
-    # doubles / floats precision paths at volume (repeated Inner.weight)
+    # doubles / int64 precision paths at volume (repeated Inner)
     shapes.append({
-        "f01String": "p" * 16500,
+        "f01String": "p" * 1000,
         "items": [{"key": f"d{j}",
                    "value": str(rng.randint(-(2**40), 2**40)),
                    "weight": rng.random() * 10 ** rng.randint(-12, 12)}
