@@ -80,6 +80,12 @@ inline uint32_t wg_dec_min() {
   return e ? (uint32_t)strtoul(e, nullptr, 10) : WG_DEC_MIN_BYTES;
 }
 
+// debug bisection: limit the wg kernel to its first N phases
+inline int wg_phases() {
+  const char* e = getenv("GGRMCP_WG_PHASES");
+  return e ? atoi(e) : 3;
+}
+
 // borrow (ptr, len) from a bytes / bytearray / memoryview element
 inline bool view_of(py::handle el, const char** ptr, size_t* len) {
   if (PyBytes_Check(el.ptr())) {
@@ -1043,7 +1049,7 @@ class Engine : public spanapi::ISpanExecutor {
           (const SlotResult*)d_results_.p, (uint8_t*)d_scratch_.p,
           (const uint32_t*)d_off + 2 * (n + 1), (uint8_t*)d_final_.p,
           (const uint32_t*)d_off + (n + 1), (DecodeResult*)d_dec_results_.p,
-          (const int32_t*)d_aux + n, tables_, n);
+          (const int32_t*)d_aux + n, tables_, n, wg_phases());
       HIP_CHECK(hipGetLastError());
     }
     HIP_CHECK(hipMemcpyAsync(h_dec_results_.p, d_dec_results_.p,

@@ -1630,7 +1630,8 @@ extern "C" __global__ void __launch_bounds__(WG_DEC_WAVES * WAVE) k_pb2json_wg(
     const SlotResult* __restrict__ enc_results, uint8_t* __restrict__ scratch,
     const uint32_t* __restrict__ scratch_off, uint8_t* __restrict__ final_out,
     const uint32_t* __restrict__ final_off, DecodeResult* __restrict__ results,
-    const int32_t* __restrict__ skip, Tables t, int n_req) {
+    const int32_t* __restrict__ skip, Tables t, int n_req, int max_phase) {
+  // max_phase: debug bisection knob (GGRMCP_WG_PHASES; 3 = full pipeline)
   int req = blockIdx.x;
   if (req >= n_req) return;
   if (!skip || skip[req] != 2) return;  // 2 = routed to this kernel
@@ -1712,7 +1713,7 @@ extern "C" __global__ void __launch_bounds__(WG_DEC_WAVES * WAVE) k_pb2json_wg(
   const int n_items = s_nitems;
 
   // ---- phase B: decode items (dynamic wave grabs) -------------------------
-  if (s_mode) {
+  if (s_mode && max_phase >= 1) {
     while (true) {
       int idx = 0;
       if (!lane) idx = atomicAdd(&s_next, 1);
@@ -1782,7 +1783,7 @@ extern "C" __global__ void __launch_bounds__(WG_DEC_WAVES * WAVE) k_pb2json_wg(
   const int n_slices = s_nslices;
 
   // ---- phase C2: escaped length per slice (wave-parallel) -----------------
-  if (s_mode && s_err == E_OK) {
+  if (s_mode && s_err == E_OK && max_phase >= 2) {
     while (true) {
       int s = 0;
       if (!lane) s = atomicAdd(&s_next, 1);
@@ -1830,7 +1831,7 @@ extern "C" __global__ void __launch_bounds__(WG_DEC_WAVES * WAVE) k_pb2json_wg(
   __syncthreads();
 
   // ---- phase D: envelope framing (wave 0) + item escapes (all waves) ------
-  if (s_mode && s_err == E_OK) {
+  if (s_mode && s_err == E_OK && max_phase >= 3) {
     if (wave == 0) {
       DCtx o;
       o.pb = nullptr;
