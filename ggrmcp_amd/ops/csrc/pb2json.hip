@@ -1712,9 +1712,10 @@ extern "C" __global__ void __launch_bounds__(WG_DEC_WAVES * WAVE) k_pb2json_wg(
   __syncthreads();
   const int n_items = s_nitems;
 
-  // ---- phase B: decode items (dynamic wave grabs) -------------------------
+  // ---- phase B: decode items (dynamic wave grabs; the guard bound makes
+  // the loop provably finite — a wave can win at most n_items grabs) ------
   if (s_mode && max_phase >= 1) {
-    while (true) {
+    for (int guard = 0; guard <= WG_DEC_MAX_ITEMS + 1; ++guard) {
       int idx = 0;
       if (!lane) idx = atomicAdd(&s_next, 1);
       idx = __shfl(idx, 0, WAVE);
@@ -1783,12 +1784,10 @@ extern "C" __global__ void __launch_bounds__(WG_DEC_WAVES * WAVE) k_pb2json_wg(
   const int n_slices = s_nslices;
 
   // ---- phase C2: escaped length per slice (wave-parallel) -----------------
+  // static wave-strided assignment: slices are uniform-size, so dynamic
+  // grabbing buys nothing over round-robin
   if (s_mode && s_err == E_OK && max_phase >= 2) {
-    while (true) {
-      int s = 0;
-      if (!lane) s = atomicAdd(&s_next, 1);
-      s = __shfl(s, 0, WAVE);
-      if (s >= n_slices) break;
+    for (int s = wave; s < n_slices; s += WG_DEC_WAVES) {
       int item = s_sl_item[s];
       uint32_t ioff = s_sl_off[s];
       uint32_t ilen = s_outlen[item] - ioff;
@@ -1861,11 +1860,7 @@ extern "C" __global__ void __launch_bounds__(WG_DEC_WAVES * WAVE) k_pb2json_wg(
         fout[cpos + 1 + i] = (uint8_t)WG_P3[i];
       if (!ok && !lane) atomicCAS(&s_err, E_OK, o.status);
     }
-    while (true) {
-      int s = 0;
-      if (!lane) s = atomicAdd(&s_next, 1);
-      s = __shfl(s, 0, WAVE);
-      if (s >= n_slices) break;
+    for (int s = wave; s < n_slices; s += WG_DEC_WAVES) {
       int item = s_sl_item[s];
       uint32_t ioff = s_sl_off[s];
       uint32_t ilen = s_outlen[item] - ioff;
