@@ -70,7 +70,9 @@ def _shapes():
     shapes.append({"attrs": {f"p{j}": "x" * 1000 for j in range(40)}})
     # escapes + unicode spread across items
     shapes.append({
-        "f01String": ('he said "hi"\n\t\\' + "é中\U0001f600") * 40,
+        # x20: json.dumps ascii-escapes the unicode, and the \uXXXX form is
+        # what the 1024-char string limit sees on the wire
+        "f01String": ('he said "hi"\n\t\\' + "é中\U0001f600") * 20,
         "f02Int32": -7,
         "f05Bool": True,
         "items": [{"key": f"i{j}", "value": str(j), "weight": j / 3}

@@ -60,3 +60,25 @@ if os.environ["GGRMCP_WG_PHASES"] == "3" and dec[0]["status"] == 0:
     assert inner == oracle, (inner, oracle)
     print("oracle match", flush=True)
 print("PROBE-OK", flush=True)
+
+def diag_shapes():
+    """kind=shapes: run the wg-decode test shapes through encode and print
+    per-slot statuses (diagnosing validation rejections)."""
+    sys.path.insert(0, str(Path(__file__).resolve().parent.parent / "tests"))
+    from test_gpu_wg_decode import _shapes
+
+    shapes = _shapes()
+    bodies = [json.dumps({"jsonrpc": "2.0", "id": i + 1, "method": "tools/call",
+                          "params": {"name": "bench_echoservice_echo",
+                                     "arguments": a}}).encode()
+              for i, a in enumerate(shapes)]
+    enc, pbs = eng.encode_batch(bodies, mode=0)
+    for i, r in enumerate(enc):
+        print(f"shape {i}: status={r['status']} aux={r['aux']} "
+              f"err_pos={r['err_pos']} body_len={len(bodies[i])}", flush=True)
+        if r["status"] != 0:
+            ctx = bodies[i][max(0, r["err_pos"] - 40):r["err_pos"] + 40]
+            print("   around err_pos:", ctx, flush=True)
+
+if kind == "shapes":
+    diag_shapes()
