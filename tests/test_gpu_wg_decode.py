@@ -127,6 +127,7 @@ def test_wg_decode_matches_protojson(env):
     pipeline, d = env
     shapes = _shapes()
     bodies = [_body(a, i + 1) for i, a in enumerate(shapes)]
+    fallbacks_before = pipeline.engine.stats.host_fallbacks
     out = pipeline.process_batch(bodies, timeout_s=30.0)
     mi = d.tools["bench_echoservice_echo"]
     for i, (args, raw) in enumerate(zip(shapes, out)):
@@ -137,7 +138,8 @@ def test_wg_decode_matches_protojson(env):
         wire = pipeline.cpu.json_to_pb(mi.input_descriptor, json.dumps(args))
         oracle = json.loads(pipeline.cpu.pb_to_json(mi.output_descriptor, wire))
         assert inner == oracle, f"slot {i} mismatch"
-    assert pipeline.engine.stats.host_fallbacks == 0
+    assert pipeline.engine.stats.host_fallbacks == fallbacks_before, (
+        "every shape must complete on the GPU path")
 
 
 def test_wg_decode_fuzz_vs_classic(env):
