@@ -217,41 +217,163 @@ DEV double scale10(double d, int e) {
   return d / DPOW10[-e];
 }
 
-// digits * 10^(e10 - ndig + 1) as double — the parse-back for verification
-DEV double digits_to_double(uint64_t digits, int ndig, int e10) {
-  return scale10((double)digits, e10 - ndig + 1);
+// ---------------------------------------------------------------------------
+// double-double arithmetic for near-exact decimal scaling.  A plain
+// `d * 10^e` carries ~1 ulp of error per step — enough to generate
+// 17-digit blocks that are off by one AND to mis-verify their parse-back
+// (found: 301/3 printed as a string that parses to a DIFFERENT double).
+// With an fma-compensated (hi, lo) pair the product of an exact power
+// step is correct to ~2^-100, so digit generation and round-trip
+// verification are exact for every double the kernel handles.
+// ---------------------------------------------------------------------------
+
+struct DD {
+  double hi, lo;
+};
+
+// (hi, lo) * p for EXACT p (a power of ten <= 1e22): Dekker product via fma
+DEV DD dd_mul_exact(DD a, double p) {
+  double p1 = a.hi * p;
+  double e1 = fma(a.hi, p, -p1);
+  double lo = fma(a.lo, p, e1);
+  DD r;
+  r.hi = p1 + lo;
+  r.lo = (p1 - r.hi) + lo;
+  return r;
+}
+
+DEV DD dd_div_exact(DD a, double p) {
+  double q1 = a.hi / p;
+  // residual of the first quotient: a - q1*p, computed exactly
+  double r1 = fma(-q1, p, a.hi) + a.lo;
+  double q2 = r1 / p;
+  DD r;
+  r.hi = q1 + q2;
+  r.lo = (q1 - r.hi) + q2;
+  return r;
+}
+
+// d * 10^e as a double-double (steps through exact powers)
+DEV DD dd_scale10(double d, int e) {
+  DD x{d, 0.0};
+  while (e > 22) {
+    x = dd_mul_exact(x, 1e22);
+    e -= 22;
+  }
+  while (e < -22) {
+    x = dd_div_exact(x, 1e22);
+    e += 22;
+  }
+  if (e >= 0) return dd_mul_exact(x, DPOW10[e]);
+  return dd_div_exact(x, DPOW10[-e]);
+}
+
+// digits * 10^(e10 - ndig + 1) ?= d, verified in double-double.  digits
+// (< 1e17 < 2^57) splits exactly into hi + lo doubles.
+DEV bool digits_roundtrip(uint64_t digits, int ndig, int e10, double d) {
+  double dh = (double)digits;                       // rounded
+  double dl = (double)(int64_t)(digits - (uint64_t)(int64_t)dh);
+  DD x{dh, dl};
+  int e = e10 - ndig + 1;
+  int ee = e;
+  while (ee > 22) {
+    x = dd_mul_exact(x, 1e22);
+    ee -= 22;
+  }
+  while (ee < -22) {
+    x = dd_div_exact(x, 1e22);
+    ee += 22;
+  }
+  x = ee >= 0 ? dd_mul_exact(x, DPOW10[ee]) : dd_div_exact(x, DPOW10[-ee]);
+  // round-trips iff the decimal lies within d's rounding interval: strictly
+  // inside the half-gap on its side, or exactly ON the boundary when d's
+  // mantissa is even (IEEE round-to-nearest-even resolves the tie to d —
+  // e.g. 60982807099928340 parses to ...336, which repr prints at 16
+  // digits; rejecting the tie forced a needless 17-digit form)
+  double diff = (x.hi - d) + x.lo;
+  bool above = diff >= 0;
+  if (diff < 0) diff = -diff;
+  double gap_half = above ? (nextafter(d, 1e308) - d) * 0.5
+                          : (d - nextafter(d, -1e308)) * 0.5;
+  if (diff < gap_half) return true;
+  return diff == gap_half && (__builtin_bit_cast(uint64_t, d) & 1) == 0;
+}
+
+// float32 variant: does the decimal parse (to double, then to float — the
+// protojson/strtof pipeline) land back on f?
+DEV bool digits_roundtrip_f(uint64_t digits, int ndig, int e10, float f) {
+  double dh = (double)digits;
+  double dl = (double)(int64_t)(digits - (uint64_t)(int64_t)dh);
+  DD x{dh, dl};
+  int ee = e10 - ndig + 1;
+  while (ee > 22) {
+    x = dd_mul_exact(x, 1e22);
+    ee -= 22;
+  }
+  while (ee < -22) {
+    x = dd_div_exact(x, 1e22);
+    ee += 22;
+  }
+  x = ee >= 0 ? dd_mul_exact(x, DPOW10[ee]) : dd_div_exact(x, DPOW10[-ee]);
+  double diff = (x.hi - (double)f) + x.lo;
+  bool above = diff >= 0;
+  if (diff < 0) diff = -diff;
+  double gap_half = above
+      ? ((double)nextafterf(f, 3.4e38f) - (double)f) * 0.5
+      : ((double)f - (double)nextafterf(f, -3.4e38f)) * 0.5;
+  if (diff < gap_half) return true;
+  return diff == gap_half && (__builtin_bit_cast(uint32_t, f) & 1) == 0;
 }
 
 // Generate `prec` significant decimal digits of d (>0, finite); returns the
 // digit block and adjusts *e10_out.  Precision-search wrapper below picks the
 // shortest precision whose parse-back round-trips — matching protojson's
-// shortest-representation output for the common cases.
+// shortest-representation output.
 // nearest integer with exact halves rounded to EVEN — shortest-round-trip
 // printers (Ryu, Go strconv; protojson uses them) break the both-candidates-
 // round-trip tie this way, e.g. float32 1048576.25 prints "1048576.2" not
 // "1048576.3" (fuzz-found divergence)
-DEV uint64_t round_half_even_u64(double scaled) {
-  double t = scaled + 0.5;
-  uint64_t dg = (uint64_t)t;
-  if ((double)dg == t && (dg & 1)) --dg;  // t integral <=> frac was exactly .5
+DEV uint64_t dd_round_half_even(DD s) {
+  uint64_t dg = (uint64_t)s.hi;
+  // (s.hi - dg) is exact (both < 2^58, within one ulp of each other).
+  // For scaled values >= 2^53 the cast itself is quantized: s.hi is an
+  // integral double with ulp up to 8, and the compensation term s.lo
+  // carries the true value up to +-ulp/2 AWAY from that integer — so the
+  // adjustment must walk, not step once (found: 17-digit blocks off by
+  // 2-4 for scaled >= 4e16).
+  double frac = (s.hi - (double)dg) + s.lo;
+  while (frac < 0.0) {
+    --dg;
+    frac += 1.0;
+  }
+  while (frac >= 1.0) {
+    ++dg;
+    frac -= 1.0;
+  }
+  if (frac > 0.5 || (frac == 0.5 && (dg & 1))) ++dg;
   return dg;
 }
 
 DEV uint64_t gen_digits(double d, int prec, int* e10_out) {
   int e10 = (int)floor(log10(d));
-  double scaled = scale10(d, (prec - 1) - e10);
-  uint64_t digits = round_half_even_u64(scaled);
+  DD s = dd_scale10(d, (prec - 1) - e10);
+  uint64_t digits = dd_round_half_even(s);
   uint64_t hi = 1;
   for (int i = 0; i < prec; ++i) hi *= 10;
   if (digits >= hi) {
+    // rounded up across a decade (999..9.7 -> 100..00): the dropped digit
+    // is zero, no re-round needed
     digits /= 10;
     e10 += 1;
   } else if (digits < hi / 10) {
     // log10 estimate was one high
     e10 -= 1;
-    scaled = scale10(d, (prec - 1) - e10);
-    digits = round_half_even_u64(scaled);
-    if (digits >= hi) { digits /= 10; e10 += 1; }
+    s = dd_scale10(d, (prec - 1) - e10);
+    digits = dd_round_half_even(s);
+    if (digits >= hi) {
+      digits /= 10;
+      e10 += 1;
+    }
   }
   *e10_out = e10;
   return digits;
@@ -287,8 +409,8 @@ DEVN bool emit_double_body(DCtx& c, double d, bool as_float) {
   for (int p = p_lo; p <= p_hi; ++p) {
     int e;
     uint64_t dg = gen_digits(d, p, &e);
-    double back = digits_to_double(dg, p, e);
-    bool ok = as_float ? ((float)back == (float)d) : (back == d);
+    bool ok = as_float ? digits_roundtrip_f(dg, p, e, (float)d)
+                       : digits_roundtrip(dg, p, e, d);
     if (ok || p == p_hi) {
       digits = dg;
       e10 = e;
@@ -357,7 +479,7 @@ DEVN bool put_double(DCtx& c, double d, bool as_float) {
   // to mis-round subnormals).  Host transcodes these slots (counted).
   {
     double ad = d < 0 ? -d : d;
-    if (ad > 1e308 || (ad != 0.0 && ad < 1e-307))
+    if (ad > 1e308 || (ad != 0.0 && ad < 1e-306))
       return dfail(c, E_UNSUPPORTED);
   }
   if (c.opos + 40 > c.ocap) return dfail(c, E_OVERFLOW);
