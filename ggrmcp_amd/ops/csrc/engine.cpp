@@ -1141,6 +1141,25 @@ class Engine : public spanapi::ISpanExecutor {
   std::vector<int32_t> tool_backend_;
 };
 
+// debug: expose the device's double-double scaling intermediates so host
+// and gfx950 arithmetic can be compared bit-for-bit (wg decode work)
+__global__ void k_dd_probe(double d, int prec, uint64_t* out) {
+  int e10 = (int)floor(log10(d));
+  DD s = dd_scale10(d, (prec - 1) - e10);
+  out[0] = __builtin_bit_cast(uint64_t, s.hi);
+  out[1] = __builtin_bit_cast(uint64_t, s.lo);
+  out[2] = dd_round_half_even(s);
+  out[3] = (uint64_t)(int64_t)e10;
+  int e2 = 0;
+  out[4] = gen_digits(d, prec, &e2);
+  out[5] = (uint64_t)(int64_t)e2;
+  // raw single-step: p1 = d*1e14, e1 = fma(d,1e14,-p1)
+  double p1 = d * 1e14;
+  double e1 = fma(d, 1e14, -p1);
+  out[6] = __builtin_bit_cast(uint64_t, p1);
+  out[7] = __builtin_bit_cast(uint64_t, e1);
+}
+
 static int device_count() {
   int n = 0;
   hipError_t e = hipGetDeviceCount(&n);
@@ -1151,6 +1170,25 @@ static int device_count() {
 PYBIND11_MODULE(_jsonproto, m) {
   m.doc() = "MI355X batch JSON<->protobuf transcode engine (gfx950 HIP kernels)";
   m.def("device_count", &device_count);
+  m.def("dd_probe", [](double d, int prec) {
+    uint64_t* dbuf;
+    HIP_CHECK(hipMalloc(&dbuf, 8 * sizeof(uint64_t)));
+    hipLaunchKernelGGL(k_dd_probe, dim3(1), dim3(1), 0, nullptr, d, prec, dbuf);
+    HIP_CHECK(hipGetLastError());
+    uint64_t h[8];
+    HIP_CHECK(hipMemcpy(h, dbuf, sizeof(h), hipMemcpyDeviceToHost));
+    (void)hipFree(dbuf);
+    py::dict r;
+    r["hi"] = py::cast(*(double*)&h[0]);
+    r["lo"] = py::cast(*(double*)&h[1]);
+    r["round"] = (unsigned long long)h[2];
+    r["e10"] = (long long)(int64_t)h[3];
+    r["digits"] = (unsigned long long)h[4];
+    r["gen_e10"] = (long long)(int64_t)h[5];
+    r["p1"] = py::cast(*(double*)&h[6]);
+    r["e1"] = py::cast(*(double*)&h[7]);
+    return r;
+  });
   // diagnostic: verify the cross-module cast of an _h2grpc Client works in
   // this process (pybind shares its type registry across extensions when
   // the class typeinfo is default-visibility)

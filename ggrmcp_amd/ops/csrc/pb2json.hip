@@ -231,8 +231,13 @@ struct DD {
   double hi, lo;
 };
 
-// (hi, lo) * p for EXACT p (a power of ten <= 1e22): Dekker product via fma
+// (hi, lo) * p for EXACT p (a power of ten <= 1e22): Dekker product via fma.
+// The fp pragmas are load-bearing: hipcc's device default
+// (-ffp-contract=fast-honor-pragmas) reassociated this fma chain and
+// applied the compensation term TWICE (measured on gfx950: 301/3 * 1e14
+// produced hi one step high and lo = e1 - 2 instead of {p1, e1}).
 DEV DD dd_mul_exact(DD a, double p) {
+#pragma clang fp contract(off) reassociate(off)
   double p1 = a.hi * p;
   double e1 = fma(a.hi, p, -p1);
   double lo = fma(a.lo, p, e1);
@@ -243,6 +248,7 @@ DEV DD dd_mul_exact(DD a, double p) {
 }
 
 DEV DD dd_div_exact(DD a, double p) {
+#pragma clang fp contract(off) reassociate(off)
   double q1 = a.hi / p;
   // residual of the first quotient: a - q1*p, computed exactly
   double r1 = fma(-q1, p, a.hi) + a.lo;
@@ -271,6 +277,7 @@ DEV DD dd_scale10(double d, int e) {
 // digits * 10^(e10 - ndig + 1) ?= d, verified in double-double.  digits
 // (< 1e17 < 2^57) splits exactly into hi + lo doubles.
 DEV bool digits_roundtrip(uint64_t digits, int ndig, int e10, double d) {
+#pragma clang fp contract(off) reassociate(off)
   double dh = (double)digits;                       // rounded
   double dl = (double)(int64_t)(digits - (uint64_t)(int64_t)dh);
   DD x{dh, dl};
@@ -302,6 +309,7 @@ DEV bool digits_roundtrip(uint64_t digits, int ndig, int e10, double d) {
 // float32 variant: does the decimal parse (to double, then to float — the
 // protojson/strtof pipeline) land back on f?
 DEV bool digits_roundtrip_f(uint64_t digits, int ndig, int e10, float f) {
+#pragma clang fp contract(off) reassociate(off)
   double dh = (double)digits;
   double dl = (double)(int64_t)(digits - (uint64_t)(int64_t)dh);
   DD x{dh, dl};
@@ -334,6 +342,7 @@ DEV bool digits_roundtrip_f(uint64_t digits, int ndig, int e10, float f) {
 // round-trip tie this way, e.g. float32 1048576.25 prints "1048576.2" not
 // "1048576.3" (fuzz-found divergence)
 DEV uint64_t dd_round_half_even(DD s) {
+#pragma clang fp contract(off) reassociate(off)
   uint64_t dg = (uint64_t)s.hi;
   // (s.hi - dg) is exact (both < 2^58, within one ulp of each other).
   // For scaled values >= 2^53 the cast itself is quantized: s.hi is an

@@ -82,3 +82,48 @@ def diag_shapes():
 
 if kind == "shapes":
     diag_shapes()
+
+def diag_shape_diff(which: int):
+    """kind=shapediffN: decode test shape N and print first oracle diff."""
+    sys.path.insert(0, str(Path(__file__).resolve().parent.parent / "tests"))
+    from test_gpu_wg_decode import _shapes
+
+    args = _shapes()[which]
+    wire = cpu.json_to_pb(mi.input_descriptor, json.dumps(args))
+    out_idx = eng.tables.msg_index[mi.output_descriptor.full_name]
+    dec, outs = eng.decode_batch([wire], [out_idx], mode=1)
+    print("dec status", dec[0]["status"], flush=True)
+    if dec[0]["status"] != 0:
+        return
+    kern = json.loads(outs[0])
+    orac = json.loads(cpu.pb_to_json(mi.output_descriptor, wire))
+    for k in orac:
+        if kern.get(k) != orac[k]:
+            a, b = kern.get(k), orac[k]
+            if isinstance(a, list):
+                for j, (x, y) in enumerate(zip(a, b)):
+                    if x != y:
+                        print(f"field {k}[{j}]:\n  kern {x}\n  orac {y}", flush=True)
+                        return
+            print(f"field {k}:\n  kern {str(a)[:200]}\n  orac {str(b)[:200]}", flush=True)
+            return
+    print("no diff", flush=True)
+
+if kind.startswith("shapediff"):
+    diag_shape_diff(int(kind[len("shapediff"):]))
+
+def diag_doubles():
+    """kind=dbl: minimal double-emission check on this device."""
+    desc = mi.input_descriptor.fields_by_name["nested"].message_type
+    idx = eng.tables.msg_index["bench.Inner"]
+    vals = [301/3, 1/3, 3333333333333333.5, 0.1+0.2, 7.479800121866815e-11,
+            2.4392533358425466e-50, 223355.53145190026, 6.098280709992834e+16]
+    wires = [cpu.json_to_pb(desc, json.dumps({"weight": v})) for v in vals]
+    dec, outs = eng.decode_batch(wires, [idx] * len(wires), mode=1)
+    for v, r, o in zip(vals, dec, outs):
+        ok = r["status"] == 0 and json.loads(o).get("weight") == v
+        print(("OK " if ok else "BAD"), repr(v), "->",
+              o if r["status"] == 0 else f"status={r['status']}", flush=True)
+
+if kind == "dbl":
+    diag_doubles()
