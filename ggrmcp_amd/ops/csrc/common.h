@@ -242,6 +242,70 @@ DEV bool get_varint(const uint8_t* p, uint32_t len, uint32_t* pos, uint64_t* out
   return false;
 }
 
+__constant__ double DD_POW10[23] = {
+    1e0,  1e1,  1e2,  1e3,  1e4,  1e5,  1e6,  1e7,  1e8,  1e9,  1e10, 1e11,
+    1e12, 1e13, 1e14, 1e15, 1e16, 1e17, 1e18, 1e19, 1e20, 1e21, 1e22};
+
+// ---------------------------------------------------------------------------
+// double-double arithmetic for near-exact decimal scaling.  A plain
+// `d * 10^e` carries ~1 ulp of error per step — enough to generate
+// 17-digit blocks that are off by one AND to mis-verify their parse-back
+// (found: 301/3 printed as a string that parses to a DIFFERENT double).
+// With an fma-compensated (hi, lo) pair the product of an exact power
+// step is correct to ~2^-100, so digit generation and round-trip
+// verification are exact for every double the kernel handles.
+// ---------------------------------------------------------------------------
+
+struct DD {
+  double hi, lo;
+};
+
+// (hi, lo) * p for EXACT p (a power of ten <= 1e22 (DD_POW10)): Dekker product via fma.
+// The fp pragmas are load-bearing: hipcc's device default
+// (-ffp-contract=fast-honor-pragmas) reassociated this fma chain and
+// applied the compensation term TWICE (measured on gfx950: 301/3 * 1e14
+// produced hi one step high and lo = e1 - 2 instead of {p1, e1}).
+DEV DD dd_mul_exact(DD a, double p) {
+#pragma clang fp contract(off) reassociate(off)
+  double p1 = a.hi * p;
+  double e1 = fma(a.hi, p, -p1);
+  double lo = fma(a.lo, p, e1);
+  DD r;
+  r.hi = p1 + lo;
+  r.lo = (p1 - r.hi) + lo;
+  return r;
+}
+
+DEV DD dd_div_exact(DD a, double p) {
+#pragma clang fp contract(off) reassociate(off)
+  double q1 = a.hi / p;
+  // residual of the first quotient: a - q1*p, computed exactly
+  double r1 = fma(-q1, p, a.hi) + a.lo;
+  double q2 = r1 / p;
+  DD r;
+  r.hi = q1 + q2;
+  r.lo = (q1 - r.hi) + q2;
+  return r;
+}
+
+// collapse a dd-scaled value d*10^e back to one double (correct to
+// ~2^-100 through composed steps; inf/zero saturate early)
+DEV double dd_scale_collapse(double hi, double lo, int e) {
+  DD x{hi, lo};
+  while (e > 22) {
+    x = dd_mul_exact(x, 1e22);
+    e -= 22;
+    if (x.hi > 1.7e308) return x.hi;  // overflowing; caller range-checks
+  }
+  while (e < -22) {
+    x = dd_div_exact(x, 1e22);
+    e += 22;
+    if (x.hi == 0.0) return 0.0;
+  }
+  x = e >= 0 ? dd_mul_exact(x, DD_POW10[e]) : dd_div_exact(x, DD_POW10[-e]);
+  return x.hi + x.lo;
+}
+
 // nesting cap for the iterative encode/decode walkers' explicit frame
 // stacks (statically-sized private arrays — no dynamic device stack).
 // Deeper nesting returns E_LIMIT and transcodes on the host (counted).

@@ -361,6 +361,8 @@ DEV bool parse_number_text(const uint8_t* s, uint32_t p, uint32_t e, NumVal* nv)
         if (mag) sig++;
         if (frac) dec_exp--;
       } else {
+        if (!overflow && dv >= 5 && mag != 0xFFFFFFFFFFFFFFFFull)
+          ++mag;  // round the kept 19-20 digits by the first dropped one
         overflow = true;
         if (!frac) dec_exp++;
       }
@@ -405,12 +407,21 @@ DEV bool parse_number_text(const uint8_t* s, uint32_t p, uint32_t e, NumVal* nv)
     } else if (dec_exp < 0 && dec_exp >= -22 && mag < (1ull << 53)) {
       d = d / POW10[-dec_exp];
     } else {
-      d = scale_by_pow10(d, dec_exp);
+      // 54..64-bit mantissas (16-20 significant digits): (double)mag
+      // rounds BEFORE the scale, and the second rounding cost 1 ulp
+      // (oracle-found: "90.33333333333333" — 9033333333333333 is just
+      // above 2^53).  Split mag EXACTLY into 32-bit halves (a signed
+      // cast of (mag - trunc) is UB past 2^63) and scale in compensated
+      // double-double arithmetic (common.h).  Longer decimals were
+      // rounded into mag at accumulation, so this path also covers them
+      // to within the 19-digit double-rounding boundary.
+      double dh = (double)(uint32_t)(mag >> 32) * 4294967296.0;  // exact
+      double dl = (double)(uint32_t)mag;                         // exact
+      d = dd_scale_collapse(dh, dl, dec_exp);
       if (isinf(d) && mag != 0) {
-        // composed scaling carries <=~4 ulp of error; within a few ulp of
-        // DBL_MAX that is enough to round to inf while strtod stays finite.
-        // Decide by decimal magnitude: value ~= mag * 10^dec_exp, so its
-        // decimal exponent is dec_exp + floor(log10(mag)).
+        // within a few ulp of DBL_MAX the scaled value can round to inf
+        // while strtod stays finite.  Decide by decimal magnitude:
+        // value ~= mag * 10^dec_exp.
         int mexp = 0;
         for (uint64_t m = mag; m >= 10; m /= 10) ++mexp;
         int vexp = dec_exp + mexp;
@@ -420,6 +431,15 @@ DEV bool parse_number_text(const uint8_t* s, uint32_t p, uint32_t e, NumVal* nv)
     }
   }
   if (neg) d = -d;
+  // correctness boundaries -> host strtod (counted fallback):
+  //  * >19-20 significant digits were rounded into mag, so values near
+  //    the 19-digit rounding boundary can double-round one ulp off
+  //  * the dd scaler loses precision approaching the subnormal range
+  if (overflow) nv->imprecise = true;
+  {
+    double ad = d < 0 ? -d : d;
+    if (ad != 0.0 && ad < 1e-306) nv->imprecise = true;
+  }
   // integral double that fits -> also expose integer view
   nv->cls = 1;
   nv->neg = neg;
@@ -927,6 +947,7 @@ DEV bool encode_packed_element(Ctx& c, const FieldEntry& f) {
       else if (nonf == -1) d = -HUGE_VAL;
       else if (nonf == 2) d = nan("");
       else d = nv.cls == 0 ? (nv.neg ? -(double)nv.mag : (double)nv.mag) : nv.d;
+      if (nv.imprecise && !nonf) return fail(c, E_UNSUPPORTED, (int)f.number);
       if (f.kind == K_DOUBLE) return emit_fixed64(c, __builtin_bit_cast(uint64_t, d));
       return emit_fixed32(c, __builtin_bit_cast(uint32_t, (float)d));
     }

@@ -217,48 +217,6 @@ DEV double scale10(double d, int e) {
   return d / DPOW10[-e];
 }
 
-// ---------------------------------------------------------------------------
-// double-double arithmetic for near-exact decimal scaling.  A plain
-// `d * 10^e` carries ~1 ulp of error per step — enough to generate
-// 17-digit blocks that are off by one AND to mis-verify their parse-back
-// (found: 301/3 printed as a string that parses to a DIFFERENT double).
-// With an fma-compensated (hi, lo) pair the product of an exact power
-// step is correct to ~2^-100, so digit generation and round-trip
-// verification are exact for every double the kernel handles.
-// ---------------------------------------------------------------------------
-
-struct DD {
-  double hi, lo;
-};
-
-// (hi, lo) * p for EXACT p (a power of ten <= 1e22): Dekker product via fma.
-// The fp pragmas are load-bearing: hipcc's device default
-// (-ffp-contract=fast-honor-pragmas) reassociated this fma chain and
-// applied the compensation term TWICE (measured on gfx950: 301/3 * 1e14
-// produced hi one step high and lo = e1 - 2 instead of {p1, e1}).
-DEV DD dd_mul_exact(DD a, double p) {
-#pragma clang fp contract(off) reassociate(off)
-  double p1 = a.hi * p;
-  double e1 = fma(a.hi, p, -p1);
-  double lo = fma(a.lo, p, e1);
-  DD r;
-  r.hi = p1 + lo;
-  r.lo = (p1 - r.hi) + lo;
-  return r;
-}
-
-DEV DD dd_div_exact(DD a, double p) {
-#pragma clang fp contract(off) reassociate(off)
-  double q1 = a.hi / p;
-  // residual of the first quotient: a - q1*p, computed exactly
-  double r1 = fma(-q1, p, a.hi) + a.lo;
-  double q2 = r1 / p;
-  DD r;
-  r.hi = q1 + q2;
-  r.lo = (q1 - r.hi) + q2;
-  return r;
-}
-
 // d * 10^e as a double-double (steps through exact powers)
 DEV DD dd_scale10(double d, int e) {
   DD x{d, 0.0};

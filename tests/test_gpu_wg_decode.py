@@ -50,6 +50,20 @@ def env():
     srv.stop()
 
 
+def _first_diff(a, b, path="$"):
+    if isinstance(a, dict) and isinstance(b, dict):
+        for k in sorted(set(a) | set(b)):
+            if a.get(k) != b.get(k):
+                return _first_diff(a.get(k), b.get(k), f"{path}.{k}")
+    if isinstance(a, list) and isinstance(b, list):
+        if len(a) != len(b):
+            return f"{path}: len {len(a)} vs {len(b)}"
+        for j, (x, y) in enumerate(zip(a, b)):
+            if x != y:
+                return _first_diff(x, y, f"{path}[{j}]")
+    return f"{path}: {a!r} != {b!r}"
+
+
 def _body(args, rid):
     return json.dumps(
         {"jsonrpc": "2.0", "id": rid, "method": "tools/call",
@@ -137,7 +151,7 @@ def test_wg_decode_matches_protojson(env):
         inner = json.loads(resp["result"]["content"][0]["text"])
         wire = pipeline.cpu.json_to_pb(mi.input_descriptor, json.dumps(args))
         oracle = json.loads(pipeline.cpu.pb_to_json(mi.output_descriptor, wire))
-        assert inner == oracle, f"slot {i} mismatch"
+        assert inner == oracle, f"slot {i}: {_first_diff(inner, oracle)}"
     assert pipeline.engine.stats.host_fallbacks == fallbacks_before, (
         "every shape must complete on the GPU path")
 

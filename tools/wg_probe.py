@@ -127,3 +127,81 @@ def diag_doubles():
 
 if kind == "dbl":
     diag_doubles()
+
+def diag_wg_diff(which: int):
+    """kind=wgdiffN: shape N through encode(mode0)+decode(mode0) — the
+    workgroup kernel path — diffed against protojson."""
+    sys.path.insert(0, str(Path(__file__).resolve().parent.parent / "tests"))
+    from test_gpu_wg_decode import _shapes
+
+    args = _shapes()[which]
+    body = json.dumps({"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                       "params": {"name": "bench_echoservice_echo",
+                                  "arguments": args}}).encode()
+    enc, pbs = eng.encode_batch([body], mode=0)
+    assert enc[0]["status"] == 0, enc[0]
+    wire = pbs[0]
+    out_idx = eng.tables.msg_index[mi.output_descriptor.full_name]
+    dec, outs = eng.decode_batch([wire], [out_idx], mode=0, skip=[False])
+    print("wire", len(wire), "dec status", dec[0]["status"], flush=True)
+    if dec[0]["status"] != 0:
+        return
+    resp = json.loads(outs[0])
+    kern = json.loads(resp["result"]["content"][0]["text"])
+    orac = json.loads(cpu.pb_to_json(mi.output_descriptor, wire))
+    for k in sorted(set(orac) | set(kern)):
+        if kern.get(k) != orac.get(k):
+            a, b = kern.get(k), orac.get(k)
+            if isinstance(a, list) and isinstance(b, list):
+                print(f"len {len(a)} vs {len(b)}", flush=True)
+                for j, (x, y) in enumerate(zip(a, b)):
+                    if x != y:
+                        print(f"field {k}[{j}]:\n  kern {x}\n  orac {y}", flush=True)
+                        return
+            print(f"field {k}:\n  kern {str(a)[:300]}\n  orac {str(b)[:300]}", flush=True)
+            return
+    print("no diff (wg path)", flush=True)
+
+if kind.startswith("wgdiff"):
+    diag_wg_diff(int(kind[len("wgdiff"):]))
+
+def diag_batch6():
+    """kind=batch6: all test shapes in ONE mode-0 batch (multi-block wg)."""
+    sys.path.insert(0, str(Path(__file__).resolve().parent.parent / "tests"))
+    from test_gpu_wg_decode import _shapes
+
+    shapes = _shapes()
+    bodies = [json.dumps({"jsonrpc": "2.0", "id": i + 1, "method": "tools/call",
+                          "params": {"name": "bench_echoservice_echo",
+                                     "arguments": a}}).encode()
+              for i, a in enumerate(shapes)]
+    enc, pbs = eng.encode_batch(bodies, mode=0)
+    out_idx = eng.tables.msg_index[mi.output_descriptor.full_name]
+    dec, outs = eng.decode_batch(pbs, [out_idx] * len(pbs), mode=0,
+                                 skip=[False] * len(pbs))
+    for i, (w, r, o) in enumerate(zip(pbs, dec, outs)):
+        if r["status"] != 0:
+            print(f"slot {i}: dec status {r['status']} (host fallback)", flush=True)
+            continue
+        resp = json.loads(o)
+        kern = json.loads(resp["result"]["content"][0]["text"])
+        orac = json.loads(cpu.pb_to_json(mi.output_descriptor, w))
+        if kern == orac:
+            print(f"slot {i}: ok", flush=True)
+            continue
+        for k in sorted(set(orac) | set(kern)):
+            if kern.get(k) != orac.get(k):
+                a, b = kern.get(k), orac.get(k)
+                if isinstance(a, list) and isinstance(b, list):
+                    for j, (x, y) in enumerate(zip(a, b)):
+                        if x != y:
+                            print(f"slot {i} field {k}[{j}]:\n  kern {x}\n  orac {y}", flush=True)
+                            break
+                    else:
+                        print(f"slot {i} field {k}: len {len(a)} vs {len(b)}", flush=True)
+                else:
+                    print(f"slot {i} field {k}:\n  kern {str(a)[:200]}\n  orac {str(b)[:200]}", flush=True)
+                break
+
+if kind == "batch6":
+    diag_batch6()
