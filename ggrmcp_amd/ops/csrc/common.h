@@ -290,12 +290,15 @@ DEV DD dd_div_exact(DD a, double p) {
 
 // collapse a dd-scaled value d*10^e back to one double (correct to
 // ~2^-100 through composed steps; inf/zero saturate early)
+// POSITIVE input only (the parser applies the sign afterwards); overflow
+// saturates to +inf for the caller's range checks
 DEV double dd_scale_collapse(double hi, double lo, int e) {
   DD x{hi, lo};
   while (e > 22) {
     x = dd_mul_exact(x, 1e22);
     e -= 22;
-    if (x.hi > 1.7e308) return x.hi;  // overflowing; caller range-checks
+    // !(x < bound) also catches the NaN a mid-chain inf cascades into
+    if (!(x.hi < 1.7e308)) return HUGE_VAL;
   }
   while (e < -22) {
     x = dd_div_exact(x, 1e22);
@@ -303,7 +306,13 @@ DEV double dd_scale_collapse(double hi, double lo, int e) {
     if (x.hi == 0.0) return 0.0;
   }
   x = e >= 0 ? dd_mul_exact(x, DD_POW10[e]) : dd_div_exact(x, DD_POW10[-e]);
-  return x.hi + x.lo;
+  double r = x.hi + x.lo;
+  // at DBL_MAX the renormalization can hit inf with a -inf compensation
+  // term, collapsing to NaN for a FINITE true value (found: the exact
+  // DBL_MAX decimal encoded as NaN); saturate so callers route the
+  // boundary to the host
+  if (r != r) return HUGE_VAL;
+  return r;
 }
 
 // nesting cap for the iterative encode/decode walkers' explicit frame

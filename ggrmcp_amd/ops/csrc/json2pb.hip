@@ -1121,6 +1121,8 @@ DEVN int encode_value_scalar_or_classify(Ctx& c, uint32_t* slot) {
   int nonf;
   if (!parse_numeric_value(c, &nv, false, &nonf)) return 0;
   double d = nv.cls == 0 ? (nv.neg ? -(double)nv.mag : (double)nv.mag) : nv.d;
+  // boundary decimals (DBL_MAX edge, >20 digits, near-denormal) -> host
+  if (nv.imprecise) return fail(c, E_UNSUPPORTED) ? 1 : 0;
   if (!emit_tag(c, 2, W_I64)) return 0;
   return emit_fixed64(c, __builtin_bit_cast(uint64_t, d)) ? 1 : 0;
 }
