@@ -98,11 +98,13 @@ def _shapes():
     # default-valued singulars between real fields (empty items -> no
     # stray commas)
     shapes.append({
+        # field-number order (the kernels' single-pass subset expects
+        # ascending wire fields; JSON order drives the encoder's emission)
         "f01String": "a" * 1000,
         "f02Int32": 0,          # proto3 default: omitted from output
         "f03Int64": "0",
-        "f05Bool": False,
         "f04Double": 1.5,
+        "f05Bool": False,
         "attrs": {f"z{j}": "w" * 900 for j in range(16)},
     })
     # doubles / int64 precision paths at volume (repeated Inner)
