@@ -1731,6 +1731,12 @@ extern "C" __global__ void __launch_bounds__(WG_DEC_WAVES * WAVE) k_pb2json_wg(
   constexpr int MAX_SLICES = 1024;
   __shared__ uint32_t s_start[WG_DEC_MAX_ITEMS + 1];
   __shared__ uint32_t s_outlen[WG_DEC_MAX_ITEMS];
+  // map / repeated runs are CHUNKED at entry boundaries so one big run
+  // doesn't serialize on a single wave; each chunk's walk emits the full
+  // `"name":{...}` / `"name":[...]` syntax and the joiner trims
+  // s_head bytes of leading name syntax and s_tail closing brackets
+  __shared__ uint16_t s_head[WG_DEC_MAX_ITEMS];
+  __shared__ uint8_t s_tail[WG_DEC_MAX_ITEMS];
   __shared__ uint16_t s_sl_item[MAX_SLICES];
   __shared__ uint32_t s_sl_off[MAX_SLICES];   // offset within the item text
   __shared__ uint32_t s_sl_esc[MAX_SLICES];   // escaped length of the slice
@@ -1750,13 +1756,21 @@ extern "C" __global__ void __launch_bounds__(WG_DEC_WAVES * WAVE) k_pb2json_wg(
   const int msg_idx = msg_idx_arr[req];
 
   // ---- phase A: top-level field-run scan (one thread) ---------------------
+  // Runs group contiguous same-number tags; MAP / REPEATED runs split into
+  // ~4 KB chunks at entry boundaries (chunk walks re-emit the name syntax;
+  // s_head/s_tail trim it at join time) so a payload-dominating map field
+  // parallelizes across waves instead of serializing on one.
   if (threadIdx.x == 0) {
     s_next = 0;
     s_err = E_OK;
     s_mode = 1;
     int n = 0;
     uint32_t pos = 0, prev_num = 0, cur = 0xFFFFFFFFu;
-    bool ok = t.msgs[msg_idx].wkt_kind == WKT_NONE;
+    uint32_t chunk_head = 0;   // head trim for continuation chunks of cur
+    bool cur_split = false;    // cur run may chunk at entry boundaries
+    uint32_t chunk_wire = 0;   // wire bytes accumulated in the open chunk
+    const MsgEntry& tm = t.msgs[msg_idx];
+    bool ok = tm.wkt_kind == WKT_NONE;
     while (ok && pos < wire_len) {
       uint32_t tag_start = pos;
       uint64_t tag;
@@ -1764,12 +1778,45 @@ extern "C" __global__ void __launch_bounds__(WG_DEC_WAVES * WAVE) k_pb2json_wg(
       uint32_t num = (uint32_t)(tag >> 3), wt = (uint32_t)(tag & 7);
       if (num == 0) { ok = false; break; }
       if (num != cur) {
+        // close the previous run: its last chunk keeps the bracket
+        if (n > 0) s_tail[n - 1] = 0;
         // new run must be strictly ascending (non-adjacent duplicates are
         // the classic path's E_UNSUPPORTED -> host fallback)
         if (num < prev_num || n >= WG_DEC_MAX_ITEMS) { ok = false; break; }
+        // field lookup decides chunkability (maps + per-entry repeated)
+        cur_split = false;
+        chunk_head = 0;
+        for (int fi = 0; fi < tm.field_count; ++fi) {
+          const FieldEntry& fe = t.fields[tm.field_start + fi];
+          if (fe.number != num) continue;
+          bool per_entry_len =
+              (fe.flags & F_MAP) ||
+              ((fe.flags & F_REPEATED) &&
+               (fe.kind == K_MESSAGE || fe.kind == K_STRING ||
+                fe.kind == K_BYTES));
+          // unpacked repeated scalars (wt != W_LEN run) also chunk
+          bool per_entry_scalar =
+              (fe.flags & F_REPEATED) && !(fe.flags & F_MAP) && wt != W_LEN;
+          if (per_entry_len || per_entry_scalar) {
+            cur_split = true;
+            // `"name":{` / `"name":[` = 1 + json_len + 2 + 1 bytes
+            chunk_head = (uint32_t)fe.json_len + 4;
+          }
+          break;
+        }
+        s_head[n] = 0;
+        s_tail[n] = (uint8_t)(cur_split ? 1 : 0);
         s_start[n++] = tag_start;
         prev_num = num;
         cur = num;
+        chunk_wire = 0;
+      } else if (cur_split && chunk_wire >= 4096) {
+        // continuation chunk of the same run at an entry boundary
+        if (n >= WG_DEC_MAX_ITEMS) { ok = false; break; }
+        s_head[n] = (uint16_t)chunk_head;
+        s_tail[n] = 1;
+        s_start[n++] = tag_start;
+        chunk_wire = 0;
       }
       if (wt == W_VARINT) {
         uint64_t v;
@@ -1789,8 +1836,10 @@ extern "C" __global__ void __launch_bounds__(WG_DEC_WAVES * WAVE) k_pb2json_wg(
         ok = false;
         break;
       }
+      chunk_wire += pos - tag_start;
     }
     if (ok && pos == wire_len && n > 0) {
+      s_tail[n - 1] = 0;  // the wire's last chunk keeps its bracket
       s_start[n] = wire_len;
       s_nitems = n;
     } else {
@@ -1830,14 +1879,17 @@ extern "C" __global__ void __launch_bounds__(WG_DEC_WAVES * WAVE) k_pb2json_wg(
       c.status = E_OK;
       c.lane = lane;
       bool ok = decode_walk(c, msg_idx, ien, /*bare_top=*/true);
-      if (!ok || c.status != E_OK) {
+      uint32_t trim = (uint32_t)s_head[idx] + s_tail[idx];
+      if (!ok || c.status != E_OK || c.opos < trim) {
         if (!lane) {
           atomicCAS(&s_err, E_OK, c.status == E_OK ? E_PARSE : c.status);
           s_outlen[idx] = 0;
         }
         continue;
       }
-      if (!lane) s_outlen[idx] = c.opos;
+      // store the TRIMMED length (chunk joins drop re-emitted name syntax
+      // and the premature closing bracket); src offsets add s_head later
+      if (!lane) s_outlen[idx] = c.opos - trim;
     }
   }
   __syncthreads();
@@ -1881,8 +1933,9 @@ extern "C" __global__ void __launch_bounds__(WG_DEC_WAVES * WAVE) k_pb2json_wg(
       uint32_t ioff = s_sl_off[s];
       uint32_t ilen = s_outlen[item] - ioff;
       if (ilen > s_slice_bytes) ilen = s_slice_bytes;
-      const uint8_t* src =
-          scr + 8u * s_start[item] + WG_DEC_ITEM_PAD * (uint32_t)item + ioff;
+      const uint8_t* src = scr + 8u * s_start[item] +
+                           WG_DEC_ITEM_PAD * (uint32_t)item + s_head[item] +
+                           ioff;
       uint32_t acc = 0;
       for (uint32_t i = lane; i < ilen; i += WAVE) acc += esc_len(src[i]);
       for (int d = WAVE / 2; d > 0; d >>= 1) acc += __shfl_down(acc, d, WAVE);
@@ -1954,8 +2007,9 @@ extern "C" __global__ void __launch_bounds__(WG_DEC_WAVES * WAVE) k_pb2json_wg(
       uint32_t ioff = s_sl_off[s];
       uint32_t ilen = s_outlen[item] - ioff;
       if (ilen > s_slice_bytes) ilen = s_slice_bytes;
-      const uint8_t* src =
-          scr + 8u * s_start[item] + WG_DEC_ITEM_PAD * (uint32_t)item + ioff;
+      const uint8_t* src = scr + 8u * s_start[item] +
+                           WG_DEC_ITEM_PAD * (uint32_t)item + s_head[item] +
+                           ioff;
       if (s_sl_comma[s] && !lane) fout[s_sl_fin[s] - 1] = ',';
       DCtx e;
       e.pb = nullptr;
