@@ -343,6 +343,18 @@ constexpr uint32_t WG_DEC_EXTRA =
     WG_DEC_ITEM_PAD * (uint32_t)WG_DEC_MAX_ITEMS + 1024;
 constexpr int WG_DEC_WAVES = 8;  // waves per request workgroup
 
+// Workgroup-cooperative ENCODE (k_json2pb_wg): JSON-RPC requests at least
+// WG_ENC_MIN_BYTES long split their arguments object into top-level-member
+// items (map/array members chunk at entry boundaries — protobuf wire
+// concatenation makes chunked runs valid without any joining fix-ups).
+// Items encode into a scratch staging area (2x the slot's pb arena span,
+// carved from the decode scratch arena, idle during encode) and compact
+// into the pb arena.
+constexpr uint32_t WG_ENC_MIN_BYTES = 16384;
+constexpr int WG_ENC_MAX_ITEMS = 384;
+constexpr uint32_t WG_ENC_ITEM_PAD = 160;
+constexpr int WG_ENC_WAVES = 8;
+
 // u64 -> decimal text without an addressable temp buffer (a local tmp[20]
 // array lands in scratch memory and costs a private-memory round trip per
 // digit; emitting MSB-first via the power table keeps everything in VGPRs)

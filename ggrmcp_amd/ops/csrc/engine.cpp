@@ -80,6 +80,11 @@ inline uint32_t wg_dec_min() {
   return e ? (uint32_t)strtoul(e, nullptr, 10) : WG_DEC_MIN_BYTES;
 }
 
+inline uint32_t wg_enc_min() {
+  const char* e = getenv("GGRMCP_WG_ENC_MIN");
+  return e ? (uint32_t)strtoul(e, nullptr, 10) : WG_ENC_MIN_BYTES;
+}
+
 // debug bisection: limit the wg kernel to its first N phases
 inline int wg_phases() {
   const char* e = getenv("GGRMCP_WG_PHASES");
@@ -943,6 +948,19 @@ class Engine : public spanapi::ISpanExecutor {
   void run_encode_device(int n, size_t in_bytes, size_t pb_bytes, bool has_idx,
                          Limits lim, int mode) {
     uint32_t* h_off = (uint32_t*)h_off_.p;
+    // route big mode-0 requests to the workgroup-cooperative encode (one
+    // request per workgroup; items stage in the decode scratch arena,
+    // which is idle during the encode phase).  h_aux is free in mode 0.
+    int n_wg = 0;
+    if (mode == 0 && !has_idx && 2 * pb_bytes <= d_scratch_.n) {
+      uint32_t thr = wg_enc_min();
+      int32_t* esk = (int32_t*)h_aux_.p;
+      for (int i = 0; i < n; ++i) {
+        uint32_t len = h_off[i + 1] - h_off[i];
+        esk[i] = len >= thr ? 2 : 0;
+        n_wg += esk[i] ? 1 : 0;
+      }
+    }
     auto g0 = std::chrono::steady_clock::now();
     HIP_CHECK(hipSetDevice(device_));
     HIP_CHECK(hipMemcpyAsync(d_in_.p, h_in_.p, in_bytes,
@@ -950,7 +968,7 @@ class Engine : public spanapi::ISpanExecutor {
     uint32_t* d_off = (uint32_t*)d_off3_.p;  // in_off | pb_off contiguous
     HIP_CHECK(hipMemcpyAsync(d_off, h_off, 2 * (n + 1) * sizeof(uint32_t),
                              hipMemcpyHostToDevice, stream_));
-    if (has_idx)
+    if (has_idx || n_wg > 0)
       HIP_CHECK(hipMemcpyAsync(d_aux2_.p, h_aux_.p, n * sizeof(int32_t),
                                hipMemcpyHostToDevice, stream_));
     int blocks = (int)cdiv(n, WPB);
@@ -961,7 +979,18 @@ class Engine : public spanapi::ISpanExecutor {
                          (const uint32_t*)d_off + (n + 1),
                          (SlotResult*)d_results_.p, (uint8_t*)d_id_slots_.p,
                          has_idx ? (const int32_t*)d_aux2_.p : nullptr,
-                         tables_, lim, n, mode);
+                         tables_, lim, n, mode,
+                         n_wg > 0 ? (const int32_t*)d_aux2_.p : nullptr);
+      HIP_CHECK(hipGetLastError());
+    }
+    if (n_wg > 0) {
+      hipLaunchKernelGGL(k_json2pb_wg, dim3(n), dim3(WG_ENC_WAVES * WAVE), 0,
+                         stream_, (const uint8_t*)d_in_.p,
+                         (const uint32_t*)d_off, (uint8_t*)d_pb_.p,
+                         (const uint32_t*)d_off + (n + 1),
+                         (SlotResult*)d_results_.p, (uint8_t*)d_id_slots_.p,
+                         (uint8_t*)d_scratch_.p, tables_, lim, n,
+                         (const int32_t*)d_aux2_.p);
       HIP_CHECK(hipGetLastError());
     }
     HIP_CHECK(hipMemcpyAsync(h_results_.p, d_results_.p,

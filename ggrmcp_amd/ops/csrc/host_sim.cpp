@@ -115,7 +115,8 @@ class HostEngine {
       for (int w = 0; w < WPB; ++w) {
         blockIdx.x = b;
         threadIdx.x = (unsigned)(w * WAVE);
-        k_json2pb(in, in_off, pb, pb_off, results, ids, mi, t_, lim, n, mode);
+        k_json2pb(in, in_off, pb, pb_off, results, ids, mi, t_, lim, n, mode,
+                  nullptr);
       }
     }
   }

@@ -1420,7 +1420,12 @@ DEVN bool encode_packed_array(Ctx& c, const FieldEntry& f) {
   return backfill_len(c, slot);
 }
 
-DEV bool encode_walk(Ctx& c, int top_msg_idx) {
+// wg_item: workgroup-encode item modes (k_json2pb_wg). 0 = classic whole
+// value; 1 = one top-level member `"key": value`; 2 = map-entry span of
+// field wg_fidx (`"k":v, ...`, no braces); 3 = array-element span of
+// wg_fidx.  Item spans END the walk at c.len instead of a closing brace.
+DEV bool encode_walk(Ctx& c, int top_msg_idx, int wg_item = 0,
+                     int wg_fidx = -1) {
   EFrame stack[MAX_RECURSE];
   int sp = 0;
   int depth_cap = (int)MAX_RECURSE;
@@ -1443,7 +1448,18 @@ DEV bool encode_walk(Ctx& c, int top_msg_idx) {
   } while (0)
 
   // ---- top-level entry ----
-  {
+  if (wg_item) {
+    uint32_t slots0[3];
+    EPUSH(EM_BODY, top_msg_idx, slots0, 0);
+    if (wg_item == 2) {
+      stack[0].cont_field = wg_fidx;
+      stack[0].resume = RB_MAPKEY;
+    } else if (wg_item == 3) {
+      stack[0].cont_field = wg_fidx;
+      stack[0].resume = RB_ARR;
+    }
+    // RB_KEY expects the cursor at the member's key quote
+  } else {
     const MsgEntry& m = c.t.msgs[top_msg_idx];
     uint32_t slots0[3];
     if (m.wkt_kind == WKT_VALUE) {
@@ -1573,6 +1589,10 @@ DEV bool encode_walk(Ctx& c, int top_msg_idx) {
       }
       case RB_SEP: {
         skip_ws(c);
+        if (wg_item == 1 && sp == 1 && c.pos >= c.len) {
+          --sp;  // member item ends at its span, no brace to consume
+          continue;
+        }
         uint8_t ch = peek(c);
         if (ch == ',') {
           c.pos++;
@@ -1622,6 +1642,10 @@ DEV bool encode_walk(Ctx& c, int top_msg_idx) {
       }
       case RB_ARRSEP: {
         skip_ws(c);
+        if (wg_item == 3 && sp == 1 && c.pos >= c.len) {
+          --sp;  // element-span item: no ']' in the span
+          continue;
+        }
         uint8_t ch = peek(c);
         if (ch == ',') {
           c.pos++;
@@ -1685,6 +1709,10 @@ DEV bool encode_walk(Ctx& c, int top_msg_idx) {
       }
       case RB_MAPSEP: {
         skip_ws(c);
+        if (wg_item == 2 && sp == 1 && c.pos >= c.len) {
+          --sp;  // entry-span item: no '}' in the span
+          continue;
+        }
         uint8_t ch = peek(c);
         if (ch == ',') {
           c.pos++;
@@ -1804,7 +1832,11 @@ DEV bool encode_walk(Ctx& c, int top_msg_idx) {
 // envelope (mode 0): full JSON-RPC tools/call request
 // ---------------------------------------------------------------------------
 
-DEV bool parse_envelope(Ctx& c, SlotResult& r, uint8_t* id_slot) {
+// wg_args non-null: stop after envelope validation + tool resolution and
+// return the arguments span as {pos, end} (0xFFFFFFFF = absent) instead of
+// encoding — the workgroup encode kernel splits the arguments itself.
+DEV bool parse_envelope(Ctx& c, SlotResult& r, uint8_t* id_slot,
+                        uint32_t* wg_args = nullptr) {
   bool saw_jsonrpc = false, jsonrpc_ok = false;
   bool saw_method = false, method_ok = false;
   bool saw_id = false;
@@ -1934,9 +1966,15 @@ DEV bool parse_envelope(Ctx& c, SlotResult& r, uint8_t* id_slot) {
   if (c.t.tools[tool].flags & 1) r.flags |= SR_SERVER_STREAMING;
 
   // ---- encode arguments ----
-  if (args_pos == 0xFFFFFFFF) return true;  // no arguments -> empty message
-  if (c.lim.enforce && args_end - args_pos > c.lim.max_args_bytes)
+  if (args_pos != 0xFFFFFFFF && c.lim.enforce &&
+      args_end - args_pos > c.lim.max_args_bytes)
     return fail(c, E_LIMIT, 3);
+  if (wg_args) {
+    wg_args[0] = args_pos;
+    wg_args[1] = args_end;
+    return true;
+  }
+  if (args_pos == 0xFFFFFFFF) return true;  // no arguments -> empty message
   uint32_t save_len = c.len;
   c.pos = args_pos;
   c.len = args_end;
@@ -1954,11 +1992,12 @@ extern "C" __global__ void __launch_bounds__(WPB * WAVE) k_json2pb(
     uint8_t* __restrict__ pb_arena, const uint32_t* __restrict__ pb_off,
     SlotResult* __restrict__ results, uint8_t* __restrict__ id_slots,
     const int32_t* __restrict__ msg_idx_in, Tables t, Limits lim, int n_req,
-    int mode) {
+    int mode, const int32_t* __restrict__ enc_skip) {
   __shared__ uint8_t keybufs[WPB][192];
   int wave_in_block = threadIdx.x / WAVE;
   for (int req = blockIdx.x * WPB + wave_in_block; req < n_req;
        req += gridDim.x * WPB) {
+    if (enc_skip && enc_skip[req]) continue;  // k_json2pb_wg owns this slot
     Ctx c;
     c.s = in_bytes + in_off[req];
     c.len = in_off[req + 1] - in_off[req];
@@ -2004,3 +2043,398 @@ extern "C" __global__ void __launch_bounds__(WPB * WAVE) k_json2pb(
     if (!c.lane) results[req] = r;
   }
 }
+
+#ifndef GGRMCP_HOST_SIM
+// ---------------------------------------------------------------------------
+// k_json2pb_wg — workgroup-per-request ENCODE for large requests
+// (BASELINE config 3: 64 KB JSON bodies).
+//
+// One workgroup (WG_ENC_WAVES wave64s) owns one request:
+//   A. wave 0 parses/validates the JSON-RPC envelope and resolves the tool
+//      (parse_envelope with wg_args), then scans the arguments object into
+//      top-level-member items; map members and non-packed repeated members
+//      chunk at entry/element boundaries (~4 KB) — protobuf wire
+//      CONCATENATION makes per-chunk emission valid with no joining
+//      fix-ups (repeated/map runs may split across tags).
+//   B. waves grab items dynamically and encode each with the SAME
+//      encode_walk machinery (wg_item modes) into private scratch regions
+//      carved from the decode scratch arena (idle during encode).
+//   C. one thread prefix-sums item lengths; D. waves compact the items
+//      into the slot's pb arena span.
+// Any anomaly (escaped member keys, duplicate fields, item errors, caps)
+// falls back to the classic single-wave encode inside the block, which
+// reproduces the exact classic error formats.  The engine routes mode-0
+// slots >= WG_ENC_MIN_BYTES here (skip tag 2).
+// ---------------------------------------------------------------------------
+
+// classic whole-slot mode-0 encode (the in-block fallback path); mirrors
+// the k_json2pb loop body exactly
+DEV void wg_enc_classic_one(const uint8_t* src, uint32_t src_len,
+                            uint8_t* pbout, uint32_t pbcap,
+                            SlotResult* results, uint8_t* id_slot,
+                            uint8_t* keybuf, Tables t, Limits lim, int req,
+                            int lane) {
+  Ctx c;
+  c.s = src;
+  c.len = src_len;
+  c.pos = 0;
+  c.out = pbout;
+  c.opos = 0;
+  c.ocap = pbcap;
+  c.t = t;
+  c.lim = lim;
+  c.status = E_OK;
+  c.err_pos = 0;
+  c.aux = 0;
+  c.lane = lane;
+  c.keybuf = keybuf;
+  SlotResult r;
+  r.status = E_OK;
+  r.tool_idx = -1;
+  r.pb_off = 0;  // caller fixes the arena offset
+  r.pb_len = 0;
+  r.err_pos = 0;
+  r.aux = 0;
+  r.id_len = 0;
+  r.flags = 0;
+  (void)parse_envelope(c, r, id_slot);
+  r.status = c.status;
+  r.err_pos = c.err_pos;
+  r.aux = c.aux;
+  r.pb_len = c.status == E_OK ? c.opos : 0;
+  if (!lane) {
+    results[req].status = r.status;
+    results[req].tool_idx = r.tool_idx;
+    results[req].pb_len = r.pb_len;
+    results[req].err_pos = r.err_pos;
+    results[req].aux = r.aux;
+    results[req].id_len = r.id_len;
+    results[req].flags = r.flags;
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(WG_ENC_WAVES * WAVE) k_json2pb_wg(
+    const uint8_t* __restrict__ in_bytes, const uint32_t* __restrict__ in_off,
+    uint8_t* __restrict__ pb_arena, const uint32_t* __restrict__ pb_off,
+    SlotResult* __restrict__ results, uint8_t* __restrict__ id_slots,
+    uint8_t* __restrict__ enc_scratch, Tables t, Limits lim, int n_req,
+    const int32_t* __restrict__ skip) {
+  int req = blockIdx.x;
+  if (req >= n_req) return;
+  if (!skip || skip[req] != 2) return;
+
+  __shared__ uint32_t s_start[WG_ENC_MAX_ITEMS];
+  __shared__ uint32_t s_end[WG_ENC_MAX_ITEMS];
+  __shared__ uint32_t s_ioff[WG_ENC_MAX_ITEMS];
+  __shared__ uint32_t s_olen[WG_ENC_MAX_ITEMS];
+  __shared__ uint32_t s_foff[WG_ENC_MAX_ITEMS];
+  __shared__ int32_t s_fidx[WG_ENC_MAX_ITEMS];
+  __shared__ uint8_t s_kind[WG_ENC_MAX_ITEMS];
+  __shared__ int s_nitems, s_next, s_mode, s_msg;
+  __shared__ SlotResult s_res;
+  __shared__ uint8_t keybufs[WG_ENC_WAVES][192];
+
+  const uint8_t* src = in_bytes + in_off[req];
+  const uint32_t src_len = in_off[req + 1] - in_off[req];
+  uint8_t* pbout = pb_arena + pb_off[req];
+  const uint32_t pbcap = pb_off[req + 1] - pb_off[req];
+  // scratch staging: 2x the slot's pb span (engine verified capacity)
+  uint8_t* scr = enc_scratch + 2ull * pb_off[req];
+  const uint32_t scr_cap = 2u * pbcap;
+  const int lane = lane_id();
+  const int wave = threadIdx.x / WAVE;
+
+  // ---- phase A: envelope + member scan (wave 0: the string helpers are
+  // wave-cooperative ballots) ----------------------------------------------
+  if (wave == 0) {
+    Ctx c;
+    c.s = src;
+    c.len = src_len;
+    c.pos = 0;
+    c.out = pbout;
+    c.opos = 0;
+    c.ocap = pbcap;
+    c.t = t;
+    c.lim = lim;
+    c.status = E_OK;
+    c.err_pos = 0;
+    c.aux = 0;
+    c.lane = lane;
+    c.keybuf = keybufs[0];
+    SlotResult r;
+    r.status = E_OK;
+    r.tool_idx = -1;
+    r.pb_off = 0;
+    r.pb_len = 0;
+    r.err_pos = 0;
+    r.aux = 0;
+    r.id_len = 0;
+    r.flags = 0;
+    uint32_t args[2] = {0xFFFFFFFFu, 0};
+    bool ok = parse_envelope(c, r, id_slots + (size_t)req * ID_SLOT_BYTES,
+                             args);
+    int mode_l;
+    int n = 0;
+    int msg_idx = 0;
+    if (!ok || c.status != E_OK) {
+      r.status = c.status;
+      r.err_pos = c.err_pos;
+      r.aux = c.aux;
+      mode_l = -1;  // final: error envelope (host/kernel error mapping)
+    } else if (args[0] == 0xFFFFFFFFu) {
+      mode_l = -1;  // no arguments -> empty message, final
+    } else {
+      msg_idx = t.tools[r.tool_idx].in_msg;
+      const MsgEntry& m = t.msgs[msg_idx];
+      c.pos = args[0];
+      c.len = args[1];
+      uint64_t seen = 0;
+      uint32_t seen_oneof = 0;
+      uint32_t acc = 0;
+      bool fallback = m.wkt_kind != WKT_NONE;
+      mode_l = 1;
+      if (!fallback && expect(c, '{')) {
+        skip_ws(c);
+        if (peek(c) == '}') {
+          c.pos++;  // empty arguments object: zero items
+        } else {
+          while (true) {
+            skip_ws(c);
+            uint32_t kst, krl;
+            bool kesc;
+            if (!string_span(c, &kst, &krl, &kesc) || kesc) {
+              fallback = true;  // escaped keys: classic handles
+              break;
+            }
+            uint64_t h = fnv1a64(c.s + kst, krl);
+            int fidx = -1;
+            for (int i = 0; i < m.field_count; ++i) {
+              const FieldEntry& fe = t.fields[m.field_start + i];
+              if ((fe.hash_json == h && fe.json_len == krl &&
+                   wave_equal(c, t.names + fe.json_off, c.s + kst, krl)) ||
+                  (fe.hash_orig == h && fe.name_len == krl &&
+                   wave_equal(c, t.names + fe.name_off, c.s + kst, krl))) {
+                fidx = i;
+                break;
+              }
+            }
+            if (fidx < 0) {  // unknown field: classic formats the error
+              fallback = true;
+              break;
+            }
+            const FieldEntry& fe = t.fields[m.field_start + fidx];
+            if (fidx < 64) {
+              if (seen & (1ull << fidx)) {
+                fallback = true;  // duplicate member -> classic error
+                break;
+              }
+              seen |= 1ull << fidx;
+            }
+            if (fe.flags & F_ONEOF) {
+              if (seen_oneof & (1u << fe.oneof_id)) {
+                fallback = true;
+                break;
+              }
+              seen_oneof |= 1u << fe.oneof_id;
+            }
+            if (!expect(c, ':')) {
+              fallback = true;
+              break;
+            }
+            skip_ws(c);
+            uint32_t member_start = kst - 1;  // include the opening quote
+            bool chunk_map = (fe.flags & F_MAP) && peek(c) == '{';
+            bool chunk_arr = (fe.flags & F_REPEATED) &&
+                             !(fe.flags & F_MAP) && !is_packable(fe.kind) &&
+                             peek(c) == '[';
+            bool big = (c.len - c.pos) >= 2048;
+            if ((chunk_map || chunk_arr) && big) {
+              // per-entry chunks: spans cover entries only (no key, no
+              // braces); chunks emit separate-but-concatenable runs
+              c.pos++;  // consume '{' / '['
+              skip_ws(c);
+              uint8_t closer = chunk_map ? '}' : ']';
+              if (peek(c) == closer) {
+                c.pos++;  // empty container member: no wire output at all
+              } else {
+                uint32_t chunk_start = c.pos;
+                bool closed = false;
+                while (!fallback && !closed) {
+                  // one entry
+                  if (chunk_map) {
+                    uint32_t ek, el;
+                    bool ee;
+                    if (!string_span(c, &ek, &el, &ee) || !expect(c, ':')) {
+                      fallback = true;
+                      break;
+                    }
+                    skip_ws(c);
+                  }
+                  if (!skip_value(c)) {
+                    fallback = true;
+                    break;
+                  }
+                  uint32_t entry_end = c.pos;
+                  skip_ws(c);
+                  uint8_t ch = peek(c);
+                  bool boundary = false;
+                  if (ch == ',') {
+                    c.pos++;
+                    skip_ws(c);
+                    if (entry_end - chunk_start >= 4096) boundary = true;
+                  } else if (ch == closer) {
+                    c.pos++;
+                    closed = true;
+                    boundary = true;
+                  } else {
+                    fallback = true;
+                    break;
+                  }
+                  if (boundary) {
+                    if (n >= WG_ENC_MAX_ITEMS) {
+                      fallback = true;
+                      break;
+                    }
+                    uint32_t span = entry_end - chunk_start;
+                    uint32_t icap = span + span / 4 + WG_ENC_ITEM_PAD;
+                    if (acc + icap > scr_cap) {
+                      fallback = true;
+                      break;
+                    }
+                    s_start[n] = chunk_start;
+                    s_end[n] = entry_end;
+                    s_ioff[n] = acc;
+                    s_fidx[n] = m.field_start + fidx;
+                    s_kind[n] = chunk_map ? 2 : 3;
+                    ++n;
+                    acc += icap;
+                    chunk_start = c.pos;
+                  }
+                }
+                if (fallback) break;
+              }
+            } else {
+              if (!skip_value(c)) {
+                fallback = true;
+                break;
+              }
+              if (n >= WG_ENC_MAX_ITEMS) {
+                fallback = true;
+                break;
+              }
+              uint32_t span = c.pos - member_start;
+              uint32_t icap = span + span / 4 + WG_ENC_ITEM_PAD;
+              if (acc + icap > scr_cap) {
+                fallback = true;
+                break;
+              }
+              s_start[n] = member_start;
+              s_end[n] = c.pos;
+              s_ioff[n] = acc;
+              s_fidx[n] = -1;
+              s_kind[n] = 1;
+              ++n;
+              acc += icap;
+            }
+            // member separator
+            skip_ws(c);
+            uint8_t ch = peek(c);
+            if (ch == ',') {
+              c.pos++;
+              continue;
+            }
+            if (ch == '}') {
+              c.pos++;
+              break;
+            }
+            fallback = true;
+            break;
+          }
+        }
+      } else {
+        fallback = true;
+      }
+      if (fallback) mode_l = 0;
+    }
+    if (!lane) {
+      s_res = r;
+      s_res.pb_off = pb_off[req];
+      s_mode = mode_l;
+      s_nitems = n;
+      s_next = 0;
+      s_msg = msg_idx;
+    }
+  }
+  __syncthreads();
+  const int n_items = s_nitems;
+  const int msg_idx = s_msg;
+
+  // ---- phase B: encode items (dynamic wave grabs, bounded) ----------------
+  if (s_mode == 1) {
+    for (int guard = 0; guard <= WG_ENC_MAX_ITEMS + 1; ++guard) {
+      int idx = 0;
+      if (!lane) idx = atomicAdd(&s_next, 1);
+      idx = __shfl(idx, 0, WAVE);
+      if (idx >= n_items) break;
+      uint32_t span = s_end[idx] - s_start[idx];
+      Ctx c;
+      c.s = src;
+      c.len = s_end[idx];
+      c.pos = s_start[idx];
+      c.out = scr + s_ioff[idx];
+      c.opos = 0;
+      c.ocap = span + span / 4 + WG_ENC_ITEM_PAD;
+      c.t = t;
+      c.lim = lim;
+      c.status = E_OK;
+      c.err_pos = 0;
+      c.aux = 0;
+      c.lane = lane;
+      c.keybuf = keybufs[wave];
+      bool ok = encode_walk(c, msg_idx, (int)s_kind[idx], s_fidx[idx]);
+      if (!ok || c.status != E_OK) {
+        // classic reproduces exact error position/detail formats
+        if (!lane) atomicCAS(&s_mode, 1, 0);
+        continue;
+      }
+      if (!lane) s_olen[idx] = c.opos;
+    }
+  }
+  __syncthreads();
+
+  // ---- phase C: offsets (one thread) --------------------------------------
+  if (s_mode == 1 && threadIdx.x == 0) {
+    uint32_t off = 0;
+    for (int i = 0; i < n_items; ++i) {
+      s_foff[i] = off;
+      off += s_olen[i];
+    }
+    if (off > pbcap) {
+      s_mode = 0;  // shouldn't happen (caps per item), but never truncate
+    } else {
+      s_res.pb_len = off;
+      s_next = 0;
+    }
+  }
+  __syncthreads();
+
+  // ---- phase D: compact items into the pb arena / classic fallback --------
+  if (s_mode == 1) {
+    for (int s = wave; s < n_items; s += WG_ENC_WAVES) {
+      const uint8_t* p = scr + s_ioff[s];
+      uint8_t* d = pbout + s_foff[s];
+      uint32_t len = s_olen[s];
+      for (uint32_t i = lane; i < len; i += WAVE) d[i] = p[i];
+    }
+  } else if (s_mode == 0 && wave == 0) {
+    wg_enc_classic_one(src, src_len, pbout, pbcap, results,
+                       id_slots + (size_t)req * ID_SLOT_BYTES, keybufs[0], t,
+                       lim, req, lane);
+    if (!lane) results[req].pb_off = pb_off[req];
+  }
+  __syncthreads();
+
+  // ---- finalize ------------------------------------------------------------
+  if (s_mode != 0 && threadIdx.x == 0) results[req] = s_res;
+}
+#endif  // GGRMCP_HOST_SIM
