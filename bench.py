@@ -67,8 +67,9 @@ def parse_args():
     ap.add_argument("--config", default="hello1k",
                     choices=["hello1k", "pipeline", "wide64", "stream",
                              "multi", "cpu"])
-    ap.add_argument("--client-threads", type=int, default=8,
-                    help="load-generator threads (serving config)")
+    ap.add_argument("--client-threads", type=int, default=16,
+                    help="load-generator threads (serving config); 16 "
+                         "beats 8 by ~11%% at 1024 sessions")
     ap.add_argument("--stream-depth", type=int, default=4096,
                     help="messages per stream (config 4)")
     ap.add_argument("--backends", type=int, default=4,
