@@ -244,3 +244,33 @@ def test_serving_soak_mixed(native_gateway):
     for t in ts:
         t.join()
     assert not errs, errs[:3]
+
+
+def test_serving_large_payloads_wg_kernels(native_gateway):
+    """64 KB tool calls through the SERVING path: the workgroup-cooperative
+    encode/decode kernels run inside the GIL-free span executor and the
+    responses stay protojson-exact."""
+    import random
+
+    from ggrmcp_amd.engine.cpu_ref import CpuTranscoder
+    from ggrmcp_amd.utils.synthetic import wide_payload
+
+    gw, port, pipeline = native_gateway[:3]
+    rng = random.Random(55)
+    cpu = CpuTranscoder()
+    d = gw.discoverer
+    mi = d.tools["bench_echoservice_echo"]
+    for i in range(4):
+        args = wide_payload(rng, target_bytes=64 * 1024)
+        body = json.dumps({"jsonrpc": "2.0", "id": 100 + i,
+                           "method": "tools/call",
+                           "params": {"name": "bench_echoservice_echo",
+                                      "arguments": args}})
+        status, resp, _ = _post(port, body)
+        assert status == 200
+        assert resp["id"] == 100 + i
+        assert resp["result"]["isError"] is False, str(resp)[:300]
+        inner = json.loads(resp["result"]["content"][0]["text"])
+        wire = cpu.json_to_pb(mi.input_descriptor, json.dumps(args))
+        oracle = json.loads(cpu.pb_to_json(mi.output_descriptor, wire))
+        assert inner == oracle
