@@ -9,7 +9,7 @@ export GGRMCP_HOSTSIM_FLAGS="-fsanitize=address -fno-omit-frame-pointer -g"
 python -c "from ggrmcp_amd.ops import build; build.build_hostsim(force=True)"
 ASAN_LIB=$(gcc -print-file-name=libasan.so)
 LD_PRELOAD=$ASAN_LIB ASAN_OPTIONS=detect_leaks=0 \
-  python -m pytest tests/test_hostsim.py tests/test_hostsim_fuzz.py -q -p no:cacheprovider "$@"
+  python -m pytest tests/test_hostsim.py tests/test_hostsim_fuzz.py tests/test_double_exactness.py -q -p no:cacheprovider "$@"
 # restore the normal build
 unset GGRMCP_HOSTSIM_FLAGS
 python -c "from ggrmcp_amd.ops import build; build.build_hostsim(force=True)"
@@ -33,4 +33,5 @@ g++ ggrmcp_amd/ops/csrc/frontend.cpp -x c++ $CXX_ASAN -pthread \
   -o ggrmcp_amd/ops/_frontend.so
 LD_PRELOAD=$ASAN_LIB ASAN_OPTIONS=detect_leaks=0 \
   python -m pytest tests/test_native_transport.py tests/test_native_frontend.py \
+  tests/test_session_table.py tests/test_decode_fallback.py \
   -q -p no:cacheprovider "$@"
