@@ -58,6 +58,22 @@ using Clock = std::chrono::steady_clock;
 // helpers
 // ---------------------------------------------------------------------------
 
+// socket buffer size for the batch transport (both directions move whole
+// 16 MB+ chunk batches; bigger kernel buffers mean fewer syscall round
+// trips).  GGRMCP_SOCKBUF overrides; 0 keeps the kernel default.
+static int transport_sockbuf() {
+  const char* e = getenv("GGRMCP_SOCKBUF");
+  return e ? atoi(e) : (4 << 20);
+}
+
+static void apply_sockbuf(int fd) {
+  int sz = transport_sockbuf();
+  if (sz > 0) {
+    setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &sz, sizeof(sz));
+    setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &sz, sizeof(sz));
+  }
+}
+
 static int connect_target(const std::string& target) {
   int fd = -1;
   if (target.rfind("unix:", 0) == 0) {
@@ -68,6 +84,7 @@ static int connect_target(const std::string& target) {
     std::string path = target.substr(5);
     if (path.size() >= sizeof(addr.sun_path)) throw std::runtime_error("uds path too long");
     memcpy(addr.sun_path, path.c_str(), path.size() + 1);
+    apply_sockbuf(fd);
     if (connect(fd, (sockaddr*)&addr, sizeof(addr)) != 0) {
       close(fd);
       throw std::runtime_error("connect failed: " + target);
@@ -86,6 +103,7 @@ static int connect_target(const std::string& target) {
       close(fd);
       throw std::runtime_error("bad host (use a literal IP): " + host);
     }
+    apply_sockbuf(fd);
     if (connect(fd, (sockaddr*)&addr, sizeof(addr)) != 0) {
       close(fd);
       throw std::runtime_error("connect failed: " + target);
@@ -988,6 +1006,7 @@ class H2Server {
       fcntl(cfd, F_SETFL, flags | O_NONBLOCK);
       int one = 1;
       setsockopt(cfd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+      apply_sockbuf(cfd);
       workers.emplace_back([this, cfd] { conn_loop(cfd); });
     }
     for (auto& t : workers)
