@@ -197,6 +197,37 @@ DEV uint64_t bcast0_u64(uint64_t v) { return __shfl(v, 0, WAVE); }
 
 DEV bool is_ws(uint8_t c) { return c == ' ' || c == '\t' || c == '\n' || c == '\r'; }
 
+// ---------------------------------------------------------------------------
+// SWAR byte classification over per-lane dwords.  The 1-byte-per-lane
+// ballot scans were memory-latency bound (~26 us/KB per wave: one
+// dependent 64 B load per iteration); loading 4 bytes per lane amortizes
+// the window latency over 4x the bytes.  Returns 0x80 in each byte
+// position that matches.
+// ---------------------------------------------------------------------------
+
+DEV uint32_t swar_zero(uint32_t w) {
+  return (w - 0x01010101u) & ~w & 0x80808080u;
+}
+DEV uint32_t swar_eq(uint32_t w, uint8_t ch) {
+  return swar_zero(w ^ (0x01010101u * ch));
+}
+
+// guarded per-lane dword load of s[base..base+4) with `fill` past n
+DEV uint32_t load4_or(const uint8_t* s, uint32_t base, uint32_t n,
+                      uint8_t fill) {
+  if (base + 4 <= n) {
+    uint32_t w;
+    __builtin_memcpy(&w, s + base, 4);  // unaligned ok on CDNA
+    return w;
+  }
+  uint32_t w = 0;
+  for (int j = 0; j < 4; ++j) {
+    uint8_t b = base + (uint32_t)j < n ? s[base + j] : fill;
+    w |= (uint32_t)b << (8 * j);
+  }
+  return w;
+}
+
 // zigzag
 DEV uint64_t zigzag64(int64_t v) { return ((uint64_t)v << 1) ^ (uint64_t)(v >> 63); }
 DEV uint32_t zigzag32(int32_t v) { return ((uint32_t)v << 1) ^ (uint32_t)(v >> 31); }
@@ -428,13 +459,18 @@ DEV bool utf8_byte_ok(const uint8_t* s, uint32_t n, uint32_t i) {
   return true;
 }
 
-// whole-span check; ASCII windows cost one ballot (the common case)
+// whole-span check; ASCII super-windows (256 B) cost one dword load + one
+// ballot per lane — the 1-byte/lane form was memory-latency bound
 DEV bool utf8_span_valid(const uint8_t* s, uint32_t n, uint32_t lane) {
-  for (uint32_t base = 0; base < n; base += WAVE) {
-    uint32_t i = base + lane;
-    uint8_t b = i < n ? s[i] : 0;
-    if (!__ballot(b >= 0x80)) continue;
-    bool ok = i < n ? utf8_byte_ok(s, n, i) : true;
+  for (uint32_t base = 0; base < n; base += 4u * WAVE) {
+    uint32_t off = base + 4u * lane;
+    uint32_t w = load4_or(s, off, n, 0);
+    if (!__ballot((w & 0x80808080u) != 0)) continue;
+    bool ok = true;
+    for (uint32_t j = 0; j < 4; ++j) {
+      uint32_t i = off + j;
+      if (i < n && !utf8_byte_ok(s, n, i)) ok = false;
+    }
     if (__ballot(!ok)) return false;
   }
   return true;

@@ -85,15 +85,20 @@ DEV bool expect(Ctx& c, uint8_t ch) {
 }
 
 // c.pos just after the opening quote; finds the closing quote.
+// 4 bytes/lane SWAR windows: the byte stream is latency-bound, so one
+// dword load per lane quadruples the bytes covered per round trip.
 DEV bool string_end(Ctx& c, uint32_t* end, bool* has_esc) {
   uint32_t p = c.pos;
   bool esc = false;
   while (p < c.len) {
-    uint32_t i = p + c.lane;
-    uint8_t ch = i < c.len ? c.s[i] : '"';
-    uint64_t m = __ballot(ch == '"' || ch == '\\');
+    uint32_t base = p + 4u * (uint32_t)c.lane;
+    uint32_t w = load4_or(c.s, base, c.len, '"');
+    uint32_t hit = swar_eq(w, '"') | swar_eq(w, '\\');
+    uint64_t m = __ballot(hit != 0);
     if (m) {
-      uint32_t k = p + __ffsll((long long)m) - 1;
+      int lf = __ffsll((long long)m) - 1;
+      uint32_t lh = (uint32_t)__shfl(hit, lf, WAVE);
+      uint32_t k = p + 4u * (uint32_t)lf + (uint32_t)(__builtin_ctz(lh) >> 3);
       if (k >= c.len) break;
       if (c.s[k] == '"') {
         *end = k;
@@ -103,7 +108,7 @@ DEV bool string_end(Ctx& c, uint32_t* end, bool* has_esc) {
       esc = true;
       p = k + 2;  // skip backslash + escaped char
     } else {
-      p += WAVE;
+      p += 4u * WAVE;
     }
   }
   c.pos = c.len;
@@ -231,7 +236,14 @@ DEV uint32_t unescape_serial(Ctx& c, const uint8_t* src, uint32_t n, uint8_t* ds
 
 // lane-parallel copy (no escapes)
 DEV void wave_copy(Ctx& c, uint8_t* dst, const uint8_t* src, uint32_t n) {
-  for (uint32_t i = c.lane; i < n; i += WAVE) dst[i] = src[i];
+  // dword-per-lane main body (unaligned ok on CDNA); byte tail
+  for (uint32_t base = 4u * (uint32_t)c.lane; base + 4 <= n;
+       base += 4u * WAVE) {
+    uint32_t w;
+    __builtin_memcpy(&w, src + base, 4);
+    __builtin_memcpy(dst + base, &w, 4);
+  }
+  for (uint32_t i = (n & ~3u) + c.lane; i < n; i += WAVE) dst[i] = src[i];
 }
 
 // lane-parallel byte compare; true if equal
@@ -570,17 +582,20 @@ DEV uint32_t b64_decode(Ctx& c, const uint8_t* src, uint32_t n, uint8_t* dst,
 DEV bool next_structural(Ctx& c, uint32_t* at) {
   uint32_t p = c.pos;
   while (p < c.len) {
-    uint32_t i = p + c.lane;
-    uint8_t ch = i < c.len ? c.s[i] : '}';
-    uint64_t m =
-        __ballot(ch == '{' || ch == '}' || ch == '[' || ch == ']' || ch == '"');
+    uint32_t base = p + 4u * (uint32_t)c.lane;
+    uint32_t w = load4_or(c.s, base, c.len, 0);
+    uint32_t hit = swar_eq(w, '{') | swar_eq(w, '}') | swar_eq(w, '[') |
+                   swar_eq(w, ']') | swar_eq(w, '"');
+    uint64_t m = __ballot(hit != 0);
     if (m) {
-      uint32_t k = p + __ffsll((long long)m) - 1;
+      int lf = __ffsll((long long)m) - 1;
+      uint32_t lh = (uint32_t)__shfl(hit, lf, WAVE);
+      uint32_t k = p + 4u * (uint32_t)lf + (uint32_t)(__builtin_ctz(lh) >> 3);
       if (k >= c.len) break;
       *at = k;
       return true;
     }
-    p += WAVE;
+    p += 4u * WAVE;
   }
   return fail(c, E_PARSE);
 }
