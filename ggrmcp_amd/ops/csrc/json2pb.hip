@@ -600,6 +600,69 @@ DEV bool next_structural(Ctx& c, uint32_t* at) {
   return fail(c, E_PARSE);
 }
 
+// Rolling-window structural scan of a container body: c.pos at the byte
+// AFTER the opener; advances to just past the matching closer.  One
+// 256 B dword-per-lane load per round trip with in-register serial hit
+// processing — the hop-per-token form (next_structural + string_end per
+// string) paid one dependent memory round trip per token and dominated
+// large-payload scans.  When `marks` is non-null, records the first
+// depth-1 comma at or past each `stride`-byte mark (entry-boundary
+// chunking for the wg encode scanner).
+DEV bool scan_container(Ctx& c, uint32_t* marks, int* n_marks, int cap,
+                        uint32_t stride) {
+  int depth = 1;
+  uint32_t p = c.pos;
+  bool in_str = false;
+  uint32_t skip_pos = 0xFFFFFFFFu;  // byte escaped by a preceding backslash
+  uint32_t next_mark = stride ? c.pos + stride : 0xFFFFFFFFu;
+  while (p < c.len) {
+    uint32_t off = p + 4u * (uint32_t)c.lane;
+    uint32_t w = load4_or(c.s, off, c.len, 0);
+    uint32_t hit = swar_eq(w, '"') | swar_eq(w, '\\') | swar_eq(w, '{') |
+                   swar_eq(w, '}') | swar_eq(w, '[') | swar_eq(w, ']') |
+                   swar_eq(w, ',');
+    uint64_t lm = __ballot(hit != 0);
+    while (lm) {
+      int lf = __ffsll((long long)lm) - 1;
+      lm &= lm - 1;
+      uint32_t lh = (uint32_t)__shfl(hit, lf, WAVE);
+      uint32_t lw = (uint32_t)__shfl(w, lf, WAVE);
+      while (lh) {
+        uint32_t bidx = (uint32_t)(__builtin_ctz(lh) >> 3);
+        lh &= lh - 1;
+        uint32_t pos = p + 4u * (uint32_t)lf + bidx;
+        if (pos >= c.len) {
+          lh = 0;
+          break;
+        }
+        if (pos == skip_pos) continue;
+        uint8_t ch = (uint8_t)(lw >> (8 * bidx));
+        if (in_str) {
+          if (ch == '\\')
+            skip_pos = pos + 1;
+          else if (ch == '"')
+            in_str = false;
+        } else if (ch == '"') {
+          in_str = true;
+        } else if (ch == '{' || ch == '[') {
+          ++depth;
+        } else if (ch == '}' || ch == ']') {
+          if (--depth == 0) {
+            c.pos = pos + 1;
+            return true;
+          }
+        } else if (ch == ',' && depth == 1 && pos >= next_mark &&
+                   marks && *n_marks < cap) {
+          marks[(*n_marks)++] = pos;
+          next_mark = pos + stride;
+        }
+      }
+    }
+    p += 4u * WAVE;
+  }
+  return fail(c, E_PARSE);
+}
+
 DEV bool skip_value(Ctx& c) {
   skip_ws(c);
   uint8_t ch = peek(c);
@@ -609,24 +672,8 @@ DEV bool skip_value(Ctx& c) {
     return string_span(c, &st, &rl, &esc);
   }
   if (ch == '{' || ch == '[') {
-    int d = 0;
-    while (true) {
-      uint32_t at;
-      if (!next_structural(c, &at)) return false;
-      uint8_t sc = c.s[at];
-      c.pos = at + 1;
-      if (sc == '"') {
-        uint32_t end;
-        bool esc;
-        if (!string_end(c, &end, &esc)) return false;
-        c.pos = end + 1;
-      } else if (sc == '{' || sc == '[') {
-        ++d;
-      } else {
-        if (--d == 0) return true;
-        if (d < 0) return fail(c, E_PARSE);
-      }
-    }
+    c.pos++;
+    return scan_container(c, nullptr, nullptr, 0, 0);
   }
   // literal / number
   uint32_t e = token_end(c);
@@ -2264,8 +2311,11 @@ extern "C" __global__ void __launch_bounds__(WG_ENC_WAVES * WAVE) k_json2pb_wg(
                              peek(c) == '[';
             bool big = (c.len - c.pos) >= 2048;
             if ((chunk_map || chunk_arr) && big) {
-              // per-entry chunks: spans cover entries only (no key, no
-              // braces); chunks emit separate-but-concatenable runs
+              // per-entry chunks: ONE bulk structural scan finds the
+              // ~4 KB entry boundaries and the closer (the per-entry
+              // string_span/skip_value loop paid a dependent memory
+              // round trip per token); spans cover entries only — chunks
+              // emit separate-but-concatenable runs
               c.pos++;  // consume '{' / '['
               skip_ws(c);
               uint8_t closer = chunk_map ? '}' : ']';
@@ -2273,58 +2323,38 @@ extern "C" __global__ void __launch_bounds__(WG_ENC_WAVES * WAVE) k_json2pb_wg(
                 c.pos++;  // empty container member: no wire output at all
               } else {
                 uint32_t chunk_start = c.pos;
-                bool closed = false;
-                while (!fallback && !closed) {
-                  // one entry
-                  if (chunk_map) {
-                    uint32_t ek, el;
-                    bool ee;
-                    if (!string_span(c, &ek, &el, &ee) || !expect(c, ':')) {
-                      fallback = true;
-                      break;
-                    }
-                    skip_ws(c);
-                  }
-                  if (!skip_value(c)) {
+                uint32_t marks[64];
+                int nm = 0;
+                if (!scan_container(c, marks, &nm, 64, 4096)) {
+                  fallback = true;
+                  break;
+                }
+                uint32_t close_pos = c.pos - 1;
+                if (c.s[close_pos] != closer) {  // '}' vs ']' mismatch
+                  fallback = true;
+                  break;
+                }
+                for (int mi2 = 0; mi2 <= nm && !fallback; ++mi2) {
+                  uint32_t cs = mi2 == 0 ? chunk_start : marks[mi2 - 1] + 1;
+                  uint32_t ce = mi2 == nm ? close_pos : marks[mi2];
+                  if (cs >= ce) continue;
+                  if (n >= WG_ENC_MAX_ITEMS) {
                     fallback = true;
                     break;
                   }
-                  uint32_t entry_end = c.pos;
-                  skip_ws(c);
-                  uint8_t ch = peek(c);
-                  bool boundary = false;
-                  if (ch == ',') {
-                    c.pos++;
-                    skip_ws(c);
-                    if (entry_end - chunk_start >= 4096) boundary = true;
-                  } else if (ch == closer) {
-                    c.pos++;
-                    closed = true;
-                    boundary = true;
-                  } else {
+                  uint32_t span = ce - cs;
+                  uint32_t icap = span + span / 4 + WG_ENC_ITEM_PAD;
+                  if (acc + icap > scr_cap) {
                     fallback = true;
                     break;
                   }
-                  if (boundary) {
-                    if (n >= WG_ENC_MAX_ITEMS) {
-                      fallback = true;
-                      break;
-                    }
-                    uint32_t span = entry_end - chunk_start;
-                    uint32_t icap = span + span / 4 + WG_ENC_ITEM_PAD;
-                    if (acc + icap > scr_cap) {
-                      fallback = true;
-                      break;
-                    }
-                    s_start[n] = chunk_start;
-                    s_end[n] = entry_end;
-                    s_ioff[n] = acc;
-                    s_fidx[n] = m.field_start + fidx;
-                    s_kind[n] = chunk_map ? 2 : 3;
-                    ++n;
-                    acc += icap;
-                    chunk_start = c.pos;
-                  }
+                  s_start[n] = cs;
+                  s_end[n] = ce;
+                  s_ioff[n] = acc;
+                  s_fidx[n] = m.field_start + fidx;
+                  s_kind[n] = chunk_map ? 2 : 3;
+                  ++n;
+                  acc += icap;
                 }
                 if (fallback) break;
               }
