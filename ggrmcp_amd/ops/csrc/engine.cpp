@@ -85,9 +85,13 @@ inline uint32_t wg_enc_min() {
   return e ? (uint32_t)strtoul(e, nullptr, 10) : WG_ENC_MIN_BYTES;
 }
 
-// debug bisection: limit the wg kernel to its first N phases
+// debug bisection: limit the wg kernels to their first N phases
 inline int wg_phases() {
   const char* e = getenv("GGRMCP_WG_PHASES");
+  return e ? atoi(e) : 3;
+}
+inline int wg_enc_phases() {
+  const char* e = getenv("GGRMCP_WG_ENC_PHASES");
   return e ? atoi(e) : 3;
 }
 
@@ -990,7 +994,7 @@ class Engine : public spanapi::ISpanExecutor {
                          (const uint32_t*)d_off + (n + 1),
                          (SlotResult*)d_results_.p, (uint8_t*)d_id_slots_.p,
                          (uint8_t*)d_scratch_.p, tables_, lim, n,
-                         (const int32_t*)d_aux2_.p);
+                         (const int32_t*)d_aux2_.p, wg_enc_phases());
       HIP_CHECK(hipGetLastError());
     }
     HIP_CHECK(hipMemcpyAsync(h_results_.p, d_results_.p,

@@ -2180,7 +2180,8 @@ extern "C" __global__ void __launch_bounds__(WG_ENC_WAVES * WAVE) k_json2pb_wg(
     uint8_t* __restrict__ pb_arena, const uint32_t* __restrict__ pb_off,
     SlotResult* __restrict__ results, uint8_t* __restrict__ id_slots,
     uint8_t* __restrict__ enc_scratch, Tables t, Limits lim, int n_req,
-    const int32_t* __restrict__ skip) {
+    const int32_t* __restrict__ skip, int max_phase) {
+  // max_phase: debug bisection (GGRMCP_WG_ENC_PHASES; 3 = full)
   int req = blockIdx.x;
   if (req >= n_req) return;
   if (!skip || skip[req] != 2) return;
@@ -2415,7 +2416,7 @@ extern "C" __global__ void __launch_bounds__(WG_ENC_WAVES * WAVE) k_json2pb_wg(
   const int msg_idx = s_msg;
 
   // ---- phase B: encode items (dynamic wave grabs, bounded) ----------------
-  if (s_mode == 1) {
+  if (s_mode == 1 && max_phase >= 1) {
     for (int guard = 0; guard <= WG_ENC_MAX_ITEMS + 1; ++guard) {
       int idx = 0;
       if (!lane) idx = atomicAdd(&s_next, 1);
@@ -2448,7 +2449,7 @@ extern "C" __global__ void __launch_bounds__(WG_ENC_WAVES * WAVE) k_json2pb_wg(
   __syncthreads();
 
   // ---- phase C: offsets (one thread) --------------------------------------
-  if (s_mode == 1 && threadIdx.x == 0) {
+  if (s_mode == 1 && threadIdx.x == 0 && max_phase >= 2) {
     uint32_t off = 0;
     for (int i = 0; i < n_items; ++i) {
       s_foff[i] = off;
@@ -2464,7 +2465,7 @@ extern "C" __global__ void __launch_bounds__(WG_ENC_WAVES * WAVE) k_json2pb_wg(
   __syncthreads();
 
   // ---- phase D: compact items into the pb arena / classic fallback --------
-  if (s_mode == 1) {
+  if (s_mode == 1 && max_phase >= 2) {
     for (int s = wave; s < n_items; s += WG_ENC_WAVES) {
       const uint8_t* p = scr + s_ioff[s];
       uint8_t* d = pbout + s_foff[s];
@@ -2480,6 +2481,9 @@ extern "C" __global__ void __launch_bounds__(WG_ENC_WAVES * WAVE) k_json2pb_wg(
   __syncthreads();
 
   // ---- finalize ------------------------------------------------------------
-  if (s_mode != 0 && threadIdx.x == 0) results[req] = s_res;
+  if (s_mode != 0 && threadIdx.x == 0) {
+    if (s_mode == 1 && max_phase < 2) s_res.pb_len = 0;  // debug phases
+    results[req] = s_res;
+  }
 }
 #endif  // GGRMCP_HOST_SIM

@@ -205,3 +205,22 @@ def diag_batch6():
 
 if kind == "batch6":
     diag_batch6()
+
+def enc_time():
+    """kind=enctime: time encode_batch alone on big wide payloads."""
+    import time
+    from ggrmcp_amd.utils.synthetic import wide_payload, jsonrpc_body
+
+    rng = random.Random(5)
+    bodies = [jsonrpc_body("bench_echoservice_echo",
+                           wide_payload(rng, target_bytes=64 * 1024), i)
+              for i in range(64)]
+    eng.encode_batch(bodies, mode=0)  # warm
+    t0 = time.perf_counter()
+    for _ in range(20):
+        eng.encode_batch(bodies, mode=0)
+    dt = (time.perf_counter() - t0) / 20
+    print(f"encode_batch(64x64KB): {dt*1e3:.2f} ms/iter", flush=True)
+
+if kind == "enctime":
+    enc_time()
