@@ -214,9 +214,13 @@ def main() -> None:
         if args.config == "hello1k" and use_gpu:
             # tuned serving defaults (profiles/serving_sweep_r02.log):
             # reactors 4->8 and span engines 2->6 lifted 1-GPU serving
-            # 242k -> 363k req/s; explicit flags still override
+            # 242k -> 363k req/s; explicit flags still override.  Multi-rank
+            # weak scaling shares the node's cores across ranks, so shed
+            # client threads there (the load generator competes hardest).
             cfg.gpu.streams = 6
-            cfg.server.reactors = 8
+            cfg.server.reactors = 8 if world == 1 else 6
+            if world > 1 and args.client_threads == 16:
+                args.client_threads = 8
         if args.streams > 0:
             cfg.gpu.streams = args.streams
         if args.reactors > 0:
