@@ -57,7 +57,12 @@ def test_gpu_format_shortest_roundtrip(env):
     rng = random.Random(31337)
     vals = [_gen_double(rng) for _ in range(8000)]
     wires = [cpu.json_to_pb(desc, json.dumps({"weight": v})) for v in vals]
-    dec, outs = eng.decode_batch(wires, [idx] * len(wires), mode=1)
+    dec, outs = [], []
+    for b in range(0, len(wires), 2048):  # engine max_batch chunks
+        d2, o2 = eng.decode_batch(wires[b:b + 2048],
+                                  [idx] * len(wires[b:b + 2048]), mode=1)
+        dec.extend(d2)
+        outs.extend(o2)
 
     def ndig(s):
         m = s.split("e")[0].split("E")[0].replace(".", "").lstrip("-0")
@@ -88,9 +93,13 @@ def test_gpu_parse_correctly_rounded(env):
 
     texts = [gen_text() for _ in range(8000)]
     payloads = [f'{{"weight": {t}}}'.encode() for t in texts]
-    enc, pbs = eng.encode_batch(payloads, mode=1,
-                                msg_indices=[idx] * len(payloads),
-                                enforce=False)
+    enc, pbs = [], []
+    for b in range(0, len(payloads), 2048):
+        part = payloads[b:b + 2048]
+        e2, p2 = eng.encode_batch(part, mode=1, msg_indices=[idx] * len(part),
+                                  enforce=False)
+        enc.extend(e2)
+        pbs.extend(p2)
     for t, r, w in zip(texts, enc, pbs):
         expect = float(t)
         if not math.isfinite(expect) or r["status"] != 0:
