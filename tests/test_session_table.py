@@ -172,3 +172,60 @@ def test_two_process_shared_state():
     finally:
         if os.path.exists(path):
             os.unlink(path)
+
+
+def test_concurrent_guard_storm():
+    """8 threads hammer one table with mixed traffic (shared session,
+    per-thread sessions, create floods, block flips): the CAS-based
+    open addressing must stay consistent — no lost sessions, call counts
+    within the expected envelope, stats scan clean."""
+    import threading
+
+    mod = load_module()
+    t = mod.SessionTable(capacity=8192, ttl_s=1800.0, path="",
+                         rate_per_min=10**9, rate_burst=0)
+    shared, _, _ = t.guard("", rate_limit=False)
+    N_THREADS, PER = 8, 4000
+    errs = []
+
+    def worker(tid):
+        try:
+            mine, _, created = t.guard(f"worker-{tid}", rate_limit=False)
+            assert created and mine == f"worker-{tid}"
+            for i in range(PER):
+                k = i % 4
+                if k == 0:
+                    _, v, _ = t.guard(shared, rate_limit=False)
+                    assert v in (0, 1)  # may race a block flip
+                elif k == 1:
+                    _, v, _ = t.guard(mine, rate_limit=False)
+                    assert v == 0
+                elif k == 2:
+                    sid, v, _ = t.guard(f"ephem-{tid}-{i}", rate_limit=False)
+                    assert v == 0 and sid == f"ephem-{tid}-{i}"
+                else:
+                    if i % 64 == 3:
+                        t.block(shared)
+                        t.unblock(shared)
+                    else:
+                        _, v, _ = t.guard(shared, rate_limit=False)
+                        assert v in (0, 1)
+        except Exception as e:  # pragma: no cover
+            errs.append(e)
+
+    ts = [threading.Thread(target=worker, args=(i,)) for i in range(N_THREADS)]
+    for th in ts:
+        th.start()
+    for th in ts:
+        th.join()
+    assert not errs, errs[:3]
+    # per-thread sessions survived the storm with exact counts
+    for tid in range(N_THREADS):
+        info = t.info(f"worker-{tid}")
+        assert info is not None
+        assert info["callCount"] == 1 + PER // 4
+    t.unblock(shared)
+    assert t.guard(shared, rate_limit=False)[1] == 0
+    s = t.stats()
+    assert s["activeSessions"] <= t.capacity
+    assert s["createdTotal"] >= N_THREADS + 1
