@@ -329,23 +329,6 @@ __constant__ double POW10[23] = {1e0,  1e1,  1e2,  1e3,  1e4,  1e5,  1e6,  1e7,
                                  1e8,  1e9,  1e10, 1e11, 1e12, 1e13, 1e14, 1e15,
                                  1e16, 1e17, 1e18, 1e19, 1e20, 1e21, 1e22};
 
-// scale d by 10^e progressively: dividing step-by-step keeps subnormal
-// results reachable (computing 10^324 first overflows to inf and turned
-// every |exp|>308 input into 0.0 — found by the hostsim fuzzer).
-DEV double scale_by_pow10(double d, int e) {
-  while (e > 22) {
-    d *= 1e22;
-    e -= 22;
-    if (d > 1.7e308) return d;  // about to overflow; caller range-checks
-  }
-  while (e < -22) {
-    d /= 1e22;
-    e += 22;
-    if (d == 0.0) return d;
-  }
-  return e >= 0 ? d * POW10[e] : d / POW10[-e];
-}
-
 // parse decimal text [p, e) -> NumVal.  Returns false if malformed.
 DEV bool parse_number_text(const uint8_t* s, uint32_t p, uint32_t e, NumVal* nv) {
   if (p >= e) return false;

@@ -237,14 +237,6 @@ __constant__ double DPOW10[23] = {1e0,  1e1,  1e2,  1e3,  1e4,  1e5,  1e6,  1e7,
                                   1e8,  1e9,  1e10, 1e11, 1e12, 1e13, 1e14, 1e15,
                                   1e16, 1e17, 1e18, 1e19, 1e20, 1e21, 1e22};
 
-// scale d by 10^e (composed; exact when |e|<=22 and mantissa fits)
-DEV double scale10(double d, int e) {
-  while (e > 22) { d *= 1e22; e -= 22; }
-  while (e < -22) { d /= 1e22; e += 22; }
-  if (e >= 0) return d * DPOW10[e];
-  return d / DPOW10[-e];
-}
-
 // d * 10^e as a double-double (steps through exact powers)
 DEV DD dd_scale10(double d, int e) {
   DD x{d, 0.0};
