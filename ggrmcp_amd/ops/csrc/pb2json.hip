@@ -278,8 +278,12 @@ DEV bool digits_roundtrip(uint64_t digits, int ndig, int e10, double d) {
   double diff = (x.hi - d) + x.lo;
   bool above = diff >= 0;
   if (diff < 0) diff = -diff;
-  double gap_half = above ? (nextafter(d, 1e308) - d) * 0.5
-                          : (d - nextafter(d, -1e308)) * 0.5;
+  // direction must be +-infinity: a finite direction constant below
+  // DBL_MAX makes nextafter step DOWNWARD for d above it (negative gap,
+  // false rejection near the top of the range — hypothesis-found with
+  // the float variant near FLT_MAX)
+  double gap_half = above ? (nextafter(d, HUGE_VAL) - d) * 0.5
+                          : (d - nextafter(d, -HUGE_VAL)) * 0.5;
   if (diff < gap_half) return true;
   return diff == gap_half && (__builtin_bit_cast(uint64_t, d) & 1) == 0;
 }
@@ -305,8 +309,8 @@ DEV bool digits_roundtrip_f(uint64_t digits, int ndig, int e10, float f) {
   bool above = diff >= 0;
   if (diff < 0) diff = -diff;
   double gap_half = above
-      ? ((double)nextafterf(f, 3.4e38f) - (double)f) * 0.5
-      : ((double)f - (double)nextafterf(f, -3.4e38f)) * 0.5;
+      ? ((double)nextafterf(f, HUGE_VALF) - (double)f) * 0.5
+      : ((double)f - (double)nextafterf(f, -HUGE_VALF)) * 0.5;
   if (diff < gap_half) return true;
   return diff == gap_half && (__builtin_bit_cast(uint32_t, f) & 1) == 0;
 }
