@@ -123,6 +123,21 @@ def test_fuzz_wide64_encode_decode(payload):
     _roundtrip("bench.Wide64", payload)
 
 
+def test_wide64_float_boundary_regressions():
+    """Pinned fuzz finds: top-of-range values where nextafter toward a
+    finite constant (3.4e38f < FLT_MAX) stepped DOWNWARD, inverting the
+    round-trip half-gap and rejecting the shortest digit form."""
+    import struct as _struct
+
+    fmax = _struct.unpack("<f", _struct.pack("<I", 0x7F7FFFFF))[0]  # FLT_MAX
+    fsub = _struct.unpack("<f", _struct.pack("<I", 0x7F7FFFFE))[0]  # pred
+    for v in (3.4028224522648084e+38, fmax, -fmax, fsub, -fsub,
+              1.7976931348623157e+308, -1.7976931348623157e+308,
+              1.0000000000000002e+308):
+        _roundtrip("bench.Wide64", {"f08Float": min(max(v, -fmax), fmax)})
+        _roundtrip("bench.Wide64", {"f04Double": v})
+
+
 @settings(max_examples=60, deadline=None)
 @given(node=node_strategy())
 def test_fuzz_node_recursive(node):
