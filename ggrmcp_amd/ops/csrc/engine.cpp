@@ -47,6 +47,8 @@ namespace py = pybind11;
 
 namespace {
 
+constexpr int PIPE_MAX_CHUNKS = 8;
+
 struct DeviceBuf {
   void* p = nullptr;
   size_t n = 0;
@@ -227,6 +229,15 @@ class Engine : public spanapi::ISpanExecutor {
     // see profiles/blocking_sync.log.)
     HIP_CHECK(hipEventCreateWithFlags(
         &sync_ev_, hipEventBlockingSync | hipEventDisableTiming));
+    // copy stream for the chunked large-batch pipeline (H2D of chunk c+1
+    // and D2H of chunk c-1 overlap chunk c's kernels on stream_)
+    HIP_CHECK(hipStreamCreateWithFlags(&stream2_, hipStreamNonBlocking));
+    HIP_CHECK(hipEventCreateWithFlags(
+        &sync_ev2_, hipEventBlockingSync | hipEventDisableTiming));
+    for (int i = 0; i < PIPE_MAX_CHUNKS; ++i) {
+      HIP_CHECK(hipEventCreateWithFlags(&ev_h2d_[i], hipEventDisableTiming));
+      HIP_CHECK(hipEventCreateWithFlags(&ev_krn_[i], hipEventDisableTiming));
+    }
 
     upload_blob(msg_table, d_msgs_);
     upload_blob(field_table, d_fields_);
@@ -275,14 +286,60 @@ class Engine : public spanapi::ISpanExecutor {
 
   ~Engine() {
     (void)hipStreamSynchronize(stream_);
+    (void)hipStreamSynchronize(stream2_);
     (void)hipEventDestroy(sync_ev_);
+    (void)hipEventDestroy(sync_ev2_);
+    for (int i = 0; i < PIPE_MAX_CHUNKS; ++i) {
+      (void)hipEventDestroy(ev_h2d_[i]);
+      (void)hipEventDestroy(ev_krn_[i]);
+    }
     (void)hipStreamDestroy(stream_);
+    (void)hipStreamDestroy(stream2_);
   }
 
   // park-the-thread stream wait (see sync_ev_ above)
   void sync_stream() {
     HIP_CHECK(hipEventRecord(sync_ev_, stream_));
     HIP_CHECK(hipEventSynchronize(sync_ev_));
+  }
+
+  void sync_stream2() {
+    HIP_CHECK(hipEventRecord(sync_ev2_, stream2_));
+    HIP_CHECK(hipEventSynchronize(sync_ev2_));
+  }
+
+  // ---- chunked copy/compute pipeline (large batches only) -----------------
+  // Wide-payload batches (wide64: 64 x 64 KB) spend ~1.1 ms/batch in
+  // serial H2D + D2H around the kernels.  Splitting the item range into
+  // byte-balanced chunks lets chunk c+1's H2D and chunk c-1's D2H run on
+  // stream2_ while chunk c's kernels run on stream_.  All per-item kernel
+  // arrays index absolutely through the offset tables, so a sub-range
+  // launch is just sliced pointers.  Gated OFF for small batches: extra
+  // HIP submissions are the known multi-instance contention point
+  // (profiles/streams_sweep.log), so serving-sized batches keep the
+  // single-shot path.
+  static int pipe_chunks() {
+    const char* e = getenv("GGRMCP_PIPE_CHUNKS");  // per-call, like wg_*_min
+    int x = e ? atoi(e) : 4;
+    return x < 1 ? 1 : (x > PIPE_MAX_CHUNKS ? PIPE_MAX_CHUNKS : x);
+  }
+  static size_t pipe_min_bytes() {
+    const char* e = getenv("GGRMCP_PIPE_MIN");
+    return e ? (size_t)atoll(e) : (size_t)262144;
+  }
+  // cut [0,n) into <=C ranges balanced by in_off bytes; returns #chunks
+  static int pipe_cuts(const uint32_t* off, int n, int C, int* cut) {
+    cut[0] = 0;
+    int k = 0, lo = 0;
+    uint64_t base = off[0], total = off[n] - off[0];
+    for (int c = 0; c < C && lo < n; ++c) {
+      int hi = (c == C - 1) ? n : lo + 1;
+      uint64_t target = base + total * (uint64_t)(c + 1) / (uint64_t)C;
+      while (hi < n && off[hi] < target) ++hi;
+      cut[++k] = hi;
+      lo = hi;
+    }
+    return k;
   }
 
   // ---- encode: JSON(-RPC) -> protobuf ------------------------------------
@@ -967,6 +1024,21 @@ class Engine : public spanapi::ISpanExecutor {
     }
     auto g0 = std::chrono::steady_clock::now();
     HIP_CHECK(hipSetDevice(device_));
+    // chunked pipeline for big batches (mode 0, no msg_idx): overlap the
+    // H2D / D2H slices with the kernels instead of serializing them
+    if (mode == 0 && !has_idx && n >= 2 && pipe_chunks() > 1 &&
+        in_bytes + pb_bytes >= pipe_min_bytes()) {
+      int cut[PIPE_MAX_CHUNKS + 1];
+      int C = pipe_cuts(h_off, n, pipe_chunks(), cut);
+      if (C > 1) {
+        run_encode_chunked(n, C, cut, n_wg, lim);
+        last_enc_gpu_ms_ = std::chrono::duration<double, std::milli>(
+                               std::chrono::steady_clock::now() - g0)
+                               .count();
+        last_batch_n_ = n;
+        return;
+      }
+    }
     HIP_CHECK(hipMemcpyAsync(d_in_.p, h_in_.p, in_bytes,
                              hipMemcpyHostToDevice, stream_));
     uint32_t* d_off = (uint32_t*)d_off3_.p;  // in_off | pb_off contiguous
@@ -1016,6 +1088,72 @@ class Engine : public spanapi::ISpanExecutor {
     last_batch_n_ = n;
   }
 
+  // Chunked encode (mode 0, no msg_idx — the serving/wide shapes): chunk
+  // c+1's input H2D and chunk c-1's pb/results D2H run on stream2_ while
+  // chunk c's kernels run on stream_.  All kernels stay on ONE stream
+  // (serialized — no extra kernel-queue contention); only copies overlap.
+  // Per-item addressing is absolute through the offset tables, so a
+  // sub-range launch is shifted pointers + a smaller n.
+  void run_encode_chunked(int n, int C, const int* cut, int n_wg, Limits lim) {
+    uint32_t* h_off = (uint32_t*)h_off_.p;
+    const uint32_t* h_pb_off = h_off + (n + 1);
+    uint32_t* d_off = (uint32_t*)d_off3_.p;  // in_off | pb_off contiguous
+    // offsets (+ wg routing flags): every chunk's kernels read them
+    HIP_CHECK(hipMemcpyAsync(d_off, h_off, 2 * (n + 1) * sizeof(uint32_t),
+                             hipMemcpyHostToDevice, stream_));
+    if (n_wg > 0)
+      HIP_CHECK(hipMemcpyAsync(d_aux2_.p, h_aux_.p, n * sizeof(int32_t),
+                               hipMemcpyHostToDevice, stream_));
+    for (int c = 0; c < C; ++c) {
+      const int lo = cut[c], hi = cut[c + 1], m = hi - lo;
+      const size_t a = h_off[lo], b = h_off[hi];
+      if (b > a)
+        HIP_CHECK(hipMemcpyAsync((uint8_t*)d_in_.p + a, (uint8_t*)h_in_.p + a,
+                                 b - a, hipMemcpyHostToDevice, stream2_));
+      HIP_CHECK(hipEventRecord(ev_h2d_[c], stream2_));
+      HIP_CHECK(hipStreamWaitEvent(stream_, ev_h2d_[c], 0));
+      int blocks = (int)cdiv(m, WPB);
+      if (blocks > 0) {
+        hipLaunchKernelGGL(
+            k_json2pb, dim3(blocks), dim3(WPB * WAVE), 0, stream_,
+            (const uint8_t*)d_in_.p, (const uint32_t*)d_off + lo,
+            (uint8_t*)d_pb_.p, (const uint32_t*)d_off + (n + 1) + lo,
+            (SlotResult*)d_results_.p + lo,
+            (uint8_t*)d_id_slots_.p + (size_t)lo * ID_SLOT_BYTES, nullptr,
+            tables_, lim, m, 0,
+            n_wg > 0 ? (const int32_t*)d_aux2_.p + lo : nullptr);
+        HIP_CHECK(hipGetLastError());
+      }
+      if (n_wg > 0) {
+        hipLaunchKernelGGL(
+            k_json2pb_wg, dim3(m), dim3(WG_ENC_WAVES * WAVE), 0, stream_,
+            (const uint8_t*)d_in_.p, (const uint32_t*)d_off + lo,
+            (uint8_t*)d_pb_.p, (const uint32_t*)d_off + (n + 1) + lo,
+            (SlotResult*)d_results_.p + lo,
+            (uint8_t*)d_id_slots_.p + (size_t)lo * ID_SLOT_BYTES,
+            (uint8_t*)d_scratch_.p, tables_, lim, m,
+            (const int32_t*)d_aux2_.p + lo, wg_enc_phases());
+        HIP_CHECK(hipGetLastError());
+      }
+      HIP_CHECK(hipEventRecord(ev_krn_[c], stream_));
+      HIP_CHECK(hipStreamWaitEvent(stream2_, ev_krn_[c], 0));
+      const size_t pa = h_pb_off[lo], pe = h_pb_off[hi];
+      if (pe > pa)
+        HIP_CHECK(hipMemcpyAsync((uint8_t*)h_pb_.p + pa, (uint8_t*)d_pb_.p + pa,
+                                 pe - pa, hipMemcpyDeviceToHost, stream2_));
+      HIP_CHECK(hipMemcpyAsync((SlotResult*)h_results_.p + lo,
+                               (SlotResult*)d_results_.p + lo,
+                               (size_t)m * sizeof(SlotResult),
+                               hipMemcpyDeviceToHost, stream2_));
+      HIP_CHECK(hipMemcpyAsync(
+          (uint8_t*)h_id_.p + (size_t)lo * ID_SLOT_BYTES,
+          (uint8_t*)d_id_slots_.p + (size_t)lo * ID_SLOT_BYTES,
+          (size_t)m * ID_SLOT_BYTES, hipMemcpyDeviceToHost, stream2_));
+    }
+    sync_stream();
+    sync_stream2();
+  }
+
   py::tuple run_encode(int n, size_t in_bytes, size_t pb_bytes, bool has_idx,
                        Limits lim, int mode) {
     {
@@ -1050,18 +1188,73 @@ class Engine : public spanapi::ISpanExecutor {
     }
     auto g0 = std::chrono::steady_clock::now();
     HIP_CHECK(hipSetDevice(device_));
-    HIP_CHECK(hipMemcpyAsync(d_resp_.p, h_resp_.p, resp_bytes,
-                             hipMemcpyHostToDevice, stream_));
     // resp_off | final_off | scratch_off contiguous, one copy; same for
     // msg_idx | skip
     uint32_t* d_off = (uint32_t*)d_off3_.p;
     int32_t* d_aux = (int32_t*)d_aux2_.p;
-    HIP_CHECK(hipMemcpyAsync(d_off, h_off, 3 * (n + 1) * sizeof(uint32_t),
-                             hipMemcpyHostToDevice, stream_));
-    HIP_CHECK(hipMemcpyAsync(d_aux, h_aux,
-                             (has_skip ? 2 : 1) * n * sizeof(int32_t),
-                             hipMemcpyHostToDevice, stream_));
-    int blocks = (int)cdiv(n, WPB);
+    // chunked pipeline for big response batches: resp H2D slices on
+    // stream2_ overlap the per-chunk kernels on stream_ (see
+    // run_encode_chunked; the compact/D2H tail below is shared)
+    bool chunked = false;
+    if (mode == 0 && has_skip && n >= 2 && pipe_chunks() > 1 &&
+        resp_bytes >= pipe_min_bytes()) {
+      int cut[PIPE_MAX_CHUNKS + 1];
+      int C = pipe_cuts(h_off, n, pipe_chunks(), cut);
+      if (C > 1) {
+        HIP_CHECK(hipMemcpyAsync(d_off, h_off, 3 * (n + 1) * sizeof(uint32_t),
+                                 hipMemcpyHostToDevice, stream_));
+        HIP_CHECK(hipMemcpyAsync(d_aux, h_aux, 2 * n * sizeof(int32_t),
+                                 hipMemcpyHostToDevice, stream_));
+        for (int c = 0; c < C; ++c) {
+          const int lo = cut[c], hi = cut[c + 1], m = hi - lo;
+          const size_t a = h_off[lo], b = h_off[hi];
+          if (b > a)
+            HIP_CHECK(hipMemcpyAsync((uint8_t*)d_resp_.p + a,
+                                     (uint8_t*)h_resp_.p + a, b - a,
+                                     hipMemcpyHostToDevice, stream2_));
+          HIP_CHECK(hipEventRecord(ev_h2d_[c], stream2_));
+          HIP_CHECK(hipStreamWaitEvent(stream_, ev_h2d_[c], 0));
+          int cblocks = (int)cdiv(m, WPB);
+          if (cblocks > 0) {
+            hipLaunchKernelGGL(
+                k_pb2json, dim3(cblocks), dim3(WPB * WAVE), 0, stream_,
+                (const uint8_t*)d_resp_.p, (const uint32_t*)d_off + lo,
+                (const int32_t*)d_aux + lo,
+                (const uint8_t*)d_id_slots_.p + (size_t)lo * ID_SLOT_BYTES,
+                (const SlotResult*)d_results_.p + lo, (uint8_t*)d_scratch_.p,
+                (const uint32_t*)d_off + 2 * (n + 1) + lo,
+                (uint8_t*)d_final_.p, (const uint32_t*)d_off + (n + 1) + lo,
+                (DecodeResult*)d_dec_results_.p + lo,
+                (const int32_t*)d_aux + n + lo, tables_, m, 0);
+            HIP_CHECK(hipGetLastError());
+          }
+          if (n_wg > 0) {
+            hipLaunchKernelGGL(
+                k_pb2json_wg, dim3(m), dim3(WG_DEC_WAVES * WAVE), 0, stream_,
+                (const uint8_t*)d_resp_.p, (const uint32_t*)d_off + lo,
+                (const int32_t*)d_aux + lo,
+                (const uint8_t*)d_id_slots_.p + (size_t)lo * ID_SLOT_BYTES,
+                (const SlotResult*)d_results_.p + lo, (uint8_t*)d_scratch_.p,
+                (const uint32_t*)d_off + 2 * (n + 1) + lo,
+                (uint8_t*)d_final_.p, (const uint32_t*)d_off + (n + 1) + lo,
+                (DecodeResult*)d_dec_results_.p + lo,
+                (const int32_t*)d_aux + n + lo, tables_, m, wg_phases());
+            HIP_CHECK(hipGetLastError());
+          }
+        }
+        chunked = true;
+      }
+    }
+    if (!chunked) {
+      HIP_CHECK(hipMemcpyAsync(d_resp_.p, h_resp_.p, resp_bytes,
+                               hipMemcpyHostToDevice, stream_));
+      HIP_CHECK(hipMemcpyAsync(d_off, h_off, 3 * (n + 1) * sizeof(uint32_t),
+                               hipMemcpyHostToDevice, stream_));
+      HIP_CHECK(hipMemcpyAsync(d_aux, h_aux,
+                               (has_skip ? 2 : 1) * n * sizeof(int32_t),
+                               hipMemcpyHostToDevice, stream_));
+    }
+    int blocks = chunked ? 0 : (int)cdiv(n, WPB);
     if (blocks > 0) {
       hipLaunchKernelGGL(
           k_pb2json, dim3(blocks), dim3(WPB * WAVE), 0, stream_,
@@ -1074,7 +1267,7 @@ class Engine : public spanapi::ISpanExecutor {
           has_skip ? (const int32_t*)d_aux + n : nullptr, tables_, n, mode);
       HIP_CHECK(hipGetLastError());
     }
-    if (n_wg > 0) {
+    if (n_wg > 0 && !chunked) {
       hipLaunchKernelGGL(
           k_pb2json_wg, dim3(n), dim3(WG_DEC_WAVES * WAVE), 0, stream_,
           (const uint8_t*)d_resp_.p, (const uint32_t*)d_off,
@@ -1160,7 +1353,11 @@ class Engine : public spanapi::ISpanExecutor {
   size_t compact_bytes_ = 0;
   size_t last_final_bytes_ = 0;
   hipStream_t stream_;
+  hipStream_t stream2_ = nullptr;
   hipEvent_t sync_ev_ = nullptr;
+  hipEvent_t sync_ev2_ = nullptr;
+  hipEvent_t ev_h2d_[PIPE_MAX_CHUNKS] = {};
+  hipEvent_t ev_krn_[PIPE_MAX_CHUNKS] = {};
   Tables tables_{};
   DeviceBuf d_msgs_, d_fields_, d_enums_, d_enum_vals_, d_tools_, d_names_;
   DeviceBuf d_in_, d_pb_, d_resp_, d_scratch_, d_final_;
