@@ -324,8 +324,11 @@ class Engine : public spanapi::ISpanExecutor {
     return x < 1 ? 1 : (x > PIPE_MAX_CHUNKS ? PIPE_MAX_CHUNKS : x);
   }
   static size_t pipe_min_bytes() {
+    // default 1 MiB: serving-shaped batches (<=0.5 MB) stay single-shot —
+    // extra submissions are the multi-instance contention point — while
+    // wide-payload batches (wide64: ~4 MB each way) always chunk
     const char* e = getenv("GGRMCP_PIPE_MIN");
-    return e ? (size_t)atoll(e) : (size_t)262144;
+    return e ? (size_t)atoll(e) : (size_t)(1u << 20);
   }
   // cut [0,n) into <=C ranges balanced by in_off bytes; returns #chunks
   static int pipe_cuts(const uint32_t* off, int n, int C, int* cut) {
