@@ -232,8 +232,11 @@ class Engine : public spanapi::ISpanExecutor {
     // copy stream for the chunked large-batch pipeline (H2D of chunk c+1
     // and D2H of chunk c-1 overlap chunk c's kernels on stream_)
     HIP_CHECK(hipStreamCreateWithFlags(&stream2_, hipStreamNonBlocking));
+    HIP_CHECK(hipStreamCreateWithFlags(&stream3_, hipStreamNonBlocking));
     HIP_CHECK(hipEventCreateWithFlags(
         &sync_ev2_, hipEventBlockingSync | hipEventDisableTiming));
+    HIP_CHECK(hipEventCreateWithFlags(
+        &sync_ev3_, hipEventBlockingSync | hipEventDisableTiming));
     for (int i = 0; i < PIPE_MAX_CHUNKS; ++i) {
       HIP_CHECK(hipEventCreateWithFlags(&ev_h2d_[i], hipEventDisableTiming));
       HIP_CHECK(hipEventCreateWithFlags(&ev_krn_[i], hipEventDisableTiming));
@@ -287,14 +290,17 @@ class Engine : public spanapi::ISpanExecutor {
   ~Engine() {
     (void)hipStreamSynchronize(stream_);
     (void)hipStreamSynchronize(stream2_);
+    (void)hipStreamSynchronize(stream3_);
     (void)hipEventDestroy(sync_ev_);
     (void)hipEventDestroy(sync_ev2_);
+    (void)hipEventDestroy(sync_ev3_);
     for (int i = 0; i < PIPE_MAX_CHUNKS; ++i) {
       (void)hipEventDestroy(ev_h2d_[i]);
       (void)hipEventDestroy(ev_krn_[i]);
     }
     (void)hipStreamDestroy(stream_);
     (void)hipStreamDestroy(stream2_);
+    (void)hipStreamDestroy(stream3_);
   }
 
   // park-the-thread stream wait (see sync_ev_ above)
@@ -306,6 +312,10 @@ class Engine : public spanapi::ISpanExecutor {
   void sync_stream2() {
     HIP_CHECK(hipEventRecord(sync_ev2_, stream2_));
     HIP_CHECK(hipEventSynchronize(sync_ev2_));
+  }
+  void sync_stream3() {
+    HIP_CHECK(hipEventRecord(sync_ev3_, stream3_));
+    HIP_CHECK(hipEventSynchronize(sync_ev3_));
   }
 
   // ---- chunked copy/compute pipeline (large batches only) -----------------
@@ -1092,7 +1102,7 @@ class Engine : public spanapi::ISpanExecutor {
   }
 
   // Chunked encode (mode 0, no msg_idx — the serving/wide shapes): chunk
-  // c+1's input H2D and chunk c-1's pb/results D2H run on stream2_ while
+  // c+1's input H2D (stream2_) and chunk c-1's pb/results D2H (stream3_)
   // chunk c's kernels run on stream_.  All kernels stay on ONE stream
   // (serialized — no extra kernel-queue contention); only copies overlap.
   // Per-item addressing is absolute through the offset tables, so a
@@ -1139,22 +1149,22 @@ class Engine : public spanapi::ISpanExecutor {
         HIP_CHECK(hipGetLastError());
       }
       HIP_CHECK(hipEventRecord(ev_krn_[c], stream_));
-      HIP_CHECK(hipStreamWaitEvent(stream2_, ev_krn_[c], 0));
+      HIP_CHECK(hipStreamWaitEvent(stream3_, ev_krn_[c], 0));
       const size_t pa = h_pb_off[lo], pe = h_pb_off[hi];
       if (pe > pa)
         HIP_CHECK(hipMemcpyAsync((uint8_t*)h_pb_.p + pa, (uint8_t*)d_pb_.p + pa,
-                                 pe - pa, hipMemcpyDeviceToHost, stream2_));
+                                 pe - pa, hipMemcpyDeviceToHost, stream3_));
       HIP_CHECK(hipMemcpyAsync((SlotResult*)h_results_.p + lo,
                                (SlotResult*)d_results_.p + lo,
                                (size_t)m * sizeof(SlotResult),
-                               hipMemcpyDeviceToHost, stream2_));
+                               hipMemcpyDeviceToHost, stream3_));
       HIP_CHECK(hipMemcpyAsync(
           (uint8_t*)h_id_.p + (size_t)lo * ID_SLOT_BYTES,
           (uint8_t*)d_id_slots_.p + (size_t)lo * ID_SLOT_BYTES,
-          (size_t)m * ID_SLOT_BYTES, hipMemcpyDeviceToHost, stream2_));
+          (size_t)m * ID_SLOT_BYTES, hipMemcpyDeviceToHost, stream3_));
     }
     sync_stream();
-    sync_stream2();
+    sync_stream3();
   }
 
   py::tuple run_encode(int n, size_t in_bytes, size_t pb_bytes, bool has_idx,
@@ -1357,8 +1367,10 @@ class Engine : public spanapi::ISpanExecutor {
   size_t last_final_bytes_ = 0;
   hipStream_t stream_;
   hipStream_t stream2_ = nullptr;
+  hipStream_t stream3_ = nullptr;
   hipEvent_t sync_ev_ = nullptr;
   hipEvent_t sync_ev2_ = nullptr;
+  hipEvent_t sync_ev3_ = nullptr;
   hipEvent_t ev_h2d_[PIPE_MAX_CHUNKS] = {};
   hipEvent_t ev_krn_[PIPE_MAX_CHUNKS] = {};
   Tables tables_{};
