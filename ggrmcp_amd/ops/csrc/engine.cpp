@@ -329,8 +329,16 @@ class Engine : public spanapi::ISpanExecutor {
   // (profiles/streams_sweep.log), so serving-sized batches keep the
   // single-shot path.
   static int pipe_chunks() {
+    // DEFAULT 1 (off).  Measured negative result (tools/pipe_probe.py,
+    // profiles/pipe_chunks_r02.md): the wg kernels are LATENCY-bound —
+    // a wide64 batch is 64 workgroups on 256 CUs, so one launch costs one
+    // request's serial depth (~2.4 ms) regardless of item count, and C
+    // sequential sub-launches cost C x that serial depth (~1.9 ms per
+    // extra chunk, single engine, independent of GPU_MAX_HW_QUEUES) —
+    // far more than the ~1 ms of copies the chunks overlap.  Kept
+    // env-gated for experiments; correctness held (byte-identical wire).
     const char* e = getenv("GGRMCP_PIPE_CHUNKS");  // per-call, like wg_*_min
-    int x = e ? atoi(e) : 4;
+    int x = e ? atoi(e) : 1;
     return x < 1 ? 1 : (x > PIPE_MAX_CHUNKS ? PIPE_MAX_CHUNKS : x);
   }
   static size_t pipe_min_bytes() {
