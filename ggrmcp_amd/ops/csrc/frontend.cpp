@@ -83,6 +83,7 @@ struct Conn {
   uint64_t next_write = 0;    // next response sequence to write
   std::unordered_map<uint64_t, std::string> ready;  // out-of-order responses
   bool closing = false;
+  bool want_out = false;  // EPOLLOUT currently armed (skip redundant MODs)
 };
 
 std::string http_response(int status, const std::string& body,
@@ -600,13 +601,20 @@ class Frontend {
     auto it = re->conns.find(id);
     if (it == re->conns.end()) return;
     Conn* c = it->second.get();
-    c->ready[seq] = std::move(payload);
-    while (true) {
-      auto rit = c->ready.find(c->next_write);
-      if (rit == c->ready.end()) break;
-      c->wbuf += rit->second;
-      c->ready.erase(rit);
+    if (seq == c->next_write && c->ready.empty()) {
+      // common case: in-order completion (closed-loop clients keep one
+      // request in flight per connection) — skip the reorder map
+      c->wbuf += payload;
       c->next_write++;
+    } else {
+      c->ready[seq] = std::move(payload);
+      while (true) {
+        auto rit = c->ready.find(c->next_write);
+        if (rit == c->ready.end()) break;
+        c->wbuf += rit->second;
+        c->ready.erase(rit);
+        c->next_write++;
+      }
     }
     flush_conn(re, id, c);
   }
@@ -617,20 +625,28 @@ class Frontend {
       if (w > 0) {
         c->wbuf.erase(0, (size_t)w);
       } else if (w < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) {
-        epoll_event ev{};
-        ev.events = EPOLLIN | EPOLLOUT;
-        ev.data.u64 = id;
-        epoll_ctl(re->epfd, EPOLL_CTL_MOD, c->fd, &ev);
+        if (!c->want_out) {
+          epoll_event ev{};
+          ev.events = EPOLLIN | EPOLLOUT;
+          ev.data.u64 = id;
+          epoll_ctl(re->epfd, EPOLL_CTL_MOD, c->fd, &ev);
+          c->want_out = true;
+        }
         return;
       } else {
         drop_conn(re, id);
         return;
       }
     }
-    epoll_event ev{};
-    ev.events = EPOLLIN;
-    ev.data.u64 = id;
-    epoll_ctl(re->epfd, EPOLL_CTL_MOD, c->fd, &ev);
+    // only disarm EPOLLOUT when it was armed: the drained-buffer case is
+    // the hot path (one MOD per response otherwise — measurable at 350k/s)
+    if (c->want_out) {
+      epoll_event ev{};
+      ev.events = EPOLLIN;
+      ev.data.u64 = id;
+      epoll_ctl(re->epfd, EPOLL_CTL_MOD, c->fd, &ev);
+      c->want_out = false;
+    }
     if (c->closing) drop_conn(re, id);
   }
 
