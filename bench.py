@@ -214,11 +214,14 @@ def main() -> None:
         if args.config == "hello1k" and use_gpu:
             # tuned serving defaults (profiles/serving_sweep_r02.log):
             # reactors 4->8 and span engines 2->6 lifted 1-GPU serving
-            # 242k -> 363k req/s; explicit flags still override.  Multi-rank
-            # weak scaling shares the node's cores across ranks, so shed
-            # client threads there (the load generator competes hardest).
+            # 242k -> 363k req/s; reactors 8->12 (after the flush-path
+            # epoll_ctl fix) a further 379k -> 426k at 800 steps on a
+            # 256-core box (gpurun_out/cf_r12t16.json).  Explicit flags
+            # still override.  Multi-rank weak scaling shares the node's
+            # cores across ranks, so shed reactors and client threads
+            # there (the load generator competes hardest).
             cfg.gpu.streams = 6
-            cfg.server.reactors = 8 if world == 1 else 6
+            cfg.server.reactors = 12 if world == 1 else 6
             if world > 1 and args.client_threads == 16:
                 args.client_threads = 8
         if args.streams > 0:
