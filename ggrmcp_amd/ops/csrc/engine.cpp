@@ -74,6 +74,13 @@ struct PinnedBuf {
 };
 
 uint64_t cdiv(uint64_t a, uint64_t b) { return (a + b - 1) / b; }
+inline size_t align64(size_t x) { return (x + 63) & ~(size_t)63; }
+
+// non-owning window into a packed DeviceBuf/PinnedBuf allocation
+struct BufView {
+  void* p = nullptr;
+  size_t n = 0;
+};
 
 // workgroup-decode routing threshold; GGRMCP_WG_DEC_MIN overrides for
 // classic-vs-cooperative differential tests (read per batch: ~100 ns)
@@ -263,28 +270,43 @@ class Engine : public spanapi::ISpanExecutor {
     d_scratch_.alloc(cap_scratch);
     d_final_.alloc(cap_final);
     size_t offs = (size_t)(max_batch + 1) * sizeof(uint32_t);
-    // offsets + aux travel as ONE H2D copy each (they are contiguous in
-    // h_off_/h_aux_): fewer HIP API calls per chunk — the submission path
-    // is the measured multi-instance contention point
-    d_off3_.alloc(offs * 3);
-    d_aux2_.alloc((size_t)max_batch * sizeof(int32_t) * 2);
+    // Packed control block (off3 | aux2 in ONE allocation each side) and
+    // packed result block (results | ids): offsets+aux travel as ONE H2D
+    // and results+ids as ONE D2H per batch.  The per-copy-op COUNT is the
+    // measured multi-instance ceiling (~3.5k batches/s aggregate across
+    // any number of engine instances, profiles/contention_r02.md), so
+    // fewer, slightly larger copies beat many small ones.
+    ctrl_aux_off_ = align64(offs * 3);
+    size_t aux_bytes = (size_t)max_batch * sizeof(int32_t) * 2;
+    d_ctrl_.alloc(ctrl_aux_off_ + aux_bytes);
+    d_off3_ = {d_ctrl_.p, offs * 3};
+    d_aux2_ = {(uint8_t*)d_ctrl_.p + ctrl_aux_off_, aux_bytes};
     d_tight_off_.alloc(offs);
-    d_results_.alloc((size_t)max_batch * sizeof(SlotResult));
+    // ids live right after the USED results span (rid_off_, set per
+    // encode) so the packed D2H has no dead middle; init to the full
+    // classic layout so mode-1 decodes before any encode stay valid
+    size_t res_cap = align64((size_t)max_batch * sizeof(SlotResult));
+    size_t id_cap = (size_t)max_batch * ID_SLOT_BYTES;
+    rid_off_ = res_cap;
+    d_rid_.alloc(res_cap + id_cap);
+    d_results_ = {d_rid_.p, (size_t)max_batch * sizeof(SlotResult)};
+    d_id_slots_ = {(uint8_t*)d_rid_.p + rid_off_, id_cap};
     d_dec_results_.alloc((size_t)max_batch * sizeof(DecodeResult));
-    d_id_slots_.alloc((size_t)max_batch * ID_SLOT_BYTES);
 
     h_in_.alloc(cap_in);
     h_pb_.alloc(cap_pb);
     h_resp_.alloc(cap_in);
     h_final_.alloc(cap_final);
-    h_results_.alloc((size_t)max_batch * sizeof(SlotResult));
     h_dec_results_.alloc((size_t)max_batch * sizeof(DecodeResult));
-    h_off_.alloc(offs * 3);
+    h_ctrl_.alloc(ctrl_aux_off_ + aux_bytes);
+    h_off_ = {h_ctrl_.p, offs * 3};
+    h_aux_ = {(uint8_t*)h_ctrl_.p + ctrl_aux_off_, aux_bytes};
     h_tight_.alloc(offs);
-    h_aux_.alloc((size_t)max_batch * sizeof(int32_t) * 2);
     // JSON-RPC id tokens on the host: the native span executor assembles
     // error envelopes in C++ and needs the ids the encode kernel captured
-    h_id_.alloc((size_t)max_batch * ID_SLOT_BYTES);
+    h_rid_.alloc(res_cap + id_cap);
+    h_results_ = {h_rid_.p, (size_t)max_batch * sizeof(SlotResult)};
+    h_id_ = {(uint8_t*)h_rid_.p + rid_off_, id_cap};
   }
 
   ~Engine() {
@@ -1043,6 +1065,10 @@ class Engine : public spanapi::ISpanExecutor {
         n_wg += esk[i] ? 1 : 0;
       }
     }
+    // pack ids right after the USED results span: one D2H returns both
+    rid_off_ = align64((size_t)n * sizeof(SlotResult));
+    d_id_slots_.p = (uint8_t*)d_rid_.p + rid_off_;
+    h_id_.p = (uint8_t*)h_rid_.p + rid_off_;
     auto g0 = std::chrono::steady_clock::now();
     HIP_CHECK(hipSetDevice(device_));
     // chunked pipeline for big batches (mode 0, no msg_idx): overlap the
@@ -1063,11 +1089,13 @@ class Engine : public spanapi::ISpanExecutor {
     HIP_CHECK(hipMemcpyAsync(d_in_.p, h_in_.p, in_bytes,
                              hipMemcpyHostToDevice, stream_));
     uint32_t* d_off = (uint32_t*)d_off3_.p;  // in_off | pb_off contiguous
-    HIP_CHECK(hipMemcpyAsync(d_off, h_off, 2 * (n + 1) * sizeof(uint32_t),
-                             hipMemcpyHostToDevice, stream_));
-    if (has_idx || n_wg > 0)
-      HIP_CHECK(hipMemcpyAsync(d_aux2_.p, h_aux_.p, n * sizeof(int32_t),
-                               hipMemcpyHostToDevice, stream_));
+    // offsets (+ aux when used) in ONE packed H2D: the dead off3 tail it
+    // spans (<=48 KB) is ~1 us of bandwidth vs ~35 us for a second op
+    HIP_CHECK(hipMemcpyAsync(
+        d_ctrl_.p, h_ctrl_.p,
+        (has_idx || n_wg > 0) ? ctrl_aux_off_ + (size_t)n * sizeof(int32_t)
+                              : 2 * (n + 1) * sizeof(uint32_t),
+        hipMemcpyHostToDevice, stream_));
     int blocks = (int)cdiv(n, WPB);
     if (blocks > 0) {
       hipLaunchKernelGGL(k_json2pb, dim3(blocks), dim3(WPB * WAVE), 0,
@@ -1090,14 +1118,12 @@ class Engine : public spanapi::ISpanExecutor {
                          (const int32_t*)d_aux2_.p, wg_enc_phases());
       HIP_CHECK(hipGetLastError());
     }
-    HIP_CHECK(hipMemcpyAsync(h_results_.p, d_results_.p,
-                             n * sizeof(SlotResult), hipMemcpyDeviceToHost,
-                             stream_));
-    // id tokens to the host: the native span's C++ error envelopes need
-    // them (mode-0 decode keeps reading the device copy)
-    HIP_CHECK(hipMemcpyAsync(h_id_.p, d_id_slots_.p,
-                             (size_t)n * ID_SLOT_BYTES, hipMemcpyDeviceToHost,
-                             stream_));
+    // results + id tokens in ONE packed D2H (ids sit at rid_off_; the
+    // native span's C++ error envelopes need them on the host, mode-0
+    // decode keeps reading the device copy)
+    HIP_CHECK(hipMemcpyAsync(h_rid_.p, d_rid_.p,
+                             rid_off_ + (size_t)n * ID_SLOT_BYTES,
+                             hipMemcpyDeviceToHost, stream_));
     HIP_CHECK(hipMemcpyAsync(h_pb_.p, d_pb_.p, pb_bytes,
                              hipMemcpyDeviceToHost, stream_));
     sync_stream();
@@ -1222,10 +1248,10 @@ class Engine : public spanapi::ISpanExecutor {
       int cut[PIPE_MAX_CHUNKS + 1];
       int C = pipe_cuts(h_off, n, pipe_chunks(), cut);
       if (C > 1) {
-        HIP_CHECK(hipMemcpyAsync(d_off, h_off, 3 * (n + 1) * sizeof(uint32_t),
-                                 hipMemcpyHostToDevice, stream_));
-        HIP_CHECK(hipMemcpyAsync(d_aux, h_aux, 2 * n * sizeof(int32_t),
-                                 hipMemcpyHostToDevice, stream_));
+        HIP_CHECK(hipMemcpyAsync(
+            d_ctrl_.p, h_ctrl_.p,
+            ctrl_aux_off_ + 2 * (size_t)n * sizeof(int32_t),
+            hipMemcpyHostToDevice, stream_));
         for (int c = 0; c < C; ++c) {
           const int lo = cut[c], hi = cut[c + 1], m = hi - lo;
           const size_t a = h_off[lo], b = h_off[hi];
@@ -1269,11 +1295,11 @@ class Engine : public spanapi::ISpanExecutor {
     if (!chunked) {
       HIP_CHECK(hipMemcpyAsync(d_resp_.p, h_resp_.p, resp_bytes,
                                hipMemcpyHostToDevice, stream_));
-      HIP_CHECK(hipMemcpyAsync(d_off, h_off, 3 * (n + 1) * sizeof(uint32_t),
-                               hipMemcpyHostToDevice, stream_));
-      HIP_CHECK(hipMemcpyAsync(d_aux, h_aux,
-                               (has_skip ? 2 : 1) * n * sizeof(int32_t),
-                               hipMemcpyHostToDevice, stream_));
+      // offsets + msg_idx/skip in ONE packed H2D (see run_encode_device)
+      HIP_CHECK(hipMemcpyAsync(
+          d_ctrl_.p, h_ctrl_.p,
+          ctrl_aux_off_ + (has_skip ? 2 : 1) * (size_t)n * sizeof(int32_t),
+          hipMemcpyHostToDevice, stream_));
     }
     int blocks = chunked ? 0 : (int)cdiv(n, WPB);
     if (blocks > 0) {
@@ -1384,11 +1410,13 @@ class Engine : public spanapi::ISpanExecutor {
   Tables tables_{};
   DeviceBuf d_msgs_, d_fields_, d_enums_, d_enum_vals_, d_tools_, d_names_;
   DeviceBuf d_in_, d_pb_, d_resp_, d_scratch_, d_final_;
-  DeviceBuf d_off3_, d_aux2_;
-  DeviceBuf d_results_, d_dec_results_, d_id_slots_,
-      d_tight_off_;
-  PinnedBuf h_in_, h_pb_, h_resp_, h_final_, h_results_, h_dec_results_, h_off_,
-      h_aux_, h_tight_, h_id_;
+  DeviceBuf d_ctrl_, d_rid_, d_dec_results_, d_tight_off_;
+  BufView d_off3_, d_aux2_, d_results_, d_id_slots_;
+  PinnedBuf h_in_, h_pb_, h_resp_, h_final_, h_dec_results_, h_ctrl_, h_rid_,
+      h_tight_;
+  BufView h_off_, h_aux_, h_results_, h_id_;
+  size_t ctrl_aux_off_ = 0;  // aux2 offset inside the ctrl block
+  size_t rid_off_ = 0;       // ids offset inside the rid block (per batch)
   std::vector<std::string> tool_paths_;
   std::vector<int32_t> tool_out_msg_;
   std::vector<int32_t> tool_backend_;
