@@ -281,7 +281,6 @@ class Engine : public spanapi::ISpanExecutor {
     d_ctrl_.alloc(ctrl_aux_off_ + aux_bytes);
     d_off3_ = {d_ctrl_.p, offs * 3};
     d_aux2_ = {(uint8_t*)d_ctrl_.p + ctrl_aux_off_, aux_bytes};
-    d_tight_off_.alloc(offs);
     // ids live right after the USED results span (rid_off_, set per
     // encode) so the packed D2H has no dead middle; init to the full
     // classic layout so mode-1 decodes before any encode stay valid
@@ -301,7 +300,6 @@ class Engine : public spanapi::ISpanExecutor {
     h_ctrl_.alloc(ctrl_aux_off_ + aux_bytes);
     h_off_ = {h_ctrl_.p, offs * 3};
     h_aux_ = {(uint8_t*)h_ctrl_.p + ctrl_aux_off_, aux_bytes};
-    h_tight_.alloc(offs);
     // JSON-RPC id tokens on the host: the native span executor assembles
     // error envelopes in C++ and needs the ids the encode kernel captured
     h_rid_.alloc(res_cap + id_cap);
@@ -1325,37 +1323,41 @@ class Engine : public spanapi::ISpanExecutor {
           (const int32_t*)d_aux + n, tables_, n, wg_phases());
       HIP_CHECK(hipGetLastError());
     }
+    // Device-side compaction: k_tight_scan rewrites out_off to a packed
+    // layout and k_compact_out gathers the used bytes, all before the
+    // first sync — no host prefix-sum round-trip, no tight-table H2D.
+    uint32_t dst_cap = (uint32_t)std::min(d_pb_.n, h_pb_.n);
+    if (n > 0) {
+      hipLaunchKernelGGL(k_tight_scan, dim3(1), dim3(256), 0, stream_,
+                         (DecodeResult*)d_dec_results_.p, n);
+      HIP_CHECK(hipGetLastError());
+      hipLaunchKernelGGL(k_compact_out, dim3(n), dim3(256), 0, stream_,
+                         (const uint8_t*)d_final_.p,
+                         (const uint32_t*)d_off + (n + 1),
+                         (const DecodeResult*)d_dec_results_.p,
+                         (uint8_t*)d_pb_.p, dst_cap, n);
+      HIP_CHECK(hipGetLastError());
+    }
     HIP_CHECK(hipMemcpyAsync(h_dec_results_.p, d_dec_results_.p,
                              n * sizeof(DecodeResult), hipMemcpyDeviceToHost,
                              stream_));
     sync_stream();
-    // Gather used bytes into a tight buffer (reusing d_pb_/h_pb_) so the
-    // D2H copy is sum(out_len), not the arena's worst-case capacity.
     DecodeResult* rs = (DecodeResult*)h_dec_results_.p;
-    uint32_t* tight = (uint32_t*)h_tight_.p;
-    uint64_t acc = 0;
-    for (int i = 0; i < n; ++i) {
-      tight[i] = (uint32_t)acc;
-      acc += (rs[i].out_len + 3u) & ~3u;
-    }
-    if (n > 0 && acc <= d_pb_.n && acc <= h_pb_.n) {
-      HIP_CHECK(hipMemcpyAsync(d_tight_off_.p, tight, n * sizeof(uint32_t),
-                               hipMemcpyHostToDevice, stream_));
-      hipLaunchKernelGGL(k_compact_out, dim3(n), dim3(256), 0, stream_,
-                         (const uint8_t*)d_final_.p,
-                         (const uint32_t*)d_off + (n + 1),
-                         (const uint32_t*)d_tight_off_.p,
-                         (const DecodeResult*)d_dec_results_.p,
-                         (uint8_t*)d_pb_.p, n);
-      HIP_CHECK(hipGetLastError());
+    uint64_t acc =
+        n > 0 ? (uint64_t)rs[n - 1].out_off + ((rs[n - 1].out_len + 3u) & ~3u)
+              : 0;
+    if (n > 0 && acc <= dst_cap) {
       if (acc)
         HIP_CHECK(hipMemcpyAsync(h_pb_.p, d_pb_.p, acc,
                                  hipMemcpyDeviceToHost, stream_));
       sync_stream();
-      for (int i = 0; i < n; ++i) rs[i].out_off = tight[i];
       compact_bytes_ = acc;
       compact_used_ = true;
     } else {
+      // packed total overflows the staging arena (k_compact_out skipped
+      // those writes): restore arena offsets and copy the whole arena
+      const uint32_t* final_off = h_off + (n + 1);
+      for (int i = 0; i < n; ++i) rs[i].out_off = final_off[i];
       HIP_CHECK(hipMemcpyAsync(h_final_.p, d_final_.p, final_bytes,
                                hipMemcpyDeviceToHost, stream_));
       sync_stream();
@@ -1410,10 +1412,9 @@ class Engine : public spanapi::ISpanExecutor {
   Tables tables_{};
   DeviceBuf d_msgs_, d_fields_, d_enums_, d_enum_vals_, d_tools_, d_names_;
   DeviceBuf d_in_, d_pb_, d_resp_, d_scratch_, d_final_;
-  DeviceBuf d_ctrl_, d_rid_, d_dec_results_, d_tight_off_;
+  DeviceBuf d_ctrl_, d_rid_, d_dec_results_;
   BufView d_off3_, d_aux2_, d_results_, d_id_slots_;
-  PinnedBuf h_in_, h_pb_, h_resp_, h_final_, h_dec_results_, h_ctrl_, h_rid_,
-      h_tight_;
+  PinnedBuf h_in_, h_pb_, h_resp_, h_final_, h_dec_results_, h_ctrl_, h_rid_;
   BufView h_off_, h_aux_, h_results_, h_id_;
   size_t ctrl_aux_off_ = 0;  // aux2 offset inside the ctrl block
   size_t rid_off_ = 0;       // ids offset inside the rid block (per batch)

@@ -1688,14 +1688,19 @@ extern "C" __global__ void __launch_bounds__(WPB * WAVE) k_pb2json(
 // time for large payloads: 268 MB vs ~25 MB used at 256 x 64 KB).
 extern "C" __global__ void k_compact_out(
     const uint8_t* __restrict__ src, const uint32_t* __restrict__ src_off,
-    const uint32_t* __restrict__ dst_off,
     const DecodeResult* __restrict__ results, uint8_t* __restrict__ dst,
-    int n_req) {
+    uint32_t dst_cap, int n_req) {
+  // destination offsets come from results[].out_off, which k_tight_scan
+  // rewrote to the packed layout; dst_cap guards the (rare) case where
+  // the packed total exceeds the staging arena — the host then takes the
+  // arena-copy fallback and this kernel must not write out of bounds
   int req = blockIdx.x;
   if (req >= n_req) return;
   uint32_t len = results[req].out_len;
+  uint32_t doff = results[req].out_off;
+  if (doff + len > dst_cap || doff + len < doff) return;
   const uint8_t* s = src + src_off[req];
-  uint8_t* d = dst + dst_off[req];
+  uint8_t* d = dst + doff;
   uint32_t t = threadIdx.x, stride = blockDim.x;
   // dword-wide main copy (both offsets 4-aligned by construction)
   uint32_t words = len >> 2;
@@ -1704,6 +1709,42 @@ extern "C" __global__ void k_compact_out(
   for (uint32_t i = t; i < words; i += stride) d4[i] = s4[i];
   for (uint32_t i = (words << 2) + t; i < len; i += stride) d[i] = s[i];
 }
+
+#ifndef GGRMCP_HOST_SIM
+// Rewrites results[].out_off from arena offsets to a PACKED (4-aligned)
+// layout entirely on device, so the host never round-trips between the
+// decode kernels and k_compact_out (the old flow synced, prefix-summed
+// out_lens on the host, and uploaded a tight-offset table — one extra
+// blocking sync + one extra H2D per batch; the copy-op count is the
+// engine-aggregate ceiling, profiles/contention_r02.md).  One block;
+// contiguous per-thread chunks then a Hillis-Steele scan of the 256
+// partials in LDS.  The packed total lands in rs[n-1] (host recomputes
+// it from out_off + out_len); if it exceeds the arena the host restores
+// arena offsets from final_off and takes the fallback copy.
+extern "C" __global__ void __launch_bounds__(256) k_tight_scan(
+    DecodeResult* __restrict__ rs, int n) {
+  __shared__ uint32_t partial[256];
+  int t = threadIdx.x;
+  int per = (n + 255) / 256;
+  int lo = t * per, hi = n < lo + per ? n : lo + per;
+  uint32_t sum = 0;
+  for (int i = lo; i < hi; ++i) sum += (rs[i].out_len + 3u) & ~3u;
+  partial[t] = sum;
+  __syncthreads();
+  for (int d = 1; d < 256; d <<= 1) {
+    uint32_t v = (t >= d) ? partial[t - d] : 0;
+    __syncthreads();
+    partial[t] += v;
+    __syncthreads();
+  }
+  uint32_t acc = (t == 0) ? 0 : partial[t - 1];
+  for (int i = lo; i < hi; ++i) {
+    uint32_t len4 = (rs[i].out_len + 3u) & ~3u;
+    rs[i].out_off = acc;
+    acc += len4;
+  }
+}
+#endif  // GGRMCP_HOST_SIM
 
 #ifndef GGRMCP_HOST_SIM
 // ---------------------------------------------------------------------------
