@@ -265,7 +265,6 @@ class Engine : public spanapi::ISpanExecutor {
     tables_.n_tools = n_tools;
 
     d_in_.alloc(cap_in);
-    d_pb_.alloc(cap_pb);
     d_resp_.alloc(cap_in);
     d_scratch_.alloc(cap_scratch);
     d_final_.alloc(cap_final);
@@ -278,33 +277,42 @@ class Engine : public spanapi::ISpanExecutor {
     // fewer, slightly larger copies beat many small ones.
     ctrl_aux_off_ = align64(offs * 3);
     size_t aux_bytes = (size_t)max_batch * sizeof(int32_t) * 2;
-    d_ctrl_.alloc(ctrl_aux_off_ + aux_bytes);
-    d_off3_ = {d_ctrl_.p, offs * 3};
-    d_aux2_ = {(uint8_t*)d_ctrl_.p + ctrl_aux_off_, aux_bytes};
-    // ids live right after the USED results span (rid_off_, set per
-    // encode) so the packed D2H has no dead middle; init to the full
-    // classic layout so mode-1 decodes before any encode stay valid
+    // control block (off3 | aux2) is ZERO-COPY: the kernels read offsets
+    // and routing flags straight from pinned host memory through its
+    // device pointer — a few latency-hidden reads per wave beat a ~50 us
+    // hipMemcpyAsync SUBMISSION (profiles/contention_r02.md: the engine
+    // span is ~85% API-submission time, so the op count is the ceiling)
+    h_ctrl_.alloc(ctrl_aux_off_ + aux_bytes);
+    void* ctrl_dev = nullptr;
+    HIP_CHECK(hipHostGetDevicePointer(&ctrl_dev, h_ctrl_.p, 0));
+    d_off3_ = {ctrl_dev, offs * 3};
+    d_aux2_ = {(uint8_t*)ctrl_dev + ctrl_aux_off_, aux_bytes};
+    h_off_ = {h_ctrl_.p, offs * 3};
+    h_aux_ = {(uint8_t*)h_ctrl_.p + ctrl_aux_off_, aux_bytes};
+    // results | ids | pb wire in ONE allocation each side: encode returns
+    // everything in ONE D2H span (ids sit right after the USED results at
+    // rid_off_, set per encode; pb at the fixed rid_cap_ — the dead gap
+    // it spans is bandwidth, not an extra submission)
     size_t res_cap = align64((size_t)max_batch * sizeof(SlotResult));
     size_t id_cap = (size_t)max_batch * ID_SLOT_BYTES;
     rid_off_ = res_cap;
-    d_rid_.alloc(res_cap + id_cap);
+    rid_cap_ = align64(res_cap + id_cap);
+    d_rid_.alloc(rid_cap_ + cap_pb);
     d_results_ = {d_rid_.p, (size_t)max_batch * sizeof(SlotResult)};
     d_id_slots_ = {(uint8_t*)d_rid_.p + rid_off_, id_cap};
+    d_pb_ = {(uint8_t*)d_rid_.p + rid_cap_, cap_pb};
     d_dec_results_.alloc((size_t)max_batch * sizeof(DecodeResult));
 
     h_in_.alloc(cap_in);
-    h_pb_.alloc(cap_pb);
     h_resp_.alloc(cap_in);
     h_final_.alloc(cap_final);
     h_dec_results_.alloc((size_t)max_batch * sizeof(DecodeResult));
-    h_ctrl_.alloc(ctrl_aux_off_ + aux_bytes);
-    h_off_ = {h_ctrl_.p, offs * 3};
-    h_aux_ = {(uint8_t*)h_ctrl_.p + ctrl_aux_off_, aux_bytes};
     // JSON-RPC id tokens on the host: the native span executor assembles
     // error envelopes in C++ and needs the ids the encode kernel captured
-    h_rid_.alloc(res_cap + id_cap);
+    h_rid_.alloc(rid_cap_ + cap_pb);
     h_results_ = {h_rid_.p, (size_t)max_batch * sizeof(SlotResult)};
     h_id_ = {(uint8_t*)h_rid_.p + rid_off_, id_cap};
+    h_pb_ = {(uint8_t*)h_rid_.p + rid_cap_, cap_pb};
   }
 
   ~Engine() {
@@ -1086,14 +1094,9 @@ class Engine : public spanapi::ISpanExecutor {
     }
     HIP_CHECK(hipMemcpyAsync(d_in_.p, h_in_.p, in_bytes,
                              hipMemcpyHostToDevice, stream_));
+    // offsets + aux are zero-copy: the kernels read them from pinned host
+    // memory (d_off3_/d_aux2_ are device pointers INTO h_ctrl_)
     uint32_t* d_off = (uint32_t*)d_off3_.p;  // in_off | pb_off contiguous
-    // offsets (+ aux when used) in ONE packed H2D: the dead off3 tail it
-    // spans (<=48 KB) is ~1 us of bandwidth vs ~35 us for a second op
-    HIP_CHECK(hipMemcpyAsync(
-        d_ctrl_.p, h_ctrl_.p,
-        (has_idx || n_wg > 0) ? ctrl_aux_off_ + (size_t)n * sizeof(int32_t)
-                              : 2 * (n + 1) * sizeof(uint32_t),
-        hipMemcpyHostToDevice, stream_));
     int blocks = (int)cdiv(n, WPB);
     if (blocks > 0) {
       hipLaunchKernelGGL(k_json2pb, dim3(blocks), dim3(WPB * WAVE), 0,
@@ -1116,13 +1119,12 @@ class Engine : public spanapi::ISpanExecutor {
                          (const int32_t*)d_aux2_.p, wg_enc_phases());
       HIP_CHECK(hipGetLastError());
     }
-    // results + id tokens in ONE packed D2H (ids sit at rid_off_; the
-    // native span's C++ error envelopes need them on the host, mode-0
-    // decode keeps reading the device copy)
-    HIP_CHECK(hipMemcpyAsync(h_rid_.p, d_rid_.p,
-                             rid_off_ + (size_t)n * ID_SLOT_BYTES,
-                             hipMemcpyDeviceToHost, stream_));
-    HIP_CHECK(hipMemcpyAsync(h_pb_.p, d_pb_.p, pb_bytes,
+    // results + id tokens + pb wire in ONE packed D2H span (ids sit at
+    // rid_off_, pb at the fixed rid_cap_; the gap between used ids and
+    // rid_cap_ is dead bandwidth, far cheaper than a second submission).
+    // The native span's C++ error envelopes need the ids on the host;
+    // mode-0 decode keeps reading the device copies.
+    HIP_CHECK(hipMemcpyAsync(h_rid_.p, d_rid_.p, rid_cap_ + pb_bytes,
                              hipMemcpyDeviceToHost, stream_));
     sync_stream();
     // pure device span (copies+kernel+sync, no GIL-reacquire wait):
@@ -1142,13 +1144,7 @@ class Engine : public spanapi::ISpanExecutor {
   void run_encode_chunked(int n, int C, const int* cut, int n_wg, Limits lim) {
     uint32_t* h_off = (uint32_t*)h_off_.p;
     const uint32_t* h_pb_off = h_off + (n + 1);
-    uint32_t* d_off = (uint32_t*)d_off3_.p;  // in_off | pb_off contiguous
-    // offsets (+ wg routing flags): every chunk's kernels read them
-    HIP_CHECK(hipMemcpyAsync(d_off, h_off, 2 * (n + 1) * sizeof(uint32_t),
-                             hipMemcpyHostToDevice, stream_));
-    if (n_wg > 0)
-      HIP_CHECK(hipMemcpyAsync(d_aux2_.p, h_aux_.p, n * sizeof(int32_t),
-                               hipMemcpyHostToDevice, stream_));
+    uint32_t* d_off = (uint32_t*)d_off3_.p;  // zero-copy (pinned host)
     for (int c = 0; c < C; ++c) {
       const int lo = cut[c], hi = cut[c + 1], m = hi - lo;
       const size_t a = h_off[lo], b = h_off[hi];
@@ -1246,10 +1242,6 @@ class Engine : public spanapi::ISpanExecutor {
       int cut[PIPE_MAX_CHUNKS + 1];
       int C = pipe_cuts(h_off, n, pipe_chunks(), cut);
       if (C > 1) {
-        HIP_CHECK(hipMemcpyAsync(
-            d_ctrl_.p, h_ctrl_.p,
-            ctrl_aux_off_ + 2 * (size_t)n * sizeof(int32_t),
-            hipMemcpyHostToDevice, stream_));
         for (int c = 0; c < C; ++c) {
           const int lo = cut[c], hi = cut[c + 1], m = hi - lo;
           const size_t a = h_off[lo], b = h_off[hi];
@@ -1293,11 +1285,7 @@ class Engine : public spanapi::ISpanExecutor {
     if (!chunked) {
       HIP_CHECK(hipMemcpyAsync(d_resp_.p, h_resp_.p, resp_bytes,
                                hipMemcpyHostToDevice, stream_));
-      // offsets + msg_idx/skip in ONE packed H2D (see run_encode_device)
-      HIP_CHECK(hipMemcpyAsync(
-          d_ctrl_.p, h_ctrl_.p,
-          ctrl_aux_off_ + (has_skip ? 2 : 1) * (size_t)n * sizeof(int32_t),
-          hipMemcpyHostToDevice, stream_));
+      // offsets + msg_idx/skip are zero-copy (pinned host ctrl block)
     }
     int blocks = chunked ? 0 : (int)cdiv(n, WPB);
     if (blocks > 0) {
@@ -1411,13 +1399,14 @@ class Engine : public spanapi::ISpanExecutor {
   hipEvent_t ev_krn_[PIPE_MAX_CHUNKS] = {};
   Tables tables_{};
   DeviceBuf d_msgs_, d_fields_, d_enums_, d_enum_vals_, d_tools_, d_names_;
-  DeviceBuf d_in_, d_pb_, d_resp_, d_scratch_, d_final_;
-  DeviceBuf d_ctrl_, d_rid_, d_dec_results_;
-  BufView d_off3_, d_aux2_, d_results_, d_id_slots_;
-  PinnedBuf h_in_, h_pb_, h_resp_, h_final_, h_dec_results_, h_ctrl_, h_rid_;
-  BufView h_off_, h_aux_, h_results_, h_id_;
+  DeviceBuf d_in_, d_resp_, d_scratch_, d_final_;
+  DeviceBuf d_rid_, d_dec_results_;
+  BufView d_off3_, d_aux2_, d_results_, d_id_slots_, d_pb_;
+  PinnedBuf h_in_, h_resp_, h_final_, h_dec_results_, h_ctrl_, h_rid_;
+  BufView h_off_, h_aux_, h_results_, h_id_, h_pb_;
   size_t ctrl_aux_off_ = 0;  // aux2 offset inside the ctrl block
-  size_t rid_off_ = 0;       // ids offset inside the rid block (per batch)
+  size_t rid_off_ = 0;       // ids offset inside the rid|pb block (per batch)
+  size_t rid_cap_ = 0;       // pb offset inside the rid|pb block (fixed)
   std::vector<std::string> tool_paths_;
   std::vector<int32_t> tool_out_msg_;
   std::vector<int32_t> tool_backend_;
