@@ -223,6 +223,15 @@ class Frontend {
     d["decodeMs"] = ns_dec_us_.load() / 1e3;
     d["encodeGpuMs"] = ns_enc_gpu_us_.load() / 1e3;
     d["decodeGpuMs"] = ns_dec_gpu_us_.load() / 1e3;
+    if (rprof_on_) {
+      d["rprofBusyMs"] = rprof_.busy_ns.load() / 1e6;
+      d["rprofRecvMs"] = rprof_.recv_ns.load() / 1e6;
+      d["rprofParseMs"] = rprof_.parse_ns.load() / 1e6;
+      d["rprofGuardMs"] = rprof_.guard_ns.load() / 1e6;
+      d["rprofFlushMs"] = rprof_.flush_ns.load() / 1e6;
+      d["rprofCycles"] = rprof_.cycles.load();
+      d["rprofReqs"] = rprof_.reqs.load();
+    }
     return d;
   }
 
@@ -362,11 +371,24 @@ class Frontend {
 
   // ---- io reactors --------------------------------------------------------
 
+  // GGRMCP_REACTOR_PROF=1: per-stage wall aggregates over all reactor
+  // threads (exported via native_stats) — how busy the reactors are and
+  // where the per-request microseconds go.  ~2 clock reads per epoll
+  // cycle + 2 per request when on; zero reads when off.
+  struct RProf {
+    std::atomic<long> busy_ns{0}, recv_ns{0}, parse_ns{0}, guard_ns{0},
+        flush_ns{0}, cycles{0}, reqs{0};
+  };
+  RProf rprof_;
+  const bool rprof_on_ = getenv("GGRMCP_REACTOR_PROF") != nullptr;
+
   void io_loop(Reactor* re) {
     std::vector<epoll_event> events(256);
     while (!stop_.load()) {
       int timeout_ms = re->pending.empty() ? 50 : 1;
       int n = epoll_wait(re->epfd, events.data(), (int)events.size(), timeout_ms);
+      Clock::time_point b0;
+      if (rprof_on_) b0 = Clock::now();
       for (int i = 0; i < n; ++i) {
         uint64_t key = events[i].data.u64;
         if (key == LISTEN_KEY) {
@@ -381,6 +403,14 @@ class Frontend {
         }
       }
       drain_completions(re);
+      if (rprof_on_) {
+        rprof_.busy_ns.fetch_add(
+            (long)std::chrono::duration_cast<std::chrono::nanoseconds>(
+                Clock::now() - b0)
+                .count(),
+            std::memory_order_relaxed);
+        rprof_.cycles.fetch_add(1, std::memory_order_relaxed);
+      }
       // Backpressure-clocked batching: while all workers are busy,
       // arrivals accumulate; dispatch as soon as a worker is idle (or the
       // batch is full).  batch_window_us additionally caps how long a
@@ -457,6 +487,8 @@ class Frontend {
       return;
     }
     if (evmask & EPOLLIN) {
+      Clock::time_point t0;
+      if (rprof_on_) t0 = Clock::now();
       char buf[1 << 16];
       while (true) {
         ssize_t r = recv(c->fd, buf, sizeof(buf), 0);
@@ -472,7 +504,21 @@ class Frontend {
           return;
         }
       }
+      Clock::time_point t1;
+      if (rprof_on_) {
+        t1 = Clock::now();
+        rprof_.recv_ns.fetch_add(
+            (long)std::chrono::duration_cast<std::chrono::nanoseconds>(t1 - t0)
+                .count(),
+            std::memory_order_relaxed);
+      }
       parse_requests(re, id, c);
+      if (rprof_on_)
+        rprof_.parse_ns.fetch_add(
+            (long)std::chrono::duration_cast<std::chrono::nanoseconds>(
+                Clock::now() - t1)
+                .count(),
+            std::memory_order_relaxed);
       if (re->conns.find(id) == re->conns.end()) return;  // dropped in parse
     }
     if (evmask & EPOLLOUT) flush_conn(re, id, c);
@@ -564,12 +610,21 @@ class Frontend {
       req.method = method;
       req.path = path;
       req.batchable = (method == "POST" && path == "/" && is_json);
+      if (rprof_on_) rprof_.reqs.fetch_add(1, std::memory_order_relaxed);
       if (req.batchable && sess_table_) {
+        Clock::time_point g0;
+        if (rprof_on_) g0 = Clock::now();
         bool created = false;
         std::string sid;
         req.verdict = sess_table_->guard(req.session.data(), req.session.size(),
                                          sess_rate_limit_, &sid, &created);
         req.session = std::move(sid);
+        if (rprof_on_)
+          rprof_.guard_ns.fetch_add(
+              (long)std::chrono::duration_cast<std::chrono::nanoseconds>(
+                  Clock::now() - g0)
+                  .count(),
+              std::memory_order_relaxed);
       }
       if (!req.batchable && method == "POST" && path == "/") {
         complete(re, id, seq,
@@ -664,7 +719,15 @@ class Frontend {
       std::lock_guard<std::mutex> lk(re->done_mu);
       done.swap(re->done);
     }
+    Clock::time_point f0;
+    if (rprof_on_ && !done.empty()) f0 = Clock::now();
     for (auto& r : done) complete(re, r.conn_id, r.seq, std::move(r.payload));
+    if (rprof_on_ && !done.empty())
+      rprof_.flush_ns.fetch_add(
+          (long)std::chrono::duration_cast<std::chrono::nanoseconds>(
+              Clock::now() - f0)
+              .count(),
+          std::memory_order_relaxed);
     // decrement AFTER complete(): when drain() sees zero the responses are
     // already in connection write buffers (its grace period covers the
     // final socket flush)
