@@ -747,9 +747,18 @@ class Frontend {
         batch = std::move(batches_.front());
         batches_.pop_front();
         // coalesce batches queued (possibly from several reactors) while
-        // the workers were busy
+        // the workers were busy — capped below max_batch: an uncapped
+        // merge after a stall produces one multi-ms span whose requests
+        // all land in the p99 tail (GGRMCP_MERGE_MAX overrides)
+        static const size_t merge_env = [] {
+          const char* e = getenv("GGRMCP_MERGE_MAX");
+          size_t v = e ? (size_t)atoll(e) : 512;
+          return v < 1 ? (size_t)1 : v;
+        }();
+        const size_t merge_cap =
+            merge_env > (size_t)max_batch_ ? (size_t)max_batch_ : merge_env;
         while (!batches_.empty() &&
-               batch.size() + batches_.front().size() <= (size_t)max_batch_) {
+               batch.size() + batches_.front().size() <= merge_cap) {
           auto& nxt = batches_.front();
           batch.insert(batch.end(), std::make_move_iterator(nxt.begin()),
                        std::make_move_iterator(nxt.end()));
