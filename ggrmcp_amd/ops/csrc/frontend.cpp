@@ -747,12 +747,13 @@ class Frontend {
         batch = std::move(batches_.front());
         batches_.pop_front();
         // coalesce batches queued (possibly from several reactors) while
-        // the workers were busy — capped below max_batch: an uncapped
-        // merge after a stall produces one multi-ms span whose requests
-        // all land in the p99 tail (GGRMCP_MERGE_MAX overrides)
+        // the workers were busy.  GGRMCP_MERGE_MAX can cap the merge; the
+        // measured default is UNCAPPED (= max_batch): capping at 512 cost
+        // ~10% throughput for no p99 gain (gpurun_out/mc_*.json — big
+        // merges amortize the span fixed costs better than they straggle)
         static const size_t merge_env = [] {
           const char* e = getenv("GGRMCP_MERGE_MAX");
-          size_t v = e ? (size_t)atoll(e) : 512;
+          size_t v = e ? (size_t)atoll(e) : (size_t)1 << 30;
           return v < 1 ? (size_t)1 : v;
         }();
         const size_t merge_cap =
