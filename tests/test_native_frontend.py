@@ -620,3 +620,64 @@ def test_native_span_blocked_session():
         assert json.loads(data)["result"]["isError"] is False
     finally:
         gw.stop()
+
+
+def test_out_of_order_completion_reordered():
+    """Two pipelined requests on ONE connection land in different batches;
+    the second batch completes FIRST (slow first batch on 2 workers).  The
+    reactor must hold the late response in the reorder map and still write
+    responses in request order (frontend.cpp complete(): the in-order fast
+    path must not bypass ordering when an earlier seq is outstanding)."""
+    import socket
+    import time as _t
+
+    from ggrmcp_amd.server.native_http import load_module
+
+    mod = load_module()
+    started = threading.Event()
+
+    def batch_cb(bodies, session_ids, headers, verdicts=None):
+        out = []
+        for b in bodies:
+            data = json.loads(b)
+            if data.get("id") == 1:
+                started.set()
+                _t.sleep(0.5)  # batch with id 1 finishes AFTER id 2's
+            resp = {"jsonrpc": "2.0", "id": data.get("id"),
+                    "result": {"content": [], "isError": False}}
+            out.append((json.dumps(resp).encode(), ""))
+        return out
+
+    def slow_cb(method, path, body, headers):
+        return 404, b"{}"
+
+    fe = mod.Frontend("127.0.0.1", 0, batch_cb, slow_cb,
+                      batch_window_us=100, max_batch=4096,
+                      workers=2, reactors=1)
+    port = fe.start()
+    try:
+        body1 = json.dumps({"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                            "params": {"name": "t", "arguments": {}}}).encode()
+        body2 = json.dumps({"jsonrpc": "2.0", "id": 2, "method": "tools/call",
+                            "params": {"name": "t", "arguments": {}}}).encode()
+        req1 = (b"POST / HTTP/1.1\r\nHost: x\r\nContent-Type: application/json\r\n"
+                + b"Content-Length: %d\r\n\r\n" % len(body1) + body1)
+        req2 = (b"POST / HTTP/1.1\r\nHost: x\r\nContent-Type: application/json\r\n"
+                + b"Content-Length: %d\r\n\r\n" % len(body2) + body2)
+        s = socket.create_connection(("127.0.0.1", port), timeout=10)
+        s.sendall(req1)
+        assert started.wait(5)  # id 1's batch is dispatched and sleeping
+        s.sendall(req2)         # separate batch; completes first on worker 2
+        data = b""
+        deadline = _t.time() + 10
+        while data.count(b"HTTP/1.1 200") < 2 and _t.time() < deadline:
+            chunk = s.recv(65536)
+            if not chunk:
+                break
+            data += chunk
+        s.close()
+        assert data.count(b"HTTP/1.1 200") == 2, data[:300]
+        assert data.index(b'"id": 1') < data.index(b'"id": 2'), (
+            "responses must be written in request order")
+    finally:
+        fe.stop()
