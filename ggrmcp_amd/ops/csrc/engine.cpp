@@ -103,6 +103,12 @@ inline int wg_enc_phases() {
   const char* e = getenv("GGRMCP_WG_ENC_PHASES");
   return e ? atoi(e) : 3;
 }
+// multi-wave speculative structural scan in the wg encode phase A
+// (GGRMCP_MW_SCAN=1 to enable; default off until proven on-device)
+inline int mw_scan_on() {
+  const char* e = getenv("GGRMCP_MW_SCAN");
+  return e ? atoi(e) : 0;
+}
 
 // borrow (ptr, len) from a bytes / bytearray / memoryview element
 inline bool view_of(py::handle el, const char** ptr, size_t* len) {
@@ -1116,7 +1122,8 @@ class Engine : public spanapi::ISpanExecutor {
                          (const uint32_t*)d_off + (n + 1),
                          (SlotResult*)d_results_.p, (uint8_t*)d_id_slots_.p,
                          (uint8_t*)d_scratch_.p, tables_, lim, n,
-                         (const int32_t*)d_aux2_.p, wg_enc_phases());
+                         (const int32_t*)d_aux2_.p, wg_enc_phases(),
+                         mw_scan_on());
       HIP_CHECK(hipGetLastError());
     }
     // results + id tokens + pb wire in ONE packed D2H span (ids sit at
@@ -1173,7 +1180,7 @@ class Engine : public spanapi::ISpanExecutor {
             (SlotResult*)d_results_.p + lo,
             (uint8_t*)d_id_slots_.p + (size_t)lo * ID_SLOT_BYTES,
             (uint8_t*)d_scratch_.p, tables_, lim, m,
-            (const int32_t*)d_aux2_.p + lo, wg_enc_phases());
+            (const int32_t*)d_aux2_.p + lo, wg_enc_phases(), mw_scan_on());
         HIP_CHECK(hipGetLastError());
       }
       HIP_CHECK(hipEventRecord(ev_krn_[c], stream_));

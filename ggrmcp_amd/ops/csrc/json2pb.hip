@@ -2163,7 +2163,7 @@ extern "C" __global__ void __launch_bounds__(WG_ENC_WAVES * WAVE) k_json2pb_wg(
     uint8_t* __restrict__ pb_arena, const uint32_t* __restrict__ pb_off,
     SlotResult* __restrict__ results, uint8_t* __restrict__ id_slots,
     uint8_t* __restrict__ enc_scratch, Tables t, Limits lim, int n_req,
-    const int32_t* __restrict__ skip, int max_phase) {
+    const int32_t* __restrict__ skip, int max_phase, int use_mw) {
   // max_phase: debug bisection (GGRMCP_WG_ENC_PHASES; 3 = full)
   int req = blockIdx.x;
   if (req >= n_req) return;
@@ -2177,6 +2177,7 @@ extern "C" __global__ void __launch_bounds__(WG_ENC_WAVES * WAVE) k_json2pb_wg(
   __shared__ int32_t s_fidx[WG_ENC_MAX_ITEMS];
   __shared__ uint8_t s_kind[WG_ENC_MAX_ITEMS];
   __shared__ int s_nitems, s_next, s_mode, s_msg;
+  __shared__ uint32_t s_args0, s_args1;
   __shared__ SlotResult s_res;
   __shared__ uint8_t keybufs[WG_ENC_WAVES][192];
 
@@ -2239,7 +2240,11 @@ extern "C" __global__ void __launch_bounds__(WG_ENC_WAVES * WAVE) k_json2pb_wg(
       uint32_t acc = 0;
       bool fallback = m.wkt_kind != WKT_NONE;
       mode_l = 1;
-      if (!fallback && expect(c, '{')) {
+      // phase A' eligibility: big arguments objects index in parallel
+      // across all waves (below); small ones keep the serial walk here
+      if (!fallback && use_mw && args[1] - args[0] >= 8192) {
+        mode_l = 2;
+      } else if (!fallback && expect(c, '{')) {
         skip_ws(c);
         if (peek(c) == '}') {
           c.pos++;  // empty arguments object: zero items
@@ -2392,9 +2397,516 @@ extern "C" __global__ void __launch_bounds__(WG_ENC_WAVES * WAVE) k_json2pb_wg(
       s_nitems = n;
       s_next = 0;
       s_msg = msg_idx;
+      s_args0 = args[0];
+      s_args1 = args[1];
     }
   }
   __syncthreads();
+
+  // ---- phase A' (s_mode == 2): multi-wave speculative structural scan ----
+  // The serial member walk above streams ~64 KB through ONE wave (~12
+  // us/KB post-SWAR) while 7 waves idle.  Here every wave summarizes
+  // 256 B-granular windows of the arguments object under the 3 possible
+  // entry states (out-of-string / in-string / in-string-with-pending-
+  // escape: for VALID JSON backslashes only occur inside strings, so
+  // exactly one global state assignment is consistent — anything
+  // contradictory falls back to classic), one thread merges the
+  // summaries into real per-window states/depths, a sparse second pass
+  // collects depth-1/2 commas in document order, and members become
+  // items IN PARALLEL (one wave per member).  Spans are constructed to
+  // match the serial scanner byte-for-byte (trailing-ws-stripped plain
+  // values; chunk edges at the same stride commas), so phases B-D and
+  // the output wire are identical.  ANY anomaly -> s_mode 0 (classic
+  // in-block path, exact classic error formats).  GGRMCP_MW_SCAN gates.
+  if (s_mode == 2) {
+    constexpr int MW_WIN = 256;
+    constexpr int MW_D2 = 512;
+    __shared__ int16_t mwdd[MW_WIN][3], mwdm[MW_WIN][3];
+    __shared__ uint8_t mwend[MW_WIN][3], mwbad[MW_WIN][3];
+    __shared__ uint8_t mwes[MW_WIN];
+    __shared__ int16_t mwed[MW_WIN];
+    __shared__ uint16_t mwc1[MW_WIN], mwc2[MW_WIN];
+    __shared__ uint32_t mwp1[WG_ENC_MAX_ITEMS], mwp2[MW_D2];
+    __shared__ uint32_t s_ob, s_close, s_winb;
+    __shared__ int s_nwin, s_cwin, s_fb, s_n1, s_n2;
+    __shared__ unsigned long long s_seenf;
+    __shared__ uint32_t s_seeno;
+    __shared__ uint16_t s_mcnt[WG_ENC_MAX_ITEMS];
+    __shared__ uint32_t s_mcap[WG_ENC_MAX_ITEMS];
+    __shared__ uint32_t s_mbase[WG_ENC_MAX_ITEMS];
+    __shared__ uint32_t s_moff[WG_ENC_MAX_ITEMS];
+
+#define MW_IS_WS(ch) ((ch) == ' ' || (ch) == '\t' || (ch) == '\n' || (ch) == '\r')
+
+    // step 0: object opener, window geometry
+    if (threadIdx.x == 0) {
+      uint32_t p = s_args0;
+      while (p < s_args1 && MW_IS_WS(src[p])) ++p;
+      if (p >= s_args1 || src[p] != '{') {
+        s_fb = 1;
+      } else {
+        s_ob = p;
+        uint32_t span = s_args1 - (p + 1);
+        uint32_t wb = (span + MW_WIN - 1) / MW_WIN;
+        wb = (wb + 255u) & ~255u;
+        if (wb == 0) wb = 256;
+        s_winb = wb;
+        s_nwin = (int)((span + wb - 1) / wb);
+        s_fb = 0;
+        s_cwin = -1;
+        s_close = 0;
+        s_n1 = 0;
+        s_n2 = 0;
+        s_seenf = 0;
+        s_seeno = 0;
+      }
+    }
+    __syncthreads();
+
+    // pass 1: per-window summaries under all 3 entry states
+    if (!s_fb) {
+      const uint32_t base = s_ob + 1, aend = s_args1, wb = s_winb;
+      for (int w = wave; w < s_nwin; w += WG_ENC_WAVES) {
+        uint32_t wst = base + (uint32_t)w * wb;
+        uint32_t wen = wst + wb;
+        if (wen > aend) wen = aend;
+        int st[3] = {0, 1, 1};
+        uint32_t sk[3] = {0xFFFFFFFFu, 0xFFFFFFFFu, wst};
+        int dd[3] = {0, 0, 0}, dm[3] = {0, 0, 0}, bad[3] = {0, 0, 0};
+        for (uint32_t p = wst; p < wen; p += 4u * WAVE) {
+          uint32_t off = p + 4u * (uint32_t)lane;
+          uint32_t wd = load4_or(src, off, wen, 0);
+          uint32_t hit = swar_eq(wd, '"') | swar_eq(wd, '\\') |
+                         swar_eq(wd, '{') | swar_eq(wd, '}') |
+                         swar_eq(wd, '[') | swar_eq(wd, ']') |
+                         swar_eq(wd, ',');
+          uint64_t lmask = __ballot(hit != 0);
+          while (lmask) {
+            int lf = __ffsll((long long)lmask) - 1;
+            lmask &= lmask - 1;
+            uint32_t lh = (uint32_t)__shfl(hit, lf, WAVE);
+            uint32_t lw = (uint32_t)__shfl(wd, lf, WAVE);
+            while (lh) {
+              uint32_t bidx = (uint32_t)(__builtin_ctz(lh) >> 3);
+              lh &= lh - 1;
+              uint32_t pos = p + 4u * (uint32_t)lf + bidx;
+              if (pos >= wen) {
+                lh = 0;
+                break;
+              }
+              uint8_t ch = (uint8_t)(lw >> (8 * bidx));
+#pragma unroll
+              for (int mi = 0; mi < 3; ++mi) {
+                if (pos == sk[mi]) continue;
+                if (st[mi]) {
+                  if (ch == '\\')
+                    sk[mi] = pos + 1;
+                  else if (ch == '"')
+                    st[mi] = 0;
+                } else if (ch == '"') {
+                  st[mi] = 1;
+                } else if (ch == '{' || ch == '[') {
+                  if (++dd[mi] > 30000) bad[mi] = 1;
+                } else if (ch == '}' || ch == ']') {
+                  if (--dd[mi] < dm[mi]) dm[mi] = dd[mi];
+                  if (dd[mi] < -30000) bad[mi] = 1;
+                } else if (ch == '\\') {
+                  bad[mi] = 1;  // backslash outside any string
+                }
+              }
+            }
+          }
+        }
+        if (!lane) {
+#pragma unroll
+          for (int mi = 0; mi < 3; ++mi) {
+            mwdd[w][mi] = (int16_t)dd[mi];
+            mwdm[w][mi] = (int16_t)dm[mi];
+            mwend[w][mi] = st[mi] ? (sk[mi] == wen ? 2 : 1) : 0;
+            mwbad[w][mi] = (uint8_t)bad[mi];
+          }
+        }
+      }
+    }
+    __syncthreads();
+
+    // merge: resolve entry state + absolute depth per window, find the
+    // window holding the object closer
+    if (!s_fb && threadIdx.x == 0) {
+      int stt = 0, dep = 1, cw = -1;
+      for (int w = 0; w < s_nwin; ++w) {
+        mwes[w] = (uint8_t)stt;
+        mwed[w] = (int16_t)dep;
+        if (mwbad[w][stt]) {
+          s_fb = 2;
+          break;
+        }
+        if (dep + (int)mwdm[w][stt] <= 0) {
+          cw = w;
+          break;
+        }
+        dep += (int)mwdd[w][stt];
+        stt = (int)mwend[w][stt];
+        if (dep > 30000) {
+          s_fb = 2;
+          break;
+        }
+      }
+      if (!s_fb) {
+        if (cw < 0)
+          s_fb = 3;  // object never closes inside the span
+        else
+          s_cwin = cw;
+      }
+    }
+    __syncthreads();
+
+    // pass 2a: count depth-1 / depth-2 commas per qualifying window
+    if (!s_fb) {
+      const uint32_t base = s_ob + 1, wb = s_winb;
+      for (int w = wave; w <= s_cwin; w += WG_ENC_WAVES) {
+        int c1 = 0, c2 = 0;
+        if ((int)mwed[w] + (int)mwdm[w][mwes[w]] <= 2) {
+          uint32_t wst = base + (uint32_t)w * wb;
+          uint32_t wen = wst + wb;
+          if (wen > s_args1) wen = s_args1;
+          int stt = (int)mwes[w], dep = (int)mwed[w];
+          uint32_t sk = (stt == 2) ? wst : 0xFFFFFFFFu;
+          if (stt == 2) stt = 1;
+          for (uint32_t p = wst; p < wen && dep > 0; p += 4u * WAVE) {
+            uint32_t off = p + 4u * (uint32_t)lane;
+            uint32_t wd = load4_or(src, off, wen, 0);
+            uint32_t hit = swar_eq(wd, '"') | swar_eq(wd, '\\') |
+                           swar_eq(wd, '{') | swar_eq(wd, '}') |
+                           swar_eq(wd, '[') | swar_eq(wd, ']') |
+                           swar_eq(wd, ',');
+            uint64_t lmask = __ballot(hit != 0);
+            while (lmask && dep > 0) {
+              int lf = __ffsll((long long)lmask) - 1;
+              lmask &= lmask - 1;
+              uint32_t lh = (uint32_t)__shfl(hit, lf, WAVE);
+              uint32_t lw = (uint32_t)__shfl(wd, lf, WAVE);
+              while (lh && dep > 0) {
+                uint32_t bidx = (uint32_t)(__builtin_ctz(lh) >> 3);
+                lh &= lh - 1;
+                uint32_t pos = p + 4u * (uint32_t)lf + bidx;
+                if (pos >= wen) {
+                  lh = 0;
+                  break;
+                }
+                uint8_t ch = (uint8_t)(lw >> (8 * bidx));
+                if (pos == sk) continue;
+                if (stt) {
+                  if (ch == '\\')
+                    sk = pos + 1;
+                  else if (ch == '"')
+                    stt = 0;
+                } else if (ch == '"') {
+                  stt = 1;
+                } else if (ch == '{' || ch == '[') {
+                  ++dep;
+                } else if (ch == '}' || ch == ']') {
+                  if (--dep == 0) {
+                    if (ch != '}') {
+                      atomicExch(&s_fb, 4);
+                    } else if (!lane) {
+                      s_close = pos;
+                    }
+                  }
+                } else if (ch == ',') {
+                  if (dep == 1)
+                    ++c1;
+                  else if (dep == 2)
+                    ++c2;
+                }
+              }
+            }
+          }
+        }
+        if (!lane) {
+          mwc1[w] = (uint16_t)c1;
+          mwc2[w] = (uint16_t)c2;
+        }
+      }
+    }
+    __syncthreads();
+
+    // pass 2b: exclusive scans -> per-window output bases; cap checks
+    if (!s_fb && threadIdx.x == 0) {
+      int a1 = 0, a2 = 0;
+      for (int w = 0; w <= s_cwin; ++w) {
+        int c1 = mwc1[w], c2 = mwc2[w];
+        mwc1[w] = (uint16_t)a1;
+        mwc2[w] = (uint16_t)a2;
+        a1 += c1;
+        a2 += c2;
+      }
+      s_n1 = a1;
+      s_n2 = a2;
+      if (a1 + 1 > WG_ENC_MAX_ITEMS || a2 > MW_D2) s_fb = 5;
+    }
+    __syncthreads();
+
+    // pass 2c: place comma positions (window bases keep document order)
+    if (!s_fb) {
+      const uint32_t base = s_ob + 1, wb = s_winb;
+      for (int w = wave; w <= s_cwin; w += WG_ENC_WAVES) {
+        if ((int)mwed[w] + (int)mwdm[w][mwes[w]] > 2) continue;
+        uint32_t wst = base + (uint32_t)w * wb;
+        uint32_t wen = wst + wb;
+        if (wen > s_args1) wen = s_args1;
+        int stt = (int)mwes[w], dep = (int)mwed[w];
+        uint32_t sk = (stt == 2) ? wst : 0xFFFFFFFFu;
+        if (stt == 2) stt = 1;
+        int o1 = mwc1[w], o2 = mwc2[w];
+        for (uint32_t p = wst; p < wen && dep > 0; p += 4u * WAVE) {
+          uint32_t off = p + 4u * (uint32_t)lane;
+          uint32_t wd = load4_or(src, off, wen, 0);
+          uint32_t hit = swar_eq(wd, '"') | swar_eq(wd, '\\') |
+                         swar_eq(wd, '{') | swar_eq(wd, '}') |
+                         swar_eq(wd, '[') | swar_eq(wd, ']') |
+                         swar_eq(wd, ',');
+          uint64_t lmask = __ballot(hit != 0);
+          while (lmask && dep > 0) {
+            int lf = __ffsll((long long)lmask) - 1;
+            lmask &= lmask - 1;
+            uint32_t lh = (uint32_t)__shfl(hit, lf, WAVE);
+            uint32_t lw = (uint32_t)__shfl(wd, lf, WAVE);
+            while (lh && dep > 0) {
+              uint32_t bidx = (uint32_t)(__builtin_ctz(lh) >> 3);
+              lh &= lh - 1;
+              uint32_t pos = p + 4u * (uint32_t)lf + bidx;
+              if (pos >= wen) {
+                lh = 0;
+                break;
+              }
+              uint8_t ch = (uint8_t)(lw >> (8 * bidx));
+              if (pos == sk) continue;
+              if (stt) {
+                if (ch == '\\')
+                  sk = pos + 1;
+                else if (ch == '"')
+                  stt = 0;
+              } else if (ch == '"') {
+                stt = 1;
+              } else if (ch == '{' || ch == '[') {
+                ++dep;
+              } else if (ch == '}' || ch == ']') {
+                --dep;
+              } else if (ch == ',') {
+                if (dep == 1) {
+                  if (!lane) mwp1[o1] = pos;
+                  ++o1;
+                } else if (dep == 2) {
+                  if (!lane) mwp2[o2] = pos;
+                  ++o2;
+                }
+              }
+            }
+          }
+        }
+      }
+    }
+    __syncthreads();
+
+    // members: M1 counts items + caps per member (validations + dup
+    // detection), a serial scan assigns bases, M2 places the items.
+    // Control flow between barriers uses a UNIFORM snapshot (s_go,
+    // written by thread 0 before each barrier): raw s_fb reads race with
+    // the atomic failure sets and could diverge the barrier counts.
+    __shared__ int s_go;
+    auto member_pass = [&](int place, int nmemb_l) {
+      for (int m_i = wave; m_i < nmemb_l; m_i += WG_ENC_WAVES) {
+        uint32_t mstart = (m_i == 0) ? s_ob + 1 : mwp1[m_i - 1] + 1;
+        uint32_t mend = (m_i == nmemb_l - 1) ? s_close : mwp1[m_i];
+        uint32_t kq = mstart;
+        while (kq < mend && MW_IS_WS(src[kq])) ++kq;
+        if (kq >= mend || src[kq] != '"') {
+          atomicExch(&s_fb, 6);
+          continue;
+        }
+        Ctx c;
+        c.s = src;
+        c.len = mend;
+        c.pos = kq;
+        c.out = nullptr;
+        c.opos = 0;
+        c.ocap = 0;
+        c.t = t;
+        c.lim = lim;
+        c.status = E_OK;
+        c.err_pos = 0;
+        c.aux = 0;
+        c.lane = lane;
+        c.keybuf = keybufs[wave];
+        uint32_t kst, krl;
+        bool kesc;
+        if (!string_span(c, &kst, &krl, &kesc) || kesc) {
+          atomicExch(&s_fb, 6);
+          continue;
+        }
+        uint64_t h = fnv1a64(c.s + kst, krl);
+        const MsgEntry& mm = t.msgs[s_msg];
+        int fidx = -1;
+        for (int i = 0; i < mm.field_count; ++i) {
+          const FieldEntry& fe = t.fields[mm.field_start + i];
+          if ((fe.hash_json == h && fe.json_len == krl &&
+               wave_equal(c, t.names + fe.json_off, c.s + kst, krl)) ||
+              (fe.hash_orig == h && fe.name_len == krl &&
+               wave_equal(c, t.names + fe.name_off, c.s + kst, krl))) {
+            fidx = i;
+            break;
+          }
+        }
+        if (fidx < 0) {
+          atomicExch(&s_fb, 6);
+          continue;
+        }
+        const FieldEntry& fe = t.fields[mm.field_start + fidx];
+        if (!place && !lane) {
+          if (fidx < 64) {
+            unsigned long long bit = 1ull << fidx;
+            if (atomicOr(&s_seenf, bit) & bit) atomicExch(&s_fb, 6);
+          }
+          if (fe.flags & F_ONEOF) {
+            uint32_t b = 1u << fe.oneof_id;
+            if (atomicOr(&s_seeno, b) & b) atomicExch(&s_fb, 6);
+          }
+        }
+        skip_ws(c);
+        if (c.pos >= mend || src[c.pos] != ':') {
+          atomicExch(&s_fb, 6);
+          continue;
+        }
+        c.pos++;
+        skip_ws(c);
+        uint32_t vstart = c.pos;
+        if (vstart >= mend) {
+          atomicExch(&s_fb, 6);
+          continue;
+        }
+        uint32_t vend = mend;
+        while (vend > vstart && MW_IS_WS(src[vend - 1])) --vend;
+        uint8_t vc = src[vstart];
+        bool chunk_map = (fe.flags & F_MAP) && vc == '{';
+        bool chunk_arr = (fe.flags & F_REPEATED) && !(fe.flags & F_MAP) &&
+                         !is_packable(fe.kind) && vc == '[';
+        bool big = (s_args1 - vstart) >= 2048;
+        int cnt = 0;
+        uint32_t cap_sum = 0;
+        uint32_t ibase = place ? s_mbase[m_i] : 0;
+        uint32_t ioff = place ? s_moff[m_i] : 0;
+        if ((chunk_map || chunk_arr) && big) {
+          uint8_t closer = chunk_map ? '}' : ']';
+          if (vend <= vstart + 1 || src[vend - 1] != closer) {
+            atomicExch(&s_fb, 6);
+            continue;
+          }
+          uint32_t close_pos = vend - 1;
+          uint32_t cstart = vstart + 1;
+          while (cstart < close_pos && MW_IS_WS(src[cstart])) ++cstart;
+          if (cstart < close_pos) {
+            // stride marks over the collected depth-2 commas (same
+            // cadence as scan_container(marks, cap=64, stride=4096))
+            int lo2 = 0, hi2 = s_n2;
+            while (lo2 < hi2) {
+              int mid = (lo2 + hi2) >> 1;
+              if (mwp2[mid] <= cstart)
+                lo2 = mid + 1;
+              else
+                hi2 = mid;
+            }
+            uint32_t prev_cs = cstart, nmark = cstart + 4096;
+            int nmk = 0;
+            for (int q = lo2; q < s_n2 && nmk < 64; ++q) {
+              uint32_t pos = mwp2[q];
+              if (pos >= close_pos) break;
+              if (pos >= nmark) {
+                if (prev_cs < pos) {
+                  uint32_t span = pos - prev_cs;
+                  uint32_t icap = span + span / 4 + WG_ENC_ITEM_PAD;
+                  if (place && !lane) {
+                    s_start[ibase + cnt] = prev_cs;
+                    s_end[ibase + cnt] = pos;
+                    s_ioff[ibase + cnt] = ioff + cap_sum;
+                    s_fidx[ibase + cnt] = mm.field_start + fidx;
+                    s_kind[ibase + cnt] = chunk_map ? 2 : 3;
+                  }
+                  ++cnt;
+                  cap_sum += icap;
+                }
+                prev_cs = pos + 1;
+                nmark = pos + 4096;
+                ++nmk;
+              }
+            }
+            if (prev_cs < close_pos) {
+              uint32_t span = close_pos - prev_cs;
+              uint32_t icap = span + span / 4 + WG_ENC_ITEM_PAD;
+              if (place && !lane) {
+                s_start[ibase + cnt] = prev_cs;
+                s_end[ibase + cnt] = close_pos;
+                s_ioff[ibase + cnt] = ioff + cap_sum;
+                s_fidx[ibase + cnt] = mm.field_start + fidx;
+                s_kind[ibase + cnt] = chunk_map ? 2 : 3;
+              }
+              ++cnt;
+              cap_sum += icap;
+            }
+          }
+        } else {
+          uint32_t span = vend - (kst - 1);
+          uint32_t icap = span + span / 4 + WG_ENC_ITEM_PAD;
+          if (place && !lane) {
+            s_start[ibase] = kst - 1;
+            s_end[ibase] = vend;
+            s_ioff[ibase] = ioff;
+            s_fidx[ibase] = -1;
+            s_kind[ibase] = 1;
+          }
+          cnt = 1;
+          cap_sum = icap;
+        }
+        if (!place && !lane) {
+          s_mcnt[m_i] = (uint16_t)cnt;
+          s_mcap[m_i] = cap_sum;
+        }
+      }
+    };
+
+    if (threadIdx.x == 0) s_go = s_fb;
+    __syncthreads();
+    int nmemb = 0;
+    if (!s_go) {
+      uint32_t p0 = s_ob + 1;
+      while (p0 < s_close && MW_IS_WS(src[p0])) ++p0;
+      nmemb = (p0 >= s_close) ? 0 : s_n1 + 1;
+      member_pass(0, nmemb);
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      if (!s_fb) {
+        uint32_t nb = 0, ob2 = 0;
+        for (int m_i = 0; m_i < nmemb; ++m_i) {
+          s_mbase[m_i] = nb;
+          s_moff[m_i] = ob2;
+          nb += s_mcnt[m_i];
+          ob2 += s_mcap[m_i];
+        }
+        if (nb > (uint32_t)WG_ENC_MAX_ITEMS || ob2 > scr_cap)
+          s_fb = 7;
+        else
+          s_nitems = (int)nb;
+      }
+      s_go = s_fb;
+    }
+    __syncthreads();
+    if (!s_go) member_pass(1, nmemb);
+    __syncthreads();
+    if (threadIdx.x == 0) s_mode = s_fb ? 0 : 1;
+    __syncthreads();
+#undef MW_IS_WS
+  }
   const int n_items = s_nitems;
   const int msg_idx = s_msg;
 
