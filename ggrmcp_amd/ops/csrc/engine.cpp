@@ -286,8 +286,12 @@ class StagePool {
 };
 
 inline int stage_threads() {
+  // default 0 (off): measured NEUTRAL on wide64 (51.9k with 3 helpers vs
+  // 53.0k without, same box) — the engines already stage in parallel
+  // across instances and the pool wakeups eat the ~0.2-0.4 ms the split
+  // saves.  Kept env-gated for single-engine deployments.
   const char* e = getenv("GGRMCP_STAGE_THREADS");
-  int v = e ? atoi(e) : 3;
+  int v = e ? atoi(e) : 0;
   return v < 0 ? 0 : (v > 15 ? 15 : v);
 }
 inline size_t stage_min_bytes() {
