@@ -121,3 +121,49 @@ def test_mw_scan_end_to_end_oracle(env):  # noqa: F811
         wire = pipeline.cpu.json_to_pb(mi.input_descriptor, json.dumps(args))
         oracle = json.loads(pipeline.cpu.pb_to_json(mi.output_descriptor, wire))
         assert inner == oracle, f"slot {i}: {_first_diff(inner, oracle)}"
+
+
+def test_mw_scan_adversarial_strings(env):  # noqa: F811
+    """The speculative scanner's hard cases: structural characters INSIDE
+    strings, backslash runs (escape state crossing 256 B window
+    boundaries), and tokens straddling window edges.  Wire must stay
+    byte-identical to the serial scanner."""
+    rng = random.Random(999)
+    bodies = []
+    rid = 1
+
+    def mk(args):
+        nonlocal rid
+        bodies.append(_body(args, rid))
+        rid += 1
+
+    # strings stuffed with JSON structural chars (validation caps strings
+    # at 1024 chars, so spread across many map values)
+    evil = '{"a":[1,2],\\\\}' + "{[,]}" * 60
+    mk({"attrs": {f"k{j}": (evil * 3)[:900] for j in range(24)}})
+    # long backslash runs with varying parity near 256-byte boundaries
+    for parity in (1, 2, 3):
+        v = ("x" * 251 + "\\" * parity + '"inner"').replace('"inner"', "")
+        # json.dumps escapes backslashes: each '\\' doubles on the wire,
+        # shifting window alignment per value
+        mk({"attrs": {f"b{j}": v + "y" * (j % 7) for j in range(20)}})
+    # quotes adjacent to escapes: \" \\" \\\" patterns
+    q = 'a\\"b' + "\\\\" + '\\"' + "c"
+    mk({"attrs": {f"q{j}": (q * 40)[:800] for j in range(20)}})
+    # member keys landing near window boundaries: tune value sizes so
+    # keys fall at 256k +- 2 offsets
+    for shift in range(-2, 3):
+        mk({"attrs": {f"w{j:03d}": "z" * (249 + shift) for j in range(40)}})
+    # deep nesting inside a big payload (within limits)
+    nest = {"name": "leaf"}
+    for _ in range(8):
+        nest = {"child": nest, "name": "n" * 50}
+    mk({"f40Node": nest,
+        "attrs": {f"n{j}": "m" * 700 for j in range(16)}})
+    # unicode multibyte spanning boundaries
+    mk({"attrs": {f"u{j}": ("é中\U0001f600" * 80)[:300] for j in range(30)}})
+
+    pipeline, d = env
+    serial = _enc(pipeline, bodies, mw=False, wg_min="2048")
+    mw = _enc(pipeline, bodies, mw=True, wg_min="2048")
+    _assert_equal(bodies, serial, mw)
