@@ -372,7 +372,7 @@ constexpr int WG_DEC_MAX_ITEMS = 768;
 constexpr uint32_t WG_DEC_ITEM_PAD = 256;
 constexpr uint32_t WG_DEC_EXTRA =
     WG_DEC_ITEM_PAD * (uint32_t)WG_DEC_MAX_ITEMS + 1024;
-constexpr int WG_DEC_WAVES = 8;  // waves per request workgroup
+constexpr int WG_DEC_WAVES = 16;  // waves per request workgroup
 
 // Workgroup-cooperative ENCODE (k_json2pb_wg): JSON-RPC requests at least
 // WG_ENC_MIN_BYTES long split their arguments object into top-level-member
@@ -384,7 +384,7 @@ constexpr int WG_DEC_WAVES = 8;  // waves per request workgroup
 constexpr uint32_t WG_ENC_MIN_BYTES = 16384;
 constexpr int WG_ENC_MAX_ITEMS = 384;
 constexpr uint32_t WG_ENC_ITEM_PAD = 160;
-constexpr int WG_ENC_WAVES = 8;
+constexpr int WG_ENC_WAVES = 16;
 
 // u64 -> decimal text without an addressable temp buffer (a local tmp[20]
 // array lands in scratch memory and costs a private-memory round trip per
