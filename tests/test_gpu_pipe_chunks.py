@@ -123,3 +123,21 @@ def test_chunked_end_to_end_errors_and_default_off(env):  # noqa: F811
     plain = pipeline.process_batch(bodies, timeout_s=30.0)
     for i, (a, b) in enumerate(zip(chunked, plain)):
         assert json.loads(a) == json.loads(b), i
+
+
+def test_staging_pool_equivalence(env):  # noqa: F811
+    """GGRMCP_STAGE_THREADS fans the span staging memcpys across a helper
+    pool (default off — measured neutral).  Forced on, the full pipeline
+    must produce identical responses."""
+    pipeline, d = env
+    bodies = _mixed_bodies(n=24, seed=77)
+    plain = pipeline.process_batch(bodies, timeout_s=30.0)
+    os.environ["GGRMCP_STAGE_THREADS"] = "3"
+    os.environ["GGRMCP_STAGE_MIN"] = "1"
+    try:
+        pooled = pipeline.process_batch(bodies, timeout_s=30.0)
+    finally:
+        os.environ.pop("GGRMCP_STAGE_THREADS", None)
+        os.environ.pop("GGRMCP_STAGE_MIN", None)
+    for i, (a, b) in enumerate(zip(plain, pooled)):
+        assert json.loads(a) == json.loads(b), i
