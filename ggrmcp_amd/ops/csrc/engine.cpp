@@ -662,21 +662,27 @@ class Engine : public spanapi::ISpanExecutor {
     uint32_t* pb_off = in_off + (n + 1);
     size_t acc = 0, pacc = 0;
     uint8_t* dst = (uint8_t*)h_in_.p;
+    std::vector<const char*> bptr(n);
     for (int i = 0; i < n; ++i) {
-      const char* ptr;
       size_t len;
-      if (!view_of(bodies[i], &ptr, &len))
+      if (!view_of(bodies[i], &bptr[i], &len))
         throw std::runtime_error("process_span: unsupported element type");
       if (acc + len > h_in_.n) throw std::runtime_error("input exceeds cap_in");
       in_off[i] = (uint32_t)acc;
       pb_off[i] = (uint32_t)pacc;
-      if (len) std::memcpy(dst + acc, ptr, len);
       acc += len;
       pacc += pb_cap(len);
     }
     in_off[n] = (uint32_t)acc;
     pb_off[n] = (uint32_t)pacc;
     if (pacc > d_pb_.n) throw std::runtime_error("pb cap exceeded");
+    // copies fan out across the staging pool for big batches (the byte
+    // buffers stay alive via the borrowed sequence refs; the caller
+    // keeps the GIL while the helpers copy)
+    stage_copies(in_off, n, acc, [&](int i) {
+      size_t len = in_off[i + 1] - in_off[i];
+      if (len) std::memcpy(dst + in_off[i], bptr[i], len);
+    });
     // per-slot metadata (headers) staged before dropping the GIL
     std::vector<std::vector<std::pair<std::string, std::string>>> metas;
     bool have_headers = !headers.is_none();
