@@ -106,7 +106,7 @@ inline int wg_enc_phases() {
 // multi-wave speculative structural scan in the wg encode phase A
 // (GGRMCP_MW_SCAN=0 disables).  Default ON: byte-identical across the
 // differential suites and +9% wide64 end-to-end (encode span -25%,
-// gpurun_out/mw_w64_on.json); spans < 8 KB keep the serial walk.
+// profiles/mw_w64_d2.json); spans < 8 KB keep the serial walk.
 inline int mw_scan_on() {
   const char* e = getenv("GGRMCP_MW_SCAN");
   return e ? atoi(e) : 1;

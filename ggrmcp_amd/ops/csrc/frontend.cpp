@@ -749,7 +749,7 @@ class Frontend {
         // coalesce batches queued (possibly from several reactors) while
         // the workers were busy.  GGRMCP_MERGE_MAX can cap the merge; the
         // measured default is UNCAPPED (= max_batch): capping at 512 cost
-        // ~10% throughput for no p99 gain (gpurun_out/mc_*.json — big
+        // ~10% throughput for no p99 gain (profiles/mc_*.json — big
         // merges amortize the span fixed costs better than they straggle)
         static const size_t merge_env = [] {
           const char* e = getenv("GGRMCP_MERGE_MAX");
