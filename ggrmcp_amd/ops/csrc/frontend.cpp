@@ -113,6 +113,16 @@ std::string http_response(int status, const std::string& body,
   return out;
 }
 
+// HTTP field values that flow into Python str / h2 metadata must be
+// visible ASCII: a high/control byte would throw inside the worker's
+// py::str conversion (process-fatal outside a catch) and is invalid
+// gRPC metadata anyway.
+bool printable_ascii(const std::string& v) {
+  for (unsigned char c : v)
+    if (c < 0x20 || c > 0x7E) return false;
+  return true;
+}
+
 bool iequal(const char* a, const char* b, size_t n) {
   for (size_t i = 0; i < n; ++i)
     if (tolower((unsigned char)a[i]) != tolower((unsigned char)b[i])) return false;
@@ -594,6 +604,17 @@ class Frontend {
       if (c->rbuf.size() < total) return;  // need more bytes
       std::string body = c->rbuf.substr(hdr_end + 4, clen);
       c->rbuf.erase(0, total);
+      // sanitize values leaving the wire-parse layer (see printable_ascii)
+      if (!session.empty() && !printable_ascii(session))
+        session.clear();  // treated as absent -> fresh session id
+      if (!hdrs.empty())
+        hdrs.erase(std::remove_if(hdrs.begin(), hdrs.end(),
+                                  [](const std::pair<std::string,
+                                                     std::string>& kv) {
+                                    return !printable_ascii(kv.first) ||
+                                           !printable_ascii(kv.second);
+                                  }),
+                   hdrs.end());
 
       uint64_t seq = c->next_seq++;
       if (!allow_rate()) {
@@ -908,18 +929,32 @@ class Frontend {
           py::list bodies, sessions, headers, verdicts;
           std::vector<size_t> batch_idx;
           bool guarded = sess_table_ != nullptr;
-          for (size_t i = 0; i < batch.size(); ++i) {
-            if (!batch[i].batchable) continue;
-            batch_idx.push_back(i);
-            bodies.append(py::bytes(batch[i].body));
-            sessions.append(batch[i].session.empty()
-                                ? py::object(py::none())
-                                : py::object(py::str(batch[i].session)));
-            if (guarded) verdicts.append(py::int_(batch[i].verdict));
-            py::dict h;
-            for (auto& kv : batch[i].headers)
-              h[py::str(kv.first)] = py::str(kv.second);
-            headers.append(h);
+          try {
+            for (size_t i = 0; i < batch.size(); ++i) {
+              if (!batch[i].batchable) continue;
+              batch_idx.push_back(i);
+              bodies.append(py::bytes(batch[i].body));
+              sessions.append(batch[i].session.empty()
+                                  ? py::object(py::none())
+                                  : py::object(py::str(batch[i].session)));
+              if (guarded) verdicts.append(py::int_(batch[i].verdict));
+              py::dict h;
+              for (auto& kv : batch[i].headers)
+                h[py::str(kv.first)] = py::str(kv.second);
+              headers.append(h);
+            }
+          } catch (const std::exception&) {
+            // conversion failure (should be impossible post-sanitize):
+            // answer the whole batch with -32603 rather than dying
+            std::string err =
+                "{\"jsonrpc\":\"2.0\",\"id\":null,\"error\":"
+                "{\"code\":-32603,\"message\":\"internal\"}}";
+            for (size_t k = 0; k < batch_idx.size(); ++k)
+              out.push_back({batch[batch_idx[k]].conn_id,
+                             batch[batch_idx[k]].seq,
+                             http_response(200, err, "")});
+            batch_idx.clear();
+            bodies = py::list();
           }
           if (py::len(bodies) > 0) {
             try {

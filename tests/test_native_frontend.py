@@ -736,3 +736,24 @@ def test_http_parser_garbage_fuzz(gateway):
         {"jsonrpc": "2.0", "id": 9, "method": "tools/call",
          "params": {"name": "t", "arguments": {"ok": True}}}))
     assert status == 200 and b'"isError": false' in data
+
+
+def test_weird_session_ids(gateway):
+    """Mcp-Session-Id edge cases: exactly at/over the 48-byte table key
+    cap, control/8-bit bytes, empty — never crash; over-cap or unknown
+    ids mint a fresh 32-hex id (manager.go GetOrCreateSession semantics:
+    a client-supplied UNKNOWN id becomes the session when it fits)."""
+    gw, port, pipe = gateway
+    body = json.dumps({"jsonrpc": "2.0", "id": 1, "method": "tools/call",
+                       "params": {"name": "t", "arguments": {}}})
+    for sid in ("x" * 47, "x" * 48, "x" * 49, "x" * 500,
+                "id with spaces", "\xfc\xf1\xe9-high-bytes", "a\tb"):
+        status, data, got = _call(port, body, session=sid)
+        assert status == 200, (sid, status)
+        assert got, sid
+        if len(sid.encode("latin-1")) <= 48:
+            # fits the table: echoed back and stable on reuse
+            status2, _, got2 = _call(port, body, session=got)
+            assert status2 == 200 and got2 == got
+        else:
+            assert len(got) == 32  # fresh crypto id
