@@ -106,7 +106,16 @@ std::string http_response(int status, const std::string& body,
   out += "Access-Control-Allow-Methods: GET, POST, OPTIONS\r\n";
   out += "Access-Control-Allow-Headers: Content-Type, Mcp-Session-Id, Authorization\r\n";
   out += "Access-Control-Expose-Headers: Mcp-Session-Id\r\n";
-  if (!session_id.empty()) out += "Mcp-Session-Id: " + session_id + "\r\n";
+  if (!session_id.empty()) {
+    // single sink for ids echoed into a header: strip anything that
+    // could split the response (defense in depth behind the parse-time
+    // sanitizer — Python callbacks also hand ids back through here)
+    std::string sid;
+    sid.reserve(session_id.size());
+    for (unsigned char ch : session_id)
+      if (ch >= 0x20 && ch != 0x7F) sid += (char)ch;
+    if (!sid.empty()) out += "Mcp-Session-Id: " + sid + "\r\n";
+  }
   if (close) out += "Connection: close\r\n";
   out += "Content-Length: " + std::to_string(body.size()) + "\r\n\r\n";
   out += body;
