@@ -36,7 +36,16 @@ class StubPipeline:
         self.batches.append(len(bodies))
         out = []
         for i, b in enumerate(bodies):
-            data = json.loads(b)
+            try:
+                data = json.loads(b)
+            except Exception:
+                # mirror the real pipeline: unparseable bodies become
+                # -32700 responses, never exceptions (the GPU engine
+                # reports E_PARSE per slot)
+                out.append(
+                    b'{"jsonrpc":"2.0","id":null,'
+                    b'"error":{"code":-32700,"message":"parse error"}}')
+                continue
             if data.get("method") != "tools/call" and self.non_toolcall_handler:
                 out.append(self.non_toolcall_handler(b, headers[i] if headers else None))
                 continue
