@@ -205,3 +205,36 @@ def test_reconnect_with_retry(backend):
     assert d.tools_version > v1  # rediscovered + atomic republish
     assert "hello_helloservice_sayhello" in d.tools
     d.close()
+
+
+def test_malformed_descriptor_blob_rejected(discoverer):
+    """A hostile/corrupt FileDescriptorSet blob must raise a clean error
+    (protobuf DecodeError surface), never publish a partial tool map."""
+    import pytest as _pytest
+
+    before = dict(discoverer.tools)
+    with _pytest.raises(Exception):
+        discoverer.load_descriptor_blob(b"\x00\xff garbage \x01\x02" * 20)
+    assert discoverer.tools == before  # atomic publish: map unchanged
+
+
+def test_corrupt_descriptor_set_file_falls_back(tmp_path, backend):
+    """descriptor_set.path pointing at corrupt bytes behaves like the
+    missing-file case: warn + reflection fallback (discovery.go:107-111)."""
+    from ggrmcp_amd.backend.discovery import ServiceDiscoverer
+    from ggrmcp_amd.config import Config
+
+    bad = tmp_path / "bad.binpb"
+    bad.write_bytes(b"\xde\xad\xbe\xef" * 64)
+    cfg = Config.default()
+    host, _, port = backend.rpartition(":")
+    cfg.grpc.host, cfg.grpc.port = host, int(port)
+    cfg.descriptor_set.enabled = True
+    cfg.descriptor_set.path = str(bad)
+    d = ServiceDiscoverer(cfg)
+    try:
+        d.connect(timeout_s=15)
+        d.discover()
+        assert "hello_helloservice_sayhello" in d.tools  # via reflection
+    finally:
+        d.close()
