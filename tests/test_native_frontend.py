@@ -766,3 +766,62 @@ def test_weird_session_ids(gateway):
             assert status2 == 200 and got2 == got
         else:
             assert len(got) == 32  # fresh crypto id
+
+
+def test_pipelined_ordering_under_random_delays():
+    """10 pipelined requests on one connection, batch handlers completing
+    in adversarial random order (2 workers, per-batch random sleeps):
+    responses must still arrive strictly in request order."""
+    import random
+    import socket
+    import time as _t
+
+    from ggrmcp_amd.server.native_http import load_module
+
+    mod = load_module()
+    rng = random.Random(7)
+
+    def batch_cb(bodies, session_ids, headers, verdicts=None):
+        _t.sleep(rng.random() * 0.12)
+        out = []
+        for b in bodies:
+            data = json.loads(b)
+            resp = {"jsonrpc": "2.0", "id": data.get("id"),
+                    "result": {"content": [], "isError": False}}
+            out.append((json.dumps(resp).encode(), ""))
+        return out
+
+    def slow_cb(method, path, body, headers):
+        return 404, b"{}"
+
+    fe = mod.Frontend("127.0.0.1", 0, batch_cb, slow_cb,
+                      batch_window_us=100, max_batch=2,  # force many batches
+                      workers=2, reactors=1)
+    port = fe.start()
+    try:
+        s = socket.create_connection(("127.0.0.1", port), timeout=15)
+        for rid in range(1, 11):
+            body = json.dumps(
+                {"jsonrpc": "2.0", "id": rid, "method": "tools/call",
+                 "params": {"name": "t", "arguments": {}}}).encode()
+            s.sendall(
+                b"POST / HTTP/1.1\r\nHost: x\r\nContent-Type: application/json\r\n"
+                + b"Content-Length: %d\r\n\r\n" % len(body) + body)
+            _t.sleep(0.02)  # spread arrivals across batches
+        data = b""
+        deadline = _t.time() + 20
+        while data.count(b"HTTP/1.1 200") < 10 and _t.time() < deadline:
+            chunk = s.recv(65536)
+            if not chunk:
+                break
+            data += chunk
+        s.close()
+        assert data.count(b"HTTP/1.1 200") == 10, data[:200]
+        last = -1
+        for rid in range(1, 11):
+            idx = data.find(b'"id": %d' % rid)
+            assert idx >= 0, rid
+            assert idx > last, f"response {rid} out of order"
+            last = idx
+    finally:
+        fe.stop()
